@@ -1,0 +1,627 @@
+/*
+ * oracle.c — CPU restatement of the OpenTenBase DataNode executor hot path.
+ * See oracle.h header comment: TEST INFRASTRUCTURE, never the product path.
+ *
+ * Structure mirrors the reference executor tuple-at-a-time loops:
+ *   - scan+qual:       ExecScan/ExecScanFetch loop, execScan.c:140-363
+ *   - hash join:       chained-bucket table; insert-at-head
+ *                      ExecHashTableInsert nodeHash.c:1828, bucket split
+ *                      ExecHashGetBucketAndBatch nodeHash.c:2141 (single
+ *                      batch: our build sides fit memory; grace batching is
+ *                      SURVEY §8f.4), probe walk ExecScanHashBucket
+ *                      nodeHash.c:2174, join FSM ExecHashJoinImpl
+ *                      nodeHashjoin.c:186 (inner join states only)
+ *   - hash aggregate:  simplehash-style open addressing, linear probe,
+ *                      fillfactor 0.9 (simplehash.h:187), lookup-or-insert
+ *                      LookupTupleHashEntry execGrouping.c:295,
+ *                      agg_fill_hash_table nodeAgg.c:2609
+ *   - transition fns:  int8inc int8.c:714 (overflow → error), float8pl
+ *                      float.c:970 (CHECKFLOATVAL float.c:58), float8_accum
+ *                      float.c:2823 (Youngs-Cramer [N,Sx,Sxx])
+ *   - NULL semantics:  strict transfns skip NULL input (nodeAgg.c:743 /
+ *                      EEOP_AGG_STRICT_TRANS checks), sum state starts NULL
+ *                      (nodeAgg.c:614: strict transfn with no initcond),
+ *                      NULL group keys compare equal for grouping
+ *                      (TupleHashTableMatch/execGrouping.c:520 via
+ *                      ExecQualAndReset with IS NOT DISTINCT semantics)
+ *   - two-phase:       float8_combine float.c:2725, int8pl; the Finalize
+ *                      Aggregate / Remote Subquery Scan / Partial Aggregate
+ *                      plan shape of opentenbase_c_aggregation.out:504-511
+ *
+ * Hashing: parity is defined on RESULTS, not hash values,
+ * so this oracle hashes with splitmix64 rather than restating hash_any
+ * (access/hash/hashfunc.c:110) — bucket population differs, result sets do not.
+ */
+#include "oracle.h"
+#include "otbx_gen.h"
+
+#include <stdlib.h>
+#include <string.h>
+#include <math.h>
+
+/* ================= generation ================= */
+
+ora_status ora_gen_lineitem(ora_lineitem *t, uint64_t seed, int64_t n_global,
+                            uint32_t rank, uint32_t nranks)
+{
+    if (nranks == 0 || n_global % nranks) return ORA_ERR_INVALID;
+    int64_t n = n_global / nranks;
+    t->n = n;
+    t->l_orderkey = malloc(n * 8);
+    t->l_quantity = malloc(n * 8);
+    t->l_extendedprice = malloc(n * 8);
+    t->l_discount = malloc(n * 8);
+    t->l_tax = malloc(n * 8);
+    t->l_returnflag = malloc(n);
+    t->l_linestatus = malloc(n);
+    t->l_shipdate = malloc(n * 4);
+    if (!t->l_orderkey || !t->l_quantity || !t->l_extendedprice || !t->l_discount ||
+        !t->l_tax || !t->l_returnflag || !t->l_linestatus || !t->l_shipdate)
+        return ORA_ERR_OOM;
+    for (int64_t l = 0; l < n; l++) {
+        uint64_t i = otbx_li_global_row((uint64_t)l, rank, nranks);
+        t->l_orderkey[l] = otbx_li_orderkey(i);
+        t->l_quantity[l] = otbx_li_quantity(seed, i);
+        t->l_extendedprice[l] = otbx_li_extendedprice(seed, i);
+        t->l_discount[l] = otbx_li_discount(seed, i);
+        t->l_tax[l] = otbx_li_tax(seed, i);
+        t->l_returnflag[l] = otbx_li_returnflag(seed, i);
+        t->l_linestatus[l] = otbx_li_linestatus(seed, i);
+        t->l_shipdate[l] = otbx_li_shipdate(seed, i);
+    }
+    return ORA_OK;
+}
+
+ora_status ora_gen_orders(ora_orders *t, uint64_t seed, int64_t n_global,
+                          int64_t ncust_global, uint32_t rank, uint32_t nranks)
+{
+    if (nranks == 0 || n_global % nranks) return ORA_ERR_INVALID;
+    int64_t n = n_global / nranks;
+    t->n = n;
+    t->o_orderkey = malloc(n * 8);
+    t->o_custkey = malloc(n * 8);
+    t->o_orderdate = malloc(n * 4);
+    t->o_shippriority = malloc(n * 4);
+    if (!t->o_orderkey || !t->o_custkey || !t->o_orderdate || !t->o_shippriority)
+        return ORA_ERR_OOM;
+    for (int64_t l = 0; l < n; l++) {
+        uint64_t i = otbx_ord_global_row((uint64_t)l, rank, nranks);
+        t->o_orderkey[l] = otbx_ord_orderkey(i);
+        t->o_custkey[l] = otbx_ord_custkey(seed, i, ncust_global);
+        t->o_orderdate[l] = otbx_ord_orderdate(seed, i);
+        t->o_shippriority[l] = otbx_ord_shippriority(i);
+    }
+    return ORA_OK;
+}
+
+ora_status ora_gen_customer(ora_customer *t, uint64_t seed, int64_t n_global,
+                            uint32_t rank, uint32_t nranks)
+{
+    if (nranks == 0 || n_global % nranks) return ORA_ERR_INVALID;
+    int64_t n = n_global / nranks;
+    t->n = n;
+    t->c_custkey = malloc(n * 8);
+    t->c_mktsegment = malloc(n);
+    if (!t->c_custkey || !t->c_mktsegment) return ORA_ERR_OOM;
+    for (int64_t l = 0; l < n; l++) {
+        uint64_t i = otbx_cust_global_row((uint64_t)l, rank, nranks);
+        t->c_custkey[l] = otbx_cust_custkey(i);
+        t->c_mktsegment[l] = otbx_cust_mktsegment(seed, i);
+    }
+    return ORA_OK;
+}
+
+void ora_free_lineitem(ora_lineitem *t)
+{
+    free(t->l_orderkey); free(t->l_quantity); free(t->l_extendedprice);
+    free(t->l_discount); free(t->l_tax); free(t->l_returnflag);
+    free(t->l_linestatus); free(t->l_shipdate);
+    memset(t, 0, sizeof(*t));
+}
+void ora_free_orders(ora_orders *t)
+{
+    free(t->o_orderkey); free(t->o_custkey); free(t->o_orderdate);
+    free(t->o_shippriority);
+    memset(t, 0, sizeof(*t));
+}
+void ora_free_customer(ora_customer *t)
+{
+    free(t->c_custkey); free(t->c_mktsegment);
+    memset(t, 0, sizeof(*t));
+}
+
+/* ================= transition functions ================= */
+
+/* float8pl, float.c:970: result = a+b; error when finite inputs overflow
+ * to inf (CHECKFLOATVAL(result, isinf(arg1)||isinf(arg2), true)) */
+static inline ora_status float8pl(double a, double b, double *out)
+{
+    double r = a + b;
+    if (isinf(r) && !(isinf(a) || isinf(b))) return ORA_ERR_OVERFLOW;
+    *out = r;
+    return ORA_OK;
+}
+
+/* int8inc, int8.c:714: +1 with signed-overflow error */
+static inline ora_status int8inc(int64_t *v)
+{
+    if (*v == INT64_MAX) return ORA_ERR_OVERFLOW;
+    (*v)++;
+    return ORA_OK;
+}
+
+/* float8_accum, float.c:2823 (Youngs-Cramer) on state [N,Sx,Sxx] */
+static ora_status float8_accum(double st[3], double newval)
+{
+    double N = st[0], Sx = st[1], Sxx = st[2], tmp;
+    N += 1.0;
+    Sx += newval;
+    if (st[0] > 0.0) {
+        tmp = newval * N - Sx;
+        Sxx += tmp * tmp / (N * st[0]);
+        if (isinf(Sx) || isinf(Sxx)) {
+            if (!isinf(st[1]) && !isinf(newval)) return ORA_ERR_OVERFLOW;
+            Sxx = NAN;
+        }
+    } else {
+        if (isinf(newval) || isnan(newval)) Sxx = NAN;
+    }
+    st[0] = N; st[1] = Sx; st[2] = Sxx;
+    return ORA_OK;
+}
+
+/* float8_combine, float.c:2725 */
+static ora_status float8_combine(double a[3], const double b[3])
+{
+    double N1 = a[0], Sx1 = a[1], Sxx1 = a[2];
+    double N2 = b[0], Sx2 = b[1], Sxx2 = b[2];
+    if (N1 == 0.0) { a[0] = N2; a[1] = Sx2; a[2] = Sxx2; return ORA_OK; }
+    if (N2 == 0.0) return ORA_OK;
+    double N = N1 + N2;
+    double Sx = Sx1 + Sx2;
+    if (isinf(Sx) && !isinf(Sx1) && !isinf(Sx2)) return ORA_ERR_OVERFLOW;
+    double tmp = Sx1 / N1 - Sx2 / N2;
+    double Sxx = Sxx1 + Sxx2 + N1 * N2 * tmp * tmp / N;
+    if (isinf(Sxx) && !isinf(Sxx1) && !isinf(Sxx2)) return ORA_ERR_OVERFLOW;
+    a[0] = N; a[1] = Sx; a[2] = Sxx;
+    return ORA_OK;
+}
+
+/* ================= config 2: scan + qual + count ================= */
+
+int64_t ora_scan_count_shipdate_le(const int32_t *shipdate, int64_t n, int32_t cutoff)
+{
+    /* ExecScan loop (execScan.c:140): fetch, ExecQual, advance aggregate.
+     * count(*) = int8inc per qualifying tuple. */
+    int64_t count = 0;
+    for (int64_t i = 0; i < n; i++)
+        if (shipdate[i] <= cutoff)
+            count++;
+    return count;
+}
+
+/* ================= Q1 partial aggregate ================= */
+
+ora_status ora_q1_partial(const ora_lineitem *t, int32_t cutoff_day,
+                          ora_q1_group out[8], int *ngroups)
+{
+    /* ≤6 possible (returnflag,linestatus) combos; the reference uses the
+     * simplehash group table (execGrouping.c:295) — with a key domain this
+     * small the oracle keeps a direct-mapped array, same lookup-or-create
+     * + advance_aggregates (nodeAgg.c:856) structure, same emit order as the
+     * plan's ORDER BY (rf,ls). */
+    ora_q1_group g[6];
+    int used[6] = {0};
+    memset(g, 0, sizeof(g));
+    ora_status st;
+
+    for (int64_t i = 0; i < t->n; i++) {
+        if (!(t->l_shipdate[i] <= cutoff_day)) /* ExecQual */
+            continue;
+        uint8_t rf = t->l_returnflag[i], ls = t->l_linestatus[i];
+        int ri = rf == 'A' ? 0 : rf == 'N' ? 1 : 2;
+        int li = ls == 'F' ? 0 : 1;
+        int k = ri * 2 + li;
+        ora_q1_group *e = &g[k];
+        if (!used[k]) {
+            used[k] = 1;
+            e->returnflag = rf;
+            e->linestatus = ls;
+            /* sum states start NULL (strict transfn, no initcond,
+             * nodeAgg.c:614); represent as has-rows-implied: count==0 */
+        }
+        double qty = t->l_quantity[i];
+        double price = t->l_extendedprice[i];
+        double disc = t->l_discount[i];
+        double disc_price = price * (1.0 - disc);           /* ExecProject */
+        double charge = disc_price * (1.0 + t->l_tax[i]);
+        if (e->count_order == 0) {
+            /* first row: strict-transfn state initialization */
+            e->sum_qty = qty;
+            e->sum_base_price = price;
+            e->sum_disc_price = disc_price;
+            e->sum_charge = charge;
+        } else {
+            if ((st = float8pl(e->sum_qty, qty, &e->sum_qty))) return st;
+            if ((st = float8pl(e->sum_base_price, price, &e->sum_base_price))) return st;
+            if ((st = float8pl(e->sum_disc_price, disc_price, &e->sum_disc_price))) return st;
+            if ((st = float8pl(e->sum_charge, charge, &e->sum_charge))) return st;
+        }
+        if ((st = float8_accum(e->qty_acc, qty))) return st;
+        if ((st = float8_accum(e->price_acc, price))) return st;
+        if ((st = float8_accum(e->disc_acc, disc))) return st;
+        if ((st = int8inc(&e->count_order))) return st;
+    }
+
+    /* emit in (rf,ls) order: A < N < R, F < O — matches Q1's ORDER BY and
+     * ASCII order of the key bytes */
+    int ng = 0;
+    for (int k = 0; k < 6; k++)
+        if (used[k])
+            out[ng++] = g[k];
+    *ngroups = ng;
+    return ORA_OK;
+}
+
+ora_status ora_q1_combine(ora_q1_group acc[8], int *nacc,
+                          const ora_q1_group *part, int npart)
+{
+    ora_status st;
+    for (int p = 0; p < npart; p++) {
+        const ora_q1_group *src = &part[p];
+        ora_q1_group *dst = NULL;
+        for (int i = 0; i < *nacc; i++)
+            if (acc[i].returnflag == src->returnflag &&
+                acc[i].linestatus == src->linestatus) { dst = &acc[i]; break; }
+        if (!dst) {
+            dst = &acc[(*nacc)++];
+            *dst = *src;
+            continue;
+        }
+        /* combinefn wiring per AGGSPLIT_FINAL_DESERIAL (nodeAgg.c:3912,4260):
+         * sum(float8) combines with float8pl, count with int8pl,
+         * avg states with float8_combine (float.c:2725). */
+        if (src->count_order > 0) {
+            if (dst->count_order == 0) {
+                dst->sum_qty = src->sum_qty;
+                dst->sum_base_price = src->sum_base_price;
+                dst->sum_disc_price = src->sum_disc_price;
+                dst->sum_charge = src->sum_charge;
+            } else {
+                if ((st = float8pl(dst->sum_qty, src->sum_qty, &dst->sum_qty))) return st;
+                if ((st = float8pl(dst->sum_base_price, src->sum_base_price, &dst->sum_base_price))) return st;
+                if ((st = float8pl(dst->sum_disc_price, src->sum_disc_price, &dst->sum_disc_price))) return st;
+                if ((st = float8pl(dst->sum_charge, src->sum_charge, &dst->sum_charge))) return st;
+            }
+        }
+        if ((st = float8_combine(dst->qty_acc, src->qty_acc))) return st;
+        if ((st = float8_combine(dst->price_acc, src->price_acc))) return st;
+        if ((st = float8_combine(dst->disc_acc, src->disc_acc))) return st;
+        if (dst->count_order > INT64_MAX - src->count_order) return ORA_ERR_OVERFLOW;
+        dst->count_order += src->count_order;
+    }
+    /* keep (rf,ls) sort order */
+    for (int i = 1; i < *nacc; i++)
+        for (int j = i; j > 0; j--) {
+            ora_q1_group *a = &acc[j - 1], *b = &acc[j];
+            if (a->returnflag > b->returnflag ||
+                (a->returnflag == b->returnflag && a->linestatus > b->linestatus)) {
+                ora_q1_group tmpv = *a; *a = *b; *b = tmpv;
+            } else break;
+        }
+    return ORA_OK;
+}
+
+void ora_q1_finalize(ora_q1_group *g, int n)
+{
+    /* float8_avg: Sx/N (NULL when N==0 — cannot occur for emitted groups) */
+    for (int i = 0; i < n; i++) {
+        g[i].avg_qty = g[i].qty_acc[0] > 0 ? g[i].qty_acc[1] / g[i].qty_acc[0] : 0;
+        g[i].avg_price = g[i].price_acc[0] > 0 ? g[i].price_acc[1] / g[i].price_acc[0] : 0;
+        g[i].avg_disc = g[i].disc_acc[0] > 0 ? g[i].disc_acc[1] / g[i].disc_acc[0] : 0;
+    }
+}
+
+/* ================= chained-bucket hash join machinery ================= */
+
+/* nodeHash.c semantics: nbuckets = power of 2 with NTUP_PER_BUCKET=1
+ * (ExecChooseHashTableSize nodeHash.c:677 region), bucketno = hash &
+ * (nbuckets-1) (ExecHashGetBucketAndBatch nodeHash.c:2141), insert at bucket
+ * head (ExecHashTableInsert nodeHash.c:1828), probe walks the chain
+ * comparing hash then key (ExecScanHashBucket nodeHash.c:2174). */
+typedef struct {
+    int64_t nbuckets;      /* power of 2 */
+    int64_t *bucket_head;  /* index into entries, -1 empty */
+    int64_t *next;         /* chain links */
+    const int64_t *keys;   /* build keys (borrowed) */
+} ora_hashtab;
+
+static inline uint64_t ora_hash_i64(int64_t k)
+{
+    return otbx_splitmix64((uint64_t)k);
+}
+
+static int64_t next_pow2(int64_t v)
+{
+    int64_t p = 1;
+    while (p < v) p <<= 1;
+    return p;
+}
+
+static ora_status ht_build(ora_hashtab *ht, const int64_t *keys,
+                           const uint8_t *knull, int64_t n)
+{
+    ht->nbuckets = next_pow2(n < 16 ? 16 : n);
+    ht->bucket_head = malloc(ht->nbuckets * 8);
+    ht->next = malloc((n > 0 ? n : 1) * 8);
+    ht->keys = keys;
+    if (!ht->bucket_head || !ht->next) return ORA_ERR_OOM;
+    for (int64_t i = 0; i < ht->nbuckets; i++) ht->bucket_head[i] = -1;
+    for (int64_t i = 0; i < n; i++) {
+        if (knull && knull[i])
+            continue; /* NULL join key never matches; hashable path drops it
+                       * (ExecHashGetHashValue nodeHash.c:2026 keep_nulls=false) */
+        int64_t b = (int64_t)(ora_hash_i64(keys[i]) & (uint64_t)(ht->nbuckets - 1));
+        ht->next[i] = ht->bucket_head[b];
+        ht->bucket_head[b] = i;
+    }
+    return ORA_OK;
+}
+
+static void ht_free(ora_hashtab *ht)
+{
+    free(ht->bucket_head);
+    free(ht->next);
+}
+
+/* probe: returns first matching build index for key, then use ht_next_match */
+static inline int64_t ht_first_match(const ora_hashtab *ht, int64_t key)
+{
+    int64_t b = (int64_t)(ora_hash_i64(key) & (uint64_t)(ht->nbuckets - 1));
+    int64_t i = ht->bucket_head[b];
+    while (i >= 0 && ht->keys[i] != key)
+        i = ht->next[i];
+    return i;
+}
+static inline int64_t ht_next_match(const ora_hashtab *ht, int64_t i, int64_t key)
+{
+    i = ht->next[i];
+    while (i >= 0 && ht->keys[i] != key)
+        i = ht->next[i];
+    return i;
+}
+
+ora_status ora_join_i64(const int64_t *bkeys, const uint8_t *bnull, int64_t nb,
+                        const int64_t *pkeys, const uint8_t *pnull, int64_t np,
+                        int64_t **out_bidx, int64_t **out_pidx, int64_t *nout)
+{
+    ora_hashtab ht;
+    ora_status st = ht_build(&ht, bkeys, bnull, nb);
+    if (st) { ht_free(&ht); return st; }
+    int64_t cap = 64, n = 0;
+    int64_t *bi = malloc(cap * 8), *pi = malloc(cap * 8);
+    if (!bi || !pi) { free(bi); free(pi); ht_free(&ht); return ORA_ERR_OOM; }
+    for (int64_t p = 0; p < np; p++) {
+        if (pnull && pnull[p])
+            continue; /* HJ_FILL_OUTER etc. not modeled: inner join only */
+        for (int64_t m = ht_first_match(&ht, pkeys[p]); m >= 0;
+             m = ht_next_match(&ht, m, pkeys[p])) {
+            if (n == cap) {
+                cap *= 2;
+                int64_t *nb2 = realloc(bi, cap * 8), *np2 = realloc(pi, cap * 8);
+                if (!nb2 || !np2) { free(nb2 ? nb2 : bi); free(np2 ? np2 : pi); ht_free(&ht); return ORA_ERR_OOM; }
+                bi = nb2; pi = np2;
+            }
+            bi[n] = m; pi[n] = p; n++;
+        }
+    }
+    ht_free(&ht);
+    *out_bidx = bi; *out_pidx = pi; *nout = n;
+    return ORA_OK;
+}
+
+/* ================= simplehash-style open-addressing agg table ============ */
+
+/* simplehash.h: linear probing, fillfactor 0.9 (SH_FILLFACTOR
+ * simplehash.h:187), size = power of 2. Used for the generic agg and Q3. */
+
+typedef struct {
+    int64_t key;
+    int used;
+    int key_isnull;
+    void *payload;
+} sh_slot;
+
+/* ---- generic agg (ora_agg_i64) ---- */
+
+ora_status ora_agg_i64(const int64_t *keys, const uint8_t *key_null,
+                       const double *vals, const uint8_t *val_null,
+                       int64_t n, ora_agg_group **out, int64_t *ngroups)
+{
+    int64_t cap = next_pow2(n < 16 ? 16 : (int64_t)((double)n / 0.85) + 1);
+    ora_agg_group *slots = calloc(cap, sizeof(ora_agg_group));
+    uint8_t *used = calloc(cap, 1);
+    if (!slots || !used) { free(slots); free(used); return ORA_ERR_OOM; }
+    int64_t ng = 0;
+    ora_status st;
+
+    for (int64_t i = 0; i < n; i++) {
+        int isnull = key_null && key_null[i];
+        int64_t k = isnull ? 0 : keys[i];
+        /* LookupTupleHashEntry (execGrouping.c:295): hash, linear probe,
+         * NULL==NULL for grouping purposes */
+        uint64_t h = isnull ? 0x9e3779b97f4a7c15ull : ora_hash_i64(k);
+        int64_t s = (int64_t)(h & (uint64_t)(cap - 1));
+        for (;;) {
+            if (!used[s]) {
+                used[s] = 1;
+                slots[s].key = k;
+                slots[s].key_isnull = isnull;
+                slots[s].sum_isnull = 1; /* strict sum starts NULL */
+                ng++;
+                break;
+            }
+            if (slots[s].key_isnull == isnull && (isnull || slots[s].key == k))
+                break;
+            s = (s + 1) & (cap - 1); /* linear probe, simplehash SH_NEXT */
+        }
+        ora_agg_group *e = &slots[s];
+        if ((st = int8inc(&e->count_star))) goto fail;         /* count(*) */
+        int vnull = val_null && val_null[i];
+        if (!vnull) {                                          /* strict aggs */
+            if ((st = int8inc(&e->count_v))) goto fail;        /* count(v) */
+            if (e->sum_isnull) {
+                e->sum_v = vals[i];                            /* init state */
+                e->sum_isnull = 0;
+            } else if ((st = float8pl(e->sum_v, vals[i], &e->sum_v))) goto fail;
+            if ((st = float8_accum(e->acc, vals[i]))) goto fail; /* avg(v) */
+        }
+    }
+
+    /* emit sorted by (key_isnull, key) for deterministic comparison */
+    ora_agg_group *res = malloc((ng > 0 ? ng : 1) * sizeof(ora_agg_group));
+    if (!res) { st = ORA_ERR_OOM; goto fail; }
+    int64_t j = 0;
+    for (int64_t s = 0; s < cap; s++)
+        if (used[s]) res[j++] = slots[s];
+    /* insertion sort adequate for test-scale group counts */
+    for (int64_t i2 = 1; i2 < ng; i2++)
+        for (int64_t j2 = i2; j2 > 0; j2--) {
+            ora_agg_group *a = &res[j2 - 1], *b = &res[j2];
+            if (a->key_isnull > b->key_isnull ||
+                (a->key_isnull == b->key_isnull && !a->key_isnull && a->key > b->key)) {
+                ora_agg_group t2 = *a; *a = *b; *b = t2;
+            } else break;
+        }
+    free(slots); free(used);
+    *out = res; *ngroups = ng;
+    return ORA_OK;
+fail:
+    free(slots); free(used);
+    return st;
+}
+
+/* ================= Q3 ================= */
+
+typedef struct {
+    int64_t orderkey;
+    int32_t orderdate;
+    int32_t shippriority;
+    double revenue;
+    int used;
+} q3_slot;
+
+ora_status ora_q3_partial(const ora_customer *c, const ora_orders *o,
+                          const ora_lineitem *l, uint8_t segment,
+                          int32_t q3date, ora_q3_row **out, int64_t *ngroups)
+{
+    ora_status st = ORA_OK;
+
+    /* --- build 1: filtered customer (c_mktsegment = seg) key set --- */
+    int64_t *ckeys = malloc((c->n > 0 ? c->n : 1) * 8);
+    if (!ckeys) return ORA_ERR_OOM;
+    int64_t ncf = 0;
+    for (int64_t i = 0; i < c->n; i++)
+        if (c->c_mktsegment[i] == segment)
+            ckeys[ncf++] = c->c_custkey[i];
+    ora_hashtab cht;
+    if ((st = ht_build(&cht, ckeys, NULL, ncf))) goto done1;
+
+    /* --- probe with orders (o_orderdate < date), collect matched orders,
+     *     build 2 keyed on o_orderkey --- */
+    int64_t *okeys = malloc((o->n > 0 ? o->n : 1) * 8);
+    int32_t *odates = malloc((o->n > 0 ? o->n : 1) * 4);
+    int32_t *oprios = malloc((o->n > 0 ? o->n : 1) * 4);
+    if (!okeys || !odates || !oprios) { st = ORA_ERR_OOM; goto done2; }
+    int64_t nof = 0;
+    for (int64_t i = 0; i < o->n; i++) {
+        if (!(o->o_orderdate[i] < q3date))
+            continue;
+        if (ht_first_match(&cht, o->o_custkey[i]) < 0)
+            continue; /* inner join, custkey unique in customer */
+        okeys[nof] = o->o_orderkey[i];
+        odates[nof] = o->o_orderdate[i];
+        oprios[nof] = o->o_shippriority[i];
+        nof++;
+    }
+    ora_hashtab oht;
+    if ((st = ht_build(&oht, okeys, NULL, nof))) { ht_free(&oht); goto done2; }
+
+    /* --- probe with lineitem (l_shipdate > date), agg revenue by group
+     *     (l_orderkey, o_orderdate, o_shippriority) ≡ l_orderkey --- */
+    int64_t cap = next_pow2(nof < 16 ? 16 : (int64_t)((double)nof / 0.85) + 1);
+    q3_slot *slots = calloc(cap, sizeof(q3_slot));
+    if (!slots) { st = ORA_ERR_OOM; ht_free(&oht); goto done2; }
+    int64_t ng = 0;
+    for (int64_t i = 0; i < l->n; i++) {
+        if (!(l->l_shipdate[i] > q3date))
+            continue;
+        int64_t k = l->l_orderkey[i];
+        int64_t m = ht_first_match(&oht, k);
+        if (m < 0)
+            continue;
+        double rev = l->l_extendedprice[i] * (1.0 - l->l_discount[i]);
+        /* group lookup-or-insert (simplehash linear probe) */
+        int64_t s = (int64_t)(ora_hash_i64(k) & (uint64_t)(cap - 1));
+        for (;;) {
+            if (!slots[s].used) {
+                slots[s].used = 1;
+                slots[s].orderkey = k;
+                slots[s].orderdate = odates[m];
+                slots[s].shippriority = oprios[m];
+                slots[s].revenue = rev; /* strict sum init */
+                ng++;
+                break;
+            }
+            if (slots[s].orderkey == k) {
+                if ((st = float8pl(slots[s].revenue, rev, &slots[s].revenue))) {
+                    free(slots); ht_free(&oht); goto done2;
+                }
+                break;
+            }
+            s = (s + 1) & (cap - 1);
+        }
+    }
+    ht_free(&oht);
+
+    ora_q3_row *res = malloc((ng > 0 ? ng : 1) * sizeof(ora_q3_row));
+    if (!res) { st = ORA_ERR_OOM; free(slots); goto done2; }
+    int64_t j = 0;
+    for (int64_t s = 0; s < cap; s++)
+        if (slots[s].used) {
+            res[j].l_orderkey = slots[s].orderkey;
+            res[j].revenue = slots[s].revenue;
+            res[j].o_orderdate = slots[s].orderdate;
+            res[j].o_shippriority = slots[s].shippriority;
+            j++;
+        }
+    free(slots);
+    *out = res;
+    *ngroups = ng;
+done2:
+    free(okeys); free(odates); free(oprios);
+    ht_free(&cht);
+done1:
+    free(ckeys);
+    return st;
+}
+
+/* ORDER BY revenue DESC, o_orderdate ASC LIMIT k (tuplesort merge analog) */
+static int q3_cmp(const void *pa, const void *pb)
+{
+    const ora_q3_row *a = pa, *b = pb;
+    if (a->revenue > b->revenue) return -1;
+    if (a->revenue < b->revenue) return 1;
+    if (a->o_orderdate < b->o_orderdate) return -1;
+    if (a->o_orderdate > b->o_orderdate) return 1;
+    /* tiebreak on orderkey for determinism (not part of SQL contract) */
+    if (a->l_orderkey < b->l_orderkey) return -1;
+    if (a->l_orderkey > b->l_orderkey) return 1;
+    return 0;
+}
+
+int64_t ora_q3_topk(ora_q3_row *rows, int64_t n, int64_t k)
+{
+    qsort(rows, (size_t)n, sizeof(ora_q3_row), q3_cmp);
+    return n < k ? n : k;
+}

@@ -1,0 +1,138 @@
+/*
+ * oracle.h — CPU oracle for the otbx offload: a standalone C restatement of
+ * the OpenTenBase/PostgreSQL DataNode executor hot path
+ * (SeqScan → HashJoin → HashAggregate) plus the two-phase combine the
+ * Coordinator applies.
+ *
+ * TEST INFRASTRUCTURE ONLY. Only tests/, __graft_entry__.smoke() and
+ * bench.py's cpu_baseline leg may call this library — it is the parity
+ * checker and the reported CPU baseline, never the shipped compute path.
+ *
+ * Reference citations are to /root/reference (OpenTenBase @2025-08-08);
+ * this is a restatement of behaviour, not a copy.
+ */
+#ifndef OTBX_ORACLE_H
+#define OTBX_ORACLE_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef enum {
+    ORA_OK = 0,
+    ORA_ERR_OVERFLOW = 1,  /* analog of ereport(ERROR, value out of range) */
+    ORA_ERR_OOM = 2,
+    ORA_ERR_INVALID = 3
+} ora_status;
+
+/* ---- columnar tables (SoA; the staged form of heap pages) ---- */
+
+typedef struct {
+    int64_t n;
+    int64_t *l_orderkey;
+    double  *l_quantity, *l_extendedprice, *l_discount, *l_tax;
+    uint8_t *l_returnflag, *l_linestatus;
+    int32_t *l_shipdate;
+} ora_lineitem;
+
+typedef struct {
+    int64_t n;
+    int64_t *o_orderkey, *o_custkey;
+    int32_t *o_orderdate, *o_shippriority;
+} ora_orders;
+
+typedef struct {
+    int64_t n;
+    int64_t *c_custkey;
+    uint8_t *c_mktsegment;
+} ora_customer;
+
+/* generation (otbx_gen.h): n_global must be divisible by nranks */
+ora_status ora_gen_lineitem(ora_lineitem *t, uint64_t seed, int64_t n_global,
+                            uint32_t rank, uint32_t nranks);
+ora_status ora_gen_orders(ora_orders *t, uint64_t seed, int64_t n_global,
+                          int64_t ncust_global, uint32_t rank, uint32_t nranks);
+ora_status ora_gen_customer(ora_customer *t, uint64_t seed, int64_t n_global,
+                            uint32_t rank, uint32_t nranks);
+void ora_free_lineitem(ora_lineitem *t);
+void ora_free_orders(ora_orders *t);
+void ora_free_customer(ora_customer *t);
+
+/* ---- config 2: SeqScan + qual + COUNT(*) ----
+ * restates ExecScan/ExecQual per-tuple loop (execScan.c:140-363) +
+ * int8inc (int8.c:714). */
+int64_t ora_scan_count_shipdate_le(const int32_t *shipdate, int64_t n, int32_t cutoff);
+
+/* ---- TPC-H Q1: scan + filter + hash aggregate, two-phase ---- */
+
+typedef struct {
+    uint8_t returnflag, linestatus;
+    /* partial (AGGSPLIT_INITIAL_SERIAL) transition states */
+    double sum_qty, sum_base_price, sum_disc_price, sum_charge; /* float8pl chains */
+    double qty_acc[3], price_acc[3], disc_acc[3]; /* float8_accum [N,Sx,Sxx] (float.c:2823) */
+    int64_t count_order;                          /* int8inc (int8.c:714) */
+    /* finalized */
+    double avg_qty, avg_price, avg_disc;          /* float8_avg: Sx/N */
+} ora_q1_group;
+
+/* DataNode side: Partial Aggregate below RemoteSubplan
+ * (agg_fill_hash_table nodeAgg.c:2609; groups sorted by (rf,ls) on emit). */
+ora_status ora_q1_partial(const ora_lineitem *t, int32_t cutoff_day,
+                          ora_q1_group out[8], int *ngroups);
+/* Coordinator side: Finalize Aggregate combine (float8pl for plain sums,
+ * float8_combine float.c:2725 for accum states, int8pl for counts;
+ * nodeAgg.c:3912,4260). */
+ora_status ora_q1_combine(ora_q1_group acc[8], int *nacc,
+                          const ora_q1_group *part, int npart);
+void ora_q1_finalize(ora_q1_group *g, int n);
+
+/* ---- TPC-H Q3: customer ⋈ orders ⋈ lineitem + hash agg + top-k ---- */
+
+typedef struct {
+    int64_t l_orderkey;
+    double revenue;        /* sum(l_extendedprice * (1 - l_discount)) */
+    int32_t o_orderdate, o_shippriority;
+} ora_q3_row;
+
+/* DataNode side: returns ALL groups (caller frees *out). Group key is
+ * (l_orderkey, o_orderdate, o_shippriority) ≡ l_orderkey (o_* functionally
+ * determined). */
+ora_status ora_q3_partial(const ora_customer *c, const ora_orders *o,
+                          const ora_lineitem *l, uint8_t segment,
+                          int32_t q3date, ora_q3_row **out, int64_t *ngroups);
+/* ORDER BY revenue DESC, o_orderdate ASC LIMIT k — sorts in place, returns
+ * min(k, n) (the Coordinator merge-sort analog, execFragment.c:4035-4059). */
+int64_t ora_q3_topk(ora_q3_row *rows, int64_t n, int64_t k);
+
+/* ---- generic entry points for NULL/edge-case parity tests ---- */
+
+typedef struct {
+    int64_t key;
+    int key_isnull;        /* NULL keys form one group (execGrouping.c match) */
+    int64_t count_star;    /* count(*)  — counts every row                    */
+    int64_t count_v;       /* count(v)  — strict, skips NULL inputs           */
+    double sum_v;          /* sum(v)    — strict float8pl chain               */
+    int sum_isnull;        /* SUM over empty/all-NULL group = NULL            */
+    double acc[3];         /* avg(v) float8_accum state                       */
+} ora_agg_group;
+
+/* hash aggregate: group by (nullable) i64 key, aggregate (nullable) f64 v.
+ * val_null/key_null may be NULL pointers (= no NULLs). Caller frees *out.
+ * Groups emitted sorted by (key_isnull, key). */
+ora_status ora_agg_i64(const int64_t *keys, const uint8_t *key_null,
+                       const double *vals, const uint8_t *val_null,
+                       int64_t n, ora_agg_group **out, int64_t *ngroups);
+
+/* inner hash join on i64 keys: build over b, probe with p; emits (bidx,pidx)
+ * pairs. NULL keys never match (strict equality op). Caller frees outputs. */
+ora_status ora_join_i64(const int64_t *bkeys, const uint8_t *bnull, int64_t nb,
+                        const int64_t *pkeys, const uint8_t *pnull, int64_t np,
+                        int64_t **out_bidx, int64_t **out_pidx, int64_t *nout);
+
+#ifdef __cplusplus
+}
+#endif
+#endif
