@@ -1,0 +1,173 @@
+"""Behavioral-invariant tests for the CPU oracle: the NULL/edge-case rules the
+reference's golden files encode: SUM over empty/all-NULL =
+NULL, COUNT = 0, NULL keys form one group, strict transfns skip NULL inputs,
+inner-join NULL keys never match; plus two-phase combine equivalence."""
+import numpy as np
+import pytest
+
+from oracle import oracle_py as ora
+
+
+def test_empty_input():
+    groups = ora.agg_i64(np.empty(0, np.int64), np.empty(0, np.float64))
+    assert groups == []
+
+
+def test_all_null_values_sum_is_null():
+    keys = np.zeros(5, np.int64)
+    vals = np.arange(5, dtype=np.float64)
+    vnull = np.ones(5, np.uint8)
+    g = ora.agg_i64(keys, vals, val_null=vnull)[0]
+    assert g.sum_isnull == 1          # SUM over all-NULL = NULL
+    assert g.count_v == 0             # count(v) skips NULLs
+    assert g.count_star == 5          # count(*) does not
+    assert g.acc[0] == 0.0            # avg state N=0 → NULL avg
+
+
+def test_strict_transfn_skips_nulls():
+    keys = np.zeros(6, np.int64)
+    vals = np.array([1.0, 2.0, 3.0, 4.0, 5.0, 6.0])
+    vnull = np.array([0, 1, 0, 1, 0, 1], np.uint8)
+    g = ora.agg_i64(keys, vals, val_null=vnull)[0]
+    assert g.sum_isnull == 0
+    assert g.sum_v == 1.0 + 3.0 + 5.0
+    assert g.count_v == 3
+    assert g.acc[0] == 3.0 and g.acc[1] == 9.0
+
+
+def test_null_keys_form_one_group():
+    keys = np.array([1, 2, 1, 99, 99], np.int64)
+    knull = np.array([0, 0, 0, 1, 1], np.uint8)
+    vals = np.ones(5, np.float64)
+    groups = ora.agg_i64(keys, vals, key_null=knull)
+    assert len(groups) == 3
+    nullg = [g for g in groups if g.key_isnull]
+    assert len(nullg) == 1 and nullg[0].count_star == 2
+
+
+def test_join_null_keys_never_match():
+    b = np.array([1, 2, 3], np.int64)
+    bn = np.array([0, 1, 0], np.uint8)
+    p = np.array([1, 2, 3, 2], np.int64)
+    pn = np.array([0, 0, 0, 1], np.uint8)
+    bi, pi = ora.join_i64(b, p, bnull=bn, pnull=pn)
+    # only keys 1 and 3 can match; the NULL build 2 and NULL probe 2 cannot
+    assert sorted(p[pi].tolist()) == [1, 3]
+
+
+def test_join_duplicate_build_keys():
+    b = np.array([7, 7, 7, 8], np.int64)
+    p = np.array([7, 8, 9], np.int64)
+    bi, pi = ora.join_i64(b, p)
+    assert len(bi) == 4  # 3 matches for 7, 1 for 8
+    pairs = sorted(zip(p[pi].tolist(), b[bi].tolist()))
+    assert pairs == [(7, 7), (7, 7), (7, 7), (8, 8)]
+
+
+def test_scan_count_conservation():
+    t = ora.gen_tables(40000)
+    sd = t["lineitem"]["l_shipdate"]
+    c = ora.scan_count(sd, 2436)
+    assert c == int((sd <= 2436).sum())
+    c_hi = ora.scan_count(sd, 10**6)
+    assert c_hi == len(sd)
+    assert ora.scan_count(sd, -1) == 0
+
+
+def test_q1_counts_match_numpy():
+    t = ora.gen_tables(40000)
+    li = t["lineitem"]
+    groups = ora.q1_partial(t)
+    mask = li["l_shipdate"] <= 2436
+    assert sum(g.count_order for g in groups) == int(mask.sum())
+    for g in groups:
+        m = mask & (li["l_returnflag"] == g.returnflag) & \
+            (li["l_linestatus"] == g.linestatus)
+        assert g.count_order == int(m.sum())
+        assert g.sum_qty == pytest.approx(float(li["l_quantity"][m].sum()), rel=1e-12)
+
+
+def test_q1_two_phase_combine_equals_single_phase():
+    """Partial per shard + Coordinator combine ≡ single-node run (the
+    Finalize/Partial Aggregate plan of opentenbase_c_aggregation.out:504)."""
+    n = 80000
+    full = ora.q1_partial(ora.gen_tables(n))
+    parts = [ora.q1_partial(ora.gen_tables(n, rank=r, nranks=4)) for r in range(4)]
+    combined = ora.q1_combine(parts)
+    fin_full = ora.q1_finalize(full)
+    fin_comb = ora.q1_finalize(combined)
+    assert len(fin_full) == len(fin_comb) == 4
+    for a, b in zip(fin_full, fin_comb):
+        assert (a.returnflag, a.linestatus) == (b.returnflag, b.linestatus)
+        assert a.count_order == b.count_order
+        for f in ("sum_qty", "sum_base_price", "sum_disc_price", "sum_charge",
+                  "avg_qty", "avg_price", "avg_disc"):
+            assert getattr(a, f) == pytest.approx(getattr(b, f), rel=1e-9)
+
+
+def test_q3_against_numpy_reference():
+    n = 200000
+    t = ora.gen_tables(n, need=("lineitem", "orders", "customer"))
+    rows = ora.q3_partial(t)
+    li, od, cu = t["lineitem"], t["orders"], t["customer"]
+    seg_cust = set(cu["c_custkey"][cu["c_mktsegment"] == 0].tolist())
+    omask = (od["o_orderdate"] < 1169) & \
+        np.isin(od["o_custkey"], np.fromiter(seg_cust, np.int64, len(seg_cust)))
+    okeys = set(od["o_orderkey"][omask].tolist())
+    lmask = (li["l_shipdate"] > 1169) & \
+        np.isin(li["l_orderkey"], np.fromiter(okeys, np.int64, len(okeys)))
+    exp_groups = set(li["l_orderkey"][lmask].tolist())
+    assert set(rows["l_orderkey"].tolist()) == exp_groups
+    # revenue of a specific group matches the scan-order float8pl chain
+    rev = li["l_extendedprice"][lmask] * (1.0 - li["l_discount"][lmask])
+    keys = li["l_orderkey"][lmask]
+    byk = {}
+    for k, r in zip(keys.tolist(), rev.tolist()):
+        byk[k] = byk.get(k, 0.0) + r
+    got = dict(zip(rows["l_orderkey"].tolist(), rows["revenue"].tolist()))
+    for k in list(byk)[:50]:
+        assert got[k] == pytest.approx(byk[k], rel=1e-12)
+
+
+def test_q3_sharded_union_equals_full():
+    """Shards are disjoint by orderkey: the union of per-shard Q3 groups must
+    equal the full-table groups (orderkey groups never span ranks —
+    SURVEY §8e)."""
+    n = 80000
+    full = ora.q3_partial(ora.gen_tables(n, need=("lineitem", "orders", "customer")))
+    parts = []
+    for r in range(2):
+        t = ora.gen_tables(n, rank=r, nranks=2,
+                           need=("lineitem", "orders", "customer"))
+        # NB: customer is sharded by custkey; Q3 needs the replicated
+        # (broadcast) customer set — emulate the all-gather by regenerating
+        # the full customer table on each rank.
+        t_full_cust = ora.gen_tables(n, need=("customer",))
+        t["customer"] = t_full_cust["customer"]
+        parts.append(ora.q3_partial(t))
+    union = np.concatenate(parts)
+    assert len(union) == len(full)
+    fs = {k: v for k, v in zip(full["l_orderkey"], full["revenue"])}
+    us = {k: v for k, v in zip(union["l_orderkey"], union["revenue"])}
+    assert fs.keys() == us.keys()
+    for k in fs:
+        assert us[k] == pytest.approx(fs[k], rel=1e-12)
+
+
+def test_q1_cli_matches_ctypes(tmp_path):
+    """The CLI (bench.py's cpu_baseline entry) agrees with the library."""
+    import json
+    import subprocess
+    import os
+    cli = os.path.join(os.path.dirname(ora.__file__), "oracle_cli")
+    if not os.path.exists(cli):
+        subprocess.run(["make", "-C", os.path.dirname(ora.__file__)], check=True,
+                       capture_output=True)
+    out = subprocess.run([cli, "q1", "--rows", "40000"], check=True,
+                         capture_output=True, text=True).stdout
+    j = json.loads(out)
+    groups = ora.q1_finalize(ora.q1_partial(ora.gen_tables(40000)))
+    assert len(j["groups"]) == len(groups)
+    for jg, g in zip(j["groups"], groups):
+        assert jg["count"] == g.count_order
+        assert jg["sum_qty"] == g.sum_qty
