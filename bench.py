@@ -1,0 +1,250 @@
+#!/usr/bin/env python3
+"""bench.py — measures the BASELINE.json metric (TPC-H Q1/Q3-shaped rows/s +
+GB/s) on MI355X.
+
+Default (no flags): N=1 GPU, workload = BASELINE config 3 — TPC-H Q1 at SF100
+on one GPU (the largest single-GPU config; the metric's home). One "step" =
+one full pass of the offloaded DN fragment (fused scan+filter+partial-agg
+kernel over the HBM-resident SF100 lineitem) + the Coordinator merge
+(all-gather of partial states + finalize). Inputs are generated on-device
+once, before the timed region.
+
+Multi-GPU: one rank per GPU (torch.distributed over RCCL/xGMI), weak scaling
+(each rank owns an SF100 shard); value = whole-job rows/s over all ranks with
+MAX-over-ranks timing.
+
+The JSON line carries:
+  roofline     — dominant kernel (the fused Q1 kernel): achieved algorithmic
+                 GB/s (38 B/row × rows ÷ HIP-event kernel time, events on the
+                 launch stream inside the C-ABI) vs 8 TB/s HBM peak
+                 (MI355X_MICROARCH.md). traffic: measured HBM bytes/launch
+                 from a committed rocprofv3 --pmc summary (profiles/), with
+                 the gfx950 FETCH_SIZE ×2 correction, else null.
+  cpu_baseline — the CPU oracle (kind "port": the reference executor
+                 restated, scalar, 1 thread) timed on this host on a bounded
+                 sample of the same workload.
+"""
+import argparse
+import ctypes as C
+import json
+import os
+import subprocess
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+Q1_BYTES_PER_ROW = 38      # shipdate 4 + flags 2 + 4×f64
+SCAN_BYTES_PER_ROW = 4
+HBM_PEAK = 8.0e12          # B/s, spec (MI355X_MICROARCH.md)
+SF_DEFAULT = 100
+LI_PER_SF = 6_000_000
+
+
+def log(*a):
+    print(*a, file=sys.stderr, flush=True)
+
+
+def load_traffic(workload):
+    p = os.path.join(REPO, "profiles", "roofline_traffic.json")
+    if not os.path.exists(p):
+        return None
+    try:
+        with open(p) as f:
+            j = json.load(f)
+        if j.get("workload") == workload:
+            return float(j["traffic_bytes_per_launch"])
+    except Exception:
+        pass
+    return None
+
+
+def cpu_baseline_q1(sample_rows=240_000_000):
+    """Oracle CLI (scalar port of the reference executor) on a bounded
+    sample; returns the cpu_baseline JSON object."""
+    cli = os.path.join(REPO, "oracle", "oracle_cli")
+    if not os.path.exists(cli):
+        subprocess.run(["make", "-C", os.path.join(REPO, "oracle")],
+                       check=True, capture_output=True)
+    out = subprocess.run([cli, "q1", "--rows", str(sample_rows)],
+                         check=True, capture_output=True, text=True).stdout
+    j = json.loads(out)
+    rows_s = j["rows"] / j["seconds"]
+    return {
+        "value": rows_s, "unit": "rows/s", "cores": 1, "kind": "port",
+        "sample": f"TPC-H Q1 executor over {sample_rows} synthetic lineitem "
+                  f"rows (SF{sample_rows // LI_PER_SF}), scalar, "
+                  f"{j['seconds']:.1f}s",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--sf", type=int, default=SF_DEFAULT,
+                    help="scale factor PER GPU (weak scaling)")
+    ap.add_argument("--workload", default="tpch_q1",
+                    choices=["tpch_q1", "scan_count", "tpch_q3"])
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--cpu-sample-rows", type=int, default=240_000_000)
+    args = ap.parse_args()
+
+    import torch
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1:
+        import torch.distributed as dist
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+    else:
+        dist = None
+
+    from opentenbase_amd import executor as ex
+    from opentenbase_amd import fragment
+
+    ex.init_device(local_rank)
+    stream = C.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+    rows_per_gpu = args.sf * LI_PER_SF
+    n_global = rows_per_gpu * world
+
+    t0 = time.time()
+    if args.workload == "tpch_q1":
+        li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world,
+                                     with_orderkey=False)
+        bytes_per_row = Q1_BYTES_PER_ROW
+    elif args.workload == "scan_count":
+        li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world,
+                                     with_orderkey=False)
+        bytes_per_row = SCAN_BYTES_PER_ROW
+    else:  # tpch_q3
+        li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world)
+        od = ex.GpuOrders.generate(n_global // 4, n_global // 40, rank=rank,
+                                   nranks=world)
+        cu = ex.GpuCustomer.generate(n_global // 40, rank=rank, nranks=world)
+        bytes_per_row = 28  # probe-side streamed bytes
+    torch.cuda.synchronize()
+    log(f"rank {rank}: staged {rows_per_gpu} rows in {time.time() - t0:.1f}s "
+        f"({li.bytes_staged() / 1e9:.1f} GB in HBM)")
+
+    kernel_ms_acc = []
+
+    def step():
+        if args.workload == "tpch_q1":
+            node = ex.GpuQ1PartialAgg(li)
+            node.BeginCustomScan()
+            node._rows = node._run()          # fused kernel
+            kernel_ms_acc.append(node.kernel_ms)
+            s, c = node.partial_state_tensors()
+            rows = fragment.merge_q1_partials(s, c)   # CN merge (collective)
+            return fragment.finalize_q1(rows)
+        elif args.workload == "scan_count":
+            out = torch.zeros(1, dtype=torch.int64, device="cuda")
+            from opentenbase_amd._lib import call
+            call("otbx_scan_count", C.c_void_p(li.t["l_shipdate"].data_ptr()),
+                 C.c_int64(li.n), C.c_int32(2436), C.c_void_p(out.data_ptr()),
+                 stream)
+            return int(out.cpu().item())
+        else:
+            import ctypes as CT
+            from opentenbase_amd._lib import call
+            keys = torch.empty(cu.n, dtype=torch.int64, device="cuda")
+            nk = torch.zeros(1, dtype=torch.int64, device="cuda")
+            call("otbx_filter_customer", CT.byref(cu.cstruct), CT.c_uint8(0),
+                 CT.c_void_p(keys.data_ptr()), CT.c_void_p(nk.data_ptr()), stream)
+            local = keys[: int(nk.cpu().item())]
+            bcast = fragment.broadcast_customer_keys(local)
+            node = ex.GpuQ3Fragment(cu, od, li,
+                                    cust_keys=bcast if world > 1 else None)
+            node.BeginCustomScan()
+            node._rows = node._run()
+            kernel_ms_acc.append(node.kernel_ms)
+            cands = ex.q3_topk(node.groups, 10)
+            return fragment.merge_q3_topk(cands, 10)
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    kernel_ms_acc.clear()
+
+    barrier_sync()
+    t_start = time.time()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.time() - t_start
+
+    # MAX over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.cpu().item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    total_rows = rows_per_gpu * world
+    value = total_rows / (elapsed / args.steps)
+    gbps = value * bytes_per_row / 1e9
+
+    kmean_ms = sum(kernel_ms_acc) / len(kernel_ms_acc) if kernel_ms_acc else None
+    roofline = None
+    if kmean_ms:
+        achieved = rows_per_gpu * bytes_per_row / (kmean_ms / 1e3)  # B/s, per GPU
+        traffic = load_traffic(args.workload)
+        roofline = {
+            "bound": "hbm",
+            "achieved": achieved / 1e9,
+            "peak": HBM_PEAK / 1e9,
+            "unit": "GB/s",
+            "frac": achieved / HBM_PEAK,
+            "traffic": traffic,
+        }
+
+    cpu = None
+    if rank == 0 and world == 1 and not args.no_cpu_baseline \
+            and args.workload == "tpch_q1":
+        log("running cpu_baseline (oracle port, 1 core)...")
+        cpu = cpu_baseline_q1(args.cpu_sample_rows)
+
+    if rank == 0:
+        result = {
+            "metric": "rows/s (TPC-H Q1 SF100-per-GPU, scan→filter→partial "
+                      "hash-agg→merge)" if args.workload == "tpch_q1"
+                      else f"rows/s ({args.workload})",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic (dbgen-shaped, seed 42, generated on-device)",
+            "gb_per_s_scanned": gbps,
+            "config": {
+                "workload": args.workload,
+                "sf_per_gpu": args.sf,
+                "rows_per_gpu": rows_per_gpu,
+                "bytes_per_row": bytes_per_row,
+                "parallelism": f"dp{world} (1 shard/GPU, RCCL merge)",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu,
+        }
+        print(json.dumps(result), flush=True)
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

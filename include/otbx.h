@@ -1,0 +1,197 @@
+/*
+ * otbx.h — C-ABI of the MI355X-native OpenTenBase executor offload.
+ *
+ * This is the drop-in boundary: the entry points a
+ * CustomScan provider (include/nodes/extensible.h:117-152, driven by
+ * executor/nodeCustom.c:31-125 in the reference) calls from
+ * BeginCustomScan / ExecCustomScan / EndCustomScan, callable equally from a
+ * standalone C harness. Plain pointers + sizes only; all device work is
+ * stream-ordered on the caller's HIP stream (passed as void*); errors are
+ * status codes the provider shim converts to ereport(ERROR) (the reference's
+ * error convention, utils/error/elog.c).
+ *
+ * Replaced reference interfaces, per entry point:
+ *   otbx_init/_finish      — _PG_init library load + per-GPU arbiter
+ *                            (utils/fmgr/dfmgr.c:98 load path; one backend
+ *                            process per DN ↔ one HIP device context)
+ *   otbx_stage_*           — heap → device-resident columnar staging; stands
+ *                            where heapgetpage/heapgettup_pagemode
+ *                            (access/heap/heapam.c:388,920) feed the scan
+ *   otbx_scan_count        — SeqScan + ExecQual + COUNT (execScan.c:140,
+ *                            execExprInterp.c:324, int8inc int8.c:714)
+ *   otbx_q1_partial        — the whole DN fragment of TPC-H Q1:
+ *                            SeqScan → qual → project → Partial HashAgg
+ *                            (nodeAgg.c:2609/856; AGGSPLIT_INITIAL_SERIAL,
+ *                            include/nodes/nodes.h:964) as one CustomScan
+ *                            covering the fragment subtree
+ *   otbx_q3_partial        — DN fragment of TPC-H Q3: two hash joins
+ *                            (nodeHash.c:1828/2174, nodeHashjoin.c:186) +
+ *                            Partial HashAgg keyed on l_orderkey
+ *   otbx_agg_i64 /
+ *   otbx_join_i64          — the composable HashAggregate / inner HashJoin
+ *                            operators for generic plan shapes (+ NULL
+ *                            semantics parity: execGrouping.c:295 NULL==NULL
+ *                            grouping; nodeHash.c:2026 NULL keys dropped)
+ *   merge (Coordinator)    — NOT here: the shard merge (execFragment.c:3877)
+ *                            is a collective over ranks, done by the host
+ *                            layer with RCCL (torch.distributed) on the
+ *                            partial-state buffers these calls return.
+ */
+#ifndef OTBX_H
+#define OTBX_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef enum {
+    OTBX_OK = 0,
+    OTBX_ERR_HIP = 1,       /* HIP runtime failure (provider → ereport)   */
+    OTBX_ERR_OOM = 2,
+    OTBX_ERR_INVALID = 3,
+    OTBX_ERR_OVERFLOW = 4,  /* numeric value out of range                 */
+    OTBX_ERR_NO_GPU = 5
+} otbx_status;
+
+const char *otbx_version(void);
+const char *otbx_status_str(otbx_status s);
+
+/* Device lifecycle. otbx_init selects the HIP device (one DN backend ↔ one
+ * GPU; the per-GPU arbiter of SURVEY §8b). */
+otbx_status otbx_init(int device);
+otbx_status otbx_finish(void);
+
+/* Stream-ordered device allocation for standalone (non-torch) callers. */
+otbx_status otbx_device_malloc(void **ptr, size_t bytes);
+otbx_status otbx_device_free(void *ptr);
+otbx_status otbx_memcpy_h2d(void *dst_dev, const void *src_host, size_t bytes, void *stream);
+otbx_status otbx_memcpy_d2h(void *dst_host, const void *src_dev, size_t bytes, void *stream);
+otbx_status otbx_stream_sync(void *stream);
+
+/* ---- staged columnar tables (device pointers; SoA — DESIGN.md §2) ---- */
+
+typedef struct {
+    int64_t n;
+    int64_t *l_orderkey;      /* may be NULL if not staged (Q1 needs none) */
+    double  *l_quantity, *l_extendedprice, *l_discount, *l_tax;
+    uint8_t *l_returnflag, *l_linestatus;
+    int32_t *l_shipdate;
+} otbx_lineitem_dev;
+
+typedef struct {
+    int64_t n;
+    int64_t *o_orderkey, *o_custkey;
+    int32_t *o_orderdate, *o_shippriority;
+} otbx_orders_dev;
+
+typedef struct {
+    int64_t n;
+    int64_t *c_custkey;
+    uint8_t *c_mktsegment;
+} otbx_customer_dev;
+
+/* On-device synthetic generation (the dbgen analog; same counter-based
+ * functions as the CPU oracle — oracle/otbx_gen.h — so tables are
+ * bit-identical on both sides). Caller provides the device buffers
+ * (column pointers in the struct, each sized for n_global/nranks rows). */
+otbx_status otbx_gen_lineitem_dev(const otbx_lineitem_dev *t, uint64_t seed,
+                                  int64_t n_global, uint32_t rank,
+                                  uint32_t nranks, void *stream);
+otbx_status otbx_gen_orders_dev(const otbx_orders_dev *t, uint64_t seed,
+                                int64_t n_global, int64_t ncust_global,
+                                uint32_t rank, uint32_t nranks, void *stream);
+otbx_status otbx_gen_customer_dev(const otbx_customer_dev *t, uint64_t seed,
+                                  int64_t n_global, uint32_t rank,
+                                  uint32_t nranks, void *stream);
+
+/* ---- config 2: SeqScan + qual + COUNT(*) (scan-bandwidth kernel) ----
+ * count_dev: one int64 device slot (zeroed by the call). */
+otbx_status otbx_scan_count(const int32_t *shipdate_dev, int64_t n,
+                            int32_t cutoff, int64_t *count_dev, void *stream);
+
+/* ---- TPC-H Q1 DN fragment ----
+ * Partial-aggregate state per (l_returnflag,l_linestatus) group, dense over
+ * the 6 possible combos, fixed slot order (A,F)(A,O)(N,F)(N,O)(R,F)(R,O):
+ *   sums_dev:   double[6][5] = {sum_qty, sum_base_price, sum_disc_price,
+ *                               sum_charge, sum_disc}
+ *   counts_dev: int64[6]     = count(*)
+ * Both zeroed by the call; the Coordinator-side merge all-gathers exactly
+ * these buffers. kernel_ms (host, may be NULL): HIP-event time of the fused
+ * kernel on `stream` (the bench's roofline numerator — DESIGN.md §4). */
+otbx_status otbx_q1_partial(const otbx_lineitem_dev *t, int32_t cutoff_day,
+                            double *sums_dev, int64_t *counts_dev,
+                            void *stream, float *kernel_ms);
+
+/* ---- TPC-H Q3 DN fragment ----
+ * customer/orders/lineitem staged on-device; custkeys of the replicated
+ * (broadcast) customer build side are read from cust_keys_dev when non-NULL
+ * (ncust_keys rows; the post-all-gather buffer), else from c->... filtered
+ * by segment locally.
+ * Outputs (device, caller-allocated):
+ *   groups_dev:  capacity cap_groups entries of otbx_q3_group
+ *   ngroups_dev: int64[1] — compacted group count
+ * ws_dev: workspace (hash tables), size from otbx_q3_workspace_bytes. */
+typedef struct {
+    int64_t l_orderkey;
+    double revenue;
+    int32_t o_orderdate, o_shippriority;
+} otbx_q3_group;
+
+otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
+                                    size_t *bytes);
+otbx_status otbx_q3_partial(const otbx_customer_dev *c,
+                            const otbx_orders_dev *o,
+                            const otbx_lineitem_dev *l,
+                            const int64_t *cust_keys_dev, int64_t ncust_keys,
+                            uint8_t segment, int32_t q3date,
+                            void *ws_dev, size_t ws_bytes,
+                            otbx_q3_group *groups_dev, int64_t cap_groups,
+                            int64_t *ngroups_dev, void *stream,
+                            float *kernel_ms);
+/* helper for the broadcast build side: compact custkeys where
+ * c_mktsegment == segment into keys_out_dev, count into nkeys_dev (zeroed). */
+otbx_status otbx_filter_customer(const otbx_customer_dev *c, uint8_t segment,
+                                 int64_t *keys_out_dev, int64_t *nkeys_dev,
+                                 void *stream);
+
+/* ---- composable operators (generic plan shapes + NULL-semantics parity) --
+
+ * Hash aggregate: group by nullable i64 key, aggregate nullable f64 value;
+ * count(*), count(v), sum(v), avg-N/Sx. Output: open-addressing table
+ * compacted to groups_dev (cap = n). Null bitmaps are byte-per-row (1 =
+ * NULL), may be NULL pointers. */
+typedef struct {
+    int64_t key;
+    int64_t count_star;
+    int64_t count_v;
+    double sum_v;
+    int32_t key_isnull;
+    int32_t sum_isnull;
+} otbx_agg_group;
+
+otbx_status otbx_agg_i64_workspace_bytes(int64_t n, size_t *bytes);
+otbx_status otbx_agg_i64(const int64_t *keys_dev, const uint8_t *key_null_dev,
+                         const double *vals_dev, const uint8_t *val_null_dev,
+                         int64_t n, void *ws_dev, size_t ws_bytes,
+                         otbx_agg_group *groups_dev, int64_t *ngroups_dev,
+                         void *stream);
+
+/* Inner hash join on i64 keys: emits (build_idx, probe_idx) pairs in
+ * arbitrary order (result-set parity; SQL imposes no order). pairs capacity
+ * cap_pairs; overflow → OTBX_ERR_INVALID reported via npairs_dev = -1. */
+otbx_status otbx_join_i64_workspace_bytes(int64_t nb, size_t *bytes);
+otbx_status otbx_join_i64(const int64_t *bkeys_dev, const uint8_t *bnull_dev,
+                          int64_t nb,
+                          const int64_t *pkeys_dev, const uint8_t *pnull_dev,
+                          int64_t np, void *ws_dev, size_t ws_bytes,
+                          int64_t *out_bidx_dev, int64_t *out_pidx_dev,
+                          int64_t cap_pairs, int64_t *npairs_dev,
+                          void *stream);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

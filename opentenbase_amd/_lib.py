@@ -1,0 +1,92 @@
+"""ctypes binding for libotbx.so (the C-ABI in include/otbx.h).
+
+The .so is built in-tree (opentenbase_amd/csrc/Makefile, hipcc gfx950) and
+travels with the repo snapshot. On a machine WITH a GPU, a missing or
+unloadable extension is a hard error — the HIP path is the only compute path
+(no CPU fallback; the oracle is test infrastructure only).
+"""
+import ctypes as C
+import os
+import subprocess
+
+_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "csrc")
+_SO = os.path.join(_DIR, "libotbx.so")
+
+OTBX_OK = 0
+_STATUS = {0: "ok", 1: "hip runtime error", 2: "out of memory",
+           3: "invalid argument", 4: "value out of range", 5: "no gpu"}
+
+
+class OtbxError(RuntimeError):
+    """Analog of ereport(ERROR): any non-OK status from the C-ABI."""
+
+    def __init__(self, status, what=""):
+        self.status = status
+        super().__init__(f"otbx error {status} ({_STATUS.get(status, '?')}) {what}")
+
+
+class LineitemDev(C.Structure):
+    _fields_ = [
+        ("n", C.c_int64),
+        ("l_orderkey", C.c_void_p),
+        ("l_quantity", C.c_void_p), ("l_extendedprice", C.c_void_p),
+        ("l_discount", C.c_void_p), ("l_tax", C.c_void_p),
+        ("l_returnflag", C.c_void_p), ("l_linestatus", C.c_void_p),
+        ("l_shipdate", C.c_void_p),
+    ]
+
+
+class OrdersDev(C.Structure):
+    _fields_ = [
+        ("n", C.c_int64),
+        ("o_orderkey", C.c_void_p), ("o_custkey", C.c_void_p),
+        ("o_orderdate", C.c_void_p), ("o_shippriority", C.c_void_p),
+    ]
+
+
+class CustomerDev(C.Structure):
+    _fields_ = [
+        ("n", C.c_int64),
+        ("c_custkey", C.c_void_p), ("c_mktsegment", C.c_void_p),
+    ]
+
+
+def build():
+    """Compile the extension (hipcc cross-compiles without a GPU)."""
+    subprocess.run(["make", "-C", _DIR], check=True)
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(_SO):
+            raise OtbxError(5, f"HIP extension not built: {_SO} missing — "
+                               "run __graft_entry__.build()")
+        _lib = C.CDLL(_SO)
+        _lib.otbx_version.restype = C.c_char_p
+        _lib.otbx_status_str.restype = C.c_char_p
+    return _lib
+
+
+def check(status, what=""):
+    if status != OTBX_OK:
+        raise OtbxError(status, what)
+
+
+def call(name, *args):
+    check(getattr(lib(), name)(*args), what=name)
+
+
+EXPORTED_SYMBOLS = [
+    "otbx_version", "otbx_status_str", "otbx_init", "otbx_finish",
+    "otbx_device_malloc", "otbx_device_free", "otbx_memcpy_h2d",
+    "otbx_memcpy_d2h", "otbx_stream_sync",
+    "otbx_gen_lineitem_dev", "otbx_gen_orders_dev", "otbx_gen_customer_dev",
+    "otbx_scan_count", "otbx_q1_partial",
+    "otbx_q3_workspace_bytes", "otbx_q3_partial", "otbx_filter_customer",
+    "otbx_agg_i64_workspace_bytes", "otbx_agg_i64",
+    "otbx_join_i64_workspace_bytes", "otbx_join_i64",
+]

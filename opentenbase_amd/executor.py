@@ -1,0 +1,354 @@
+"""Host-side mirror of the reference's CustomScan provider contract.
+
+The reference drives a custom executor node through CustomExecMethods
+(include/nodes/extensible.h:117-152; nodeCustom.c:31-125):
+BeginCustomScan(estate, eflags) → repeated ExecCustomScan(node) returning one
+tuple per call (None = done) → ReScanCustomScan → EndCustomScan. The classes
+here keep those names, argument meanings and the error convention (OtbxError ↔
+ereport(ERROR)) so the parity tests read like the reference's node lifecycle;
+INTEGRATION.md shows the C provider these map onto.
+
+Device memory, streams: torch (plumbing only — all compute is in libotbx.so;
+there is NO torch/CPU fallback for any operator here).
+"""
+import ctypes as C
+
+import torch
+
+from ._lib import (CustomerDev, LineitemDev, OrdersDev, OtbxError, call, check,
+                   lib)
+
+Q1_SLOT_ORDER = [(b"A", b"F"), (b"A", b"O"), (b"N", b"F"),
+                 (b"N", b"O"), (b"R", b"F"), (b"R", b"O")]
+Q1_CUTOFF_DEFAULT = 2436   # date '1998-12-01' - interval '90 day'
+Q3_DATE_DEFAULT = 1169     # date '1995-03-15'
+SEED_DEFAULT = 42
+
+
+def _stream():
+    return C.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def init_device(device=0):
+    torch.cuda.set_device(device)
+    call("otbx_init", device)
+
+
+def device_synchronize():
+    torch.cuda.synchronize()
+
+
+# ---------------- staged tables (device-resident column cache) -------------
+
+class GpuLineitem:
+    COLS = [("l_orderkey", torch.int64), ("l_quantity", torch.float64),
+            ("l_extendedprice", torch.float64), ("l_discount", torch.float64),
+            ("l_tax", torch.float64), ("l_returnflag", torch.uint8),
+            ("l_linestatus", torch.uint8), ("l_shipdate", torch.int32)]
+
+    def __init__(self, n, with_orderkey=True, device="cuda"):
+        self.n = n
+        self.t = {}
+        for name, dt in self.COLS:
+            if name == "l_orderkey" and not with_orderkey:
+                self.t[name] = None
+                continue
+            self.t[name] = torch.empty(n, dtype=dt, device=device)
+        self.cstruct = LineitemDev(
+            n=n, **{k: C.c_void_p(0 if v is None else v.data_ptr())
+                    for k, v in self.t.items()})
+
+    @classmethod
+    def generate(cls, n_global, rank=0, nranks=1, seed=SEED_DEFAULT,
+                 with_orderkey=True):
+        t = cls(n_global // nranks, with_orderkey=with_orderkey)
+        call("otbx_gen_lineitem_dev", C.byref(t.cstruct), C.c_uint64(seed),
+             C.c_int64(n_global), C.c_uint32(rank), C.c_uint32(nranks),
+             _stream())
+        return t
+
+    def bytes_staged(self):
+        return sum(v.numel() * v.element_size()
+                   for v in self.t.values() if v is not None)
+
+
+class GpuOrders:
+    def __init__(self, n, device="cuda"):
+        self.n = n
+        self.t = {
+            "o_orderkey": torch.empty(n, dtype=torch.int64, device=device),
+            "o_custkey": torch.empty(n, dtype=torch.int64, device=device),
+            "o_orderdate": torch.empty(n, dtype=torch.int32, device=device),
+            "o_shippriority": torch.empty(n, dtype=torch.int32, device=device),
+        }
+        self.cstruct = OrdersDev(
+            n=n, **{k: C.c_void_p(v.data_ptr()) for k, v in self.t.items()})
+
+    @classmethod
+    def generate(cls, n_global, ncust_global, rank=0, nranks=1,
+                 seed=SEED_DEFAULT):
+        t = cls(n_global // nranks)
+        call("otbx_gen_orders_dev", C.byref(t.cstruct), C.c_uint64(seed),
+             C.c_int64(n_global), C.c_int64(ncust_global), C.c_uint32(rank),
+             C.c_uint32(nranks), _stream())
+        return t
+
+
+class GpuCustomer:
+    def __init__(self, n, device="cuda"):
+        self.n = n
+        self.t = {
+            "c_custkey": torch.empty(n, dtype=torch.int64, device=device),
+            "c_mktsegment": torch.empty(n, dtype=torch.uint8, device=device),
+        }
+        self.cstruct = CustomerDev(
+            n=n, **{k: C.c_void_p(v.data_ptr()) for k, v in self.t.items()})
+
+    @classmethod
+    def generate(cls, n_global, rank=0, nranks=1, seed=SEED_DEFAULT):
+        t = cls(n_global // nranks)
+        call("otbx_gen_customer_dev", C.byref(t.cstruct), C.c_uint64(seed),
+             C.c_int64(n_global), C.c_uint32(rank), C.c_uint32(nranks),
+             _stream())
+        return t
+
+
+# ---------------- CustomScan lifecycle ----------------
+
+class CustomScanState:
+    """Base lifecycle, mirroring CustomExecMethods (extensible.h:117-152)."""
+
+    def __init__(self):
+        self._begun = False
+
+    def BeginCustomScan(self, estate=None, eflags=0):
+        self._begun = True
+        self._begin(estate, eflags)
+        self._rows = None
+        self._pos = 0
+
+    def ExecCustomScan(self):
+        """One tuple per call; None = end of stream (nodeCustom.c:104)."""
+        if not self._begun:
+            raise OtbxError(3, "ExecCustomScan before BeginCustomScan")
+        if self._rows is None:
+            self._rows = self._run()
+        if self._pos >= len(self._rows):
+            return None
+        row = self._rows[self._pos]
+        self._pos += 1
+        return row
+
+    def ReScanCustomScan(self):
+        self._rows = None
+        self._pos = 0
+
+    def EndCustomScan(self):
+        self._begun = False
+        self._end()
+
+    # subclass hooks
+    def _begin(self, estate, eflags): pass
+    def _run(self): raise NotImplementedError
+    def _end(self): pass
+
+
+class GpuSeqScanCount(CustomScanState):
+    """SeqScan + qual + COUNT(*) (BASELINE config 2). Replaces the
+    SeqScan→Agg(count) fragment (execScan.c:140 + nodeAgg.c)."""
+
+    def __init__(self, lineitem, cutoff=Q1_CUTOFF_DEFAULT):
+        super().__init__()
+        self.li = lineitem
+        self.cutoff = cutoff
+
+    def _run(self):
+        out = torch.zeros(1, dtype=torch.int64, device="cuda")
+        call("otbx_scan_count", C.c_void_p(self.li.t["l_shipdate"].data_ptr()),
+             C.c_int64(self.li.n), C.c_int32(self.cutoff),
+             C.c_void_p(out.data_ptr()), _stream())
+        return [(int(out.cpu().item()),)]
+
+
+class GpuQ1PartialAgg(CustomScanState):
+    """The DN fragment of TPC-H Q1: SeqScan → qual → project → Partial
+    HashAgg (AGGSPLIT_INITIAL_SERIAL), one fused kernel. Emits one partial
+    group state per call, ordered by (l_returnflag, l_linestatus).
+
+    partial_state_tensors() exposes the dense device buffers the Coordinator
+    merge all-gathers (fragment.py)."""
+
+    def __init__(self, lineitem, cutoff=Q1_CUTOFF_DEFAULT):
+        super().__init__()
+        self.li = lineitem
+        self.cutoff = cutoff
+        self.kernel_ms = None
+        self.sums = None
+        self.counts = None
+
+    def _run(self):
+        self.sums = torch.empty((6, 5), dtype=torch.float64, device="cuda")
+        self.counts = torch.empty(6, dtype=torch.int64, device="cuda")
+        ms = C.c_float(0.0)
+        call("otbx_q1_partial", C.byref(self.li.cstruct), C.c_int32(self.cutoff),
+             C.c_void_p(self.sums.data_ptr()), C.c_void_p(self.counts.data_ptr()),
+             _stream(), C.byref(ms))
+        self.kernel_ms = ms.value
+        return q1_rows_from_state(self.sums, self.counts)
+
+    def partial_state_tensors(self):
+        return self.sums, self.counts
+
+
+def q1_rows_from_state(sums, counts):
+    """Dense [6,5] sums + [6] counts → partial group rows (host)."""
+    s = sums.cpu().numpy() if hasattr(sums, "cpu") else sums
+    c = counts.cpu().numpy() if hasattr(counts, "cpu") else counts
+    rows = []
+    for g, (rf, ls) in enumerate(Q1_SLOT_ORDER):
+        if c[g] == 0:
+            continue
+        rows.append({
+            "l_returnflag": rf.decode(), "l_linestatus": ls.decode(),
+            "sum_qty": float(s[g][0]), "sum_base_price": float(s[g][1]),
+            "sum_disc_price": float(s[g][2]), "sum_charge": float(s[g][3]),
+            "sum_disc": float(s[g][4]), "count_order": int(c[g]),
+        })
+    return rows
+
+
+def q1_finalize(rows):
+    """Finalize Aggregate (CN): avg = Sx/N (float8_avg semantics)."""
+    out = []
+    for r in rows:
+        n = r["count_order"]
+        fin = dict(r)
+        fin["avg_qty"] = r["sum_qty"] / n
+        fin["avg_price"] = r["sum_base_price"] / n
+        fin["avg_disc"] = r["sum_disc"] / n
+        out.append(fin)
+    return out
+
+
+class GpuQ3Fragment(CustomScanState):
+    """The DN fragment of TPC-H Q3: customer⋈orders⋈lineitem + Partial
+    HashAgg keyed on l_orderkey. cust_keys: optional device int64 tensor of
+    the REPLICATED (post-broadcast) filtered customer keys;
+    without it, the local customer shard is filtered in-kernel."""
+
+    def __init__(self, customer, orders, lineitem, segment=0,
+                 date=Q3_DATE_DEFAULT, cust_keys=None):
+        super().__init__()
+        self.cu, self.od, self.li = customer, orders, lineitem
+        self.segment, self.date = segment, date
+        self.cust_keys = cust_keys
+        self.kernel_ms = None
+        self.groups = None   # structured numpy array of all partial groups
+
+    def _run(self):
+        L = lib()
+        ncust = self.cu.n if self.cust_keys is None else len(self.cust_keys)
+        ws_bytes = C.c_size_t(0)
+        check(L.otbx_q3_workspace_bytes(C.c_int64(ncust), C.c_int64(self.od.n),
+                                        C.byref(ws_bytes)))
+        ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
+        cap = self.od.n if self.od.n > 0 else 1
+        groups = torch.empty(cap * 24, dtype=torch.uint8, device="cuda")
+        ng = torch.zeros(1, dtype=torch.int64, device="cuda")
+        ms = C.c_float(0.0)
+        ck = C.c_void_p(self.cust_keys.data_ptr()) if self.cust_keys is not None else None
+        nck = C.c_int64(0 if self.cust_keys is None else len(self.cust_keys))
+        call("otbx_q3_partial", C.byref(self.cu.cstruct), C.byref(self.od.cstruct),
+             C.byref(self.li.cstruct), ck, nck, C.c_uint8(self.segment),
+             C.c_int32(self.date), C.c_void_p(ws.data_ptr()),
+             C.c_size_t(ws_bytes.value), C.c_void_p(groups.data_ptr()),
+             C.c_int64(cap), C.c_void_p(ng.data_ptr()), _stream(), C.byref(ms))
+        self.kernel_ms = ms.value
+        n = int(ng.cpu().item())
+        import numpy as np
+        dt = np.dtype([("l_orderkey", "i8"), ("revenue", "f8"),
+                       ("o_orderdate", "i4"), ("o_shippriority", "i4")])
+        raw = groups[: n * 24].cpu().numpy().tobytes()
+        self.groups = np.frombuffer(raw, dtype=dt).copy()
+        # partial top-k candidates (LIMIT 10 below the merge); full ORDER BY
+        # offload is SURVEY §8f.2
+        return [tuple(r) for r in q3_topk(self.groups, 10)]
+
+
+def q3_topk(groups, k=10):
+    """ORDER BY revenue DESC, o_orderdate ASC LIMIT k (host-side; the
+    coordinator merge-sort analog, execFragment.c:4035)."""
+    import numpy as np
+    order = np.lexsort((groups["l_orderkey"], groups["o_orderdate"],
+                        -groups["revenue"]))
+    return groups[order][:k]
+
+
+class GpuHashAgg(CustomScanState):
+    """Composable HashAggregate over i64 keys / f64 values with full NULL
+    semantics (execGrouping.c:295 + nodeAgg.c:743). Inputs: device tensors;
+    null bitmaps optional uint8 tensors (1 = NULL)."""
+
+    def __init__(self, keys, vals, key_null=None, val_null=None):
+        super().__init__()
+        self.keys, self.vals = keys, vals
+        self.key_null, self.val_null = key_null, val_null
+
+    def _run(self):
+        import numpy as np
+        L = lib()
+        n = len(self.keys)
+        ws_bytes = C.c_size_t(0)
+        check(L.otbx_agg_i64_workspace_bytes(C.c_int64(n), C.byref(ws_bytes)))
+        ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8, device="cuda")
+        out = torch.empty(max(n, 1) * 40, dtype=torch.uint8, device="cuda")
+        ng = torch.zeros(1, dtype=torch.int64, device="cuda")
+        kn = C.c_void_p(self.key_null.data_ptr()) if self.key_null is not None else None
+        vn = C.c_void_p(self.val_null.data_ptr()) if self.val_null is not None else None
+        call("otbx_agg_i64", C.c_void_p(self.keys.data_ptr()), kn,
+             C.c_void_p(self.vals.data_ptr()), vn, C.c_int64(n),
+             C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+             C.c_void_p(out.data_ptr()), C.c_void_p(ng.data_ptr()), _stream())
+        ngroups = int(ng.cpu().item())
+        dt = np.dtype([("key", "i8"), ("count_star", "i8"), ("count_v", "i8"),
+                       ("sum_v", "f8"), ("key_isnull", "i4"), ("sum_isnull", "i4")])
+        raw = out[: ngroups * 40].cpu().numpy().tobytes()
+        arr = np.frombuffer(raw, dtype=dt).copy()
+        arr.sort(order=["key_isnull", "key"])
+        return list(arr)
+
+
+class GpuHashJoin(CustomScanState):
+    """Composable inner HashJoin on i64 keys (nodeHash.c/nodeHashjoin.c).
+    Emits (build_idx, probe_idx) pairs; result-set parity (order-free)."""
+
+    def __init__(self, build_keys, probe_keys, build_null=None, probe_null=None,
+                 cap_pairs=None):
+        super().__init__()
+        self.bk, self.pk = build_keys, probe_keys
+        self.bn, self.pn = build_null, probe_null
+        self.cap_pairs = cap_pairs
+
+    def _run(self):
+        import numpy as np
+        L = lib()
+        nb, npr = len(self.bk), len(self.pk)
+        ws_bytes = C.c_size_t(0)
+        check(L.otbx_join_i64_workspace_bytes(C.c_int64(nb), C.byref(ws_bytes)))
+        ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8, device="cuda")
+        cap = self.cap_pairs if self.cap_pairs else max(4 * max(nb, npr), 64)
+        ob = torch.empty(cap, dtype=torch.int64, device="cuda")
+        op = torch.empty(cap, dtype=torch.int64, device="cuda")
+        npairs = torch.zeros(1, dtype=torch.int64, device="cuda")
+        bn = C.c_void_p(self.bn.data_ptr()) if self.bn is not None else None
+        pn = C.c_void_p(self.pn.data_ptr()) if self.pn is not None else None
+        call("otbx_join_i64", C.c_void_p(self.bk.data_ptr()), bn, C.c_int64(nb),
+             C.c_void_p(self.pk.data_ptr()), pn, C.c_int64(npr),
+             C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+             C.c_void_p(ob.data_ptr()), C.c_void_p(op.data_ptr()),
+             C.c_int64(cap), C.c_void_p(npairs.data_ptr()), _stream())
+        n = int(npairs.cpu().item())
+        if n > cap:
+            raise OtbxError(3, f"join pair overflow: {n} > cap {cap}")
+        return list(zip(ob[:n].cpu().numpy().tolist(),
+                        op[:n].cpu().numpy().tolist()))

@@ -1,0 +1,92 @@
+"""Coordinator shard-merge over RCCL/xGMI — the MI355X-native replacement of
+the reference's RemoteSubplan recv + FN transport (execFragment.c:3877,
+forward/*; SURVEY §2 'Fragment exchange'). Semantics: per-shard partial-agg
+streams are CONCATENATED at the CN and combined by the Finalize Aggregate
+(float8pl / int8pl / float8_combine — nodeAgg.c:3912,4260).
+
+One process per GPU (1 DN shard ↔ 1 GPU); collectives via torch.distributed —
+backend "nccl" IS RCCL on ROCm (xGMI inside the node); CPU tests use "gloo"
+with the same code path.
+"""
+import torch
+import torch.distributed as dist
+
+from .executor import Q1_SLOT_ORDER, q1_finalize, q1_rows_from_state
+
+
+def is_dist():
+    return dist.is_available() and dist.is_initialized()
+
+
+def merge_q1_partials(sums, counts):
+    """All-gather the dense per-rank Q1 partial states (6×5 sums + 6 counts —
+    the 'Remote Subquery Scan' payload: 4 rows/DN) and combine on every rank
+    (combine is cheap and keeping it symmetric avoids a broadcast back).
+
+    Dense slot layout makes the combine a pure elementwise sum: float8pl for
+    sums, int8pl for counts, slot-aligned groups."""
+    if not is_dist() or dist.get_world_size() == 1:
+        return q1_rows_from_state(sums, counts)
+    world = dist.get_world_size()
+    gs = [torch.empty_like(sums) for _ in range(world)]
+    gc = [torch.empty_like(counts) for _ in range(world)]
+    dist.all_gather(gs, sums)
+    dist.all_gather(gc, counts)
+    tot_s = torch.zeros_like(sums)
+    tot_c = torch.zeros_like(counts)
+    for s in gs:
+        tot_s += s            # float8pl chain over shards
+    for c in gc:
+        tot_c += c            # int8pl
+    return q1_rows_from_state(tot_s, tot_c)
+
+
+def allgather_variable(t):
+    """All-gather a 1-D tensor with per-rank variable length (the FN-page
+    concatenation semantics). Returns the concatenation over ranks."""
+    if not is_dist() or dist.get_world_size() == 1:
+        return t
+    world = dist.get_world_size()
+    n = torch.tensor([t.numel()], dtype=torch.int64, device=t.device)
+    ns = [torch.empty_like(n) for _ in range(world)]
+    dist.all_gather(ns, n)
+    counts = [int(x.item()) for x in ns]
+    mx = max(counts) if counts else 0
+    pad = torch.empty(mx, dtype=t.dtype, device=t.device)
+    pad[: t.numel()] = t
+    outs = [torch.empty_like(pad) for _ in range(world)]
+    dist.all_gather(outs, pad)
+    return torch.cat([o[:c] for o, c in zip(outs, counts)])
+
+
+def broadcast_customer_keys(local_keys):
+    """Q3 replicated build side: all-gather each rank's filtered customer
+    keys (reference analog: replicated distribution / 'Distribute results by'
+    exchange, xl_join.out:13-20). Shards are disjoint (custkey % nranks), so
+    concatenation = the full filtered key set."""
+    return allgather_variable(local_keys)
+
+
+def merge_q3_topk(candidates, k=10):
+    """All-gather per-rank top-k candidate rows (as raw bytes) and take the
+    global top-k (merge-sorted recv analog, execFragment.c:4035-4059)."""
+    import numpy as np
+    dt = np.dtype([("l_orderkey", "i8"), ("revenue", "f8"),
+                   ("o_orderdate", "i4"), ("o_shippriority", "i4")])
+    if is_dist() and dist.get_world_size() > 1:
+        raw = torch.from_numpy(
+            np.ascontiguousarray(candidates).view(np.uint8).reshape(-1).copy())
+        dev = "cuda" if torch.cuda.is_available() and \
+            dist.get_backend() == "nccl" else "cpu"
+        raw = raw.to(dev)
+        allraw = allgather_variable(raw).cpu().numpy().tobytes()
+        cands = np.frombuffer(allraw, dtype=dt).copy()
+    else:
+        cands = np.asarray(candidates, dtype=dt)
+    order = np.lexsort((cands["l_orderkey"], cands["o_orderdate"],
+                        -cands["revenue"]))
+    return cands[order][:k]
+
+
+def finalize_q1(rows):
+    return q1_finalize(rows)

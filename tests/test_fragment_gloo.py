@@ -1,0 +1,93 @@
+"""Multi-process CPU tests of the Coordinator shard-merge (fragment.py) over
+the gloo backend, world_size 2 — covers the N>1 distributed path that the
+driver's round-end scaling bench exercises over RCCL."""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _init(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+
+
+def _worker_q1(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    # rank-local partial states, dense slot layout
+    sums = torch.zeros((6, 5), dtype=torch.float64)
+    counts = torch.zeros(6, dtype=torch.int64)
+    sums[0, 0] = 1.5 * (rank + 1)
+    counts[0] = 10 * (rank + 1)
+    sums[3, 2] = 2.0
+    counts[3] = rank  # only rank 1 populates slot 3
+    rows = fragment.merge_q1_partials(sums, counts)
+    if rank == 0:
+        q.put(rows)
+    torch.distributed.destroy_process_group()
+
+
+def _worker_allgatherv(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    t = torch.arange(3 + 2 * rank, dtype=torch.int64) + 100 * rank
+    out = fragment.allgather_variable(t)
+    if rank == 0:
+        q.put(out.numpy())
+    torch.distributed.destroy_process_group()
+
+
+def _worker_topk(rank, world, port, q):
+    from opentenbase_amd import fragment
+    dt = np.dtype([("l_orderkey", "i8"), ("revenue", "f8"),
+                   ("o_orderdate", "i4"), ("o_shippriority", "i4")])
+    _init(rank, world, port)
+    cands = np.zeros(3, dtype=dt)
+    cands["l_orderkey"] = np.arange(3) + 10 * rank
+    cands["revenue"] = [5.0 + rank, 1.0, 3.0 + 2 * rank]
+    top = fragment.merge_q3_topk(cands, k=4)
+    if rank == 0:
+        q.put(top)
+    torch.distributed.destroy_process_group()
+
+
+def _run(worker):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29511 + np.random.randint(0, 400)
+    procs = [ctx.Process(target=worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = q.get(timeout=120)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    return res
+
+
+@pytest.mark.timeout(180)
+def test_merge_q1_partials_two_ranks():
+    rows = _run(_worker_q1)
+    # slot 0 = (A,F): counts 10+20, sums 1.5+3.0; slot 3 = (N,O): count 0+1
+    af = [r for r in rows if r["l_returnflag"] == "A"][0]
+    assert af["count_order"] == 30
+    assert af["sum_qty"] == 4.5
+    no = [r for r in rows if r["l_returnflag"] == "N"][0]
+    assert no["count_order"] == 1
+    assert no["sum_disc_price"] == 4.0  # 2.0 from each rank's tensor
+
+
+@pytest.mark.timeout(180)
+def test_allgather_variable():
+    out = _run(_worker_allgatherv)
+    assert out.tolist() == [0, 1, 2, 100, 101, 102, 103, 104]
+
+
+@pytest.mark.timeout(180)
+def test_merge_q3_topk():
+    top = _run(_worker_topk)
+    assert top["revenue"].tolist() == [6.0, 5.0, 5.0, 3.0]
