@@ -46,14 +46,14 @@ def log(*a):
     print(*a, file=sys.stderr, flush=True)
 
 
-def load_traffic(workload):
+def load_traffic(workload, sf):
     p = os.path.join(REPO, "profiles", "roofline_traffic.json")
     if not os.path.exists(p):
         return None
     try:
         with open(p) as f:
             j = json.load(f)
-        if j.get("workload") == workload:
+        if j.get("workload") == workload and j.get("sf_per_gpu") == sf:
             return float(j["traffic_bytes_per_launch"])
     except Exception:
         pass
@@ -89,7 +89,8 @@ def main():
     ap.add_argument("--workload", default="tpch_q1",
                     choices=["tpch_q1", "scan_count", "tpch_q3"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
-    ap.add_argument("--cpu-sample-rows", type=int, default=240_000_000)
+    ap.add_argument("--cpu-sample-rows", type=int, default=1_200_000_000,
+                    help="~10 s of scalar CPU work on the target host")
     args = ap.parse_args()
 
     import torch
@@ -197,7 +198,7 @@ def main():
     roofline = None
     if kmean_ms:
         achieved = rows_per_gpu * bytes_per_row / (kmean_ms / 1e3)  # B/s, per GPU
-        traffic = load_traffic(args.workload)
+        traffic = load_traffic(args.workload, args.sf)
         roofline = {
             "bound": "hbm",
             "achieved": achieved / 1e9,
