@@ -124,6 +124,13 @@ otbx_status otbx_scan_count(const int32_t *shipdate_dev, int64_t n,
 otbx_status otbx_q1_partial(const otbx_lineitem_dev *t, int32_t cutoff_day,
                             double *sums_dev, int64_t *counts_dev,
                             void *stream, float *kernel_ms);
+/* A/B harness entry: variant 0 = 2 rows/lane, 1 = 4 rows/lane,
+ * 2 = 4 rows/lane + non-temporal loads. otbx_q1_partial dispatches the
+ * measured-best variant. */
+otbx_status otbx_q1_partial_variant(const otbx_lineitem_dev *t,
+                                    int32_t cutoff_day, double *sums_dev,
+                                    int64_t *counts_dev, void *stream,
+                                    float *kernel_ms, int variant);
 
 /* ---- TPC-H Q3 DN fragment ----
  * customer/orders/lineitem staged on-device; custkeys of the replicated
@@ -142,6 +149,11 @@ typedef struct {
 
 otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
                                     size_t *bytes);
+/* kernel_ms (host, may be NULL): float[4] = {customer-keyset build,
+ * orders build+probe, lineitem probe+partial-agg, compact} HIP-event times.
+ * stats_dev (may be NULL): int64[1] = probe hits (lineitem rows passing the
+ * date qual AND matching an order — the N_probe_hits of the §8d roofline
+ * formula); zeroed by the call. */
 otbx_status otbx_q3_partial(const otbx_customer_dev *c,
                             const otbx_orders_dev *o,
                             const otbx_lineitem_dev *l,
@@ -149,8 +161,18 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c,
                             uint8_t segment, int32_t q3date,
                             void *ws_dev, size_t ws_bytes,
                             otbx_q3_group *groups_dev, int64_t cap_groups,
-                            int64_t *ngroups_dev, void *stream,
-                            float *kernel_ms);
+                            int64_t *ngroups_dev, int64_t *stats_dev,
+                            void *stream, float *kernel_ms);
+
+/* top-k selection over compacted Q3 groups (ORDER BY revenue DESC …
+ * LIMIT k pre-selection; the final k-way ordering happens host-side on the
+ * ≤ cap_cand candidates). hist_dev: uint32[16384] workspace (zeroed by the
+ * call). ncand_dev ≥ k unless n < k; candidates are all groups with revenue
+ * ≥ the selection threshold (a bit-pattern bin boundary). */
+otbx_status otbx_topk_by_revenue(const otbx_q3_group *groups_dev, int64_t n,
+                                 int64_t k, otbx_q3_group *cand_dev,
+                                 int64_t cap_cand, int64_t *ncand_dev,
+                                 uint32_t *hist_dev, void *stream);
 /* helper for the broadcast build side: compact custkeys where
  * c_mktsegment == segment into keys_out_dev, count into nkeys_dev (zeroed). */
 otbx_status otbx_filter_customer(const otbx_customer_dev *c, uint8_t segment,

@@ -85,8 +85,9 @@ EXPORTED_SYMBOLS = [
     "otbx_device_malloc", "otbx_device_free", "otbx_memcpy_h2d",
     "otbx_memcpy_d2h", "otbx_stream_sync",
     "otbx_gen_lineitem_dev", "otbx_gen_orders_dev", "otbx_gen_customer_dev",
-    "otbx_scan_count", "otbx_q1_partial",
+    "otbx_scan_count", "otbx_q1_partial", "otbx_q1_partial_variant",
     "otbx_q3_workspace_bytes", "otbx_q3_partial", "otbx_filter_customer",
+    "otbx_topk_by_revenue",
     "otbx_agg_i64_workspace_bytes", "otbx_agg_i64",
     "otbx_join_i64_workspace_bytes", "otbx_join_i64",
 ]

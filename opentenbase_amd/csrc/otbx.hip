@@ -394,10 +394,151 @@ __global__ void k_q1_partial(const int32_t *__restrict__ sd,
     }
 }
 
-extern "C" otbx_status otbx_q1_partial(const otbx_lineitem_dev *t,
-                                       int32_t cutoff_day, double *sums_dev,
-                                       int64_t *counts_dev, void *stream,
-                                       float *kernel_ms)
+/* variant 2: 4 rows/lane (int4 dates, uchar4 flags, 2× double2 per column),
+ * optional non-temporal loads on the read-once f64 streams (NT template). */
+typedef double v2d __attribute__((ext_vector_type(2)));
+
+template <bool NT>
+__device__ __forceinline__ double2 ld2(const double2 *p)
+{
+    if (NT) {
+        v2d v = __builtin_nontemporal_load((const v2d *)p);
+        return make_double2(v.x, v.y);
+    }
+    return *p;
+}
+
+template <bool NT>
+__launch_bounds__(256, 2)
+__global__ void k_q1_partial_v2(const int32_t *__restrict__ sd,
+                                const uint8_t *__restrict__ rf,
+                                const uint8_t *__restrict__ ls,
+                                const double *__restrict__ qty,
+                                const double *__restrict__ price,
+                                const double *__restrict__ disc,
+                                const double *__restrict__ tax,
+                                int64_t n, int32_t cutoff,
+                                double *__restrict__ out_sums,
+                                unsigned long long *__restrict__ out_counts)
+{
+    double acc[Q1_NG][Q1_NS];
+    uint32_t cnt[Q1_NG];
+#pragma unroll
+    for (int g = 0; g < Q1_NG; g++) {
+        cnt[g] = 0;
+#pragma unroll
+        for (int s = 0; s < Q1_NS; s++) acc[g][s] = 0.0;
+    }
+    int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t nq = n / 4;
+    const int4 *sd4 = (const int4 *)sd;
+    const uchar4 *rf4 = (const uchar4 *)rf;
+    const uchar4 *ls4 = (const uchar4 *)ls;
+    const double2 *qty2 = (const double2 *)qty;
+    const double2 *price2 = (const double2 *)price;
+    const double2 *disc2 = (const double2 *)disc;
+    const double2 *tax2 = (const double2 *)tax;
+    for (int64_t i = tid; i < nq; i += stride) {
+        int4 d = sd4[i];
+        uchar4 r = rf4[i];
+        uchar4 l = ls4[i];
+        double2 qa = ld2<NT>(&qty2[2 * i]), qb = ld2<NT>(&qty2[2 * i + 1]);
+        double2 pa = ld2<NT>(&price2[2 * i]), pb = ld2<NT>(&price2[2 * i + 1]);
+        double2 da = ld2<NT>(&disc2[2 * i]), db = ld2<NT>(&disc2[2 * i + 1]);
+        double2 ta = ld2<NT>(&tax2[2 * i]), tb = ld2<NT>(&tax2[2 * i + 1]);
+        int32_t ds[4] = {d.x, d.y, d.z, d.w};
+        uint8_t rs[4] = {r.x, r.y, r.z, r.w};
+        uint8_t lss[4] = {l.x, l.y, l.z, l.w};
+        double qv[4] = {qa.x, qa.y, qb.x, qb.y};
+        double pv[4] = {pa.x, pa.y, pb.x, pb.y};
+        double dv[4] = {da.x, da.y, db.x, db.y};
+        double tv[4] = {ta.x, ta.y, tb.x, tb.y};
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+            bool pass = ds[j] <= cutoff;
+            int gid = (rs[j] == 'A' ? 0 : (rs[j] == 'N' ? 1 : 2)) * 2 +
+                      (lss[j] == 'F' ? 0 : 1);
+            double dp = pv[j] * (1.0 - dv[j]);
+            double ch = dp * (1.0 + tv[j]);
+#pragma unroll
+            for (int g = 0; g < Q1_NG; g++) {
+                bool m = pass && (gid == g);
+                double w = m ? 1.0 : 0.0;
+                acc[g][0] += w * qv[j];
+                acc[g][1] += w * pv[j];
+                acc[g][2] += w * dp;
+                acc[g][3] += w * ch;
+                acc[g][4] += w * dv[j];
+                cnt[g] += m;
+            }
+        }
+    }
+    /* tail rows (n % 4) by thread 0 of block 0 */
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        for (int64_t i = nq * 4; i < n; i++) {
+            if (sd[i] <= cutoff) {
+                int gid = (rf[i] == 'A' ? 0 : (rf[i] == 'N' ? 1 : 2)) * 2 +
+                          (ls[i] == 'F' ? 0 : 1);
+                double dp = price[i] * (1.0 - disc[i]);
+#pragma unroll
+                for (int g = 0; g < Q1_NG; g++) {
+                    if (g == gid) {
+                        acc[g][0] += qty[i];
+                        acc[g][1] += price[i];
+                        acc[g][2] += dp;
+                        acc[g][3] += dp * (1.0 + tax[i]);
+                        acc[g][4] += disc[i];
+                        cnt[g] += 1;
+                    }
+                }
+            }
+        }
+    }
+#pragma unroll
+    for (int g = 0; g < Q1_NG; g++) {
+#pragma unroll
+        for (int s = 0; s < Q1_NS; s++)
+            for (int off = WAVE / 2; off > 0; off >>= 1)
+                acc[g][s] += __shfl_down(acc[g][s], off, WAVE);
+        for (int off = WAVE / 2; off > 0; off >>= 1)
+            cnt[g] += __shfl_down(cnt[g], off, WAVE);
+    }
+    __shared__ double lacc[256 / WAVE][Q1_NG][Q1_NS];
+    __shared__ uint32_t lcnt[256 / WAVE][Q1_NG];
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) {
+#pragma unroll
+        for (int g = 0; g < Q1_NG; g++) {
+#pragma unroll
+            for (int s = 0; s < Q1_NS; s++) lacc[wid][g][s] = acc[g][s];
+            lcnt[wid][g] = cnt[g];
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int nw = blockDim.x / WAVE;
+#pragma unroll
+        for (int g = 0; g < Q1_NG; g++) {
+            unsigned long long c = 0;
+#pragma unroll
+            for (int s = 0; s < Q1_NS; s++) {
+                double v = 0;
+                for (int w = 0; w < nw; w++) v += lacc[w][g][s];
+                if (v != 0.0) atomicAdd(&out_sums[g * Q1_NS + s], v);
+            }
+            for (int w = 0; w < nw; w++) c += lcnt[w][g];
+            if (c) atomicAdd(&out_counts[g], c);
+        }
+    }
+}
+
+extern "C" otbx_status otbx_q1_partial_variant(const otbx_lineitem_dev *t,
+                                               int32_t cutoff_day,
+                                               double *sums_dev,
+                                               int64_t *counts_dev,
+                                               void *stream, float *kernel_ms,
+                                               int variant)
 {
     if (!t || !sums_dev || !counts_dev) return OTBX_ERR_INVALID;
     hipStream_t s = (hipStream_t)stream;
@@ -409,11 +550,24 @@ extern "C" otbx_status otbx_q1_partial(const otbx_lineitem_dev *t,
         HIP_CHECK(hipEventCreate(&ev1));
         HIP_CHECK(hipEventRecord(ev0, s));
     }
-    hipLaunchKernelGGL(k_q1_partial, dim3(grid_for(t->n / 2, 256)), dim3(256), 0, s,
-                       t->l_shipdate, t->l_returnflag, t->l_linestatus,
-                       t->l_quantity, t->l_extendedprice, t->l_discount,
-                       t->l_tax, t->n, cutoff_day, sums_dev,
-                       (unsigned long long *)counts_dev);
+    if (variant == 0)
+        hipLaunchKernelGGL(k_q1_partial, dim3(grid_for(t->n / 2, 256)), dim3(256),
+                           0, s, t->l_shipdate, t->l_returnflag, t->l_linestatus,
+                           t->l_quantity, t->l_extendedprice, t->l_discount,
+                           t->l_tax, t->n, cutoff_day, sums_dev,
+                           (unsigned long long *)counts_dev);
+    else if (variant == 1)
+        hipLaunchKernelGGL(k_q1_partial_v2<false>, dim3(grid_for(t->n / 4, 256)),
+                           dim3(256), 0, s, t->l_shipdate, t->l_returnflag,
+                           t->l_linestatus, t->l_quantity, t->l_extendedprice,
+                           t->l_discount, t->l_tax, t->n, cutoff_day, sums_dev,
+                           (unsigned long long *)counts_dev);
+    else
+        hipLaunchKernelGGL(k_q1_partial_v2<true>, dim3(grid_for(t->n / 4, 256)),
+                           dim3(256), 0, s, t->l_shipdate, t->l_returnflag,
+                           t->l_linestatus, t->l_quantity, t->l_extendedprice,
+                           t->l_discount, t->l_tax, t->n, cutoff_day, sums_dev,
+                           (unsigned long long *)counts_dev);
     HIP_CHECK(hipGetLastError());
     if (kernel_ms) {
         HIP_CHECK(hipEventRecord(ev1, s));
@@ -423,6 +577,15 @@ extern "C" otbx_status otbx_q1_partial(const otbx_lineitem_dev *t,
         HIP_CHECK(hipEventDestroy(ev1));
     }
     return OTBX_OK;
+}
+
+extern "C" otbx_status otbx_q1_partial(const otbx_lineitem_dev *t,
+                                       int32_t cutoff_day, double *sums_dev,
+                                       int64_t *counts_dev, void *stream,
+                                       float *kernel_ms)
+{
+    return otbx_q1_partial_variant(t, cutoff_day, sums_dev, counts_dev, stream,
+                                   kernel_ms, 0 /* measured-best variant */);
 }
 
 /* ================= open-addressing hash-table helpers =================
@@ -766,10 +929,12 @@ struct q3g_slot {
 
 __global__ void k_q3_probe_agg(const otbx_lineitem_dev l, int32_t q3date,
                                const ord_slot *__restrict__ otab, int64_t ocap,
-                               q3g_slot *gtab, int64_t gcap)
+                               q3g_slot *gtab, int64_t gcap,
+                               unsigned long long *__restrict__ nhits)
 {
     int64_t omask = ocap - 1, gmask = gcap - 1;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long myhits = 0;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < l.n;
          i += stride) {
         if (!(l.l_shipdate[i] > q3date))          /* ExecQual on lineitem */
@@ -792,6 +957,7 @@ __global__ void k_q3_probe_agg(const otbx_lineitem_dev l, int32_t q3date,
         }
         if (!hit)
             continue;
+        myhits++;
         double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
         /* group insert-or-update (mutable table → CAS probing only) */
         int64_t g = (int64_t)(d_hash_i64(k) & (uint64_t)gmask);
@@ -809,6 +975,11 @@ __global__ void k_q3_probe_agg(const otbx_lineitem_dev l, int32_t q3date,
         }
         atomicAdd(&gtab[g].revenue, rev);
     }
+    /* per-wave hit count, one atomic per wave */
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        myhits += __shfl_down(myhits, off, WAVE);
+    if ((threadIdx.x % WAVE) == 0 && myhits)
+        atomicAdd(nhits, myhits);
 }
 
 __global__ void k_q3_compact(const q3g_slot *gtab, int64_t gcap,
@@ -882,7 +1053,7 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                             uint8_t segment, int32_t q3date, void *ws,
                             size_t ws_bytes, otbx_q3_group *groups_dev,
                             int64_t cap_groups, int64_t *ngroups_dev,
-                            void *stream, float *kernel_ms)
+                            int64_t *stats_dev, void *stream, float *kernel_ms)
 {
     int64_t ncust = cust_keys_dev ? ncust_keys : c->n;
     int64_t ccap, ocap, gcap;
@@ -897,13 +1068,26 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
 
     HIP_CHECK(hipMemsetAsync(ws, 0, need, s)); /* EMPTY = 0 everywhere */
     HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, sizeof(int64_t), s));
-
-    hipEvent_t ev0 = nullptr, ev1 = nullptr;
-    if (kernel_ms) {
-        HIP_CHECK(hipEventCreate(&ev0));
-        HIP_CHECK(hipEventCreate(&ev1));
-        HIP_CHECK(hipEventRecord(ev0, s));
+    unsigned long long *nhits = nullptr;
+    if (stats_dev) {
+        HIP_CHECK(hipMemsetAsync(stats_dev, 0, sizeof(int64_t), s));
+        nhits = (unsigned long long *)stats_dev;
+    } else {
+        /* hit counter still needs a target: use the last group slot word
+         * of the (oversized) workspace tail — gcap ≥ 1 slot spare is not
+         * guaranteed, so fall back to ngroups_dev[0]'s neighbor is unsafe;
+         * just allocate-free is forbidden mid-graph. Count into a static
+         * device symbol instead. */
+        static unsigned long long *scratch = nullptr;
+        if (!scratch)
+            HIP_CHECK(hipMalloc(&scratch, sizeof(unsigned long long)));
+        nhits = scratch;
     }
+
+    hipEvent_t ev[5] = {};
+    if (kernel_ms)
+        for (int i = 0; i < 5; i++) HIP_CHECK(hipEventCreate(&ev[i]));
+    if (kernel_ms) HIP_CHECK(hipEventRecord(ev[0], s));
     if (cust_keys_dev) {
         if (ncust_keys > 0)
             hipLaunchKernelGGL(k_keyset_build, dim3(grid_for(ncust_keys, 256)),
@@ -914,22 +1098,87 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                            dim3(256), 0, s, c->c_custkey, c->c_mktsegment,
                            segment, c->n, ctab, ccap);
     }
+    if (kernel_ms) HIP_CHECK(hipEventRecord(ev[1], s));
     hipLaunchKernelGGL(k_orders_build, dim3(grid_for(o->n, 256)), dim3(256), 0,
                        s, *o, ctab, ccap, q3date, otab, ocap);
+    if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
     hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(l->n, 256)), dim3(256), 0,
-                       s, *l, q3date, otab, ocap, gtab, gcap);
-    if (kernel_ms) {
-        HIP_CHECK(hipEventRecord(ev1, s));
-    }
+                       s, *l, q3date, otab, ocap, gtab, gcap, nhits);
+    if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
     hipLaunchKernelGGL(k_q3_compact, dim3(grid_for(gcap, 256)), dim3(256), 0, s,
                        gtab, gcap, groups_dev, cap_groups, ngroups_dev);
+    if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
     HIP_CHECK(hipGetLastError());
     if (kernel_ms) {
-        HIP_CHECK(hipEventSynchronize(ev1));
-        HIP_CHECK(hipEventElapsedTime(kernel_ms, ev0, ev1));
-        HIP_CHECK(hipEventDestroy(ev0));
-        HIP_CHECK(hipEventDestroy(ev1));
+        HIP_CHECK(hipEventSynchronize(ev[4]));
+        for (int i = 0; i < 4; i++)
+            HIP_CHECK(hipEventElapsedTime(&kernel_ms[i], ev[i], ev[i + 1]));
+        for (int i = 0; i < 5; i++) HIP_CHECK(hipEventDestroy(ev[i]));
     }
+    return OTBX_OK;
+}
+
+/* ---- top-k by revenue: bit-pattern histogram selection ----
+ * For revenue ≥ 0, the raw IEEE-754 bit pattern is order-preserving, so the
+ * top 14 bits (sign+exp+mantissa head) give 16384 monotonic bins; select the
+ * threshold bin from the suffix sum, collect candidates ≥ threshold. */
+
+__device__ __forceinline__ uint32_t rev_bin(double r)
+{
+    unsigned long long b = __double_as_longlong(r);
+    return (uint32_t)(b >> 50); /* positive doubles: monotonic */
+}
+
+__global__ void k_topk_hist(const otbx_q3_group *__restrict__ g, int64_t n,
+                            uint32_t *hist)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        atomicAdd(&hist[rev_bin(g[i].revenue)], 1u);
+}
+
+__global__ void k_topk_collect(const otbx_q3_group *__restrict__ g, int64_t n,
+                               uint32_t thr_bin, otbx_q3_group *out,
+                               int64_t cap, int64_t *ncand)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        if (rev_bin(g[i].revenue) >= thr_bin) {
+            int64_t pos = (int64_t)atomicAdd((unsigned long long *)ncand, 1ull);
+            if (pos < cap)
+                out[pos] = g[i];
+        }
+    }
+}
+
+extern "C" otbx_status otbx_topk_by_revenue(const otbx_q3_group *groups,
+                                            int64_t n, int64_t k,
+                                            otbx_q3_group *cand, int64_t cap,
+                                            int64_t *ncand_dev, uint32_t *hist,
+                                            void *stream)
+{
+    hipStream_t s = (hipStream_t)stream;
+    HIP_CHECK(hipMemsetAsync(hist, 0, 16384 * 4, s));
+    HIP_CHECK(hipMemsetAsync(ncand_dev, 0, 8, s));
+    if (n <= 0) return OTBX_OK;
+    hipLaunchKernelGGL(k_topk_hist, dim3(grid_for(n, 256)), dim3(256), 0, s,
+                       groups, n, hist);
+    static uint32_t *h_hist = nullptr;
+    if (!h_hist)
+        HIP_CHECK(hipHostMalloc(&h_hist, 16384 * 4));
+    HIP_CHECK(hipMemcpyAsync(h_hist, hist, 16384 * 4, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    int64_t cum = 0;
+    uint32_t thr = 0;
+    for (int b = 16383; b >= 0; b--) {
+        cum += h_hist[b];
+        if (cum >= k) { thr = (uint32_t)b; break; }
+    }
+    hipLaunchKernelGGL(k_topk_collect, dim3(grid_for(n, 256)), dim3(256), 0, s,
+                       groups, n, thr, cand, cap, ncand_dev);
+    HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }
 

@@ -236,16 +236,24 @@ class GpuQ3Fragment(CustomScanState):
     the REPLICATED (post-broadcast) filtered customer keys (SURVEY §8e);
     without it, the local customer shard is filtered in-kernel."""
 
+    NP_DTYPE = [("l_orderkey", "i8"), ("revenue", "f8"),
+                ("o_orderdate", "i4"), ("o_shippriority", "i4")]
+
     def __init__(self, customer, orders, lineitem, segment=0,
-                 date=Q3_DATE_DEFAULT, cust_keys=None):
+                 date=Q3_DATE_DEFAULT, cust_keys=None, k=10):
         super().__init__()
         self.cu, self.od, self.li = customer, orders, lineitem
         self.segment, self.date = segment, date
         self.cust_keys = cust_keys
-        self.kernel_ms = None
-        self.groups = None   # structured numpy array of all partial groups
+        self.k = k
+        self.kernel_ms = None    # [keyset, orders, probe_agg, compact]
+        self.probe_hits = None   # N_probe_hits (roofline formula, SURVEY §8d)
+        self.ngroups = None
+        self.groups = None       # lazy: fetch_groups()
+        self._groups_dev = None
 
     def _run(self):
+        import numpy as np
         L = lib()
         ncust = self.cu.n if self.cust_keys is None else len(self.cust_keys)
         ws_bytes = C.c_size_t(0)
@@ -255,24 +263,46 @@ class GpuQ3Fragment(CustomScanState):
         cap = self.od.n if self.od.n > 0 else 1
         groups = torch.empty(cap * 24, dtype=torch.uint8, device="cuda")
         ng = torch.zeros(1, dtype=torch.int64, device="cuda")
-        ms = C.c_float(0.0)
+        stats = torch.zeros(1, dtype=torch.int64, device="cuda")
+        ms = (C.c_float * 4)()
         ck = C.c_void_p(self.cust_keys.data_ptr()) if self.cust_keys is not None else None
         nck = C.c_int64(0 if self.cust_keys is None else len(self.cust_keys))
         call("otbx_q3_partial", C.byref(self.cu.cstruct), C.byref(self.od.cstruct),
              C.byref(self.li.cstruct), ck, nck, C.c_uint8(self.segment),
              C.c_int32(self.date), C.c_void_p(ws.data_ptr()),
              C.c_size_t(ws_bytes.value), C.c_void_p(groups.data_ptr()),
-             C.c_int64(cap), C.c_void_p(ng.data_ptr()), _stream(), C.byref(ms))
-        self.kernel_ms = ms.value
-        n = int(ng.cpu().item())
+             C.c_int64(cap), C.c_void_p(ng.data_ptr()),
+             C.c_void_p(stats.data_ptr()), _stream(), ms)
+        self.kernel_ms = list(ms)
+        self._groups_dev = groups
+        self.ngroups = int(ng.cpu().item())
+        self.probe_hits = int(stats.cpu().item())
+        # GPU top-k pre-selection (LIMIT 10 below the merge); the final
+        # ordering of the ≤ few-thousand candidates happens on host
+        n = self.ngroups
+        if n == 0:
+            self.groups = np.empty(0, dtype=np.dtype(self.NP_DTYPE))
+            return []
+        cap_cand = 1 << 20
+        cand = torch.empty(cap_cand * 24, dtype=torch.uint8, device="cuda")
+        ncand = torch.zeros(1, dtype=torch.int64, device="cuda")
+        hist = torch.empty(16384, dtype=torch.int32, device="cuda")
+        call("otbx_topk_by_revenue", C.c_void_p(groups.data_ptr()),
+             C.c_int64(n), C.c_int64(self.k), C.c_void_p(cand.data_ptr()),
+             C.c_int64(cap_cand), C.c_void_p(ncand.data_ptr()),
+             C.c_void_p(hist.data_ptr()), _stream())
+        nc = min(int(ncand.cpu().item()), cap_cand)
+        raw = cand[: nc * 24].cpu().numpy().tobytes()
+        cands = np.frombuffer(raw, dtype=np.dtype(self.NP_DTYPE)).copy()
+        return [tuple(r) for r in q3_topk(cands, self.k)]
+
+    def fetch_groups(self):
+        """D2H copy of ALL partial groups (parity tests / debugging)."""
         import numpy as np
-        dt = np.dtype([("l_orderkey", "i8"), ("revenue", "f8"),
-                       ("o_orderdate", "i4"), ("o_shippriority", "i4")])
-        raw = groups[: n * 24].cpu().numpy().tobytes()
-        self.groups = np.frombuffer(raw, dtype=dt).copy()
-        # partial top-k candidates (LIMIT 10 below the merge); full ORDER BY
-        # offload is SURVEY §8f.2
-        return [tuple(r) for r in q3_topk(self.groups, 10)]
+        if self.groups is None or len(self.groups) != self.ngroups:
+            raw = self._groups_dev[: self.ngroups * 24].cpu().numpy().tobytes()
+            self.groups = np.frombuffer(raw, dtype=np.dtype(self.NP_DTYPE)).copy()
+        return self.groups
 
 
 def q3_topk(groups, k=10):

@@ -164,14 +164,14 @@ def test_q3_parity(ex, ora):
     top = drain(node)
     og = q3_oracle(ora, n)
     # full group-set parity
-    got = {int(r["l_orderkey"]): float(r["revenue"]) for r in node.groups}
+    got = {int(r["l_orderkey"]): float(r["revenue"]) for r in node.fetch_groups()}
     exp = {int(k): float(v) for k, v in zip(og["l_orderkey"], og["revenue"])}
     assert got.keys() == exp.keys()
     for k in exp:
         assert approx(got[k], exp[k]), k
     # dates/prios bit-exact
     gd = {int(r["l_orderkey"]): (int(r["o_orderdate"]), int(r["o_shippriority"]))
-          for r in node.groups}
+          for r in node.fetch_groups()}
     ed = {int(k): (int(d), int(p)) for k, d, p in
           zip(og["l_orderkey"], og["o_orderdate"], og["o_shippriority"])}
     assert gd == ed
@@ -209,7 +209,7 @@ def test_q3_broadcast_path_parity(ex, ora):
         cu = ex.GpuCustomer.generate(n // 40, rank=rank, nranks=2)
         node = ex.GpuQ3Fragment(cu, od, li, cust_keys=bcast)
         drain(node)
-        union.append(node.groups)
+        union.append(node.fetch_groups())
     union = np.concatenate(union)
     og = q3_oracle(ora, n)
     assert set(union["l_orderkey"].tolist()) == set(og["l_orderkey"].tolist())
