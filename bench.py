@@ -152,16 +152,18 @@ def main():
                  stream)
             return int(out.cpu().item())
         else:
-            import ctypes as CT
-            from opentenbase_amd._lib import call
-            keys = torch.empty(cu.n, dtype=torch.int64, device="cuda")
-            nk = torch.zeros(1, dtype=torch.int64, device="cuda")
-            call("otbx_filter_customer", CT.byref(cu.cstruct), CT.c_uint8(0),
-                 CT.c_void_p(keys.data_ptr()), CT.c_void_p(nk.data_ptr()), stream)
-            local = keys[: int(nk.cpu().item())]
-            bcast = fragment.broadcast_customer_keys(local)
-            node = ex.GpuQ3Fragment(cu, od, li,
-                                    cust_keys=bcast if world > 1 else None)
+            bcast = None
+            if world > 1:
+                import ctypes as CT
+                from opentenbase_amd._lib import call
+                keys = torch.empty(cu.n, dtype=torch.int64, device="cuda")
+                nk = torch.zeros(1, dtype=torch.int64, device="cuda")
+                call("otbx_filter_customer", CT.byref(cu.cstruct), CT.c_uint8(0),
+                     CT.c_void_p(keys.data_ptr()), CT.c_void_p(nk.data_ptr()),
+                     stream)
+                local = keys[: int(nk.cpu().item())]
+                bcast = fragment.broadcast_customer_keys(local)
+            node = ex.GpuQ3Fragment(cu, od, li, cust_keys=bcast)
             node.BeginCustomScan()
             node._rows = node._run()
             # probe+partial-agg kernel is the roofline-dominant phase

@@ -584,8 +584,11 @@ extern "C" otbx_status otbx_q1_partial(const otbx_lineitem_dev *t,
                                        int64_t *counts_dev, void *stream,
                                        float *kernel_ms)
 {
+    /* variant 2 (4 rows/lane + non-temporal loads) measured best:
+     * 3.85 ms vs 4.25 ms (v0) at SF100 — within-probe interleaved A/B,
+     * 8 rounds (profiles/r01_q1_ab.txt) */
     return otbx_q1_partial_variant(t, cutoff_day, sums_dev, counts_dev, stream,
-                                   kernel_ms, 0 /* measured-best variant */);
+                                   kernel_ms, 2);
 }
 
 /* ================= open-addressing hash-table helpers =================
@@ -604,6 +607,24 @@ extern "C" otbx_status otbx_q1_partial(const otbx_lineitem_dev *t,
 __device__ __forceinline__ uint64_t d_hash_i64(int64_t k)
 {
     return otbx_splitmix64((uint64_t)k);
+}
+
+/* wave-aggregated append: one atomicAdd per wave instead of per lane
+ * (Guideline 12). Returns this lane's output slot, or -1 if !pred.
+ * blockDim must be a multiple of 64. */
+__device__ __forceinline__ int64_t wave_append(int64_t *counter, bool pred)
+{
+    unsigned long long mask = __ballot(pred);
+    if (!mask) return -1;
+    int lane = (int)(threadIdx.x % WAVE);
+    int leader = __ffsll((long long)mask) - 1;
+    int rank = __popcll(mask & ((1ull << lane) - 1ull));
+    long long base = 0;
+    if (lane == leader)
+        base = (long long)atomicAdd((unsigned long long *)counter,
+                                    (unsigned long long)__popcll(mask));
+    base = __shfl(base, leader, WAVE);
+    return pred ? base + rank : -1;
 }
 
 static inline int64_t next_pow2_host(int64_t v)
@@ -679,8 +700,9 @@ __global__ void k_agg_compact(const agg_slot *tab, int64_t cap,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < cap;
          i += stride) {
-        if (tab[i].key != AGG_EMPTY) {
-            int64_t pos = (int64_t)atomicAdd((unsigned long long *)ngroups, 1ull);
+        bool used = tab[i].key != AGG_EMPTY;
+        int64_t pos = wave_append(ngroups, used);
+        if (used) {
             out[pos].key = tab[i].key;
             out[pos].key_isnull = 0;
             out[pos].count_star = (int64_t)tab[i].count_star;
@@ -989,14 +1011,13 @@ __global__ void k_q3_compact(const q3g_slot *gtab, int64_t gcap,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < gcap;
          i += stride) {
-        if (gtab[i].okey != 0ull) {
-            int64_t pos = (int64_t)atomicAdd((unsigned long long *)ngroups, 1ull);
-            if (pos < cap_out) {
-                out[pos].l_orderkey = (int64_t)gtab[i].okey;
-                out[pos].revenue = gtab[i].revenue;
-                out[pos].o_orderdate = gtab[i].date;
-                out[pos].o_shippriority = gtab[i].prio;
-            }
+        bool used = gtab[i].okey != 0ull;
+        int64_t pos = wave_append(ngroups, used);
+        if (used && pos < cap_out) {
+            out[pos].l_orderkey = (int64_t)gtab[i].okey;
+            out[pos].revenue = gtab[i].revenue;
+            out[pos].o_orderdate = gtab[i].date;
+            out[pos].o_shippriority = gtab[i].prio;
         }
     }
 }
@@ -1008,11 +1029,46 @@ __global__ void k_filter_customer(const otbx_customer_dev c, uint8_t want,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < c.n;
          i += stride) {
-        if (c.c_mktsegment[i] == want) {
-            int64_t pos = (int64_t)atomicAdd((unsigned long long *)nkeys, 1ull);
+        bool m = c.c_mktsegment[i] == want;
+        int64_t pos = wave_append(nkeys, m);
+        if (m)
             out_keys[pos] = c.c_custkey[i];
-        }
     }
+}
+
+/* count rows passing {segment} / {date + customer-keyset} predicates —
+ * sizes the right-fit hash tables before building (count-then-build). */
+__global__ void k_count_customer_seg(const otbx_customer_dev c, uint8_t want,
+                                     int64_t *count)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long my = 0;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < c.n;
+         i += stride)
+        my += c.c_mktsegment[i] == want;
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        my += __shfl_down(my, off, WAVE);
+    if ((threadIdx.x % WAVE) == 0 && my)
+        atomicAdd((unsigned long long *)count, my);
+}
+
+__global__ void k_count_orders_filtered(const otbx_orders_dev o,
+                                        const unsigned long long *__restrict__ ckeys,
+                                        int64_t ccap, int32_t q3date,
+                                        int64_t *count)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long my = 0;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < o.n;
+         i += stride) {
+        if (o.o_orderdate[i] < q3date &&
+            d_keyset_probe(ckeys, ccap, o.o_custkey[i]))
+            my++;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        my += __shfl_down(my, off, WAVE);
+    if ((threadIdx.x % WAVE) == 0 && my)
+        atomicAdd((unsigned long long *)count, my);
 }
 
 extern "C" {
@@ -1029,21 +1085,19 @@ otbx_status otbx_filter_customer(const otbx_customer_dev *c, uint8_t segment,
     return OTBX_OK;
 }
 
-static void q3_caps(int64_t ncust, int64_t norders, int64_t *ccap,
-                    int64_t *ocap, int64_t *gcap)
+static int64_t fit_cap(int64_t n)
 {
-    *ccap = next_pow2_host(ncust < 16 ? 16 : (int64_t)(ncust / 0.7) + 1);
-    *ocap = next_pow2_host(norders < 16 ? 16 : (int64_t)(norders / 0.7) + 1);
-    *gcap = *ocap; /* groups ⊆ filtered orders */
+    return next_pow2_host(n < 16 ? 16 : (int64_t)(n / 0.7) + 1);
 }
 
 otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
                                     size_t *bytes)
 {
-    int64_t ccap, ocap, gcap;
-    q3_caps(ncust, norders, &ccap, &ocap, &gcap);
-    *bytes = (size_t)ccap * 8 + (size_t)ocap * sizeof(ord_slot) +
-             (size_t)gcap * sizeof(q3g_slot);
+    /* worst case: tables are right-sized at run time (count-then-build),
+     * the caller allocates the no-filter upper bound */
+    int64_t ccap = fit_cap(ncust), ocap = fit_cap(norders);
+    *bytes = 64 + (size_t)ccap * 8 + (size_t)ocap * sizeof(ord_slot) +
+             (size_t)ocap * sizeof(q3g_slot);
     return OTBX_OK;
 }
 
@@ -1056,28 +1110,25 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                             int64_t *stats_dev, void *stream, float *kernel_ms)
 {
     int64_t ncust = cust_keys_dev ? ncust_keys : c->n;
-    int64_t ccap, ocap, gcap;
-    q3_caps(ncust, o->n, &ccap, &ocap, &gcap);
-    size_t need = (size_t)ccap * 8 + (size_t)ocap * sizeof(ord_slot) +
-                  (size_t)gcap * sizeof(q3g_slot);
-    if (ws_bytes < need) return OTBX_ERR_INVALID;
+    {
+        size_t worst;
+        otbx_q3_workspace_bytes(ncust, o->n, &worst);
+        if (ws_bytes < worst) return OTBX_ERR_INVALID;
+    }
     hipStream_t s = (hipStream_t)stream;
-    unsigned long long *ctab = (unsigned long long *)ws;
-    ord_slot *otab = (ord_slot *)((char *)ws + (size_t)ccap * 8);
-    q3g_slot *gtab = (q3g_slot *)((char *)otab + (size_t)ocap * sizeof(ord_slot));
+    int64_t *hdr = (int64_t *)ws;           /* [0]=ncust_f, [1]=nof */
+    char *tabs = (char *)ws + 64;
+    static int64_t *h_cnt = nullptr;        /* pinned host readback */
+    if (!h_cnt)
+        HIP_CHECK(hipHostMalloc(&h_cnt, 2 * sizeof(int64_t)));
 
-    HIP_CHECK(hipMemsetAsync(ws, 0, need, s)); /* EMPTY = 0 everywhere */
+    HIP_CHECK(hipMemsetAsync(hdr, 0, 64, s));
     HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, sizeof(int64_t), s));
     unsigned long long *nhits = nullptr;
     if (stats_dev) {
         HIP_CHECK(hipMemsetAsync(stats_dev, 0, sizeof(int64_t), s));
         nhits = (unsigned long long *)stats_dev;
     } else {
-        /* hit counter still needs a target: use the last group slot word
-         * of the (oversized) workspace tail — gcap ≥ 1 slot spare is not
-         * guaranteed, so fall back to ngroups_dev[0]'s neighbor is unsafe;
-         * just allocate-free is forbidden mid-graph. Count into a static
-         * device symbol instead. */
         static unsigned long long *scratch = nullptr;
         if (!scratch)
             HIP_CHECK(hipMalloc(&scratch, sizeof(unsigned long long)));
@@ -1088,6 +1139,20 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     if (kernel_ms)
         for (int i = 0; i < 5; i++) HIP_CHECK(hipEventCreate(&ev[i]));
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[0], s));
+
+    /* ---- phase 1: size + build the customer keyset (count-then-build) */
+    int64_t ccap;
+    if (cust_keys_dev) {
+        ccap = fit_cap(ncust_keys);
+    } else {
+        hipLaunchKernelGGL(k_count_customer_seg, dim3(grid_for(c->n, 256)),
+                           dim3(256), 0, s, *c, segment, &hdr[0]);
+        HIP_CHECK(hipMemcpyAsync(h_cnt, hdr, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        ccap = fit_cap(h_cnt[0]);
+    }
+    unsigned long long *ctab = (unsigned long long *)tabs;
+    HIP_CHECK(hipMemsetAsync(ctab, 0, (size_t)ccap * 8, s));
     if (cust_keys_dev) {
         if (ncust_keys > 0)
             hipLaunchKernelGGL(k_keyset_build, dim3(grid_for(ncust_keys, 256)),
@@ -1099,12 +1164,28 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                            segment, c->n, ctab, ccap);
     }
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[1], s));
+
+    /* ---- phase 2: size + build the filtered-orders table */
+    hipLaunchKernelGGL(k_count_orders_filtered, dim3(grid_for(o->n, 256)),
+                       dim3(256), 0, s, *o, ctab, ccap, q3date, &hdr[1]);
+    HIP_CHECK(hipMemcpyAsync(h_cnt + 1, hdr + 1, 8, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    int64_t nof = h_cnt[1];
+    int64_t ocap = fit_cap(nof), gcap = ocap;
+    ord_slot *otab = (ord_slot *)(tabs + (size_t)ccap * 8);
+    q3g_slot *gtab = (q3g_slot *)((char *)otab + (size_t)ocap * sizeof(ord_slot));
+    HIP_CHECK(hipMemsetAsync(otab, 0, (size_t)ocap * sizeof(ord_slot) +
+                                          (size_t)gcap * sizeof(q3g_slot), s));
     hipLaunchKernelGGL(k_orders_build, dim3(grid_for(o->n, 256)), dim3(256), 0,
                        s, *o, ctab, ccap, q3date, otab, ocap);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
+
+    /* ---- phase 3: lineitem probe + partial agg */
     hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(l->n, 256)), dim3(256), 0,
                        s, *l, q3date, otab, ocap, gtab, gcap, nhits);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
+
+    /* ---- phase 4: compact groups */
     hipLaunchKernelGGL(k_q3_compact, dim3(grid_for(gcap, 256)), dim3(256), 0, s,
                        gtab, gcap, groups_dev, cap_groups, ngroups_dev);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
@@ -1132,10 +1213,22 @@ __device__ __forceinline__ uint32_t rev_bin(double r)
 __global__ void k_topk_hist(const otbx_q3_group *__restrict__ g, int64_t n,
                             uint32_t *hist)
 {
+    /* per-block LDS histogram: the global bins are few and hot (revenues
+     * cluster in a couple of octaves), device-atomic contention would
+     * serialize — accumulate in LDS, flush nonzero bins once per block */
+    __shared__ uint32_t lh[16384];
+    for (int i = threadIdx.x; i < 16384; i += blockDim.x) lh[i] = 0;
+    __syncthreads();
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride)
-        atomicAdd(&hist[rev_bin(g[i].revenue)], 1u);
+        atomicAdd(&lh[rev_bin(g[i].revenue)], 1u);
+    __syncthreads();
+    for (int i = threadIdx.x; i < 16384; i += blockDim.x) {
+        uint32_t v = lh[i];
+        if (v)
+            atomicAdd(&hist[i], v);
+    }
 }
 
 __global__ void k_topk_collect(const otbx_q3_group *__restrict__ g, int64_t n,
@@ -1145,11 +1238,10 @@ __global__ void k_topk_collect(const otbx_q3_group *__restrict__ g, int64_t n,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride) {
-        if (rev_bin(g[i].revenue) >= thr_bin) {
-            int64_t pos = (int64_t)atomicAdd((unsigned long long *)ncand, 1ull);
-            if (pos < cap)
-                out[pos] = g[i];
-        }
+        bool m = rev_bin(g[i].revenue) >= thr_bin;
+        int64_t pos = wave_append(ncand, m);
+        if (m && pos < cap)
+            out[pos] = g[i];
     }
 }
 
