@@ -148,7 +148,7 @@ typedef struct {
 } otbx_q3_group;
 
 otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
-                                    size_t *bytes);
+                                    int64_t nlineitem, size_t *bytes);
 /* kernel_ms (host, may be NULL): float[4] = {customer-keyset build,
  * orders build+probe, lineitem probe+partial-agg, compact} HIP-event times.
  * stats_dev (may be NULL): int64[1] = probe hits (lineitem rows passing the

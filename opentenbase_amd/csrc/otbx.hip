@@ -976,24 +976,65 @@ struct q3g_slot {
     double revenue;
 };
 
-__global__ void k_q3_probe_agg(const otbx_lineitem_dev l, int32_t q3date,
+/* probe phase A: stream l_shipdate + l_orderkey, qual + bloom test, emit a
+ * COMPACTED candidate row-id list. Without compaction the probe loop is
+ * exec-mask divergent (~5% of lanes active: a 64-lane wave walks the table
+ * for ~3 lanes of work); the dense list gives phase B full lanes and full
+ * memory-level parallelism. Block-aggregated two-phase append, one global
+ * atomic per block, contiguous chunk per block. */
+__global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
+                                 const unsigned long long *__restrict__ bloom,
+                                 int64_t bloom_words,
+                                 int64_t *__restrict__ cand, int64_t *ncand)
+{
+    int64_t per_block = (l.n + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < l.n ? lo + per_block : l.n;
+    __shared__ int64_t tcnt[256];
+    __shared__ int64_t tbase[257];
+    int64_t my = 0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        my += (l.l_shipdate[i] > q3date) &&
+              d_bloom_test(bloom, bloom_words, l.l_orderkey[i]);
+    tcnt[threadIdx.x] = my;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int64_t tot = 0;
+        for (int t = 0; t < (int)blockDim.x; t++) {
+            tbase[t] = tot;
+            tot += tcnt[t];
+        }
+        tbase[256] = tot ? (int64_t)atomicAdd((unsigned long long *)ncand,
+                                              (unsigned long long)tot)
+                         : 0;
+    }
+    __syncthreads();
+    int64_t pos = tbase[256] + tbase[threadIdx.x];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        if ((l.l_shipdate[i] > q3date) &&
+            d_bloom_test(bloom, bloom_words, l.l_orderkey[i]))
+            cand[pos++] = i;
+    }
+}
+
+/* probe phase B: dense over candidates: gather key, probe orders table,
+ * gather payload columns, aggregate into the group table. */
+__global__ void k_q3_probe_agg(const otbx_lineitem_dev l,
+                               const int64_t *__restrict__ cand,
+                               const int64_t *__restrict__ ncand_p,
                                const ord_slot *__restrict__ otab, int64_t ocap,
                                q3g_slot *gtab, int64_t gcap,
-                               const unsigned long long *__restrict__ bloom,
-                               int64_t bloom_words,
                                unsigned long long *__restrict__ nhits)
 {
     int64_t omask = ocap - 1, gmask = gcap - 1;
+    int64_t n = *ncand_p;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     unsigned long long myhits = 0;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < l.n;
-         i += stride) {
-        if (!(l.l_shipdate[i] > q3date))          /* ExecQual on lineitem */
-            continue;
+    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
+         ci += stride) {
+        int64_t i = cand[ci];
         int64_t k = l.l_orderkey[i];
-        if (!d_bloom_test(bloom, bloom_words, k)) /* bloom pre-rejection */
-            continue;
-        /* probe orders table (read-only this launch → plain loads) */
+        /* probe orders table (read-only this launch -> plain loads) */
         int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)omask);
         int32_t date = 0, prio = 0;
         bool hit = false;
@@ -1012,7 +1053,7 @@ __global__ void k_q3_probe_agg(const otbx_lineitem_dev l, int32_t q3date,
             continue;
         myhits++;
         double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
-        /* group insert-or-update (mutable table → CAS probing only) */
+        /* group insert-or-update (mutable table -> CAS probing only) */
         int64_t g = (int64_t)(d_hash_i64(k) & (uint64_t)gmask);
         for (;;) {
             unsigned long long old = atomicCAS(&gtab[g].okey, 0ull,
@@ -1148,13 +1189,14 @@ static int64_t fit_cap(int64_t n)
 }
 
 otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
-                                    size_t *bytes)
+                                    int64_t nlineitem, size_t *bytes)
 {
     /* worst case: tables are right-sized at run time (count-then-build),
      * the caller allocates the no-filter upper bound */
     int64_t ccap = fit_cap(ncust), ocap = fit_cap(norders);
     *bytes = 64 + ((size_t)1 << 23) * 8 /* bloom */ + (size_t)ccap * 8 +
-             (size_t)ocap * sizeof(ord_slot) + (size_t)ocap * sizeof(q3g_slot);
+             (size_t)ocap * sizeof(ord_slot) + (size_t)ocap * sizeof(q3g_slot) +
+             (size_t)nlineitem * 8 /* candidate row ids */;
     return OTBX_OK;
 }
 
@@ -1176,7 +1218,7 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     int64_t ncust = cust_keys_dev ? ncust_keys : c->n;
     {
         size_t worst;
-        otbx_q3_workspace_bytes(ncust, o->n, &worst);
+        otbx_q3_workspace_bytes(ncust, o->n, l->n, &worst);
         if (ws_bytes < worst) return OTBX_ERR_INVALID;
     }
     hipStream_t s = (hipStream_t)stream;
@@ -1247,10 +1289,13 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                        s, *o, ctab, ccap, q3date, otab, ocap, bloom, bwords);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
 
-    /* ---- phase 3: lineitem probe + partial agg */
+    /* ---- phase 3: lineitem scan+filter (compacted candidates), then the
+     * dense probe + partial agg */
+    int64_t *cand = (int64_t *)((char *)gtab + (size_t)gcap * sizeof(q3g_slot));
+    hipLaunchKernelGGL(k_q3_scan_filter, dim3(grid_for(l->n, 256)), dim3(256),
+                       0, s, *l, q3date, bloom, bwords, cand, &hdr[2]);
     hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(l->n, 256)), dim3(256), 0,
-                       s, *l, q3date, otab, ocap, gtab, gcap, bloom, bwords,
-                       nhits);
+                       s, *l, cand, &hdr[2], otab, ocap, gtab, gcap, nhits);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
 
     /* ---- phase 4: compact groups */

@@ -258,7 +258,7 @@ class GpuQ3Fragment(CustomScanState):
         ncust = self.cu.n if self.cust_keys is None else len(self.cust_keys)
         ws_bytes = C.c_size_t(0)
         check(L.otbx_q3_workspace_bytes(C.c_int64(ncust), C.c_int64(self.od.n),
-                                        C.byref(ws_bytes)))
+                                        C.c_int64(self.li.n), C.byref(ws_bytes)))
         ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
         cap = self.od.n if self.od.n > 0 else 1
         groups = torch.empty(cap * 24, dtype=torch.uint8, device="cuda")
