@@ -987,33 +987,44 @@ __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
                                  int64_t bloom_words,
                                  int64_t *__restrict__ cand, int64_t *ncand)
 {
-    int64_t per_block = (l.n + gridDim.x - 1) / gridDim.x;
-    int64_t lo = blockIdx.x * per_block;
-    int64_t hi = lo + per_block < l.n ? lo + per_block : l.n;
-    __shared__ int64_t tcnt[256];
-    __shared__ int64_t tbase[257];
-    int64_t my = 0;
-    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
-        my += (l.l_shipdate[i] > q3date) &&
-              d_bloom_test(bloom, bloom_words, l.l_orderkey[i]);
-    tcnt[threadIdx.x] = my;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-        int64_t tot = 0;
-        for (int t = 0; t < (int)blockDim.x; t++) {
-            tbase[t] = tot;
-            tot += tcnt[t];
+    /* single pass: per-wave LDS staging buffer, one global atomic per 1024
+     * candidates (a direct wave-level append funnels ~500k atomics through
+     * one word; a two-phase block scheme re-streams the columns) */
+    const int BUF = 1024;
+    __shared__ int64_t buf[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0; /* wave-uniform */
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < l.n;
+         i += stride) {
+        bool m = (l.l_shipdate[i] > q3date) &&
+                 d_bloom_test(bloom, bloom_words, l.l_orderkey[i]);
+        unsigned long long mask = __ballot(m);
+        int cnt = __popcll(mask);
+        if (nbuf + cnt > BUF) {
+            long long base = 0;
+            if (lane == 0)
+                base = (long long)atomicAdd((unsigned long long *)ncand,
+                                            (unsigned long long)nbuf);
+            base = __shfl(base, 0, WAVE);
+            for (int j = lane; j < nbuf; j += WAVE)
+                cand[base + j] = buf[wid][j];
+            nbuf = 0;
         }
-        tbase[256] = tot ? (int64_t)atomicAdd((unsigned long long *)ncand,
-                                              (unsigned long long)tot)
-                         : 0;
+        if (m) {
+            int rank = __popcll(mask & ((1ull << lane) - 1ull));
+            buf[wid][nbuf + rank] = i;
+        }
+        nbuf += cnt;
     }
-    __syncthreads();
-    int64_t pos = tbase[256] + tbase[threadIdx.x];
-    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        if ((l.l_shipdate[i] > q3date) &&
-            d_bloom_test(bloom, bloom_words, l.l_orderkey[i]))
-            cand[pos++] = i;
+    if (nbuf) {
+        long long base = 0;
+        if (lane == 0)
+            base = (long long)atomicAdd((unsigned long long *)ncand,
+                                        (unsigned long long)nbuf);
+        base = __shfl(base, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE)
+            cand[base + j] = buf[wid][j];
     }
 }
 
