@@ -943,20 +943,109 @@ __device__ __forceinline__ bool d_bloom_test(const unsigned long long *bloom,
     return (bloom[w] & bits) == bits;
 }
 
-__global__ void k_orders_build(const otbx_orders_dev o,
-                               const unsigned long long *__restrict__ ckeys,
-                               int64_t ccap, int32_t q3date, ord_slot *tab,
-                               int64_t cap, unsigned long long *bloom,
-                               int64_t bloom_words)
+/* orders side, candidate-list pipeline (same rationale as the lineitem
+ * scan): date-filter to a dense row-id list, dense customer-keyset probe to
+ * a matched list (count = table size), dense insert into the right-sized
+ * table + bloom. Replaces a divergent fused build + a duplicated counting
+ * pass. */
+__global__ void k_ord_filter_date(const otbx_orders_dev o, int32_t q3date,
+                                  int64_t *__restrict__ out, int64_t *ncand)
 {
-    int64_t mask = cap - 1;
+    const int BUF = 1024;
+    __shared__ int64_t buf[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < o.n;
          i += stride) {
-        if (!(o.o_orderdate[i] < q3date))        /* ExecQual on orders */
-            continue;
-        if (!d_keyset_probe(ckeys, ccap, o.o_custkey[i]))  /* ⋈ customer */
-            continue;
+        bool m = o.o_orderdate[i] < q3date;       /* ExecQual on orders */
+        unsigned long long mask = __ballot(m);
+        int cnt = __popcll(mask);
+        if (nbuf + cnt > BUF) {
+            long long base = 0;
+            if (lane == 0)
+                base = (long long)atomicAdd((unsigned long long *)ncand,
+                                            (unsigned long long)nbuf);
+            base = __shfl(base, 0, WAVE);
+            for (int j = lane; j < nbuf; j += WAVE)
+                out[base + j] = buf[wid][j];
+            nbuf = 0;
+        }
+        if (m) {
+            int rank = __popcll(mask & ((1ull << lane) - 1ull));
+            buf[wid][nbuf + rank] = i;
+        }
+        nbuf += cnt;
+    }
+    if (nbuf) {
+        long long base = 0;
+        if (lane == 0)
+            base = (long long)atomicAdd((unsigned long long *)ncand,
+                                        (unsigned long long)nbuf);
+        base = __shfl(base, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE)
+            out[base + j] = buf[wid][j];
+    }
+}
+
+__global__ void k_ord_probe_cust(const otbx_orders_dev o,
+                                 const int64_t *__restrict__ cand,
+                                 const int64_t *__restrict__ ncand_p,
+                                 const unsigned long long *__restrict__ ckeys,
+                                 int64_t ccap, int64_t *__restrict__ out,
+                                 int64_t *nout)
+{
+    const int BUF = 1024;
+    __shared__ int64_t buf[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0;
+    int64_t n = *ncand_p;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
+         ci += stride) {
+        int64_t i = cand[ci];
+        bool m = d_keyset_probe(ckeys, ccap, o.o_custkey[i]); /* ⋈ customer */
+        unsigned long long mask = __ballot(m);
+        int cnt = __popcll(mask);
+        if (nbuf + cnt > BUF) {
+            long long base = 0;
+            if (lane == 0)
+                base = (long long)atomicAdd((unsigned long long *)nout,
+                                            (unsigned long long)nbuf);
+            base = __shfl(base, 0, WAVE);
+            for (int j = lane; j < nbuf; j += WAVE)
+                out[base + j] = buf[wid][j];
+            nbuf = 0;
+        }
+        if (m) {
+            int rank = __popcll(mask & ((1ull << lane) - 1ull));
+            buf[wid][nbuf + rank] = i;
+        }
+        nbuf += cnt;
+    }
+    if (nbuf) {
+        long long base = 0;
+        if (lane == 0)
+            base = (long long)atomicAdd((unsigned long long *)nout,
+                                        (unsigned long long)nbuf);
+        base = __shfl(base, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE)
+            out[base + j] = buf[wid][j];
+    }
+}
+
+__global__ void k_ord_insert(const otbx_orders_dev o,
+                             const int64_t *__restrict__ cand,
+                             const int64_t *__restrict__ ncand_p,
+                             ord_slot *tab, int64_t cap,
+                             unsigned long long *bloom, int64_t bloom_words)
+{
+    int64_t mask = cap - 1;
+    int64_t n = *ncand_p;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
+         ci += stride) {
+        int64_t i = cand[ci];
         unsigned long long k = (unsigned long long)o.o_orderkey[i];
         d_bloom_set(bloom, bloom_words, (int64_t)k);
         int64_t s = (int64_t)(d_hash_i64((int64_t)k) & (uint64_t)mask);
@@ -1202,12 +1291,14 @@ static int64_t fit_cap(int64_t n)
 otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
                                     int64_t nlineitem, size_t *bytes)
 {
-    /* worst case: tables are right-sized at run time (count-then-build),
-     * the caller allocates the no-filter upper bound */
-    int64_t ccap = fit_cap(ncust), ocap = fit_cap(norders);
-    *bytes = 64 + ((size_t)1 << 23) * 8 /* bloom */ + (size_t)ccap * 8 +
-             (size_t)ocap * sizeof(ord_slot) + (size_t)ocap * sizeof(q3g_slot) +
-             (size_t)nlineitem * 8 /* candidate row ids */;
+    /* fixed region layout at worst-case caps; tables inside each region are
+     * right-sized at run time (count-then-build) so only the live prefix is
+     * cleared/touched */
+    int64_t ccap_w = fit_cap(ncust), ocap_w = fit_cap(norders);
+    *bytes = 64 + (size_t)ccap_w * 8 + ((size_t)1 << 23) * 8 /* bloom */ +
+             (size_t)ocap_w * sizeof(ord_slot) + (size_t)ocap_w * sizeof(q3g_slot) +
+             (size_t)nlineitem * 8 /* lineitem candidates */ +
+             (size_t)norders * 16 /* orders candidate lists (2) */;
     return OTBX_OK;
 }
 
@@ -1233,11 +1324,21 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
         if (ws_bytes < worst) return OTBX_ERR_INVALID;
     }
     hipStream_t s = (hipStream_t)stream;
-    int64_t *hdr = (int64_t *)ws;           /* [0]=ncust_f, [1]=nof */
-    char *tabs = (char *)ws + 64;
+    /* fixed region offsets at worst-case caps */
+    int64_t ccap_w = fit_cap(ncust), ocap_w = fit_cap(o->n);
+    int64_t *hdr = (int64_t *)ws;  /* [0]=ncust_f [1]=nof [2]=li cand [3]=ord date cand */
+    char *p0 = (char *)ws + 64;
+    unsigned long long *ctab = (unsigned long long *)p0;
+    unsigned long long *bloom = (unsigned long long *)(p0 + (size_t)ccap_w * 8);
+    ord_slot *otab = (ord_slot *)((char *)bloom + ((size_t)1 << 23) * 8);
+    q3g_slot *gtab = (q3g_slot *)((char *)otab + (size_t)ocap_w * sizeof(ord_slot));
+    int64_t *cand_li = (int64_t *)((char *)gtab + (size_t)ocap_w * sizeof(q3g_slot));
+    int64_t *cand_o1 = cand_li + l->n;
+    int64_t *cand_o2 = cand_o1 + o->n;
+
     static int64_t *h_cnt = nullptr;        /* pinned host readback */
     if (!h_cnt)
-        HIP_CHECK(hipHostMalloc(&h_cnt, 2 * sizeof(int64_t)));
+        HIP_CHECK(hipHostMalloc(&h_cnt, 4 * sizeof(int64_t)));
 
     HIP_CHECK(hipMemsetAsync(hdr, 0, 64, s));
     HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, sizeof(int64_t), s));
@@ -1268,7 +1369,6 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
         HIP_CHECK(hipStreamSynchronize(s));
         ccap = fit_cap(h_cnt[0]);
     }
-    unsigned long long *ctab = (unsigned long long *)tabs;
     HIP_CHECK(hipMemsetAsync(ctab, 0, (size_t)ccap * 8, s));
     if (cust_keys_dev) {
         if (ncust_keys > 0)
@@ -1282,31 +1382,31 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     }
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[1], s));
 
-    /* ---- phase 2: size + build the filtered-orders table + bloom */
-    hipLaunchKernelGGL(k_count_orders_filtered, dim3(grid_for(o->n, 256)),
-                       dim3(256), 0, s, *o, ctab, ccap, q3date, &hdr[1]);
+    /* ---- phase 2: orders pipeline — date filter (dense list), keyset
+     * probe (dense matched list = exact count), right-sized insert + bloom */
+    hipLaunchKernelGGL(k_ord_filter_date, dim3(grid_for(o->n, 256)), dim3(256),
+                       0, s, *o, q3date, cand_o1, &hdr[3]);
+    hipLaunchKernelGGL(k_ord_probe_cust, dim3(grid_for(o->n, 256)), dim3(256),
+                       0, s, *o, cand_o1, &hdr[3], ctab, ccap, cand_o2,
+                       &hdr[1]);
     HIP_CHECK(hipMemcpyAsync(h_cnt + 1, hdr + 1, 8, hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     int64_t nof = h_cnt[1];
     int64_t ocap = fit_cap(nof), gcap = ocap;
     int64_t bwords = bloom_words_for(nof);
-    unsigned long long *bloom = (unsigned long long *)(tabs + (size_t)ccap * 8);
-    ord_slot *otab = (ord_slot *)((char *)bloom + (size_t)bwords * 8);
-    q3g_slot *gtab = (q3g_slot *)((char *)otab + (size_t)ocap * sizeof(ord_slot));
-    HIP_CHECK(hipMemsetAsync(bloom, 0,
-                             (size_t)bwords * 8 + (size_t)ocap * sizeof(ord_slot) +
-                                 (size_t)gcap * sizeof(q3g_slot), s));
-    hipLaunchKernelGGL(k_orders_build, dim3(grid_for(o->n, 256)), dim3(256), 0,
-                       s, *o, ctab, ccap, q3date, otab, ocap, bloom, bwords);
+    HIP_CHECK(hipMemsetAsync(bloom, 0, (size_t)bwords * 8, s));
+    HIP_CHECK(hipMemsetAsync(otab, 0, (size_t)ocap * sizeof(ord_slot), s));
+    HIP_CHECK(hipMemsetAsync(gtab, 0, (size_t)gcap * sizeof(q3g_slot), s));
+    hipLaunchKernelGGL(k_ord_insert, dim3(grid_for(o->n, 256)), dim3(256), 0,
+                       s, *o, cand_o2, &hdr[1], otab, ocap, bloom, bwords);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
 
-    /* ---- phase 3: lineitem scan+filter (compacted candidates), then the
-     * dense probe + partial agg */
-    int64_t *cand = (int64_t *)((char *)gtab + (size_t)gcap * sizeof(q3g_slot));
+    /* ---- phase 3: lineitem scan+filter (compacted candidates), dense
+     * probe + partial agg */
     hipLaunchKernelGGL(k_q3_scan_filter, dim3(grid_for(l->n, 256)), dim3(256),
-                       0, s, *l, q3date, bloom, bwords, cand, &hdr[2]);
+                       0, s, *l, q3date, bloom, bwords, cand_li, &hdr[2]);
     hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(l->n, 256)), dim3(256), 0,
-                       s, *l, cand, &hdr[2], otab, ocap, gtab, gcap, nhits);
+                       s, *l, cand_li, &hdr[2], otab, ocap, gtab, gcap, nhits);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
 
     /* ---- phase 4: compact groups */
