@@ -1076,49 +1076,76 @@ __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
                                  int64_t bloom_words,
                                  int64_t *__restrict__ cand, int64_t *ncand)
 {
-    /* single pass: per-wave LDS staging buffer, one global atomic per 1024
-     * candidates (a direct wave-level append funnels ~500k atomics through
-     * one word; a two-phase block scheme re-streams the columns) */
+    /* single pass, 8 rows per lane per batch: the 8 bloom loads pipeline
+     * (a row-at-a-time loop waits a full random-load latency per row because
+     * the ballot converges the wave); date-failing lanes read bloom word 0
+     * (stays hot in L1) so all 8 loads issue unconditionally. Per-wave LDS
+     * staging, one global atomic per 1024 candidates. */
     const int BUF = 1024;
+    const int B = 8;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
     int nbuf = 0; /* wave-uniform */
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < l.n;
-         i += stride) {
-        bool m = (l.l_shipdate[i] > q3date) &&
-                 d_bloom_test(bloom, bloom_words, l.l_orderkey[i]);
-        unsigned long long mask = __ballot(m);
-        int cnt = __popcll(mask);
-        if (nbuf + cnt > BUF) {
-            long long base = 0;
-            if (lane == 0)
-                base = (long long)atomicAdd((unsigned long long *)ncand,
-                                            (unsigned long long)nbuf);
-            base = __shfl(base, 0, WAVE);
-            for (int j = lane; j < nbuf; j += WAVE)
-                cand[base + j] = buf[wid][j];
-            nbuf = 0;
+    int64_t chunk = (int64_t)blockDim.x * B;
+    int64_t stride = (int64_t)gridDim.x * chunk;
+    for (int64_t base = (int64_t)blockIdx.x * chunk; base < l.n;
+         base += stride) {
+        bool m[B];
+        int64_t rows[B];
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            int64_t i = base + (int64_t)k * blockDim.x + threadIdx.x;
+            rows[k] = i;
+            bool datepass = false;
+            uint64_t w = 0;
+            unsigned long long bits = 0;
+            int64_t key = 0;
+            if (i < l.n) {
+                datepass = l.l_shipdate[i] > q3date;
+                key = l.l_orderkey[i];
+                uint64_t h = otbx_splitmix64((uint64_t)key);
+                w = datepass ? ((h >> 12) & (uint64_t)(bloom_words - 1)) : 0;
+                bits = (1ull << (h & 63)) | (1ull << ((h >> 6) & 63));
+            }
+            unsigned long long v = bloom[w];  /* w=0 broadcast line if fail */
+            m[k] = datepass && ((v & bits) == bits);
         }
-        if (m) {
-            int rank = __popcll(mask & ((1ull << lane) - 1ull));
-            buf[wid][nbuf + rank] = i;
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            unsigned long long mask = __ballot(m[k]);
+            int cnt = __popcll(mask);
+            if (nbuf + cnt > BUF) {
+                long long bpos = 0;
+                if (lane == 0)
+                    bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                                (unsigned long long)nbuf);
+                bpos = __shfl(bpos, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE)
+                    cand[bpos + j] = buf[wid][j];
+                nbuf = 0;
+            }
+            if (m[k]) {
+                int rank = __popcll(mask & ((1ull << lane) - 1ull));
+                buf[wid][nbuf + rank] = rows[k];
+            }
+            nbuf += cnt;
         }
-        nbuf += cnt;
     }
     if (nbuf) {
-        long long base = 0;
+        long long bpos = 0;
         if (lane == 0)
-            base = (long long)atomicAdd((unsigned long long *)ncand,
+            bpos = (long long)atomicAdd((unsigned long long *)ncand,
                                         (unsigned long long)nbuf);
-        base = __shfl(base, 0, WAVE);
+        bpos = __shfl(bpos, 0, WAVE);
         for (int j = lane; j < nbuf; j += WAVE)
-            cand[base + j] = buf[wid][j];
+            cand[bpos + j] = buf[wid][j];
     }
 }
 
-/* probe phase B: dense over candidates: gather key, probe orders table,
- * gather payload columns, aggregate into the group table. */
+/* probe phase B: dense over candidates, 4 per lane per batch so the key
+ * gathers and first-slot table loads pipeline; the (rare) collision walk and
+ * the group-table atomics stay scalar. The 16-B slot is read as one load
+ * (okey | date,prio). */
 __global__ void k_q3_probe_agg(const otbx_lineitem_dev l,
                                const int64_t *__restrict__ cand,
                                const int64_t *__restrict__ ncand_p,
@@ -1126,61 +1153,78 @@ __global__ void k_q3_probe_agg(const otbx_lineitem_dev l,
                                q3g_slot *gtab, int64_t gcap,
                                unsigned long long *__restrict__ nhits)
 {
+    const int B = 4;
     int64_t omask = ocap - 1, gmask = gcap - 1;
     int64_t n = *ncand_p;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t chunk = (int64_t)blockDim.x * B;
+    int64_t stride = (int64_t)gridDim.x * chunk;
     unsigned long long myhits = 0;
-    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
-         ci += stride) {
-        int64_t i = cand[ci];
-        int64_t k = l.l_orderkey[i];
-        /* probe orders table (read-only this launch -> plain loads) */
-        int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)omask);
-        int32_t date = 0, prio = 0;
-        bool hit = false;
-        for (;;) {
-            unsigned long long v = otab[s].okey;
-            if (v == 0ull) break;
-            if (v == (unsigned long long)k) {
-                hit = true;
-                date = otab[s].date;
-                prio = otab[s].prio;
-                break;
-            }
-            s = (s + 1) & omask;
+    const ulonglong2 *otab2 = (const ulonglong2 *)otab;
+    for (int64_t base = (int64_t)blockIdx.x * chunk; base < n; base += stride) {
+        int64_t row[B], key[B], slot[B];
+        ulonglong2 sv[B];
+        bool valid[B];
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            int64_t ci = base + (int64_t)k * blockDim.x + threadIdx.x;
+            valid[k] = ci < n;
+            row[k] = valid[k] ? cand[ci] : 0;
         }
-        if (!hit)
-            continue;
-        myhits++;
-        double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
-        /* group insert-or-update (mutable table -> CAS probing only) */
-        int64_t g = (int64_t)(d_hash_i64(k) & (uint64_t)gmask);
-        for (;;) {
-            unsigned long long old = atomicCAS(&gtab[g].okey, 0ull,
-                                               (unsigned long long)k);
-            if (old == 0ull) {
-                gtab[g].date = date;   /* winner writes payload; readers are
-                                        * the compact kernel (next launch) */
-                gtab[g].prio = prio;
-                break;
-            }
-            if (old == (unsigned long long)k) break;
-            g = (g + 1) & gmask;
+#pragma unroll
+        for (int k = 0; k < B; k++)
+            key[k] = l.l_orderkey[row[k]];
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            slot[k] = (int64_t)(d_hash_i64(key[k]) & (uint64_t)omask);
+            sv[k] = otab2[slot[k]];
         }
-        atomicAdd(&gtab[g].revenue, rev);
+        double ep[B], dc[B];
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            ep[k] = l.l_extendedprice[row[k]];
+            dc[k] = l.l_discount[row[k]];
+        }
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            if (!valid[k])
+                continue;
+            unsigned long long v = sv[k].x;
+            unsigned long long payload = sv[k].y;
+            int64_t s = slot[k];
+            while (v != 0ull && v != (unsigned long long)key[k]) {
+                s = (s + 1) & omask;          /* rare: collision walk */
+                ulonglong2 sv2 = otab2[s];
+                v = sv2.x;
+                payload = sv2.y;
+            }
+            if (v == 0ull)
+                continue;                     /* bloom false positive */
+            myhits++;
+            int32_t date = (int32_t)(payload & 0xffffffffull);
+            int32_t prio = (int32_t)(payload >> 32);
+            double rev = ep[k] * (1.0 - dc[k]);
+            int64_t g = (int64_t)(d_hash_i64(key[k]) & (uint64_t)gmask);
+            for (;;) {
+                unsigned long long old = atomicCAS(&gtab[g].okey, 0ull,
+                                                   (unsigned long long)key[k]);
+                if (old == 0ull) {
+                    gtab[g].date = date;  /* winner writes payload; read by
+                                           * the compact kernel (next launch) */
+                    gtab[g].prio = prio;
+                    break;
+                }
+                if (old == (unsigned long long)key[k]) break;
+                g = (g + 1) & gmask;
+            }
+            atomicAdd(&gtab[g].revenue, rev);
+        }
     }
-    /* per-wave hit count, one atomic per wave */
     for (int off = WAVE / 2; off > 0; off >>= 1)
         myhits += __shfl_down(myhits, off, WAVE);
     if ((threadIdx.x % WAVE) == 0 && myhits)
         atomicAdd(nhits, myhits);
 }
 
-/* block-aggregated compaction: one global atomic per BLOCK (a wave-level
- * append still funnels ~500k atomics through one word at SF100 — measured
- * 6.3 ms; this runs at stream bandwidth). Each block owns a contiguous
- * chunk: pass 1 counts, one atomicAdd reserves the block's output range,
- * pass 2 writes at per-thread prefix offsets. */
 __global__ void k_q3_compact(const q3g_slot *gtab, int64_t gcap,
                              otbx_q3_group *out, int64_t cap_out,
                              int64_t *ngroups)
