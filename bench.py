@@ -99,8 +99,12 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     if world > 1:
         import torch.distributed as dist
+        # OTBX_DIST_BACKEND=gloo lets the full multi-rank path run on a
+        # single-GPU box (CI); production is nccl (= RCCL over xGMI)
+        backend = os.environ.get("OTBX_DIST_BACKEND", "nccl")
+        local_rank = local_rank % max(torch.cuda.device_count(), 1)
         torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        dist.init_process_group(backend)
     else:
         dist = None
 
@@ -162,7 +166,7 @@ def main():
                      CT.c_void_p(keys.data_ptr()), CT.c_void_p(nk.data_ptr()),
                      stream)
                 local = keys[: int(nk.cpu().item())]
-                bcast = fragment.broadcast_customer_keys(local)
+                bcast = fragment.broadcast_customer_keys(local).to("cuda")
             node = ex.GpuQ3Fragment(cu, od, li, cust_keys=bcast)
             node.BeginCustomScan()
             node._rows = node._run()
@@ -191,7 +195,8 @@ def main():
 
     # MAX over ranks
     if dist is not None:
-        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.cpu().item())
 

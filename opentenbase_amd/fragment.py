@@ -27,6 +27,8 @@ def merge_q1_partials(sums, counts):
     sums, int8pl for counts, slot-aligned groups."""
     if not is_dist() or dist.get_world_size() == 1:
         return q1_rows_from_state(sums, counts)
+    if dist.get_backend() == "gloo" and sums.is_cuda:
+        sums, counts = sums.cpu(), counts.cpu()
     world = dist.get_world_size()
     gs = [torch.empty_like(sums) for _ in range(world)]
     gc = [torch.empty_like(counts) for _ in range(world)]
@@ -46,6 +48,8 @@ def allgather_variable(t):
     concatenation semantics). Returns the concatenation over ranks."""
     if not is_dist() or dist.get_world_size() == 1:
         return t
+    if dist.get_backend() == "gloo" and t.is_cuda:
+        t = t.cpu()
     world = dist.get_world_size()
     n = torch.tensor([t.numel()], dtype=torch.int64, device=t.device)
     ns = [torch.empty_like(n) for _ in range(world)]
