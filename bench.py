@@ -53,8 +53,10 @@ def load_traffic(workload, sf):
     try:
         with open(p) as f:
             j = json.load(f)
-        if j.get("workload") == workload and j.get("sf_per_gpu") == sf:
-            return float(j["traffic_bytes_per_launch"])
+        entries = j if isinstance(j, list) else [j]
+        for e in entries:
+            if e.get("workload") == workload and e.get("sf_per_gpu") == sf:
+                return float(e["traffic_bytes_per_launch"])
     except Exception:
         pass
     return None
