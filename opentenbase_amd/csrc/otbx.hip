@@ -1076,13 +1076,11 @@ __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
                                  int64_t bloom_words,
                                  int64_t *__restrict__ cand, int64_t *ncand)
 {
-    /* 8 CONSECUTIVE rows per lane: lineitem is clustered ~4 rows per
-     * orderkey (the reference's heap order), so within a lane batch only
-     * run-leading keys pay a bloom-word load (dupes reuse the leader's
-     * test; non-leaders read word 0, which stays hot in L1). Candidates are
-     * appended ORDER-PRESERVING (wave prefix-sum over per-lane counts) so
-     * the probe kernel sees key runs intact and can deduplicate probes.
-     * Per-wave LDS staging keeps it at one global atomic per 1024. */
+    /* single pass, 8 rows per lane per batch: the 8 bloom loads pipeline
+     * (a row-at-a-time loop waits a full random-load latency per row because
+     * the ballot converges the wave); date-failing lanes read bloom word 0
+     * (stays hot in L1) so all 8 loads issue unconditionally. Per-wave LDS
+     * staging, one global atomic per 1024 candidates. */
     const int BUF = 1024;
     const int B = 8;
     __shared__ int64_t buf[256 / WAVE][BUF];
@@ -1092,61 +1090,46 @@ __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
     int64_t stride = (int64_t)gridDim.x * chunk;
     for (int64_t base = (int64_t)blockIdx.x * chunk; base < l.n;
          base += stride) {
-        int64_t r0 = base + (int64_t)threadIdx.x * B;
         bool m[B];
-        int64_t ky[B];
-        bool dp[B];
+        int64_t rows[B];
 #pragma unroll
-        for (int j = 0; j < B; j++) {
-            int64_t i = r0 + j;
-            bool inb = i < l.n;
-            dp[j] = inb && (l.l_shipdate[inb ? i : 0] > q3date);
-            ky[j] = l.l_orderkey[inb ? i : 0];
+        for (int k = 0; k < B; k++) {
+            int64_t i = base + (int64_t)k * blockDim.x + threadIdx.x;
+            rows[k] = i;
+            bool datepass = false;
+            uint64_t w = 0;
+            unsigned long long bits = 0;
+            int64_t key = 0;
+            if (i < l.n) {
+                datepass = l.l_shipdate[i] > q3date;
+                key = l.l_orderkey[i];
+                uint64_t h = otbx_splitmix64((uint64_t)key);
+                w = datepass ? ((h >> 12) & (uint64_t)(bloom_words - 1)) : 0;
+                bits = (1ull << (h & 63)) | (1ull << ((h >> 6) & 63));
+            }
+            unsigned long long v = bloom[w];  /* w=0 broadcast line if fail */
+            m[k] = datepass && ((v & bits) == bits);
         }
-        unsigned long long v[B];
-        unsigned long long bits[B];
-        bool dup[B];
 #pragma unroll
-        for (int j = 0; j < B; j++) {
-            dup[j] = j > 0 && ky[j] == ky[j - 1];
-            uint64_t h = otbx_splitmix64((uint64_t)ky[j]);
-            uint64_t w = dup[j] ? 0 : ((h >> 12) & (uint64_t)(bloom_words - 1));
-            bits[j] = (1ull << (h & 63)) | (1ull << ((h >> 6) & 63));
-            v[j] = bloom[w];
+        for (int k = 0; k < B; k++) {
+            unsigned long long mask = __ballot(m[k]);
+            int cnt = __popcll(mask);
+            if (nbuf + cnt > BUF) {
+                long long bpos = 0;
+                if (lane == 0)
+                    bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                                (unsigned long long)nbuf);
+                bpos = __shfl(bpos, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE)
+                    cand[bpos + j] = buf[wid][j];
+                nbuf = 0;
+            }
+            if (m[k]) {
+                int rank = __popcll(mask & ((1ull << lane) - 1ull));
+                buf[wid][nbuf + rank] = rows[k];
+            }
+            nbuf += cnt;
         }
-        bool ok[B];
-#pragma unroll
-        for (int j = 0; j < B; j++)
-            ok[j] = dup[j] ? ok[j - 1] : ((v[j] & bits[j]) == bits[j]);
-        int mycnt = 0;
-#pragma unroll
-        for (int j = 0; j < B; j++) {
-            m[j] = dp[j] && ok[j];
-            mycnt += m[j];
-        }
-        /* wave-inclusive prefix sum over per-lane counts */
-        int incl = mycnt;
-        for (int off = 1; off < WAVE; off <<= 1) {
-            int up = __shfl_up(incl, off, WAVE);
-            if (lane >= off) incl += up;
-        }
-        int tot = __shfl(incl, WAVE - 1, WAVE);
-        if (nbuf + tot > BUF) {
-            long long bpos = 0;
-            if (lane == 0)
-                bpos = (long long)atomicAdd((unsigned long long *)ncand,
-                                            (unsigned long long)nbuf);
-            bpos = __shfl(bpos, 0, WAVE);
-            for (int j = lane; j < nbuf; j += WAVE)
-                cand[bpos + j] = buf[wid][j];
-            nbuf = 0;
-        }
-        int pos = nbuf + incl - mycnt;
-#pragma unroll
-        for (int j = 0; j < B; j++)
-            if (m[j])
-                buf[wid][pos++] = r0 + j;
-        nbuf += tot;
     }
     if (nbuf) {
         long long bpos = 0;
@@ -1159,9 +1142,10 @@ __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
     }
 }
 
-/* probe phase: 4 CONSECUTIVE candidates per lane — the order-preserving
- * scan keeps key runs intact, so only run leaders probe the orders table and
- * a run's revenue is accumulated in-register before ONE group-table atomic. */
+/* probe phase B: dense over candidates, 4 per lane per batch so the key
+ * gathers and first-slot table loads pipeline; the (rare) collision walk and
+ * the group-table atomics stay scalar. The 16-B slot is read as one load
+ * (okey | date,prio). */
 __global__ void k_q3_probe_agg(const otbx_lineitem_dev l,
                                const int64_t *__restrict__ cand,
                                const int64_t *__restrict__ ncand_p,
@@ -1177,100 +1161,62 @@ __global__ void k_q3_probe_agg(const otbx_lineitem_dev l,
     unsigned long long myhits = 0;
     const ulonglong2 *otab2 = (const ulonglong2 *)otab;
     for (int64_t base = (int64_t)blockIdx.x * chunk; base < n; base += stride) {
-        int64_t ci0 = base + (int64_t)threadIdx.x * B;
-        int64_t row[B], key[B];
-        bool valid[B], dup[B];
-#pragma unroll
-        for (int j = 0; j < B; j++) {
-            int64_t ci = ci0 + j;
-            valid[j] = ci < n;
-            row[j] = cand[valid[j] ? ci : 0];
-        }
-#pragma unroll
-        for (int j = 0; j < B; j++)
-            key[j] = l.l_orderkey[row[j]];
+        int64_t row[B], key[B], slot[B];
         ulonglong2 sv[B];
+        bool valid[B];
 #pragma unroll
-        for (int j = 0; j < B; j++) {
-            dup[j] = j > 0 && key[j] == key[j - 1];
-            int64_t s = dup[j] ? 0
-                               : (int64_t)(d_hash_i64(key[j]) & (uint64_t)omask);
-            sv[j] = otab2[s];
+        for (int k = 0; k < B; k++) {
+            int64_t ci = base + (int64_t)k * blockDim.x + threadIdx.x;
+            valid[k] = ci < n;
+            row[k] = valid[k] ? cand[ci] : 0;
         }
-        bool hit[B];
-        unsigned long long payload[B];
 #pragma unroll
-        for (int j = 0; j < B; j++) {
-            if (dup[j]) {
-                hit[j] = hit[j - 1];
-                payload[j] = payload[j - 1];
-                continue;
-            }
-            int64_t s = (int64_t)(d_hash_i64(key[j]) & (uint64_t)omask);
-            unsigned long long v = sv[j].x;
-            unsigned long long pl = sv[j].y;
-            while (v != 0ull && v != (unsigned long long)key[j]) {
-                s = (s + 1) & omask;          /* rare: collision walk */
-                ulonglong2 sv2 = otab2[s];
-                v = sv2.x;
-                pl = sv2.y;
-            }
-            hit[j] = v != 0ull;
-            payload[j] = pl;
+        for (int k = 0; k < B; k++)
+            key[k] = l.l_orderkey[row[k]];
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            slot[k] = (int64_t)(d_hash_i64(key[k]) & (uint64_t)omask);
+            sv[k] = otab2[slot[k]];
         }
         double ep[B], dc[B];
 #pragma unroll
-        for (int j = 0; j < B; j++) {
-            ep[j] = l.l_extendedprice[row[j]];
-            dc[j] = l.l_discount[row[j]];
+        for (int k = 0; k < B; k++) {
+            ep[k] = l.l_extendedprice[row[k]];
+            dc[k] = l.l_discount[row[k]];
         }
-        /* run-accumulate revenue; one group atomic per run */
-        double runsum = 0.0;
-        int64_t runkey = 0;
-        unsigned long long runpl = 0;
 #pragma unroll
-        for (int j = 0; j < B; j++) {
-            if (j > 0 && !dup[j] && runsum != 0.0) {
-                int32_t date = (int32_t)(runpl & 0xffffffffull);
-                int32_t prio = (int32_t)(runpl >> 32);
-                int64_t g = (int64_t)(d_hash_i64(runkey) & (uint64_t)gmask);
-                for (;;) {
-                    unsigned long long old = atomicCAS(
-                        &gtab[g].okey, 0ull, (unsigned long long)runkey);
-                    if (old == 0ull) {
-                        gtab[g].date = date;
-                        gtab[g].prio = prio;
-                        break;
-                    }
-                    if (old == (unsigned long long)runkey) break;
-                    g = (g + 1) & gmask;
-                }
-                atomicAdd(&gtab[g].revenue, runsum);
-                runsum = 0.0;
+        for (int k = 0; k < B; k++) {
+            if (!valid[k])
+                continue;
+            unsigned long long v = sv[k].x;
+            unsigned long long payload = sv[k].y;
+            int64_t s = slot[k];
+            while (v != 0ull && v != (unsigned long long)key[k]) {
+                s = (s + 1) & omask;          /* rare: collision walk */
+                ulonglong2 sv2 = otab2[s];
+                v = sv2.x;
+                payload = sv2.y;
             }
-            if (valid[j] && hit[j]) {
-                myhits++;
-                runsum += ep[j] * (1.0 - dc[j]);
-                runkey = key[j];
-                runpl = payload[j];
-            }
-        }
-        if (runsum != 0.0) {
-            int32_t date = (int32_t)(runpl & 0xffffffffull);
-            int32_t prio = (int32_t)(runpl >> 32);
-            int64_t g = (int64_t)(d_hash_i64(runkey) & (uint64_t)gmask);
+            if (v == 0ull)
+                continue;                     /* bloom false positive */
+            myhits++;
+            int32_t date = (int32_t)(payload & 0xffffffffull);
+            int32_t prio = (int32_t)(payload >> 32);
+            double rev = ep[k] * (1.0 - dc[k]);
+            int64_t g = (int64_t)(d_hash_i64(key[k]) & (uint64_t)gmask);
             for (;;) {
                 unsigned long long old = atomicCAS(&gtab[g].okey, 0ull,
-                                                   (unsigned long long)runkey);
+                                                   (unsigned long long)key[k]);
                 if (old == 0ull) {
-                    gtab[g].date = date;
+                    gtab[g].date = date;  /* winner writes payload; read by
+                                           * the compact kernel (next launch) */
                     gtab[g].prio = prio;
                     break;
                 }
-                if (old == (unsigned long long)runkey) break;
+                if (old == (unsigned long long)key[k]) break;
                 g = (g + 1) & gmask;
             }
-            atomicAdd(&gtab[g].revenue, runsum);
+            atomicAdd(&gtab[g].revenue, rev);
         }
     }
     for (int off = WAVE / 2; off > 0; off >>= 1)
