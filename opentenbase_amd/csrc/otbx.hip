@@ -1097,18 +1097,30 @@ __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
             int64_t i = base + (int64_t)k * blockDim.x + threadIdx.x;
             rows[k] = i;
             bool datepass = false;
-            uint64_t w = 0;
-            unsigned long long bits = 0;
             int64_t key = 0;
             if (i < l.n) {
                 datepass = l.l_shipdate[i] > q3date;
                 key = l.l_orderkey[i];
-                uint64_t h = otbx_splitmix64((uint64_t)key);
-                w = datepass ? ((h >> 12) & (uint64_t)(bloom_words - 1)) : 0;
-                bits = (1ull << (h & 63)) | (1ull << ((h >> 6) & 63));
             }
-            unsigned long long v = bloom[w];  /* w=0 broadcast line if fail */
-            m[k] = datepass && ((v & bits) == bits);
+            /* consecutive rows sit in consecutive lanes: only run-leading
+             * lanes load a bloom word, dupes take the leader's result via a
+             * nearest-leader max-scan (6 shuffles) — lineitem is clustered
+             * ~4 rows/orderkey so this ~halves the random bloom lines */
+            int64_t kprev = __shfl_up((long long)key, 1, WAVE);
+            bool isleader = (lane == 0) || key != kprev;
+            uint64_t h = otbx_splitmix64((uint64_t)key);
+            uint64_t w = isleader ? ((h >> 12) & (uint64_t)(bloom_words - 1)) : 0;
+            unsigned long long bits =
+                (1ull << (h & 63)) | (1ull << ((h >> 6) & 63));
+            unsigned long long v = bloom[w]; /* word 0 stays hot for dupes */
+            int okv = (v & bits) == bits;
+            int lidx = isleader ? lane : -1;
+            for (int st = 1; st < WAVE; st <<= 1) {
+                int u = __shfl_up(lidx, st, WAVE);
+                if (lane >= st && u > lidx) lidx = u;
+            }
+            int ok = __shfl(okv, lidx, WAVE);
+            m[k] = datepass && ok;
         }
 #pragma unroll
         for (int k = 0; k < B; k++) {
@@ -1173,10 +1185,17 @@ __global__ void k_q3_probe_agg(const otbx_lineitem_dev l,
 #pragma unroll
         for (int k = 0; k < B; k++)
             key[k] = l.l_orderkey[row[k]];
+        /* consecutive candidates sit in consecutive lanes (the scan append
+         * preserves row order): only run-leading lanes load a table slot,
+         * dupes take the leader's line via nearest-leader broadcast */
+        int lane = (int)(threadIdx.x % WAVE);
+        bool lead[B];
 #pragma unroll
         for (int k = 0; k < B; k++) {
+            int64_t kprev = __shfl_up((long long)key[k], 1, WAVE);
+            lead[k] = (lane == 0) || key[k] != kprev;
             slot[k] = (int64_t)(d_hash_i64(key[k]) & (uint64_t)omask);
-            sv[k] = otab2[slot[k]];
+            sv[k] = otab2[lead[k] ? slot[k] : 0];
         }
         double ep[B], dc[B];
 #pragma unroll
@@ -1186,19 +1205,30 @@ __global__ void k_q3_probe_agg(const otbx_lineitem_dev l,
         }
 #pragma unroll
         for (int k = 0; k < B; k++) {
-            if (!valid[k])
-                continue;
+            /* leaders resolve their slot (collision walk); every lane then
+             * broadcasts from its nearest leader */
             unsigned long long v = sv[k].x;
             unsigned long long payload = sv[k].y;
-            int64_t s = slot[k];
-            while (v != 0ull && v != (unsigned long long)key[k]) {
-                s = (s + 1) & omask;          /* rare: collision walk */
-                ulonglong2 sv2 = otab2[s];
-                v = sv2.x;
-                payload = sv2.y;
+            if (lead[k]) {
+                int64_t s = slot[k];
+                while (v != 0ull && v != (unsigned long long)key[k]) {
+                    s = (s + 1) & omask;      /* rare: collision walk */
+                    ulonglong2 sv2 = otab2[s];
+                    v = sv2.x;
+                    payload = sv2.y;
+                }
             }
-            if (v == 0ull)
-                continue;                     /* bloom false positive */
+            int hitv = v == (unsigned long long)key[k];
+            int lidx = lead[k] ? lane : -1;
+            for (int st = 1; st < WAVE; st <<= 1) {
+                int u = __shfl_up(lidx, st, WAVE);
+                if (lane >= st && u > lidx) lidx = u;
+            }
+            int hit = __shfl(hitv, lidx, WAVE);
+            payload = (unsigned long long)__shfl(
+                (long long)payload, lidx, WAVE);
+            if (!valid[k] || !hit)
+                continue;
             myhits++;
             int32_t date = (int32_t)(payload & 0xffffffffull);
             int32_t prio = (int32_t)(payload >> 32);
