@@ -152,6 +152,13 @@ class CustomScanState:
     def _run(self): raise NotImplementedError
     def _end(self): pass
 
+    def explain(self):
+        """EXPLAIN ANALYZE analog: per-kernel HIP-event timings for this
+        node, mirroring the reference's per-plan-node Instrumentation
+        (include/executor/instrument.h:45, merged at the CN by
+        commands/explain_dist.c). Returns {phase: ms} or None before run."""
+        return None
+
 
 class GpuSeqScanCount(CustomScanState):
     """SeqScan + qual + COUNT(*) (BASELINE config 2). Replaces the
@@ -198,6 +205,12 @@ class GpuQ1PartialAgg(CustomScanState):
 
     def partial_state_tensors(self):
         return self.sums, self.counts
+
+    def explain(self):
+        if self.kernel_ms is None:
+            return None
+        return {"k_q1_partial (SeqScan+Qual+Project+PartialAgg, fused)":
+                self.kernel_ms}
 
 
 def q1_rows_from_state(sums, counts):
@@ -295,6 +308,15 @@ class GpuQ3Fragment(CustomScanState):
         raw = cand[: nc * 24].cpu().numpy().tobytes()
         cands = np.frombuffer(raw, dtype=np.dtype(self.NP_DTYPE)).copy()
         return [tuple(r) for r in q3_topk(cands, self.k)]
+
+    def explain(self):
+        if self.kernel_ms is None:
+            return None
+        names = ["customer keyset build (Hash build side 1)",
+                 "orders filter+probe+insert (HashJoin 1 + Hash build 2)",
+                 "lineitem scan-filter + probe + partial agg (HashJoin 2 + PartialAgg)",
+                 "group compaction (emit)"]
+        return dict(zip(names, self.kernel_ms))
 
     def fetch_groups(self):
         """D2H copy of ALL partial groups (parity tests / debugging)."""
