@@ -91,6 +91,8 @@ def main():
     ap.add_argument("--workload", default="tpch_q1",
                     choices=["tpch_q1", "scan_count", "tpch_q3"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--skew", action="store_true",
+                    help="config-5 skewed distribution keys (hot custkeys)")
     ap.add_argument("--cpu-sample-rows", type=int, default=1_200_000_000,
                     help="~10 s of scalar CPU work on the target host")
     args = ap.parse_args()
@@ -131,7 +133,7 @@ def main():
     else:  # tpch_q3
         li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world)
         od = ex.GpuOrders.generate(n_global // 4, n_global // 40, rank=rank,
-                                   nranks=world)
+                                   nranks=world, skew=args.skew)
         cu = ex.GpuCustomer.generate(n_global // 40, rank=rank, nranks=world)
         bytes_per_row = 28  # probe-side streamed bytes
     torch.cuda.synchronize()

@@ -102,7 +102,8 @@ otbx_status otbx_gen_lineitem_dev(const otbx_lineitem_dev *t, uint64_t seed,
                                   uint32_t nranks, void *stream);
 otbx_status otbx_gen_orders_dev(const otbx_orders_dev *t, uint64_t seed,
                                 int64_t n_global, int64_t ncust_global,
-                                uint32_t rank, uint32_t nranks, void *stream);
+                                uint32_t rank, uint32_t nranks, int skew,
+                                void *stream);
 otbx_status otbx_gen_customer_dev(const otbx_customer_dev *t, uint64_t seed,
                                   int64_t n_global, uint32_t rank,
                                   uint32_t nranks, void *stream);

@@ -133,14 +133,15 @@ __global__ void k_gen_lineitem(otbx_lineitem_dev t, uint64_t seed,
 }
 
 __global__ void k_gen_orders(otbx_orders_dev t, uint64_t seed, int64_t ncust,
-                             uint32_t rank, uint32_t nranks)
+                             uint32_t rank, uint32_t nranks, int skew)
 {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t l = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; l < t.n;
          l += stride) {
         uint64_t i = otbx_ord_global_row((uint64_t)l, rank, nranks);
         t.o_orderkey[l] = otbx_ord_orderkey(i);
-        t.o_custkey[l] = otbx_ord_custkey(seed, i, ncust);
+        t.o_custkey[l] = skew ? otbx_ord_custkey_skewed(seed, i, ncust)
+                              : otbx_ord_custkey(seed, i, ncust);
         t.o_orderdate[l] = otbx_ord_orderdate(seed, i);
         t.o_shippriority[l] = otbx_ord_shippriority(i);
     }
@@ -182,12 +183,14 @@ otbx_status otbx_gen_lineitem_dev(const otbx_lineitem_dev *t, uint64_t seed,
 
 otbx_status otbx_gen_orders_dev(const otbx_orders_dev *t, uint64_t seed,
                                 int64_t n_global, int64_t ncust_global,
-                                uint32_t rank, uint32_t nranks, void *stream)
+                                uint32_t rank, uint32_t nranks, int skew,
+                                void *stream)
 {
     if (!t || nranks == 0 || n_global % nranks || t->n != n_global / nranks)
         return OTBX_ERR_INVALID;
     hipLaunchKernelGGL(k_gen_orders, dim3(grid_for(t->n, 256)), dim3(256), 0,
-                       (hipStream_t)stream, *t, seed, ncust_global, rank, nranks);
+                       (hipStream_t)stream, *t, seed, ncust_global, rank, nranks,
+                       skew);
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }

@@ -86,11 +86,11 @@ class GpuOrders:
 
     @classmethod
     def generate(cls, n_global, ncust_global, rank=0, nranks=1,
-                 seed=SEED_DEFAULT):
+                 seed=SEED_DEFAULT, skew=False):
         t = cls(n_global // nranks)
         call("otbx_gen_orders_dev", C.byref(t.cstruct), C.c_uint64(seed),
              C.c_int64(n_global), C.c_int64(ncust_global), C.c_uint32(rank),
-             C.c_uint32(nranks), _stream())
+             C.c_uint32(nranks), C.c_int(1 if skew else 0), _stream())
         return t
 
 

@@ -73,7 +73,8 @@ ora_status ora_gen_lineitem(ora_lineitem *t, uint64_t seed, int64_t n_global,
 }
 
 ora_status ora_gen_orders(ora_orders *t, uint64_t seed, int64_t n_global,
-                          int64_t ncust_global, uint32_t rank, uint32_t nranks)
+                          int64_t ncust_global, uint32_t rank, uint32_t nranks,
+                          int skew)
 {
     if (nranks == 0 || n_global % nranks) return ORA_ERR_INVALID;
     int64_t n = n_global / nranks;
@@ -87,7 +88,8 @@ ora_status ora_gen_orders(ora_orders *t, uint64_t seed, int64_t n_global,
     for (int64_t l = 0; l < n; l++) {
         uint64_t i = otbx_ord_global_row((uint64_t)l, rank, nranks);
         t->o_orderkey[l] = otbx_ord_orderkey(i);
-        t->o_custkey[l] = otbx_ord_custkey(seed, i, ncust_global);
+        t->o_custkey[l] = skew ? otbx_ord_custkey_skewed(seed, i, ncust_global)
+                               : otbx_ord_custkey(seed, i, ncust_global);
         t->o_orderdate[l] = otbx_ord_orderdate(seed, i);
         t->o_shippriority[l] = otbx_ord_shippriority(i);
     }

@@ -54,7 +54,8 @@ typedef struct {
 ora_status ora_gen_lineitem(ora_lineitem *t, uint64_t seed, int64_t n_global,
                             uint32_t rank, uint32_t nranks);
 ora_status ora_gen_orders(ora_orders *t, uint64_t seed, int64_t n_global,
-                          int64_t ncust_global, uint32_t rank, uint32_t nranks);
+                          int64_t ncust_global, uint32_t rank, uint32_t nranks,
+                          int skew /* config-5 hot-custkey skew */);
 ora_status ora_gen_customer(ora_customer *t, uint64_t seed, int64_t n_global,
                             uint32_t rank, uint32_t nranks);
 void ora_free_lineitem(ora_lineitem *t);

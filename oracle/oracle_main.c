@@ -73,7 +73,7 @@ int main(int argc, char **argv)
     } else if (!strcmp(query, "q3")) {
         ora_orders od;
         ora_customer cu;
-        if (ora_gen_orders(&od, seed, rows / 4, rows / 40, rank, nranks) ||
+        if (ora_gen_orders(&od, seed, rows / 4, rows / 40, rank, nranks, 0) ||
             ora_gen_customer(&cu, seed, rows / 40, rank, nranks)) {
             fprintf(stderr, "gen failed\n"); return 1;
         }

@@ -111,7 +111,8 @@ def _p(arr, ctype):
     return arr.ctypes.data_as(C.POINTER(ctype))
 
 
-def gen_tables(n_lineitem, rank=0, nranks=1, seed=42, need=("lineitem",)):
+def gen_tables(n_lineitem, rank=0, nranks=1, seed=42, need=("lineitem",),
+               skew=False):
     """Generate tables via the oracle's generator; returns dict of numpy views
     plus the C structs (kept alive). n_lineitem is the GLOBAL row count."""
     L = lib()
@@ -138,7 +139,8 @@ def gen_tables(n_lineitem, rank=0, nranks=1, seed=42, need=("lineitem",)):
     if "orders" in need:
         t = _Orders()
         st = L.ora_gen_orders(C.byref(t), C.c_uint64(seed), C.c_int64(n_lineitem // 4),
-                              C.c_int64(n_lineitem // 40), rank, nranks)
+                              C.c_int64(n_lineitem // 40), rank, nranks,
+                              1 if skew else 0)
         assert st == 0, st
         n = t.n
         out["orders"] = {

@@ -141,6 +141,20 @@ OTBX_FN int64_t otbx_ord_custkey(uint64_t seed, uint64_t i, int64_t ncust)
     return 1 + (int64_t)(r % (uint64_t)ncust);
 }
 
+/* BASELINE config 5: skewed distribution keys — 20% of custkeys ("hot",
+ * keys 1..ncust/5) receive 80% of the orders. Pure integer arithmetic:
+ * bit-identical on CPU and GPU. */
+OTBX_FN int64_t otbx_ord_custkey_skewed(uint64_t seed, uint64_t i, int64_t ncust)
+{
+    uint64_t r = otbx_rnd(seed, OTBX_T_ORDERS, OTBX_C_CUSTKEY, i);
+    uint64_t r2 = otbx_splitmix64(r);
+    int64_t nhot = ncust / 5 > 0 ? ncust / 5 : 1;
+    if ((int32_t)(r % 10) < 8)
+        return 1 + (int64_t)(r2 % (uint64_t)nhot);
+    int64_t ncold = ncust - nhot > 0 ? ncust - nhot : 1;
+    return 1 + nhot + (int64_t)(r2 % (uint64_t)ncold);
+}
+
 OTBX_FN int32_t otbx_ord_orderdate(uint64_t seed, uint64_t i)
 {
     uint64_t r = otbx_rnd(seed, OTBX_T_ORDERS, OTBX_C_ODATE, i);

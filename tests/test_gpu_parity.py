@@ -305,3 +305,30 @@ def test_golden_tenk_join_gpu(ex):
     node = ex.GpuHashJoin(dev(bk), dev(pk), cap_pairs=200000)
     pairs = drain(node)
     assert len(pairs) == expjs["tenk_selfjoin"]["count"]
+
+
+# ---------------- config-5 skewed distribution keys ----------------
+
+def test_skewed_custkey_datagen_and_q3_parity(ex, ora):
+    """Skewed orders (20% hot custkeys get 80% of orders, config 5):
+    bit-identical GPU/CPU generation and full Q3 result parity."""
+    n = 400000
+    od = ex.GpuOrders.generate(n // 4, n // 40, skew=True)
+    ot = ora.gen_tables(n, need=("orders",), skew=True)["orders"]
+    ck = od.t["o_custkey"].cpu().numpy()
+    assert np.array_equal(ck, ot["o_custkey"])
+    # skew property: hot 20% of custkeys hold ~80% of orders
+    ncust = n // 40
+    hot = (ck <= ncust // 5).mean()
+    assert 0.75 < hot < 0.85
+    li = ex.GpuLineitem.generate(n)
+    cu = ex.GpuCustomer.generate(n // 40)
+    node = ex.GpuQ3Fragment(cu, od, li)
+    drain(node)
+    t = ora.gen_tables(n, need=("lineitem", "orders", "customer"), skew=True)
+    og = ora.q3_partial(t)
+    got = {int(r["l_orderkey"]): float(r["revenue"]) for r in node.fetch_groups()}
+    exp = dict(zip(og["l_orderkey"].tolist(), og["revenue"].tolist()))
+    assert got.keys() == exp.keys()
+    for k in list(exp)[:200]:
+        assert approx(got[k], exp[k])
