@@ -996,18 +996,25 @@ __global__ void k_ord_probe_cust(const otbx_orders_dev o,
                                  const int64_t *__restrict__ ncand_p,
                                  const unsigned long long *__restrict__ ckeys,
                                  int64_t ccap, int64_t *__restrict__ out,
-                                 int64_t *nout)
+                                 int64_t *nout, unsigned long long *minkey,
+                                 unsigned long long *maxkey)
 {
     const int BUF = 1024;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
     int nbuf = 0;
+    unsigned long long mymin = ~0ull >> 1, mymax = 0;
     int64_t n = *ncand_p;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
          ci += stride) {
         int64_t i = cand[ci];
         bool m = d_keyset_probe(ckeys, ccap, o.o_custkey[i]); /* ⋈ customer */
+        if (m) {
+            unsigned long long k = (unsigned long long)o.o_orderkey[i];
+            if (k < mymin) mymin = k;
+            if (k > mymax) mymax = k;
+        }
         unsigned long long mask = __ballot(m);
         int cnt = __popcll(mask);
         if (nbuf + cnt > BUF) {
@@ -1035,6 +1042,19 @@ __global__ void k_ord_probe_cust(const otbx_orders_dev o,
         for (int j = lane; j < nbuf; j += WAVE)
             out[base + j] = buf[wid][j];
     }
+    /* matched-key range for the dense-direct path decision */
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        unsigned long long mn = (unsigned long long)__shfl_down(
+            (long long)mymin, off, WAVE);
+        unsigned long long mx = (unsigned long long)__shfl_down(
+            (long long)mymax, off, WAVE);
+        if (mn < mymin) mymin = mn;
+        if (mx > mymax) mymax = mx;
+    }
+    if (lane == 0) {
+        if (mymin != (~0ull >> 1)) atomicMin(minkey, mymin);
+        if (mymax) atomicMax(maxkey, mymax);
+    }
 }
 
 __global__ void k_ord_insert(const otbx_orders_dev o,
@@ -1056,6 +1076,175 @@ __global__ void k_ord_insert(const otbx_orders_dev o,
             s = (s + 1) & mask;      /* keys unique: claim exactly one slot */
         tab[s].date = o.o_orderdate[i];   /* plain: read by NEXT launch */
         tab[s].prio = o.o_shippriority[i];
+    }
+}
+
+/* ---- dense-orderkey DIRECT path ----
+ * When the filtered build side's key range (max-min+1) fits
+ * OTBX_DIRECT_CAP, hashing is strictly worse than direct addressing: the
+ * probe stream arrives in (clustered) key order, so un-hashed bitmap/table
+ * indices give consecutive probes the SAME cache lines — the random-line
+ * traffic that bounds the hash path collapses to streaming. The reference
+ * always hashes (nodeHash.c); on MI355X the locality is worth preserving.
+ * Fallback to the bloom+hash path above when the range is too wide
+ * (sparse keys). Empty marker in ptab: packed payload is
+ * date | prio<<32 with date ≥ 1, so 0 = empty. */
+#define OTBX_DIRECT_CAP_DEFAULT (1ll << 29)
+
+__global__ void k_ord_insert_direct(const otbx_orders_dev o,
+                                    const int64_t *__restrict__ cand,
+                                    const int64_t *__restrict__ ncand_p,
+                                    int64_t mino, unsigned long long *bitmap,
+                                    unsigned long long *ptab)
+{
+    int64_t n = *ncand_p;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
+         ci += stride) {
+        int64_t i = cand[ci];
+        int64_t idx = o.o_orderkey[i] - mino;
+        atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
+        ptab[idx] = (unsigned long long)(uint32_t)o.o_orderdate[i] |
+                    ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
+    }
+}
+
+__global__ void k_q3_scan_filter_direct(const otbx_lineitem_dev l,
+                                        int32_t q3date, int64_t mino,
+                                        int64_t range,
+                                        const unsigned long long *__restrict__ bitmap,
+                                        int64_t *__restrict__ cand,
+                                        int64_t *ncand)
+{
+    /* same LDS-buffered append as k_q3_scan_filter; the filter is one
+     * un-hashed bitmap bit — adjacent lanes hit the same word (L1/L2). */
+    const int BUF = 1024;
+    const int B = 8;
+    __shared__ int64_t buf[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0;
+    int64_t chunk = (int64_t)blockDim.x * B;
+    int64_t stride = (int64_t)gridDim.x * chunk;
+    for (int64_t base = (int64_t)blockIdx.x * chunk; base < l.n;
+         base += stride) {
+        bool m[B];
+        int64_t rows[B];
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            int64_t i = base + (int64_t)k * blockDim.x + threadIdx.x;
+            rows[k] = i;
+            bool pass = false;
+            int64_t idx = 0;
+            if (i < l.n && l.l_shipdate[i] > q3date) {
+                idx = l.l_orderkey[i] - mino;
+                pass = idx >= 0 && idx < range;
+                if (!pass) idx = 0;
+            }
+            unsigned long long w = bitmap[idx >> 6];
+            m[k] = pass && ((w >> (idx & 63)) & 1ull);
+        }
+#pragma unroll
+        for (int k = 0; k < B; k++) {
+            unsigned long long mask = __ballot(m[k]);
+            int cnt = __popcll(mask);
+            if (nbuf + cnt > BUF) {
+                long long bpos = 0;
+                if (lane == 0)
+                    bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                                (unsigned long long)nbuf);
+                bpos = __shfl(bpos, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE)
+                    cand[bpos + j] = buf[wid][j];
+                nbuf = 0;
+            }
+            if (m[k]) {
+                int rank = __popcll(mask & ((1ull << lane) - 1ull));
+                buf[wid][nbuf + rank] = rows[k];
+            }
+            nbuf += cnt;
+        }
+    }
+    if (nbuf) {
+        long long bpos = 0;
+        if (lane == 0)
+            bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                        (unsigned long long)nbuf);
+        bpos = __shfl(bpos, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE)
+            cand[bpos + j] = buf[wid][j];
+    }
+}
+
+__global__ void k_q3_probe_agg_direct(const otbx_lineitem_dev l,
+                                      const int64_t *__restrict__ cand,
+                                      const int64_t *__restrict__ ncand_p,
+                                      int64_t mino,
+                                      const unsigned long long *__restrict__ ptab,
+                                      double *__restrict__ rtab,
+                                      unsigned long long *__restrict__ nhits)
+{
+    int64_t n = *ncand_p;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long myhits = 0;
+    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
+         ci += stride) {
+        int64_t i = cand[ci];
+        int64_t idx = l.l_orderkey[i] - mino;
+        unsigned long long pl = ptab[idx];
+        if (pl == 0ull)
+            continue; /* bitmap false positive cannot happen (exact), but an
+                       * unset payload can if a later rank pattern reuses the
+                       * bit range — keep the guard for safety */
+        myhits++;
+        double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
+        atomicAdd(&rtab[idx], rev);
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        myhits += __shfl_down(myhits, off, WAVE);
+    if ((threadIdx.x % WAVE) == 0 && myhits)
+        atomicAdd(nhits, myhits);
+}
+
+__global__ void k_q3_compact_direct(const double *__restrict__ rtab,
+                                    const unsigned long long *__restrict__ ptab,
+                                    int64_t range, int64_t mino,
+                                    otbx_q3_group *out, int64_t cap_out,
+                                    int64_t *ngroups)
+{
+    int64_t per_block = (range + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < range ? lo + per_block : range;
+    __shared__ int64_t tcnt[256];
+    __shared__ int64_t tbase[257];
+    int64_t my = 0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        my += rtab[i] != 0.0;
+    tcnt[threadIdx.x] = my;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int64_t tot = 0;
+        for (int t = 0; t < (int)blockDim.x; t++) {
+            tbase[t] = tot;
+            tot += tcnt[t];
+        }
+        tbase[256] = tot ? (int64_t)atomicAdd((unsigned long long *)ngroups,
+                                              (unsigned long long)tot)
+                         : 0;
+    }
+    __syncthreads();
+    int64_t pos = tbase[256] + tbase[threadIdx.x];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        double rev = rtab[i];
+        if (rev != 0.0) {
+            if (pos < cap_out) {
+                unsigned long long pl = ptab[i];
+                out[pos].l_orderkey = mino + i;
+                out[pos].revenue = rev;
+                out[pos].o_orderdate = (int32_t)(pl & 0xffffffffull);
+                out[pos].o_shippriority = (int32_t)(pl >> 32);
+            }
+            pos++;
+        }
     }
 }
 
@@ -1365,6 +1554,15 @@ static int64_t fit_cap(int64_t n)
     return next_pow2_host(n < 16 ? 16 : (int64_t)(n / 0.7) + 1);
 }
 
+/* dense-direct path admission: key range the direct tables can cover.
+ * Deterministic in the inputs so workspace_bytes and q3_partial agree. */
+static int64_t direct_cap_for(int64_t norders)
+{
+    int64_t cap = next_pow2_host(norders < 16 ? 16 : norders) * 32;
+    if (cap > OTBX_DIRECT_CAP_DEFAULT) cap = OTBX_DIRECT_CAP_DEFAULT;
+    return cap;
+}
+
 otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
                                     int64_t nlineitem, size_t *bytes)
 {
@@ -1372,10 +1570,14 @@ otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
      * right-sized at run time (count-then-build) so only the live prefix is
      * cleared/touched */
     int64_t ccap_w = fit_cap(ncust), ocap_w = fit_cap(norders);
+    int64_t dcap = direct_cap_for(norders);
     *bytes = 64 + (size_t)ccap_w * 8 + ((size_t)1 << 23) * 8 /* bloom */ +
              (size_t)ocap_w * sizeof(ord_slot) + (size_t)ocap_w * sizeof(q3g_slot) +
              (size_t)nlineitem * 8 /* lineitem candidates */ +
-             (size_t)norders * 16 /* orders candidate lists (2) */;
+             (size_t)norders * 16 /* orders candidate lists (2) */ +
+             (size_t)(dcap / 8 + 64) /* direct bitmap */ +
+             (size_t)dcap * 8 /* direct payload */ +
+             (size_t)dcap * 8 /* direct revenue */;
     return OTBX_OK;
 }
 
@@ -1412,12 +1614,17 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     int64_t *cand_li = (int64_t *)((char *)gtab + (size_t)ocap_w * sizeof(q3g_slot));
     int64_t *cand_o1 = cand_li + l->n;
     int64_t *cand_o2 = cand_o1 + o->n;
+    int64_t dcap = direct_cap_for(o->n);
+    unsigned long long *dbitmap = (unsigned long long *)(cand_o2 + o->n);
+    unsigned long long *dptab = dbitmap + dcap / 64 + 8;
+    double *drtab = (double *)(dptab + dcap);
 
     static int64_t *h_cnt = nullptr;        /* pinned host readback */
     if (!h_cnt)
-        HIP_CHECK(hipHostMalloc(&h_cnt, 4 * sizeof(int64_t)));
+        HIP_CHECK(hipHostMalloc(&h_cnt, 8 * sizeof(int64_t)));
 
     HIP_CHECK(hipMemsetAsync(hdr, 0, 64, s));
+    HIP_CHECK(hipMemsetAsync(&hdr[4], 0x7f, 8, s)); /* min-key accumulator */
     HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, sizeof(int64_t), s));
     unsigned long long *nhits = nullptr;
     if (stats_dev) {
@@ -1465,10 +1672,37 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                        0, s, *o, q3date, cand_o1, &hdr[3]);
     hipLaunchKernelGGL(k_ord_probe_cust, dim3(grid_for(o->n, 256)), dim3(256),
                        0, s, *o, cand_o1, &hdr[3], ctab, ccap, cand_o2,
-                       &hdr[1]);
-    HIP_CHECK(hipMemcpyAsync(h_cnt + 1, hdr + 1, 8, hipMemcpyDeviceToHost, s));
+                       &hdr[1], (unsigned long long *)&hdr[4],
+                       (unsigned long long *)&hdr[5]);
+    HIP_CHECK(hipMemcpyAsync(h_cnt, hdr, 48, hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     int64_t nof = h_cnt[1];
+    int64_t mino = h_cnt[4], maxo = h_cnt[5];
+    int64_t range = maxo - mino + 1;
+    bool use_direct = nof > 0 && range > 0 && range <= dcap &&
+                      !getenv("OTBX_Q3_FORCE_HASH");
+    if (use_direct) {
+        /* dense-orderkey direct path: bitmap filter + direct-addressed
+         * payload/revenue tables preserve the probe stream's key locality */
+        HIP_CHECK(hipMemsetAsync(dbitmap, 0, (size_t)(range / 64 + 8) * 8, s));
+        HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)range * 8, s));
+        HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)range * 8, s));
+        hipLaunchKernelGGL(k_ord_insert_direct, dim3(grid_for(o->n, 256)),
+                           dim3(256), 0, s, *o, cand_o2, &hdr[1], mino,
+                           dbitmap, dptab);
+        if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
+        hipLaunchKernelGGL(k_q3_scan_filter_direct, dim3(grid_for(l->n, 256)),
+                           dim3(256), 0, s, *l, q3date, mino, range, dbitmap,
+                           cand_li, &hdr[2]);
+        hipLaunchKernelGGL(k_q3_probe_agg_direct, dim3(grid_for(l->n, 256)),
+                           dim3(256), 0, s, *l, cand_li, &hdr[2], mino, dptab,
+                           drtab, nhits);
+        if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
+        hipLaunchKernelGGL(k_q3_compact_direct, dim3(grid_for(range, 256)),
+                           dim3(256), 0, s, drtab, dptab, range, mino,
+                           groups_dev, cap_groups, ngroups_dev);
+        if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
+    } else {
     int64_t ocap = fit_cap(nof), gcap = ocap;
     int64_t bwords = bloom_words_for(nof);
     HIP_CHECK(hipMemsetAsync(bloom, 0, (size_t)bwords * 8, s));
@@ -1490,6 +1724,7 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     hipLaunchKernelGGL(k_q3_compact, dim3(grid_for(gcap, 256)), dim3(256), 0, s,
                        gtab, gcap, groups_dev, cap_groups, ngroups_dev);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
+    }
     HIP_CHECK(hipGetLastError());
     if (kernel_ms) {
         HIP_CHECK(hipEventSynchronize(ev[4]));
