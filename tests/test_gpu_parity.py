@@ -332,3 +332,33 @@ def test_skewed_custkey_datagen_and_q3_parity(ex, ora):
     assert got.keys() == exp.keys()
     for k in list(exp)[:200]:
         assert approx(got[k], exp[k])
+
+
+def test_q3_hash_fallback_path(ex, ora):
+    """The wide-range hash+bloom fallback (forced via env) produces identical
+    results to the dense-direct default path and the oracle."""
+    import os
+    n = 400000
+    li = ex.GpuLineitem.generate(n)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    cu = ex.GpuCustomer.generate(n // 40)
+    os.environ["OTBX_Q3_FORCE_HASH"] = "1"
+    try:
+        node = ex.GpuQ3Fragment(cu, od, li)
+        drain(node)
+        hash_groups = {int(k): float(v) for k, v in
+                       zip(node.fetch_groups()["l_orderkey"],
+                           node.fetch_groups()["revenue"])}
+    finally:
+        del os.environ["OTBX_Q3_FORCE_HASH"]
+    node2 = ex.GpuQ3Fragment(cu, od, li)
+    drain(node2)
+    direct_groups = {int(k): float(v) for k, v in
+                     zip(node2.fetch_groups()["l_orderkey"],
+                         node2.fetch_groups()["revenue"])}
+    assert hash_groups.keys() == direct_groups.keys()
+    for k in hash_groups:
+        assert approx(hash_groups[k], direct_groups[k])
+    og = q3_oracle(ora, n)
+    exp = dict(zip(og["l_orderkey"].tolist(), og["revenue"].tolist()))
+    assert direct_groups.keys() == exp.keys()
