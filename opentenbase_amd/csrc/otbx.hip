@@ -1116,38 +1116,59 @@ __global__ void k_q3_scan_filter_direct(const otbx_lineitem_dev l,
                                         int64_t *__restrict__ cand,
                                         int64_t *ncand)
 {
-    /* same LDS-buffered append as k_q3_scan_filter; the filter is one
-     * un-hashed bitmap bit — adjacent lanes hit the same word (L1/L2). */
+    /* 4 consecutive rows per lane via 16-B vector loads (the Q1 kernel's
+     * coalescing pattern: lanes stride 16 B); the bitmap filter is an
+     * un-hashed bit so a wave's 1024 rows touch ~4 bitmap words (L1).
+     * Candidate append is ORDER-PRESERVING (wave prefix sum over per-lane
+     * counts) with per-wave LDS staging, one global atomic per 1024. */
     const int BUF = 1024;
-    const int B = 8;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
-    int nbuf = 0;
-    int64_t chunk = (int64_t)blockDim.x * B;
-    int64_t stride = (int64_t)gridDim.x * chunk;
-    for (int64_t base = (int64_t)blockIdx.x * chunk; base < l.n;
-         base += stride) {
-        bool m[B];
-        int64_t rows[B];
+    int nbuf = 0; /* wave-uniform */
+    int64_t nq = l.n / 4;
+    const int4 *sd4 = (const int4 *)l.l_shipdate;
+    const longlong2 *ok2 = (const longlong2 *)l.l_orderkey;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;; q += stride) {
+        bool m[4];
+        int64_t r0 = q * 4;
+        int mycnt = 0;
+        if (q < nq) {
+            int4 d = sd4[q];
+            longlong2 ka = ok2[2 * q], kb = ok2[2 * q + 1];
+            int32_t ds[4] = {d.x, d.y, d.z, d.w};
+            int64_t ky[4] = {ka.x, ka.y, kb.x, kb.y};
 #pragma unroll
-        for (int k = 0; k < B; k++) {
-            int64_t i = base + (int64_t)k * blockDim.x + threadIdx.x;
-            rows[k] = i;
-            bool pass = false;
-            int64_t idx = 0;
-            if (i < l.n && l.l_shipdate[i] > q3date) {
-                idx = l.l_orderkey[i] - mino;
-                pass = idx >= 0 && idx < range;
-                if (!pass) idx = 0;
+            for (int j = 0; j < 4; j++) {
+                int64_t idx = ky[j] - mino;
+                bool pass = (ds[j] > q3date) && idx >= 0 && idx < range;
+                unsigned long long w = bitmap[pass ? (idx >> 6) : 0];
+                m[j] = pass && ((w >> (idx & 63)) & 1ull);
+                mycnt += m[j];
             }
-            unsigned long long w = bitmap[idx >> 6];
-            m[k] = pass && ((w >> (idx & 63)) & 1ull);
-        }
+        } else {
 #pragma unroll
-        for (int k = 0; k < B; k++) {
-            unsigned long long mask = __ballot(m[k]);
-            int cnt = __popcll(mask);
-            if (nbuf + cnt > BUF) {
+            for (int j = 0; j < 4; j++) m[j] = false;
+        }
+        /* tail rows (l.n % 4) handled by the lane owning q == nq */
+        if (q == nq) {
+            for (int64_t i = nq * 4; i < l.n; i++) {
+                int j = (int)(i - nq * 4);
+                int64_t idx = l.l_orderkey[i] - mino;
+                bool pass = (l.l_shipdate[i] > q3date) && idx >= 0 && idx < range;
+                m[j] = pass && ((bitmap[pass ? (idx >> 6) : 0] >> (idx & 63)) & 1ull);
+                mycnt += m[j];
+            }
+        }
+        /* wave-exclusive prefix for order-preserving append */
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        int tot = __shfl(incl, WAVE - 1, WAVE);
+        if (tot) {
+            if (nbuf + tot > BUF) {
                 long long bpos = 0;
                 if (lane == 0)
                     bpos = (long long)atomicAdd((unsigned long long *)ncand,
@@ -1157,12 +1178,16 @@ __global__ void k_q3_scan_filter_direct(const otbx_lineitem_dev l,
                     cand[bpos + j] = buf[wid][j];
                 nbuf = 0;
             }
-            if (m[k]) {
-                int rank = __popcll(mask & ((1ull << lane) - 1ull));
-                buf[wid][nbuf + rank] = rows[k];
-            }
-            nbuf += cnt;
+            int pos = nbuf + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j])
+                    buf[wid][pos++] = r0 + j;
+            nbuf += tot;
         }
+        /* uniform loop exit AFTER the tail/flush work */
+        if (__all(q >= nq))
+            break;
     }
     if (nbuf) {
         long long bpos = 0;

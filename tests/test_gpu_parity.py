@@ -155,8 +155,8 @@ def q3_oracle(ora, n, rank=0, nranks=1, replicate_customer=False):
     return ora.q3_partial(t)
 
 
-def test_q3_parity(ex, ora):
-    n = 400000
+@pytest.mark.parametrize("n", [400000, 400004, 399998])
+def test_q3_parity(ex, ora, n):
     li = ex.GpuLineitem.generate(n)
     od = ex.GpuOrders.generate(n // 4, n // 40)
     cu = ex.GpuCustomer.generate(n // 40)
