@@ -954,40 +954,64 @@ __device__ __forceinline__ bool d_bloom_test(const unsigned long long *bloom,
 __global__ void k_ord_filter_date(const otbx_orders_dev o, int32_t q3date,
                                   int64_t *__restrict__ out, int64_t *ncand)
 {
+    /* 4 rows per lane via one int4 load; per-wave LDS-staged append */
     const int BUF = 1024;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
     int nbuf = 0;
+    int64_t nq = o.n / 4;
+    const int4 *od4 = (const int4 *)o.o_orderdate;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < o.n;
-         i += stride) {
-        bool m = o.o_orderdate[i] < q3date;       /* ExecQual on orders */
-        unsigned long long mask = __ballot(m);
-        int cnt = __popcll(mask);
-        if (nbuf + cnt > BUF) {
-            long long base = 0;
-            if (lane == 0)
-                base = (long long)atomicAdd((unsigned long long *)ncand,
-                                            (unsigned long long)nbuf);
-            base = __shfl(base, 0, WAVE);
-            for (int j = lane; j < nbuf; j += WAVE)
-                out[base + j] = buf[wid][j];
-            nbuf = 0;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         q += stride) {
+        bool m[4] = {false, false, false, false};
+        int64_t r0 = q * 4;
+        if (q < nq) {
+            int4 d = od4[q];
+            m[0] = d.x < q3date;
+            m[1] = d.y < q3date;
+            m[2] = d.z < q3date;
+            m[3] = d.w < q3date;
+        } else if (q == nq) {
+            for (int64_t i = nq * 4; i < o.n; i++)
+                m[(int)(i - nq * 4)] = o.o_orderdate[i] < q3date;
         }
-        if (m) {
-            int rank = __popcll(mask & ((1ull << lane) - 1ull));
-            buf[wid][nbuf + rank] = i;
+        int mycnt = m[0] + m[1] + m[2] + m[3];
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
         }
-        nbuf += cnt;
+        int tot = __shfl(incl, WAVE - 1, WAVE);
+        if (tot) {
+            if (nbuf + tot > BUF) {
+                long long bpos = 0;
+                if (lane == 0)
+                    bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                                (unsigned long long)nbuf);
+                bpos = __shfl(bpos, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE)
+                    out[bpos + j] = buf[wid][j];
+                nbuf = 0;
+            }
+            int pos = nbuf + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j])
+                    buf[wid][pos++] = r0 + j;
+            nbuf += tot;
+        }
+        if (__all(q >= nq))
+            break;
     }
     if (nbuf) {
-        long long base = 0;
+        long long bpos = 0;
         if (lane == 0)
-            base = (long long)atomicAdd((unsigned long long *)ncand,
+            bpos = (long long)atomicAdd((unsigned long long *)ncand,
                                         (unsigned long long)nbuf);
-        base = __shfl(base, 0, WAVE);
+        bpos = __shfl(bpos, 0, WAVE);
         for (int j = lane; j < nbuf; j += WAVE)
-            out[base + j] = buf[wid][j];
+            out[bpos + j] = buf[wid][j];
     }
 }
 
@@ -995,7 +1019,10 @@ __global__ void k_ord_probe_cust(const otbx_orders_dev o,
                                  const int64_t *__restrict__ cand,
                                  const int64_t *__restrict__ ncand_p,
                                  const unsigned long long *__restrict__ ckeys,
-                                 int64_t ccap, int64_t *__restrict__ out,
+                                 int64_t ccap,
+                                 const unsigned long long *__restrict__ cbitmap,
+                                 int64_t cmin, int64_t crange,
+                                 int64_t *__restrict__ out,
                                  int64_t *nout, unsigned long long *minkey,
                                  unsigned long long *maxkey)
 {
@@ -1009,7 +1036,14 @@ __global__ void k_ord_probe_cust(const otbx_orders_dev o,
     for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
          ci += stride) {
         int64_t i = cand[ci];
-        bool m = d_keyset_probe(ckeys, ccap, o.o_custkey[i]); /* ⋈ customer */
+        bool m;
+        if (cbitmap) {
+            int64_t idx = o.o_custkey[i] - cmin;
+            m = idx >= 0 && idx < crange &&
+                ((cbitmap[idx >> 6] >> (idx & 63)) & 1ull);
+        } else {
+            m = d_keyset_probe(ckeys, ccap, o.o_custkey[i]); /* ⋈ customer */
+        }
         if (m) {
             unsigned long long k = (unsigned long long)o.o_orderkey[i];
             if (k < mymin) mymin = k;
@@ -1528,17 +1562,88 @@ __global__ void k_filter_customer(const otbx_customer_dev c, uint8_t want,
 /* count rows passing {segment} / {date + customer-keyset} predicates —
  * sizes the right-fit hash tables before building (count-then-build). */
 __global__ void k_count_customer_seg(const otbx_customer_dev c, uint8_t want,
-                                     int64_t *count)
+                                     int64_t *count,
+                                     unsigned long long *minkey,
+                                     unsigned long long *maxkey)
 {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    unsigned long long my = 0;
+    unsigned long long my = 0, mymin = ~0ull >> 1, mymax = 0;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < c.n;
-         i += stride)
-        my += c.c_mktsegment[i] == want;
-    for (int off = WAVE / 2; off > 0; off >>= 1)
+         i += stride) {
+        if (c.c_mktsegment[i] == want) {
+            my++;
+            unsigned long long k = (unsigned long long)c.c_custkey[i];
+            if (k < mymin) mymin = k;
+            if (k > mymax) mymax = k;
+        }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
         my += __shfl_down(my, off, WAVE);
-    if ((threadIdx.x % WAVE) == 0 && my)
-        atomicAdd((unsigned long long *)count, my);
+        unsigned long long mn = (unsigned long long)__shfl_down(
+            (long long)mymin, off, WAVE);
+        unsigned long long mx = (unsigned long long)__shfl_down(
+            (long long)mymax, off, WAVE);
+        if (mn < mymin) mymin = mn;
+        if (mx > mymax) mymax = mx;
+    }
+    if ((threadIdx.x % WAVE) == 0) {
+        if (my) atomicAdd((unsigned long long *)count, my);
+        if (mymin != (~0ull >> 1)) atomicMin(minkey, mymin);
+        if (mymax) atomicMax(maxkey, mymax);
+    }
+}
+
+__global__ void k_minmax_i64(const int64_t *__restrict__ keys, int64_t n,
+                             unsigned long long *minkey,
+                             unsigned long long *maxkey)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long mymin = ~0ull >> 1, mymax = 0;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        unsigned long long k = (unsigned long long)keys[i];
+        if (k < mymin) mymin = k;
+        if (k > mymax) mymax = k;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        unsigned long long mn = (unsigned long long)__shfl_down(
+            (long long)mymin, off, WAVE);
+        unsigned long long mx = (unsigned long long)__shfl_down(
+            (long long)mymax, off, WAVE);
+        if (mn < mymin) mymin = mn;
+        if (mx > mymax) mymax = mx;
+    }
+    if ((threadIdx.x % WAVE) == 0) {
+        if (mymin != (~0ull >> 1)) atomicMin(minkey, mymin);
+        if (mymax) atomicMax(maxkey, mymax);
+    }
+}
+
+/* customer-key bitmap build (dense-custkey direct filter; 1-2 MB and
+ * L2-resident vs a 64+ MB hashed keyset — same reasoning as the orderkey
+ * direct path) */
+__global__ void k_cust_bitmap_filter(const otbx_customer_dev c, uint8_t want,
+                                     int64_t minc, unsigned long long *bitmap)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < c.n;
+         i += stride) {
+        if (c.c_mktsegment[i] == want) {
+            int64_t idx = c.c_custkey[i] - minc;
+            atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
+        }
+    }
+}
+
+__global__ void k_cust_bitmap_keys(const int64_t *__restrict__ keys, int64_t n,
+                                   int64_t minc, unsigned long long *bitmap)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        int64_t idx = keys[i] - minc;
+        atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
+    }
 }
 
 __global__ void k_count_orders_filtered(const otbx_orders_dev o,
@@ -1596,13 +1701,15 @@ otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
      * cleared/touched */
     int64_t ccap_w = fit_cap(ncust), ocap_w = fit_cap(norders);
     int64_t dcap = direct_cap_for(norders);
+    int64_t dcap_c = direct_cap_for(ncust);
     *bytes = 64 + (size_t)ccap_w * 8 + ((size_t)1 << 23) * 8 /* bloom */ +
              (size_t)ocap_w * sizeof(ord_slot) + (size_t)ocap_w * sizeof(q3g_slot) +
              (size_t)nlineitem * 8 /* lineitem candidates */ +
              (size_t)norders * 16 /* orders candidate lists (2) */ +
-             (size_t)(dcap / 8 + 64) /* direct bitmap */ +
+             (size_t)(dcap / 8 + 64) /* direct orderkey bitmap */ +
              (size_t)dcap * 8 /* direct payload */ +
-             (size_t)dcap * 8 /* direct revenue */;
+             (size_t)dcap * 8 /* direct revenue */ +
+             (size_t)(dcap_c / 8 + 64) /* direct custkey bitmap */;
     return OTBX_OK;
 }
 
@@ -1640,16 +1747,19 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     int64_t *cand_o1 = cand_li + l->n;
     int64_t *cand_o2 = cand_o1 + o->n;
     int64_t dcap = direct_cap_for(o->n);
+    int64_t dcap_c = direct_cap_for(ncust);
     unsigned long long *dbitmap = (unsigned long long *)(cand_o2 + o->n);
     unsigned long long *dptab = dbitmap + dcap / 64 + 8;
     double *drtab = (double *)(dptab + dcap);
+    unsigned long long *cbitmap_buf = (unsigned long long *)(drtab + dcap);
 
     static int64_t *h_cnt = nullptr;        /* pinned host readback */
     if (!h_cnt)
         HIP_CHECK(hipHostMalloc(&h_cnt, 8 * sizeof(int64_t)));
 
     HIP_CHECK(hipMemsetAsync(hdr, 0, 64, s));
-    HIP_CHECK(hipMemsetAsync(&hdr[4], 0x7f, 8, s)); /* min-key accumulator */
+    HIP_CHECK(hipMemsetAsync(&hdr[4], 0x7f, 8, s)); /* min orderkey */
+    HIP_CHECK(hipMemsetAsync(&hdr[6], 0x7f, 8, s)); /* min custkey  */
     HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, sizeof(int64_t), s));
     unsigned long long *nhits = nullptr;
     if (stats_dev) {
@@ -1667,27 +1777,54 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
         for (int i = 0; i < 5; i++) HIP_CHECK(hipEventCreate(&ev[i]));
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[0], s));
 
-    /* ---- phase 1: size + build the customer keyset (count-then-build) */
-    int64_t ccap;
-    if (cust_keys_dev) {
-        ccap = fit_cap(ncust_keys);
-    } else {
-        hipLaunchKernelGGL(k_count_customer_seg, dim3(grid_for(c->n, 256)),
-                           dim3(256), 0, s, *c, segment, &hdr[0]);
-        HIP_CHECK(hipMemcpyAsync(h_cnt, hdr, 8, hipMemcpyDeviceToHost, s));
-        HIP_CHECK(hipStreamSynchronize(s));
-        ccap = fit_cap(h_cnt[0]);
-    }
-    HIP_CHECK(hipMemsetAsync(ctab, 0, (size_t)ccap * 8, s));
+    /* ---- phase 1: customer build side — count + key range, then either a
+     * dense custkey BITMAP (direct filter, L2-resident) or the hashed
+     * keyset fallback for wide key ranges */
+    bool force_hash = getenv("OTBX_Q3_FORCE_HASH") != nullptr;
+    int64_t ccap = 16, cmin = 0, crange = 0;
+    unsigned long long *cbitmap = nullptr;
     if (cust_keys_dev) {
         if (ncust_keys > 0)
-            hipLaunchKernelGGL(k_keyset_build, dim3(grid_for(ncust_keys, 256)),
-                               dim3(256), 0, s, cust_keys_dev, ncust_keys, ctab,
-                               ccap);
+            hipLaunchKernelGGL(k_minmax_i64, dim3(grid_for(ncust_keys, 256)),
+                               dim3(256), 0, s, cust_keys_dev, ncust_keys,
+                               (unsigned long long *)&hdr[6],
+                               (unsigned long long *)&hdr[7]);
     } else {
-        hipLaunchKernelGGL(k_keyset_build_filter, dim3(grid_for(c->n, 256)),
-                           dim3(256), 0, s, c->c_custkey, c->c_mktsegment,
-                           segment, c->n, ctab, ccap);
+        hipLaunchKernelGGL(k_count_customer_seg, dim3(grid_for(c->n, 256)),
+                           dim3(256), 0, s, *c, segment, &hdr[0],
+                           (unsigned long long *)&hdr[6],
+                           (unsigned long long *)&hdr[7]);
+    }
+    HIP_CHECK(hipMemcpyAsync(h_cnt, hdr, 64, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    int64_t ncust_f = cust_keys_dev ? ncust_keys : h_cnt[0];
+    cmin = h_cnt[6];
+    crange = h_cnt[7] - h_cnt[6] + 1;
+    bool cust_direct = ncust_f > 0 && crange > 0 && crange <= dcap_c &&
+                       !force_hash;
+    if (cust_direct) {
+        cbitmap = cbitmap_buf;
+        HIP_CHECK(hipMemsetAsync(cbitmap, 0, (size_t)(crange / 64 + 8) * 8, s));
+        if (cust_keys_dev)
+            hipLaunchKernelGGL(k_cust_bitmap_keys,
+                               dim3(grid_for(ncust_keys, 256)), dim3(256), 0, s,
+                               cust_keys_dev, ncust_keys, cmin, cbitmap);
+        else
+            hipLaunchKernelGGL(k_cust_bitmap_filter, dim3(grid_for(c->n, 256)),
+                               dim3(256), 0, s, *c, segment, cmin, cbitmap);
+    } else {
+        ccap = fit_cap(ncust_f > 0 ? ncust_f : 16);
+        HIP_CHECK(hipMemsetAsync(ctab, 0, (size_t)ccap * 8, s));
+        if (cust_keys_dev) {
+            if (ncust_keys > 0)
+                hipLaunchKernelGGL(k_keyset_build,
+                                   dim3(grid_for(ncust_keys, 256)), dim3(256),
+                                   0, s, cust_keys_dev, ncust_keys, ctab, ccap);
+        } else {
+            hipLaunchKernelGGL(k_keyset_build_filter, dim3(grid_for(c->n, 256)),
+                               dim3(256), 0, s, c->c_custkey, c->c_mktsegment,
+                               segment, c->n, ctab, ccap);
+        }
     }
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[1], s));
 
@@ -1696,16 +1833,16 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     hipLaunchKernelGGL(k_ord_filter_date, dim3(grid_for(o->n, 256)), dim3(256),
                        0, s, *o, q3date, cand_o1, &hdr[3]);
     hipLaunchKernelGGL(k_ord_probe_cust, dim3(grid_for(o->n, 256)), dim3(256),
-                       0, s, *o, cand_o1, &hdr[3], ctab, ccap, cand_o2,
-                       &hdr[1], (unsigned long long *)&hdr[4],
+                       0, s, *o, cand_o1, &hdr[3], ctab, ccap, cbitmap, cmin,
+                       crange, cand_o2, &hdr[1],
+                       (unsigned long long *)&hdr[4],
                        (unsigned long long *)&hdr[5]);
     HIP_CHECK(hipMemcpyAsync(h_cnt, hdr, 48, hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     int64_t nof = h_cnt[1];
     int64_t mino = h_cnt[4], maxo = h_cnt[5];
     int64_t range = maxo - mino + 1;
-    bool use_direct = nof > 0 && range > 0 && range <= dcap &&
-                      !getenv("OTBX_Q3_FORCE_HASH");
+    bool use_direct = nof > 0 && range > 0 && range <= dcap && !force_hash;
     if (use_direct) {
         /* dense-orderkey direct path: bitmap filter + direct-addressed
          * payload/revenue tables preserve the probe stream's key locality */
