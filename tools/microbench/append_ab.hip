@@ -1,0 +1,173 @@
+/* Microbench: isolate what bounds the LDS-staged candidate-append kernels
+ * (k_ord_filter_date shape: 150M int32 rows, ~49% selectivity).
+ * Variants: 0 = loads+predicate+popcount only (no append)
+ *           1 = + wave prefix sum (no LDS, no output)
+ *           2 = full LDS-staged order-preserving append (the product shape)
+ *           3 = plain per-lane global atomicAdd-free scatter via ballot
+ *               wave_append (order-free)
+ */
+#include <hip/hip_runtime.h>
+#include <stdio.h>
+#define WAVE 64
+
+__global__ void k_v0(const int32_t *__restrict__ d, int64_t n, int32_t c,
+                     unsigned long long *out)
+{
+    int64_t nq = n / 4;
+    const int4 *d4 = (const int4 *)d;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long my = 0;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
+         q += stride) {
+        int4 v = d4[q];
+        my += (v.x < c) + (v.y < c) + (v.z < c) + (v.w < c);
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        my += __shfl_down(my, off, WAVE);
+    if ((threadIdx.x % WAVE) == 0 && my) atomicAdd(out, my);
+}
+
+__global__ void k_v1(const int32_t *__restrict__ d, int64_t n, int32_t c,
+                     unsigned long long *out)
+{
+    int64_t nq = n / 4;
+    const int4 *d4 = (const int4 *)d;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int lane = (int)(threadIdx.x % WAVE);
+    unsigned long long acc = 0;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
+         q += stride) {
+        int4 v = d4[q];
+        int mycnt = (v.x < c) + (v.y < c) + (v.z < c) + (v.w < c);
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        acc += __shfl(incl, WAVE - 1, WAVE);
+    }
+    if (lane == 0) atomicAdd(out, acc);
+}
+
+__global__ void k_v2(const int32_t *__restrict__ d, int64_t n, int32_t c,
+                     int64_t *cand, int64_t *ncand)
+{
+    const int BUF = 1024;
+    __shared__ int64_t buf[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0;
+    int64_t nq = n / 4;
+    const int4 *d4 = (const int4 *)d;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         q += stride) {
+        bool m[4] = {false, false, false, false};
+        int64_t r0 = q * 4;
+        if (q < nq) {
+            int4 v = d4[q];
+            m[0] = v.x < c; m[1] = v.y < c; m[2] = v.z < c; m[3] = v.w < c;
+        }
+        int mycnt = m[0] + m[1] + m[2] + m[3];
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        int tot = __shfl(incl, WAVE - 1, WAVE);
+        if (tot) {
+            if (nbuf + tot > BUF) {
+                long long bpos = 0;
+                if (lane == 0)
+                    bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                                (unsigned long long)nbuf);
+                bpos = __shfl(bpos, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE)
+                    cand[bpos + j] = buf[wid][j];
+                nbuf = 0;
+            }
+            int pos = nbuf + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j]) buf[wid][pos++] = r0 + j;
+            nbuf += tot;
+        }
+        if (__all(q >= nq)) break;
+    }
+    if (nbuf) {
+        long long bpos = 0;
+        if (lane == 0)
+            bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                        (unsigned long long)nbuf);
+        bpos = __shfl(bpos, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE)
+            cand[bpos + j] = buf[wid][j];
+    }
+}
+
+__global__ void k_v3(const int32_t *__restrict__ d, int64_t n, int32_t c,
+                     int64_t *cand, int64_t *ncand)
+{
+    int64_t nq = n / 4;
+    const int4 *d4 = (const int4 *)d;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int lane = (int)(threadIdx.x % WAVE);
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
+         q += stride) {
+        int4 v = d4[q];
+        bool m[4] = {v.x < c, v.y < c, v.z < c, v.w < c};
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+            unsigned long long mask = __ballot(m[j]);
+            if (!mask) continue;
+            int leader = __ffsll((long long)mask) - 1;
+            int rank = __popcll(mask & ((1ull << lane) - 1ull));
+            long long base = 0;
+            if (lane == leader)
+                base = (long long)atomicAdd((unsigned long long *)ncand,
+                                            (unsigned long long)__popcll(mask));
+            base = __shfl(base, leader, WAVE);
+            if (m[j]) cand[base + rank] = q * 4 + j;
+        }
+    }
+}
+
+extern "C" int run(int64_t n, int reps)
+{
+    int32_t *d;
+    int64_t *cand, *ncand;
+    unsigned long long *out;
+    hipMalloc(&d, n * 4);
+    hipMalloc(&cand, n * 8);
+    hipMalloc(&ncand, 8);
+    hipMalloc(&out, 8);
+    hipMemset(d, 0x11, n * 4); /* ~49% pass vs cutoff below */
+    int32_t cutoff = 0x11111112;
+    /* fill with pseudo-random via simple kernel-free trick: memset pattern is
+     * constant; instead use alternating via a second memset on half */
+    hipMemset((char *)d, 0x22, n * 2); /* first half fails, second passes */
+    dim3 grid(2048), blk(256);
+    hipDeviceSynchronize();
+    for (int v = 0; v < 4; v++) {
+        hipEvent_t e0, e1;
+        hipEventCreate(&e0); hipEventCreate(&e1);
+        float best = 1e9f;
+        for (int r = 0; r < reps; r++) {
+            hipMemset(ncand, 0, 8);
+            hipEventRecord(e0);
+            if (v == 0) hipLaunchKernelGGL(k_v0, grid, blk, 0, 0, d, n, cutoff, out);
+            if (v == 1) hipLaunchKernelGGL(k_v1, grid, blk, 0, 0, d, n, cutoff, out);
+            if (v == 2) hipLaunchKernelGGL(k_v2, grid, blk, 0, 0, d, n, cutoff, cand, ncand);
+            if (v == 3) hipLaunchKernelGGL(k_v3, grid, blk, 0, 0, d, n, cutoff, cand, ncand);
+            hipEventRecord(e1);
+            hipEventSynchronize(e1);
+            float ms; hipEventElapsedTime(&ms, e0, e1);
+            if (ms < best) best = ms;
+        }
+        printf("variant %d: %.3f ms  (%.0f GB/s read)\n", v, best,
+               n * 4.0 / best / 1e6);
+    }
+    hipDeviceSynchronize();
+    return 0;
+}
+
+int main() { return run(150000000, 6); }
