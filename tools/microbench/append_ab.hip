@@ -131,6 +131,120 @@ __global__ void k_v3(const int32_t *__restrict__ d, int64_t n, int32_t c,
     }
 }
 
+
+/* v4: BUF=2048 (64KB LDS/block, halves flush atomics, 2 blocks/CU) */
+__global__ void k_v4(const int32_t *__restrict__ d, int64_t n, int32_t c,
+                     int64_t *cand, int64_t *ncand)
+{
+    const int BUF = 2048;
+    __shared__ int64_t buf[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0;
+    int64_t nq = n / 4;
+    const int4 *d4 = (const int4 *)d;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         q += stride) {
+        bool m[4] = {false, false, false, false};
+        int64_t r0 = q * 4;
+        if (q < nq) {
+            int4 v = d4[q];
+            m[0] = v.x < c; m[1] = v.y < c; m[2] = v.z < c; m[3] = v.w < c;
+        }
+        int mycnt = m[0] + m[1] + m[2] + m[3];
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        int tot = __shfl(incl, WAVE - 1, WAVE);
+        if (tot) {
+            if (nbuf + tot > BUF) {
+                long long bpos = 0;
+                if (lane == 0)
+                    bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                                (unsigned long long)nbuf);
+                bpos = __shfl(bpos, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE)
+                    cand[bpos + j] = buf[wid][j];
+                nbuf = 0;
+            }
+            int pos = nbuf + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j]) buf[wid][pos++] = r0 + j;
+            nbuf += tot;
+        }
+        if (__all(q >= nq)) break;
+    }
+    if (nbuf) {
+        long long bpos = 0;
+        if (lane == 0)
+            bpos = (long long)atomicAdd((unsigned long long *)ncand,
+                                        (unsigned long long)nbuf);
+        bpos = __shfl(bpos, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE)
+            cand[bpos + j] = buf[wid][j];
+    }
+}
+
+/* v5: register-mask two-phase, one atomic per BLOCK. Fixed 1024 rows/thread
+ * window per block (16 u64 mask words); phase 1 computes masks + counts,
+ * block reduce + one atomicAdd, phase 2 replays from registers. */
+__global__ void k_v5(const int32_t *__restrict__ d, int64_t n, int32_t c,
+                     int64_t *cand, int64_t *ncand)
+{
+    const int MW = 16; /* 1024 rows per thread */
+    unsigned long long mask[MW];
+    int64_t win = (int64_t)blockDim.x * 1024;
+    const int4 *d4 = (const int4 *)d;
+    __shared__ int64_t tcnt[256];
+    __shared__ int64_t tbase[257];
+    for (int64_t w0 = (int64_t)blockIdx.x * win; w0 < n;
+         w0 += (int64_t)gridDim.x * win) {
+        int my = 0;
+#pragma unroll
+        for (int mwi = 0; mwi < MW; mwi++) mask[mwi] = 0;
+        for (int it = 0; it < 256; it++) {
+            int64_t q = (w0 >> 2) + (int64_t)it * blockDim.x + threadIdx.x;
+            if (q * 4 >= n) break;
+            int4 v = d4[q];
+            unsigned long long mm = (unsigned long long)(v.x < c) |
+                                    ((unsigned long long)(v.y < c) << 1) |
+                                    ((unsigned long long)(v.z < c) << 2) |
+                                    ((unsigned long long)(v.w < c) << 3);
+            mask[it >> 4] |= mm << ((it & 15) * 4);
+            my += __popcll(mm);
+        }
+        tcnt[threadIdx.x] = my;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int64_t tot = 0;
+            for (int t = 0; t < (int)blockDim.x; t++) {
+                tbase[t] = tot;
+                tot += tcnt[t];
+            }
+            tbase[256] = tot ? (int64_t)atomicAdd((unsigned long long *)ncand,
+                                                  (unsigned long long)tot)
+                             : 0;
+        }
+        __syncthreads();
+        int64_t pos = tbase[256] + tbase[threadIdx.x];
+#pragma unroll
+        for (int mwi = 0; mwi < MW; mwi++) {
+            unsigned long long mm = mask[mwi];
+            while (mm) {
+                int b = __ffsll((long long)mm) - 1;
+                mm &= mm - 1;
+                int it = mwi * 16 + (b >> 2);
+                int64_t q = (w0 >> 2) + (int64_t)it * blockDim.x + threadIdx.x;
+                cand[pos++] = q * 4 + (b & 3);
+            }
+        }
+        __syncthreads();
+    }
+}
+
 extern "C" int run(int64_t n, int reps)
 {
     int32_t *d;
@@ -147,7 +261,7 @@ extern "C" int run(int64_t n, int reps)
     hipMemset((char *)d, 0x22, n * 2); /* first half fails, second passes */
     dim3 grid(2048), blk(256);
     hipDeviceSynchronize();
-    for (int v = 0; v < 4; v++) {
+    for (int v = 0; v < 6; v++) {
         hipEvent_t e0, e1;
         hipEventCreate(&e0); hipEventCreate(&e1);
         float best = 1e9f;
@@ -158,6 +272,8 @@ extern "C" int run(int64_t n, int reps)
             if (v == 1) hipLaunchKernelGGL(k_v1, grid, blk, 0, 0, d, n, cutoff, out);
             if (v == 2) hipLaunchKernelGGL(k_v2, grid, blk, 0, 0, d, n, cutoff, cand, ncand);
             if (v == 3) hipLaunchKernelGGL(k_v3, grid, blk, 0, 0, d, n, cutoff, cand, ncand);
+            if (v == 4) hipLaunchKernelGGL(k_v4, grid, blk, 0, 0, d, n, cutoff, cand, ncand);
+            if (v == 5) hipLaunchKernelGGL(k_v5, dim3((n + 256*1024 - 1) / (256*1024)), blk, 0, 0, d, n, cutoff, cand, ncand);
             hipEventRecord(e1);
             hipEventSynchronize(e1);
             float ms; hipEventElapsedTime(&ms, e0, e1);
