@@ -955,7 +955,7 @@ __global__ void k_ord_filter_date(const otbx_orders_dev o, int32_t q3date,
                                   int64_t *__restrict__ out, int64_t *ncand)
 {
     /* 4 rows per lane via one int4 load; per-wave LDS-staged append */
-    const int BUF = 2048; /* 64 KB LDS/block: halves the flush reservations on the single output counter (the measured bottleneck - tools/microbench) */
+    const int BUF = 1024;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
     int nbuf = 0;
@@ -1026,7 +1026,7 @@ __global__ void k_ord_probe_cust(const otbx_orders_dev o,
                                  int64_t *nout, unsigned long long *minkey,
                                  unsigned long long *maxkey)
 {
-    const int BUF = 2048; /* 64 KB LDS/block: halves the flush reservations on the single output counter (the measured bottleneck - tools/microbench) */
+    const int BUF = 1024;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
     int nbuf = 0;
@@ -1155,7 +1155,7 @@ __global__ void k_q3_scan_filter_direct(const otbx_lineitem_dev l,
      * un-hashed bit so a wave's 1024 rows touch ~4 bitmap words (L1).
      * Candidate append is ORDER-PRESERVING (wave prefix sum over per-lane
      * counts) with per-wave LDS staging, one global atomic per 1024. */
-    const int BUF = 2048; /* 64 KB LDS/block: halves the flush reservations on the single output counter (the measured bottleneck - tools/microbench) */
+    const int BUF = 1024;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
     int nbuf = 0; /* wave-uniform */
@@ -1332,7 +1332,7 @@ __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
      * the ballot converges the wave); date-failing lanes read bloom word 0
      * (stays hot in L1) so all 8 loads issue unconditionally. Per-wave LDS
      * staging, one global atomic per 1024 candidates. */
-    const int BUF = 2048; /* 64 KB LDS/block: halves the flush reservations on the single output counter (the measured bottleneck - tools/microbench) */
+    const int BUF = 1024;
     const int B = 8;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
