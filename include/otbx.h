@@ -202,6 +202,29 @@ otbx_status otbx_agg_i64(const int64_t *keys_dev, const uint8_t *key_null_dev,
                          otbx_agg_group *groups_dev, int64_t *ngroups_dev,
                          void *stream);
 
+/* ---- repartition exchange ----
+ * The GPU half of the reference's "Distribute results by H: col" exchange
+ * (make_remotesubplan, optimizer/plan/createplan.c:8671; locator semantics
+ * shardid → node, pgxc/shard/shardmap.c:2231 restated as key % nranks for
+ * the dense-key locator of DESIGN.md §2): groups rows by owning rank into a
+ * permutation with contiguous per-rank segments; the host layer then
+ * all-to-alls the gathered segments over RCCL (fragment.py). nranks ≤ 64.
+ * counts_host: int64[nranks], written synchronously (the call syncs). */
+otbx_status otbx_partition_by_key(const int64_t *keys_dev, int64_t n,
+                                  uint32_t nranks, int64_t *perm_dev,
+                                  int64_t *counts_host, void *stream);
+
+/* permutation gathers (dst[i] = src[perm[i]]) for applying the partition to
+ * payload columns without leaving the native path */
+otbx_status otbx_gather_i64(const int64_t *src, const int64_t *perm, int64_t n,
+                            int64_t *dst, void *stream);
+otbx_status otbx_gather_f64(const double *src, const int64_t *perm, int64_t n,
+                            double *dst, void *stream);
+otbx_status otbx_gather_i32(const int32_t *src, const int64_t *perm, int64_t n,
+                            int32_t *dst, void *stream);
+otbx_status otbx_gather_u8(const uint8_t *src, const int64_t *perm, int64_t n,
+                           uint8_t *dst, void *stream);
+
 /* Inner hash join on i64 keys: emits (build_idx, probe_idx) pairs in
  * arbitrary order (result-set parity; SQL imposes no order). pairs capacity
  * cap_pairs; overflow → OTBX_ERR_INVALID reported via npairs_dev = -1. */

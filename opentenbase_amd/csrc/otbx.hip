@@ -759,6 +759,155 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
 
 } /* extern "C" */
 
+/* ================= repartition exchange ================= */
+
+/* owner = key % nranks (dense-key locator restatement of shardid→node,
+ * shardmap.c:2231/1147) */
+#define PART_MAX_RANKS 64
+
+__global__ void k_part_count(const int64_t *__restrict__ keys, int64_t n,
+                             uint32_t nranks, unsigned long long *counts)
+{
+    __shared__ unsigned int bc[PART_MAX_RANKS];
+    for (int r = threadIdx.x; r < (int)nranks; r += blockDim.x) bc[r] = 0;
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        atomicAdd(&bc[(uint32_t)(((uint64_t)keys[i]) % nranks)], 1u);
+    __syncthreads();
+    for (int r = threadIdx.x; r < (int)nranks; r += blockDim.x)
+        if (bc[r])
+            atomicAdd(&counts[r], (unsigned long long)bc[r]);
+}
+
+/* scatter: contiguous chunk per block; pass 1 counts the block's rows per
+ * rank, one global cursor reservation per (block, rank), pass 2 places rows
+ * at LDS-allocated in-block offsets (re-reads the chunk from L2). */
+__global__ void k_part_scatter(const int64_t *__restrict__ keys, int64_t n,
+                               uint32_t nranks,
+                               unsigned long long *__restrict__ cursor,
+                               int64_t *__restrict__ perm)
+{
+    __shared__ unsigned int lcur[PART_MAX_RANKS];
+    __shared__ long long base[PART_MAX_RANKS];
+    int64_t per_block = (n + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < n ? lo + per_block : n;
+    for (int r = threadIdx.x; r < (int)nranks; r += blockDim.x) lcur[r] = 0;
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        atomicAdd(&lcur[(uint32_t)(((uint64_t)keys[i]) % nranks)], 1u);
+    __syncthreads();
+    for (int r = threadIdx.x; r < (int)nranks; r += blockDim.x) {
+        base[r] = lcur[r]
+                      ? (long long)atomicAdd(&cursor[r],
+                                             (unsigned long long)lcur[r])
+                      : 0;
+        lcur[r] = 0;
+    }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint32_t owner = (uint32_t)(((uint64_t)keys[i]) % nranks);
+        unsigned int off = atomicAdd(&lcur[owner], 1u);
+        perm[base[owner] + off] = i;
+    }
+}
+
+template <typename T>
+__global__ void k_gather(const T *__restrict__ src,
+                         const int64_t *__restrict__ perm, int64_t n,
+                         T *__restrict__ dst)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        dst[i] = src[perm[i]];
+}
+
+extern "C" {
+
+otbx_status otbx_partition_by_key(const int64_t *keys, int64_t n,
+                                  uint32_t nranks, int64_t *perm,
+                                  int64_t *counts_host, void *stream)
+{
+    if (nranks == 0 || nranks > PART_MAX_RANKS || !counts_host)
+        return OTBX_ERR_INVALID;
+    hipStream_t s = (hipStream_t)stream;
+    static unsigned long long *scratch = nullptr; /* counts + cursors */
+    if (!scratch)
+        HIP_CHECK(hipMalloc(&scratch, PART_MAX_RANKS * 2 * 8));
+    unsigned long long *counts = scratch, *cursor = scratch + PART_MAX_RANKS;
+    HIP_CHECK(hipMemsetAsync(counts, 0, PART_MAX_RANKS * 8, s));
+    if (n > 0)
+        hipLaunchKernelGGL(k_part_count, dim3(grid_for(n, 256)), dim3(256), 0,
+                           s, keys, n, nranks, counts);
+    static int64_t *h_counts = nullptr;
+    if (!h_counts)
+        HIP_CHECK(hipHostMalloc(&h_counts, PART_MAX_RANKS * 8));
+    HIP_CHECK(hipMemcpyAsync(h_counts, counts, nranks * 8,
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    int64_t off = 0;
+    static int64_t *h_cursor = nullptr;
+    if (!h_cursor)
+        HIP_CHECK(hipHostMalloc(&h_cursor, PART_MAX_RANKS * 8));
+    for (uint32_t r = 0; r < nranks; r++) {
+        counts_host[r] = h_counts[r];
+        h_cursor[r] = off;
+        off += h_counts[r];
+    }
+    HIP_CHECK(hipMemcpyAsync(cursor, h_cursor, nranks * 8,
+                             hipMemcpyHostToDevice, s));
+    if (n > 0)
+        hipLaunchKernelGGL(k_part_scatter, dim3(grid_for(n, 256)), dim3(256),
+                           0, s, keys, n, nranks, cursor, perm);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+otbx_status otbx_gather_i64(const int64_t *src, const int64_t *perm, int64_t n,
+                            int64_t *dst, void *stream)
+{
+    if (n > 0)
+        hipLaunchKernelGGL(k_gather<int64_t>, dim3(grid_for(n, 256)), dim3(256),
+                           0, (hipStream_t)stream, src, perm, n, dst);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+otbx_status otbx_gather_f64(const double *src, const int64_t *perm, int64_t n,
+                            double *dst, void *stream)
+{
+    if (n > 0)
+        hipLaunchKernelGGL(k_gather<double>, dim3(grid_for(n, 256)), dim3(256),
+                           0, (hipStream_t)stream, src, perm, n, dst);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+otbx_status otbx_gather_i32(const int32_t *src, const int64_t *perm, int64_t n,
+                            int32_t *dst, void *stream)
+{
+    if (n > 0)
+        hipLaunchKernelGGL(k_gather<int32_t>, dim3(grid_for(n, 256)), dim3(256),
+                           0, (hipStream_t)stream, src, perm, n, dst);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+otbx_status otbx_gather_u8(const uint8_t *src, const int64_t *perm, int64_t n,
+                           uint8_t *dst, void *stream)
+{
+    if (n > 0)
+        hipLaunchKernelGGL(k_gather<uint8_t>, dim3(grid_for(n, 256)), dim3(256),
+                           0, (hipStream_t)stream, src, perm, n, dst);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+} /* extern "C" */
+
 /* ================= generic inner hash join (otbx_join_i64) ================= */
 
 /* build slot: {idx (claim word, -1 empty), key}; duplicates occupy their own

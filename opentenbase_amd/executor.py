@@ -336,6 +336,36 @@ def q3_topk(groups, k=10):
     return groups[order][:k]
 
 
+def partition_by_key(keys):
+    """Repartition exchange, GPU half (the 'Distribute results
+    by H: col' exchange of make_remotesubplan, createplan.c:8671): returns
+    (perm, counts) where perm groups row indices into contiguous per-rank
+    segments (owner = key % world) and counts[r] is segment r's length.
+    fragment.exchange_rows() moves the gathered segments over RCCL."""
+    import torch.distributed as dist
+    world = dist.get_world_size() if dist.is_available() and \
+        dist.is_initialized() else 1
+    n = len(keys)
+    perm = torch.empty(n, dtype=torch.int64, device="cuda")
+    counts = (C.c_int64 * max(world, 1))()
+    call("otbx_partition_by_key", C.c_void_p(keys.data_ptr()), C.c_int64(n),
+         C.c_uint32(world), C.c_void_p(perm.data_ptr()), counts, _stream())
+    return perm, list(counts)
+
+
+_GATHER_FN = {torch.int64: "otbx_gather_i64", torch.float64: "otbx_gather_f64",
+              torch.int32: "otbx_gather_i32", torch.uint8: "otbx_gather_u8"}
+
+
+def gather(src, perm):
+    """dst[i] = src[perm[i]] via the native gather kernels."""
+    dst = torch.empty(len(perm), dtype=src.dtype, device="cuda")
+    call(_GATHER_FN[src.dtype], C.c_void_p(src.data_ptr()),
+         C.c_void_p(perm.data_ptr()), C.c_int64(len(perm)),
+         C.c_void_p(dst.data_ptr()), _stream())
+    return dst
+
+
 class GpuHashAgg(CustomScanState):
     """Composable HashAggregate over i64 keys / f64 values with full NULL
     semantics (execGrouping.c:295 + nodeAgg.c:743). Inputs: device tensors;

@@ -94,3 +94,54 @@ def merge_q3_topk(candidates, k=10):
 
 def finalize_q1(rows):
     return q1_finalize(rows)
+
+
+def all_to_all_variable(t, send_counts):
+    """All-to-all with per-rank variable segment lengths: rank r sends
+    t[offsets[r]:offsets[r]+send_counts[r]] to rank r and receives the
+    concatenation of every rank's segment addressed to it. The RCCL
+    all-to-allv of the repartition exchange (FN-transport
+    semantics of forward/*). nccl: dist.all_to_all_single with splits;
+    gloo (CPU tests): emulated via all-gather + slicing (gloo has no
+    all-to-all)."""
+    if not is_dist() or dist.get_world_size() == 1:
+        return t
+    world = dist.get_world_size()
+    assert len(send_counts) == world
+    if dist.get_backend() == "nccl":
+        sc = torch.tensor(send_counts, dtype=torch.int64, device=t.device)
+        rc = torch.empty_like(sc)
+        dist.all_to_all_single(rc, sc)
+        recv_counts = [int(x) for x in rc.cpu()]
+        out = torch.empty(sum(recv_counts), dtype=t.dtype, device=t.device)
+        dist.all_to_all_single(out, t, output_split_sizes=recv_counts,
+                               input_split_sizes=list(send_counts))
+        return out
+    # gloo emulation: gather everything + counts, slice my segments
+    if t.is_cuda:
+        t = t.cpu()
+    rank = dist.get_rank()
+    cnts = torch.tensor(send_counts, dtype=torch.int64)
+    all_cnts = [torch.empty_like(cnts) for _ in range(world)]
+    dist.all_gather(all_cnts, cnts)
+    alldata = allgather_variable(t)
+    out = []
+    pos = 0
+    for src in range(world):
+        c = [int(x) for x in all_cnts[src]]
+        seg_start = pos + sum(c[:rank])
+        out.append(alldata[seg_start: seg_start + c[rank]])
+        pos += sum(c)
+    return torch.cat(out) if out else t[:0]
+
+
+def exchange_rows(keys, payload_tensors):
+    """Full repartition exchange: GPU partition by key % world, native
+    gathers, all-to-allv of every column. Returns (keys', payloads') — the
+    rows this rank owns after redistribution."""
+    from . import executor as ex
+    perm, counts = ex.partition_by_key(keys)
+    out_keys = all_to_all_variable(ex.gather(keys, perm), counts)
+    out_payloads = [all_to_all_variable(ex.gather(p, perm), counts)
+                    for p in payload_tensors]
+    return out_keys, out_payloads

@@ -91,3 +91,22 @@ def test_allgather_variable():
 def test_merge_q3_topk():
     top = _run(_worker_topk)
     assert top["revenue"].tolist() == [6.0, 5.0, 5.0, 3.0]
+
+
+def _worker_a2a(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    # rank r sends [r*100 + 0..2] to rank 0 and [r*100 + 10..12] to rank 1
+    t = torch.tensor([rank * 100 + i for i in (0, 1, 2)] +
+                     [rank * 100 + i for i in (10, 11, 12)], dtype=torch.int64)
+    out = fragment.all_to_all_variable(t, [3, 3])
+    if rank == 0:
+        q.put(out.numpy())
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_all_to_all_variable_gloo():
+    out = _run(_worker_a2a)
+    # rank 0 receives rank 0's first segment then rank 1's first segment
+    assert out.tolist() == [0, 1, 2, 100, 101, 102]

@@ -362,3 +362,39 @@ def test_q3_hash_fallback_path(ex, ora):
     og = q3_oracle(ora, n)
     exp = dict(zip(og["l_orderkey"].tolist(), og["revenue"].tolist()))
     assert direct_groups.keys() == exp.keys()
+
+
+# ---------------- repartition exchange ----------------
+
+def test_partition_by_key_and_gather(ex):
+    """GPU partition groups rows into contiguous per-rank segments
+    (owner = key % nranks) with exact counts; native gather applies the
+    permutation. Checked against numpy."""
+    import ctypes as Ct
+    from opentenbase_amd._lib import call
+    rng = np.random.default_rng(3)
+    n = 1_000_000
+    keys_h = rng.integers(1, 10_000_000, n)
+    keys = torch.as_tensor(keys_h, dtype=torch.int64, device="cuda")
+    vals = torch.as_tensor(keys_h * 2 + 1, dtype=torch.int64, device="cuda")
+    nranks = 4
+    perm = torch.empty(n, dtype=torch.int64, device="cuda")
+    counts = (Ct.c_int64 * nranks)()
+    call("otbx_partition_by_key", Ct.c_void_p(keys.data_ptr()), Ct.c_int64(n),
+         Ct.c_uint32(nranks), Ct.c_void_p(perm.data_ptr()), counts,
+         Ct.c_void_p(torch.cuda.current_stream().cuda_stream))
+    counts = list(counts)
+    exp_counts = [int((keys_h % nranks == r).sum()) for r in range(nranks)]
+    assert counts == exp_counts
+    pk = ex.gather(keys, perm).cpu().numpy()
+    pv = ex.gather(vals, perm).cpu().numpy()
+    off = 0
+    for r in range(nranks):
+        seg = pk[off:off + counts[r]]
+        assert (seg % nranks == r).all()
+        assert sorted(seg.tolist()) == sorted(
+            keys_h[keys_h % nranks == r].tolist())
+        assert (pv[off:off + counts[r]] == seg * 2 + 1).all()
+        off += counts[r]
+    # the permutation is a bijection over [0, n)
+    assert np.array_equal(np.sort(perm.cpu().numpy()), np.arange(n))
