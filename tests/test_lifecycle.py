@@ -26,13 +26,21 @@ def drain(node):
 
 
 def test_rescan_identical(ex):
+    """ReScan re-runs the fragment: group keys/counts bit-identical; float
+    sums within the 1e-6 contract (atomic f64 accumulation order varies
+    between runs — BASELINE.md parity gate)."""
     li = ex.GpuLineitem.generate(200000, with_orderkey=False)
     node = ex.GpuQ1PartialAgg(li)
     node.BeginCustomScan()
     first = drain(node)
     node.ReScanCustomScan()          # ReScanCustomScan (extensible.h:131)
     second = drain(node)
-    assert first == second
+    assert len(first) == len(second)
+    for a, b in zip(first, second):
+        assert a["l_returnflag"] == b["l_returnflag"]
+        assert a["count_order"] == b["count_order"]
+        for f in ("sum_qty", "sum_base_price", "sum_disc_price", "sum_charge"):
+            assert abs(a[f] - b[f]) <= 1e-6 * abs(a[f])
     node.EndCustomScan()
 
 
