@@ -1277,7 +1277,8 @@ __global__ void k_ord_insert(const otbx_orders_dev o,
 __global__ void k_ord_insert_direct(const otbx_orders_dev o,
                                     const int64_t *__restrict__ cand,
                                     const int64_t *__restrict__ ncand_p,
-                                    int64_t mino, unsigned long long *bitmap,
+                                    int64_t mino, int64_t range,
+                                    unsigned long long *bitmap,
                                     unsigned long long *ptab)
 {
     int64_t n = *ncand_p;
@@ -1286,6 +1287,8 @@ __global__ void k_ord_insert_direct(const otbx_orders_dev o,
          ci += stride) {
         int64_t i = cand[ci];
         int64_t idx = o.o_orderkey[i] - mino;
+        if (idx < 0 || idx >= range)
+            continue; /* outside this grace pass's key sub-range */
         atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
         ptab[idx] = (unsigned long long)(uint32_t)o.o_orderdate[i] |
                     ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
@@ -1839,6 +1842,11 @@ static int64_t direct_cap_for(int64_t norders)
 {
     int64_t cap = next_pow2_host(norders < 16 ? 16 : norders) * 32;
     if (cap > OTBX_DIRECT_CAP_DEFAULT) cap = OTBX_DIRECT_CAP_DEFAULT;
+    const char *env = getenv("OTBX_DIRECT_CAP"); /* test hook */
+    if (env) {
+        int64_t e = atoll(env);
+        if (e >= (1 << 14) && e < cap) cap = e;
+    }
     return cap;
 }
 
@@ -1991,27 +1999,56 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     int64_t nof = h_cnt[1];
     int64_t mino = h_cnt[4], maxo = h_cnt[5];
     int64_t range = maxo - mino + 1;
-    bool use_direct = nof > 0 && range > 0 && range <= dcap && !force_hash;
+    /* grace-style multi-pass (SURVEY §8f.4 analog): when the build side's
+     * key range exceeds the direct tables' capacity, split it into ≤64
+     * sub-ranges and run the direct pipeline once per sub-range — bounded
+     * memory per pass, P passes over the probe stream (the reference's
+     * batch spill tradeoff, nodeHash.c:1086). */
+    int64_t npasses = 1;
+    if (nof > 0 && range > dcap) {
+        npasses = (range + dcap - 1) / dcap;
+        if (npasses > 64) npasses = 1; /* too sparse: hash path */
+    }
+    bool use_direct = nof > 0 && range > 0 &&
+                      (range <= dcap || npasses > 1) && !force_hash;
     if (use_direct) {
         /* dense-orderkey direct path: bitmap filter + direct-addressed
-         * payload/revenue tables preserve the probe stream's key locality */
-        HIP_CHECK(hipMemsetAsync(dbitmap, 0, (size_t)(range / 64 + 8) * 8, s));
-        HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)range * 8, s));
-        HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)range * 8, s));
-        hipLaunchKernelGGL(k_ord_insert_direct, dim3(grid_for(o->n, 256)),
-                           dim3(256), 0, s, *o, cand_o2, &hdr[1], mino,
-                           dbitmap, dptab);
-        if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
-        hipLaunchKernelGGL(k_q3_scan_filter_direct, dim3(grid_for(l->n, 256)),
-                           dim3(256), 0, s, *l, q3date, mino, range, dbitmap,
-                           cand_li, &hdr[2]);
-        hipLaunchKernelGGL(k_q3_probe_agg_direct, dim3(grid_for(l->n, 256)),
-                           dim3(256), 0, s, *l, cand_li, &hdr[2], mino, dptab,
-                           drtab, nhits);
-        if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
-        hipLaunchKernelGGL(k_q3_compact_direct, dim3(grid_for(range, 256)),
-                           dim3(256), 0, s, drtab, dptab, range, mino,
-                           groups_dev, cap_groups, ngroups_dev);
+         * payload/revenue tables preserve the probe stream's key locality;
+         * npasses > 1 = the grace multi-pass over key sub-ranges */
+        bool rec2 = false, rec3 = false;
+        for (int64_t pass = 0; pass < npasses; pass++) {
+            int64_t pmin = mino + pass * dcap;
+            int64_t prange = range - pass * dcap < dcap ? range - pass * dcap
+                                                        : dcap;
+            HIP_CHECK(hipMemsetAsync(dbitmap, 0,
+                                     (size_t)(prange / 64 + 8) * 8, s));
+            HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)prange * 8, s));
+            HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)prange * 8, s));
+            if (pass > 0)
+                HIP_CHECK(hipMemsetAsync(&hdr[2], 0, 8, s)); /* reset cands */
+            hipLaunchKernelGGL(k_ord_insert_direct, dim3(grid_for(o->n, 256)),
+                               dim3(256), 0, s, *o, cand_o2, &hdr[1], pmin,
+                               prange, dbitmap, dptab);
+            if (kernel_ms && !rec2) {
+                HIP_CHECK(hipEventRecord(ev[2], s));
+                rec2 = true;
+            }
+            hipLaunchKernelGGL(k_q3_scan_filter_direct,
+                               dim3(grid_for(l->n, 256)), dim3(256), 0, s, *l,
+                               q3date, pmin, prange, dbitmap, cand_li,
+                               &hdr[2]);
+            hipLaunchKernelGGL(k_q3_probe_agg_direct,
+                               dim3(grid_for(l->n, 256)), dim3(256), 0, s, *l,
+                               cand_li, &hdr[2], pmin, dptab, drtab, nhits);
+            if (kernel_ms && !rec3 && pass == npasses - 1) {
+                HIP_CHECK(hipEventRecord(ev[3], s));
+                rec3 = true;
+            }
+            hipLaunchKernelGGL(k_q3_compact_direct,
+                               dim3(grid_for(prange, 256)), dim3(256), 0, s,
+                               drtab, dptab, prange, pmin, groups_dev,
+                               cap_groups, ngroups_dev);
+        }
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
     } else {
     int64_t ocap = fit_cap(nof), gcap = ocap;

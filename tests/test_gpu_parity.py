@@ -398,3 +398,28 @@ def test_partition_by_key_and_gather(ex):
         off += counts[r]
     # the permutation is a bijection over [0, n)
     assert np.array_equal(np.sort(perm.cpu().numpy()), np.arange(n))
+
+
+def test_q3_grace_multipass_parity(ex, ora):
+    """Grace-style multi-pass (SURVEY §8f.4 analog): shrink the direct-table
+    capacity so the key range needs ~7 passes; results must equal the
+    single-pass run and the oracle."""
+    import os
+    n = 400000
+    li = ex.GpuLineitem.generate(n)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    cu = ex.GpuCustomer.generate(n // 40)
+    os.environ["OTBX_DIRECT_CAP"] = "16384"   # norders=100k → ~7 passes
+    try:
+        node = ex.GpuQ3Fragment(cu, od, li)
+        drain(node)
+        got = {int(k): float(v) for k, v in
+               zip(node.fetch_groups()["l_orderkey"],
+                   node.fetch_groups()["revenue"])}
+    finally:
+        del os.environ["OTBX_DIRECT_CAP"]
+    og = q3_oracle(ora, n)
+    exp = dict(zip(og["l_orderkey"].tolist(), og["revenue"].tolist()))
+    assert got.keys() == exp.keys()
+    for k in exp:
+        assert approx(got[k], exp[k]), k
