@@ -58,6 +58,13 @@ def main():
         tenk_fivethous=tenk["fivethous"],
     )
 
+    # aggtest (4 rows): src/test/regress/data/agg.data — a int2, b float4
+    agg_rows = []
+    with open(os.path.join(REF, "data", "agg.data")) as f:
+        for line in f:
+            a, b = line.split("\t")
+            agg_rows.append([int(a), float(b)])
+
     # test_having rows transcribed from sql/select_having.sql INSERTs
     test_having = [
         (0, 1, "XXXX", "A"), (1, 2, "AAAA", "b"), (2, 2, "AAAA", "c"),
@@ -67,6 +74,15 @@ def main():
     ]
 
     meta = {
+        "aggtest": {
+            "rows": agg_rows,
+            # aggregates.out:12 avg(a) where a<100 = 32.666...; :38 sum(a)=198
+            # :44 sum(b)=431.773 (float4 display); :24 avg(b)≈107.943
+            "avg_a_lt100": 32.666666666666666,
+            "sum_a": 198,
+            "sum_b_3dp": 431.773,
+            "avg_b_3dp": 107.943,
+        },
         "onek": {
             "nrows": int(len(onek["four"])),
             "avg_four": 1.5,          # aggregates.out:6
