@@ -1468,68 +1468,67 @@ struct q3g_slot {
     double revenue;
 };
 
-/* probe phase A: stream l_shipdate + l_orderkey, qual + bloom test, emit a
- * COMPACTED candidate row-id list. Without compaction the probe loop is
- * exec-mask divergent (~5% of lanes active: a 64-lane wave walks the table
- * for ~3 lanes of work); the dense list gives phase B full lanes and full
- * memory-level parallelism. Block-aggregated two-phase append, one global
- * atomic per block, contiguous chunk per block. */
 __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
                                  const unsigned long long *__restrict__ bloom,
                                  int64_t bloom_words,
                                  int64_t *__restrict__ cand, int64_t *ncand)
 {
-    /* single pass, 8 rows per lane per batch: the 8 bloom loads pipeline
-     * (a row-at-a-time loop waits a full random-load latency per row because
-     * the ballot converges the wave); date-failing lanes read bloom word 0
-     * (stays hot in L1) so all 8 loads issue unconditionally. Per-wave LDS
-     * staging, one global atomic per 1024 candidates. */
+    /* hash-fallback scan (sparse/wide key ranges): same 16-B-vector,
+     * order-preserving structure as k_q3_scan_filter_direct, with the
+     * hashed two-bit bloom test (nodeHashjoin.c:1862 analog) instead of the
+     * dense bitmap; date-failing lanes read bloom word 0 (hot in L1). */
     const int BUF = 1024;
-    const int B = 8;
     __shared__ int64_t buf[256 / WAVE][BUF];
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
     int nbuf = 0; /* wave-uniform */
-    int64_t chunk = (int64_t)blockDim.x * B;
-    int64_t stride = (int64_t)gridDim.x * chunk;
-    for (int64_t base = (int64_t)blockIdx.x * chunk; base < l.n;
-         base += stride) {
-        bool m[B];
-        int64_t rows[B];
+    int64_t nq = l.n / 4;
+    const int4 *sd4 = (const int4 *)l.l_shipdate;
+    const longlong2 *ok2 = (const longlong2 *)l.l_orderkey;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         q += stride) {
+        bool m[4] = {false, false, false, false};
+        int64_t r0 = q * 4;
+        int mycnt = 0;
+        if (q < nq) {
+            int4 d = sd4[q];
+            longlong2 ka = ok2[2 * q], kb = ok2[2 * q + 1];
+            int32_t ds[4] = {d.x, d.y, d.z, d.w};
+            int64_t ky[4] = {ka.x, ka.y, kb.x, kb.y};
 #pragma unroll
-        for (int k = 0; k < B; k++) {
-            int64_t i = base + (int64_t)k * blockDim.x + threadIdx.x;
-            rows[k] = i;
-            bool datepass = false;
-            int64_t key = 0;
-            if (i < l.n) {
-                datepass = l.l_shipdate[i] > q3date;
-                key = l.l_orderkey[i];
+            for (int j = 0; j < 4; j++) {
+                bool pass = ds[j] > q3date;
+                uint64_t h = otbx_splitmix64((uint64_t)ky[j]);
+                uint64_t w =
+                    pass ? ((h >> 12) & (uint64_t)(bloom_words - 1)) : 0;
+                unsigned long long bits =
+                    (1ull << (h & 63)) | (1ull << ((h >> 6) & 63));
+                unsigned long long v = bloom[w];
+                m[j] = pass && ((v & bits) == bits);
+                mycnt += m[j];
             }
-            /* consecutive rows sit in consecutive lanes: only run-leading
-             * lanes load a bloom word, dupes take the leader's result via a
-             * nearest-leader max-scan (6 shuffles) — lineitem is clustered
-             * ~4 rows/orderkey so this ~halves the random bloom lines */
-            int64_t kprev = __shfl_up((long long)key, 1, WAVE);
-            bool isleader = (lane == 0) || key != kprev;
-            uint64_t h = otbx_splitmix64((uint64_t)key);
-            uint64_t w = isleader ? ((h >> 12) & (uint64_t)(bloom_words - 1)) : 0;
-            unsigned long long bits =
-                (1ull << (h & 63)) | (1ull << ((h >> 6) & 63));
-            unsigned long long v = bloom[w]; /* word 0 stays hot for dupes */
-            int okv = (v & bits) == bits;
-            int lidx = isleader ? lane : -1;
-            for (int st = 1; st < WAVE; st <<= 1) {
-                int u = __shfl_up(lidx, st, WAVE);
-                if (lane >= st && u > lidx) lidx = u;
-            }
-            int ok = __shfl(okv, lidx, WAVE);
-            m[k] = datepass && ok;
         }
-#pragma unroll
-        for (int k = 0; k < B; k++) {
-            unsigned long long mask = __ballot(m[k]);
-            int cnt = __popcll(mask);
-            if (nbuf + cnt > BUF) {
+        if (q == nq) { /* tail rows (l.n % 4) */
+            for (int64_t i = nq * 4; i < l.n; i++) {
+                int j = (int)(i - nq * 4);
+                bool pass = l.l_shipdate[i] > q3date;
+                uint64_t h = otbx_splitmix64((uint64_t)l.l_orderkey[i]);
+                uint64_t w =
+                    pass ? ((h >> 12) & (uint64_t)(bloom_words - 1)) : 0;
+                unsigned long long bits =
+                    (1ull << (h & 63)) | (1ull << ((h >> 6) & 63));
+                m[j] = pass && ((bloom[w] & bits) == bits);
+                mycnt += m[j];
+            }
+        }
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        int tot = __shfl(incl, WAVE - 1, WAVE);
+        if (tot) {
+            if (nbuf + tot > BUF) {
                 long long bpos = 0;
                 if (lane == 0)
                     bpos = (long long)atomicAdd((unsigned long long *)ncand,
@@ -1539,12 +1538,15 @@ __global__ void k_q3_scan_filter(const otbx_lineitem_dev l, int32_t q3date,
                     cand[bpos + j] = buf[wid][j];
                 nbuf = 0;
             }
-            if (m[k]) {
-                int rank = __popcll(mask & ((1ull << lane) - 1ull));
-                buf[wid][nbuf + rank] = rows[k];
-            }
-            nbuf += cnt;
+            int pos = nbuf + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j])
+                    buf[wid][pos++] = r0 + j;
+            nbuf += tot;
         }
+        if (__all(q >= nq))
+            break;
     }
     if (nbuf) {
         long long bpos = 0;
