@@ -202,6 +202,16 @@ otbx_status otbx_agg_i64(const int64_t *keys_dev, const uint8_t *key_null_dev,
                          otbx_agg_group *groups_dev, int64_t *ngroups_dev,
                          void *stream);
 
+/* ---- GPU ORDER BY ----
+ * Full sort of Q3 group rows by (revenue DESC, o_orderdate ASC) — the
+ * tuplesort.c analog for the no-LIMIT ORDER BY case (LIMIT queries use
+ * otbx_topk_by_revenue). Stable LSD radix sort; out_dev must not alias
+ * groups_dev. */
+otbx_status otbx_order_groups_workspace_bytes(int64_t n, size_t *bytes);
+otbx_status otbx_order_groups(const otbx_q3_group *groups_dev, int64_t n,
+                              otbx_q3_group *out_dev, void *ws,
+                              size_t ws_bytes, void *stream);
+
 /* ---- repartition exchange ----
  * The GPU half of the reference's "Distribute results by H: col" exchange
  * (make_remotesubplan, optimizer/plan/createplan.c:8671; locator semantics

@@ -88,6 +88,7 @@ EXPORTED_SYMBOLS = [
     "otbx_scan_count", "otbx_q1_partial", "otbx_q1_partial_variant",
     "otbx_q3_workspace_bytes", "otbx_q3_partial", "otbx_filter_customer",
     "otbx_topk_by_revenue",
+    "otbx_order_groups_workspace_bytes", "otbx_order_groups",
     "otbx_agg_i64_workspace_bytes", "otbx_agg_i64",
     "otbx_partition_by_key", "otbx_gather_i64", "otbx_gather_f64",
     "otbx_gather_i32", "otbx_gather_u8",

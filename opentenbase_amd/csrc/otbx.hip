@@ -759,6 +759,214 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
 
 } /* extern "C" */
 
+/* ================= GPU ORDER BY =================
+ *
+ * Full ordering of Q3 group rows: ORDER BY revenue DESC, o_orderdate ASC
+ * (tuplesort.c analog). Stable LSD radix sort over (key u64, idx u32) pairs,
+ * 8-bit digits, two stages chained by stability:
+ *   stage A: date ASC  (2 passes — dates < 2^16)
+ *   stage B: ~bits(revenue) ASC ≡ revenue DESC (8 passes; revenue > 0 so the
+ *            raw IEEE bit pattern is order-preserving)
+ * Stable scatter: per-wave 8-ballot multi-split rank + per-digit wave/round
+ * prefix counters in LDS — lane order preserved, so every pass is stable.
+ */
+
+#define RS_BLOCK 256
+#define RS_BINS 256
+
+__global__ void k_rs_hist(const unsigned long long *__restrict__ keys,
+                          int64_t n, int shift, uint32_t *hist /* [bins][nblocks] */)
+{
+    __shared__ uint32_t lh[RS_BINS];
+    for (int i = threadIdx.x; i < RS_BINS; i += blockDim.x) lh[i] = 0;
+    __syncthreads();
+    int64_t per_block = (n + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < n ? lo + per_block : n;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        atomicAdd(&lh[(keys[i] >> shift) & 0xff], 1u);
+    __syncthreads();
+    for (int i = threadIdx.x; i < RS_BINS; i += blockDim.x)
+        hist[(size_t)i * gridDim.x + blockIdx.x] = lh[i];
+}
+
+/* stable scatter: block processes its chunk in rounds of 256 elements;
+ * within a round, rank among equal digits = wave multi-split (8 ballots,
+ * lane-ordered) + per-digit wave-offset prefix; across rounds, per-digit
+ * running counters. base[digit][block] comes from the scanned histogram. */
+__global__ void k_rs_scatter(const unsigned long long *__restrict__ keys,
+                             const uint32_t *__restrict__ vals, int64_t n,
+                             int shift,
+                             const uint32_t *__restrict__ base /* [bins][nblocks] */,
+                             unsigned long long *__restrict__ okeys,
+                             uint32_t *__restrict__ ovals)
+{
+    __shared__ uint32_t run[RS_BINS];        /* per-digit running count */
+    __shared__ uint32_t wcnt[4][RS_BINS];    /* per-wave per-digit counts */
+    for (int i = threadIdx.x; i < RS_BINS; i += blockDim.x) run[i] = 0;
+    __syncthreads();
+    int64_t per_block = (n + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < n ? lo + per_block : n;
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    for (int64_t r0 = lo; r0 < hi; r0 += blockDim.x) {
+        int64_t i = r0 + threadIdx.x;
+        bool valid = i < hi;
+        unsigned long long k = valid ? keys[i] : 0;
+        uint32_t v = valid ? vals[i] : 0;
+        int digit = (int)((k >> shift) & 0xff);
+        /* wave multi-split: match = lanes with my digit (8 ballots) */
+        unsigned long long match = ~0ull;
+#pragma unroll
+        for (int b = 0; b < 8; b++) {
+            unsigned long long bb = __ballot((digit >> b) & 1);
+            match &= ((digit >> b) & 1) ? bb : ~bb;
+        }
+        unsigned long long vb = __ballot(valid);
+        match &= vb;
+        int wrank = __popcll(match & ((1ull << lane) - 1ull));
+        int wtotal = __popcll(match);
+        /* per-wave digit counts (leader writes) */
+        for (int i2 = threadIdx.x; i2 < 4 * RS_BINS; i2 += blockDim.x)
+            ((uint32_t *)wcnt)[i2] = 0;
+        __syncthreads();
+        if (valid && lane == (__ffsll((long long)match) - 1))
+            wcnt[wid][digit] = (uint32_t)wtotal;
+        __syncthreads();
+        if (valid) {
+            uint32_t before = 0;
+            for (int w = 0; w < wid; w++) before += wcnt[w][digit];
+            int64_t pos = (int64_t)base[(size_t)digit * gridDim.x + blockIdx.x] +
+                          run[digit] + before + wrank;
+            okeys[pos] = k;
+            ovals[pos] = v;
+        }
+        __syncthreads();
+        /* advance running counters by this round's totals */
+        for (int d = threadIdx.x; d < RS_BINS; d += blockDim.x) {
+            uint32_t t = 0;
+            for (int w = 0; w < 4; w++) t += wcnt[w][d];
+            run[d] += t;
+        }
+        __syncthreads();
+    }
+}
+
+__global__ void k_rs_key_date(const otbx_q3_group *__restrict__ g,
+                              const uint32_t *__restrict__ idx, int64_t n,
+                              unsigned long long *keys)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        keys[i] = (unsigned long long)(uint32_t)g[idx[i]].o_orderdate;
+}
+
+__global__ void k_rs_key_revdesc(const otbx_q3_group *__restrict__ g,
+                                 const uint32_t *__restrict__ idx, int64_t n,
+                                 unsigned long long *keys)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        keys[i] = ~(unsigned long long)__double_as_longlong(g[idx[i]].revenue);
+}
+
+__global__ void k_rs_iota(uint32_t *idx, int64_t n)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        idx[i] = (uint32_t)i;
+}
+
+__global__ void k_rs_apply(const otbx_q3_group *__restrict__ g,
+                           const uint32_t *__restrict__ idx, int64_t n,
+                           otbx_q3_group *__restrict__ out)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        out[i] = g[idx[i]];
+}
+
+extern "C" {
+
+otbx_status otbx_order_groups_workspace_bytes(int64_t n, size_t *bytes)
+{
+    int64_t nb = grid_for(n, RS_BLOCK);
+    *bytes = (size_t)n * (8 + 8 + 4 + 4) /* key ping-pong + idx ping-pong */ +
+             (size_t)RS_BINS * nb * 4 * 2 /* hist + scanned base */ + 4096;
+    return OTBX_OK;
+}
+
+/* sorts groups by (revenue DESC, o_orderdate ASC) into out_dev (may not
+ * alias groups_dev). ORDER BY without LIMIT (§8f.2); ties in both keys keep
+ * arbitrary (but stable-by-input) order, as SQL allows. */
+otbx_status otbx_order_groups(const otbx_q3_group *groups_dev, int64_t n,
+                              otbx_q3_group *out_dev, void *ws,
+                              size_t ws_bytes, void *stream)
+{
+    if (n < 0) return OTBX_ERR_INVALID;
+    hipStream_t s = (hipStream_t)stream;
+    if (n == 0) return OTBX_OK;
+    int nb = grid_for(n, RS_BLOCK);
+    size_t need;
+    otbx_order_groups_workspace_bytes(n, &need);
+    if (ws_bytes < need) return OTBX_ERR_INVALID;
+    unsigned long long *kA = (unsigned long long *)ws;
+    unsigned long long *kB = kA + n;
+    uint32_t *iA = (uint32_t *)(kB + n);
+    uint32_t *iB = iA + n;
+    uint32_t *hist = iB + n;
+    uint32_t *basep = hist + (size_t)RS_BINS * nb;
+    static uint32_t *h_hist = nullptr;
+    static size_t h_cap = 0;
+    if (h_cap < (size_t)RS_BINS * nb) {
+        if (h_hist) HIP_CHECK(hipHostFree(h_hist));
+        h_cap = (size_t)RS_BINS * nb;
+        HIP_CHECK(hipHostMalloc(&h_hist, h_cap * 4));
+    }
+    hipLaunchKernelGGL(k_rs_iota, dim3(grid_for(n, 256)), dim3(256), 0, s, iA, n);
+
+    /* stage A: date ASC (2 passes), stage B: ~rev bits ASC (8 passes) */
+    for (int stage = 0; stage < 2; stage++) {
+        int passes = stage == 0 ? 2 : 8;
+        if (stage == 0)
+            hipLaunchKernelGGL(k_rs_key_date, dim3(grid_for(n, 256)), dim3(256),
+                               0, s, groups_dev, iA, n, kA);
+        else
+            hipLaunchKernelGGL(k_rs_key_revdesc, dim3(grid_for(n, 256)),
+                               dim3(256), 0, s, groups_dev, iA, n, kA);
+        for (int p = 0; p < passes; p++) {
+            int shift = p * 8;
+            hipLaunchKernelGGL(k_rs_hist, dim3(nb), dim3(RS_BLOCK), 0, s, kA, n,
+                               shift, hist);
+            HIP_CHECK(hipMemcpyAsync(h_hist, hist, (size_t)RS_BINS * nb * 4,
+                                     hipMemcpyDeviceToHost, s));
+            HIP_CHECK(hipStreamSynchronize(s));
+            uint32_t acc = 0;
+            for (size_t i = 0; i < (size_t)RS_BINS * nb; i++) {
+                uint32_t c = h_hist[i];
+                h_hist[i] = acc;
+                acc += c;
+            }
+            HIP_CHECK(hipMemcpyAsync(basep, h_hist, (size_t)RS_BINS * nb * 4,
+                                     hipMemcpyHostToDevice, s));
+            hipLaunchKernelGGL(k_rs_scatter, dim3(nb), dim3(RS_BLOCK), 0, s, kA,
+                               iA, n, shift, basep, kB, iB);
+            unsigned long long *tk = kA; kA = kB; kB = tk;
+            uint32_t *ti = iA; iA = iB; iB = ti;
+        }
+    }
+    hipLaunchKernelGGL(k_rs_apply, dim3(grid_for(n, 256)), dim3(256), 0, s,
+                       groups_dev, iA, n, out_dev);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+} /* extern "C" */
+
 /* ================= repartition exchange ================= */
 
 /* owner = key % nranks (dense-key locator restatement of shardid→node,

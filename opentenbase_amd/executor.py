@@ -336,6 +336,23 @@ def q3_topk(groups, k=10):
     return groups[order][:k]
 
 
+def order_groups(groups_dev_u8, n):
+    """ORDER BY revenue DESC, o_orderdate ASC over device q3 groups (raw
+    24-B-record uint8 tensor) — the full-sort operator.
+    Returns a sorted structured numpy array."""
+    import numpy as np
+    L = lib()
+    ws_bytes = C.c_size_t(0)
+    check(L.otbx_order_groups_workspace_bytes(C.c_int64(n), C.byref(ws_bytes)))
+    ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8, device="cuda")
+    out = torch.empty(max(n, 1) * 24, dtype=torch.uint8, device="cuda")
+    call("otbx_order_groups", C.c_void_p(groups_dev_u8.data_ptr()),
+         C.c_int64(n), C.c_void_p(out.data_ptr()), C.c_void_p(ws.data_ptr()),
+         C.c_size_t(ws_bytes.value), _stream())
+    raw = out[: n * 24].cpu().numpy().tobytes()
+    return np.frombuffer(raw, dtype=np.dtype(GpuQ3Fragment.NP_DTYPE)).copy()
+
+
 def partition_by_key(keys):
     """Repartition exchange, GPU half (the 'Distribute results
     by H: col' exchange of make_remotesubplan, createplan.c:8671): returns
