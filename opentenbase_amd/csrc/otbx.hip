@@ -945,6 +945,16 @@ otbx_status otbx_order_groups(const otbx_q3_group *groups_dev, int64_t n,
             HIP_CHECK(hipMemcpyAsync(h_hist, hist, (size_t)RS_BINS * nb * 4,
                                      hipMemcpyDeviceToHost, s));
             HIP_CHECK(hipStreamSynchronize(s));
+            /* identity pass (every key shares this digit)? skip the scatter —
+             * common for the high bytes of same-magnitude doubles */
+            int nonzero_digits = 0;
+            for (int d = 0; d < RS_BINS && nonzero_digits < 2; d++) {
+                uint32_t t = 0;
+                for (int b = 0; b < nb; b++) t += h_hist[(size_t)d * nb + b];
+                if (t) nonzero_digits++;
+            }
+            if (nonzero_digits < 2)
+                continue;
             uint32_t acc = 0;
             for (size_t i = 0; i < (size_t)RS_BINS * nb; i++) {
                 uint32_t c = h_hist[i];
@@ -2006,25 +2016,6 @@ __global__ void k_cust_bitmap_keys(const int64_t *__restrict__ keys, int64_t n,
         int64_t idx = keys[i] - minc;
         atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
     }
-}
-
-__global__ void k_count_orders_filtered(const otbx_orders_dev o,
-                                        const unsigned long long *__restrict__ ckeys,
-                                        int64_t ccap, int32_t q3date,
-                                        int64_t *count)
-{
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    unsigned long long my = 0;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < o.n;
-         i += stride) {
-        if (o.o_orderdate[i] < q3date &&
-            d_keyset_probe(ckeys, ccap, o.o_custkey[i]))
-            my++;
-    }
-    for (int off = WAVE / 2; off > 0; off >>= 1)
-        my += __shfl_down(my, off, WAVE);
-    if ((threadIdx.x % WAVE) == 0 && my)
-        atomicAdd((unsigned long long *)count, my);
 }
 
 extern "C" {
