@@ -305,8 +305,7 @@ class GpuQ3Fragment(CustomScanState):
              C.c_int64(cap_cand), C.c_void_p(ncand.data_ptr()),
              C.c_void_p(hist.data_ptr()), _stream())
         nc = min(int(ncand.cpu().item()), cap_cand)
-        raw = cand[: nc * 24].cpu().numpy().tobytes()
-        cands = np.frombuffer(raw, dtype=np.dtype(self.NP_DTYPE)).copy()
+        cands = cand[: nc * 24].cpu().numpy().view(np.dtype(self.NP_DTYPE))
         return [tuple(r) for r in q3_topk(cands, self.k)]
 
     def explain(self):
@@ -322,8 +321,8 @@ class GpuQ3Fragment(CustomScanState):
         """D2H copy of ALL partial groups (parity tests / debugging)."""
         import numpy as np
         if self.groups is None or len(self.groups) != self.ngroups:
-            raw = self._groups_dev[: self.ngroups * 24].cpu().numpy().tobytes()
-            self.groups = np.frombuffer(raw, dtype=np.dtype(self.NP_DTYPE)).copy()
+            self.groups = self._groups_dev[: self.ngroups * 24].cpu().numpy() \
+                .view(np.dtype(self.NP_DTYPE))
         return self.groups
 
 
@@ -349,8 +348,7 @@ def order_groups(groups_dev_u8, n):
     call("otbx_order_groups", C.c_void_p(groups_dev_u8.data_ptr()),
          C.c_int64(n), C.c_void_p(out.data_ptr()), C.c_void_p(ws.data_ptr()),
          C.c_size_t(ws_bytes.value), _stream())
-    raw = out[: n * 24].cpu().numpy().tobytes()
-    return np.frombuffer(raw, dtype=np.dtype(GpuQ3Fragment.NP_DTYPE)).copy()
+    return out[: n * 24].cpu().numpy().view(np.dtype(GpuQ3Fragment.NP_DTYPE))
 
 
 def partition_by_key(keys):
@@ -411,8 +409,7 @@ class GpuHashAgg(CustomScanState):
         ngroups = int(ng.cpu().item())
         dt = np.dtype([("key", "i8"), ("count_star", "i8"), ("count_v", "i8"),
                        ("sum_v", "f8"), ("key_isnull", "i4"), ("sum_isnull", "i4")])
-        raw = out[: ngroups * 40].cpu().numpy().tobytes()
-        arr = np.frombuffer(raw, dtype=dt).copy()
+        arr = out[: ngroups * 40].cpu().numpy().view(dt).copy()
         arr.sort(order=["key_isnull", "key"])
         return list(arr)
 
