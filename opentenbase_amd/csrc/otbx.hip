@@ -1313,6 +1313,137 @@ __device__ __forceinline__ bool d_bloom_test(const unsigned long long *bloom,
     return (bloom[w] & bits) == bits;
 }
 
+/* orders side, FUSED single pass (when the customer filter is the dense
+ * bitmap, its test is an L2-resident word — no reason to materialize the
+ * date-filtered candidate list first): 4 rows/lane via 16-B vector loads,
+ * date qual + custkey filter, LDS-staged append of matched row ids +
+ * matched-orderkey min/max for the direct-path decision. */
+__global__ void k_ord_filter_probe_fused(const otbx_orders_dev o,
+                                         int32_t q3date,
+                                         const unsigned long long *__restrict__ ckeys,
+                                         int64_t ccap,
+                                         const unsigned long long *__restrict__ cbitmap,
+                                         int64_t cmin, int64_t crange,
+                                         int64_t *__restrict__ out,
+                                         int64_t *nout,
+                                         unsigned long long *minkey,
+                                         unsigned long long *maxkey)
+{
+    const int BUF = 1024;
+    __shared__ int64_t buf[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0;
+    unsigned long long mymin = ~0ull >> 1, mymax = 0;
+    int64_t nq = o.n / 4;
+    const int4 *od4 = (const int4 *)o.o_orderdate;
+    const longlong2 *ck2 = (const longlong2 *)o.o_custkey;
+    const longlong2 *okk2 = (const longlong2 *)o.o_orderkey;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         q += stride) {
+        bool m[4] = {false, false, false, false};
+        int64_t r0 = q * 4;
+        int mycnt = 0;
+        if (q < nq) {
+            int4 d = od4[q];
+            longlong2 ca = ck2[2 * q], cb = ck2[2 * q + 1];
+            longlong2 ka = okk2[2 * q], kb = okk2[2 * q + 1];
+            int32_t ds[4] = {d.x, d.y, d.z, d.w};
+            int64_t ck[4] = {ca.x, ca.y, cb.x, cb.y};
+            int64_t ok[4] = {ka.x, ka.y, kb.x, kb.y};
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                bool pass = ds[j] < q3date;       /* ExecQual on orders */
+                if (pass) {
+                    if (cbitmap) {
+                        int64_t idx = ck[j] - cmin;
+                        pass = idx >= 0 && idx < crange &&
+                               ((cbitmap[idx >> 6] >> (idx & 63)) & 1ull);
+                    } else {
+                        pass = d_keyset_probe(ckeys, ccap, ck[j]);
+                    }
+                }
+                m[j] = pass;
+                if (pass) {
+                    mycnt++;
+                    unsigned long long k = (unsigned long long)ok[j];
+                    if (k < mymin) mymin = k;
+                    if (k > mymax) mymax = k;
+                }
+            }
+        } else if (q == nq) { /* tail rows */
+            for (int64_t i = nq * 4; i < o.n; i++) {
+                int j = (int)(i - nq * 4);
+                bool pass = o.o_orderdate[i] < q3date;
+                if (pass) {
+                    int64_t ckv = o.o_custkey[i];
+                    if (cbitmap) {
+                        int64_t idx = ckv - cmin;
+                        pass = idx >= 0 && idx < crange &&
+                               ((cbitmap[idx >> 6] >> (idx & 63)) & 1ull);
+                    } else {
+                        pass = d_keyset_probe(ckeys, ccap, ckv);
+                    }
+                }
+                m[j] = pass;
+                if (pass) {
+                    mycnt++;
+                    unsigned long long k = (unsigned long long)o.o_orderkey[i];
+                    if (k < mymin) mymin = k;
+                    if (k > mymax) mymax = k;
+                }
+            }
+        }
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        int tot = __shfl(incl, WAVE - 1, WAVE);
+        if (tot) {
+            if (nbuf + tot > BUF) {
+                long long bpos = 0;
+                if (lane == 0)
+                    bpos = (long long)atomicAdd((unsigned long long *)nout,
+                                                (unsigned long long)nbuf);
+                bpos = __shfl(bpos, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE)
+                    out[bpos + j] = buf[wid][j];
+                nbuf = 0;
+            }
+            int pos = nbuf + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j])
+                    buf[wid][pos++] = r0 + j;
+            nbuf += tot;
+        }
+        if (__all(q >= nq))
+            break;
+    }
+    if (nbuf) {
+        long long bpos = 0;
+        if (lane == 0)
+            bpos = (long long)atomicAdd((unsigned long long *)nout,
+                                        (unsigned long long)nbuf);
+        bpos = __shfl(bpos, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE)
+            out[bpos + j] = buf[wid][j];
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        unsigned long long mn = (unsigned long long)__shfl_down(
+            (long long)mymin, off, WAVE);
+        unsigned long long mx = (unsigned long long)__shfl_down(
+            (long long)mymax, off, WAVE);
+        if (mn < mymin) mymin = mn;
+        if (mx > mymax) mymax = mx;
+    }
+    if (lane == 0) {
+        if (mymin != (~0ull >> 1)) atomicMin(minkey, mymin);
+        if (mymax) atomicMax(maxkey, mymax);
+    }
+}
+
 /* orders side, candidate-list pipeline (same rationale as the lineitem
  * scan): date-filter to a dense row-id list, dense customer-keyset probe to
  * a matched list (count = table size), dense insert into the right-sized
@@ -2186,12 +2317,10 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     }
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev[1], s));
 
-    /* ---- phase 2: orders pipeline — date filter (dense list), keyset
-     * probe (dense matched list = exact count), right-sized insert + bloom */
-    hipLaunchKernelGGL(k_ord_filter_date, dim3(grid_for(o->n, 256)), dim3(256),
-                       0, s, *o, q3date, cand_o1, &hdr[3]);
-    hipLaunchKernelGGL(k_ord_probe_cust, dim3(grid_for(o->n, 256)), dim3(256),
-                       0, s, *o, cand_o1, &hdr[3], ctab, ccap, cbitmap, cmin,
+    /* ---- phase 2: orders side, fused date qual + customer filter →
+     * matched row-id list (exact count for the right-sized build) */
+    hipLaunchKernelGGL(k_ord_filter_probe_fused, dim3(grid_for(o->n, 256)),
+                       dim3(256), 0, s, *o, q3date, ctab, ccap, cbitmap, cmin,
                        crange, cand_o2, &hdr[1],
                        (unsigned long long *)&hdr[4],
                        (unsigned long long *)&hdr[5]);
