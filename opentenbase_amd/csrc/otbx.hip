@@ -1449,146 +1449,6 @@ __global__ void k_ord_filter_probe_fused(const otbx_orders_dev o,
  * a matched list (count = table size), dense insert into the right-sized
  * table + bloom. Replaces a divergent fused build + a duplicated counting
  * pass. */
-__global__ void k_ord_filter_date(const otbx_orders_dev o, int32_t q3date,
-                                  int64_t *__restrict__ out, int64_t *ncand)
-{
-    /* 4 rows per lane via one int4 load; per-wave LDS-staged append */
-    const int BUF = 1024;
-    __shared__ int64_t buf[256 / WAVE][BUF];
-    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
-    int nbuf = 0;
-    int64_t nq = o.n / 4;
-    const int4 *od4 = (const int4 *)o.o_orderdate;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
-         q += stride) {
-        bool m[4] = {false, false, false, false};
-        int64_t r0 = q * 4;
-        if (q < nq) {
-            int4 d = od4[q];
-            m[0] = d.x < q3date;
-            m[1] = d.y < q3date;
-            m[2] = d.z < q3date;
-            m[3] = d.w < q3date;
-        } else if (q == nq) {
-            for (int64_t i = nq * 4; i < o.n; i++)
-                m[(int)(i - nq * 4)] = o.o_orderdate[i] < q3date;
-        }
-        int mycnt = m[0] + m[1] + m[2] + m[3];
-        int incl = mycnt;
-        for (int off = 1; off < WAVE; off <<= 1) {
-            int up = __shfl_up(incl, off, WAVE);
-            if (lane >= off) incl += up;
-        }
-        int tot = __shfl(incl, WAVE - 1, WAVE);
-        if (tot) {
-            if (nbuf + tot > BUF) {
-                long long bpos = 0;
-                if (lane == 0)
-                    bpos = (long long)atomicAdd((unsigned long long *)ncand,
-                                                (unsigned long long)nbuf);
-                bpos = __shfl(bpos, 0, WAVE);
-                for (int j = lane; j < nbuf; j += WAVE)
-                    out[bpos + j] = buf[wid][j];
-                nbuf = 0;
-            }
-            int pos = nbuf + incl - mycnt;
-#pragma unroll
-            for (int j = 0; j < 4; j++)
-                if (m[j])
-                    buf[wid][pos++] = r0 + j;
-            nbuf += tot;
-        }
-        if (__all(q >= nq))
-            break;
-    }
-    if (nbuf) {
-        long long bpos = 0;
-        if (lane == 0)
-            bpos = (long long)atomicAdd((unsigned long long *)ncand,
-                                        (unsigned long long)nbuf);
-        bpos = __shfl(bpos, 0, WAVE);
-        for (int j = lane; j < nbuf; j += WAVE)
-            out[bpos + j] = buf[wid][j];
-    }
-}
-
-__global__ void k_ord_probe_cust(const otbx_orders_dev o,
-                                 const int64_t *__restrict__ cand,
-                                 const int64_t *__restrict__ ncand_p,
-                                 const unsigned long long *__restrict__ ckeys,
-                                 int64_t ccap,
-                                 const unsigned long long *__restrict__ cbitmap,
-                                 int64_t cmin, int64_t crange,
-                                 int64_t *__restrict__ out,
-                                 int64_t *nout, unsigned long long *minkey,
-                                 unsigned long long *maxkey)
-{
-    const int BUF = 1024;
-    __shared__ int64_t buf[256 / WAVE][BUF];
-    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
-    int nbuf = 0;
-    unsigned long long mymin = ~0ull >> 1, mymax = 0;
-    int64_t n = *ncand_p;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
-         ci += stride) {
-        int64_t i = cand[ci];
-        bool m;
-        if (cbitmap) {
-            int64_t idx = o.o_custkey[i] - cmin;
-            m = idx >= 0 && idx < crange &&
-                ((cbitmap[idx >> 6] >> (idx & 63)) & 1ull);
-        } else {
-            m = d_keyset_probe(ckeys, ccap, o.o_custkey[i]); /* ⋈ customer */
-        }
-        if (m) {
-            unsigned long long k = (unsigned long long)o.o_orderkey[i];
-            if (k < mymin) mymin = k;
-            if (k > mymax) mymax = k;
-        }
-        unsigned long long mask = __ballot(m);
-        int cnt = __popcll(mask);
-        if (nbuf + cnt > BUF) {
-            long long base = 0;
-            if (lane == 0)
-                base = (long long)atomicAdd((unsigned long long *)nout,
-                                            (unsigned long long)nbuf);
-            base = __shfl(base, 0, WAVE);
-            for (int j = lane; j < nbuf; j += WAVE)
-                out[base + j] = buf[wid][j];
-            nbuf = 0;
-        }
-        if (m) {
-            int rank = __popcll(mask & ((1ull << lane) - 1ull));
-            buf[wid][nbuf + rank] = i;
-        }
-        nbuf += cnt;
-    }
-    if (nbuf) {
-        long long base = 0;
-        if (lane == 0)
-            base = (long long)atomicAdd((unsigned long long *)nout,
-                                        (unsigned long long)nbuf);
-        base = __shfl(base, 0, WAVE);
-        for (int j = lane; j < nbuf; j += WAVE)
-            out[base + j] = buf[wid][j];
-    }
-    /* matched-key range for the dense-direct path decision */
-    for (int off = WAVE / 2; off > 0; off >>= 1) {
-        unsigned long long mn = (unsigned long long)__shfl_down(
-            (long long)mymin, off, WAVE);
-        unsigned long long mx = (unsigned long long)__shfl_down(
-            (long long)mymax, off, WAVE);
-        if (mn < mymin) mymin = mn;
-        if (mx > mymax) mymax = mx;
-    }
-    if (lane == 0) {
-        if (mymin != (~0ull >> 1)) atomicMin(minkey, mymin);
-        if (mymax) atomicMax(maxkey, mymax);
-    }
-}
-
 __global__ void k_ord_insert(const otbx_orders_dev o,
                              const int64_t *__restrict__ cand,
                              const int64_t *__restrict__ ncand_p,
@@ -2194,7 +2054,7 @@ otbx_status otbx_q3_workspace_bytes(int64_t ncust, int64_t norders,
     *bytes = 64 + (size_t)ccap_w * 8 + ((size_t)1 << 23) * 8 /* bloom */ +
              (size_t)ocap_w * sizeof(ord_slot) + (size_t)ocap_w * sizeof(q3g_slot) +
              (size_t)nlineitem * 8 /* lineitem candidates */ +
-             (size_t)norders * 16 /* orders candidate lists (2) */ +
+             (size_t)norders * 8 /* matched-orders candidate list */ +
              (size_t)(dcap / 8 + 64) /* direct orderkey bitmap */ +
              (size_t)dcap * 8 /* direct payload */ +
              (size_t)dcap * 8 /* direct revenue */ +
@@ -2233,8 +2093,7 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     ord_slot *otab = (ord_slot *)((char *)bloom + ((size_t)1 << 23) * 8);
     q3g_slot *gtab = (q3g_slot *)((char *)otab + (size_t)ocap_w * sizeof(ord_slot));
     int64_t *cand_li = (int64_t *)((char *)gtab + (size_t)ocap_w * sizeof(q3g_slot));
-    int64_t *cand_o1 = cand_li + l->n;
-    int64_t *cand_o2 = cand_o1 + o->n;
+    int64_t *cand_o2 = cand_li + l->n;
     int64_t dcap = direct_cap_for(o->n);
     int64_t dcap_c = direct_cap_for(ncust);
     unsigned long long *dbitmap = (unsigned long long *)(cand_o2 + o->n);
