@@ -400,20 +400,6 @@ __global__ void k_q1_partial(const int32_t *__restrict__ sd,
 /* variant 2: 4 rows/lane (int4 dates, uchar4 flags, 2× double2 per column),
  * optional non-temporal loads on the read-once f64 streams (NT template). */
 typedef double v2d __attribute__((ext_vector_type(2)));
-typedef int v4i __attribute__((ext_vector_type(4)));
-typedef long long v2l __attribute__((ext_vector_type(2)));
-
-__device__ __forceinline__ int4 ld4i_nt(const int4 *p)
-{
-    v4i v = __builtin_nontemporal_load((const v4i *)p);
-    return make_int4(v.x, v.y, v.z, v.w);
-}
-
-__device__ __forceinline__ longlong2 ld2l_nt(const longlong2 *p)
-{
-    v2l v = __builtin_nontemporal_load((const v2l *)p);
-    return make_longlong2(v.x, v.y);
-}
 
 template <bool NT>
 __device__ __forceinline__ double2 ld2(const double2 *p)
@@ -1359,9 +1345,9 @@ __global__ void k_ord_filter_probe_fused(const otbx_orders_dev o,
         int64_t r0 = q * 4;
         int mycnt = 0;
         if (q < nq) {
-            int4 d = ld4i_nt(&od4[q]);      /* read-once streams: NT */
-            longlong2 ca = ld2l_nt(&ck2[2 * q]), cb = ld2l_nt(&ck2[2 * q + 1]);
-            longlong2 ka = ld2l_nt(&okk2[2 * q]), kb = ld2l_nt(&okk2[2 * q + 1]);
+            int4 d = od4[q];
+            longlong2 ca = ck2[2 * q], cb = ck2[2 * q + 1];
+            longlong2 ka = okk2[2 * q], kb = okk2[2 * q + 1];
             int32_t ds[4] = {d.x, d.y, d.z, d.w};
             int64_t ck[4] = {ca.x, ca.y, cb.x, cb.y};
             int64_t ok[4] = {ka.x, ka.y, kb.x, kb.y};
@@ -1543,8 +1529,8 @@ __global__ void k_q3_scan_filter_direct(const otbx_lineitem_dev l,
         int64_t r0 = q * 4;
         int mycnt = 0;
         if (q < nq) {
-            int4 d = ld4i_nt(&sd4[q]);      /* read-once streams: NT */
-            longlong2 ka = ld2l_nt(&ok2[2 * q]), kb = ld2l_nt(&ok2[2 * q + 1]);
+            int4 d = sd4[q];
+            longlong2 ka = ok2[2 * q], kb = ok2[2 * q + 1];
             int32_t ds[4] = {d.x, d.y, d.z, d.w};
             int64_t ky[4] = {ka.x, ka.y, kb.x, kb.y};
 #pragma unroll
