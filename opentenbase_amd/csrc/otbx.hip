@@ -852,6 +852,71 @@ __global__ void k_rs_scatter(const unsigned long long *__restrict__ keys,
     }
 }
 
+/* device-side exclusive scan of the [bins][nblocks] histogram (≤ 524k
+ * entries): block partials → single-block scan of partials → add back.
+ * Removes the per-pass host round trip. */
+#define RS_SCAN_CHUNK 1024
+
+__global__ void k_rs_scan_partials(const uint32_t *__restrict__ in, int64_t n,
+                                   uint32_t *__restrict__ sums)
+{
+    __shared__ uint32_t sh[256];
+    int64_t lo = (int64_t)blockIdx.x * RS_SCAN_CHUNK;
+    uint32_t my = 0;
+    for (int i = threadIdx.x; i < RS_SCAN_CHUNK; i += blockDim.x) {
+        int64_t j = lo + i;
+        my += j < n ? in[j] : 0;
+    }
+    sh[threadIdx.x] = my;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t t = 0;
+        for (int i = 0; i < 256; i++) t += sh[i];
+        sums[blockIdx.x] = t;
+    }
+}
+
+__global__ void k_rs_scan_sums(uint32_t *sums, int64_t nb)
+{
+    /* single block: exclusive scan of ≤512 partials */
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        uint32_t acc = 0;
+        for (int64_t i = 0; i < nb; i++) {
+            uint32_t c = sums[i];
+            sums[i] = acc;
+            acc += c;
+        }
+    }
+}
+
+__global__ void k_rs_scan_apply(const uint32_t *__restrict__ in, int64_t n,
+                                const uint32_t *__restrict__ sums,
+                                uint32_t *__restrict__ out)
+{
+    /* per chunk: sequential exclusive scan by thread 0 over LDS-staged
+     * values (1024 adds — trivial next to the global traffic) */
+    __shared__ uint32_t sh[RS_SCAN_CHUNK];
+    int64_t lo = (int64_t)blockIdx.x * RS_SCAN_CHUNK;
+    for (int i = threadIdx.x; i < RS_SCAN_CHUNK; i += blockDim.x) {
+        int64_t j = lo + i;
+        sh[i] = j < n ? in[j] : 0;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t acc = sums[blockIdx.x];
+        for (int i = 0; i < RS_SCAN_CHUNK; i++) {
+            uint32_t c = sh[i];
+            sh[i] = acc;
+            acc += c;
+        }
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < RS_SCAN_CHUNK; i += blockDim.x) {
+        int64_t j = lo + i;
+        if (j < n) out[j] = sh[i];
+    }
+}
+
 __global__ void k_rs_key_date(const otbx_q3_group *__restrict__ g,
                               const uint32_t *__restrict__ idx, int64_t n,
                               unsigned long long *keys)
@@ -896,7 +961,8 @@ otbx_status otbx_order_groups_workspace_bytes(int64_t n, size_t *bytes)
 {
     int64_t nb = grid_for(n, RS_BLOCK);
     *bytes = (size_t)n * (8 + 8 + 4 + 4) /* key ping-pong + idx ping-pong */ +
-             (size_t)RS_BINS * nb * 4 * 2 /* hist + scanned base */ + 4096;
+             (size_t)RS_BINS * nb * 4 * 2 /* hist + scanned base */ +
+             4096 /* scan partial sums */ + 4096;
     return OTBX_OK;
 }
 
@@ -920,13 +986,7 @@ otbx_status otbx_order_groups(const otbx_q3_group *groups_dev, int64_t n,
     uint32_t *iB = iA + n;
     uint32_t *hist = iB + n;
     uint32_t *basep = hist + (size_t)RS_BINS * nb;
-    static uint32_t *h_hist = nullptr;
-    static size_t h_cap = 0;
-    if (h_cap < (size_t)RS_BINS * nb) {
-        if (h_hist) HIP_CHECK(hipHostFree(h_hist));
-        h_cap = (size_t)RS_BINS * nb;
-        HIP_CHECK(hipHostMalloc(&h_hist, h_cap * 4));
-    }
+    uint32_t *scansums = basep + (size_t)RS_BINS * nb;
     hipLaunchKernelGGL(k_rs_iota, dim3(grid_for(n, 256)), dim3(256), 0, s, iA, n);
 
     /* stage A: date ASC (2 passes), stage B: ~rev bits ASC (8 passes) */
@@ -942,27 +1002,15 @@ otbx_status otbx_order_groups(const otbx_q3_group *groups_dev, int64_t n,
             int shift = p * 8;
             hipLaunchKernelGGL(k_rs_hist, dim3(nb), dim3(RS_BLOCK), 0, s, kA, n,
                                shift, hist);
-            HIP_CHECK(hipMemcpyAsync(h_hist, hist, (size_t)RS_BINS * nb * 4,
-                                     hipMemcpyDeviceToHost, s));
-            HIP_CHECK(hipStreamSynchronize(s));
-            /* identity pass (every key shares this digit)? skip the scatter —
-             * common for the high bytes of same-magnitude doubles */
-            int nonzero_digits = 0;
-            for (int d = 0; d < RS_BINS && nonzero_digits < 2; d++) {
-                uint32_t t = 0;
-                for (int b = 0; b < nb; b++) t += h_hist[(size_t)d * nb + b];
-                if (t) nonzero_digits++;
-            }
-            if (nonzero_digits < 2)
-                continue;
-            uint32_t acc = 0;
-            for (size_t i = 0; i < (size_t)RS_BINS * nb; i++) {
-                uint32_t c = h_hist[i];
-                h_hist[i] = acc;
-                acc += c;
-            }
-            HIP_CHECK(hipMemcpyAsync(basep, h_hist, (size_t)RS_BINS * nb * 4,
-                                     hipMemcpyHostToDevice, s));
+            /* device-side exclusive scan of the [bins][nblocks] histogram */
+            int64_t hn = (int64_t)RS_BINS * nb;
+            int64_t snb = (hn + RS_SCAN_CHUNK - 1) / RS_SCAN_CHUNK;
+            hipLaunchKernelGGL(k_rs_scan_partials, dim3((uint32_t)snb),
+                               dim3(256), 0, s, hist, hn, scansums);
+            hipLaunchKernelGGL(k_rs_scan_sums, dim3(1), dim3(64), 0, s,
+                               scansums, snb);
+            hipLaunchKernelGGL(k_rs_scan_apply, dim3((uint32_t)snb), dim3(256),
+                               0, s, hist, hn, scansums, basep);
             hipLaunchKernelGGL(k_rs_scatter, dim3(nb), dim3(RS_BLOCK), 0, s, kA,
                                iA, n, shift, basep, kB, iB);
             unsigned long long *tk = kA; kA = kB; kB = tk;
