@@ -1977,10 +1977,29 @@ __global__ void k_count_customer_seg(const otbx_customer_dev c, uint8_t want,
                                      unsigned long long *minkey,
                                      unsigned long long *maxkey)
 {
+    /* 4 rows/lane: uchar4 segment loads; custkey loaded only for matches */
+    int64_t nq = c.n / 4;
+    const uchar4 *seg4 = (const uchar4 *)c.c_mktsegment;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     unsigned long long my = 0, mymin = ~0ull >> 1, mymax = 0;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < c.n;
-         i += stride) {
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
+         q += stride) {
+        uchar4 sv = seg4[q];
+        uint8_t ss[4] = {sv.x, sv.y, sv.z, sv.w};
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+            if (ss[j] == want) {
+                my++;
+                unsigned long long k =
+                    (unsigned long long)c.c_custkey[q * 4 + j];
+                if (k < mymin) mymin = k;
+                if (k > mymax) mymax = k;
+            }
+        }
+    }
+    /* tail */
+    for (int64_t i = nq * 4 + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < c.n; i += stride) {
         if (c.c_mktsegment[i] == want) {
             my++;
             unsigned long long k = (unsigned long long)c.c_custkey[i];
