@@ -676,6 +676,10 @@ __global__ void k_agg_build(const int64_t *__restrict__ keys,
         agg_slot *e;
         if (kn) {
             e = nullgrp; /* NULL keys form one group (execGrouping.c:520) */
+        } else if (keys[i] == AGG_EMPTY) {
+            /* the open-addressing empty sentinel value gets a dedicated
+             * accumulator (nullgrp[1]) so ANY i64 key is supported */
+            e = nullgrp + 1;
         } else {
             int64_t k = keys[i];
             int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)mask);
@@ -714,14 +718,25 @@ __global__ void k_agg_compact(const agg_slot *tab, int64_t cap,
             out[pos].sum_isnull = tab[i].count_v == 0;
         }
     }
-    if (blockIdx.x == 0 && threadIdx.x == 0 && nullgrp->count_star > 0) {
-        int64_t pos = (int64_t)atomicAdd((unsigned long long *)ngroups, 1ull);
-        out[pos].key = 0;
-        out[pos].key_isnull = 1;
-        out[pos].count_star = (int64_t)nullgrp->count_star;
-        out[pos].count_v = (int64_t)nullgrp->count_v;
-        out[pos].sum_v = nullgrp->sum_v;
-        out[pos].sum_isnull = nullgrp->count_v == 0;
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        if (nullgrp->count_star > 0) {
+            int64_t pos = (int64_t)atomicAdd((unsigned long long *)ngroups, 1ull);
+            out[pos].key = 0;
+            out[pos].key_isnull = 1;
+            out[pos].count_star = (int64_t)nullgrp->count_star;
+            out[pos].count_v = (int64_t)nullgrp->count_v;
+            out[pos].sum_v = nullgrp->sum_v;
+            out[pos].sum_isnull = nullgrp->count_v == 0;
+        }
+        if (nullgrp[1].count_star > 0) { /* the AGG_EMPTY-valued key group */
+            int64_t pos = (int64_t)atomicAdd((unsigned long long *)ngroups, 1ull);
+            out[pos].key = AGG_EMPTY;
+            out[pos].key_isnull = 0;
+            out[pos].count_star = (int64_t)nullgrp[1].count_star;
+            out[pos].count_v = (int64_t)nullgrp[1].count_v;
+            out[pos].sum_v = nullgrp[1].sum_v;
+            out[pos].sum_isnull = nullgrp[1].count_v == 0;
+        }
     }
 }
 
@@ -730,7 +745,8 @@ extern "C" {
 otbx_status otbx_agg_i64_workspace_bytes(int64_t n, size_t *bytes)
 {
     int64_t cap = next_pow2_host(n < 16 ? 16 : (int64_t)(n / 0.7) + 1);
-    *bytes = (size_t)(cap + 1) * sizeof(agg_slot);
+    *bytes = (size_t)(cap + 2) * sizeof(agg_slot); /* + NULL and
+                                                    * sentinel-key groups */
     return OTBX_OK;
 }
 
@@ -740,12 +756,12 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
                          int64_t *ngroups_dev, void *stream)
 {
     int64_t cap = next_pow2_host(n < 16 ? 16 : (int64_t)(n / 0.7) + 1);
-    if (ws_bytes < (size_t)(cap + 1) * sizeof(agg_slot)) return OTBX_ERR_INVALID;
+    if (ws_bytes < (size_t)(cap + 2) * sizeof(agg_slot)) return OTBX_ERR_INVALID;
     hipStream_t s = (hipStream_t)stream;
     agg_slot *tab = (agg_slot *)ws;
     agg_slot *nullgrp = tab + cap;
-    hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap + 1, 256)), dim3(256), 0, s,
-                       tab, cap + 1);
+    hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap + 2, 256)), dim3(256), 0, s,
+                       tab, cap + 2);
     HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, sizeof(int64_t), s));
     if (n > 0) {
         hipLaunchKernelGGL(k_agg_build, dim3(grid_for(n, 256)), dim3(256), 0, s,

@@ -476,3 +476,18 @@ def test_order_groups_edges(ex):
     geq = torch.from_numpy(eq.view(np.uint8).reshape(-1).copy()).cuda()
     out = ex.order_groups(geq, 1000)
     assert (np.diff(out["o_orderdate"]) >= 0).all()
+
+
+def test_agg_int64min_key(ex, ora):
+    """ANY i64 key is groupable, including the open-addressing sentinel
+    value INT64_MIN (routed to a dedicated accumulator)."""
+    imin = -(2**63)
+    keys = np.array([imin, 5, imin, 7, imin], dtype=np.int64)
+    vals = np.array([1.0, 2.0, 3.0, 4.0, 5.0])
+    got = _agg(ex, keys, vals)
+    exp = ora.agg_i64(keys, vals)
+    assert len(got) == len(exp) == 3
+    gm = {int(g["key"]): (int(g["count_star"]), float(g["sum_v"])) for g in got}
+    assert gm[imin] == (3, 9.0)
+    assert gm[5] == (1, 2.0)
+    assert gm[7] == (1, 4.0)
