@@ -190,6 +190,7 @@ def main():
     for _ in range(args.warmup):
         step()
     kernel_ms_acc.clear()
+    hits_acc.clear()
 
     barrier_sync()
     t_start = time.time()
