@@ -143,6 +143,7 @@ def main():
 
     kernel_ms_acc = []
     hits_acc = []
+    phase_ms = []
 
     def step():
         if args.workload == "tpch_q1":
@@ -178,6 +179,7 @@ def main():
             # probe+partial-agg kernel is the roofline-dominant phase
             kernel_ms_acc.append(node.kernel_ms[2])
             hits_acc.append(node.probe_hits)
+            phase_ms.append(node.kernel_ms)
             import numpy as np
             cands = np.array(node._rows, dtype=np.dtype(node.NP_DTYPE))
             return fragment.merge_q3_topk(cands, 10)
@@ -191,6 +193,7 @@ def main():
         step()
     kernel_ms_acc.clear()
     hits_acc.clear()
+    phase_ms.clear()
 
     barrier_sync()
     t_start = time.time()
@@ -266,6 +269,10 @@ def main():
                 # cold-cache cost: one-time on-device staging,
                 # outside the timed region; the timed steps are hot-cache
                 "staging_s": round(staging_s, 3),
+                # Q3 only: per-phase HIP-event ms of the last step
+                # [customer build, orders side, probe window, compact]
+                "phase_ms": [round(x, 3) for x in phase_ms[-1]]
+                if phase_ms else None,
             },
             "roofline": roofline,
             "cpu_baseline": cpu,
