@@ -26,6 +26,7 @@
 #include <hip/hip_runtime.h>
 #include <limits.h>
 #include <stdio.h>
+#include <stdlib.h>  /* getenv, atoll */
 #include <string.h>
 
 #include "../../include/otbx.h"
