@@ -1514,11 +1514,16 @@ __global__ void k_ord_filter_probe_fused(const otbx_orders_dev o,
  * a matched list (count = table size), dense insert into the right-sized
  * table + bloom. Replaces a divergent fused build + a duplicated counting
  * pass. */
+/* pass/npasses: grace batching for the hash path — a pass
+ * owns the keys with hash % npasses == pass, so per-pass table memory is
+ * bounded; the scan's bloom filter is rebuilt per pass and naturally admits
+ * only that pass's keys. */
 __global__ void k_ord_insert(const otbx_orders_dev o,
                              const int64_t *__restrict__ cand,
                              const int64_t *__restrict__ ncand_p,
                              ord_slot *tab, int64_t cap,
-                             unsigned long long *bloom, int64_t bloom_words)
+                             unsigned long long *bloom, int64_t bloom_words,
+                             uint32_t pass, uint32_t npasses)
 {
     int64_t mask = cap - 1;
     int64_t n = *ncand_p;
@@ -1527,13 +1532,36 @@ __global__ void k_ord_insert(const otbx_orders_dev o,
          ci += stride) {
         int64_t i = cand[ci];
         unsigned long long k = (unsigned long long)o.o_orderkey[i];
+        uint64_t h = d_hash_i64((int64_t)k);
+        if (npasses > 1 && (uint32_t)(h >> 32) % npasses != pass)
+            continue;
         d_bloom_set(bloom, bloom_words, (int64_t)k);
-        int64_t s = (int64_t)(d_hash_i64((int64_t)k) & (uint64_t)mask);
+        int64_t s = (int64_t)(h & (uint64_t)mask);
         while (atomicCAS(&tab[s].okey, 0ull, k) != 0ull)
             s = (s + 1) & mask;      /* keys unique: claim exactly one slot */
         tab[s].date = o.o_orderdate[i];   /* plain: read by NEXT launch */
         tab[s].prio = o.o_shippriority[i];
     }
+}
+
+__global__ void k_ord_count_pass(const otbx_orders_dev o,
+                                 const int64_t *__restrict__ cand,
+                                 const int64_t *__restrict__ ncand_p,
+                                 uint32_t pass, uint32_t npasses,
+                                 int64_t *count)
+{
+    int64_t n = *ncand_p;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long my = 0;
+    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
+         ci += stride) {
+        uint64_t h = d_hash_i64(o.o_orderkey[cand[ci]]);
+        my += (uint32_t)(h >> 32) % npasses == pass;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        my += __shfl_down(my, off, WAVE);
+    if ((threadIdx.x % WAVE) == 0 && my)
+        atomicAdd((unsigned long long *)count, my);
 }
 
 /* ---- dense-orderkey DIRECT path ----
@@ -2324,27 +2352,65 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
         }
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
     } else {
-    int64_t ocap = fit_cap(nof), gcap = ocap;
-    int64_t bwords = bloom_words_for(nof);
-    HIP_CHECK(hipMemsetAsync(bloom, 0, (size_t)bwords * 8, s));
-    HIP_CHECK(hipMemsetAsync(otab, 0, (size_t)ocap * sizeof(ord_slot), s));
-    HIP_CHECK(hipMemsetAsync(gtab, 0, (size_t)gcap * sizeof(q3g_slot), s));
-    hipLaunchKernelGGL(k_ord_insert, dim3(grid_for(o->n, 256)), dim3(256), 0,
-                       s, *o, cand_o2, &hdr[1], otab, ocap, bloom, bwords);
-    if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
-
-    /* ---- phase 3: lineitem scan+filter (compacted candidates), dense
-     * probe + partial agg */
-    hipLaunchKernelGGL(k_q3_scan_filter, dim3(grid_for(l->n, 256)), dim3(256),
-                       0, s, *l, q3date, bloom, bwords, cand_li, &hdr[2]);
-    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(l->n, 256)), dim3(256), 0,
-                       s, *l, cand_li, &hdr[2], otab, ocap, gtab, gcap, nhits);
-    if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
-
-    /* ---- phase 4: compact groups */
-    hipLaunchKernelGGL(k_q3_compact, dim3(grid_for(gcap, 256)), dim3(256), 0, s,
-                       gtab, gcap, groups_dev, cap_groups, ngroups_dev);
-    if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
+        /* hash+bloom fallback (wide/sparse key ranges), with grace batching
+         * when the build side exceeds the per-pass table budget: keys are
+         * hash-partitioned over P passes, each pass builds a bounded table
+         * and re-scans the probe stream (nodeHash.c:1086 batch tradeoff). */
+        int64_t hash_budget = 1ll << 30;  /* slots per pass (~40 GB tables) */
+        const char *hb = getenv("OTBX_Q3_HASH_BUDGET"); /* test hook */
+        if (hb) {
+            int64_t e = atoll(hb);
+            if (e >= 1024) hash_budget = e;
+        }
+        uint32_t nph = (uint32_t)((nof + hash_budget - 1) / hash_budget);
+        if (nph < 1) nph = 1;
+        if (nph > 64) nph = 64;
+        bool rec2 = false, rec3 = false;
+        for (uint32_t pass = 0; pass < nph; pass++) {
+            int64_t nof_p = nof;
+            if (nph > 1) {
+                HIP_CHECK(hipMemsetAsync(&hdr[3], 0, 8, s));
+                hipLaunchKernelGGL(k_ord_count_pass, dim3(grid_for(o->n, 256)),
+                                   dim3(256), 0, s, *o, cand_o2, &hdr[1], pass,
+                                   nph, &hdr[3]);
+                HIP_CHECK(hipMemcpyAsync(h_cnt + 3, hdr + 3, 8,
+                                         hipMemcpyDeviceToHost, s));
+                HIP_CHECK(hipStreamSynchronize(s));
+                nof_p = h_cnt[3];
+                if (nof_p == 0)
+                    continue;
+                HIP_CHECK(hipMemsetAsync(&hdr[2], 0, 8, s)); /* reset cands */
+            }
+            int64_t ocap = fit_cap(nof_p), gcap = ocap;
+            int64_t bwords = bloom_words_for(nof_p);
+            HIP_CHECK(hipMemsetAsync(bloom, 0, (size_t)bwords * 8, s));
+            HIP_CHECK(hipMemsetAsync(otab, 0, (size_t)ocap * sizeof(ord_slot), s));
+            HIP_CHECK(hipMemsetAsync(gtab, 0, (size_t)gcap * sizeof(q3g_slot), s));
+            hipLaunchKernelGGL(k_ord_insert, dim3(grid_for(o->n, 256)),
+                               dim3(256), 0, s, *o, cand_o2, &hdr[1], otab,
+                               ocap, bloom, bwords, pass, nph);
+            if (kernel_ms && !rec2) {
+                HIP_CHECK(hipEventRecord(ev[2], s));
+                rec2 = true;
+            }
+            /* lineitem scan+filter (compacted candidates), dense probe +
+             * partial agg */
+            hipLaunchKernelGGL(k_q3_scan_filter, dim3(grid_for(l->n, 256)),
+                               dim3(256), 0, s, *l, q3date, bloom, bwords,
+                               cand_li, &hdr[2]);
+            hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(l->n, 256)),
+                               dim3(256), 0, s, *l, cand_li, &hdr[2], otab,
+                               ocap, gtab, gcap, nhits);
+            if (kernel_ms && !rec3 && pass == nph - 1) {
+                HIP_CHECK(hipEventRecord(ev[3], s));
+                rec3 = true;
+            }
+            /* compact this pass's groups (append) */
+            hipLaunchKernelGGL(k_q3_compact, dim3(grid_for(gcap, 256)),
+                               dim3(256), 0, s, gtab, gcap, groups_dev,
+                               cap_groups, ngroups_dev);
+        }
+        if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
     }
     HIP_CHECK(hipGetLastError());
     if (kernel_ms) {

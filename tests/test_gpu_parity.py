@@ -491,3 +491,32 @@ def test_agg_int64min_key(ex, ora):
     assert gm[imin] == (3, 9.0)
     assert gm[5] == (1, 2.0)
     assert gm[7] == (1, 4.0)
+
+
+def test_q3_hash_grace_multipass_parity(ex, ora):
+    """Grace batching on the HASH fallback: force both the hash path and a
+    tiny per-pass table budget (→ ~13 passes); results must equal the
+    oracle."""
+    import os
+    n = 400000
+    li = ex.GpuLineitem.generate(n)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    cu = ex.GpuCustomer.generate(n // 40)
+    os.environ["OTBX_Q3_FORCE_HASH"] = "1"
+    os.environ["OTBX_Q3_HASH_BUDGET"] = "1024"
+    try:
+        node = ex.GpuQ3Fragment(cu, od, li)
+        drain(node)
+        got = {int(k): float(v) for k, v in
+               zip(node.fetch_groups()["l_orderkey"],
+                   node.fetch_groups()["revenue"])}
+        hits = node.probe_hits
+    finally:
+        del os.environ["OTBX_Q3_FORCE_HASH"]
+        del os.environ["OTBX_Q3_HASH_BUDGET"]
+    og = q3_oracle(ora, n)
+    exp = dict(zip(og["l_orderkey"].tolist(), og["revenue"].tolist()))
+    assert got.keys() == exp.keys()
+    for k in exp:
+        assert approx(got[k], exp[k]), k
+    assert hits >= len(exp)
