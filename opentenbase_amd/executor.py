@@ -317,9 +317,14 @@ class GpuQ3Fragment(CustomScanState):
                  "group compaction (emit)"]
         return dict(zip(names, self.kernel_ms))
 
-    def fetch_groups(self):
-        """D2H copy of ALL partial groups (parity tests / debugging)."""
+    def fetch_groups(self, ordered=False):
+        """D2H copy of ALL partial groups (parity tests / debugging).
+        ordered=True returns them fully sorted by (revenue DESC,
+        o_orderdate ASC) via the GPU radix sort (the no-LIMIT ORDER BY
+        path)."""
         import numpy as np
+        if ordered:
+            return order_groups(self._groups_dev, self.ngroups)
         if self.groups is None or len(self.groups) != self.ngroups:
             self.groups = self._groups_dev[: self.ngroups * 24].cpu().numpy() \
                 .view(np.dtype(self.NP_DTYPE))

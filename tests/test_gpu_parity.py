@@ -520,3 +520,16 @@ def test_q3_hash_grace_multipass_parity(ex, ora):
     for k in exp:
         assert approx(got[k], exp[k]), k
     assert hits >= len(exp)
+
+
+def test_fetch_groups_ordered(ex):
+    n = 400000
+    li = ex.GpuLineitem.generate(n)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    cu = ex.GpuCustomer.generate(n // 40)
+    node = ex.GpuQ3Fragment(cu, od, li)
+    drain(node)
+    g = node.fetch_groups(ordered=True)
+    key = list(zip((-g["revenue"]).tolist(), g["o_orderdate"].tolist()))
+    assert key == sorted(key)
+    assert len(g) == node.ngroups
