@@ -1295,23 +1295,71 @@ __global__ void k_join_probe(const join_slot *__restrict__ tab, int64_t cap,
                              int64_t *__restrict__ out_p, int64_t cap_pairs,
                              int64_t *npairs)
 {
+    /* wave-uniform chain walk (all lanes step while any is walking, so the
+     * match ballot is wave-converged) + per-wave LDS pair buffer: one
+     * global counter reservation per 512 pairs instead of per pair. */
+    const int BUF = 512;
+    __shared__ int64_t bufb[256 / WAVE][BUF];
+    __shared__ int64_t bufp[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0; /* wave-uniform */
     int64_t mask = cap - 1;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < np;
-         i += stride) {
-        if (pnull && pnull[i])
-            continue;
-        int64_t k = pkeys[i];
+    for (int64_t i0 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         i0 += stride) {
+        bool active = i0 < np && !(pnull && pnull[i0]);
+        int64_t k = pkeys[active ? i0 : 0];
         int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)mask);
-        while (tab[s].idx >= 0) {  /* walk to first unclaimed slot */
-            if (tab[s].key == k) {
-                int64_t pos = (int64_t)atomicAdd((unsigned long long *)npairs, 1ull);
-                if (pos < cap_pairs) {
-                    out_b[pos] = tab[s].idx;
-                    out_p[pos] = i;
+        bool walking = active;
+        while (__any(walking)) {
+            long long bidx = walking ? tab[s].idx : -1;
+            bool have = walking && bidx >= 0;
+            bool match = have && tab[s].key == k;
+            /* buffered pair append (converged point) */
+            unsigned long long mmask = __ballot(match);
+            int cnt = __popcll(mmask);
+            if (cnt) {
+                if (nbuf + cnt > BUF) {
+                    long long base = 0;
+                    if (lane == 0)
+                        base = (long long)atomicAdd(
+                            (unsigned long long *)npairs,
+                            (unsigned long long)nbuf);
+                    base = __shfl(base, 0, WAVE);
+                    for (int j = lane; j < nbuf; j += WAVE) {
+                        int64_t pos = base + j;
+                        if (pos < cap_pairs) {
+                            out_b[pos] = bufb[wid][j];
+                            out_p[pos] = bufp[wid][j];
+                        }
+                    }
+                    nbuf = 0;
                 }
+                if (match) {
+                    int rank = __popcll(mmask & ((1ull << lane) - 1ull));
+                    bufb[wid][nbuf + rank] = bidx;
+                    bufp[wid][nbuf + rank] = i0;
+                }
+                nbuf += cnt;
             }
             s = (s + 1) & mask;
+            walking = have;
+        }
+        if (__all(i0 >= np))
+            break;
+    }
+    if (nbuf) {
+        long long base = 0;
+        if (lane == 0)
+            base = (long long)atomicAdd((unsigned long long *)npairs,
+                                        (unsigned long long)nbuf);
+        base = __shfl(base, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE) {
+            int64_t pos = base + j;
+            if (pos < cap_pairs) {
+                out_b[pos] = bufb[wid][j];
+                out_p[pos] = bufp[wid][j];
+            }
         }
     }
 }
