@@ -15,6 +15,18 @@ def ex():
     return executor
 
 
+def drain(node):
+    node.BeginCustomScan()
+    rows = []
+    while True:
+        r = node.ExecCustomScan()
+        if r is None:
+            break
+        rows.append(r)
+    node.EndCustomScan()
+    return rows
+
+
 @pytest.fixture(scope="module")
 def ora():
     from oracle import oracle_py
@@ -35,7 +47,6 @@ def test_fuzz_agg(ex, ora, seed, n, kspace, nullp):
     kn = (rng.random(n) < nullp).astype(np.uint8)
     vn = (rng.random(n) < nullp).astype(np.uint8)
     dev = lambda a, dt: torch.as_tensor(a, dtype=dt, device="cuda")  # noqa
-    from tests.test_gpu_parity import drain  # reuse helper
     node = ex.GpuHashAgg(dev(keys, torch.int64), dev(vals, torch.float64),
                          dev(kn, torch.uint8), dev(vn, torch.uint8))
     got = drain(node)
@@ -64,7 +75,6 @@ def test_fuzz_join(ex, ora, seed, nb, np_, kspace):
     pk = rng.integers(0, kspace, np_)
     dev = lambda a: torch.as_tensor(np.ascontiguousarray(a),  # noqa
                                     dtype=torch.int64, device="cuda")
-    from tests.test_gpu_parity import drain
     node = ex.GpuHashJoin(dev(bk), dev(pk), cap_pairs=max(20 * np_, 64))
     pairs = drain(node)
     obi, opi = ora.join_i64(bk, pk)
