@@ -11,7 +11,7 @@ with the same code path.
 import torch
 import torch.distributed as dist
 
-from .executor import Q1_SLOT_ORDER, q1_finalize, q1_rows_from_state
+from .executor import q1_finalize, q1_rows_from_state
 
 
 def is_dist():
