@@ -801,13 +801,233 @@ __global__ void k_agg_compact(const agg_slot *tab, int64_t cap,
     }
 }
 
+template <typename T>
+__global__ void k_gather(const T *__restrict__ src,
+                         const int64_t *__restrict__ perm, int64_t n,
+                         T *__restrict__ dst); /* defined in the exchange section */
+
+/* ---- partitioned aggregation (mid-cardinality; round-2 item 1) ----
+ * When the distinct-key estimate exceeds what one block's LDS table can
+ * absorb but is far below n, hash-partition the rows into P bucket segments
+ * (each LDS-table-resident) and aggregate one bucket per block — turning
+ * per-row random global atomics into sequential traffic + per-bucket LDS
+ * work. Bucket = high hash bits (decorrelated from the table-slot bits). */
+#define AGGP_MAX_BUCKETS 2048
+
+__device__ __forceinline__ uint32_t d_agg_bucket(int64_t k, uint32_t nb)
+{
+    return (uint32_t)((d_hash_i64(k) >> 40) & (uint64_t)(nb - 1));
+}
+
+/* linear-counting distinct estimate over a strided sample (one block) */
+__global__ void k_agg_sample_distinct(const int64_t *__restrict__ keys,
+                                      const uint8_t *__restrict__ knull,
+                                      int64_t n, int64_t *est_out)
+{
+    const int MBITS = 1 << 16; /* 8 KB bitset */
+    __shared__ unsigned int bits[MBITS / 32];
+    for (int i = threadIdx.x; i < MBITS / 32; i += blockDim.x) bits[i] = 0;
+    __syncthreads();
+    const int SAMPLE = 1 << 16;
+    int64_t stride = n > SAMPLE ? n / SAMPLE : 1;
+    for (int64_t j = threadIdx.x; j < SAMPLE; j += blockDim.x) {
+        int64_t i = j * stride;
+        if (i >= n) break;
+        if (knull && knull[i]) continue;
+        uint32_t h = (uint32_t)(d_hash_i64(keys[i]) >> 16) & (MBITS - 1);
+        atomicOr(&bits[h >> 5], 1u << (h & 31));
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int64_t set = 0;
+        for (int i = 0; i < MBITS / 32; i++) set += __popc(bits[i]);
+        double m = MBITS, x = (double)set;
+        double sampled = n > SAMPLE ? SAMPLE : n;
+        double d_sample =
+            x >= m ? m : -m * log(1.0 - x / m); /* linear counting */
+        /* scale the sample estimate to the population (capped at n);
+         * downstream only needs an order of magnitude */
+        double scale = (double)n / (sampled > 0 ? sampled : 1);
+        double est = d_sample * (scale > 1.0 ? scale : 1.0);
+        *est_out = (int64_t)(est < (double)n ? est : (double)n);
+    }
+}
+
+__global__ void k_aggp_count(const int64_t *__restrict__ keys,
+                             const uint8_t *__restrict__ knull, int64_t n,
+                             uint32_t nb, unsigned long long *counts)
+{
+    __shared__ unsigned int bc[AGGP_MAX_BUCKETS];
+    for (int i = threadIdx.x; i < (int)nb; i += blockDim.x) bc[i] = 0;
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        if (knull && knull[i]) continue;
+        atomicAdd(&bc[d_agg_bucket(keys[i], nb)], 1u);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < (int)nb; i += blockDim.x)
+        if (bc[i]) atomicAdd(&counts[i], (unsigned long long)bc[i]);
+}
+
+__global__ void k_aggp_scatter(const int64_t *__restrict__ keys,
+                               const uint8_t *__restrict__ knull, int64_t n,
+                               uint32_t nb, unsigned long long *cursor,
+                               int64_t *__restrict__ perm)
+{
+    __shared__ unsigned int lcur[AGGP_MAX_BUCKETS];
+    __shared__ long long base[AGGP_MAX_BUCKETS];
+    int64_t per_block = (n + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < n ? lo + per_block : n;
+    for (int i = threadIdx.x; i < (int)nb; i += blockDim.x) lcur[i] = 0;
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        if (knull && knull[i]) continue;
+        atomicAdd(&lcur[d_agg_bucket(keys[i], nb)], 1u);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < (int)nb; i += blockDim.x) {
+        base[i] = lcur[i] ? (long long)atomicAdd(
+                                &cursor[i], (unsigned long long)lcur[i])
+                          : 0;
+        lcur[i] = 0;
+    }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        if (knull && knull[i]) continue;
+        uint32_t b = d_agg_bucket(keys[i], nb);
+        unsigned int off = atomicAdd(&lcur[b], 1u);
+        perm[base[b] + off] = i;
+    }
+}
+
+/* accumulate NULL-key rows (and AGG_EMPTY-valued keys, which the partition
+ * passes route normally but the per-bucket table cannot hold) */
+__global__ void k_aggp_specials(const int64_t *__restrict__ keys,
+                                const uint8_t *__restrict__ knull,
+                                const double *__restrict__ vals,
+                                const uint8_t *__restrict__ vnull, int64_t n,
+                                agg_slot *nullgrp)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long cs0 = 0, cv0 = 0, cs1 = 0, cv1 = 0;
+    double s0 = 0, s1 = 0;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        bool kn = knull && knull[i];
+        bool sent = !kn && keys[i] == AGG_EMPTY;
+        if (!kn && !sent) continue;
+        bool vn = vnull && vnull[i];
+        double v = vn ? 0.0 : vals[i];
+        if (kn) {
+            cs0++;
+            if (!vn) { cv0++; s0 += v; }
+        } else {
+            cs1++;
+            if (!vn) { cv1++; s1 += v; }
+        }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        cs0 += __shfl_down(cs0, off, WAVE);
+        cv0 += __shfl_down(cv0, off, WAVE);
+        s0 += __shfl_down(s0, off, WAVE);
+        cs1 += __shfl_down(cs1, off, WAVE);
+        cv1 += __shfl_down(cv1, off, WAVE);
+        s1 += __shfl_down(s1, off, WAVE);
+    }
+    if ((threadIdx.x % WAVE) == 0) {
+        if (cs0) {
+            atomicAdd(&nullgrp[0].count_star, cs0);
+            if (cv0) {
+                atomicAdd(&nullgrp[0].count_v, cv0);
+                atomicAdd(&nullgrp[0].sum_v, s0);
+            }
+        }
+        if (cs1) {
+            atomicAdd(&nullgrp[1].count_star, cs1);
+            if (cv1) {
+                atomicAdd(&nullgrp[1].count_v, cv1);
+                atomicAdd(&nullgrp[1].sum_v, s1);
+            }
+        }
+    }
+}
+
+/* per-bucket aggregation: block b owns the partitioned segment of bucket b
+ * (offsets from the scanned counts); rows were gathered into partitioned
+ * order so the reads stream. LDS table + global spill (same update fn). */
+__global__ void k_aggp_bucket_agg(const int64_t *__restrict__ pkeys,
+                                  const double *__restrict__ pvals,
+                                  const uint8_t *__restrict__ pvnull,
+                                  const unsigned long long *__restrict__ offs,
+                                  const unsigned long long *__restrict__ cnts,
+                                  agg_slot *tab, int64_t cap)
+{
+    const int LSLOTS = 1024;
+    __shared__ agg_lds_slot ltab[LSLOTS];
+    for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
+        ltab[s].key = AGG_EMPTY;
+        ltab[s].cs = 0;
+        ltab[s].cv = 0;
+        ltab[s].sum = 0.0;
+    }
+    __syncthreads();
+    int64_t mask = cap - 1;
+    int64_t lo = (int64_t)offs[blockIdx.x];
+    int64_t hi = lo + (int64_t)cnts[blockIdx.x];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        int64_t k = pkeys[i];
+        bool vn = pvnull && pvnull[i];
+        double v = vn ? 0.0 : pvals[i];
+        if (k == AGG_EMPTY) /* handled by k_aggp_specials */
+            continue;
+        int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)(LSLOTS - 1));
+        bool placed = false;
+        for (int t = 0; t < 8; t++) {
+            long long old = atomicCAS((unsigned long long *)&ltab[s].key,
+                                      (unsigned long long)AGG_EMPTY,
+                                      (unsigned long long)k);
+            if (old == AGG_EMPTY || old == k) {
+                atomicAdd(&ltab[s].cs, 1u);
+                if (!vn) {
+                    atomicAdd(&ltab[s].cv, 1u);
+                    atomicAdd(&ltab[s].sum, v);
+                }
+                placed = true;
+                break;
+            }
+            s = (s + 1) & (LSLOTS - 1);
+        }
+        if (!placed)
+            d_agg_global_update(tab, mask, k, 1ull, vn ? 0ull : 1ull, v);
+    }
+    __syncthreads();
+    for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
+        if (ltab[s].key != AGG_EMPTY)
+            d_agg_global_update(tab, mask, ltab[s].key,
+                                (unsigned long long)ltab[s].cs,
+                                (unsigned long long)ltab[s].cv, ltab[s].sum);
+    }
+}
+
 extern "C" {
+
+#define AGGP_THRESHOLD (8ll << 20) /* rows: partitioned path considered */
 
 otbx_status otbx_agg_i64_workspace_bytes(int64_t n, size_t *bytes)
 {
     int64_t cap = next_pow2_host(n < 16 ? 16 : (int64_t)(n / 0.7) + 1);
-    *bytes = (size_t)(cap + 2) * sizeof(agg_slot); /* + NULL and
-                                                    * sentinel-key groups */
+    size_t b = (size_t)(cap + 2) * sizeof(agg_slot); /* + NULL and
+                                                      * sentinel-key groups */
+    if (n >= AGGP_THRESHOLD) {
+        /* partitioned-path buffers: perm + partitioned key/val/vnull copies
+         * + bucket counts/cursor/offsets */
+        b += (size_t)n * (8 + 8 + 8 + 1) + (size_t)AGGP_MAX_BUCKETS * 8 * 3 +
+             4096;
+    }
+    *bytes = b;
     return OTBX_OK;
 }
 
@@ -817,19 +1037,98 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
                          int64_t *ngroups_dev, void *stream)
 {
     int64_t cap = next_pow2_host(n < 16 ? 16 : (int64_t)(n / 0.7) + 1);
-    if (ws_bytes < (size_t)(cap + 2) * sizeof(agg_slot)) return OTBX_ERR_INVALID;
+    {
+        size_t need;
+        otbx_agg_i64_workspace_bytes(n, &need);
+        if (ws_bytes < need) return OTBX_ERR_INVALID;
+    }
     hipStream_t s = (hipStream_t)stream;
     agg_slot *tab = (agg_slot *)ws;
     agg_slot *nullgrp = tab + cap;
     hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap + 2, 256)), dim3(256), 0, s,
                        tab, cap + 2);
     HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, sizeof(int64_t), s));
-    if (n > 0) {
+    if (n == 0) {
+        HIP_CHECK(hipGetLastError());
+        return OTBX_OK;
+    }
+
+    bool partitioned = false;
+    if (n >= AGGP_THRESHOLD) {
+        /* estimate the distinct-key count; partition when the per-block LDS
+         * table cannot absorb the key domain but buckets would be
+         * LDS-resident */
+        static int64_t *d_est = nullptr;
+        static int64_t *h_est = nullptr;
+        if (!d_est) {
+            HIP_CHECK(hipMalloc(&d_est, 8));
+            HIP_CHECK(hipHostMalloc(&h_est, 8));
+        }
+        hipLaunchKernelGGL(k_agg_sample_distinct, dim3(1), dim3(256), 0, s,
+                           keys, knull, n, d_est);
+        HIP_CHECK(hipMemcpyAsync(h_est, d_est, 8, hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        int64_t est = *h_est;
+        if (est > 1536 && est <= (32ll << 20)) {
+            partitioned = true;
+            uint32_t nb = (uint32_t)next_pow2_host(est / 256 < 64 ? 64
+                                                                  : est / 256);
+            if (nb > AGGP_MAX_BUCKETS) nb = AGGP_MAX_BUCKETS;
+            char *p = (char *)(nullgrp + 2);
+            int64_t *perm = (int64_t *)p;
+            int64_t *pkeys = perm + n;
+            double *pvals = (double *)(pkeys + n);
+            uint8_t *pvnull = (uint8_t *)(pvals + n);
+            unsigned long long *cnts =
+                (unsigned long long *)(pvnull + ((n + 63) & ~63ll));
+            unsigned long long *cursor = cnts + AGGP_MAX_BUCKETS;
+            unsigned long long *offs = cursor + AGGP_MAX_BUCKETS;
+            HIP_CHECK(hipMemsetAsync(cnts, 0, (size_t)nb * 8, s));
+            hipLaunchKernelGGL(k_aggp_count, dim3(grid_for(n, 256)), dim3(256),
+                               0, s, keys, knull, n, nb, cnts);
+            static unsigned long long *h_cnts = nullptr;
+            if (!h_cnts)
+                HIP_CHECK(hipHostMalloc(&h_cnts, AGGP_MAX_BUCKETS * 8 * 2));
+            HIP_CHECK(hipMemcpyAsync(h_cnts, cnts, (size_t)nb * 8,
+                                     hipMemcpyDeviceToHost, s));
+            HIP_CHECK(hipStreamSynchronize(s));
+            unsigned long long *h_offs = h_cnts + AGGP_MAX_BUCKETS;
+            unsigned long long acc = 0;
+            for (uint32_t b = 0; b < nb; b++) {
+                h_offs[b] = acc;
+                acc += h_cnts[b];
+            }
+            HIP_CHECK(hipMemcpyAsync(offs, h_offs, (size_t)nb * 8,
+                                     hipMemcpyHostToDevice, s));
+            HIP_CHECK(hipMemcpyAsync(cursor, h_offs, (size_t)nb * 8,
+                                     hipMemcpyHostToDevice, s));
+            hipLaunchKernelGGL(k_aggp_scatter, dim3(grid_for(n, 256)),
+                               dim3(256), 0, s, keys, knull, n, nb, cursor,
+                               perm);
+            int64_t nr = (int64_t)acc; /* partitioned (non-NULL-key) rows */
+            if (nr > 0) {
+                hipLaunchKernelGGL(k_gather<int64_t>, dim3(grid_for(nr, 256)),
+                                   dim3(256), 0, s, keys, perm, nr, pkeys);
+                hipLaunchKernelGGL(k_gather<double>, dim3(grid_for(nr, 256)),
+                                   dim3(256), 0, s, vals, perm, nr, pvals);
+                if (vnull)
+                    hipLaunchKernelGGL(k_gather<uint8_t>,
+                                       dim3(grid_for(nr, 256)), dim3(256), 0,
+                                       s, vnull, perm, nr, pvnull);
+                hipLaunchKernelGGL(k_aggp_bucket_agg, dim3(nb), dim3(256), 0,
+                                   s, pkeys, pvals, vnull ? pvnull : nullptr,
+                                   offs, cnts, tab, cap);
+            }
+            hipLaunchKernelGGL(k_aggp_specials, dim3(grid_for(n, 256)),
+                               dim3(256), 0, s, keys, knull, vals, vnull, n,
+                               nullgrp);
+        }
+    }
+    if (!partitioned)
         hipLaunchKernelGGL(k_agg_build, dim3(grid_for(n, 256)), dim3(256), 0, s,
                            keys, knull, vals, vnull, n, tab, cap, nullgrp);
-        hipLaunchKernelGGL(k_agg_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
-                           s, tab, cap, nullgrp, out, ngroups_dev);
-    }
+    hipLaunchKernelGGL(k_agg_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
+                       s, tab, cap, nullgrp, out, ngroups_dev);
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }
