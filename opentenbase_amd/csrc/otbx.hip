@@ -842,13 +842,22 @@ __global__ void k_agg_sample_distinct(const int64_t *__restrict__ keys,
         int64_t set = 0;
         for (int i = 0; i < MBITS / 32; i++) set += __popc(bits[i]);
         double m = MBITS, x = (double)set;
-        double sampled = n > SAMPLE ? SAMPLE : n;
+        double s_n = n > SAMPLE ? SAMPLE : n;
         double d_sample =
             x >= m ? m : -m * log(1.0 - x / m); /* linear counting */
-        /* scale the sample estimate to the population (capped at n);
-         * downstream only needs an order of magnitude */
-        double scale = (double)n / (sampled > 0 ? sampled : 1);
-        double est = d_sample * (scale > 1.0 ? scale : 1.0);
+        /* population-distinct estimate (order of magnitude is enough):
+         * - few distinct in the sample → the sample saturated the key
+         *   domain: D ≈ d_sample (hot keys do NOT scale with n)
+         * - mostly-distinct sample → birthday estimator from the collision
+         *   count: D ≈ s² / (2·(s − d)) */
+        double est;
+        if (d_sample < s_n / 2) {
+            est = d_sample;
+        } else {
+            double c = s_n - d_sample;
+            if (c < 1.0) c = 1.0;
+            est = s_n * s_n / (2.0 * c);
+        }
         *est_out = (int64_t)(est < (double)n ? est : (double)n);
     }
 }
