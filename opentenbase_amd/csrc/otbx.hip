@@ -912,7 +912,93 @@ __global__ void k_aggp_scatter(const int64_t *__restrict__ keys,
     }
 }
 
-/* accumulate NULL-key rows (and AGG_EMPTY-valued keys, which the partition
+/* direct (key,val) record scatter — the no-NULL fast path: one pass, no
+ * index gathers (an index permutation would make every downstream gather a
+ * random 64-B line read — measured 2x slower end to end) */
+__global__ void k_aggp_scatter_kv(const int64_t *__restrict__ keys,
+                                  const double *__restrict__ vals, int64_t n,
+                                  uint32_t nb, unsigned long long *cursor,
+                                  ulonglong2 *__restrict__ recs)
+{
+    __shared__ unsigned int lcur[AGGP_MAX_BUCKETS];
+    __shared__ long long base[AGGP_MAX_BUCKETS];
+    int64_t per_block = (n + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < n ? lo + per_block : n;
+    for (int i = threadIdx.x; i < (int)nb; i += blockDim.x) lcur[i] = 0;
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        atomicAdd(&lcur[d_agg_bucket(keys[i], nb)], 1u);
+    __syncthreads();
+    for (int i = threadIdx.x; i < (int)nb; i += blockDim.x) {
+        base[i] = lcur[i] ? (long long)atomicAdd(
+                                &cursor[i], (unsigned long long)lcur[i])
+                          : 0;
+        lcur[i] = 0;
+    }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        int64_t k = keys[i];
+        uint32_t b = d_agg_bucket(k, nb);
+        unsigned int off = atomicAdd(&lcur[b], 1u);
+        ulonglong2 r;
+        r.x = (unsigned long long)k;
+        r.y = (unsigned long long)__double_as_longlong(vals[i]);
+        recs[base[b] + off] = r;
+    }
+}
+
+__global__ void k_aggp_bucket_agg_kv(const ulonglong2 *__restrict__ recs,
+                                     const unsigned long long *__restrict__ offs,
+                                     const unsigned long long *__restrict__ cnts,
+                                     agg_slot *tab, int64_t cap)
+{
+    const int LSLOTS = 1024;
+    __shared__ agg_lds_slot ltab[LSLOTS];
+    for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
+        ltab[s].key = AGG_EMPTY;
+        ltab[s].cs = 0;
+        ltab[s].cv = 0;
+        ltab[s].sum = 0.0;
+    }
+    __syncthreads();
+    int64_t mask = cap - 1;
+    int64_t lo = (int64_t)offs[blockIdx.x];
+    int64_t hi = lo + (int64_t)cnts[blockIdx.x];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        ulonglong2 r = recs[i];
+        int64_t k = (int64_t)r.x;
+        double v = __longlong_as_double((long long)r.y);
+        if (k == AGG_EMPTY)
+            continue; /* handled by k_aggp_specials */
+        int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)(LSLOTS - 1));
+        bool placed = false;
+        for (int t = 0; t < 8; t++) {
+            long long old = atomicCAS((unsigned long long *)&ltab[s].key,
+                                      (unsigned long long)AGG_EMPTY,
+                                      (unsigned long long)k);
+            if (old == AGG_EMPTY || old == k) {
+                atomicAdd(&ltab[s].cs, 1u);
+                atomicAdd(&ltab[s].cv, 1u);
+                atomicAdd(&ltab[s].sum, v);
+                placed = true;
+                break;
+            }
+            s = (s + 1) & (LSLOTS - 1);
+        }
+        if (!placed)
+            d_agg_global_update(tab, mask, k, 1ull, 1ull, v);
+    }
+    __syncthreads();
+    for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
+        if (ltab[s].key != AGG_EMPTY)
+            d_agg_global_update(tab, mask, ltab[s].key,
+                                (unsigned long long)ltab[s].cs,
+                                (unsigned long long)ltab[s].cv, ltab[s].sum);
+    }
+}
+
+/* accumulate NULL-key rows/* accumulate NULL-key rows (and AGG_EMPTY-valued keys, which the partition
  * passes route normally but the per-bucket table cannot hold) */
 __global__ void k_aggp_specials(const int64_t *__restrict__ keys,
                                 const uint8_t *__restrict__ knull,
@@ -1111,22 +1197,38 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
                                      hipMemcpyHostToDevice, s));
             HIP_CHECK(hipMemcpyAsync(cursor, h_offs, (size_t)nb * 8,
                                      hipMemcpyHostToDevice, s));
-            hipLaunchKernelGGL(k_aggp_scatter, dim3(grid_for(n, 256)),
-                               dim3(256), 0, s, keys, knull, n, nb, cursor,
-                               perm);
             int64_t nr = (int64_t)acc; /* partitioned (non-NULL-key) rows */
-            if (nr > 0) {
-                hipLaunchKernelGGL(k_gather<int64_t>, dim3(grid_for(nr, 256)),
-                                   dim3(256), 0, s, keys, perm, nr, pkeys);
-                hipLaunchKernelGGL(k_gather<double>, dim3(grid_for(nr, 256)),
-                                   dim3(256), 0, s, vals, perm, nr, pvals);
-                if (vnull)
-                    hipLaunchKernelGGL(k_gather<uint8_t>,
+            if (!knull && !vnull) {
+                /* fast path: scatter (key,val) records directly — no index
+                 * gathers */
+                ulonglong2 *recs = (ulonglong2 *)perm; /* n×16 ≤ region */
+                hipLaunchKernelGGL(k_aggp_scatter_kv, dim3(grid_for(n, 256)),
+                                   dim3(256), 0, s, keys, vals, n, nb, cursor,
+                                   recs);
+                if (nr > 0)
+                    hipLaunchKernelGGL(k_aggp_bucket_agg_kv, dim3(nb),
+                                       dim3(256), 0, s, recs, offs, cnts, tab,
+                                       cap);
+            } else {
+                hipLaunchKernelGGL(k_aggp_scatter, dim3(grid_for(n, 256)),
+                                   dim3(256), 0, s, keys, knull, n, nb, cursor,
+                                   perm);
+                if (nr > 0) {
+                    hipLaunchKernelGGL(k_gather<int64_t>,
                                        dim3(grid_for(nr, 256)), dim3(256), 0,
-                                       s, vnull, perm, nr, pvnull);
-                hipLaunchKernelGGL(k_aggp_bucket_agg, dim3(nb), dim3(256), 0,
-                                   s, pkeys, pvals, vnull ? pvnull : nullptr,
-                                   offs, cnts, tab, cap);
+                                       s, keys, perm, nr, pkeys);
+                    hipLaunchKernelGGL(k_gather<double>,
+                                       dim3(grid_for(nr, 256)), dim3(256), 0,
+                                       s, vals, perm, nr, pvals);
+                    if (vnull)
+                        hipLaunchKernelGGL(k_gather<uint8_t>,
+                                           dim3(grid_for(nr, 256)), dim3(256),
+                                           0, s, vnull, perm, nr, pvnull);
+                    hipLaunchKernelGGL(k_aggp_bucket_agg, dim3(nb), dim3(256),
+                                       0, s, pkeys, pvals,
+                                       vnull ? pvnull : nullptr, offs, cnts,
+                                       tab, cap);
+                }
             }
             hipLaunchKernelGGL(k_aggp_specials, dim3(grid_for(n, 256)),
                                dim3(256), 0, s, keys, knull, vals, vnull, n,
