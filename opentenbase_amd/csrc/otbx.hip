@@ -998,7 +998,7 @@ __global__ void k_aggp_bucket_agg_kv(const ulonglong2 *__restrict__ recs,
     }
 }
 
-/* accumulate NULL-key rows/* accumulate NULL-key rows (and AGG_EMPTY-valued keys, which the partition
+/* accumulate NULL-key rows (and AGG_EMPTY-valued keys, which the partition
  * passes route normally but the per-bucket table cannot hold) */
 __global__ void k_aggp_specials(const int64_t *__restrict__ keys,
                                 const uint8_t *__restrict__ knull,
