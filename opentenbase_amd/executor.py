@@ -67,6 +67,26 @@ class GpuLineitem:
              _stream())
         return t
 
+    @classmethod
+    def from_host(cls, cols, with_orderkey=True):
+        """The real otbx_stage_table flow: host columnar
+        arrays (the provider's heap→SoA staging output) → device column
+        cache via otbx_memcpy_h2d. cols: dict of numpy arrays keyed like
+        COLS. Returns the staged table; PCIe-inclusive, one-time."""
+        import numpy as np
+        n = len(cols["l_shipdate"])
+        t = cls(n, with_orderkey=with_orderkey)
+        for name, dt in cls.COLS:
+            if t.t[name] is None:
+                continue
+            src = np.ascontiguousarray(cols[name])
+            assert src.nbytes == t.t[name].numel() * t.t[name].element_size()
+            call("otbx_memcpy_h2d", C.c_void_p(t.t[name].data_ptr()),
+                 src.ctypes.data_as(C.c_void_p), C.c_size_t(src.nbytes),
+                 _stream())
+        call("otbx_stream_sync", _stream())  # host buffers may be freed
+        return t
+
     def bytes_staged(self):
         return sum(v.numel() * v.element_size()
                    for v in self.t.values() if v is not None)
