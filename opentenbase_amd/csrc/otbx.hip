@@ -2206,6 +2206,75 @@ __global__ void k_q3_scan_filter_direct(const otbx_lineitem_dev l,
     }
 }
 
+/* fully fused direct-path probe: scan + bitmap filter + payload read +
+ * revenue accumulation in ONE pass. The candidate-list split exists for the
+ * HASH path's divergent chain walk; on the direct path the "probe" is a
+ * single clustered table read + one atomic, so materializing candidates
+ * (write + re-read + key re-gather) only costs bandwidth. */
+__global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
+                                           int32_t q3date, int64_t mino,
+                                           int64_t range,
+                                           const unsigned long long *__restrict__ bitmap,
+                                           const unsigned long long *__restrict__ ptab,
+                                           double *__restrict__ rtab,
+                                           unsigned long long *__restrict__ nhits)
+{
+    int64_t nq = l.n / 4;
+    const int4 *sd4 = (const int4 *)l.l_shipdate;
+    const longlong2 *ok2 = (const longlong2 *)l.l_orderkey;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned long long myhits = 0;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
+         q += stride) {
+        int4 d = sd4[q];
+        longlong2 ka = ok2[2 * q], kb = ok2[2 * q + 1];
+        int32_t ds[4] = {d.x, d.y, d.z, d.w};
+        int64_t ky[4] = {ka.x, ka.y, kb.x, kb.y};
+        unsigned long long pl[4];
+        bool m[4];
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+            int64_t idx = ky[j] - mino;
+            bool pass = (ds[j] > q3date) && idx >= 0 && idx < range;
+            int64_t bidx = pass ? idx : 0;
+            unsigned long long w = bitmap[bidx >> 6];
+            m[j] = pass && ((w >> (bidx & 63)) & 1ull);
+            /* payload read only where the bitmap passed (exact filter):
+             * clustered keys keep these on few lines */
+            pl[j] = m[j] ? ptab[idx] : 0ull;
+        }
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+            if (!m[j] || pl[j] == 0ull)
+                continue;
+            myhits++;
+            int64_t i = q * 4 + j;
+            double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
+            atomicAdd(&rtab[ky[j] - mino], rev);
+        }
+    }
+    /* tail rows */
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        for (int64_t i = nq * 4; i < l.n; i++) {
+            int64_t idx = l.l_orderkey[i] - mino;
+            if (!(l.l_shipdate[i] > q3date) || idx < 0 || idx >= range)
+                continue;
+            if (!((bitmap[idx >> 6] >> (idx & 63)) & 1ull))
+                continue;
+            unsigned long long pv = ptab[idx];
+            if (pv == 0ull)
+                continue;
+            myhits++;
+            double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
+            atomicAdd(&rtab[idx], rev);
+        }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        myhits += __shfl_down(myhits, off, WAVE);
+    if ((threadIdx.x % WAVE) == 0 && myhits)
+        atomicAdd(nhits, myhits);
+}
+
 __global__ void k_q3_probe_agg_direct(const otbx_lineitem_dev l,
                                       const int64_t *__restrict__ cand,
                                       const int64_t *__restrict__ ncand_p,
@@ -2852,13 +2921,10 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                 HIP_CHECK(hipEventRecord(ev[2], s));
                 rec2 = true;
             }
-            hipLaunchKernelGGL(k_q3_scan_filter_direct,
-                               dim3(grid_for(l->n, 256)), dim3(256), 0, s, *l,
-                               q3date, pmin, prange, dbitmap, cand_li,
-                               &hdr[2]);
-            hipLaunchKernelGGL(k_q3_probe_agg_direct,
-                               dim3(grid_for(l->n, 256)), dim3(256), 0, s, *l,
-                               cand_li, &hdr[2], pmin, dptab, drtab, nhits);
+            hipLaunchKernelGGL(k_q3_scan_probe_agg_direct,
+                               dim3(grid_for(l->n / 4, 256)), dim3(256), 0, s,
+                               *l, q3date, pmin, prange, dbitmap, dptab,
+                               drtab, nhits);
             if (kernel_ms && !rec3 && pass == npasses - 1) {
                 HIP_CHECK(hipEventRecord(ev[3], s));
                 rec3 = true;
