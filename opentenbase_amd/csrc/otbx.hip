@@ -2108,105 +2108,7 @@ __global__ void k_ord_insert_direct(const otbx_orders_dev o,
         int64_t i = cand[ci];
         int64_t idx = o.o_orderkey[i] - mino;
         if (idx < 0 || idx >= range)
-            continue; /* outside this grace pass's key sub-range */
-        atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
-        ptab[idx] = (unsigned long long)(uint32_t)o.o_orderdate[i] |
-                    ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
-    }
-}
-
-__global__ void k_q3_scan_filter_direct(const otbx_lineitem_dev l,
-                                        int32_t q3date, int64_t mino,
-                                        int64_t range,
-                                        const unsigned long long *__restrict__ bitmap,
-                                        int64_t *__restrict__ cand,
-                                        int64_t *ncand)
-{
-    /* 4 consecutive rows per lane via 16-B vector loads (the Q1 kernel's
-     * coalescing pattern: lanes stride 16 B); the bitmap filter is an
-     * un-hashed bit so a wave's 1024 rows touch ~4 bitmap words (L1).
-     * Candidate append is ORDER-PRESERVING (wave prefix sum over per-lane
-     * counts) with per-wave LDS staging, one global atomic per 1024. */
-    const int BUF = 1024;
-    __shared__ int64_t buf[256 / WAVE][BUF];
-    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
-    int nbuf = 0; /* wave-uniform */
-    int64_t nq = l.n / 4;
-    const int4 *sd4 = (const int4 *)l.l_shipdate;
-    const longlong2 *ok2 = (const longlong2 *)l.l_orderkey;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;; q += stride) {
-        bool m[4];
-        int64_t r0 = q * 4;
-        int mycnt = 0;
-        if (q < nq) {
-            int4 d = sd4[q];
-            longlong2 ka = ok2[2 * q], kb = ok2[2 * q + 1];
-            int32_t ds[4] = {d.x, d.y, d.z, d.w};
-            int64_t ky[4] = {ka.x, ka.y, kb.x, kb.y};
-#pragma unroll
-            for (int j = 0; j < 4; j++) {
-                int64_t idx = ky[j] - mino;
-                bool pass = (ds[j] > q3date) && idx >= 0 && idx < range;
-                unsigned long long w = bitmap[pass ? (idx >> 6) : 0];
-                m[j] = pass && ((w >> (idx & 63)) & 1ull);
-                mycnt += m[j];
-            }
-        } else {
-#pragma unroll
-            for (int j = 0; j < 4; j++) m[j] = false;
-        }
-        /* tail rows (l.n % 4) handled by the lane owning q == nq */
-        if (q == nq) {
-            for (int64_t i = nq * 4; i < l.n; i++) {
-                int j = (int)(i - nq * 4);
-                int64_t idx = l.l_orderkey[i] - mino;
-                bool pass = (l.l_shipdate[i] > q3date) && idx >= 0 && idx < range;
-                m[j] = pass && ((bitmap[pass ? (idx >> 6) : 0] >> (idx & 63)) & 1ull);
-                mycnt += m[j];
-            }
-        }
-        /* wave-exclusive prefix for order-preserving append */
-        int incl = mycnt;
-        for (int off = 1; off < WAVE; off <<= 1) {
-            int up = __shfl_up(incl, off, WAVE);
-            if (lane >= off) incl += up;
-        }
-        int tot = __shfl(incl, WAVE - 1, WAVE);
-        if (tot) {
-            if (nbuf + tot > BUF) {
-                long long bpos = 0;
-                if (lane == 0)
-                    bpos = (long long)atomicAdd((unsigned long long *)ncand,
-                                                (unsigned long long)nbuf);
-                bpos = __shfl(bpos, 0, WAVE);
-                for (int j = lane; j < nbuf; j += WAVE)
-                    cand[bpos + j] = buf[wid][j];
-                nbuf = 0;
-            }
-            int pos = nbuf + incl - mycnt;
-#pragma unroll
-            for (int j = 0; j < 4; j++)
-                if (m[j])
-                    buf[wid][pos++] = r0 + j;
-            nbuf += tot;
-        }
-        /* uniform loop exit AFTER the tail/flush work */
-        if (__all(q >= nq))
-            break;
-    }
-    if (nbuf) {
-        long long bpos = 0;
-        if (lane == 0)
-            bpos = (long long)atomicAdd((unsigned long long *)ncand,
-                                        (unsigned long long)nbuf);
-        bpos = __shfl(bpos, 0, WAVE);
-        for (int j = lane; j < nbuf; j += WAVE)
-            cand[bpos + j] = buf[wid][j];
-    }
-}
-
-/* fully fused direct-path probe: scan + bitmap filter + payload read +
+            continue; /* fully fused direct-path probe: scan + bitmap filter + payload read +
  * revenue accumulation in ONE pass. The candidate-list split exists for the
  * HASH path's divergent chain walk; on the direct path the "probe" is a
  * single clustered table read + one atomic, so materializing candidates
@@ -2268,36 +2170,6 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
             double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
             atomicAdd(&rtab[idx], rev);
         }
-    }
-    for (int off = WAVE / 2; off > 0; off >>= 1)
-        myhits += __shfl_down(myhits, off, WAVE);
-    if ((threadIdx.x % WAVE) == 0 && myhits)
-        atomicAdd(nhits, myhits);
-}
-
-__global__ void k_q3_probe_agg_direct(const otbx_lineitem_dev l,
-                                      const int64_t *__restrict__ cand,
-                                      const int64_t *__restrict__ ncand_p,
-                                      int64_t mino,
-                                      const unsigned long long *__restrict__ ptab,
-                                      double *__restrict__ rtab,
-                                      unsigned long long *__restrict__ nhits)
-{
-    int64_t n = *ncand_p;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    unsigned long long myhits = 0;
-    for (int64_t ci = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; ci < n;
-         ci += stride) {
-        int64_t i = cand[ci];
-        int64_t idx = l.l_orderkey[i] - mino;
-        unsigned long long pl = ptab[idx];
-        if (pl == 0ull)
-            continue; /* bitmap false positive cannot happen (exact), but an
-                       * unset payload can if a later rank pattern reuses the
-                       * bit range — keep the guard for safety */
-        myhits++;
-        double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
-        atomicAdd(&rtab[idx], rev);
     }
     for (int off = WAVE / 2; off > 0; off >>= 1)
         myhits += __shfl_down(myhits, off, WAVE);
