@@ -2108,7 +2108,14 @@ __global__ void k_ord_insert_direct(const otbx_orders_dev o,
         int64_t i = cand[ci];
         int64_t idx = o.o_orderkey[i] - mino;
         if (idx < 0 || idx >= range)
-            continue; /* fully fused direct-path probe: scan + bitmap filter + payload read +
+            continue; /* outside this grace pass's key sub-range */
+        atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
+        ptab[idx] = (unsigned long long)(uint32_t)o.o_orderdate[i] |
+                    ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
+    }
+}
+
+/* fully fused direct-path probe: scan + bitmap filter + payload read +
  * revenue accumulation in ONE pass. The candidate-list split exists for the
  * HASH path's divergent chain walk; on the direct path the "probe" is a
  * single clustered table read + one atomic, so materializing candidates
