@@ -142,3 +142,37 @@ def test_topk_edge_cases(ex):
          C.c_int64(10), C.c_void_p(cand.data_ptr()), C.c_int64(64),
          C.c_void_p(ncand.data_ptr()), C.c_void_p(hist.data_ptr()), stream)
     assert int(ncand.cpu().item()) == 0
+
+
+def test_q1_properties_at_full_size_sf100(ex):
+    """Size-independent properties at the BASELINE full size (SF100, 600 M
+    rows): count conservation between two independent kernels, the exact
+    4-group domain, and 8-shard merge == single-shard run (counts bit-exact,
+    sums within 1e-9)."""
+    n = 600_000_000
+    li = ex.GpuLineitem.generate(n, with_orderkey=False)
+    cnt = drain(ex.GpuSeqScanCount(li, cutoff=2436))[0][0]
+    node = ex.GpuQ1PartialAgg(li)
+    rows = drain(node)
+    assert sum(r["count_order"] for r in rows) == cnt
+    assert [r["l_returnflag"] + r["l_linestatus"] for r in rows] == \
+        ["AF", "NF", "NO", "RF"]
+    full_s, full_c = node.partial_state_tensors()
+    del li
+    torch.cuda.empty_cache()
+    tot_s = torch.zeros_like(full_s)
+    tot_c = torch.zeros_like(full_c)
+    for r in range(8):
+        shard = ex.GpuLineitem.generate(n, rank=r, nranks=8,
+                                        with_orderkey=False)
+        nd = ex.GpuQ1PartialAgg(shard)
+        drain(nd)
+        s, c = nd.partial_state_tensors()
+        tot_s += s
+        tot_c += c
+        del shard
+        torch.cuda.empty_cache()
+    assert torch.equal(tot_c, full_c)
+    rel = ((tot_s - full_s).abs() /
+           full_s.abs().clamp(min=1e-300)).max().item()
+    assert rel < 1e-9
