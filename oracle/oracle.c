@@ -55,8 +55,10 @@ ora_status ora_gen_lineitem(ora_lineitem *t, uint64_t seed, int64_t n_global,
     t->l_returnflag = malloc(n);
     t->l_linestatus = malloc(n);
     t->l_shipdate = malloc(n * 4);
+    t->l_partkey = malloc(n * 8);
     if (!t->l_orderkey || !t->l_quantity || !t->l_extendedprice || !t->l_discount ||
-        !t->l_tax || !t->l_returnflag || !t->l_linestatus || !t->l_shipdate)
+        !t->l_tax || !t->l_returnflag || !t->l_linestatus || !t->l_shipdate ||
+        !t->l_partkey)
         return ORA_ERR_OOM;
     for (int64_t l = 0; l < n; l++) {
         uint64_t i = otbx_li_global_row((uint64_t)l, rank, nranks);
@@ -68,6 +70,9 @@ ora_status ora_gen_lineitem(ora_lineitem *t, uint64_t seed, int64_t n_global,
         t->l_returnflag[l] = otbx_li_returnflag(seed, i);
         t->l_linestatus[l] = otbx_li_linestatus(seed, i);
         t->l_shipdate[l] = otbx_li_shipdate(seed, i);
+        /* nparts = SF×200k = n_global/30 (same formula as the GPU side) */
+        t->l_partkey[l] = otbx_li_partkey(
+            seed, i, n_global / 30 > 0 ? n_global / 30 : 1);
     }
     return ORA_OK;
 }
@@ -117,7 +122,26 @@ void ora_free_lineitem(ora_lineitem *t)
 {
     free(t->l_orderkey); free(t->l_quantity); free(t->l_extendedprice);
     free(t->l_discount); free(t->l_tax); free(t->l_returnflag);
-    free(t->l_linestatus); free(t->l_shipdate);
+    free(t->l_linestatus); free(t->l_shipdate); free(t->l_partkey);
+    memset(t, 0, sizeof(*t));
+}
+
+ora_status ora_gen_part(ora_part *t, uint64_t seed, int64_t n_global)
+{
+    t->n = n_global;
+    t->p_partkey = malloc((n_global > 0 ? n_global : 1) * 8);
+    t->p_type = malloc(n_global > 0 ? n_global : 1);
+    if (!t->p_partkey || !t->p_type) return ORA_ERR_OOM;
+    for (int64_t i = 0; i < n_global; i++) {
+        t->p_partkey[i] = otbx_part_partkey((uint64_t)i);
+        t->p_type[i] = otbx_part_type(seed, (uint64_t)i);
+    }
+    return ORA_OK;
+}
+
+void ora_free_part(ora_part *t)
+{
+    free(t->p_partkey); free(t->p_type);
     memset(t, 0, sizeof(*t));
 }
 void ora_free_orders(ora_orders *t)
@@ -500,6 +524,74 @@ ora_status ora_agg_i64(const int64_t *keys, const uint8_t *key_null,
     return ORA_OK;
 fail:
     free(slots); free(used);
+    return st;
+}
+
+/* ================= Q9-mix partial aggregate ================= */
+
+ora_status ora_q9_partial(const ora_part *p, const ora_orders *o,
+                          const ora_lineitem *l, uint8_t typemod,
+                          uint8_t typeval, ora_q9_group out[8], int *ngroups)
+{
+    /* same executor structure as Q3 (two chained-bucket hash joins feeding
+     * a hash aggregate, nodeHash.c/nodeHashjoin.c/nodeAgg.c) but the group
+     * key is COMPUTED (year(o_orderdate)) — the Q9 plan shape. The tiny
+     * year domain makes the group table a direct-mapped array (the
+     * simplehash instance degenerates). */
+    ora_status st = ORA_OK;
+
+    /* build 1: filtered part key set */
+    int64_t *pkeys = malloc((p->n > 0 ? p->n : 1) * 8);
+    if (!pkeys) return ORA_ERR_OOM;
+    int64_t npf = 0;
+    for (int64_t i = 0; i < p->n; i++)
+        if (p->p_type[i] % typemod == typeval)
+            pkeys[npf++] = p->p_partkey[i];
+    ora_hashtab pht;
+    if ((st = ht_build(&pht, pkeys, NULL, npf))) goto done1;
+
+    /* build 2: orders keyed on o_orderkey (no filter in Q9) */
+    {
+        ora_hashtab oht;
+        if ((st = ht_build(&oht, o->o_orderkey, NULL, o->n))) {
+            ht_free(&oht);
+            goto done2;
+        }
+        ora_q9_group g[8];
+        memset(g, 0, sizeof(g));
+        for (int64_t i = 0; i < l->n; i++) {
+            if (ht_first_match(&pht, l->l_partkey[i]) < 0)
+                continue; /* ⋈ part (inner) */
+            int64_t m = ht_first_match(&oht, l->l_orderkey[i]);
+            if (m < 0)
+                continue; /* ⋈ orders */
+            int32_t y = otbx_year_of_day(o->o_orderdate[m]);
+            double rev = l->l_extendedprice[i] * (1.0 - l->l_discount[i]);
+            if (g[y].count_rows == 0)
+                g[y].revenue = rev;        /* strict sum init */
+            else if ((st = float8pl(g[y].revenue, rev, &g[y].revenue))) {
+                ht_free(&oht);
+                goto done2;
+            }
+            if ((st = int8inc(&g[y].count_rows))) {
+                ht_free(&oht);
+                goto done2;
+            }
+        }
+        ht_free(&oht);
+        int ng = 0;
+        for (int y = 0; y < 8; y++)
+            if (g[y].count_rows > 0) {
+                out[ng] = g[y];
+                out[ng].year = y;
+                ng++;
+            }
+        *ngroups = ng;
+    }
+done2:
+    ht_free(&pht);
+done1:
+    free(pkeys);
     return st;
 }
 

@@ -36,6 +36,7 @@ typedef struct {
     double  *l_quantity, *l_extendedprice, *l_discount, *l_tax;
     uint8_t *l_returnflag, *l_linestatus;
     int32_t *l_shipdate;
+    int64_t *l_partkey;   /* appended: Q9-mix FK into part */
 } ora_lineitem;
 
 typedef struct {
@@ -50,6 +51,12 @@ typedef struct {
     uint8_t *c_mktsegment;
 } ora_customer;
 
+typedef struct {
+    int64_t n;
+    int64_t *p_partkey;
+    uint8_t *p_type;
+} ora_part;
+
 /* generation (otbx_gen.h): n_global must be divisible by nranks */
 ora_status ora_gen_lineitem(ora_lineitem *t, uint64_t seed, int64_t n_global,
                             uint32_t rank, uint32_t nranks);
@@ -58,9 +65,12 @@ ora_status ora_gen_orders(ora_orders *t, uint64_t seed, int64_t n_global,
                           int skew /* config-5 hot-custkey skew */);
 ora_status ora_gen_customer(ora_customer *t, uint64_t seed, int64_t n_global,
                             uint32_t rank, uint32_t nranks);
+/* part is replicated (small dimension table): every rank generates all rows */
+ora_status ora_gen_part(ora_part *t, uint64_t seed, int64_t n_global);
 void ora_free_lineitem(ora_lineitem *t);
 void ora_free_orders(ora_orders *t);
 void ora_free_customer(ora_customer *t);
+void ora_free_part(ora_part *t);
 
 /* ---- config 2: SeqScan + qual + COUNT(*) ----
  * restates ExecScan/ExecQual per-tuple loop (execScan.c:140-363) +
@@ -107,6 +117,21 @@ ora_status ora_q3_partial(const ora_customer *c, const ora_orders *o,
 /* ORDER BY revenue DESC, o_orderdate ASC LIMIT k — sorts in place, returns
  * min(k, n) (the Coordinator merge-sort analog, execFragment.c:4035-4059). */
 int64_t ora_q3_topk(ora_q3_row *rows, int64_t n, int64_t k);
+
+/* ---- Q9-mix (BASELINE config 5): revenue by order year over parts of a
+ * filtered type class — lineitem ⋈ part (p_type % typemod == typeval) ⋈
+ * orders, GROUP BY year(o_orderdate). The Q9 shape: two joins below an
+ * aggregate over a COMPUTED key. ---- */
+
+typedef struct {
+    int32_t year;          /* 0 = 1992 … 6 = 1998 */
+    double revenue;        /* sum(l_extendedprice * (1 - l_discount)) */
+    int64_t count_rows;
+} ora_q9_group;
+
+ora_status ora_q9_partial(const ora_part *p, const ora_orders *o,
+                          const ora_lineitem *l, uint8_t typemod,
+                          uint8_t typeval, ora_q9_group out[8], int *ngroups);
 
 /* ---- generic entry points for NULL/edge-case parity tests ---- */
 

@@ -46,6 +46,11 @@
 #define OTBX_T_LINEITEM 1u
 #define OTBX_T_ORDERS   2u
 #define OTBX_T_CUSTOMER 3u
+#define OTBX_T_PART     4u
+
+#define OTBX_PART_PER_SF 200000ll
+#define OTBX_PART_NTYPES 150    /* p_type domain; Q9's p_name LIKE filter is
+                                 * restated as p_type % 17 == 0 (~6%) */
 
 /* column ids (per table) */
 #define OTBX_C_SHIPDATE  1u
@@ -54,6 +59,7 @@
 #define OTBX_C_DISCOUNT  4u
 #define OTBX_C_TAX       5u
 #define OTBX_C_RFLAG     6u
+#define OTBX_C_PARTKEY   7u
 #define OTBX_C_ODATE     1u
 #define OTBX_C_CUSTKEY   2u
 #define OTBX_C_MKTSEG    1u
@@ -129,6 +135,33 @@ OTBX_FN uint8_t otbx_li_returnflag(uint64_t seed, uint64_t i)
         return (uint8_t)'N';
     uint64_t r = otbx_rnd(seed, OTBX_T_LINEITEM, OTBX_C_RFLAG, i);
     return (r & 1) ? (uint8_t)'A' : (uint8_t)'R';
+}
+
+OTBX_FN int64_t otbx_li_partkey(uint64_t seed, uint64_t i, int64_t nparts)
+{
+    uint64_t r = otbx_rnd(seed, OTBX_T_LINEITEM, OTBX_C_PARTKEY, i);
+    return 1 + (int64_t)(r % (uint64_t)nparts);
+}
+
+/* ---- part: global row i (0-based) ---- */
+
+OTBX_FN int64_t otbx_part_partkey(uint64_t i) { return (int64_t)i + 1; }
+
+OTBX_FN uint8_t otbx_part_type(uint64_t seed, uint64_t i)
+{
+    uint64_t r = otbx_rnd(seed, OTBX_T_PART, 1u, i);
+    return (uint8_t)(r % OTBX_PART_NTYPES);
+}
+
+/* calendar year index of a day number (day 0 = 1992-01-01); boundaries use
+ * the real 1992..1998 lengths (leap 1992/1996) */
+OTBX_FN int32_t otbx_year_of_day(int32_t d)
+{
+    const int32_t b[8] = {0, 366, 731, 1096, 1461, 1827, 2192, 2558};
+    int32_t y = 0;
+    for (int i = 1; i < 8; i++)
+        if (d >= b[i]) y = i;
+    return y; /* 0 = 1992 … 6 = 1998 */
 }
 
 /* ---- orders: global row i (0-based); o_orderkey dense 1..norders ---- */
