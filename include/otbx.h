@@ -79,6 +79,7 @@ typedef struct {
     double  *l_quantity, *l_extendedprice, *l_discount, *l_tax;
     uint8_t *l_returnflag, *l_linestatus;
     int32_t *l_shipdate;
+    int64_t *l_partkey;       /* appended; may be NULL (Q9-mix only) */
 } otbx_lineitem_dev;
 
 typedef struct {
@@ -92,6 +93,12 @@ typedef struct {
     int64_t *c_custkey;
     uint8_t *c_mktsegment;
 } otbx_customer_dev;
+
+typedef struct {
+    int64_t n;
+    int64_t *p_partkey;
+    uint8_t *p_type;
+} otbx_part_dev;
 
 /* On-device synthetic generation (the dbgen analog; same counter-based
  * functions as the CPU oracle — oracle/otbx_gen.h — so tables are
@@ -107,6 +114,9 @@ otbx_status otbx_gen_orders_dev(const otbx_orders_dev *t, uint64_t seed,
 otbx_status otbx_gen_customer_dev(const otbx_customer_dev *t, uint64_t seed,
                                   int64_t n_global, uint32_t rank,
                                   uint32_t nranks, void *stream);
+/* part is a replicated dimension table: every rank holds all n_global rows */
+otbx_status otbx_gen_part_dev(const otbx_part_dev *t, uint64_t seed,
+                              int64_t n_global, void *stream);
 
 /* ---- config 2: SeqScan + qual + COUNT(*) (scan-bandwidth kernel) ----
  * count_dev: one int64 device slot (zeroed by the call). */
@@ -179,6 +189,21 @@ otbx_status otbx_topk_by_revenue(const otbx_q3_group *groups_dev, int64_t n,
 otbx_status otbx_filter_customer(const otbx_customer_dev *c, uint8_t segment,
                                  int64_t *keys_out_dev, int64_t *nkeys_dev,
                                  void *stream);
+
+/* ---- Q9-mix DN fragment (BASELINE config 5's second query shape) ----
+ * lineitem ⋈ part (p_type % typemod == typeval) ⋈ orders, partial aggregate
+ * GROUP BY year(o_orderdate) — two joins under an aggregate on a COMPUTED
+ * key. Dense outputs over the 7 order years (0 = 1992):
+ *   sums_dev: double[7] revenue, counts_dev: int64[7]; both zeroed by the
+ * call; the Coordinator merge all-gathers them like Q1's states.
+ * ws: otbx_q9_workspace_bytes(nparts, norders_local, nranks). */
+otbx_status otbx_q9_workspace_bytes(int64_t nparts, int64_t norders,
+                                    uint32_t nranks, size_t *bytes);
+otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
+                            const otbx_lineitem_dev *l, uint8_t typemod,
+                            uint8_t typeval, void *ws, size_t ws_bytes,
+                            double *sums_dev, int64_t *counts_dev,
+                            void *stream, float *kernel_ms);
 
 /* ---- composable operators (generic plan shapes + NULL-semantics parity) --
 

@@ -33,6 +33,7 @@ class LineitemDev(C.Structure):
         ("l_discount", C.c_void_p), ("l_tax", C.c_void_p),
         ("l_returnflag", C.c_void_p), ("l_linestatus", C.c_void_p),
         ("l_shipdate", C.c_void_p),
+        ("l_partkey", C.c_void_p),
     ]
 
 
@@ -48,6 +49,13 @@ class CustomerDev(C.Structure):
     _fields_ = [
         ("n", C.c_int64),
         ("c_custkey", C.c_void_p), ("c_mktsegment", C.c_void_p),
+    ]
+
+
+class PartDev(C.Structure):
+    _fields_ = [
+        ("n", C.c_int64),
+        ("p_partkey", C.c_void_p), ("p_type", C.c_void_p),
     ]
 
 
@@ -85,6 +93,7 @@ EXPORTED_SYMBOLS = [
     "otbx_device_malloc", "otbx_device_free", "otbx_memcpy_h2d",
     "otbx_memcpy_d2h", "otbx_stream_sync",
     "otbx_gen_lineitem_dev", "otbx_gen_orders_dev", "otbx_gen_customer_dev",
+    "otbx_gen_part_dev", "otbx_q9_workspace_bytes", "otbx_q9_partial",
     "otbx_scan_count", "otbx_q1_partial", "otbx_q1_partial_variant",
     "otbx_q3_workspace_bytes", "otbx_q3_partial", "otbx_filter_customer",
     "otbx_topk_by_revenue",

@@ -15,8 +15,8 @@ import ctypes as C
 
 import torch
 
-from ._lib import (CustomerDev, LineitemDev, OrdersDev, OtbxError, call, check,
-                   lib)
+from ._lib import (CustomerDev, LineitemDev, OrdersDev, OtbxError, PartDev,
+                   call, check, lib)
 
 Q1_SLOT_ORDER = [(b"A", b"F"), (b"A", b"O"), (b"N", b"F"),
                  (b"N", b"O"), (b"R", b"F"), (b"R", b"O")]
@@ -44,13 +44,15 @@ class GpuLineitem:
     COLS = [("l_orderkey", torch.int64), ("l_quantity", torch.float64),
             ("l_extendedprice", torch.float64), ("l_discount", torch.float64),
             ("l_tax", torch.float64), ("l_returnflag", torch.uint8),
-            ("l_linestatus", torch.uint8), ("l_shipdate", torch.int32)]
+            ("l_linestatus", torch.uint8), ("l_shipdate", torch.int32),
+            ("l_partkey", torch.int64)]
 
-    def __init__(self, n, with_orderkey=True, device="cuda"):
+    def __init__(self, n, with_orderkey=True, with_partkey=False,
+                 device="cuda"):
         self.n = n
         self.t = {}
         for name, dt in self.COLS:
-            if name == "l_orderkey" and not with_orderkey:
+            if (name == "l_orderkey" and not with_orderkey) or                (name == "l_partkey" and not with_partkey):
                 self.t[name] = None
                 continue
             self.t[name] = torch.empty(n, dtype=dt, device=device)
@@ -60,22 +62,23 @@ class GpuLineitem:
 
     @classmethod
     def generate(cls, n_global, rank=0, nranks=1, seed=SEED_DEFAULT,
-                 with_orderkey=True):
-        t = cls(n_global // nranks, with_orderkey=with_orderkey)
+                 with_orderkey=True, with_partkey=False):
+        t = cls(n_global // nranks, with_orderkey=with_orderkey,
+                with_partkey=with_partkey)
         call("otbx_gen_lineitem_dev", C.byref(t.cstruct), C.c_uint64(seed),
              C.c_int64(n_global), C.c_uint32(rank), C.c_uint32(nranks),
              _stream())
         return t
 
     @classmethod
-    def from_host(cls, cols, with_orderkey=True):
+    def from_host(cls, cols, with_orderkey=True, with_partkey=False):
         """The real otbx_stage_table flow: host columnar
         arrays (the provider's heap→SoA staging output) → device column
         cache via otbx_memcpy_h2d. cols: dict of numpy arrays keyed like
         COLS. Returns the staged table; PCIe-inclusive, one-time."""
         import numpy as np
         n = len(cols["l_shipdate"])
-        t = cls(n, with_orderkey=with_orderkey)
+        t = cls(n, with_orderkey=with_orderkey, with_partkey=with_partkey)
         for name, dt in cls.COLS:
             if t.t[name] is None:
                 continue
@@ -130,6 +133,27 @@ class GpuCustomer:
         call("otbx_gen_customer_dev", C.byref(t.cstruct), C.c_uint64(seed),
              C.c_int64(n_global), C.c_uint32(rank), C.c_uint32(nranks),
              _stream())
+        return t
+
+
+class GpuPart:
+    """Replicated dimension table (every rank holds all rows — the locator
+    'R' / replicated-relation case of the reference's shard layout)."""
+
+    def __init__(self, n, device="cuda"):
+        self.n = n
+        self.t = {
+            "p_partkey": torch.empty(n, dtype=torch.int64, device=device),
+            "p_type": torch.empty(n, dtype=torch.uint8, device=device),
+        }
+        self.cstruct = PartDev(
+            n=n, **{k: C.c_void_p(v.data_ptr()) for k, v in self.t.items()})
+
+    @classmethod
+    def generate(cls, n_global, seed=SEED_DEFAULT):
+        t = cls(n_global)
+        call("otbx_gen_part_dev", C.byref(t.cstruct), C.c_uint64(seed),
+             C.c_int64(n_global), _stream())
         return t
 
 
@@ -374,6 +398,65 @@ def order_groups(groups_dev_u8, n):
          C.c_int64(n), C.c_void_p(out.data_ptr()), C.c_void_p(ws.data_ptr()),
          C.c_size_t(ws_bytes.value), _stream())
     return out[: n * 24].cpu().numpy().view(np.dtype(GpuQ3Fragment.NP_DTYPE))
+
+
+class GpuQ9Fragment(CustomScanState):
+    """The DN fragment of the Q9-shaped mix query (BASELINE config 5):
+    lineitem ⋈ part[p_type % typemod == typeval] ⋈ orders, partial agg
+    keyed on year(o_orderdate) — two HashJoins under a Partial HashAgg
+    with a tiny dense group domain (7 years). Emits one partial state row
+    per year; the Coordinator merge is elementwise (fragment.py
+    merge_q9_partials), exactly the Q1 pattern."""
+
+    YEARS = list(range(1992, 1999))
+
+    def __init__(self, part, orders, lineitem, typemod=17, typeval=0,
+                 nranks=1):
+        super().__init__()
+        self.pt, self.od, self.li = part, orders, lineitem
+        self.typemod, self.typeval = typemod, typeval
+        self.nranks = nranks
+        self.kernel_ms = None
+        self.sums = None
+        self.counts = None
+
+    def _run(self):
+        L = lib()
+        ws_bytes = C.c_size_t(0)
+        check(L.otbx_q9_workspace_bytes(C.c_int64(self.pt.n),
+                                        C.c_int64(self.od.n),
+                                        C.c_uint32(self.nranks),
+                                        C.byref(ws_bytes)))
+        ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
+        self.sums = torch.empty(7, dtype=torch.float64, device="cuda")
+        self.counts = torch.empty(7, dtype=torch.int64, device="cuda")
+        ms = C.c_float(0.0)
+        call("otbx_q9_partial", C.byref(self.pt.cstruct),
+             C.byref(self.od.cstruct), C.byref(self.li.cstruct),
+             C.c_uint8(self.typemod), C.c_uint8(self.typeval),
+             C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+             C.c_void_p(self.sums.data_ptr()),
+             C.c_void_p(self.counts.data_ptr()), _stream(), C.byref(ms))
+        self.kernel_ms = ms.value
+        return q9_rows_from_state(self.sums, self.counts)
+
+    def partial_state_tensors(self):
+        return self.sums, self.counts
+
+    def explain(self):
+        if self.kernel_ms is None:
+            return None
+        return {"k_q9_fused (Scan+2×HashJoin probe+PartialAgg, fused)":
+                self.kernel_ms}
+
+
+def q9_rows_from_state(sums, counts):
+    """Dense [7] sums + [7] counts → partial rows ordered by year."""
+    s = sums.cpu().numpy() if hasattr(sums, "cpu") else sums
+    c = counts.cpu().numpy() if hasattr(counts, "cpu") else counts
+    return [{"o_year": 1992 + g, "sum_revenue": float(s[g]),
+             "count_rows": int(c[g])}
+            for g in range(7) if c[g] != 0]
 
 
 def partition_by_key(keys):

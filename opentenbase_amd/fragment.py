@@ -43,6 +43,19 @@ def merge_q1_partials(sums, counts):
     return q1_rows_from_state(tot_s, tot_c)
 
 
+def merge_q9_partials(sums, counts):
+    """Q9-mix shard merge: all-gather + elementwise combine of the dense
+    [7]-year partial states (same RemoteSubplan payload shape as Q1)."""
+    from .executor import q9_rows_from_state
+    if not is_dist() or dist.get_world_size() == 1:
+        return q9_rows_from_state(sums, counts)
+    if dist.get_backend() == "gloo" and sums.is_cuda:
+        sums, counts = sums.cpu(), counts.cpu()
+    dist.all_reduce(sums)      # float8pl over shards
+    dist.all_reduce(counts)    # int8pl
+    return q9_rows_from_state(sums, counts)
+
+
 def allgather_variable(t):
     """All-gather a 1-D tensor with per-rank variable length (the FN-page
     concatenation semantics). Returns the concatenation over ranks."""

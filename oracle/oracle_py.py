@@ -54,6 +54,7 @@ class _Lineitem(C.Structure):
         ("l_returnflag", C.POINTER(C.c_uint8)),
         ("l_linestatus", C.POINTER(C.c_uint8)),
         ("l_shipdate", C.POINTER(C.c_int32)),
+        ("l_partkey", C.POINTER(C.c_int64)),
     ]
 
 
@@ -72,6 +73,21 @@ class _Customer(C.Structure):
         ("n", C.c_int64),
         ("c_custkey", C.POINTER(C.c_int64)),
         ("c_mktsegment", C.POINTER(C.c_uint8)),
+    ]
+
+
+class _Part(C.Structure):
+    _fields_ = [
+        ("n", C.c_int64),
+        ("p_partkey", C.POINTER(C.c_int64)),
+        ("p_type", C.POINTER(C.c_uint8)),
+    ]
+
+
+class OraQ9Group(C.Structure):
+    _fields_ = [
+        ("year", C.c_int32), ("revenue", C.c_double),
+        ("count_rows", C.c_int64),
     ]
 
 
@@ -134,6 +150,7 @@ def gen_tables(n_lineitem, rank=0, nranks=1, seed=42, need=("lineitem",),
             "l_returnflag": np.ctypeslib.as_array(t.l_returnflag, (n,)),
             "l_linestatus": np.ctypeslib.as_array(t.l_linestatus, (n,)),
             "l_shipdate": np.ctypeslib.as_array(t.l_shipdate, (n,)),
+            "l_partkey": np.ctypeslib.as_array(t.l_partkey, (n,)),
         }
         keep.append(t)
     if "orders" in need:
@@ -149,6 +166,17 @@ def gen_tables(n_lineitem, rank=0, nranks=1, seed=42, need=("lineitem",),
             "o_custkey": np.ctypeslib.as_array(t.o_custkey, (n,)),
             "o_orderdate": np.ctypeslib.as_array(t.o_orderdate, (n,)),
             "o_shippriority": np.ctypeslib.as_array(t.o_shippriority, (n,)),
+        }
+    if "part" in need:
+        t = _Part()
+        st = L.ora_gen_part(C.byref(t), C.c_uint64(seed),
+                            C.c_int64(max(n_lineitem // 30, 1)))
+        assert st == 0, st
+        n = t.n
+        out["part"] = {
+            "struct": t,
+            "p_partkey": np.ctypeslib.as_array(t.p_partkey, (n,)),
+            "p_type": np.ctypeslib.as_array(t.p_type, (n,)),
         }
     if "customer" in need:
         t = _Customer()
@@ -191,6 +219,19 @@ def q1_finalize(groups):
     arr = (OraQ1Group * len(groups))(*groups)
     L.ora_q1_finalize(arr, len(groups))
     return list(arr)
+
+
+def q9_partial(tables, typemod=17, typeval=0):
+    L = lib()
+    g = (OraQ9Group * 8)()
+    ng = C.c_int(0)
+    st = L.ora_q9_partial(C.byref(tables["part"]["struct"]),
+                          C.byref(tables["orders"]["struct"]),
+                          C.byref(tables["lineitem"]["struct"]),
+                          C.c_uint8(typemod), C.c_uint8(typeval),
+                          g, C.byref(ng))
+    assert st == 0, st
+    return [g[i] for i in range(ng.value)]
 
 
 def q3_partial(tables, segment=0, date=1169):

@@ -218,6 +218,80 @@ def test_q3_broadcast_path_parity(ex, ora):
         assert approx(float(v), float(exp[k]))
 
 
+# ---------------- Q9-mix fragment (BASELINE config 5) ----------------
+
+def test_q9_datagen_bit_identical(ex, ora):
+    n = 120000
+    li = ex.GpuLineitem.generate(n, with_partkey=True)
+    pt = ex.GpuPart.generate(n // 30)
+    t = ora.gen_tables(n, need=("lineitem", "part"))
+    assert np.array_equal(li.t["l_partkey"].cpu().numpy(),
+                          t["lineitem"]["l_partkey"])
+    assert np.array_equal(pt.t["p_partkey"].cpu().numpy(),
+                          t["part"]["p_partkey"])
+    assert np.array_equal(pt.t["p_type"].cpu().numpy(), t["part"]["p_type"])
+
+
+@pytest.mark.parametrize("n", [400000, 400004, 399998])
+def test_q9_parity(ex, ora, n):
+    li = ex.GpuLineitem.generate(n, with_partkey=True)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    pt = ex.GpuPart.generate(max(n // 30, 1))
+    node = ex.GpuQ9Fragment(pt, od, li)
+    rows = drain(node)
+    t = ora.gen_tables(n, need=("lineitem", "orders", "part"))
+    exp = ora.q9_partial(t)
+    assert [r["o_year"] - 1992 for r in rows] == [g.year for g in exp]
+    for r, g in zip(rows, exp):
+        assert r["count_rows"] == g.count_rows          # bit-exact
+        assert approx(r["sum_revenue"], g.revenue)      # float8 ≤ 1e-6 rel
+    assert node.kernel_ms is not None and node.kernel_ms > 0
+
+
+def test_q9_typemod_variants(ex, ora):
+    """Different part-filter selectivities (the p_type % m == v predicate
+    family) agree with the oracle, including empty-result cases."""
+    n = 200000
+    li = ex.GpuLineitem.generate(n, with_partkey=True)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    pt = ex.GpuPart.generate(n // 30)
+    t = ora.gen_tables(n, need=("lineitem", "orders", "part"))
+    for m, v in [(2, 1), (150, 149), (151, 150), (1, 0)]:
+        rows = drain(ex.GpuQ9Fragment(pt, od, li, typemod=m, typeval=v))
+        exp = ora.q9_partial(t, typemod=m, typeval=v)
+        assert len(rows) == len(exp), (m, v)
+        for r, g in zip(rows, exp):
+            assert r["o_year"] - 1992 == g.year
+            assert r["count_rows"] == g.count_rows
+            assert approx(r["sum_revenue"], g.revenue)
+
+
+def test_q9_sharded_merge_parity(ex, ora):
+    """2-shard dense-state combine == full run (counts bit-exact, sums
+    ≤1e-9 relative): the RemoteSubplan merge payload for Q9."""
+    n = 400000
+    full = ex.GpuQ9Fragment(ex.GpuPart.generate(n // 30),
+                            ex.GpuOrders.generate(n // 4, n // 40),
+                            ex.GpuLineitem.generate(n, with_partkey=True))
+    drain(full)
+    fs, fc = full.partial_state_tensors()
+    tot_s = torch.zeros_like(fs)
+    tot_c = torch.zeros_like(fc)
+    for r in range(2):
+        node = ex.GpuQ9Fragment(
+            ex.GpuPart.generate(n // 30),
+            ex.GpuOrders.generate(n // 4, n // 40, rank=r, nranks=2),
+            ex.GpuLineitem.generate(n, rank=r, nranks=2, with_partkey=True),
+            nranks=2)
+        drain(node)
+        s, c = node.partial_state_tensors()
+        tot_s += s
+        tot_c += c
+    assert torch.equal(tot_c, fc)
+    rel = ((tot_s - fs).abs() / fs.abs().clamp(min=1e-300)).max().item()
+    assert rel < 1e-9
+
+
 # ---------------- composable operators: NULL semantics on GPU -------------
 
 def _agg(ex, keys, vals, kn=None, vn=None):

@@ -129,6 +129,54 @@ def test_q3_against_numpy_reference():
         assert got[k] == pytest.approx(byk[k], rel=1e-12)
 
 
+def test_q9_against_numpy_reference():
+    """Pin ora_q9_partial against an independent numpy restatement of
+    lineitem ⋈ part[p_type%17==0] ⋈ orders GROUP BY year(o_orderdate)."""
+    n = 120000
+    t = ora.gen_tables(n, need=("lineitem", "orders", "part"))
+    li, od, pt = t["lineitem"], t["orders"], t["part"]
+    # part filter: partkeys are dense 1..nparts
+    sel = pt["p_type"] % 17 == 0
+    part_ok = np.zeros(len(pt["p_partkey"]) + 1, dtype=bool)
+    part_ok[pt["p_partkey"][sel]] = True
+    # orderkey → orderdate (dense keys)
+    odate = np.zeros(int(od["o_orderkey"].max()) + 1, dtype=np.int32)
+    odate[od["o_orderkey"]] = od["o_orderdate"]
+    mask = part_ok[li["l_partkey"]]
+    d = odate[li["l_orderkey"][mask]]
+    rev = (li["l_extendedprice"][mask] * (1.0 - li["l_discount"][mask]))
+    bounds = np.array([0, 366, 731, 1096, 1461, 1827, 2192, 2558])
+    yr = np.searchsorted(bounds, d, side="right") - 1
+    got = ora.q9_partial(t)
+    exp = {}
+    for y in range(7):
+        m = yr == y
+        if m.sum():
+            exp[y] = (float(rev[m].sum()), int(m.sum()))
+    assert {g.year for g in got} == set(exp)
+    for g in got:
+        assert g.count_rows == exp[g.year][1]
+        assert abs(g.revenue - exp[g.year][0]) <= 1e-9 * abs(exp[g.year][0])
+
+
+def test_q9_sharded_union_equals_full():
+    """2-shard Q9 partial states combine (float8pl/int8pl) to the full run
+    (part replicated; lineitem+orders co-sharded by orderkey)."""
+    n = 120000
+    full = ora.q9_partial(ora.gen_tables(n, need=("lineitem", "orders", "part")))
+    merged = {}
+    for r in range(2):
+        t = ora.gen_tables(n, rank=r, nranks=2,
+                           need=("lineitem", "orders", "part"))
+        for g in ora.q9_partial(t):
+            rv, c = merged.get(g.year, (0.0, 0))
+            merged[g.year] = (rv + g.revenue, c + g.count_rows)
+    assert {g.year for g in full} == set(merged)
+    for g in full:
+        assert merged[g.year][1] == g.count_rows
+        assert abs(merged[g.year][0] - g.revenue) <= 1e-9 * abs(g.revenue)
+
+
 def test_q3_sharded_union_equals_full():
     """Shards are disjoint by orderkey: the union of per-shard Q3 groups must
     equal the full-table groups (orderkey groups never span ranks —
