@@ -89,7 +89,8 @@ def main():
     ap.add_argument("--sf", type=int, default=SF_DEFAULT,
                     help="scale factor PER GPU (weak scaling)")
     ap.add_argument("--workload", default="tpch_q1",
-                    choices=["tpch_q1", "scan_count", "tpch_q3"])
+                    choices=["tpch_q1", "scan_count", "tpch_q3",
+                             "tpch_q9mix"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--skew", action="store_true",
                     help="config-5 skewed distribution keys (hot custkeys)")
@@ -130,6 +131,16 @@ def main():
         li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world,
                                      with_orderkey=False)
         bytes_per_row = SCAN_BYTES_PER_ROW
+    elif args.workload == "tpch_q9mix":
+        # BASELINE config 5's second shape: lineitem ⋈ part ⋈ orders,
+        # GROUP BY year. Streamed probe-side bytes: partkey 8 + orderkey 8
+        # + extendedprice 8 + discount 8 = 32 B/row.
+        li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world,
+                                     with_partkey=True)
+        od = ex.GpuOrders.generate(n_global // 4, n_global // 40, rank=rank,
+                                   nranks=world, skew=args.skew)
+        pt = ex.GpuPart.generate(max(n_global // 30, 1))
+        bytes_per_row = 32
     else:  # tpch_q3
         li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world)
         od = ex.GpuOrders.generate(n_global // 4, n_global // 40, rank=rank,
@@ -161,6 +172,13 @@ def main():
                  C.c_int64(li.n), C.c_int32(2436), C.c_void_p(out.data_ptr()),
                  stream)
             return int(out.cpu().item())
+        elif args.workload == "tpch_q9mix":
+            node = ex.GpuQ9Fragment(pt, od, li, nranks=world)
+            node.BeginCustomScan()
+            node._rows = node._run()
+            kernel_ms_acc.append(node.kernel_ms)
+            s, c = node.partial_state_tensors()
+            return fragment.merge_q9_partials(s, c)
         else:
             bcast = None
             if world > 1:
@@ -236,10 +254,11 @@ def main():
 
     cpu = None
     if rank == 0 and world == 1 and not args.no_cpu_baseline \
-            and args.workload in ("tpch_q1", "tpch_q3"):
+            and args.workload in ("tpch_q1", "tpch_q3", "tpch_q9mix"):
         log("running cpu_baseline (oracle port, 1 core)...")
-        q = "q1" if args.workload == "tpch_q1" else "q3"
-        # q3's tuple-at-a-time executor is ~4x slower per row: smaller sample
+        q = {"tpch_q1": "q1", "tpch_q3": "q3",
+             "tpch_q9mix": "q9"}[args.workload]
+        # join executors are ~4x slower per row: smaller sample
         rows = args.cpu_sample_rows if q == "q1" else args.cpu_sample_rows // 4
         cpu = cpu_baseline(q, rows)
 

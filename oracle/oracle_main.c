@@ -6,6 +6,7 @@
  *   oracle_cli q1   --rows N [--rank R --nranks W] [--cutoff D]
  *   oracle_cli q3   --rows N [--rank R --nranks W] [--date D] [--segment S]
  *   oracle_cli scan --rows N [--rank R --nranks W] [--cutoff D]
+ *   oracle_cli q9   --rows N [--rank R --nranks W]
  * --rows is the GLOBAL lineitem row count (orders = rows/4, customer = rows/40).
  * Prints one JSON line: timing covers the executor only, not generation.
  */
@@ -95,6 +96,29 @@ int main(int argc, char **argv)
         free(out);
         ora_free_orders(&od);
         ora_free_customer(&cu);
+    } else if (!strcmp(query, "q9")) {
+        ora_orders od;
+        ora_part pt;
+        if (ora_gen_orders(&od, seed, rows / 4, rows / 40, rank, nranks, 0) ||
+            ora_gen_part(&pt, seed, rows / 30 > 0 ? rows / 30 : 1)) {
+            fprintf(stderr, "gen failed\n"); return 1;
+        }
+        ora_q9_group g[8];
+        int ng = 0;
+        double t0 = now_s();
+        if (ora_q9_partial(&pt, &od, &li, 17, 0, g, &ng)) {
+            fprintf(stderr, "q9 failed\n"); return 1;
+        }
+        double dt = now_s() - t0;
+        printf("{\"query\":\"q9\",\"rows\":%lld,\"seconds\":%.6f,\"groups\":[",
+               (long long)li.n, dt);
+        for (int i = 0; i < ng; i++)
+            printf("%s{\"year\":%d,\"revenue\":%.17g,\"count\":%lld}",
+                   i ? "," : "", 1992 + g[i].year, g[i].revenue,
+                   (long long)g[i].count_rows);
+        printf("]}\n");
+        ora_free_orders(&od);
+        ora_free_part(&pt);
     } else {
         fprintf(stderr, "unknown query %s\n", query);
         return 2;
