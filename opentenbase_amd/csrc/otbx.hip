@@ -1306,7 +1306,6 @@ __global__ void k_q9_odate_build(const otbx_orders_dev o, int64_t mino,
 /* fused probe: per lineitem row — part-bitmap semi-join, orders date lookup,
  * computed-year partial aggregate in per-lane registers (the Q1 pattern:
  * 7-year domain, compile-time indexed). */
-__launch_bounds__(256, 2)
 __global__ void k_q9_fused(const otbx_lineitem_dev l,
                            const unsigned long long *__restrict__ pbitmap,
                            int64_t nparts, const int32_t *__restrict__ dtab,
@@ -1324,31 +1323,36 @@ __global__ void k_q9_fused(const otbx_lineitem_dev l,
     int64_t nq = l.n / 2;
     const longlong2 *pk2 = (const longlong2 *)l.l_partkey;
     const longlong2 *ok2 = (const longlong2 *)l.l_orderkey;
-    const double2 *ep2 = (const double2 *)l.l_extendedprice;
-    const double2 *dc2 = (const double2 *)l.l_discount;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
          q += stride) {
         longlong2 pk = pk2[q];
         longlong2 ok = ok2[q];
-        double2 ep = ep2[q];
-        double2 dc = dc2[q];
         int64_t pks[2] = {pk.x, pk.y};
         int64_t oks[2] = {ok.x, ok.y};
-        double eps[2] = {ep.x, ep.y};
-        double dcs[2] = {dc.x, dc.y};
 #pragma unroll
         for (int j = 0; j < 2; j++) {
             int64_t pidx = pks[j] - 1;
             bool m = pidx >= 0 && pidx < nparts &&
                      ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
-            int64_t oidx = oks[j] - mino;
-            bool ob = m && oidx >= 0 && oidx < orange;
-            int32_t date = dtab[ob ? oidx : 0];
-            if (!ob || date == 0)
+            /* real branch, not a select: with ~6% of lanes surviving the
+             * part filter, exec-masking skips the dtab/payload requests of
+             * the other 94% (a `dtab[ob ? oidx : 0]` select forces a
+             * request from every lane — measured slower) */
+            if (!m)
                 continue;
+            int64_t oidx = oks[j] - mino;
+            if (oidx < 0 || oidx >= orange)
+                continue;
+            int32_t date = dtab[oidx];
+            if (date == 0)
+                continue;
+            /* payload columns only for surviving rows: the filter columns
+             * (16 B/row) are the only full streams; payload traffic is
+             * per-hit (SURVEY §8d style) */
             int32_t y = otbx_year_of_day(date);
-            double rev = eps[j] * (1.0 - dcs[j]);
+            int64_t i = 2 * q + j;
+            double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
 #pragma unroll
             for (int yy = 0; yy < 7; yy++) {
                 bool hit = yy == y;
@@ -1357,24 +1361,25 @@ __global__ void k_q9_fused(const otbx_lineitem_dev l,
             }
         }
     }
-    /* tail row (odd n) */
+    /* tail rows (n % 2) */
     if ((l.n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
-        int64_t i = l.n - 1;
-        int64_t pidx = l.l_partkey[i] - 1;
-        if (pidx >= 0 && pidx < nparts &&
-            ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull)) {
-            int64_t oidx = l.l_orderkey[i] - mino;
-            if (oidx >= 0 && oidx < orange) {
-                int32_t date = dtab[oidx];
-                if (date != 0) {
-                    int32_t y = otbx_year_of_day(date);
-                    double rev =
-                        l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
+        for (int64_t i = nq * 2; i < l.n; i++) {
+            int64_t pidx = l.l_partkey[i] - 1;
+            if (pidx >= 0 && pidx < nparts &&
+                ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull)) {
+                int64_t oidx = l.l_orderkey[i] - mino;
+                if (oidx >= 0 && oidx < orange) {
+                    int32_t date = dtab[oidx];
+                    if (date != 0) {
+                        int32_t y = otbx_year_of_day(date);
+                        double rev =
+                            l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
 #pragma unroll
-                    for (int yy = 0; yy < 7; yy++) {
-                        if (yy == y) {
-                            acc[yy] += rev;
-                            cnt[yy] += 1;
+                        for (int yy = 0; yy < 7; yy++) {
+                            if (yy == y) {
+                                acc[yy] += rev;
+                                cnt[yy] += 1;
+                            }
                         }
                     }
                 }
