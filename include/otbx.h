@@ -198,7 +198,8 @@ otbx_status otbx_filter_customer(const otbx_customer_dev *c, uint8_t segment,
  * call; the Coordinator merge all-gathers them like Q1's states.
  * ws: otbx_q9_workspace_bytes(nparts, norders_local, nranks). */
 otbx_status otbx_q9_workspace_bytes(int64_t nparts, int64_t norders,
-                                    uint32_t nranks, size_t *bytes);
+                                    int64_t nlineitem, uint32_t nranks,
+                                    size_t *bytes);
 otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
                             const otbx_lineitem_dev *l, uint8_t typemod,
                             uint8_t typeval, void *ws, size_t ws_bytes,

@@ -1303,14 +1303,102 @@ __global__ void k_q9_odate_build(const otbx_orders_dev o, int64_t mino,
         dtab[o.o_orderkey[i] - mino] = o.o_orderdate[i];
 }
 
-/* fused probe: per lineitem row — part-bitmap semi-join, orders date lookup,
- * computed-year partial aggregate in per-lane registers (the Q1 pattern:
- * 7-year domain, compile-time indexed). */
-__global__ void k_q9_fused(const otbx_lineitem_dev l,
-                           const unsigned long long *__restrict__ pbitmap,
-                           int64_t nparts, const int32_t *__restrict__ dtab,
-                           int64_t mino, int64_t orange,
-                           double *__restrict__ out_sums,   /* [7] */
+/* Q9 probe, split in two phases. The single fused kernel measured as the
+ * SUM of its component costs with no overlap (request-bound;
+ * profiles/r01_q9mix_microbench.txt), so splitting is free — and phase 1
+ * then streams only l_partkey (8 B/row), while phase 2 runs dense over the
+ * ~6% survivors.
+ *
+ * phase 1: part-bitmap semi-join over the l_partkey stream; wave-aggregated
+ * append (Guideline 12) of surviving row ids (u32 — otbx_q9_partial guards
+ * n < 2^32). */
+__global__ void k_q9_filter(const int64_t *__restrict__ pk, int64_t n,
+                            const unsigned long long *__restrict__ pbitmap,
+                            int64_t nparts, uint32_t *__restrict__ hits,
+                            int64_t *nhits)
+{
+    /* LDS-staged per-wave append, one global reservation per ~BUF rows —
+     * a bare wave_append here would make ~n/64 single-counter reservations,
+     * which caps at ~88/µs (the measured law behind every append kernel in
+     * this file; a first cut of this kernel measured 107 ms at SF100 for
+     * exactly that reason). */
+    const int BUF = 1024;
+    __shared__ uint32_t buf[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0;
+    int64_t nq = n / 4;
+    const longlong2 *pk2 = (const longlong2 *)pk;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         q += stride) {
+        bool m[4] = {false, false, false, false};
+        int64_t r0 = q * 4;
+        int mycnt = 0;
+        if (q < nq) {
+            longlong2 pa = pk2[2 * q], pb = pk2[2 * q + 1];
+            int64_t pks[4] = {pa.x, pa.y, pb.x, pb.y};
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int64_t pidx = pks[j] - 1;
+                m[j] = pidx >= 0 && pidx < nparts &&
+                       ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
+                mycnt += m[j];
+            }
+        } else if (q == nq) { /* tail rows (n % 4) */
+            for (int64_t i = nq * 4; i < n; i++) {
+                int j = (int)(i - nq * 4);
+                int64_t pidx = pk[i] - 1;
+                m[j] = pidx >= 0 && pidx < nparts &&
+                       ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
+                mycnt += m[j];
+            }
+        }
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        int tot = __shfl(incl, WAVE - 1, WAVE);
+        if (tot) {
+            if (nbuf + tot > BUF) {
+                long long bpos = 0;
+                if (lane == 0)
+                    bpos = (long long)atomicAdd((unsigned long long *)nhits,
+                                                (unsigned long long)nbuf);
+                bpos = __shfl(bpos, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE)
+                    hits[bpos + j] = buf[wid][j];
+                nbuf = 0;
+            }
+            int pos = nbuf + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j])
+                    buf[wid][pos++] = (uint32_t)(r0 + j);
+            nbuf += tot;
+        }
+        if (__all(q >= nq))
+            break;
+    }
+    if (nbuf) {
+        long long bpos = 0;
+        if (lane == 0)
+            bpos = (long long)atomicAdd((unsigned long long *)nhits,
+                                        (unsigned long long)nbuf);
+        bpos = __shfl(bpos, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE)
+            hits[bpos + j] = buf[wid][j];
+    }
+}
+
+/* phase 2: dense pass over the survivors — orders date lookup + year
+ * partial aggregate in per-lane registers (the Q1 pattern: 7-year domain,
+ * compile-time indexed), wave+block reduce, one atomic per (year, block). */
+__global__ void k_q9_probe(const otbx_lineitem_dev l,
+                           const uint32_t *__restrict__ hits,
+                           const int64_t *__restrict__ nhits_p,
+                           const int32_t *__restrict__ dtab, int64_t mino,
+                           int64_t orange, double *__restrict__ out_sums,
                            unsigned long long *__restrict__ out_counts)
 {
     double acc[7];
@@ -1320,70 +1408,24 @@ __global__ void k_q9_fused(const otbx_lineitem_dev l,
         acc[y] = 0.0;
         cnt[y] = 0;
     }
-    int64_t nq = l.n / 2;
-    const longlong2 *pk2 = (const longlong2 *)l.l_partkey;
-    const longlong2 *ok2 = (const longlong2 *)l.l_orderkey;
+    int64_t nh = *nhits_p;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
-         q += stride) {
-        longlong2 pk = pk2[q];
-        longlong2 ok = ok2[q];
-        int64_t pks[2] = {pk.x, pk.y};
-        int64_t oks[2] = {ok.x, ok.y};
+    for (int64_t h = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; h < nh;
+         h += stride) {
+        int64_t i = (int64_t)hits[h];
+        int64_t oidx = l.l_orderkey[i] - mino;
+        if (oidx < 0 || oidx >= orange)
+            continue;
+        int32_t date = dtab[oidx];
+        if (date == 0)
+            continue;
+        int32_t y = otbx_year_of_day(date);
+        double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
 #pragma unroll
-        for (int j = 0; j < 2; j++) {
-            int64_t pidx = pks[j] - 1;
-            bool m = pidx >= 0 && pidx < nparts &&
-                     ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
-            /* real branch, not a select: with ~6% of lanes surviving the
-             * part filter, exec-masking skips the dtab/payload requests of
-             * the other 94% (a `dtab[ob ? oidx : 0]` select forces a
-             * request from every lane — measured slower) */
-            if (!m)
-                continue;
-            int64_t oidx = oks[j] - mino;
-            if (oidx < 0 || oidx >= orange)
-                continue;
-            int32_t date = dtab[oidx];
-            if (date == 0)
-                continue;
-            /* payload columns only for surviving rows: the filter columns
-             * (16 B/row) are the only full streams; payload traffic is
-             * per-hit (SURVEY §8d style) */
-            int32_t y = otbx_year_of_day(date);
-            int64_t i = 2 * q + j;
-            double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
-#pragma unroll
-            for (int yy = 0; yy < 7; yy++) {
-                bool hit = yy == y;
-                acc[yy] += hit ? rev : 0.0;
-                cnt[yy] += hit;
-            }
-        }
-    }
-    /* tail rows (n % 2) */
-    if ((l.n & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
-        for (int64_t i = nq * 2; i < l.n; i++) {
-            int64_t pidx = l.l_partkey[i] - 1;
-            if (pidx >= 0 && pidx < nparts &&
-                ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull)) {
-                int64_t oidx = l.l_orderkey[i] - mino;
-                if (oidx >= 0 && oidx < orange) {
-                    int32_t date = dtab[oidx];
-                    if (date != 0) {
-                        int32_t y = otbx_year_of_day(date);
-                        double rev =
-                            l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
-#pragma unroll
-                        for (int yy = 0; yy < 7; yy++) {
-                            if (yy == y) {
-                                acc[yy] += rev;
-                                cnt[yy] += 1;
-                            }
-                        }
-                    }
-                }
-            }
+        for (int yy = 0; yy < 7; yy++) {
+            bool hit = yy == y;
+            acc[yy] += hit ? rev : 0.0;
+            cnt[yy] += hit;
         }
     }
     /* wave + block reduction, one atomic per (year) per block */
@@ -1423,12 +1465,17 @@ __global__ void k_q9_fused(const otbx_lineitem_dev l,
 
 extern "C" {
 
+static inline size_t align64_sz(size_t x) { return (x + 63) / 64 * 64; }
+
 otbx_status otbx_q9_workspace_bytes(int64_t nparts, int64_t norders,
-                                    uint32_t nranks, size_t *bytes)
+                                    int64_t nlineitem, uint32_t nranks,
+                                    size_t *bytes)
 {
     int64_t orange = norders * (int64_t)(nranks ? nranks : 1);
-    *bytes = (size_t)(nparts / 8 + 64) /* part bitmap */ +
-             (size_t)orange * 4 + 64 /* odate direct table */;
+    *bytes = align64_sz(8 * (size_t)((nparts + 63) / 64)) /* part bitmap */ +
+             align64_sz((size_t)orange * 4) /* odate direct table */ +
+             64 /* hit counter */ +
+             (size_t)(nlineitem > 0 ? nlineitem : 1) * 4 /* hit row ids */;
     return OTBX_OK;
 }
 
@@ -1440,10 +1487,15 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
 {
     if (!p || !o || !l || !l->l_partkey || !l->l_orderkey || typemod == 0)
         return OTBX_ERR_INVALID;
+    if (l->n >= (int64_t)UINT32_MAX)
+        return OTBX_ERR_INVALID;  /* u32 hit row ids; shard larger tables */
     hipStream_t s = (hipStream_t)stream;
-    /* orderkey range from the shard layout: local keys span ~n×nranks; use
-     * min/max via the dense layout assumption (keys ≥ 1) — compute exactly
-     * with a minmax kernel to stay layout-agnostic */
+    /* orderkey range from a minmax kernel (layout-agnostic; the dense shard
+     * layout makes the direct odate table the size of the global orders
+     * table). Running the two small build kernels CONCURRENTLY with the
+     * filter on a second stream was measured neutral (6.33 → 6.46 ms step):
+     * on this request-bound pipeline concurrent kernels just split the same
+     * resource — so everything stays on the caller's stream. */
     static int64_t *h_mm = nullptr;
     static unsigned long long *d_mm = nullptr;
     if (!h_mm) {
@@ -1462,14 +1514,20 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
         mino = 1;
         orange = 1;
     }
-    size_t need = (size_t)(p->n / 8 + 64) + (size_t)orange * 4 + 64;
+    size_t bm_bytes = align64_sz(8 * (size_t)((p->n + 63) / 64));
+    size_t dt_bytes = align64_sz((size_t)orange * 4);
+    size_t need = bm_bytes + dt_bytes + 64 +
+                  (size_t)(l->n > 0 ? l->n : 1) * 4;
     if (ws_bytes < need) return OTBX_ERR_INVALID;
     unsigned long long *pbitmap = (unsigned long long *)ws;
-    int32_t *dtab = (int32_t *)((char *)ws + (size_t)(p->n / 8 + 64));
+    int32_t *dtab = (int32_t *)((char *)ws + bm_bytes);
+    int64_t *nhits = (int64_t *)((char *)ws + bm_bytes + dt_bytes);
+    uint32_t *hits = (uint32_t *)((char *)ws + bm_bytes + dt_bytes + 64);
     HIP_CHECK(hipMemsetAsync(sums_dev, 0, 7 * 8, s));
     HIP_CHECK(hipMemsetAsync(counts_dev, 0, 7 * 8, s));
-    HIP_CHECK(hipMemsetAsync(pbitmap, 0, (size_t)(p->n / 8 + 64), s));
+    HIP_CHECK(hipMemsetAsync(pbitmap, 0, bm_bytes, s));
     HIP_CHECK(hipMemsetAsync(dtab, 0, (size_t)orange * 4, s));
+    HIP_CHECK(hipMemsetAsync(nhits, 0, 8, s));
     hipEvent_t ev0 = nullptr, ev1 = nullptr;
     if (kernel_ms) {
         HIP_CHECK(hipEventCreate(&ev0));
@@ -1481,8 +1539,10 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
         hipLaunchKernelGGL(k_q9_odate_build, dim3(grid_for(o->n, 256)),
                            dim3(256), 0, s, *o, mino, dtab);
     if (kernel_ms) HIP_CHECK(hipEventRecord(ev0, s));
-    hipLaunchKernelGGL(k_q9_fused, dim3(grid_for(l->n / 2, 256)), dim3(256), 0,
-                       s, *l, pbitmap, p->n, dtab, mino, orange, sums_dev,
+    hipLaunchKernelGGL(k_q9_filter, dim3(grid_for(l->n / 4, 256)), dim3(256),
+                       0, s, l->l_partkey, l->n, pbitmap, p->n, hits, nhits);
+    hipLaunchKernelGGL(k_q9_probe, dim3(grid_for(l->n / 8, 256)), dim3(256),
+                       0, s, *l, hits, nhits, dtab, mino, orange, sums_dev,
                        (unsigned long long *)counts_dev);
     HIP_CHECK(hipGetLastError());
     if (kernel_ms) {

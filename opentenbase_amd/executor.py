@@ -425,6 +425,7 @@ class GpuQ9Fragment(CustomScanState):
         ws_bytes = C.c_size_t(0)
         check(L.otbx_q9_workspace_bytes(C.c_int64(self.pt.n),
                                         C.c_int64(self.od.n),
+                                        C.c_int64(self.li.n),
                                         C.c_uint32(self.nranks),
                                         C.byref(ws_bytes)))
         ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
