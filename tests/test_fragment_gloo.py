@@ -93,6 +93,33 @@ def test_merge_q3_topk():
     assert top["revenue"].tolist() == [6.0, 5.0, 5.0, 3.0]
 
 
+def _worker_q9(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    sums = torch.zeros(7, dtype=torch.float64)
+    counts = torch.zeros(7, dtype=torch.int64)
+    sums[0] = 1.25 * (rank + 1)
+    counts[0] = 7 * (rank + 1)
+    sums[6] = 3.0
+    counts[6] = rank  # only rank 1 populates 1998
+    rows = fragment.merge_q9_partials(sums, counts)
+    if rank == 0:
+        q.put(rows)
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_merge_q9_partials_two_ranks():
+    rows = _run(_worker_q9)
+    y92 = [r for r in rows if r["o_year"] == 1992][0]
+    assert y92["count_rows"] == 21
+    assert y92["sum_revenue"] == 3.75
+    y98 = [r for r in rows if r["o_year"] == 1998][0]
+    assert y98["count_rows"] == 1
+    assert y98["sum_revenue"] == 6.0
+    assert len(rows) == 2
+
+
 def _worker_a2a(rank, world, port, q):
     from opentenbase_amd import fragment
     _init(rank, world, port)
