@@ -789,12 +789,46 @@ __global__ void k_agg_compact(const agg_slot *tab, int64_t cap,
                               const agg_slot *nullgrp,
                               otbx_agg_group *out, int64_t *ngroups)
 {
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < cap;
-         i += stride) {
+    /* block-aggregated two-phase emission: ONE global reservation per block
+     * (a per-wave wave_append over a 1-billion-slot table made ~16 M
+     * global-counter reservations — measured 201 ms of a 339 ms call at
+     * cap 2^30; this form costs one extra pass over the block's range). */
+    __shared__ unsigned long long lbase;
+    __shared__ unsigned int lcnt, ltot;
+    if (threadIdx.x == 0) lcnt = 0;
+    __syncthreads();
+    int64_t per_block = (cap + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < cap ? lo + per_block : cap;
+    int lane = (int)(threadIdx.x % WAVE);
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        unsigned long long mask = __ballot(tab[i].key != AGG_EMPTY);
+        if (lane == 0 && mask)
+            atomicAdd(&lcnt, (unsigned int)__popcll(mask));
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        ltot = lcnt;
+        lbase = lcnt ? (unsigned long long)atomicAdd(
+                           (unsigned long long *)ngroups,
+                           (unsigned long long)lcnt)
+                     : 0;
+        lcnt = 0;
+    }
+    __syncthreads();
+    if (ltot == 0) goto specials; /* low-cardinality tables: most blocks
+                                   * scan an all-empty range — skip pass 2 */
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
         bool used = tab[i].key != AGG_EMPTY;
-        int64_t pos = wave_append(ngroups, used);
+        unsigned long long mask = __ballot(used);
+        if (!mask) continue;
+        unsigned int wbase = 0;
+        if (lane == 0)
+            wbase = atomicAdd(&lcnt, (unsigned int)__popcll(mask));
+        wbase = (unsigned int)__shfl((int)wbase, 0, WAVE);
         if (used) {
+            int64_t pos = (int64_t)lbase + wbase +
+                          __popcll(mask & ((1ull << lane) - 1ull));
             out[pos].key = tab[i].key;
             out[pos].key_isnull = 0;
             out[pos].count_star = (int64_t)tab[i].count_star;
@@ -803,6 +837,7 @@ __global__ void k_agg_compact(const agg_slot *tab, int64_t cap,
             out[pos].sum_isnull = tab[i].count_v == 0;
         }
     }
+specials:
     if (blockIdx.x == 0 && threadIdx.x == 0) {
         if (nullgrp->count_star > 0) {
             int64_t pos = (int64_t)atomicAdd((unsigned long long *)ngroups, 1ull);
@@ -843,46 +878,69 @@ __device__ __forceinline__ uint32_t d_agg_bucket(int64_t k, uint32_t nb)
     return (uint32_t)((d_hash_i64(k) >> 40) & (uint64_t)(nb - 1));
 }
 
-/* linear-counting distinct estimate over a strided sample (one block) */
-__global__ void k_agg_sample_distinct(const int64_t *__restrict__ keys,
-                                      const uint8_t *__restrict__ knull,
-                                      int64_t n, int64_t *est_out)
+/* exact distinct count over a strided sample, via a global open-addressing
+ * key set (2^21 slots, staged in the — still unused — perm region of the
+ * workspace). The earlier one-block linear-counting estimator saturated
+ * (65k-bit set vs a 65k sample), and its noisy collision count made the
+ * birthday estimate swing 4× at 100 M groups — which undersized nb2 and
+ * sent EVERY bucket down the flagged-recompute path (102 ms of a 162 ms
+ * call). The birthday estimator needs the sample''s distinct count EXACTLY;
+ * only a key-exact set gives it. */
+#define AGG_SCAP (1ll << 21)
+#define AGG_SAMPLE (1ll << 20)
+
+__global__ void k_fill_i64(int64_t *a, int64_t n, int64_t v)
 {
-    const int MBITS = 1 << 16; /* 8 KB bitset */
-    __shared__ unsigned int bits[MBITS / 32];
-    for (int i = threadIdx.x; i < MBITS / 32; i += blockDim.x) bits[i] = 0;
-    __syncthreads();
-    const int SAMPLE = 1 << 16;
-    int64_t stride = n > SAMPLE ? n / SAMPLE : 1;
-    for (int64_t j = threadIdx.x; j < SAMPLE; j += blockDim.x) {
-        int64_t i = j * stride;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        a[i] = v;
+}
+
+__global__ void k_agg_sample_exact(const int64_t *__restrict__ keys,
+                                   const uint8_t *__restrict__ knull,
+                                   int64_t n, int64_t *__restrict__ stab,
+                                   unsigned long long *out2 /* {d, s_eff} */)
+{
+    int64_t stride_n = n > AGG_SAMPLE ? n / AGG_SAMPLE : 1;
+    unsigned long long mynew = 0, myproc = 0;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         j < AGG_SAMPLE; j += stride) {
+        int64_t i = j * stride_n;
         if (i >= n) break;
         if (knull && knull[i]) continue;
-        uint32_t h = (uint32_t)(d_hash_i64(keys[i]) >> 16) & (MBITS - 1);
-        atomicOr(&bits[h >> 5], 1u << (h & 31));
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-        int64_t set = 0;
-        for (int i = 0; i < MBITS / 32; i++) set += __popc(bits[i]);
-        double m = MBITS, x = (double)set;
-        double s_n = n > SAMPLE ? SAMPLE : n;
-        double d_sample =
-            x >= m ? m : -m * log(1.0 - x / m); /* linear counting */
-        /* population-distinct estimate (order of magnitude is enough):
-         * - few distinct in the sample → the sample saturated the key
-         *   domain: D ≈ d_sample (hot keys do NOT scale with n)
-         * - mostly-distinct sample → birthday estimator from the collision
-         *   count: D ≈ s² / (2·(s − d)) */
-        double est;
-        if (d_sample < s_n / 2) {
-            est = d_sample;
-        } else {
-            double c = s_n - d_sample;
-            if (c < 1.0) c = 1.0;
-            est = s_n * s_n / (2.0 * c);
+        myproc++;
+        int64_t k = keys[i];
+        int64_t pos = (int64_t)(d_hash_i64(k) & (uint64_t)(AGG_SCAP - 1));
+        for (;;) {
+            /* plain read first: hot keys resolve with a broadcast load —
+             * 1 M CASes on 4 slots measured +3.7 ms on the 4-group case */
+            long long cur = __hip_atomic_load(
+                (long long *)&stab[pos], __ATOMIC_RELAXED,
+                __HIP_MEMORY_SCOPE_AGENT);
+            if (cur == k) break;
+            if (cur == AGG_EMPTY) {
+                long long old = atomicCAS((unsigned long long *)&stab[pos],
+                                          (unsigned long long)AGG_EMPTY,
+                                          (unsigned long long)k);
+                if (old == AGG_EMPTY) {
+                    mynew++;
+                    break;
+                }
+                if (old == k) break;
+                continue; /* someone else claimed it: re-read this slot */
+            }
+            pos = (pos + 1) & (AGG_SCAP - 1);
         }
-        *est_out = (int64_t)(est < (double)n ? est : (double)n);
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        mynew += __shfl_down(mynew, off, WAVE);
+        myproc += __shfl_down(myproc, off, WAVE);
+    }
+    if ((threadIdx.x % WAVE) == 0 && (mynew | myproc)) {
+        atomicAdd(&out2[0], mynew);
+        atomicAdd(&out2[1], myproc);
     }
 }
 
@@ -972,21 +1030,78 @@ __global__ void k_aggp_scatter_kv(const int64_t *__restrict__ keys,
     }
 }
 
-__global__ void k_aggp_bucket_agg_kv(const ulonglong2 *__restrict__ recs,
-                                     const unsigned long long *__restrict__ offs,
-                                     const unsigned long long *__restrict__ cnts,
-                                     agg_slot *tab, int64_t cap)
+/* level-2 bucket bits: decorrelated from both the level-1 bits (>>40) and
+ * the LDS-slot bits (low) */
+__device__ __forceinline__ uint32_t d_agg_bucket2(int64_t k, uint32_t nb2)
+{
+    return (uint32_t)((d_hash_i64(k) >> 28) & (uint64_t)(nb2 - 1));
+}
+
+/* second partition level (kv fast path): block b re-partitions level-1
+ * bucket b's record segment into nb2 sub-buckets, in place of the random
+ * global spills that dominate when est/nb1 overflows the 1024-slot LDS
+ * table (measured 1.9 Grows/s at 100 M groups). Segment-local: the block
+ * counts, serial-scans, then scatters — 3 streaming passes. */
+__global__ void k_aggp_scatter_kv2(const ulonglong2 *__restrict__ recs,
+                                   const unsigned long long *__restrict__ offs,
+                                   const unsigned long long *__restrict__ cnts,
+                                   uint32_t nb2, ulonglong2 *__restrict__ recs2,
+                                   unsigned long long *__restrict__ offs2,
+                                   unsigned long long *__restrict__ cnts2)
+{
+    __shared__ unsigned int c2[AGGP_MAX_BUCKETS];
+    __shared__ unsigned long long b2[AGGP_MAX_BUCKETS];
+    int64_t lo = (int64_t)offs[blockIdx.x];
+    int64_t hi = lo + (int64_t)cnts[blockIdx.x];
+    for (int j = threadIdx.x; j < (int)nb2; j += blockDim.x) c2[j] = 0;
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        atomicAdd(&c2[d_agg_bucket2((int64_t)recs[i].x, nb2)], 1u);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        unsigned long long acc = (unsigned long long)lo;
+        for (uint32_t j = 0; j < nb2; j++) {
+            b2[j] = acc;
+            offs2[(size_t)blockIdx.x * nb2 + j] = acc;
+            cnts2[(size_t)blockIdx.x * nb2 + j] = c2[j];
+            acc += c2[j];
+            c2[j] = 0;
+        }
+    }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        ulonglong2 r = recs[i];
+        uint32_t j = d_agg_bucket2((int64_t)r.x, nb2);
+        unsigned int off = atomicAdd(&c2[j], 1u);
+        recs2[b2[j] + off] = r;
+    }
+}
+
+/* direct-emit per-bucket aggregation: two-level buckets hold DISJOINT key
+ * sets, so a bucket that fits its LDS table emits its groups straight to
+ * the output — no shared global table, no cap-sized init or compact scan
+ * (those dominated the 100 M-group case: 34 GB init + scan + 100 M random
+ * CAS flushes). A bucket whose LDS table overflows (est noise/skew) is
+ * FLAGGED and recomputed by k_aggp_bucket_agg_kv_flagged through the
+ * global table; its LDS result is discarded. */
+__global__ void k_aggp_bucket_agg_kv_direct(
+    const ulonglong2 *__restrict__ recs,
+    const unsigned long long *__restrict__ offs,
+    const unsigned long long *__restrict__ cnts, uint8_t *__restrict__ flags,
+    unsigned int *nflagged, otbx_agg_group *__restrict__ out,
+    int64_t *ngroups)
 {
     const int LSLOTS = 1024;
     __shared__ agg_lds_slot ltab[LSLOTS];
-    for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
-        ltab[s].key = AGG_EMPTY;
-        ltab[s].cs = 0;
-        ltab[s].cv = 0;
-        ltab[s].sum = 0.0;
+    __shared__ int bfail;
+    if (threadIdx.x == 0) bfail = 0;
+    for (int t = threadIdx.x; t < LSLOTS; t += blockDim.x) {
+        ltab[t].key = AGG_EMPTY;
+        ltab[t].cs = 0;
+        ltab[t].cv = 0;
+        ltab[t].sum = 0.0;
     }
     __syncthreads();
-    int64_t mask = cap - 1;
     int64_t lo = (int64_t)offs[blockIdx.x];
     int64_t hi = lo + (int64_t)cnts[blockIdx.x];
     for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
@@ -995,30 +1110,124 @@ __global__ void k_aggp_bucket_agg_kv(const ulonglong2 *__restrict__ recs,
         double v = __longlong_as_double((long long)r.y);
         if (k == AGG_EMPTY)
             continue; /* handled by k_aggp_specials */
-        int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)(LSLOTS - 1));
+        uint64_t h = d_hash_i64(k);
+        int64_t t0 = (int64_t)(h & (uint64_t)(LSLOTS - 1));
+        /* odd-step double hashing: linear runs at ~0.4 load flagged whole
+         * buckets often enough that their recompute dominated */
+        int64_t step = (int64_t)(((h >> 52) & (uint64_t)(LSLOTS - 2)) | 1ull);
         bool placed = false;
-        for (int t = 0; t < 8; t++) {
-            long long old = atomicCAS((unsigned long long *)&ltab[s].key,
+        for (int t = 0; t < 64; t++) {
+            long long old = atomicCAS((unsigned long long *)&ltab[t0].key,
                                       (unsigned long long)AGG_EMPTY,
                                       (unsigned long long)k);
             if (old == AGG_EMPTY || old == k) {
-                atomicAdd(&ltab[s].cs, 1u);
-                atomicAdd(&ltab[s].cv, 1u);
-                atomicAdd(&ltab[s].sum, v);
+                atomicAdd(&ltab[t0].cs, 1u);
+                atomicAdd(&ltab[t0].cv, 1u);
+                atomicAdd(&ltab[t0].sum, v);
                 placed = true;
                 break;
             }
-            s = (s + 1) & (LSLOTS - 1);
+            t0 = (t0 + step) & (LSLOTS - 1);
         }
-        if (!placed)
-            d_agg_global_update(tab, mask, k, 1ull, 1ull, v);
+        if (!placed) bfail = 1; /* plain store: any-lane sets, read after
+                                 * barrier */
     }
     __syncthreads();
-    for (int s = threadIdx.x; s < LSLOTS; s += blockDim.x) {
-        if (ltab[s].key != AGG_EMPTY)
-            d_agg_global_update(tab, mask, ltab[s].key,
-                                (unsigned long long)ltab[s].cs,
-                                (unsigned long long)ltab[s].cv, ltab[s].sum);
+    if (bfail) {
+        if (threadIdx.x == 0) {
+            flags[blockIdx.x] = 1;
+            atomicAdd(nflagged, 1u);
+        }
+        return;
+    }
+    /* block-aggregated flush: ONE global reservation per block (per-wave
+     * wave_append over 262k blocks × 16 waves made 4.2 M single-counter
+     * reservations = 47 of this kernel''s 50 ms at 100 M groups) */
+    __shared__ unsigned long long fbase;
+    __shared__ unsigned int fcnt;
+    if (threadIdx.x == 0) fcnt = 0;
+    __syncthreads();
+    int lane = (int)(threadIdx.x % WAVE);
+    for (int t = threadIdx.x; t < LSLOTS; t += blockDim.x) {
+        unsigned long long m = __ballot(ltab[t].key != AGG_EMPTY);
+        if (lane == 0 && m) atomicAdd(&fcnt, (unsigned int)__popcll(m));
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        fbase = fcnt ? (unsigned long long)atomicAdd(
+                           (unsigned long long *)ngroups,
+                           (unsigned long long)fcnt)
+                     : 0;
+        fcnt = 0;
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < LSLOTS; t += blockDim.x) {
+        bool used = ltab[t].key != AGG_EMPTY;
+        unsigned long long m = __ballot(used);
+        if (!m) continue;
+        unsigned int wbase = 0;
+        if (lane == 0) wbase = atomicAdd(&fcnt, (unsigned int)__popcll(m));
+        wbase = (unsigned int)__shfl((int)wbase, 0, WAVE);
+        if (used) {
+            int64_t pos = (int64_t)fbase + wbase +
+                          __popcll(m & ((1ull << lane) - 1ull));
+            out[pos].key = ltab[t].key;
+            out[pos].key_isnull = 0;
+            out[pos].count_star = (int64_t)ltab[t].cs;
+            out[pos].count_v = (int64_t)ltab[t].cv;
+            out[pos].sum_v = ltab[t].sum;
+            out[pos].sum_isnull = ltab[t].cv == 0;
+        }
+    }
+}
+
+/* recompute pass for flagged (overflowed) buckets: straight through the
+ * shared global table; k_agg_compact then appends those groups. */
+__global__ void k_aggp_bucket_agg_kv_flagged(
+    const ulonglong2 *__restrict__ recs,
+    const unsigned long long *__restrict__ offs,
+    const unsigned long long *__restrict__ cnts,
+    const uint8_t *__restrict__ flags, agg_slot *tab, int64_t cap)
+{
+    if (!flags[blockIdx.x]) return;
+    int64_t mask = cap - 1;
+    int64_t lo = (int64_t)offs[blockIdx.x];
+    int64_t hi = lo + (int64_t)cnts[blockIdx.x];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        ulonglong2 r = recs[i];
+        int64_t k = (int64_t)r.x;
+        if (k == AGG_EMPTY) continue;
+        d_agg_global_update(tab, mask, k, 1ull, 1ull,
+                            __longlong_as_double((long long)r.y));
+    }
+}
+
+/* append the two special groups (NULL key, AGG_EMPTY-sentinel key) when the
+ * direct path skipped k_agg_compact */
+__global__ void k_agg_emit_specials(const agg_slot *nullgrp,
+                                    otbx_agg_group *out, int64_t *ngroups)
+{
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        if (nullgrp->count_star > 0) {
+            int64_t pos =
+                (int64_t)atomicAdd((unsigned long long *)ngroups, 1ull);
+            out[pos].key = 0;
+            out[pos].key_isnull = 1;
+            out[pos].count_star = (int64_t)nullgrp->count_star;
+            out[pos].count_v = (int64_t)nullgrp->count_v;
+            out[pos].sum_v = nullgrp->sum_v;
+            out[pos].sum_isnull = nullgrp->count_v == 0;
+        }
+        if (nullgrp[1].count_star > 0) {
+            int64_t pos =
+                (int64_t)atomicAdd((unsigned long long *)ngroups, 1ull);
+            out[pos].key = AGG_EMPTY;
+            out[pos].key_isnull = 0;
+            out[pos].count_star = (int64_t)nullgrp[1].count_star;
+            out[pos].count_v = (int64_t)nullgrp[1].count_v;
+            out[pos].sum_v = nullgrp[1].sum_v;
+            out[pos].sum_isnull = nullgrp[1].count_v == 0;
+        }
     }
 }
 
@@ -1142,9 +1351,11 @@ otbx_status otbx_agg_i64_workspace_bytes(int64_t n, size_t *bytes)
                                                       * sentinel-key groups */
     if (n >= AGGP_THRESHOLD) {
         /* partitioned-path buffers: perm + partitioned key/val/vnull copies
-         * + bucket counts/cursor/offsets */
-        b += (size_t)n * (8 + 8 + 8 + 1) + (size_t)AGGP_MAX_BUCKETS * 8 * 3 +
-             4096;
+         * + bucket counts/cursor/offsets, plus the level-2 record buffer
+         * and sub-bucket arrays of the two-level kv path */
+        b += (size_t)n * (8 + 8 + 8 + 1 + 16) +
+             (size_t)AGGP_MAX_BUCKETS * 8 * 3 +
+             (size_t)(1 << 19) * (8 * 2 + 1) + 8192;
     }
     *bytes = b;
     return OTBX_OK;
@@ -1164,106 +1375,180 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
     hipStream_t s = (hipStream_t)stream;
     agg_slot *tab = (agg_slot *)ws;
     agg_slot *nullgrp = tab + cap;
-    hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap + 2, 256)), dim3(256), 0, s,
-                       tab, cap + 2);
     HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, sizeof(int64_t), s));
     if (n == 0) {
         HIP_CHECK(hipGetLastError());
         return OTBX_OK;
     }
 
+    /* path decision first: the two-level direct path never touches the
+     * cap-sized global table, so its init (34 GB of writes at n=600 M) is
+     * skipped unless actually needed */
+    int64_t est = 0;
     bool partitioned = false;
     if (n >= AGGP_THRESHOLD) {
-        /* estimate the distinct-key count; partition when the per-block LDS
-         * table cannot absorb the key domain but buckets would be
-         * LDS-resident */
-        static int64_t *d_est = nullptr;
-        static int64_t *h_est = nullptr;
-        if (!d_est) {
-            HIP_CHECK(hipMalloc(&d_est, 8));
-            HIP_CHECK(hipHostMalloc(&h_est, 8));
+        static unsigned long long *d_ds = nullptr;
+        static unsigned long long *h_ds = nullptr;
+        if (!d_ds) {
+            HIP_CHECK(hipMalloc(&d_ds, 16));
+            HIP_CHECK(hipHostMalloc(&h_ds, 16));
         }
-        hipLaunchKernelGGL(k_agg_sample_distinct, dim3(1), dim3(256), 0, s,
-                           keys, knull, n, d_est);
-        HIP_CHECK(hipMemcpyAsync(h_est, d_est, 8, hipMemcpyDeviceToHost, s));
+        /* sample set lives in the (still unwritten) perm region */
+        int64_t *stab = (int64_t *)((char *)(nullgrp + 2));
+        HIP_CHECK(hipMemsetAsync(d_ds, 0, 16, s));
+        hipLaunchKernelGGL(k_fill_i64, dim3(grid_for(AGG_SCAP, 256)),
+                           dim3(256), 0, s, stab, AGG_SCAP, AGG_EMPTY);
+        hipLaunchKernelGGL(k_agg_sample_exact,
+                           dim3(grid_for(AGG_SAMPLE, 256)), dim3(256), 0, s,
+                           keys, knull, n, stab, d_ds);
+        HIP_CHECK(hipMemcpyAsync(h_ds, d_ds, 16, hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipStreamSynchronize(s));
-        int64_t est = *h_est;
-        if (est > 1536 && est <= (32ll << 20)) {
-            partitioned = true;
-            uint32_t nb = (uint32_t)next_pow2_host(est / 256 < 64 ? 64
-                                                                  : est / 256);
-            if (nb > AGGP_MAX_BUCKETS) nb = AGGP_MAX_BUCKETS;
-            char *p = (char *)(nullgrp + 2);
-            int64_t *perm = (int64_t *)p;
-            int64_t *pkeys = perm + n;
-            double *pvals = (double *)(pkeys + n);
-            uint8_t *pvnull = (uint8_t *)(pvals + n);
-            unsigned long long *cnts =
-                (unsigned long long *)(pvnull + ((n + 63) & ~63ll));
-            unsigned long long *cursor = cnts + AGGP_MAX_BUCKETS;
-            unsigned long long *offs = cursor + AGGP_MAX_BUCKETS;
-            HIP_CHECK(hipMemsetAsync(cnts, 0, (size_t)nb * 8, s));
-            hipLaunchKernelGGL(k_aggp_count, dim3(grid_for(n, 256)), dim3(256),
-                               0, s, keys, knull, n, nb, cnts);
-            static unsigned long long *h_cnts = nullptr;
-            if (!h_cnts)
-                HIP_CHECK(hipHostMalloc(&h_cnts, AGGP_MAX_BUCKETS * 8 * 2));
-            HIP_CHECK(hipMemcpyAsync(h_cnts, cnts, (size_t)nb * 8,
-                                     hipMemcpyDeviceToHost, s));
-            HIP_CHECK(hipStreamSynchronize(s));
-            unsigned long long *h_offs = h_cnts + AGGP_MAX_BUCKETS;
-            unsigned long long acc = 0;
-            for (uint32_t b = 0; b < nb; b++) {
-                h_offs[b] = acc;
-                acc += h_cnts[b];
-            }
-            HIP_CHECK(hipMemcpyAsync(offs, h_offs, (size_t)nb * 8,
-                                     hipMemcpyHostToDevice, s));
-            HIP_CHECK(hipMemcpyAsync(cursor, h_offs, (size_t)nb * 8,
-                                     hipMemcpyHostToDevice, s));
-            int64_t nr = (int64_t)acc; /* partitioned (non-NULL-key) rows */
-            if (!knull && !vnull) {
-                /* fast path: scatter (key,val) records directly — no index
-                 * gathers */
-                ulonglong2 *recs = (ulonglong2 *)perm; /* n×16 ≤ region */
-                hipLaunchKernelGGL(k_aggp_scatter_kv, dim3(grid_for(n, 256)),
-                                   dim3(256), 0, s, keys, vals, n, nb, cursor,
-                                   recs);
-                if (nr > 0)
-                    hipLaunchKernelGGL(k_aggp_bucket_agg_kv, dim3(nb),
-                                       dim3(256), 0, s, recs, offs, cnts, tab,
-                                       cap);
-            } else {
-                hipLaunchKernelGGL(k_aggp_scatter, dim3(grid_for(n, 256)),
-                                   dim3(256), 0, s, keys, knull, n, nb, cursor,
-                                   perm);
-                if (nr > 0) {
-                    hipLaunchKernelGGL(k_gather<int64_t>,
-                                       dim3(grid_for(nr, 256)), dim3(256), 0,
-                                       s, keys, perm, nr, pkeys);
-                    hipLaunchKernelGGL(k_gather<double>,
-                                       dim3(grid_for(nr, 256)), dim3(256), 0,
-                                       s, vals, perm, nr, pvals);
-                    if (vnull)
-                        hipLaunchKernelGGL(k_gather<uint8_t>,
-                                           dim3(grid_for(nr, 256)), dim3(256),
-                                           0, s, vnull, perm, nr, pvnull);
-                    hipLaunchKernelGGL(k_aggp_bucket_agg, dim3(nb), dim3(256),
-                                       0, s, pkeys, pvals,
-                                       vnull ? pvnull : nullptr, offs, cnts,
-                                       tab, cap);
-                }
-            }
-            hipLaunchKernelGGL(k_aggp_specials, dim3(grid_for(n, 256)),
-                               dim3(256), 0, s, keys, knull, vals, vnull, n,
-                               nullgrp);
+        double d = (double)h_ds[0], se = (double)h_ds[1];
+        if (se < 1.0) se = 1.0;
+        if (d < se / 2.0) {
+            est = (int64_t)d; /* sample saturated the key domain */
+        } else {
+            double c = se - d;
+            if (c < 1.0) c = 1.0;
+            est = (int64_t)(se * se / (2.0 * c)); /* birthday */
+        }
+        if (est > n) est = n;
+        /* two-level partitioning (kv fast path only) extends the reach far
+         * past the single-level 2048×~768 ≈ 1.5 M-group ceiling */
+        int64_t est_cap = (!knull && !vnull) ? (256ll << 20) : (32ll << 20);
+        partitioned = est > 1536 && est <= est_cap;
+    }
+
+    uint32_t nb = 0, nb2 = 1;
+    if (partitioned) {
+        nb = (uint32_t)next_pow2_host(est / 256 < 64 ? 64 : est / 256);
+        if (nb > AGGP_MAX_BUCKETS) nb = AGGP_MAX_BUCKETS;
+        int64_t per_b = est / nb;
+        if (!knull && !vnull && per_b > 768) {
+            nb2 = (uint32_t)next_pow2_host(per_b / 384 + 1);
+            if (nb2 > AGGP_MAX_BUCKETS) nb2 = AGGP_MAX_BUCKETS;
+            while ((size_t)nb * nb2 > (1 << 19)) nb2 >>= 1;
         }
     }
-    if (!partitioned)
-        hipLaunchKernelGGL(k_agg_build, dim3(grid_for(n, 256)), dim3(256), 0, s,
-                           keys, knull, vals, vnull, n, tab, cap, nullgrp);
-    hipLaunchKernelGGL(k_agg_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
-                       s, tab, cap, nullgrp, out, ngroups_dev);
+    /* buckets are key-disjoint at either level, so the whole kv fast path
+     * emits direct — no global table, no cap-sized init or compact scan */
+    bool direct = partitioned && !knull && !vnull;
+    if (direct)
+        HIP_CHECK(hipMemsetAsync(nullgrp, 0, 2 * sizeof(agg_slot), s));
+    else
+        hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap + 2, 256)), dim3(256),
+                           0, s, tab, cap + 2);
+
+    bool compact_tab = !direct; /* does the final compact scan the table? */
+    if (partitioned) {
+        char *p = (char *)(nullgrp + 2);
+        int64_t *perm = (int64_t *)p;
+        int64_t *pkeys = perm + n;
+        double *pvals = (double *)(pkeys + n);
+        uint8_t *pvnull = (uint8_t *)(pvals + n);
+        unsigned long long *cnts =
+            (unsigned long long *)(pvnull + ((n + 63) & ~63ll));
+        unsigned long long *cursor = cnts + AGGP_MAX_BUCKETS;
+        unsigned long long *offs = cursor + AGGP_MAX_BUCKETS;
+        unsigned long long *cnts2 = offs + AGGP_MAX_BUCKETS;
+        unsigned long long *offs2 = cnts2 + (1 << 19);
+        uint8_t *flags = (uint8_t *)(offs2 + (1 << 19));
+        ulonglong2 *recs2 = (ulonglong2 *)(flags + (1 << 19));
+        HIP_CHECK(hipMemsetAsync(cnts, 0, (size_t)nb * 8, s));
+        hipLaunchKernelGGL(k_aggp_count, dim3(grid_for(n, 256)), dim3(256), 0,
+                           s, keys, knull, n, nb, cnts);
+        static unsigned long long *h_cnts = nullptr;
+        if (!h_cnts)
+            HIP_CHECK(hipHostMalloc(&h_cnts, AGGP_MAX_BUCKETS * 8 * 2));
+        HIP_CHECK(hipMemcpyAsync(h_cnts, cnts, (size_t)nb * 8,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        unsigned long long *h_offs = h_cnts + AGGP_MAX_BUCKETS;
+        unsigned long long acc = 0;
+        for (uint32_t b = 0; b < nb; b++) {
+            h_offs[b] = acc;
+            acc += h_cnts[b];
+        }
+        HIP_CHECK(hipMemcpyAsync(offs, h_offs, (size_t)nb * 8,
+                                 hipMemcpyHostToDevice, s));
+        HIP_CHECK(hipMemcpyAsync(cursor, h_offs, (size_t)nb * 8,
+                                 hipMemcpyHostToDevice, s));
+        int64_t nr = (int64_t)acc; /* partitioned (non-NULL-key) rows */
+        if (!knull && !vnull) {
+            /* fast path: scatter (key,val) records directly — no index
+             * gathers */
+            ulonglong2 *recs = (ulonglong2 *)perm; /* n×16 ≤ region */
+            hipLaunchKernelGGL(k_aggp_scatter_kv, dim3(grid_for(n, 256)),
+                               dim3(256), 0, s, keys, vals, n, nb, cursor,
+                               recs);
+            if (nr > 0) {
+                static unsigned int *d_nflag = nullptr;
+                static unsigned int *h_nflag = nullptr;
+                if (!d_nflag) {
+                    HIP_CHECK(hipMalloc(&d_nflag, 4));
+                    HIP_CHECK(hipHostMalloc(&h_nflag, 4));
+                }
+                const ulonglong2 *frecs = recs;
+                const unsigned long long *foffs = offs, *fcnts = cnts;
+                uint32_t fgrid = nb;
+                if (nb2 > 1) { /* second partition level */
+                    hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nb),
+                                       dim3(256), 0, s, recs, offs, cnts, nb2,
+                                       recs2, offs2, cnts2);
+                    frecs = recs2;
+                    foffs = offs2;
+                    fcnts = cnts2;
+                    fgrid = (uint32_t)(nb * nb2);
+                }
+                HIP_CHECK(hipMemsetAsync(d_nflag, 0, 4, s));
+                HIP_CHECK(hipMemsetAsync(flags, 0, (size_t)fgrid, s));
+                hipLaunchKernelGGL(k_aggp_bucket_agg_kv_direct, dim3(fgrid),
+                                   dim3(256), 0, s, frecs, foffs, fcnts,
+                                   flags, d_nflag, out, ngroups_dev);
+                HIP_CHECK(hipMemcpyAsync(h_nflag, d_nflag, 4,
+                                         hipMemcpyDeviceToHost, s));
+                HIP_CHECK(hipStreamSynchronize(s));
+                if (*h_nflag > 0) {
+                    /* rare: est noise/skew overflowed some LDS tables —
+                     * recompute those buckets through the global table */
+                    hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap, 256)),
+                                       dim3(256), 0, s, tab, cap);
+                    hipLaunchKernelGGL(k_aggp_bucket_agg_kv_flagged,
+                                       dim3(fgrid), dim3(256), 0, s, frecs,
+                                       foffs, fcnts, flags, tab, cap);
+                    compact_tab = true;
+                }
+            }
+        } else {
+            hipLaunchKernelGGL(k_aggp_scatter, dim3(grid_for(n, 256)),
+                               dim3(256), 0, s, keys, knull, n, nb, cursor,
+                               perm);
+            if (nr > 0) {
+                hipLaunchKernelGGL(k_gather<int64_t>, dim3(grid_for(nr, 256)),
+                                   dim3(256), 0, s, keys, perm, nr, pkeys);
+                hipLaunchKernelGGL(k_gather<double>, dim3(grid_for(nr, 256)),
+                                   dim3(256), 0, s, vals, perm, nr, pvals);
+                if (vnull)
+                    hipLaunchKernelGGL(k_gather<uint8_t>,
+                                       dim3(grid_for(nr, 256)), dim3(256), 0,
+                                       s, vnull, perm, nr, pvnull);
+                hipLaunchKernelGGL(k_aggp_bucket_agg, dim3(nb), dim3(256), 0,
+                                   s, pkeys, pvals, vnull ? pvnull : nullptr,
+                                   offs, cnts, tab, cap);
+            }
+        }
+        hipLaunchKernelGGL(k_aggp_specials, dim3(grid_for(n, 256)), dim3(256),
+                           0, s, keys, knull, vals, vnull, n, nullgrp);
+    } else {
+        hipLaunchKernelGGL(k_agg_build, dim3(grid_for(n, 256)), dim3(256), 0,
+                           s, keys, knull, vals, vnull, n, tab, cap, nullgrp);
+    }
+    if (compact_tab)
+        hipLaunchKernelGGL(k_agg_compact, dim3(grid_for(cap, 256)), dim3(256),
+                           0, s, tab, cap, nullgrp, out, ngroups_dev);
+    else
+        hipLaunchKernelGGL(k_agg_emit_specials, dim3(1), dim3(64), 0, s,
+                           nullgrp, out, ngroups_dev);
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }
