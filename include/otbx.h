@@ -68,6 +68,26 @@ otbx_status otbx_finish(void);
 otbx_status otbx_device_malloc(void **ptr, size_t bytes);
 otbx_status otbx_device_free(void *ptr);
 otbx_status otbx_memcpy_h2d(void *dst_dev, const void *src_host, size_t bytes, void *stream);
+
+/* ---- heap-page → SoA staging shim (host-side; the provider's
+ * BeginCustomScan staging step — INTEGRATION.md §4). Walks PostgreSQL-format
+ * heap pages (PageHeaderData storage/bufpage.h:157; ItemIdData
+ * storage/itemid.h:25; HeapTupleHeaderData access/htup_details.h:118) and
+ * deforms each LP_NORMAL tuple (heap_deform_tuple,
+ * access/common/heaptuple.c:936) into caller-provided per-column host
+ * arrays, ready for otbx_memcpy_h2d. Fixed-width attributes only; MVCC
+ * visibility stays server-side (LP_NORMAL = post-heapgetpage state).
+ * out_nulls[a] (byte-per-row, may be NULL per column) receives the NULL
+ * bitmap; a NULL attribute with no out_nulls[a] is an error. */
+typedef struct {
+    uint16_t attlen;   /* attribute width: 1, 2, 4 or 8 bytes */
+    uint16_t attalign; /* typalign in bytes: 1, 2, 4 or 8 */
+} otbx_attdesc;
+otbx_status otbx_stage_pages(const void *pages_host, int64_t npages,
+                             size_t page_size, const otbx_attdesc *atts,
+                             int32_t natts, void **out_cols_host,
+                             uint8_t **out_nulls_host, int64_t cap_rows,
+                             int64_t *nrows_out);
 otbx_status otbx_memcpy_d2h(void *dst_host, const void *src_dev, size_t bytes, void *stream);
 otbx_status otbx_stream_sync(void *stream);
 

@@ -94,6 +94,7 @@ EXPORTED_SYMBOLS = [
     "otbx_memcpy_d2h", "otbx_stream_sync",
     "otbx_gen_lineitem_dev", "otbx_gen_orders_dev", "otbx_gen_customer_dev",
     "otbx_gen_part_dev", "otbx_q9_workspace_bytes", "otbx_q9_partial",
+    "otbx_stage_pages",
     "otbx_scan_count", "otbx_q1_partial", "otbx_q1_partial_variant",
     "otbx_q3_workspace_bytes", "otbx_q3_partial", "otbx_filter_customer",
     "otbx_topk_by_revenue",
