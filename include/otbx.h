@@ -284,7 +284,8 @@ otbx_status otbx_gather_u8(const uint8_t *src, const int64_t *perm, int64_t n,
 /* Inner hash join on i64 keys: emits (build_idx, probe_idx) pairs in
  * arbitrary order (result-set parity; SQL imposes no order). pairs capacity
  * cap_pairs; overflow → OTBX_ERR_INVALID reported via npairs_dev = -1. */
-otbx_status otbx_join_i64_workspace_bytes(int64_t nb, size_t *bytes);
+otbx_status otbx_join_i64_workspace_bytes(int64_t nb, int64_t np,
+                                           size_t *bytes);
 otbx_status otbx_join_i64(const int64_t *bkeys_dev, const uint8_t *bnull_dev,
                           int64_t nb,
                           const int64_t *pkeys_dev, const uint8_t *pnull_dev,

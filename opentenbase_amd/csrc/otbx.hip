@@ -2370,12 +2370,278 @@ __global__ void k_join_probe(const join_slot *__restrict__ tab, int64_t cap,
     }
 }
 
+/* ---- partitioned join (build ≥ JOINP_THRESHOLD rows) ----
+ * The single shared table runs at 4.8 Gprobes/s at a 150 M-row build
+ * (random 64-B line per probe step). Hash-partition BOTH sides with the
+ * same bucket bits (reusing the aggregation partitioner: level-1
+ * k_joinp_scatter, level-2 k_aggp_scatter_kv2 — records are (key, row)
+ * pairs) into key-disjoint buckets whose build side fits an LDS chained
+ * table; each block joins one bucket with streaming reads. A bucket whose
+ * build side exceeds the LDS capacity (row skew, or an INT64_MIN key — the
+ * LDS table's claim sentinel) is FLAGGED and joined by the global-table
+ * fallback kernels instead. */
+#define JOINP_THRESHOLD (8ll << 20) /* measured: partitioned wins 28.7 vs
+                                     * 49.5 ms at a 15 M-row build, 38.7 vs
+                                     * 125 ms at 150 M (600 M probes) */
+#define JB_CAP 768 /* max build rows per final bucket (LDS arrays) */
+
+__global__ void k_joinp_scatter(const int64_t *__restrict__ keys,
+                                const uint8_t *__restrict__ knull, int64_t n,
+                                uint32_t nbuk, unsigned long long *cursor,
+                                ulonglong2 *__restrict__ recs)
+{
+    __shared__ unsigned int lcur[AGGP_MAX_BUCKETS];
+    __shared__ long long base[AGGP_MAX_BUCKETS];
+    int64_t per_block = (n + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < n ? lo + per_block : n;
+    for (int i = threadIdx.x; i < (int)nbuk; i += blockDim.x) lcur[i] = 0;
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        if (knull && knull[i]) continue; /* NULL keys never match */
+        atomicAdd(&lcur[d_agg_bucket(keys[i], nbuk)], 1u);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < (int)nbuk; i += blockDim.x) {
+        base[i] = lcur[i] ? (long long)atomicAdd(
+                                &cursor[i], (unsigned long long)lcur[i])
+                          : 0;
+        lcur[i] = 0;
+    }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        if (knull && knull[i]) continue;
+        int64_t k = keys[i];
+        uint32_t b = d_agg_bucket(k, nbuk);
+        unsigned int off = atomicAdd(&lcur[b], 1u);
+        ulonglong2 r;
+        r.x = (unsigned long long)k;
+        r.y = (unsigned long long)i;
+        recs[base[b] + off] = r;
+    }
+}
+
+/* per-bucket LDS chained join. Build: open-addressing key slots (claim
+ * sentinel INT64_MIN → such keys force the fallback) with a chain head per
+ * slot; entries chain through lnext. Probe runs TWICE: a counting pass,
+ * then ONE global reservation per block, then an emitting pass writing
+ * pairs at exact offsets through an LDS cursor — per-wave buffer flushes
+ * to the single pair counter measured 2.3 M reservations ≈ 27 ms at 600 M
+ * pairs (the ~88/µs single-counter law); this form makes one per block. */
+__global__ void k_joinp_bucket(const ulonglong2 *__restrict__ brecs,
+                               const unsigned long long *__restrict__ boffs,
+                               const unsigned long long *__restrict__ bcnts,
+                               const ulonglong2 *__restrict__ precs,
+                               const unsigned long long *__restrict__ poffs,
+                               const unsigned long long *__restrict__ pcnts,
+                               uint8_t *__restrict__ flags,
+                               unsigned int *nflagged,
+                               int64_t *__restrict__ out_b,
+                               int64_t *__restrict__ out_p, int64_t cap_pairs,
+                               int64_t *npairs)
+{
+    const int HSLOTS = 1024;
+    __shared__ long long hkey[HSLOTS];
+    __shared__ int hhead[HSLOTS];
+    __shared__ int lnext[JB_CAP];
+    __shared__ int bfail;
+    __shared__ unsigned long long bbase;
+    __shared__ unsigned int bcur;
+    int64_t blo = (int64_t)boffs[blockIdx.x];
+    int64_t nbb = (int64_t)bcnts[blockIdx.x];
+    if (threadIdx.x == 0) {
+        bfail = nbb > JB_CAP;
+        bcur = 0;
+    }
+    for (int t = threadIdx.x; t < HSLOTS; t += blockDim.x) {
+        hkey[t] = INT64_MIN;
+        hhead[t] = -1;
+    }
+    __syncthreads();
+    if (!bfail) {
+        for (int e = threadIdx.x; e < (int)nbb; e += blockDim.x) {
+            int64_t k = (int64_t)brecs[blo + e].x;
+            if (k == INT64_MIN) {
+                bfail = 1;
+                break;
+            }
+            uint64_t h = d_hash_i64(k);
+            int t0 = (int)(h & (HSLOTS - 1));
+            int step = (int)(((h >> 52) & (HSLOTS - 2)) | 1ull);
+            bool placed = false;
+            for (int t = 0; t < HSLOTS; t++) {
+                long long old = atomicCAS((unsigned long long *)&hkey[t0],
+                                          (unsigned long long)INT64_MIN,
+                                          (unsigned long long)k);
+                if (old == INT64_MIN || old == k) {
+                    lnext[e] = atomicExch(&hhead[t0], e);
+                    placed = true;
+                    break;
+                }
+                t0 = (t0 + step) & (HSLOTS - 1);
+            }
+            if (!placed) bfail = 1; /* cannot happen at load ≤ 0.75 */
+        }
+    }
+    __syncthreads();
+    if (bfail) {
+        if (threadIdx.x == 0) {
+            flags[blockIdx.x] = 1;
+            atomicAdd(nflagged, 1u);
+        }
+        return;
+    }
+    int64_t plo = (int64_t)poffs[blockIdx.x];
+    int64_t npb = (int64_t)pcnts[blockIdx.x];
+    int lane = (int)(threadIdx.x % WAVE);
+    /* pass 1: count matches */
+    unsigned int my = 0;
+    for (int64_t i0 = threadIdx.x; i0 < npb; i0 += blockDim.x) {
+        int64_t k = (int64_t)precs[plo + i0].x;
+        if (k == INT64_MIN) continue;
+        uint64_t h = d_hash_i64(k);
+        int t0 = (int)(h & (HSLOTS - 1));
+        int step = (int)(((h >> 52) & (HSLOTS - 2)) | 1ull);
+        int e = -1;
+        for (;;) {
+            long long sk = hkey[t0];
+            if (sk == k) {
+                e = hhead[t0];
+                break;
+            }
+            if (sk == INT64_MIN) break;
+            t0 = (t0 + step) & (HSLOTS - 1);
+        }
+        while (e >= 0) {
+            my++;
+            e = lnext[e];
+        }
+    }
+    {
+        unsigned int w = my;
+        for (int off = WAVE / 2; off > 0; off >>= 1)
+            w += (unsigned int)__shfl_down((int)w, off, WAVE);
+        if (lane == 0 && w) atomicAdd(&bcur, w);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        bbase = bcur ? (unsigned long long)atomicAdd(
+                           (unsigned long long *)npairs,
+                           (unsigned long long)bcur)
+                     : 0;
+        bcur = 0;
+    }
+    __syncthreads();
+    /* pass 2: emit at exact offsets (LDS cursor, zero global atomics) */
+    for (int64_t i0 = threadIdx.x; i0 < npb; i0 += blockDim.x) {
+        ulonglong2 pr = precs[plo + i0];
+        int64_t k = (int64_t)pr.x;
+        if (k == INT64_MIN) continue;
+        uint64_t h = d_hash_i64(k);
+        int t0 = (int)(h & (HSLOTS - 1));
+        int step = (int)(((h >> 52) & (HSLOTS - 2)) | 1ull);
+        int e = -1;
+        for (;;) {
+            long long sk = hkey[t0];
+            if (sk == k) {
+                e = hhead[t0];
+                break;
+            }
+            if (sk == INT64_MIN) break;
+            t0 = (t0 + step) & (HSLOTS - 1);
+        }
+        if (e < 0) continue;
+        unsigned int cnt = 0;
+        for (int e2 = e; e2 >= 0; e2 = lnext[e2]) cnt++;
+        int64_t pos = (int64_t)bbase + atomicAdd(&bcur, cnt);
+        for (; e >= 0; e = lnext[e]) {
+            if (pos < cap_pairs) {
+                out_b[pos] = (long long)brecs[blo + e].y;
+                out_p[pos] = (long long)pr.y;
+            }
+            pos++;
+        }
+    }
+}
+
+/* fallback for flagged buckets: build/probe through the global idx-claim
+ * table (key-agnostic, handles INT64_MIN and any skew) */
+__global__ void k_joinp_build_flagged(const ulonglong2 *__restrict__ brecs,
+                                      const unsigned long long *__restrict__ boffs,
+                                      const unsigned long long *__restrict__ bcnts,
+                                      const uint8_t *__restrict__ flags,
+                                      join_slot *tab, int64_t cap)
+{
+    if (!flags[blockIdx.x]) return;
+    int64_t mask = cap - 1;
+    int64_t lo = (int64_t)boffs[blockIdx.x];
+    int64_t hi = lo + (int64_t)bcnts[blockIdx.x];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        ulonglong2 r = brecs[i];
+        int64_t k = (int64_t)r.x;
+        int64_t t = (int64_t)(d_hash_i64(k) & (uint64_t)mask);
+        while (atomicCAS((unsigned long long *)&tab[t].idx,
+                         (unsigned long long)(-1ll), r.y) !=
+               (unsigned long long)(-1ll))
+            t = (t + 1) & mask;
+        tab[t].key = k;
+    }
+}
+
+__global__ void k_joinp_probe_flagged(const join_slot *__restrict__ tab,
+                                      int64_t cap,
+                                      const ulonglong2 *__restrict__ precs,
+                                      const unsigned long long *__restrict__ poffs,
+                                      const unsigned long long *__restrict__ pcnts,
+                                      const uint8_t *__restrict__ flags,
+                                      int64_t *__restrict__ out_b,
+                                      int64_t *__restrict__ out_p,
+                                      int64_t cap_pairs, int64_t *npairs)
+{
+    if (!flags[blockIdx.x]) return;
+    int64_t mask = cap - 1;
+    int64_t lo = (int64_t)poffs[blockIdx.x];
+    int64_t npb = (int64_t)pcnts[blockIdx.x];
+    for (int64_t i0 = threadIdx.x;; i0 += blockDim.x) {
+        bool active = i0 < npb;
+        ulonglong2 pr;
+        pr.x = 0;
+        pr.y = 0;
+        if (active) pr = precs[lo + i0];
+        int64_t k = (int64_t)pr.x;
+        int64_t t = (int64_t)(d_hash_i64(k) & (uint64_t)mask);
+        bool walking = active;
+        while (__any(walking)) {
+            long long bidx = walking ? tab[t].idx : -1;
+            bool have = walking && bidx >= 0;
+            bool match = have && tab[t].key == k;
+            int64_t pos = wave_append(npairs, match);
+            if (match && pos < cap_pairs) {
+                out_b[pos] = bidx;
+                out_p[pos] = (int64_t)pr.y;
+            }
+            t = (t + 1) & mask;
+            walking = have;
+        }
+        if (__all(i0 >= npb)) break;
+    }
+}
+
 extern "C" {
 
-otbx_status otbx_join_i64_workspace_bytes(int64_t nb, size_t *bytes)
+otbx_status otbx_join_i64_workspace_bytes(int64_t nb, int64_t np,
+                                           size_t *bytes)
 {
     int64_t cap = next_pow2_host(nb < 16 ? 16 : (int64_t)(nb / 0.7) + 1);
-    *bytes = (size_t)cap * sizeof(join_slot);
+    size_t b = (size_t)cap * sizeof(join_slot);
+    const char *fj = getenv("OTBX_JOINP_FORCE");
+    if (nb >= JOINP_THRESHOLD || (fj && atoi(fj))) {
+        /* two partition levels of (key,row) records for both sides +
+         * level-1/level-2 bucket arrays + flags */
+        b += (size_t)(nb + np) * 32 + (size_t)AGGP_MAX_BUCKETS * 8 * 6 +
+             (size_t)(1 << 19) * (8 * 4 + 1) + 8192;
+    }
+    *bytes = b;
     return OTBX_OK;
 }
 
@@ -2386,12 +2652,128 @@ otbx_status otbx_join_i64(const int64_t *bkeys, const uint8_t *bnull, int64_t nb
                           void *stream)
 {
     int64_t cap = next_pow2_host(nb < 16 ? 16 : (int64_t)(nb / 0.7) + 1);
-    if (ws_bytes < (size_t)cap * sizeof(join_slot)) return OTBX_ERR_INVALID;
+    {
+        size_t need;
+        otbx_join_i64_workspace_bytes(nb, np, &need);
+        if (ws_bytes < need) return OTBX_ERR_INVALID;
+    }
     hipStream_t s = (hipStream_t)stream;
     join_slot *tab = (join_slot *)ws;
+    HIP_CHECK(hipMemsetAsync(npairs_dev, 0, sizeof(int64_t), s));
+
+    const char *fj = getenv("OTBX_JOINP_FORCE"); /* test hook: force the
+                                                   * partitioned path */
+    if ((nb >= JOINP_THRESHOLD || (fj && atoi(fj))) && np > 0 && nb > 0) {
+        /* partitioned path: bucket count from build ROWS (duplicates
+         * occupy chain entries), aiming ≤ ~384 rows per final bucket */
+        uint32_t nbuk = AGGP_MAX_BUCKETS;
+        int64_t per_b = nb / nbuk;
+        uint32_t nb2 = 1;
+        if (per_b > 384) {
+            nb2 = (uint32_t)next_pow2_host(per_b / 384 + 1);
+            if (nb2 > AGGP_MAX_BUCKETS) nb2 = AGGP_MAX_BUCKETS;
+            while ((size_t)nbuk * nb2 > (1 << 19)) nb2 >>= 1;
+        }
+        char *p = (char *)(tab + cap);
+        ulonglong2 *brecs = (ulonglong2 *)p;
+        ulonglong2 *brecs2 = brecs + nb;
+        ulonglong2 *precs = brecs2 + nb;
+        ulonglong2 *precs2 = precs + np;
+        unsigned long long *cnts_b = (unsigned long long *)(precs2 + np);
+        unsigned long long *cursor_b = cnts_b + AGGP_MAX_BUCKETS;
+        unsigned long long *offs_b = cursor_b + AGGP_MAX_BUCKETS;
+        unsigned long long *cnts_p = offs_b + AGGP_MAX_BUCKETS;
+        unsigned long long *cursor_p = cnts_p + AGGP_MAX_BUCKETS;
+        unsigned long long *offs_p = cursor_p + AGGP_MAX_BUCKETS;
+        unsigned long long *bcnts2 = offs_p + AGGP_MAX_BUCKETS;
+        unsigned long long *boffs2 = bcnts2 + (1 << 19);
+        unsigned long long *pcnts2 = boffs2 + (1 << 19);
+        unsigned long long *poffs2 = pcnts2 + (1 << 19);
+        uint8_t *flags = (uint8_t *)(poffs2 + (1 << 19));
+        HIP_CHECK(hipMemsetAsync(cnts_b, 0, (size_t)nbuk * 8, s));
+        HIP_CHECK(hipMemsetAsync(cnts_p, 0, (size_t)nbuk * 8, s));
+        hipLaunchKernelGGL(k_aggp_count, dim3(grid_for(nb, 256)), dim3(256),
+                           0, s, bkeys, bnull, nb, nbuk, cnts_b);
+        hipLaunchKernelGGL(k_aggp_count, dim3(grid_for(np, 256)), dim3(256),
+                           0, s, pkeys, pnull, np, nbuk, cnts_p);
+        static unsigned long long *h_j = nullptr;
+        if (!h_j) HIP_CHECK(hipHostMalloc(&h_j, AGGP_MAX_BUCKETS * 8 * 4));
+        HIP_CHECK(hipMemcpyAsync(h_j, cnts_b, (size_t)nbuk * 8,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipMemcpyAsync(h_j + AGGP_MAX_BUCKETS, cnts_p,
+                                 (size_t)nbuk * 8, hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        unsigned long long *h_ob = h_j + 2 * AGGP_MAX_BUCKETS;
+        unsigned long long *h_op = h_j + 3 * AGGP_MAX_BUCKETS;
+        unsigned long long ab = 0, ap = 0;
+        for (uint32_t b = 0; b < nbuk; b++) {
+            h_ob[b] = ab;
+            ab += h_j[b];
+            h_op[b] = ap;
+            ap += h_j[AGGP_MAX_BUCKETS + b];
+        }
+        HIP_CHECK(hipMemcpyAsync(offs_b, h_ob, (size_t)nbuk * 8,
+                                 hipMemcpyHostToDevice, s));
+        HIP_CHECK(hipMemcpyAsync(cursor_b, h_ob, (size_t)nbuk * 8,
+                                 hipMemcpyHostToDevice, s));
+        HIP_CHECK(hipMemcpyAsync(offs_p, h_op, (size_t)nbuk * 8,
+                                 hipMemcpyHostToDevice, s));
+        HIP_CHECK(hipMemcpyAsync(cursor_p, h_op, (size_t)nbuk * 8,
+                                 hipMemcpyHostToDevice, s));
+        hipLaunchKernelGGL(k_joinp_scatter, dim3(grid_for(nb, 256)),
+                           dim3(256), 0, s, bkeys, bnull, nb, nbuk, cursor_b,
+                           brecs);
+        hipLaunchKernelGGL(k_joinp_scatter, dim3(grid_for(np, 256)),
+                           dim3(256), 0, s, pkeys, pnull, np, nbuk, cursor_p,
+                           precs);
+        const ulonglong2 *fb = brecs, *fp = precs;
+        const unsigned long long *fbo = offs_b, *fbc = cnts_b;
+        const unsigned long long *fpo = offs_p, *fpc = cnts_p;
+        uint32_t fgrid = nbuk;
+        if (nb2 > 1) {
+            hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nbuk), dim3(256), 0,
+                               s, brecs, offs_b, cnts_b, nb2, brecs2, boffs2,
+                               bcnts2);
+            hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nbuk), dim3(256), 0,
+                               s, precs, offs_p, cnts_p, nb2, precs2, poffs2,
+                               pcnts2);
+            fb = brecs2;
+            fbo = boffs2;
+            fbc = bcnts2;
+            fp = precs2;
+            fpo = poffs2;
+            fpc = pcnts2;
+            fgrid = (uint32_t)(nbuk * nb2);
+        }
+        static unsigned int *d_nflag = nullptr;
+        static unsigned int *h_nflag = nullptr;
+        if (!d_nflag) {
+            HIP_CHECK(hipMalloc(&d_nflag, 4));
+            HIP_CHECK(hipHostMalloc(&h_nflag, 4));
+        }
+        HIP_CHECK(hipMemsetAsync(d_nflag, 0, 4, s));
+        HIP_CHECK(hipMemsetAsync(flags, 0, (size_t)fgrid, s));
+        hipLaunchKernelGGL(k_joinp_bucket, dim3(fgrid), dim3(256), 0, s, fb,
+                           fbo, fbc, fp, fpo, fpc, flags, d_nflag, out_b,
+                           out_p, cap_pairs, npairs_dev);
+        HIP_CHECK(hipMemcpyAsync(h_nflag, d_nflag, 4, hipMemcpyDeviceToHost,
+                                 s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        if (*h_nflag > 0) {
+            hipLaunchKernelGGL(k_join_init, dim3(grid_for(cap, 256)),
+                               dim3(256), 0, s, tab, cap);
+            hipLaunchKernelGGL(k_joinp_build_flagged, dim3(fgrid), dim3(256),
+                               0, s, fb, fbo, fbc, flags, tab, cap);
+            hipLaunchKernelGGL(k_joinp_probe_flagged, dim3(fgrid), dim3(256),
+                               0, s, tab, cap, fp, fpo, fpc, flags, out_b,
+                               out_p, cap_pairs, npairs_dev);
+        }
+        HIP_CHECK(hipGetLastError());
+        return OTBX_OK;
+    }
+
     hipLaunchKernelGGL(k_join_init, dim3(grid_for(cap, 256)), dim3(256), 0, s,
                        tab, cap);
-    HIP_CHECK(hipMemsetAsync(npairs_dev, 0, sizeof(int64_t), s));
     if (nb > 0)
         hipLaunchKernelGGL(k_join_build, dim3(grid_for(nb, 256)), dim3(256), 0, s,
                            bkeys, bnull, nb, tab, cap);
@@ -2786,14 +3168,23 @@ __global__ void k_q3_compact_direct(const double *__restrict__ rtab,
                                     otbx_q3_group *out, int64_t cap_out,
                                     int64_t *ngroups)
 {
-    int64_t per_block = (range + gridDim.x - 1) / gridDim.x;
+    /* 16-B vector loads on both passes: the scalar-8B version ran at 1.01
+     * ms for a 150 M-entry table — the guide's 0.54-0.7× scalar-load
+     * penalty — vs the ~0.45 ms two-pass stream floor */
+    int64_t nq = range / 2;
+    const v2d *rt2 = (const v2d *)rtab;
+    int64_t per_block = (nq + gridDim.x - 1) / gridDim.x;
     int64_t lo = blockIdx.x * per_block;
-    int64_t hi = lo + per_block < range ? lo + per_block : range;
+    int64_t hi = lo + per_block < nq ? lo + per_block : nq;
     __shared__ int64_t tcnt[256];
     __shared__ int64_t tbase[257];
     int64_t my = 0;
-    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
-        my += rtab[i] != 0.0;
+    for (int64_t q = lo + threadIdx.x; q < hi; q += blockDim.x) {
+        v2d r = rt2[q];
+        my += (r.x != 0.0) + (r.y != 0.0);
+    }
+    if ((range & 1) && blockIdx.x == 0 && threadIdx.x == 0)
+        my += rtab[range - 1] != 0.0;
     tcnt[threadIdx.x] = my;
     __syncthreads();
     if (threadIdx.x == 0) {
@@ -2808,7 +3199,26 @@ __global__ void k_q3_compact_direct(const double *__restrict__ rtab,
     }
     __syncthreads();
     int64_t pos = tbase[256] + tbase[threadIdx.x];
-    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    for (int64_t q = lo + threadIdx.x; q < hi; q += blockDim.x) {
+        v2d r = rt2[q];
+        double revs[2] = {r.x, r.y};
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+            if (revs[j] != 0.0) {
+                int64_t i = 2 * q + j;
+                if (pos < cap_out) {
+                    unsigned long long pl = ptab[i];
+                    out[pos].l_orderkey = mino + i;
+                    out[pos].revenue = revs[j];
+                    out[pos].o_orderdate = (int32_t)(pl & 0xffffffffull);
+                    out[pos].o_shippriority = (int32_t)(pl >> 32);
+                }
+                pos++;
+            }
+        }
+    }
+    if ((range & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
+        int64_t i = range - 1;
         double rev = rtab[i];
         if (rev != 0.0) {
             if (pos < cap_out) {

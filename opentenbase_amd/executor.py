@@ -539,7 +539,8 @@ class GpuHashJoin(CustomScanState):
         L = lib()
         nb, npr = len(self.bk), len(self.pk)
         ws_bytes = C.c_size_t(0)
-        check(L.otbx_join_i64_workspace_bytes(C.c_int64(nb), C.byref(ws_bytes)))
+        check(L.otbx_join_i64_workspace_bytes(C.c_int64(nb), C.c_int64(npr),
+                                              C.byref(ws_bytes)))
         ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8, device="cuda")
         cap = self.cap_pairs if self.cap_pairs else max(4 * max(nb, npr), 64)
         ob = torch.empty(cap, dtype=torch.int64, device="cuda")

@@ -352,6 +352,56 @@ def test_join_parity_gpu(ex, ora):
     assert got == exp
 
 
+def test_join_partitioned_parity(ex, ora, monkeypatch):
+    """The partitioned join path (build ≥ 8M rows in production; forced via
+    OTBX_JOINP_FORCE here) against the oracle — including buckets flagged
+    to the global fallback by INT64_MIN build keys (the LDS table's
+    reserved sentinel) and NULL keys on both sides."""
+    monkeypatch.setenv("OTBX_JOINP_FORCE", "1")
+    rng = np.random.default_rng(17)
+    nb, npr = 300000, 900000
+    bk = rng.integers(0, 200000, nb)
+    pk = rng.integers(0, 200000, npr)
+    bk[::5000] = np.iinfo(np.int64).min   # sentinel keys -> flagged buckets
+    pk[::7000] = np.iinfo(np.int64).min
+    bn = (rng.random(nb) < 0.03).astype(np.uint8)
+    pn = (rng.random(npr) < 0.03).astype(np.uint8)
+    dev = lambda a, dt: torch.as_tensor(a, dtype=dt, device="cuda")  # noqa: E731
+    node = ex.GpuHashJoin(dev(bk, torch.int64), dev(pk, torch.int64),
+                          dev(bn, torch.uint8), dev(pn, torch.uint8),
+                          cap_pairs=8 * npr)
+    pairs = drain(node)
+    obi, opi = ora.join_i64(bk, pk, bnull=bn, pnull=pn)
+    assert len(pairs) == len(obi)
+    assert sorted(pairs) == sorted(zip(obi.tolist(), opi.tolist()))
+
+
+def test_join_partitioned_large_count(ex):
+    """Pair-count parity at a size that takes the partitioned path without
+    forcing (build 10M > the 8M threshold): expected count computed from
+    the key histograms (sum over keys of cb[k]*cp[k])."""
+    g = torch.Generator(device="cuda").manual_seed(5)
+    nb, npr = 10_000_000, 20_000_000
+    dom = 4_000_000
+    bk = torch.randint(0, dom, (nb,), dtype=torch.int64, device="cuda",
+                       generator=g)
+    pk = torch.randint(0, dom, (npr,), dtype=torch.int64, device="cuda",
+                       generator=g)
+    cb = torch.bincount(bk, minlength=dom)
+    cp = torch.bincount(pk, minlength=dom)
+    expected = int((cb * cp).sum().item())
+    node = ex.GpuHashJoin(bk, pk, cap_pairs=expected + 64)
+    node.BeginCustomScan()
+    node._rows = node._run()
+    assert len(node._rows) == expected
+    # spot-check: every emitted pair joins equal keys
+    import numpy as np_
+    pairs = np_.array(node._rows[:100000], dtype=np_.int64)
+    bkh = bk.cpu().numpy()
+    pkh = pk.cpu().numpy()
+    assert (bkh[pairs[:, 0]] == pkh[pairs[:, 1]]).all()
+
+
 # ---------------- golden vectors on the GPU path ----------------
 
 def test_golden_onek_gpu(ex):

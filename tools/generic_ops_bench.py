@@ -47,7 +47,8 @@ def bench_join(nb, np_):
                        generator=g)
     L = lib()
     ws_bytes = C.c_size_t(0)
-    L.otbx_join_i64_workspace_bytes(C.c_int64(nb), C.byref(ws_bytes))
+    L.otbx_join_i64_workspace_bytes(C.c_int64(nb), C.c_int64(np_),
+                                    C.byref(ws_bytes))
     ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
     cap = int(np_ * 2.2)
     ob = torch.empty(cap, dtype=torch.int64, device="cuda")
