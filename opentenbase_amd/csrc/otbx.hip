@@ -722,6 +722,42 @@ __device__ __forceinline__ void d_agg_global_update(agg_slot *tab, int64_t mask,
     }
 }
 
+/* bounded variant for the small-table first attempt: gives up after
+ * maxprobe steps (table effectively full → the sample-based sizing was
+ * wrong) and reports failure instead of spinning */
+__device__ __forceinline__ bool
+d_agg_global_update_b(agg_slot *tab, int64_t mask, int64_t k,
+                      unsigned long long cs, unsigned long long cv,
+                      double sum, int maxprobe)
+{
+    int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)mask);
+    for (int t = 0; t < maxprobe; t++) {
+        long long old = atomicCAS((unsigned long long *)&tab[s].key,
+                                  (unsigned long long)AGG_EMPTY,
+                                  (unsigned long long)k);
+        if (old == AGG_EMPTY || old == k) {
+            atomicAdd(&tab[s].count_star, cs);
+            if (cv) {
+                atomicAdd(&tab[s].count_v, cv);
+                atomicAdd(&tab[s].sum_v, sum);
+            }
+            return true;
+        }
+        s = (s + 1) & mask;
+    }
+    return false;
+}
+
+/* small-table build attempt: k_agg_build with bounded global probes + an
+ * abort flag. When the estimator-sized table overflows (rare-key tail the
+ * sample missed), the whole call is redone against the full-size table. */
+__global__ void k_agg_build_bounded(const int64_t *__restrict__ keys,
+                                    const uint8_t *__restrict__ knull,
+                                    const double *__restrict__ vals,
+                                    const uint8_t *__restrict__ vnull,
+                                    int64_t n, agg_slot *tab, int64_t cap,
+                                    agg_slot *nullgrp, unsigned int *abort);
+
 __global__ void k_agg_build(const int64_t *__restrict__ keys,
                             const uint8_t *__restrict__ knull,
                             const double *__restrict__ vals,
@@ -782,6 +818,74 @@ __global__ void k_agg_build(const int64_t *__restrict__ keys,
             d_agg_global_update(tab, mask, ltab[s].key,
                                 (unsigned long long)ltab[s].cs,
                                 (unsigned long long)ltab[s].cv, ltab[s].sum);
+    }
+}
+
+__global__ void k_agg_build_bounded(const int64_t *__restrict__ keys,
+                                    const uint8_t *__restrict__ knull,
+                                    const double *__restrict__ vals,
+                                    const uint8_t *__restrict__ vnull,
+                                    int64_t n, agg_slot *tab, int64_t cap,
+                                    agg_slot *nullgrp, unsigned int *abort)
+{
+    __shared__ agg_lds_slot ltab[AGG_LDS_SLOTS];
+    for (int s = threadIdx.x; s < AGG_LDS_SLOTS; s += blockDim.x) {
+        ltab[s].key = AGG_EMPTY;
+        ltab[s].cs = 0;
+        ltab[s].cv = 0;
+        ltab[s].sum = 0.0;
+    }
+    __syncthreads();
+    int64_t mask = cap - 1;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    bool dead = false;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < n && !dead; i += stride) {
+        bool kn = knull && knull[i];
+        bool vn = vnull && vnull[i];
+        double v = vn ? 0.0 : vals[i];
+        if (kn || keys[i] == AGG_EMPTY) {
+            agg_slot *e = kn ? nullgrp : nullgrp + 1;
+            atomicAdd(&e->count_star, 1ull);
+            if (!vn) {
+                atomicAdd(&e->count_v, 1ull);
+                atomicAdd(&e->sum_v, v);
+            }
+            continue;
+        }
+        int64_t k = keys[i];
+        int64_t s = (int64_t)(d_hash_i64(k) & (uint64_t)(AGG_LDS_SLOTS - 1));
+        bool placed = false;
+        for (int t = 0; t < AGG_LDS_PROBES; t++) {
+            long long old = atomicCAS((unsigned long long *)&ltab[s].key,
+                                      (unsigned long long)AGG_EMPTY,
+                                      (unsigned long long)k);
+            if (old == AGG_EMPTY || old == k) {
+                atomicAdd(&ltab[s].cs, 1u);
+                if (!vn) {
+                    atomicAdd(&ltab[s].cv, 1u);
+                    atomicAdd(&ltab[s].sum, v);
+                }
+                placed = true;
+                break;
+            }
+            s = (s + 1) & (AGG_LDS_SLOTS - 1);
+        }
+        if (!placed &&
+            !d_agg_global_update_b(tab, mask, k, 1ull, vn ? 0ull : 1ull, v,
+                                   256)) {
+            atomicAdd(abort, 1u); /* undersized: caller redoes at full cap */
+            dead = true;
+        }
+    }
+    __syncthreads();
+    for (int s = threadIdx.x; s < AGG_LDS_SLOTS; s += blockDim.x) {
+        if (ltab[s].key != AGG_EMPTY &&
+            !d_agg_global_update_b(tab, mask, ltab[s].key,
+                                   (unsigned long long)ltab[s].cs,
+                                   (unsigned long long)ltab[s].cv,
+                                   ltab[s].sum, 256))
+            atomicAdd(abort, 1u);
     }
 }
 
@@ -1433,9 +1537,10 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
     /* buckets are key-disjoint at either level, so the whole kv fast path
      * emits direct — no global table, no cap-sized init or compact scan */
     bool direct = partitioned && !knull && !vnull;
+    bool smalltab = !partitioned && est > 0 && est <= 1536;
     if (direct)
         HIP_CHECK(hipMemsetAsync(nullgrp, 0, 2 * sizeof(agg_slot), s));
-    else
+    else if (!smalltab) /* the small-table attempt inits its own size */
         hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap + 2, 256)), dim3(256),
                            0, s, tab, cap + 2);
 
@@ -1539,6 +1644,39 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
         }
         hipLaunchKernelGGL(k_aggp_specials, dim3(grid_for(n, 256)), dim3(256),
                            0, s, keys, knull, vals, vnull, n, nullgrp);
+    } else if (smalltab) {
+        /* low-cardinality (exact-sampled): try an estimator-sized table —
+         * the full cap-sized init + compact scan cost ~12 ms of the
+         * 4-group case's 17 ms. The rare-key tail the sample missed can
+         * overflow it: bounded probes set an abort flag and the call
+         * redoes against the full table. */
+        static unsigned int *d_ab = nullptr;
+        static unsigned int *h_ab = nullptr;
+        if (!d_ab) {
+            HIP_CHECK(hipMalloc(&d_ab, 4));
+            HIP_CHECK(hipHostMalloc(&h_ab, 4));
+        }
+        int64_t cap_s = next_pow2_host(est * 64 < 4096 ? 4096 : est * 64);
+        if (cap_s > cap) cap_s = cap;
+        HIP_CHECK(hipMemsetAsync(d_ab, 0, 4, s));
+        hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap_s, 256)),
+                           dim3(256), 0, s, tab, cap_s);
+        /* nullgrp sits at tab+cap (FULL cap), outside the small init */
+        HIP_CHECK(hipMemsetAsync(nullgrp, 0, 2 * sizeof(agg_slot), s));
+        hipLaunchKernelGGL(k_agg_build_bounded, dim3(grid_for(n, 256)),
+                           dim3(256), 0, s, keys, knull, vals, vnull, n, tab,
+                           cap_s, nullgrp, d_ab);
+        HIP_CHECK(hipMemcpyAsync(h_ab, d_ab, 4, hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        if (*h_ab == 0) {
+            cap = cap_s; /* compact scans only the small table */
+        } else {
+            hipLaunchKernelGGL(k_agg_init, dim3(grid_for(cap + 2, 256)),
+                               dim3(256), 0, s, tab, cap + 2);
+            hipLaunchKernelGGL(k_agg_build, dim3(grid_for(n, 256)), dim3(256),
+                               0, s, keys, knull, vals, vnull, n, tab, cap,
+                               nullgrp);
+        }
     } else {
         hipLaunchKernelGGL(k_agg_build, dim3(grid_for(n, 256)), dim3(256), 0,
                            s, keys, knull, vals, vnull, n, tab, cap, nullgrp);

@@ -352,6 +352,41 @@ def test_join_parity_gpu(ex, ora):
     assert got == exp
 
 
+def test_agg_small_table_abort_path(ex):
+    """Low-cardinality estimate with a rare-key tail the sample misses: the
+    estimator-sized table aborts (bounded probes) and the call redoes at
+    full capacity — group count and count conservation must hold."""
+    import ctypes as C
+    from opentenbase_amd._lib import call, lib
+    n = 8_400_000  # ≥ AGGP_THRESHOLD so the estimator runs; stride = n/2^20
+    stride = n // (1 << 20)
+    keys_np = np.arange(n, dtype=np.int64) + 1_000_000
+    keys_np[::stride] = 7  # every sampled position is the hot key
+    keys = torch.as_tensor(keys_np, device="cuda")
+    vals = torch.ones(n, dtype=torch.float64, device="cuda")
+    L = lib()
+    wsb = C.c_size_t(0)
+    L.otbx_agg_i64_workspace_bytes(C.c_int64(n), C.byref(wsb))
+    ws = torch.empty(wsb.value, dtype=torch.uint8, device="cuda")
+    out = torch.empty(n * 40, dtype=torch.uint8, device="cuda")
+    ng = torch.zeros(1, dtype=torch.int64, device="cuda")
+    stream = C.c_void_p(torch.cuda.current_stream().cuda_stream)
+    call("otbx_agg_i64", C.c_void_p(keys.data_ptr()), None,
+         C.c_void_p(vals.data_ptr()), None, C.c_int64(n),
+         C.c_void_p(ws.data_ptr()), C.c_size_t(wsb.value),
+         C.c_void_p(out.data_ptr()), C.c_void_p(ng.data_ptr()), stream)
+    g = int(ng.cpu().item())
+    assert g == len(np.unique(keys_np))
+    dt = np.dtype([("key", "i8"), ("count_star", "i8"), ("count_v", "i8"),
+                   ("sum_v", "f8"), ("key_isnull", "i4"),
+                   ("sum_isnull", "i4")])
+    groups = out[: g * 40].cpu().numpy().view(dt)
+    assert int(groups["count_star"].sum()) == n
+    hot = groups[groups["key"] == 7]
+    assert len(hot) == 1
+    assert int(hot["count_star"][0]) == len(np.arange(0, n, stride))
+
+
 def test_join_partitioned_parity(ex, ora, monkeypatch):
     """The partitioned join path (build ≥ 8M rows in production; forced via
     OTBX_JOINP_FORCE here) against the oracle — including buckets flagged
