@@ -106,6 +106,11 @@ typedef struct {
     int64_t n;
     int64_t *o_orderkey, *o_custkey;
     int32_t *o_orderdate, *o_shippriority;
+    /* staged-table metadata (zone-map style, computed once at staging —
+     * otbx_gen_orders_dev fills it; a provider computes it while walking
+     * pages). has_minmax = 0 → executors run their own minmax kernel. */
+    int64_t okey_min, okey_max;
+    int32_t has_minmax;
 } otbx_orders_dev;
 
 typedef struct {
@@ -127,7 +132,7 @@ typedef struct {
 otbx_status otbx_gen_lineitem_dev(const otbx_lineitem_dev *t, uint64_t seed,
                                   int64_t n_global, uint32_t rank,
                                   uint32_t nranks, void *stream);
-otbx_status otbx_gen_orders_dev(const otbx_orders_dev *t, uint64_t seed,
+otbx_status otbx_gen_orders_dev(otbx_orders_dev *t, uint64_t seed,
                                 int64_t n_global, int64_t ncust_global,
                                 uint32_t rank, uint32_t nranks, int skew,
                                 void *stream);

@@ -42,6 +42,8 @@ class OrdersDev(C.Structure):
         ("n", C.c_int64),
         ("o_orderkey", C.c_void_p), ("o_custkey", C.c_void_p),
         ("o_orderdate", C.c_void_p), ("o_shippriority", C.c_void_p),
+        ("okey_min", C.c_int64), ("okey_max", C.c_int64),
+        ("has_minmax", C.c_int32),
     ]
 
 
