@@ -132,6 +132,62 @@ def _worker_a2a(rank, world, port, q):
     torch.distributed.destroy_process_group()
 
 
+def _run_n(worker, world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29911 + np.random.randint(0, 400)
+    procs = [ctx.Process(target=worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    res = q.get(timeout=180)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    return res
+
+
+def _worker_q1_w4(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    sums = torch.zeros((6, 5), dtype=torch.float64)
+    counts = torch.zeros(6, dtype=torch.int64)
+    sums[0, 0] = float(rank + 1)
+    counts[0] = rank + 1
+    rows = fragment.merge_q1_partials(sums, counts)
+    if rank == 0:
+        q.put(rows)
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_merge_q1_partials_four_ranks():
+    """World-size 4 (the driver's N=4 scale point): elementwise combine
+    over four shards."""
+    rows = _run_n(_worker_q1_w4, 4)
+    af = [r for r in rows if r["l_returnflag"] == "A"][0]
+    assert af["count_order"] == 1 + 2 + 3 + 4
+    assert af["sum_qty"] == 1.0 + 2.0 + 3.0 + 4.0
+
+
+def _worker_a2a_w4(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    # rank r sends one element tagged (r, dest) to every dest
+    t = torch.tensor([rank * 10 + d for d in range(world)], dtype=torch.int64)
+    out = fragment.all_to_all_variable(t, [1] * world)
+    if rank == 1:
+        q.put(out.numpy())
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_all_to_all_variable_four_ranks():
+    out = _run_n(_worker_a2a_w4, 4)
+    # rank 1 receives element (r*10 + 1) from every rank r, in rank order
+    assert out.tolist() == [1, 11, 21, 31]
+
+
 @pytest.mark.timeout(180)
 def test_all_to_all_variable_gloo():
     out = _run(_worker_a2a)

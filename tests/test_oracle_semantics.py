@@ -202,6 +202,28 @@ def test_q3_sharded_union_equals_full():
         assert us[k] == pytest.approx(fs[k], rel=1e-12)
 
 
+def test_q9_cli_matches_ctypes():
+    """The q9 CLI entry (bench.py's Q9 cpu_baseline) agrees with the
+    library path."""
+    import json
+    import subprocess
+    import os
+    cli = os.path.join(os.path.dirname(ora.__file__), "oracle_cli")
+    if not os.path.exists(cli):
+        subprocess.run(["make", "-C", os.path.dirname(ora.__file__)],
+                       check=True, capture_output=True)
+    out = subprocess.run([cli, "q9", "--rows", "120000"], check=True,
+                         capture_output=True, text=True).stdout
+    j = json.loads(out)
+    exp = ora.q9_partial(ora.gen_tables(120000,
+                                        need=("lineitem", "orders", "part")))
+    assert len(j["groups"]) == len(exp)
+    for jg, g in zip(j["groups"], exp):
+        assert jg["year"] == 1992 + g.year
+        assert jg["count"] == g.count_rows
+        assert jg["revenue"] == g.revenue  # same scan order: bit-exact
+
+
 def test_q1_cli_matches_ctypes(tmp_path):
     """The CLI (bench.py's cpu_baseline entry) agrees with the library."""
     import json

@@ -210,6 +210,52 @@ def test_capacity_exceeded_is_error():
     assert st == 3
 
 
+def test_staging_fuzz_mixed_pages():
+    """Randomized pages: variable tuples per page, random dead/unused
+    items, random NULL masks, a dropped-column tuple — the walker's output
+    must equal a straightforward Python reconstruction."""
+    rng = np.random.default_rng(23)
+    atts = [(8, 8), (4, 4), (8, 8), (1, 1)]
+    ncols = len(atts)
+    pages = []
+    exp_rows = []
+    for _ in range(20):
+        tups, flags = [], []
+        for _ in range(int(rng.integers(0, 40))):
+            flag = int(rng.choice([LP_NORMAL, LP_NORMAL, LP_NORMAL,
+                                   LP_DEAD, LP_UNUSED]))
+            nm = [bool(rng.random() < 0.2) for _ in range(ncols)]
+            vals_raw = [int(rng.integers(-2**40, 2**40)),
+                        int(rng.integers(-2**20, 2**20)),
+                        int(rng.integers(0, 2**30)),
+                        int(rng.integers(0, 256))]
+            packs = [np.int64(vals_raw[0]).tobytes(),
+                     np.int32(vals_raw[1]).tobytes(),
+                     np.int64(vals_raw[2]).tobytes(),
+                     bytes([vals_raw[3]])]
+            packed = [None if nm[a] else packs[a] for a in range(ncols)]
+            tups.append(pack_tuple(packed, atts,
+                                   null_mask=nm if any(nm) else None))
+            flags.append(flag)
+            if flag == LP_NORMAL:
+                exp_rows.append((vals_raw, nm))
+        pages.append(pack_page(tups, flags=flags))
+    blob = b"".join(pages)
+    cap = len(exp_rows) + 8
+    st, nrows, cols, nulls = stage(blob, len(pages), atts, ncols, cap,
+                                   with_nulls=True)
+    assert st == 0
+    assert nrows == len(exp_rows)
+    dts = [np.int64, np.int32, np.int64, np.uint8]
+    for a in range(ncols):
+        got = np.frombuffer(bytes(cols[a])[: nrows * atts[a][0]],
+                            dtype=dts[a])
+        for r, (vals_raw, nm) in enumerate(exp_rows):
+            assert nulls[a][r] == (1 if nm[a] else 0), (a, r)
+            if not nm[a]:
+                assert int(got[r]) == int(dts[a](vals_raw[a])), (a, r)
+
+
 @pytest.mark.gpu
 def test_staged_pages_q1_parity():
     """End to end: heap pages → walker → from_host staging → fused Q1 kernel
