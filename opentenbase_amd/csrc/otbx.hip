@@ -433,6 +433,7 @@ __global__ void k_q1_partial(const int32_t *__restrict__ sd,
 /* variant 2: 4 rows/lane (int4 dates, uchar4 flags, 2× double2 per column),
  * optional non-temporal loads on the read-once f64 streams (NT template). */
 typedef double v2d __attribute__((ext_vector_type(2)));
+typedef long long v2l __attribute__((ext_vector_type(2)));
 
 template <bool NT>
 __device__ __forceinline__ double2 ld2(const double2 *p)
@@ -1707,17 +1708,22 @@ __global__ void k_minmax_i64(const int64_t *__restrict__ keys, int64_t n,
                              unsigned long long *minkey,
                              unsigned long long *maxkey); /* defined below */
 
-/* part bitmap: dense p_partkey 1..nparts, bit set where
- * p_type % typemod == typeval (the p_name LIKE filter restatement) */
+/* part bitmap for partkey slice [lo, hi): dense p_partkey 1..nparts, bit
+ * (idx - lo) set where p_type % typemod == typeval (the p_name LIKE filter
+ * restatement). Sliced so the bitmap stays inside a per-XCD L2: at SF300
+ * a full 7.5 MB bitmap spilled to HBM and the gather rate collapsed
+ * (DESIGN.md §7 SF300 table). */
 __global__ void k_q9_part_bitmap(const otbx_part_dev p, uint8_t typemod,
-                                 uint8_t typeval, unsigned long long *bitmap)
+                                 uint8_t typeval, int64_t lo, int64_t hi,
+                                 unsigned long long *bitmap)
 {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < p.n;
          i += stride) {
         if (p.p_type[i] % typemod == typeval) {
             int64_t idx = p.p_partkey[i] - 1;
-            atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
+            if (idx >= lo && idx < hi)
+                atomicOr(&bitmap[(idx - lo) >> 6], 1ull << ((idx - lo) & 63));
         }
     }
 }
@@ -1745,8 +1751,8 @@ __global__ void k_q9_odate_build(const otbx_orders_dev o, int64_t mino,
  * n < 2^32). */
 __global__ void k_q9_filter(const int64_t *__restrict__ pk, int64_t n,
                             const unsigned long long *__restrict__ pbitmap,
-                            int64_t nparts, uint32_t *__restrict__ hits,
-                            int64_t *nhits)
+                            int64_t lo, int64_t hi,
+                            uint32_t *__restrict__ hits, int64_t *nhits)
 {
     /* LDS-staged per-wave append, one global reservation per ~BUF rows —
      * a bare wave_append here would make ~n/64 single-counter reservations,
@@ -1760,7 +1766,7 @@ __global__ void k_q9_filter(const int64_t *__restrict__ pk, int64_t n,
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
     int nbuf = 0;
     int64_t nq = n / 4;
-    const longlong2 *pk2 = (const longlong2 *)pk;
+    const v2l *pk2 = (const v2l *)pk;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
          q += stride) {
@@ -1768,20 +1774,23 @@ __global__ void k_q9_filter(const int64_t *__restrict__ pk, int64_t n,
         int64_t r0 = q * 4;
         int mycnt = 0;
         if (q < nq) {
-            longlong2 pa = pk2[2 * q], pb = pk2[2 * q + 1];
+            /* non-temporal: the single-use pk stream must not evict the
+             * L2-resident slice bitmap it races against */
+            v2l pa = __builtin_nontemporal_load(&pk2[2 * q]);
+            v2l pb = __builtin_nontemporal_load(&pk2[2 * q + 1]);
             int64_t pks[4] = {pa.x, pa.y, pb.x, pb.y};
 #pragma unroll
             for (int j = 0; j < 4; j++) {
-                int64_t pidx = pks[j] - 1;
-                m[j] = pidx >= 0 && pidx < nparts &&
+                int64_t pidx = pks[j] - 1 - lo;
+                m[j] = pidx >= 0 && pidx < hi - lo &&
                        ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
                 mycnt += m[j];
             }
         } else if (q == nq) { /* tail rows (n % 4) */
             for (int64_t i = nq * 4; i < n; i++) {
                 int j = (int)(i - nq * 4);
-                int64_t pidx = pk[i] - 1;
-                m[j] = pidx >= 0 && pidx < nparts &&
+                int64_t pidx = pk[i] - 1 - lo;
+                m[j] = pidx >= 0 && pidx < hi - lo &&
                        ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
                 mycnt += m[j];
             }
@@ -1905,7 +1914,8 @@ otbx_status otbx_q9_workspace_bytes(int64_t nparts, int64_t norders,
                                     size_t *bytes)
 {
     int64_t orange = norders * (int64_t)(nranks ? nranks : 1);
-    *bytes = align64_sz(8 * (size_t)((nparts + 63) / 64)) /* part bitmap */ +
+    int64_t slice = nparts < (24ll << 20) ? nparts : (24ll << 20);
+    *bytes = align64_sz(8 * (size_t)((slice + 63) / 64)) /* part bitmap */ +
              align64_sz((size_t)orange * 4) /* odate direct table */ +
              64 /* hit counter */ +
              (size_t)(nlineitem > 0 ? nlineitem : 1) * 4 /* hit row ids */;
@@ -1952,7 +1962,22 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
         mino = h_mm[0];
         orange = h_mm[1] - h_mm[0] + 1;
     }
-    size_t bm_bytes = align64_sz(8 * (size_t)((p->n + 63) / 64));
+    /* keep the slice bitmap inside a per-XCD L2 (4 MB): a full SF300
+     * bitmap (7.5 MB) spilled the gathers to HBM. Grace-style partkey
+     * slices: each pass re-streams l_partkey (cheap) and gathers only the
+     * rows whose key falls in the slice (each row gathers in exactly one
+     * pass). Test hook: OTBX_Q9_BITMAP_BITS. */
+    int64_t cap_bits = 24ll << 20;
+    const char *cb = getenv("OTBX_Q9_BITMAP_BITS");
+    if (cb) {
+        int64_t e = atoll(cb);
+        if (e >= 64) cap_bits = e;
+    }
+    int64_t slice = p->n < cap_bits ? p->n : cap_bits;
+    if (slice < 1) slice = 1; /* empty part table: one no-match pass */
+    int64_t npasses = (p->n + slice - 1) / slice;
+    if (npasses < 1) npasses = 1;
+    size_t bm_bytes = align64_sz(8 * (size_t)((slice + 63) / 64));
     size_t dt_bytes = align64_sz((size_t)orange * 4);
     size_t need = bm_bytes + dt_bytes + 64 +
                   (size_t)(l->n > 0 ? l->n : 1) * 4;
@@ -1963,25 +1988,35 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
     uint32_t *hits = (uint32_t *)((char *)ws + bm_bytes + dt_bytes + 64);
     HIP_CHECK(hipMemsetAsync(sums_dev, 0, 7 * 8, s));
     HIP_CHECK(hipMemsetAsync(counts_dev, 0, 7 * 8, s));
-    HIP_CHECK(hipMemsetAsync(pbitmap, 0, bm_bytes, s));
     HIP_CHECK(hipMemsetAsync(dtab, 0, (size_t)orange * 4, s));
-    HIP_CHECK(hipMemsetAsync(nhits, 0, 8, s));
     hipEvent_t ev0 = nullptr, ev1 = nullptr;
     if (kernel_ms) {
         HIP_CHECK(hipEventCreate(&ev0));
         HIP_CHECK(hipEventCreate(&ev1));
     }
-    hipLaunchKernelGGL(k_q9_part_bitmap, dim3(grid_for(p->n, 256)), dim3(256),
-                       0, s, *p, typemod, typeval, pbitmap);
     if (o->n > 0)
         hipLaunchKernelGGL(k_q9_odate_build, dim3(grid_for(o->n, 256)),
                            dim3(256), 0, s, *o, mino, dtab);
-    if (kernel_ms) HIP_CHECK(hipEventRecord(ev0, s));
-    hipLaunchKernelGGL(k_q9_filter, dim3(grid_for(l->n / 4, 256)), dim3(256),
-                       0, s, l->l_partkey, l->n, pbitmap, p->n, hits, nhits);
-    hipLaunchKernelGGL(k_q9_probe, dim3(grid_for(l->n / 8, 256)), dim3(256),
-                       0, s, *l, hits, nhits, dtab, mino, orange, sums_dev,
-                       (unsigned long long *)counts_dev);
+    bool rec0 = false;
+    for (int64_t pass = 0; pass < npasses; pass++) {
+        int64_t lo = pass * slice;
+        int64_t hi = lo + slice < p->n ? lo + slice : p->n;
+        HIP_CHECK(hipMemsetAsync(pbitmap, 0, bm_bytes, s));
+        HIP_CHECK(hipMemsetAsync(nhits, 0, 8, s));
+        hipLaunchKernelGGL(k_q9_part_bitmap, dim3(grid_for(p->n, 256)),
+                           dim3(256), 0, s, *p, typemod, typeval, lo, hi,
+                           pbitmap);
+        if (kernel_ms && !rec0) {
+            HIP_CHECK(hipEventRecord(ev0, s));
+            rec0 = true;
+        }
+        hipLaunchKernelGGL(k_q9_filter, dim3(grid_for(l->n / 4, 256)),
+                           dim3(256), 0, s, l->l_partkey, l->n, pbitmap, lo,
+                           hi, hits, nhits);
+        hipLaunchKernelGGL(k_q9_probe, dim3(grid_for(l->n / 8, 256)),
+                           dim3(256), 0, s, *l, hits, nhits, dtab, mino,
+                           orange, sums_dev, (unsigned long long *)counts_dev);
+    }
     HIP_CHECK(hipGetLastError());
     if (kernel_ms) {
         HIP_CHECK(hipEventRecord(ev1, s));

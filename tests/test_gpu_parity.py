@@ -266,6 +266,23 @@ def test_q9_typemod_variants(ex, ora):
             assert approx(r["sum_revenue"], g.revenue)
 
 
+def test_q9_bitmap_slice_multipass(ex, ora, monkeypatch):
+    """Grace-style part-bitmap slice passes (production: bitmap > the
+    ~3 MB L2 budget at SF300+; forced tiny here) must match the oracle."""
+    monkeypatch.setenv("OTBX_Q9_BITMAP_BITS", "4096")  # 400k parts -> 4 passes
+    n = 400000
+    li = ex.GpuLineitem.generate(n, with_partkey=True)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    pt = ex.GpuPart.generate(n // 30)
+    rows = drain(ex.GpuQ9Fragment(pt, od, li))
+    exp = ora.q9_partial(ora.gen_tables(n, need=("lineitem", "orders",
+                                                 "part")))
+    assert [r["o_year"] - 1992 for r in rows] == [g.year for g in exp]
+    for r, g in zip(rows, exp):
+        assert r["count_rows"] == g.count_rows
+        assert approx(r["sum_revenue"], g.revenue)
+
+
 def test_q9_sharded_merge_parity(ex, ora):
     """2-shard dense-state combine == full run (counts bit-exact, sums
     ≤1e-9 relative): the RemoteSubplan merge payload for Q9."""
