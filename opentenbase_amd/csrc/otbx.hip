@@ -1909,12 +1909,32 @@ extern "C" {
 
 static inline size_t align64_sz(size_t x) { return (x + 63) / 64 * 64; }
 
+/* Part-bitmap slice width in bits. Measured on MI355X: at
+ * SF300 a 3 MB L2-resident slice in 3 passes LOST to the single-pass
+ * 7.5 MB bitmap (27.95 vs 23.54 ms/step) — re-streaming 14.4 GB of
+ * l_partkey per extra pass costs more than the partially-spilled gathers
+ * it avoids. So the default cap is 1 Gbit (128 MB bitmap, ~17× SF300's
+ * part table): slicing only engages for part tables far beyond benchmark
+ * sizes, where the workspace itself would balloon. OTBX_Q9_BITMAP_BITS
+ * overrides (tests force multipass with tiny values). */
+static inline int64_t q9_slice_bits(void)
+{
+    int64_t cap_bits = 1ll << 30;
+    const char *cb = getenv("OTBX_Q9_BITMAP_BITS");
+    if (cb) {
+        int64_t e = atoll(cb);
+        if (e >= 64) cap_bits = e;
+    }
+    return cap_bits;
+}
+
 otbx_status otbx_q9_workspace_bytes(int64_t nparts, int64_t norders,
                                     int64_t nlineitem, uint32_t nranks,
                                     size_t *bytes)
 {
     int64_t orange = norders * (int64_t)(nranks ? nranks : 1);
-    int64_t slice = nparts < (24ll << 20) ? nparts : (24ll << 20);
+    int64_t cap = q9_slice_bits();
+    int64_t slice = nparts < cap ? nparts : cap;
     *bytes = align64_sz(8 * (size_t)((slice + 63) / 64)) /* part bitmap */ +
              align64_sz((size_t)orange * 4) /* odate direct table */ +
              64 /* hit counter */ +
@@ -1962,17 +1982,13 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
         mino = h_mm[0];
         orange = h_mm[1] - h_mm[0] + 1;
     }
-    /* keep the slice bitmap inside a per-XCD L2 (4 MB): a full SF300
-     * bitmap (7.5 MB) spilled the gathers to HBM. Grace-style partkey
-     * slices: each pass re-streams l_partkey (cheap) and gathers only the
-     * rows whose key falls in the slice (each row gathers in exactly one
-     * pass). Test hook: OTBX_Q9_BITMAP_BITS. */
-    int64_t cap_bits = 24ll << 20;
-    const char *cb = getenv("OTBX_Q9_BITMAP_BITS");
-    if (cb) {
-        int64_t e = atoll(cb);
-        if (e >= 64) cap_bits = e;
-    }
+    /* Grace-style partkey slices: each pass rebuilds the bitmap for one
+     * key range, re-streams l_partkey, and gathers only rows whose key
+     * falls in the slice (each row gathers in exactly one pass). With the
+     * default 1 Gbit cap (see q9_slice_bits) this is single-pass for any
+     * realistic part table; slicing engages only for huge key domains or
+     * under the OTBX_Q9_BITMAP_BITS test hook. */
+    int64_t cap_bits = q9_slice_bits();
     int64_t slice = p->n < cap_bits ? p->n : cap_bits;
     if (slice < 1) slice = 1; /* empty part table: one no-match pass */
     int64_t npasses = (p->n + slice - 1) / slice;
