@@ -1190,6 +1190,206 @@ __global__ void k_aggp_scatter_kv2(const ulonglong2 *__restrict__ recs,
     }
 }
 
+/* ---- tile-staged counting-sort partitioner (OTBX_PART_TILE) ----------
+ * Replaces the cursor scatters above (§8b.0; default ON, =0 restores
+ * them). A/B-measured in tools/microbench/scatter_ab.hip
+ * (profiles/r01_scatter_ab.txt): at 600 M rows the tile sort at fanout 256
+ * runs 7.30 ms vs 14.14 ms for the shipped cursor-scatter shape — the
+ * random 16-B record writes become coalesced per-bucket runs (avg 32
+ * records = 512 B at nb=256; fanout 2048 keeps runs at 4 records and loses
+ * the gain, hence the level-1 fanout drops to 256 and level-2 takes up to
+ * 2048). Per tile of 8192 rows: LDS histogram -> serial exclusive scan ->
+ * LDS-sorted stage -> one global cursor reservation per (tile, bucket) ->
+ * linear coalesced write-out. delta[] folds (segment run base - tile-local
+ * base) into one u32 per bucket (mod-2^32 arithmetic — requires n < 2^32,
+ * which the workspace sizes already imply). */
+#define PT_TILE 8192
+
+/* level 1: keys[] (+ optional knull skip list) -> (key, payload) records;
+ * vals == NULL emits the row index as payload (the join shape), else the
+ * double bits (the agg kv shape). nb ≤ 256. */
+__global__ __launch_bounds__(1024) void k_tile_scatter1(
+    const int64_t *__restrict__ keys, const uint8_t *__restrict__ knull,
+    const double *__restrict__ vals, int64_t n, uint32_t nb,
+    unsigned long long *__restrict__ cursor, ulonglong2 *__restrict__ recs)
+{
+    __shared__ ulonglong2 stage[PT_TILE];
+    __shared__ unsigned int hist[256], excl[256], delta[256];
+    __shared__ unsigned int tot;
+    int64_t ntiles = (n + PT_TILE - 1) / PT_TILE;
+    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+        int64_t lo = t * (int64_t)PT_TILE;
+        int tn = (int)(n - lo < PT_TILE ? n - lo : PT_TILE);
+        for (int j = threadIdx.x; j < (int)nb; j += blockDim.x) hist[j] = 0;
+        __syncthreads();
+        for (int i = threadIdx.x; i < tn; i += blockDim.x) {
+            if (knull && knull[lo + i]) continue;
+            atomicAdd(&hist[d_agg_bucket(keys[lo + i], nb)], 1u);
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            unsigned int acc = 0;
+            for (uint32_t j = 0; j < nb; j++) {
+                excl[j] = acc;
+                acc += hist[j];
+            }
+            tot = acc;
+        }
+        __syncthreads();
+        for (int j = threadIdx.x; j < (int)nb; j += blockDim.x) {
+            unsigned int rb =
+                hist[j] ? (unsigned int)atomicAdd(&cursor[j],
+                                                  (unsigned long long)hist[j])
+                        : 0u;
+            delta[j] = rb - excl[j]; /* mod 2^32 */
+            hist[j] = 0;             /* reused as the stage cursor */
+        }
+        __syncthreads();
+        for (int i = threadIdx.x; i < tn; i += blockDim.x) {
+            if (knull && knull[lo + i]) continue;
+            int64_t k = keys[lo + i];
+            uint32_t b = d_agg_bucket(k, nb);
+            unsigned int r = excl[b] + atomicAdd(&hist[b], 1u);
+            ulonglong2 rec;
+            rec.x = (unsigned long long)k;
+            rec.y = vals ? (unsigned long long)__double_as_longlong(
+                               vals[lo + i])
+                         : (unsigned long long)(lo + i);
+            stage[r] = rec;
+        }
+        __syncthreads();
+        int wtot = (int)tot;
+        for (int p = threadIdx.x; p < wtot; p += blockDim.x) {
+            ulonglong2 rec = stage[p];
+            uint32_t b = d_agg_bucket((int64_t)rec.x, nb);
+            recs[(size_t)(uint32_t)(delta[b] + (unsigned int)p)] = rec;
+        }
+        __syncthreads();
+    }
+}
+
+/* level-2 sub-bucket histogram: grid (blocks-per-segment, nb segments);
+ * cnts2 must be zeroed first */
+__global__ void k_tile_count2(const ulonglong2 *__restrict__ recs,
+                              const unsigned long long *__restrict__ offs,
+                              const unsigned long long *__restrict__ cnts,
+                              uint32_t nb2,
+                              unsigned long long *__restrict__ cnts2)
+{
+    __shared__ unsigned int h[AGGP_MAX_BUCKETS];
+    int sg = blockIdx.y;
+    int64_t lo = (int64_t)offs[sg];
+    int64_t hi = lo + (int64_t)cnts[sg];
+    for (int j = threadIdx.x; j < (int)nb2; j += blockDim.x) h[j] = 0;
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = lo + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < hi; i += stride)
+        atomicAdd(&h[d_agg_bucket2((int64_t)recs[i].x, nb2)], 1u);
+    __syncthreads();
+    for (int j = threadIdx.x; j < (int)nb2; j += blockDim.x)
+        if (h[j])
+            atomicAdd(&cnts2[(size_t)sg * nb2 + j], (unsigned long long)h[j]);
+}
+
+/* per-segment serial exclusive scan: offs2[s][j] = offs[s] + Σ cnts2[s][<j]
+ * (one block per segment; nb2 ≤ 2048 so a serial loop is cheap) */
+__global__ void k_tile_scan2(const unsigned long long *__restrict__ offs,
+                             uint32_t nb2,
+                             const unsigned long long *__restrict__ cnts2,
+                             unsigned long long *__restrict__ offs2)
+{
+    if (threadIdx.x != 0) return;
+    unsigned long long acc = offs[blockIdx.x];
+    for (uint32_t j = 0; j < nb2; j++) {
+        offs2[(size_t)blockIdx.x * nb2 + j] = acc;
+        acc += cnts2[(size_t)blockIdx.x * nb2 + j];
+    }
+}
+
+/* level 2: tile sort within each level-1 segment (records in, records
+ * out). cursor2 is offs2 itself, mutated by the run reservations —
+ * k_tile_restore_offs subtracts cnts2 afterwards to recover the segment
+ * starts for the bucket kernels. nb2 ≤ 2048. */
+__global__ __launch_bounds__(1024) void k_tile_scatter2(
+    const ulonglong2 *__restrict__ recs,
+    const unsigned long long *__restrict__ offs,
+    const unsigned long long *__restrict__ cnts, uint32_t nb2,
+    unsigned long long *__restrict__ cursor2, ulonglong2 *__restrict__ recs2)
+{
+    __shared__ ulonglong2 stage[PT_TILE];
+    __shared__ unsigned int hist[AGGP_MAX_BUCKETS];
+    __shared__ unsigned int excl[AGGP_MAX_BUCKETS];
+    __shared__ unsigned int delta[AGGP_MAX_BUCKETS];
+    int sg = blockIdx.y;
+    int64_t slo = (int64_t)offs[sg];
+    int64_t scnt = (int64_t)cnts[sg];
+    int64_t ntiles = (scnt + PT_TILE - 1) / PT_TILE;
+    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+        int64_t lo = slo + t * (int64_t)PT_TILE;
+        int tn = (int)(slo + scnt - lo < PT_TILE ? slo + scnt - lo : PT_TILE);
+        for (int j = threadIdx.x; j < (int)nb2; j += blockDim.x) hist[j] = 0;
+        __syncthreads();
+        for (int i = threadIdx.x; i < tn; i += blockDim.x)
+            atomicAdd(&hist[d_agg_bucket2((int64_t)recs[lo + i].x, nb2)], 1u);
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            unsigned int acc = 0;
+            for (uint32_t j = 0; j < nb2; j++) {
+                excl[j] = acc;
+                acc += hist[j];
+            }
+        }
+        __syncthreads();
+        for (int j = threadIdx.x; j < (int)nb2; j += blockDim.x) {
+            unsigned int rb =
+                hist[j]
+                    ? (unsigned int)atomicAdd(&cursor2[(size_t)sg * nb2 + j],
+                                              (unsigned long long)hist[j])
+                    : 0u;
+            delta[j] = rb - excl[j]; /* mod 2^32 */
+            hist[j] = 0;
+        }
+        __syncthreads();
+        for (int i = threadIdx.x; i < tn; i += blockDim.x) {
+            ulonglong2 rec = recs[lo + i];
+            uint32_t b = d_agg_bucket2((int64_t)rec.x, nb2);
+            unsigned int r = excl[b] + atomicAdd(&hist[b], 1u);
+            stage[r] = rec;
+        }
+        __syncthreads();
+        for (int p = threadIdx.x; p < tn; p += blockDim.x) {
+            ulonglong2 rec = stage[p];
+            uint32_t b = d_agg_bucket2((int64_t)rec.x, nb2);
+            recs2[(size_t)(uint32_t)(delta[b] + (unsigned int)p)] = rec;
+        }
+        __syncthreads();
+    }
+}
+
+__global__ void k_tile_restore_offs(unsigned long long *__restrict__ offs2,
+                                    const unsigned long long *__restrict__
+                                        cnts2,
+                                    int64_t n)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        offs2[i] -= cnts2[i];
+}
+
+/* default ON (measured on MI355X, gpurun_out/tile_ops_bench.txt →
+ * profiles/r01_tile_ops.txt: agg-100M 29.4→22.8 ms, join-150M 38.7→34.2,
+ * join-15M 28.8→23.2, agg-1M 19.9→18.5, all parity-green);
+ * OTBX_PART_TILE=0 selects the legacy cursor scatters (kept as the
+ * fallback for NULL-carrying agg inputs and n ≥ 2^32, and pinned by the
+ * *_legacy_parity tests) */
+static inline bool part_tile_enabled(void)
+{
+    const char *e = getenv("OTBX_PART_TILE");
+    return !e || atoi(e);
+}
+
 /* direct-emit per-bucket aggregation: two-level buckets hold DISJOINT key
  * sets, so a bucket that fits its LDS table emits its groups straight to
  * the output — no shared global table, no cap-sized init or compact scan
@@ -1532,10 +1732,16 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
         partitioned = est > 1536 && est <= est_cap;
     }
 
+    /* tile-sort partitioner (§8b.0, default ON — see part_tile_enabled):
+     * level-1 fanout drops to 256 so the per-bucket write runs coalesce;
+     * level-2 fanout grows to compensate (same total nb*nb2). */
+    bool tile = part_tile_enabled() && !knull && !vnull &&
+                n < (int64_t)UINT32_MAX;
     uint32_t nb = 0, nb2 = 1;
     if (partitioned) {
         nb = (uint32_t)next_pow2_host(est / 256 < 64 ? 64 : est / 256);
         if (nb > AGGP_MAX_BUCKETS) nb = AGGP_MAX_BUCKETS;
+        if (tile && nb > 256) nb = 256;
         int64_t per_b = est / nb;
         if (!knull && !vnull && per_b > 768) {
             nb2 = (uint32_t)next_pow2_host(per_b / 384 + 1);
@@ -1592,9 +1798,14 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
             /* fast path: scatter (key,val) records directly — no index
              * gathers */
             ulonglong2 *recs = (ulonglong2 *)perm; /* n×16 ≤ region */
-            hipLaunchKernelGGL(k_aggp_scatter_kv, dim3(grid_for(n, 256)),
-                               dim3(256), 0, s, keys, vals, n, nb, cursor,
-                               recs);
+            if (tile)
+                hipLaunchKernelGGL(k_tile_scatter1, dim3(2048), dim3(1024), 0,
+                                   s, keys, nullptr, vals, n, nb, cursor,
+                                   recs);
+            else
+                hipLaunchKernelGGL(k_aggp_scatter_kv, dim3(grid_for(n, 256)),
+                                   dim3(256), 0, s, keys, vals, n, nb, cursor,
+                                   recs);
             if (nr > 0) {
                 static unsigned int *d_nflag = nullptr;
                 static unsigned int *h_nflag = nullptr;
@@ -1606,9 +1817,26 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
                 const unsigned long long *foffs = offs, *fcnts = cnts;
                 uint32_t fgrid = nb;
                 if (nb2 > 1) { /* second partition level */
-                    hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nb),
-                                       dim3(256), 0, s, recs, offs, cnts, nb2,
-                                       recs2, offs2, cnts2);
+                    if (tile) {
+                        HIP_CHECK(hipMemsetAsync(cnts2, 0,
+                                                 (size_t)nb * nb2 * 8, s));
+                        hipLaunchKernelGGL(k_tile_count2, dim3(16, nb),
+                                           dim3(256), 0, s, recs, offs, cnts,
+                                           nb2, cnts2);
+                        hipLaunchKernelGGL(k_tile_scan2, dim3(nb), dim3(64),
+                                           0, s, offs, nb2, cnts2, offs2);
+                        hipLaunchKernelGGL(k_tile_scatter2, dim3(8, nb),
+                                           dim3(1024), 0, s, recs, offs, cnts,
+                                           nb2, offs2, recs2);
+                        hipLaunchKernelGGL(
+                            k_tile_restore_offs,
+                            dim3(grid_for((int64_t)nb * nb2, 256)), dim3(256),
+                            0, s, offs2, cnts2, (int64_t)nb * nb2);
+                    } else {
+                        hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nb),
+                                           dim3(256), 0, s, recs, offs, cnts,
+                                           nb2, recs2, offs2, cnts2);
+                    }
                     frecs = recs2;
                     foffs = offs2;
                     fcnts = cnts2;
@@ -2870,7 +3098,9 @@ otbx_status otbx_join_i64(const int64_t *bkeys, const uint8_t *bnull, int64_t nb
     if ((nb >= JOINP_THRESHOLD || (fj && atoi(fj))) && np > 0 && nb > 0) {
         /* partitioned path: bucket count from build ROWS (duplicates
          * occupy chain entries), aiming ≤ ~384 rows per final bucket */
-        uint32_t nbuk = AGGP_MAX_BUCKETS;
+        bool tile = part_tile_enabled() && nb < (int64_t)UINT32_MAX &&
+                    np < (int64_t)UINT32_MAX;
+        uint32_t nbuk = tile ? 256 : AGGP_MAX_BUCKETS;
         int64_t per_b = nb / nbuk;
         uint32_t nb2 = 1;
         if (per_b > 384) {
@@ -2924,23 +3154,61 @@ otbx_status otbx_join_i64(const int64_t *bkeys, const uint8_t *bnull, int64_t nb
                                  hipMemcpyHostToDevice, s));
         HIP_CHECK(hipMemcpyAsync(cursor_p, h_op, (size_t)nbuk * 8,
                                  hipMemcpyHostToDevice, s));
-        hipLaunchKernelGGL(k_joinp_scatter, dim3(grid_for(nb, 256)),
-                           dim3(256), 0, s, bkeys, bnull, nb, nbuk, cursor_b,
-                           brecs);
-        hipLaunchKernelGGL(k_joinp_scatter, dim3(grid_for(np, 256)),
-                           dim3(256), 0, s, pkeys, pnull, np, nbuk, cursor_p,
-                           precs);
+        if (tile) {
+            hipLaunchKernelGGL(k_tile_scatter1, dim3(2048), dim3(1024), 0, s,
+                               bkeys, bnull, nullptr, nb, nbuk, cursor_b,
+                               brecs);
+            hipLaunchKernelGGL(k_tile_scatter1, dim3(2048), dim3(1024), 0, s,
+                               pkeys, pnull, nullptr, np, nbuk, cursor_p,
+                               precs);
+        } else {
+            hipLaunchKernelGGL(k_joinp_scatter, dim3(grid_for(nb, 256)),
+                               dim3(256), 0, s, bkeys, bnull, nb, nbuk,
+                               cursor_b, brecs);
+            hipLaunchKernelGGL(k_joinp_scatter, dim3(grid_for(np, 256)),
+                               dim3(256), 0, s, pkeys, pnull, np, nbuk,
+                               cursor_p, precs);
+        }
         const ulonglong2 *fb = brecs, *fp = precs;
         const unsigned long long *fbo = offs_b, *fbc = cnts_b;
         const unsigned long long *fpo = offs_p, *fpc = cnts_p;
         uint32_t fgrid = nbuk;
         if (nb2 > 1) {
-            hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nbuk), dim3(256), 0,
-                               s, brecs, offs_b, cnts_b, nb2, brecs2, boffs2,
-                               bcnts2);
-            hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nbuk), dim3(256), 0,
-                               s, precs, offs_p, cnts_p, nb2, precs2, poffs2,
-                               pcnts2);
+            if (tile) {
+                HIP_CHECK(hipMemsetAsync(bcnts2, 0, (size_t)nbuk * nb2 * 8,
+                                         s));
+                HIP_CHECK(hipMemsetAsync(pcnts2, 0, (size_t)nbuk * nb2 * 8,
+                                         s));
+                hipLaunchKernelGGL(k_tile_count2, dim3(16, nbuk), dim3(256),
+                                   0, s, brecs, offs_b, cnts_b, nb2, bcnts2);
+                hipLaunchKernelGGL(k_tile_scan2, dim3(nbuk), dim3(64), 0, s,
+                                   offs_b, nb2, bcnts2, boffs2);
+                hipLaunchKernelGGL(k_tile_scatter2, dim3(8, nbuk), dim3(1024),
+                                   0, s, brecs, offs_b, cnts_b, nb2, boffs2,
+                                   brecs2);
+                hipLaunchKernelGGL(k_tile_restore_offs,
+                                   dim3(grid_for((int64_t)nbuk * nb2, 256)),
+                                   dim3(256), 0, s, boffs2, bcnts2,
+                                   (int64_t)nbuk * nb2);
+                hipLaunchKernelGGL(k_tile_count2, dim3(16, nbuk), dim3(256),
+                                   0, s, precs, offs_p, cnts_p, nb2, pcnts2);
+                hipLaunchKernelGGL(k_tile_scan2, dim3(nbuk), dim3(64), 0, s,
+                                   offs_p, nb2, pcnts2, poffs2);
+                hipLaunchKernelGGL(k_tile_scatter2, dim3(8, nbuk), dim3(1024),
+                                   0, s, precs, offs_p, cnts_p, nb2, poffs2,
+                                   precs2);
+                hipLaunchKernelGGL(k_tile_restore_offs,
+                                   dim3(grid_for((int64_t)nbuk * nb2, 256)),
+                                   dim3(256), 0, s, poffs2, pcnts2,
+                                   (int64_t)nbuk * nb2);
+            } else {
+                hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nbuk), dim3(256),
+                                   0, s, brecs, offs_b, cnts_b, nb2, brecs2,
+                                   boffs2, bcnts2);
+                hipLaunchKernelGGL(k_aggp_scatter_kv2, dim3(nbuk), dim3(256),
+                                   0, s, precs, offs_p, cnts_p, nb2, precs2,
+                                   poffs2, pcnts2);
+            }
             fb = brecs2;
             fbo = boffs2;
             fbc = bcnts2;

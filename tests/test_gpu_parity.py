@@ -751,6 +751,75 @@ def test_agg_partitioned_path_parity(ex, ora):
                 1e-9 * max(abs(e.sum_v), 1e-6)
 
 
+def test_agg_partitioned_tile_level2_parity(ex, ora):
+    """The tile-staged counting-sort partitioner (default path, §8b.0,
+    A/B in profiles/r01_scatter_ab.txt) through BOTH partition levels
+    (600k distinct -> nb=256, nb2=8), incl. sentinel keys. Asserts sample
+    the group map (the full 600k-iteration Python loop was the slow part,
+    not the GPU) plus count-conservation invariants."""
+    rng = np.random.default_rng(41)
+    n = 12_000_000
+    keys = rng.integers(0, 600_000, n)
+    keys[::1_000_000] = -(2**63)
+    vals = rng.random(n) * 1e3
+    got = _agg(ex, keys, vals)
+    exp = ora.agg_i64(keys, vals)
+    assert len(got) == len(exp)
+    assert sum(g["count_star"] for g in got) == n
+    gm = {(int(g["key"]), bool(g["key_isnull"])): g for g in got}
+    for e in exp[::29]:  # ~21k sampled groups
+        g = gm[(e.key if not e.key_isnull else 0, bool(e.key_isnull))]
+        assert g["count_star"] == e.count_star
+        assert g["count_v"] == e.count_v
+        assert abs(float(g["sum_v"]) - e.sum_v) <= 1e-9 * abs(e.sum_v)
+
+
+def test_agg_partitioned_legacy_parity(ex, ora, monkeypatch):
+    """OTBX_PART_TILE=0 pins the legacy cursor-scatter partitioner (the
+    fallback for NULL-carrying inputs and n >= 2^32) on the same two-level
+    shape."""
+    monkeypatch.setenv("OTBX_PART_TILE", "0")
+    rng = np.random.default_rng(43)
+    n = 10_000_000
+    keys = rng.integers(0, 600_000, n)
+    vals = rng.random(n) * 1e3
+    got = _agg(ex, keys, vals)
+    exp = ora.agg_i64(keys, vals)
+    assert len(got) == len(exp)
+    assert sum(g["count_star"] for g in got) == n
+    gm = {int(g["key"]): g for g in got}
+    for e in exp[::29]:
+        g = gm[e.key]
+        assert g["count_star"] == e.count_star
+        assert abs(float(g["sum_v"]) - e.sum_v) <= 1e-9 * abs(e.sum_v)
+
+
+def test_join_partitioned_legacy_parity(ex, ora, monkeypatch):
+    """OTBX_PART_TILE=0 pins the legacy k_joinp_scatter/kv2 partitioner;
+    the default-path equivalent (tile scatter with row-index payload and
+    NULL skip list) is covered by test_join_partitioned_parity and
+    test_join_partitioned_large_count above. Sentinel keys still flag
+    buckets to the global fallback."""
+    monkeypatch.setenv("OTBX_JOINP_FORCE", "1")
+    monkeypatch.setenv("OTBX_PART_TILE", "0")
+    rng = np.random.default_rng(19)
+    nb, npr = 300000, 900000
+    bk = rng.integers(0, 200000, nb)
+    pk = rng.integers(0, 200000, npr)
+    bk[::5000] = np.iinfo(np.int64).min
+    pk[::7000] = np.iinfo(np.int64).min
+    bn = (rng.random(nb) < 0.03).astype(np.uint8)
+    pn = (rng.random(npr) < 0.03).astype(np.uint8)
+    dev = lambda a, dt: torch.as_tensor(a, dtype=dt, device="cuda")  # noqa: E731
+    node = ex.GpuHashJoin(dev(bk, torch.int64), dev(pk, torch.int64),
+                          dev(bn, torch.uint8), dev(pn, torch.uint8),
+                          cap_pairs=8 * npr)
+    pairs = drain(node)
+    obi, opi = ora.join_i64(bk, pk, bnull=bn, pnull=pn)
+    assert len(pairs) == len(obi)
+    assert sorted(pairs) == sorted(zip(obi.tolist(), opi.tolist()))
+
+
 def test_host_staging_path(ex, ora):
     """otbx_stage_table flow: oracle-generated HOST columns staged through
     otbx_memcpy_h2d, then the Q1 fragment — exercises the real provider

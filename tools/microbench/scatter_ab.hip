@@ -1,0 +1,359 @@
+/* Microbench: partition-scatter strategies for the two-level partitioned
+ * agg/join (DESIGN.md §8b priority 0). The product's level-1 scatter
+ * (k_aggp_scatter_kv shape: read keys+vals, write 16-B (key,val) records
+ * into per-bucket segments via global cursors) runs at ~4x its stream
+ * floor — random 16-B writes across 2048 bucket cursors. Candidates keep
+ * the same output contract (bucket-contiguous segments, any order within
+ * a bucket) and vary the write pattern:
+ *
+ *   v0  stream floor: read k+v, write rec at same index (no partition)
+ *   v1  product shape: per-block count -> cursor reserve -> random 16-B
+ *       scatter; nb=2048, blockDim 256 (what otbx_agg_i64/join ship today)
+ *   v2  v1 at nb=256 (fewer cursors, longer L2 write-combining runs)
+ *   v3  tile-staged LDS counting sort, nb=256, T=4096, blockDim 512
+ *       (2 wg/CU): per tile: hist -> scan -> LDS-sorted stage -> reserve
+ *       runs -> coalesced run writes (avg run 16 recs = 256 B)
+ *   v4  tile-staged, nb=2048, T=8192, blockDim 1024, ~152 KB LDS
+ *       (1 wg/CU; avg run 4 recs = 64-B full lines)
+ *   v5  tile-staged, nb=256, T=8192, blockDim 1024 (avg run 32 recs)
+ *   v6  v1 at nb=2048 but blockDim 1024 (bigger per-block bucket runs)
+ *
+ * Validation per variant: every output record lands in its own bucket's
+ * segment, and the key-sum (mod 2^64) is conserved.
+ *
+ * Build: hipcc --offload-arch=gfx950 -O3 scatter_ab.hip -o scatter_ab
+ */
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include <stdio.h>
+
+#define WAVE 64
+typedef unsigned long long u64;
+typedef unsigned int u32;
+
+#define CHK(x)                                                              \
+    do {                                                                    \
+        hipError_t e_ = (x);                                                \
+        if (e_ != hipSuccess) {                                             \
+            fprintf(stderr, "HIP error %s at line %d\n",                    \
+                    hipGetErrorString(e_), __LINE__);                       \
+            return 1;                                                       \
+        }                                                                   \
+    } while (0)
+
+__device__ __host__ __forceinline__ u64 smix(u64 x)
+{
+    x += 0x9e3779b97f4a7c15ull;
+    x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ull;
+    x = (x ^ (x >> 27)) * 0x94d049bb133111ebull;
+    return x ^ (x >> 31);
+}
+
+/* same bucket bits as the product (d_agg_bucket: splitmix >> 40) */
+__device__ __forceinline__ u32 bucket_of(long long k, u32 nb)
+{
+    return (u32)((smix((u64)k) >> 40) & (u64)(nb - 1));
+}
+
+__global__ void k_init(long long *keys, double *vals, int64_t n)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        keys[i] = (long long)smix((u64)i); /* uniform 64-bit key domain */
+        vals[i] = (double)(i & 0xffff);
+    }
+}
+
+__global__ void k_keysum(const long long *keys, int64_t n, u64 *sum)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    u64 my = 0;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        my += (u64)keys[i];
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        my += __shfl_down(my, off, WAVE);
+    if ((threadIdx.x % WAVE) == 0 && my) atomicAdd(sum, my);
+}
+
+/* global per-bucket histogram (the product's k_aggp_count shape) */
+__global__ void k_count(const long long *__restrict__ keys, int64_t n, u32 nb,
+                        u64 *__restrict__ cnts)
+{
+    __shared__ u32 h[2048];
+    for (int j = threadIdx.x; j < (int)nb; j += blockDim.x) h[j] = 0;
+    __syncthreads();
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        atomicAdd(&h[bucket_of(keys[i], nb)], 1u);
+    __syncthreads();
+    for (int j = threadIdx.x; j < (int)nb; j += blockDim.x)
+        if (h[j]) atomicAdd(&cnts[j], (u64)h[j]);
+}
+
+/* v0: no partition — the 32 B/row stream floor */
+__global__ void k_v0(const long long *__restrict__ keys,
+                     const double *__restrict__ vals, int64_t n,
+                     ulonglong2 *__restrict__ out)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        ulonglong2 r;
+        r.x = (u64)keys[i];
+        r.y = (u64)__double_as_longlong(vals[i]);
+        out[i] = r;
+    }
+}
+
+/* v1/v2/v6: the product's cursor scatter (count chunk, reserve per-bucket
+ * segment, random 16-B writes) */
+__global__ void k_cursor_scatter(const long long *__restrict__ keys,
+                                 const double *__restrict__ vals, int64_t n,
+                                 u32 nb, u64 *__restrict__ cursor,
+                                 ulonglong2 *__restrict__ out)
+{
+    __shared__ u32 lcur[2048];
+    __shared__ long long base[2048];
+    int64_t per_block = (n + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < n ? lo + per_block : n;
+    for (int j = threadIdx.x; j < (int)nb; j += blockDim.x) lcur[j] = 0;
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        atomicAdd(&lcur[bucket_of(keys[i], nb)], 1u);
+    __syncthreads();
+    for (int j = threadIdx.x; j < (int)nb; j += blockDim.x) {
+        base[j] = lcur[j] ? (long long)atomicAdd(&cursor[j], (u64)lcur[j]) : 0;
+        lcur[j] = 0;
+    }
+    __syncthreads();
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        long long k = keys[i];
+        u32 b = bucket_of(k, nb);
+        u32 off = atomicAdd(&lcur[b], 1u);
+        ulonglong2 r;
+        r.x = (u64)k;
+        r.y = (u64)__double_as_longlong(vals[i]);
+        out[(int64_t)base[b] + off] = r;
+    }
+}
+
+/* v3/v4/v5: tile-staged counting sort. Per tile of T rows: LDS histogram,
+ * serial exclusive scan, LDS-sorted stage, per-bucket global run
+ * reservation, then a coalesced linear write-out — consecutive threads in
+ * a bucket run write consecutive global addresses. delta[] folds
+ * (runbase - excl) so write-out needs one u32 per bucket (mod-2^32
+ * arithmetic; valid while n < 2^32). */
+template <int T, int NB>
+__global__ void k_tile_sort(const long long *__restrict__ keys,
+                            const double *__restrict__ vals, int64_t n,
+                            u64 *__restrict__ cursor,
+                            ulonglong2 *__restrict__ out)
+{
+    __shared__ ulonglong2 stage[T];
+    __shared__ u32 hist[NB]; /* reused as the scatter cursor after reserve */
+    __shared__ u32 excl[NB];
+    __shared__ u32 delta[NB];
+    int64_t ntiles = (n + T - 1) / T;
+    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+        int64_t lo = t * (int64_t)T;
+        int tn = (int)(n - lo < T ? n - lo : T);
+        for (int j = threadIdx.x; j < NB; j += blockDim.x) hist[j] = 0;
+        __syncthreads();
+        for (int i = threadIdx.x; i < tn; i += blockDim.x)
+            atomicAdd(&hist[bucket_of(keys[lo + i], NB)], 1u);
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            u32 acc = 0;
+            for (int j = 0; j < NB; j++) {
+                excl[j] = acc;
+                acc += hist[j];
+            }
+        }
+        __syncthreads();
+        for (int j = threadIdx.x; j < NB; j += blockDim.x) {
+            u32 rb = hist[j] ? (u32)atomicAdd(&cursor[j], (u64)hist[j]) : 0u;
+            delta[j] = rb - excl[j]; /* mod 2^32 */
+            hist[j] = 0;             /* becomes the stage cursor */
+        }
+        __syncthreads();
+        for (int i = threadIdx.x; i < tn; i += blockDim.x) {
+            long long k = keys[lo + i];
+            u32 b = bucket_of(k, NB);
+            u32 r = excl[b] + atomicAdd(&hist[b], 1u);
+            ulonglong2 rec;
+            rec.x = (u64)k;
+            rec.y = (u64)__double_as_longlong(vals[lo + i]);
+            stage[r] = rec;
+        }
+        __syncthreads();
+        for (int p = threadIdx.x; p < tn; p += blockDim.x) {
+            ulonglong2 rec = stage[p];
+            u32 b = bucket_of((long long)rec.x, NB);
+            out[(u32)(delta[b] + (u32)p)] = rec;
+        }
+        __syncthreads();
+    }
+}
+
+/* validation: each record sits in its own bucket's segment; key-sum
+ * conserved */
+__global__ void k_check(const ulonglong2 *__restrict__ out,
+                        const u64 *__restrict__ offs, u32 nb, int64_t n,
+                        u64 *errs, u64 *sum)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    u64 my = 0, myerr = 0;
+    for (int64_t b = blockIdx.y; b < nb; b += gridDim.y) {
+        int64_t lo = (int64_t)offs[b];
+        int64_t hi = b + 1 < nb ? (int64_t)offs[b + 1] : n;
+        for (int64_t i = lo + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+             i < hi; i += stride) {
+            ulonglong2 r = out[i];
+            my += r.x;
+            myerr += bucket_of((long long)r.x, nb) != (u32)b;
+        }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        my += __shfl_down(my, off, WAVE);
+        myerr += __shfl_down(myerr, off, WAVE);
+    }
+    if ((threadIdx.x % WAVE) == 0) {
+        if (my) atomicAdd(sum, my);
+        if (myerr) atomicAdd(errs, myerr);
+    }
+}
+
+int main(int argc, char **argv)
+{
+    int64_t n = 600000000;
+    int reps = 4;
+    if (argc > 1) n = atoll(argv[1]);
+    long long *keys;
+    double *vals;
+    ulonglong2 *out;
+    u64 *cnts256, *cnts2048, *offs256, *offs2048, *cursor, *scal;
+    CHK(hipMalloc(&keys, n * 8));
+    CHK(hipMalloc(&vals, n * 8));
+    CHK(hipMalloc(&out, n * 16));
+    CHK(hipMalloc(&cnts256, 256 * 8));
+    CHK(hipMalloc(&cnts2048, 2048 * 8));
+    CHK(hipMalloc(&offs256, 256 * 8));
+    CHK(hipMalloc(&offs2048, 2048 * 8));
+    CHK(hipMalloc(&cursor, 2048 * 8));
+    CHK(hipMalloc(&scal, 3 * 8)); /* keysum_in, errs, keysum_out */
+    hipLaunchKernelGGL(k_init, dim3(4096), dim3(256), 0, 0, keys, vals, n);
+    CHK(hipMemset(scal, 0, 24));
+    hipLaunchKernelGGL(k_keysum, dim3(2048), dim3(256), 0, 0, keys, n, scal);
+    CHK(hipMemset(cnts256, 0, 256 * 8));
+    CHK(hipMemset(cnts2048, 0, 2048 * 8));
+    hipLaunchKernelGGL(k_count, dim3(2048), dim3(256), 0, 0, keys, n, 256,
+                       cnts256);
+    hipLaunchKernelGGL(k_count, dim3(2048), dim3(256), 0, 0, keys, n, 2048,
+                       cnts2048);
+    /* host exclusive scans */
+    static u64 h_c[2048], h_o[2048];
+    u64 h_keysum = 0;
+    CHK(hipMemcpy(&h_keysum, scal, 8, hipMemcpyDeviceToHost));
+    CHK(hipMemcpy(h_c, cnts256, 256 * 8, hipMemcpyDeviceToHost));
+    u64 acc = 0;
+    for (int j = 0; j < 256; j++) {
+        h_o[j] = acc;
+        acc += h_c[j];
+    }
+    CHK(hipMemcpy(offs256, h_o, 256 * 8, hipMemcpyHostToDevice));
+    CHK(hipMemcpy(h_c, cnts2048, 2048 * 8, hipMemcpyDeviceToHost));
+    acc = 0;
+    for (int j = 0; j < 2048; j++) {
+        h_o[j] = acc;
+        acc += h_c[j];
+    }
+    CHK(hipMemcpy(offs2048, h_o, 2048 * 8, hipMemcpyHostToDevice));
+
+    const char *names[] = {
+        "v0 stream floor (no partition)          ",
+        "v1 cursor scatter nb=2048 bd=256 (prod) ",
+        "v2 cursor scatter nb=256  bd=256        ",
+        "v3 tile sort nb=256  T=4096 bd=512      ",
+        "v4 tile sort nb=2048 T=8192 bd=1024     ",
+        "v5 tile sort nb=256  T=8192 bd=1024     ",
+        "v6 cursor scatter nb=2048 bd=1024       ",
+    };
+    hipEvent_t e0, e1;
+    CHK(hipEventCreate(&e0));
+    CHK(hipEventCreate(&e1));
+    for (int v = 0; v < 7; v++) {
+        u32 nb = (v == 2 || v == 3 || v == 5) ? 256 : 2048;
+        u64 *offs = nb == 256 ? offs256 : offs2048;
+        float best = 1e9f;
+        for (int r = 0; r < reps; r++) {
+            CHK(hipMemcpy(cursor, offs, nb * 8, hipMemcpyDeviceToDevice));
+            CHK(hipMemset(out, 0, n > 16 ? 256 : n * 16)); /* touch only */
+            CHK(hipDeviceSynchronize());
+            CHK(hipEventRecord(e0));
+            switch (v) {
+            case 0:
+                hipLaunchKernelGGL(k_v0, dim3(2048), dim3(256), 0, 0, keys,
+                                   vals, n, out);
+                break;
+            case 1:
+                hipLaunchKernelGGL(k_cursor_scatter, dim3(2048), dim3(256), 0,
+                                   0, keys, vals, n, 2048, cursor, out);
+                break;
+            case 2:
+                hipLaunchKernelGGL(k_cursor_scatter, dim3(2048), dim3(256), 0,
+                                   0, keys, vals, n, 256, cursor, out);
+                break;
+            case 3:
+                hipLaunchKernelGGL((k_tile_sort<4096, 256>), dim3(4096),
+                                   dim3(512), 0, 0, keys, vals, n, cursor,
+                                   out);
+                break;
+            case 4:
+                hipLaunchKernelGGL((k_tile_sort<8192, 2048>), dim3(2048),
+                                   dim3(1024), 0, 0, keys, vals, n, cursor,
+                                   out);
+                break;
+            case 5:
+                hipLaunchKernelGGL((k_tile_sort<8192, 256>), dim3(2048),
+                                   dim3(1024), 0, 0, keys, vals, n, cursor,
+                                   out);
+                break;
+            case 6:
+                hipLaunchKernelGGL(k_cursor_scatter, dim3(512), dim3(1024), 0,
+                                   0, keys, vals, n, 2048, cursor, out);
+                break;
+            }
+            CHK(hipEventRecord(e1));
+            CHK(hipEventSynchronize(e1));
+            CHK(hipGetLastError());
+            float ms;
+            CHK(hipEventElapsedTime(&ms, e0, e1));
+            if (ms < best) best = ms;
+        }
+        /* validate the last rep's output */
+        CHK(hipMemset(scal + 1, 0, 16));
+        if (v == 0) {
+            /* v0 keeps input order: just key-sum via segments of one big
+             * bucket — check with nb=1 semantics: skip homogeneity */
+            hipLaunchKernelGGL(k_keysum, dim3(2048), dim3(256), 0, 0,
+                               (const long long *)out, 2 * n,
+                               scal + 2); /* sums keys+vals interleaved */
+            CHK(hipDeviceSynchronize());
+            printf("%s %8.3f ms  %7.0f GB/s  (order-preserving)\n", names[v],
+                   best, 32.0 * n / best / 1e6);
+            continue;
+        }
+        hipLaunchKernelGGL(k_check, dim3(32, 64), dim3(256), 0, 0, out, offs,
+                           nb, n, scal + 1, scal + 2);
+        u64 h_s[3];
+        CHK(hipMemcpy(h_s, scal, 24, hipMemcpyDeviceToHost));
+        printf("%s %8.3f ms  %7.0f GB/s  errs=%llu sum%s\n", names[v], best,
+               32.0 * n / best / 1e6, (unsigned long long)h_s[1],
+               h_s[2] == h_keysum ? "=ok" : "=MISMATCH");
+    }
+    return 0;
+}
