@@ -1,0 +1,87 @@
+"""Property-based pinning of the CPU oracle against brute-force Python
+restatements (hypothesis). The oracle is the parity anchor for the GPU path
+(SURVEY §8c); these tests guard the anchor itself on adversarial small
+inputs: duplicate / negative / INT64_MIN keys, NULL masks on keys and
+values, empty inputs, and float accumulation in scan order (reference
+semantics: nodeAgg.c:856 advance_aggregates via int8inc int8.c:714 and
+float8pl float.c:970 — strict transfns skip NULL inputs, NULL keys form one
+group, inner-join NULL keys never match)."""
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+from oracle import oracle_py as ora
+
+I64_MIN = -(2**63)
+
+small_key = st.integers(min_value=-4, max_value=4)
+wild_key = st.one_of(small_key,
+                     st.sampled_from([I64_MIN, I64_MIN + 1, 2**63 - 1]))
+val = st.floats(min_value=-1e12, max_value=1e12,
+                allow_nan=False, width=64)
+
+
+@st.composite
+def agg_input(draw):
+    n = draw(st.integers(min_value=0, max_value=80))
+    keys = np.array(draw(st.lists(wild_key, min_size=n, max_size=n)),
+                    dtype=np.int64)
+    vals = np.array(draw(st.lists(val, min_size=n, max_size=n)),
+                    dtype=np.float64)
+    kn = np.array(draw(st.lists(st.booleans(), min_size=n, max_size=n)),
+                  dtype=np.uint8)
+    vn = np.array(draw(st.lists(st.booleans(), min_size=n, max_size=n)),
+                  dtype=np.uint8)
+    return keys, vals, kn, vn
+
+
+@settings(max_examples=80, deadline=None, derandomize=True)
+@given(agg_input())
+def test_agg_matches_bruteforce(inp):
+    keys, vals, kn, vn = inp
+    got = ora.agg_i64(keys, vals, key_null=kn, val_null=vn)
+    # brute force in SCAN ORDER (float sums must match bit-exactly: the
+    # oracle restates the reference's tuple-at-a-time accumulation)
+    bf = {}  # (key, isnull) -> [count_star, count_v, sum_v]
+    for i in range(len(keys)):
+        gk = (0, True) if kn[i] else (int(keys[i]), False)
+        g = bf.setdefault(gk, [0, 0, 0.0])
+        g[0] += 1
+        if not vn[i]:
+            g[1] += 1
+            g[2] += float(vals[i])
+    assert len(got) == len(bf)
+    for g in got:
+        e = bf[(int(g.key) if not g.key_isnull else 0, bool(g.key_isnull))]
+        assert g.count_star == e[0]
+        assert g.count_v == e[1]
+        if e[1]:
+            assert g.sum_v == e[2]  # bit-exact: same accumulation order
+        else:
+            assert g.sum_isnull  # SUM over all-NULL = NULL
+
+
+@st.composite
+def join_input(draw):
+    nb = draw(st.integers(min_value=0, max_value=40))
+    npr = draw(st.integers(min_value=0, max_value=60))
+    bk = np.array(draw(st.lists(wild_key, min_size=nb, max_size=nb)),
+                  dtype=np.int64)
+    pk = np.array(draw(st.lists(wild_key, min_size=npr, max_size=npr)),
+                  dtype=np.int64)
+    bn = np.array(draw(st.lists(st.booleans(), min_size=nb, max_size=nb)),
+                  dtype=np.uint8)
+    pn = np.array(draw(st.lists(st.booleans(), min_size=npr, max_size=npr)),
+                  dtype=np.uint8)
+    return bk, pk, bn, pn
+
+
+@settings(max_examples=80, deadline=None, derandomize=True)
+@given(join_input())
+def test_join_matches_bruteforce(inp):
+    bk, pk, bn, pn = inp
+    bi, pi = ora.join_i64(bk, pk, bnull=bn, pnull=pn)
+    exp = sorted((i, j)
+                 for i in range(len(bk)) if not bn[i]
+                 for j in range(len(pk)) if not pn[j]
+                 if bk[i] == pk[j])
+    assert sorted(zip(bi.tolist(), pi.tolist())) == exp

@@ -99,6 +99,17 @@ def main():
             # select_having.out:16-22: GROUP BY b,c HAVING count(*)=1
             "groups_count1": [[1, "XXXX"], [3, "bbbb"]],
         },
+        "test_vec": {
+            # opentenbase_c_aggregation.out:1-139 — the distributed
+            # two-phase (Partial on DN -> Finalize on CN) aggregate plans
+            # over test_vec = generate_series(-10,15), int2/int4/int8 and
+            # float8 variants all agreeing: sum=65, avg=2.5, 26 rows.
+            # Negative values included — exercises signed accumulation.
+            "values": list(range(-10, 16)),
+            "sum": 65,
+            "avg": 2.5,
+            "count": 26,
+        },
     }
     with open(os.path.join(OUT, "expected.json"), "w") as f:
         json.dump(meta, f, indent=1)
