@@ -199,17 +199,93 @@ __global__ void k_tile_sort(const long long *__restrict__ keys,
     }
 }
 
+/* v7: tile sort nb=256 with 2 rows/thread and 16-B vectorized key loads
+ * in the hist and stage passes (tn is even except possibly the last tile;
+ * the odd tail row is handled scalar) */
+typedef long long ll2 __attribute__((ext_vector_type(2)));
+template <int T, int NB>
+__global__ __launch_bounds__(1024) void k_tile_sort_v2(
+    const long long *__restrict__ keys, const double *__restrict__ vals,
+    int64_t n, u64 *__restrict__ cursor, ulonglong2 *__restrict__ out)
+{
+    __shared__ ulonglong2 stage[T];
+    __shared__ u32 hist[NB], excl[NB], delta[NB];
+    int64_t ntiles = (n + T - 1) / T;
+    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+        int64_t lo = t * (int64_t)T;
+        int tn = (int)(n - lo < T ? n - lo : T);
+        int tq = tn / 2;
+        const ll2 *k2 = (const ll2 *)(keys + lo);
+        for (int j = threadIdx.x; j < NB; j += blockDim.x) hist[j] = 0;
+        __syncthreads();
+        for (int q = threadIdx.x; q < tq; q += blockDim.x) {
+            ll2 kk = k2[q];
+            atomicAdd(&hist[bucket_of(kk.x, NB)], 1u);
+            atomicAdd(&hist[bucket_of(kk.y, NB)], 1u);
+        }
+        if (threadIdx.x == 0 && (tn & 1))
+            atomicAdd(&hist[bucket_of(keys[lo + tn - 1], NB)], 1u);
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            u32 acc = 0;
+            for (int j = 0; j < NB; j++) {
+                excl[j] = acc;
+                acc += hist[j];
+            }
+        }
+        __syncthreads();
+        for (int j = threadIdx.x; j < NB; j += blockDim.x) {
+            u32 rb = hist[j] ? (u32)atomicAdd(&cursor[j], (u64)hist[j]) : 0u;
+            delta[j] = rb - excl[j];
+            hist[j] = 0;
+        }
+        __syncthreads();
+        for (int q = threadIdx.x; q < tq; q += blockDim.x) {
+            ll2 kk = k2[q];
+            double2 vv = ((const double2 *)(vals + lo))[q];
+            u32 b0 = bucket_of(kk.x, NB), b1 = bucket_of(kk.y, NB);
+            u32 r0 = excl[b0] + atomicAdd(&hist[b0], 1u);
+            u32 r1 = excl[b1] + atomicAdd(&hist[b1], 1u);
+            ulonglong2 ra, rb;
+            ra.x = (u64)kk.x;
+            ra.y = (u64)__double_as_longlong(vv.x);
+            rb.x = (u64)kk.y;
+            rb.y = (u64)__double_as_longlong(vv.y);
+            stage[r0] = ra;
+            stage[r1] = rb;
+        }
+        if (threadIdx.x == 0 && (tn & 1)) {
+            long long k = keys[lo + tn - 1];
+            u32 b = bucket_of(k, NB);
+            u32 r = excl[b] + atomicAdd(&hist[b], 1u);
+            ulonglong2 rr;
+            rr.x = (u64)k;
+            rr.y = (u64)__double_as_longlong(vals[lo + tn - 1]);
+            stage[r] = rr;
+        }
+        __syncthreads();
+        for (int p = threadIdx.x; p < tn; p += blockDim.x) {
+            ulonglong2 rec = stage[p];
+            u32 b = bucket_of((long long)rec.x, NB);
+            out[(size_t)(u32)(delta[b] + (u32)p)] = rec;
+        }
+        __syncthreads();
+    }
+}
+
 /* validation: each record sits in its own bucket's segment; key-sum
- * conserved */
+ * conserved. Segments given explicitly as (offs[b], cnts[b]) so padded
+ * layouts (v8) validate too. */
 __global__ void k_check(const ulonglong2 *__restrict__ out,
-                        const u64 *__restrict__ offs, u32 nb, int64_t n,
+                        const u64 *__restrict__ offs,
+                        const u64 *__restrict__ cnts, u32 nb, int64_t n,
                         u64 *errs, u64 *sum)
 {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     u64 my = 0, myerr = 0;
     for (int64_t b = blockIdx.y; b < nb; b += gridDim.y) {
         int64_t lo = (int64_t)offs[b];
-        int64_t hi = b + 1 < nb ? (int64_t)offs[b + 1] : n;
+        int64_t hi = lo + (int64_t)cnts[b];
         for (int64_t i = lo + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
              i < hi; i += stride) {
             ulonglong2 r = out[i];
@@ -236,22 +312,41 @@ int main(int argc, char **argv)
     double *vals;
     ulonglong2 *out;
     u64 *cnts256, *cnts2048, *offs256, *offs2048, *cursor, *scal;
+    u64 *pad_offs, *pad_cnts;
+    /* pad: fixed-stride segments for v8 (no pre-count); 1.3x the uniform
+     * average absorbs the binomial tail at 256 buckets */
+    u64 pad = (u64)((n / 256) * 1.3) + 64;
     CHK(hipMalloc(&keys, n * 8));
     CHK(hipMalloc(&vals, n * 8));
-    CHK(hipMalloc(&out, n * 16));
+    CHK(hipMalloc(&out, (size_t)(pad * 256) * 16 > (size_t)n * 16
+                            ? (size_t)(pad * 256) * 16
+                            : (size_t)n * 16));
     CHK(hipMalloc(&cnts256, 256 * 8));
     CHK(hipMalloc(&cnts2048, 2048 * 8));
     CHK(hipMalloc(&offs256, 256 * 8));
     CHK(hipMalloc(&offs2048, 2048 * 8));
     CHK(hipMalloc(&cursor, 2048 * 8));
+    CHK(hipMalloc(&pad_offs, 256 * 8));
+    CHK(hipMalloc(&pad_cnts, 256 * 8));
     CHK(hipMalloc(&scal, 3 * 8)); /* keysum_in, errs, keysum_out */
     hipLaunchKernelGGL(k_init, dim3(4096), dim3(256), 0, 0, keys, vals, n);
     CHK(hipMemset(scal, 0, 24));
     hipLaunchKernelGGL(k_keysum, dim3(2048), dim3(256), 0, 0, keys, n, scal);
     CHK(hipMemset(cnts256, 0, 256 * 8));
     CHK(hipMemset(cnts2048, 0, 2048 * 8));
+    hipEvent_t ec0, ec1;
+    CHK(hipEventCreate(&ec0));
+    CHK(hipEventCreate(&ec1));
+    CHK(hipDeviceSynchronize());
+    CHK(hipEventRecord(ec0));
     hipLaunchKernelGGL(k_count, dim3(2048), dim3(256), 0, 0, keys, n, 256,
                        cnts256);
+    CHK(hipEventRecord(ec1));
+    CHK(hipEventSynchronize(ec1));
+    float count_ms;
+    CHK(hipEventElapsedTime(&count_ms, ec0, ec1));
+    printf("pre-count pass (nb=256, what v8 eliminates): %.3f ms\n",
+           count_ms);
     hipLaunchKernelGGL(k_count, dim3(2048), dim3(256), 0, 0, keys, n, 2048,
                        cnts2048);
     /* host exclusive scans */
@@ -281,13 +376,19 @@ int main(int argc, char **argv)
         "v4 tile sort nb=2048 T=8192 bd=1024     ",
         "v5 tile sort nb=256  T=8192 bd=1024     ",
         "v6 cursor scatter nb=2048 bd=1024       ",
+        "v7 tile sort v2 (2 rows/thr, 16-B lds)  ",
+        "v8 tile sort, padded segs (no pre-count)",
     };
+    /* v8 fixed-stride cursor init values */
+    static u64 h_pad[256];
+    for (int j = 0; j < 256; j++) h_pad[j] = (u64)j * pad;
+    CHK(hipMemcpy(pad_offs, h_pad, 256 * 8, hipMemcpyHostToDevice));
     hipEvent_t e0, e1;
     CHK(hipEventCreate(&e0));
     CHK(hipEventCreate(&e1));
-    for (int v = 0; v < 7; v++) {
-        u32 nb = (v == 2 || v == 3 || v == 5) ? 256 : 2048;
-        u64 *offs = nb == 256 ? offs256 : offs2048;
+    for (int v = 0; v < 9; v++) {
+        u32 nb = (v == 2 || v == 3 || v == 5 || v >= 7) ? 256 : 2048;
+        u64 *offs = v == 8 ? pad_offs : (nb == 256 ? offs256 : offs2048);
         float best = 1e9f;
         for (int r = 0; r < reps; r++) {
             CHK(hipMemcpy(cursor, offs, nb * 8, hipMemcpyDeviceToDevice));
@@ -326,6 +427,16 @@ int main(int argc, char **argv)
                 hipLaunchKernelGGL(k_cursor_scatter, dim3(512), dim3(1024), 0,
                                    0, keys, vals, n, 2048, cursor, out);
                 break;
+            case 7:
+                hipLaunchKernelGGL((k_tile_sort_v2<8192, 256>), dim3(2048),
+                                   dim3(1024), 0, 0, keys, vals, n, cursor,
+                                   out);
+                break;
+            case 8: /* same kernel as v5; only the cursor layout differs */
+                hipLaunchKernelGGL((k_tile_sort<8192, 256>), dim3(2048),
+                                   dim3(1024), 0, 0, keys, vals, n, cursor,
+                                   out);
+                break;
             }
             CHK(hipEventRecord(e1));
             CHK(hipEventSynchronize(e1));
@@ -347,8 +458,24 @@ int main(int argc, char **argv)
                    best, 32.0 * n / best / 1e6);
             continue;
         }
+        u64 *vcnts = nb == 256 ? cnts256 : cnts2048;
+        if (v == 8) {
+            /* recover per-bucket counts from the mutated cursor; also
+             * verify no padded segment overflowed */
+            static u64 h_cur[256], h_pc[256];
+            CHK(hipMemcpy(h_cur, cursor, 256 * 8, hipMemcpyDeviceToHost));
+            u64 ovf = 0;
+            for (int j = 0; j < 256; j++) {
+                h_pc[j] = h_cur[j] - (u64)j * pad;
+                if (h_pc[j] > pad) ovf++;
+            }
+            if (ovf) printf("v8: %llu segments OVERFLOWED pad!\n",
+                            (unsigned long long)ovf);
+            CHK(hipMemcpy(pad_cnts, h_pc, 256 * 8, hipMemcpyHostToDevice));
+            vcnts = pad_cnts;
+        }
         hipLaunchKernelGGL(k_check, dim3(32, 64), dim3(256), 0, 0, out, offs,
-                           nb, n, scal + 1, scal + 2);
+                           vcnts, nb, n, scal + 1, scal + 2);
         u64 h_s[3];
         CHK(hipMemcpy(h_s, scal, 24, hipMemcpyDeviceToHost));
         printf("%s %8.3f ms  %7.0f GB/s  errs=%llu sum%s\n", names[v], best,
