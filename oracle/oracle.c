@@ -460,6 +460,16 @@ typedef struct {
 
 /* ---- generic agg (ora_agg_i64) ---- */
 
+/* emit order (key_isnull, key) — deterministic comparison only, not a
+ * semantic of the reference (hash iteration order there is arbitrary) */
+static int agg_group_cmp(const void *pa, const void *pb)
+{
+    const ora_agg_group *a = pa, *b = pb;
+    if (a->key_isnull != b->key_isnull) return a->key_isnull ? 1 : -1;
+    if (a->key_isnull) return 0;
+    return a->key < b->key ? -1 : a->key > b->key ? 1 : 0;
+}
+
 ora_status ora_agg_i64(const int64_t *keys, const uint8_t *key_null,
                        const double *vals, const uint8_t *val_null,
                        int64_t n, ora_agg_group **out, int64_t *ngroups)
@@ -510,15 +520,7 @@ ora_status ora_agg_i64(const int64_t *keys, const uint8_t *key_null,
     int64_t j = 0;
     for (int64_t s = 0; s < cap; s++)
         if (used[s]) res[j++] = slots[s];
-    /* insertion sort adequate for test-scale group counts */
-    for (int64_t i2 = 1; i2 < ng; i2++)
-        for (int64_t j2 = i2; j2 > 0; j2--) {
-            ora_agg_group *a = &res[j2 - 1], *b = &res[j2];
-            if (a->key_isnull > b->key_isnull ||
-                (a->key_isnull == b->key_isnull && !a->key_isnull && a->key > b->key)) {
-                ora_agg_group t2 = *a; *a = *b; *b = t2;
-            } else break;
-        }
+    qsort(res, (size_t)ng, sizeof(ora_agg_group), agg_group_cmp);
     free(slots); free(used);
     *out = res; *ngroups = ng;
     return ORA_OK;
