@@ -1,0 +1,151 @@
+"""Logic-level simulation of the tile-staged counting-sort partitioner
+(opentenbase_amd/csrc/otbx.hip k_tile_scatter1/k_tile_count2/k_tile_scan2/
+k_tile_scatter2/k_tile_restore_offs). Mirrors the kernel arithmetic
+faithfully — per-tile histogram, cursor run reservation, the mod-2^32
+delta trick (delta[b] = runbase - excl[b] as u32; dst = delta[b] + p), and
+the offs2-mutate-then-subtract level-2 bookkeeping — under ARBITRARY tile
+processing orders (blocks race on the cursors; any interleaving must yield
+bucket-contiguous segments). Catches algorithm/bookkeeping bugs without a
+GPU; the HIP-level races are covered by the GPU parity tests and the
+microbench validation (profiles/r01_scatter_ab.txt errs=0)."""
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+U32 = np.uint32
+TILE = 16  # small tile so multi-tile interleavings are exercised
+
+
+def bucket_of(keys, nb, shift):
+    # d_agg_bucket / d_agg_bucket2: splitmix64 finalizer bits
+    x = np.uint64(keys.astype(np.int64).view(np.uint64))
+    x = (x + np.uint64(0x9E3779B97F4A7C15)) & np.uint64(0xFFFFFFFFFFFFFFFF)
+    x = ((x ^ (x >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)) & \
+        np.uint64(0xFFFFFFFFFFFFFFFF)
+    x = ((x ^ (x >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)) & \
+        np.uint64(0xFFFFFFFFFFFFFFFF)
+    x = x ^ (x >> np.uint64(31))
+    return ((x >> np.uint64(shift)) & np.uint64(nb - 1)).astype(np.int64)
+
+
+def tile_scatter(keys, knull, nb, shift, cursor, out, out_b, tile_order):
+    """k_tile_scatter1's logic: tiles processed in tile_order (simulating
+    block scheduling), one cursor reservation per (tile, bucket), stage
+    ranks by counting sort, dst via u32 delta arithmetic."""
+    n = len(keys)
+    for t in tile_order:
+        lo, hi = t * TILE, min(t * TILE + TILE, n)
+        rows = np.arange(lo, hi)
+        if knull is not None:
+            rows = rows[~knull[lo:hi]]
+        b = bucket_of(keys[rows], nb, shift)
+        hist = np.bincount(b, minlength=nb)
+        excl = np.concatenate([[0], np.cumsum(hist)[:-1]]).astype(U32)
+        delta = np.zeros(nb, dtype=U32)
+        for j in range(nb):
+            if hist[j]:
+                rb = U32(cursor[j] & 0xFFFFFFFF)
+                delta[j] = rb - excl[j]  # wraps mod 2^32, as the kernel
+                cursor[j] += hist[j]
+        # stage = counting sort of the tile's rows by bucket (stable not
+        # required); p = stage position
+        order = np.argsort(b, kind="stable")
+        for p, ri in enumerate(order):
+            dst = int((delta[b[ri]] + U32(p)) & U32(0xFFFFFFFF))
+            out[dst] = keys[rows[ri]]
+            out_b[dst] = b[ri]
+
+
+@st.composite
+def l1_case(draw):
+    n = draw(st.integers(min_value=0, max_value=200))
+    keys = np.array(draw(st.lists(
+        st.integers(min_value=-(2**63), max_value=2**63 - 1),
+        min_size=n, max_size=n)), dtype=np.int64)
+    nb = draw(st.sampled_from([1, 2, 8, 32]))
+    with_null = draw(st.booleans())
+    knull = None
+    if with_null:
+        knull = np.array(draw(st.lists(st.booleans(), min_size=n,
+                                       max_size=n)), dtype=bool)
+    ntiles = (n + TILE - 1) // TILE
+    perm = draw(st.permutations(list(range(ntiles))))
+    return keys, knull, nb, perm
+
+
+@settings(max_examples=60, deadline=None, derandomize=True)
+@given(l1_case())
+def test_level1_any_tile_order_gives_bucket_contiguous_segments(case):
+    keys, knull, nb, perm = case
+    live = np.ones(len(keys), dtype=bool) if knull is None else ~knull
+    b_all = bucket_of(keys[live], nb, 40)
+    cnts = np.bincount(b_all, minlength=nb)
+    offs = np.concatenate([[0], np.cumsum(cnts)[:-1]])
+    cursor = offs.copy()
+    out = np.zeros(int(cnts.sum()), dtype=np.int64)
+    out_b = np.full(int(cnts.sum()), -1, dtype=np.int64)
+    tile_scatter(keys, knull, nb, 40, cursor, out, out_b, perm)
+    # every slot written exactly once, segments bucket-homogeneous,
+    # multiset of keys preserved per bucket
+    assert (cursor == offs + cnts).all()
+    for j in range(nb):
+        seg = slice(int(offs[j]), int(offs[j] + cnts[j]))
+        assert (out_b[seg] == j).all()
+        exp = np.sort(keys[live][b_all == j])
+        assert (np.sort(out[seg]) == exp).all()
+
+
+@settings(max_examples=40, deadline=None, derandomize=True)
+@given(l1_case())
+def test_level2_mutate_restore_bookkeeping(case):
+    """Level-2 flow on level-1 output: count2 per (segment, sub-bucket),
+    scan2 (offs2[s][j] = offs[s] + prefix), scatter with offs2 AS the
+    cursor, then restore offs2 -= cnts2 — final (offs2, cnts2) must tile
+    the array exactly and stay segment-nested."""
+    keys, knull, nb, perm = case
+    nb2 = 4
+    live = np.ones(len(keys), dtype=bool) if knull is None else ~knull
+    lk = keys[live]
+    b1 = bucket_of(lk, nb, 40)
+    cnts = np.bincount(b1, minlength=nb)
+    offs = np.concatenate([[0], np.cumsum(cnts)[:-1]])
+    # level-1 result (any bucket-contiguous layout works; use sorted)
+    recs = lk[np.argsort(b1, kind="stable")]
+    # count2 + scan2
+    b2_all = bucket_of(recs, nb2, 28)
+    cnts2 = np.zeros(nb * nb2, dtype=np.int64)
+    offs2 = np.zeros(nb * nb2, dtype=np.int64)
+    for s in range(nb):
+        seg = recs[int(offs[s]):int(offs[s] + cnts[s])]
+        h = np.bincount(bucket_of(seg, nb2, 28), minlength=nb2)
+        cnts2[s * nb2:(s + 1) * nb2] = h
+        offs2[s * nb2:(s + 1) * nb2] = offs[s] + \
+            np.concatenate([[0], np.cumsum(h)[:-1]])
+    # scatter2: offs2 mutated as the cursor, per-segment tiles in the
+    # drawn order (reuse perm modulo the segment's tile count)
+    out = np.zeros(len(recs), dtype=np.int64)
+    out_b = np.full(len(recs), -1, dtype=np.int64)
+    start = offs2.copy()
+    for s in range(nb):
+        seg_keys = recs[int(offs[s]):int(offs[s] + cnts[s])]
+        ntiles = (len(seg_keys) + TILE - 1) // TILE
+        order = [p % ntiles for p in perm if p < ntiles] or list(
+            range(ntiles))
+        seen = []
+        order = [t for t in order if not (t in seen or seen.append(t))]
+        order += [t for t in range(ntiles) if t not in order]
+        tile_scatter(seg_keys, None, nb2, 28,
+                     offs2[s * nb2:(s + 1) * nb2], out, out_b, order)
+    # restore (k_tile_restore_offs)
+    offs2 -= cnts2
+    assert (offs2 == start).all()
+    # final buckets tile the array exactly, nested inside their segment
+    for s in range(nb):
+        for j in range(nb2):
+            i = s * nb2 + j
+            seg = slice(int(offs2[i]), int(offs2[i] + cnts2[i]))
+            assert offs2[i] >= offs[s]
+            assert offs2[i] + cnts2[i] <= offs[s] + cnts[s]
+            assert (out_b[seg] == j).all()
+    b2_sorted = bucket_of(out, nb2, 28)
+    assert (b2_sorted == out_b).all() or len(out) == 0
+    assert (np.sort(out) == np.sort(lk)).all()
