@@ -149,3 +149,52 @@ def test_level2_mutate_restore_bookkeeping(case):
     b2_sorted = bucket_of(out, nb2, 28)
     assert (b2_sorted == out_b).all() or len(out) == 0
     assert (np.sort(out) == np.sort(lk)).all()
+
+
+@st.composite
+def padded_case(draw):
+    n = draw(st.integers(min_value=0, max_value=200))
+    # narrow domain -> heavy buckets; exercises the overflow-detection arm
+    keys = np.array(draw(st.lists(
+        st.integers(min_value=-30, max_value=30), min_size=n, max_size=n)),
+        dtype=np.int64)
+    nb = draw(st.sampled_from([2, 8]))
+    pad = draw(st.integers(min_value=4, max_value=64))
+    ntiles = (n + TILE - 1) // TILE
+    perm = draw(st.permutations(list(range(ntiles))))
+    return keys, nb, pad, perm
+
+
+@settings(max_examples=60, deadline=None, derandomize=True)
+@given(padded_case())
+def test_padded_segments_no_precount(case):
+    """The microbench-v8 layout (round-2 candidate): fixed-stride padded
+    segments, cursor[b] = b*pad, NO pre-count pass. Counts are recovered
+    post-hoc as cursor[b] - b*pad. If every bucket fits its pad, the
+    result is a valid partition; if any overflows, the post-hoc check
+    MUST detect it (that detection is the product integration's
+    retry-with-exact-counts trigger — records spill into the next
+    segment, so overflow without detection would be silent corruption)."""
+    keys, nb, pad, perm = case
+    b_all = bucket_of(keys, nb, 40)
+    true_cnts = np.bincount(b_all, minlength=nb)
+    cursor = np.arange(nb, dtype=np.int64) * pad
+    # NB: the LAST bucket's overflow writes past nb*pad — the sim sizes
+    # the buffer with slack to observe it; the product integration must
+    # allocate that slack (or clamp) since the kernel cannot know counts
+    # in advance. This is the hazard the detection arm below exists for.
+    out = np.full(nb * pad + len(keys), -1, dtype=np.int64)
+    out_b = np.full(nb * pad + len(keys), -1, dtype=np.int64)
+    # the scatter itself would corrupt on overflow; emulate the kernel
+    # faithfully (writes beyond the pad land in the next segment)
+    tile_scatter(keys, None, nb, 40, cursor, out, out_b, perm)
+    rec_cnts = cursor - np.arange(nb, dtype=np.int64) * pad
+    assert (rec_cnts == true_cnts).all()  # counts always recoverable
+    if (true_cnts <= pad).all():
+        for j in range(nb):
+            seg = slice(j * pad, j * pad + int(true_cnts[j]))
+            assert (out_b[seg] == j).all()
+            assert (np.sort(out[seg]) ==
+                    np.sort(keys[b_all == j])).all()
+    else:
+        assert (rec_cnts > pad).any()  # the detection arm fires
