@@ -193,3 +193,45 @@ def test_all_to_all_variable_gloo():
     out = _run(_worker_a2a)
     # rank 0 receives rank 0's first segment then rank 1's first segment
     assert out.tolist() == [0, 1, 2, 100, 101, 102]
+
+
+def _worker_q1_w8(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    sums = torch.zeros((6, 5), dtype=torch.float64)
+    counts = torch.zeros(6, dtype=torch.int64)
+    sums[0, 0] = float(rank + 1)
+    counts[0] = rank + 1
+    rows = fragment.merge_q1_partials(sums, counts)
+    if rank == 0:
+        q.put(rows)
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_merge_q1_partials_eight_ranks():
+    """World-size 8 — the driver's full-node scale point (8 DataNode
+    shards / 8 GPUs): the Q1 finalize combine over eight shards."""
+    rows = _run_n(_worker_q1_w8, 8)
+    af = [r for r in rows if r["l_returnflag"] == "A"][0]
+    assert af["count_order"] == sum(range(1, 9))
+    assert af["sum_qty"] == float(sum(range(1, 9)))
+
+
+def _worker_a2a_w8(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    t = torch.tensor([rank * 10 + d for d in range(world)], dtype=torch.int64)
+    out = fragment.all_to_all_variable(t, [1] * world)
+    if rank == 3:
+        q.put(out.numpy())
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_all_to_all_variable_eight_ranks():
+    """World-size 8 repartition exchange (the §8f.1 path at the full-node
+    width): rank 3 receives the (r, 3)-tagged element from every rank in
+    rank order."""
+    out = _run_n(_worker_a2a_w8, 8)
+    assert out.tolist() == [3, 13, 23, 33, 43, 53, 63, 73]
