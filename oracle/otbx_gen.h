@@ -170,6 +170,9 @@ OTBX_FN int64_t otbx_ord_orderkey(uint64_t i) { return (int64_t)i + 1; }
 
 OTBX_FN int64_t otbx_ord_custkey(uint64_t seed, uint64_t i, int64_t ncust)
 {
+    /* ncust = n_lineitem/40 is 0 for tiny tables; clamp like li_partkey's
+     * call sites do (a bare % 0 is SIGFPE on the CPU side, UB on GPU) */
+    if (ncust < 1) ncust = 1;
     uint64_t r = otbx_rnd(seed, OTBX_T_ORDERS, OTBX_C_CUSTKEY, i);
     return 1 + (int64_t)(r % (uint64_t)ncust);
 }
@@ -179,6 +182,7 @@ OTBX_FN int64_t otbx_ord_custkey(uint64_t seed, uint64_t i, int64_t ncust)
  * bit-identical on CPU and GPU. */
 OTBX_FN int64_t otbx_ord_custkey_skewed(uint64_t seed, uint64_t i, int64_t ncust)
 {
+    if (ncust < 1) ncust = 1;
     uint64_t r = otbx_rnd(seed, OTBX_T_ORDERS, OTBX_C_CUSTKEY, i);
     uint64_t r2 = otbx_splitmix64(r);
     int64_t nhot = ncust / 5 > 0 ? ncust / 5 : 1;

@@ -113,3 +113,45 @@ def test_q1_partial_matches_numpy_at_any_size(n, cutoff, seed_shift):
         if m.any():
             assert abs(g.sum_qty - float(li["l_quantity"][m].sum())) <= \
                 1e-9 * max(abs(g.sum_qty), 1.0)
+
+
+@settings(max_examples=20, deadline=None, derandomize=True)
+@given(st.integers(min_value=1, max_value=3000),
+       st.integers(min_value=1, max_value=50),
+       st.integers(min_value=0, max_value=49))
+def test_q9_partial_matches_numpy_any_size_and_filter(n, typemod, typeval):
+    """q9_partial across sizes and part-filter parameters (p_type %
+    typemod == typeval), incl. typeval >= typemod (matches nothing) and
+    typemod=1 (matches everything) — the selectivity extremes the fixed
+    reference test does not reach."""
+    t = ora.gen_tables(n, need=("lineitem", "orders", "part"))
+    li, od, pt = t["lineitem"], t["orders"], t["part"]
+    sel = pt["p_type"] % typemod == typeval
+    part_ok = np.zeros(int(pt["p_partkey"].max(initial=0)) + 1, dtype=bool)
+    part_ok[pt["p_partkey"][sel]] = True
+    # orderkey -> orderdate, with -1 for keys absent from orders (tiny
+    # tables: norders = n//4 may not cover every l_orderkey; the inner
+    # join drops those rows)
+    kmax = max(int(od["o_orderkey"].max(initial=0)),
+               int(li["l_orderkey"].max(initial=0)))
+    odate = np.full(kmax + 1, -1, dtype=np.int32)
+    odate[od["o_orderkey"]] = od["o_orderdate"]
+    mask = part_ok[li["l_partkey"]] & (odate[li["l_orderkey"]] >= 0)
+    got = ora.q9_partial(t, typemod=typemod, typeval=typeval)
+    if not mask.any():
+        assert got == []
+        return
+    d = odate[li["l_orderkey"][mask]]
+    rev = li["l_extendedprice"][mask] * (1.0 - li["l_discount"][mask])
+    bounds = np.array([0, 366, 731, 1096, 1461, 1827, 2192, 2558])
+    yr = np.searchsorted(bounds, d, side="right") - 1
+    exp = {}
+    for y in range(7):
+        m = yr == y
+        if m.sum():
+            exp[y] = (float(rev[m].sum()), int(m.sum()))
+    assert {g.year for g in got} == set(exp)
+    for g in got:
+        assert g.count_rows == exp[g.year][1]
+        assert abs(g.revenue - exp[g.year][0]) <= \
+            1e-9 * max(abs(exp[g.year][0]), 1e-9)
