@@ -155,3 +155,31 @@ def test_q9_partial_matches_numpy_any_size_and_filter(n, typemod, typeval):
         assert g.count_rows == exp[g.year][1]
         assert abs(g.revenue - exp[g.year][0]) <= \
             1e-9 * max(abs(exp[g.year][0]), 1e-9)
+
+
+@settings(max_examples=20, deadline=None, derandomize=True)
+@given(st.integers(min_value=1, max_value=3000),
+       st.integers(min_value=0, max_value=5),
+       st.integers(min_value=-1, max_value=3000))
+def test_q3_partial_matches_numpy_any_size(n, segment, date):
+    """q3_partial across sizes (incl. customer/orders tables of 0 rows at
+    n<40), market segments (incl. 5 = matches nothing; domain is 0..4) and
+    date cutoffs at both extremes."""
+    t = ora.gen_tables(n, need=("lineitem", "orders", "customer"))
+    li, od, cu = t["lineitem"], t["orders"], t["customer"]
+    rows = ora.q3_partial(t, segment=segment, date=date)
+    seg_keys = cu["c_custkey"][cu["c_mktsegment"] == segment]
+    omask = (od["o_orderdate"] < date) & np.isin(od["o_custkey"], seg_keys)
+    okeys = od["o_orderkey"][omask]
+    lmask = (li["l_shipdate"] > date) & np.isin(li["l_orderkey"], okeys)
+    exp_groups = set(li["l_orderkey"][lmask].tolist())
+    assert set(rows["l_orderkey"].tolist()) == exp_groups
+    if lmask.any():
+        rev = li["l_extendedprice"][lmask] * (1.0 - li["l_discount"][lmask])
+        byk = {}
+        for k, r in zip(li["l_orderkey"][lmask].tolist(), rev.tolist()):
+            byk[k] = byk.get(k, 0.0) + r
+        got = dict(zip(rows["l_orderkey"].tolist(),
+                       rows["revenue"].tolist()))
+        for k, v in byk.items():
+            assert abs(got[k] - v) <= 1e-9 * max(abs(v), 1e-9)
