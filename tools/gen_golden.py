@@ -99,6 +99,25 @@ def main():
             # select_having.out:16-22: GROUP BY b,c HAVING count(*)=1
             "groups_count1": [[1, "XXXX"], [3, "bbbb"]],
         },
+        "xc_fqs_join": {
+            # xc_FQS_join.out — tab1_rep/tab2_rep = the 5x5 grid
+            # (val, val2) in 1..5 x 1..5 (generate_series cross join, :51)
+            # :89-105  multi-key self-join (val, val2) with 1<val<4
+            #          -> exactly 10 matched pairs (grid keys are unique)
+            # :257-263 avg(val) over the natural 3-way self-join with
+            #          val>0 and val<3 -> 1.5
+            "join_filtered_pairs": 10,
+            "avg_val_filtered": 1.5,
+        },
+        "xl_join": {
+            # xl_join.out:20-27 — chained inner join through the
+            # "Distribute results by H" repartition exchange plan:
+            # t1.val1 in {1,2} joined on t2.val2 in {30,40} -> 0 rows
+            "t1": [[1, 10], [2, 20]],
+            "t2": [[3, 30], [4, 40]],
+            "t3": [[5, 50], [6, 60]],
+            "chained_rows": 0,
+        },
         "test_vec": {
             # opentenbase_c_aggregation.out:1-139 — the distributed
             # two-phase (Partial on DN -> Finalize on CN) aggregate plans
