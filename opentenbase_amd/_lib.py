@@ -78,6 +78,15 @@ def lib():
         _lib = C.CDLL(_SO)
         _lib.otbx_version.restype = C.c_char_p
         _lib.otbx_status_str.restype = C.c_char_p
+        # argtypes for the pure-host sizing functions: a wrong arity from
+        # a caller becomes a ctypes error instead of a segfault (the
+        # compute entry points keep the explicit-ctypes call style)
+        i64, u32, szp = C.c_int64, C.c_uint32, C.POINTER(C.c_size_t)
+        _lib.otbx_agg_i64_workspace_bytes.argtypes = [i64, szp]
+        _lib.otbx_join_i64_workspace_bytes.argtypes = [i64, i64, szp]
+        _lib.otbx_order_groups_workspace_bytes.argtypes = [i64, szp]
+        _lib.otbx_q3_workspace_bytes.argtypes = [i64, i64, i64, szp]
+        _lib.otbx_q9_workspace_bytes.argtypes = [i64, i64, i64, u32, szp]
     return _lib
 
 

@@ -59,3 +59,69 @@ def test_oracle_is_not_imported_by_product_code():
             "bad=[m for m in sys.modules if 'oracle' in m]; "
             "assert not bad, bad")
     subprocess.run([sys.executable, "-c", code], check=True, cwd=REPO)
+
+
+def _ws(so, fn, *args):
+    import ctypes as C
+    out = C.c_size_t(0)
+    st = getattr(so, fn)(*[C.c_int64(a) if not isinstance(a, C.c_uint32)
+                           else a for a in args], C.byref(out))
+    assert st == 0
+    return out.value
+
+
+def test_workspace_sizing_monotone_and_sane(so):
+    """The *_workspace_bytes functions are pure host code in libotbx.so —
+    callable without a GPU. Sizing must be monotone in the row counts
+    (a caller growing its input must never get a SMALLER requirement —
+    the kernels check ws_bytes >= need and return OTBX_ERR_INVALID on a
+    mismatch) and nonzero for nonzero inputs."""
+    import ctypes as C
+    prev = 0
+    for n in (0, 1, 1000, 10**6, 8 << 20, 100 * 10**6):
+        b = _ws(so, "otbx_agg_i64_workspace_bytes", n)
+        assert b >= prev
+        prev = b
+    prev = 0
+    for nb in (0, 1, 1000, 8 << 20, 50 * 10**6):
+        b = _ws(so, "otbx_join_i64_workspace_bytes", nb, 4 * nb)
+        assert b >= prev
+        prev = b
+    prev = 0
+    for n in (1, 1000, 10**6, 10**8):
+        b = _ws(so, "otbx_order_groups_workspace_bytes", n)
+        assert b >= prev
+        prev = b
+    b3 = _ws(so, "otbx_q3_workspace_bytes", 150_000, 1_500_000, 6_000_000)
+    assert b3 > 0
+    assert _ws(so, "otbx_q3_workspace_bytes", 1_500_000, 15_000_000,
+               60_000_000) > b3
+
+
+def test_q9_workspace_respects_bitmap_slice_cap(so, monkeypatch):
+    """otbx_q9_workspace_bytes and otbx_q9_partial share q9_slice_bits():
+    the workspace for a huge part table must stop growing once the slice
+    cap is hit (multipass), and the OTBX_Q9_BITMAP_BITS hook must shrink
+    it further — if the two sides ever diverged, every q9 call at SF300+
+    would fail with OTBX_ERR_INVALID on the GPU."""
+    import ctypes as C
+
+    def q9(nparts):
+        out = C.c_size_t(0)
+        st = so.otbx_q9_workspace_bytes(C.c_int64(nparts),
+                                        C.c_int64(1_000_000),
+                                        C.c_int64(4_000_000),
+                                        C.c_uint32(1), C.byref(out))
+        assert st == 0
+        return out.value
+
+    cap_bits = 1 << 30
+    small = q9(cap_bits // 2)
+    at_cap = q9(cap_bits)
+    beyond = q9(4 * cap_bits)
+    assert small < at_cap  # still single-pass: bitmap grows with nparts
+    assert beyond == at_cap  # sliced: bitmap capped at the slice width
+    monkeypatch.setenv("OTBX_Q9_BITMAP_BITS", "4096")
+    assert q9(cap_bits) < small  # the test hook shrinks the bitmap
+    monkeypatch.delenv("OTBX_Q9_BITMAP_BITS")
+    assert q9(cap_bits) == at_cap
