@@ -1,0 +1,24 @@
+#!/bin/bash
+# Round-2 first GPU call: run the prepared microbench
+# variants and re-baseline the generic ops, writing everything under
+# gpurun_out/ for triage. Usage:
+#   /usr/local/graft/bin/gpurun --timeout 600 -- 'bash tools/round2_kickoff.sh'
+set -x
+mkdir -p gpurun_out
+
+# 1. Scatter A/B: v7 (2-row vectorized tile sort) and v8 (padded segments,
+#    no pre-count) vs the shipped v5 — fold the winner into
+#    k_tile_scatter1/2 (v8 also needs the slack+retry notes in §8b).
+timeout 120 ./tools/microbench/scatter_ab | tee gpurun_out/r2_scatter_ab.txt
+
+# 2. Append plateau: v6 (tile-staged compaction) vs the shipped per-wave
+#    staged appenders (v2/v4) — if v6 wins, it drops into
+#    k_ord_filter_date (Q3) and k_q9_filter (Q9-mix).
+timeout 120 ./tools/microbench/append_ab | tee gpurun_out/r2_append_ab.txt
+
+# 3. Current generic-op baselines on this box (tile default ON), for
+#    apples-to-apples before any kernel edits.
+timeout 180 python tools/generic_ops_bench.py 2>&1 | tee gpurun_out/r2_ops_baseline.txt
+
+# 4. Quick suite sanity on the fresh box.
+timeout 240 python -m pytest tests -q -m gpu -x 2>&1 | tail -3 | tee gpurun_out/r2_suite.txt
