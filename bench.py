@@ -62,22 +62,45 @@ def load_traffic(workload, sf):
     return None
 
 
-def cpu_baseline(query, sample_rows):
+def cpu_baseline(query, sample_rows, shards=1):
     """Oracle CLI (scalar port of the reference executor) on a bounded
-    sample; returns the cpu_baseline JSON object."""
+    sample; returns the cpu_baseline JSON object. shards > 1 runs one CLI
+    process per shard concurrently (--rank/--nranks) — the DataNode
+    deployment analog (each backend is single-threaded; parallelism in
+    production is process-per-shard, SURVEY §8b threading note) — and
+    reports whole-job rows/s = global rows / max-over-shards seconds,
+    with cores = shards."""
     cli = os.path.join(REPO, "oracle", "oracle_cli")
     if not os.path.exists(cli):
         subprocess.run(["make", "-C", os.path.join(REPO, "oracle")],
                        check=True, capture_output=True)
-    out = subprocess.run([cli, query, "--rows", str(sample_rows)],
-                         check=True, capture_output=True, text=True).stdout
-    j = json.loads(out)
-    rows_s = j["rows"] / j["seconds"]
+    sample_rows -= sample_rows % shards  # generator: rows % nranks == 0
+    if shards <= 1:
+        out = subprocess.run([cli, query, "--rows", str(sample_rows)],
+                             check=True, capture_output=True,
+                             text=True).stdout
+        j = json.loads(out)
+        rows_s = j["rows"] / j["seconds"]
+        secs = j["seconds"]
+        tag = "scalar"
+    else:
+        procs = [subprocess.Popen(
+            [cli, query, "--rows", str(sample_rows), "--rank", str(r),
+             "--nranks", str(shards)],
+            stdout=subprocess.PIPE, text=True) for r in range(shards)]
+        js = []
+        for p in procs:
+            out, _ = p.communicate()
+            assert p.returncode == 0, p.returncode
+            js.append(json.loads(out))
+        secs = max(j["seconds"] for j in js)
+        rows_s = sum(j["rows"] for j in js) / secs
+        tag = f"{shards} shard processes"
     return {
-        "value": rows_s, "unit": "rows/s", "cores": 1, "kind": "port",
+        "value": rows_s, "unit": "rows/s", "cores": shards, "kind": "port",
         "sample": f"TPC-H {query.upper()} executor over {sample_rows} "
                   f"synthetic lineitem rows (SF{sample_rows // LI_PER_SF}), "
-                  f"scalar, {j['seconds']:.1f}s",
+                  f"{tag}, {secs:.1f}s",
     }
 
 
@@ -96,6 +119,9 @@ def main():
                     help="config-5 skewed distribution keys (hot custkeys)")
     ap.add_argument("--cpu-sample-rows", type=int, default=1_200_000_000,
                     help="~10 s of scalar CPU work on the target host")
+    ap.add_argument("--cpu-shards", type=int, default=1,
+                    help="oracle baseline processes (DataNode deployment "
+                         "analog; cores reported = this)")
     args = ap.parse_args()
 
     import torch
@@ -270,7 +296,7 @@ def main():
              "tpch_q9mix": "q9"}[args.workload]
         # join executors are ~4x slower per row: smaller sample
         rows = args.cpu_sample_rows if q == "q1" else args.cpu_sample_rows // 4
-        cpu = cpu_baseline(q, rows)
+        cpu = cpu_baseline(q, rows, shards=args.cpu_shards)
 
     if rank == 0:
         result = {

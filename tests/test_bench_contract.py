@@ -48,3 +48,24 @@ def test_committed_bench_snapshots_schema():
             assert k in d, k
         assert d["roofline"]["bound"] == "hbm"
         assert 0 < d["roofline"]["frac"] < 1
+
+
+def test_cpu_baseline_sharded():
+    """shards>1 = one CLI process per shard (DataNode deployment analog):
+    whole-job rows/s over max-shard time, cores = shards, and roughly
+    additive throughput vs the 1-core run on an idle host."""
+    b = _bench()
+    one = b.cpu_baseline("q1", 2_000_000)
+    four = b.cpu_baseline("q1", 2_000_000, shards=4)
+    assert four["cores"] == 4 and four["kind"] == "port"
+    assert "4 shard processes" in four["sample"]
+    # parallel shards must beat one core (loose: >=1.5x, CI boxes share)
+    assert four["value"] > 1.5 * one["value"]
+
+
+def test_cpu_baseline_shard_rounding():
+    """sample_rows not divisible by shards is rounded down (generator
+    requires rows % nranks == 0)."""
+    b = _bench()
+    cb = b.cpu_baseline("q1", 1_000_001, shards=4)
+    assert cb["value"] > 0 and "1000000 " in cb["sample"]
