@@ -22,3 +22,12 @@ timeout 180 python tools/generic_ops_bench.py 2>&1 | tee gpurun_out/r2_ops_basel
 
 # 4. Quick suite sanity on the fresh box.
 timeout 240 python -m pytest tests -q -m gpu -x 2>&1 | tail -3 | tee gpurun_out/r2_suite.txt
+timeout 120 python bench.py --no-cpu-baseline --steps 0 --warmup 0 2>/dev/null; timeout 180 python -c "
+import sys; sys.argv=['bench']
+import importlib.util, os, json
+spec = importlib.util.spec_from_file_location('bench', 'bench.py'); b = importlib.util.module_from_spec(spec); spec.loader.exec_module(b)
+import multiprocessing as mp
+for w in (1, 32, mp.cpu_count()):
+    cb = b.cpu_baseline('q1', 600_000_000 if w>1 else 150_000_000, shards=w)
+    print(json.dumps(cb))
+" | tee gpurun_out/r2_cpu_baseline_cores.txt
