@@ -22,7 +22,10 @@ timeout 180 python tools/generic_ops_bench.py 2>&1 | tee gpurun_out/r2_ops_basel
 
 # 4. Quick suite sanity on the fresh box.
 timeout 240 python -m pytest tests -q -m gpu -x 2>&1 | tail -3 | tee gpurun_out/r2_suite.txt
-timeout 120 python bench.py --no-cpu-baseline --steps 0 --warmup 0 2>/dev/null; timeout 180 python -c "
+
+# 5. All-cores CPU baseline (report shard-per-process totals
+#    next to the 1-core scalar number; cores stated in the object).
+timeout 300 python -c "
 import sys; sys.argv=['bench']
 import importlib.util, os, json
 spec = importlib.util.spec_from_file_location('bench', 'bench.py'); b = importlib.util.module_from_spec(spec); spec.loader.exec_module(b)
