@@ -74,7 +74,9 @@ def cpu_baseline(query, sample_rows, shards=1):
     if not os.path.exists(cli):
         subprocess.run(["make", "-C", os.path.join(REPO, "oracle")],
                        check=True, capture_output=True)
-    sample_rows -= sample_rows % shards  # generator: rows % nranks == 0
+    # generators require rows % nranks == 0 for EVERY sharded table:
+    # lineitem = rows, orders = rows/4, customer = rows/40
+    sample_rows -= sample_rows % (40 * shards)
     if shards <= 1:
         out = subprocess.run([cli, query, "--rows", str(sample_rows)],
                              check=True, capture_output=True,
