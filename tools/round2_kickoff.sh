@@ -6,6 +6,14 @@
 set -x
 mkdir -p gpurun_out
 
+# (re)build the microbenches if the binaries are absent — they are
+# git-ignored; normally the in-tree binaries travel with the snapshot
+for mb in scatter_ab append_ab; do
+  [ -x "tools/microbench/$mb" ] || \
+    hipcc --offload-arch=gfx950 -O3 "tools/microbench/$mb.hip" \
+          -o "tools/microbench/$mb"
+done
+
 # 1. Scatter A/B: v7 (2-row vectorized tile sort) and v8 (padded segments,
 #    no pre-count) vs the shipped v5 — fold the winner into
 #    k_tile_scatter1/2 (v8 also needs the slack+retry notes in §8b).
