@@ -235,3 +235,38 @@ def test_all_to_all_variable_eight_ranks():
     rank order."""
     out = _run_n(_worker_a2a_w8, 8)
     assert out.tolist() == [3, 13, 23, 33, 43, 53, 63, 73]
+
+
+def _worker_a2a_zero(rank, world, port, q):
+    from opentenbase_amd import fragment
+    _init(rank, world, port)
+    # skewed repartition edge: rank 1 has NO rows at all; rank 0 sends
+    # nothing to itself and 3 rows to rank 1
+    if rank == 0:
+        t = torch.tensor([10, 11, 12], dtype=torch.int64)
+        out = fragment.all_to_all_variable(t, [0, 3])
+    else:
+        t = torch.empty(0, dtype=torch.int64)
+        out = fragment.all_to_all_variable(t, [0, 0])
+    q.put((rank, out.numpy()))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_all_to_all_zero_length_segments():
+    """Zero-length splits and a completely empty rank (the skewed
+    distribution-key case of BASELINE config 5 at its extreme) must
+    round-trip: rank 0 ends with nothing, rank 1 with rank 0's rows."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29911 + np.random.randint(400, 800)
+    procs = [ctx.Process(target=_worker_a2a_zero, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = dict(q.get(timeout=180) for _ in range(2))
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert res[0].tolist() == []
+    assert res[1].tolist() == [10, 11, 12]
