@@ -270,3 +270,28 @@ def test_all_to_all_zero_length_segments():
         assert p.exitcode == 0
     assert res[0].tolist() == []
     assert res[1].tolist() == [10, 11, 12]
+
+
+def _worker_topk_empty(rank, world, port, q):
+    from opentenbase_amd import fragment
+    dt = np.dtype([("l_orderkey", "i8"), ("revenue", "f8"),
+                   ("o_orderdate", "i4"), ("o_shippriority", "i4")])
+    _init(rank, world, port)
+    # rank 1's shard has NO qualifying candidates (selective segment on a
+    # small shard) — the merge must still return rank 0's rows
+    n = 3 if rank == 0 else 0
+    cands = np.zeros(n, dtype=dt)
+    if n:
+        cands["l_orderkey"] = [1, 2, 3]
+        cands["revenue"] = [5.0, 9.0, 7.0]
+    top = fragment.merge_q3_topk(cands, k=2)
+    if rank == 0:
+        q.put(top)
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_merge_q3_topk_empty_shard():
+    top = _run(_worker_topk_empty)
+    assert top["revenue"].tolist() == [9.0, 7.0]
+    assert top["l_orderkey"].tolist() == [2, 3]
