@@ -36,6 +36,21 @@
  *                            is a collective over ranks, done by the host
  *                            layer with RCCL (torch.distributed) on the
  *                            partial-state buffers these calls return.
+ *
+ * Tuning/test environment variables (read at call time; all optional — the
+ * GUC analog of the provider shim, guc.c):
+ *   OTBX_PART_TILE=0       — select the legacy cursor-scatter partitioner
+ *                            in otbx_agg_i64/otbx_join_i64 (default is the
+ *                            tile-staged counting sort)
+ *   OTBX_JOINP_FORCE=1     — force the partitioned join path below its
+ *                            8 M-row build threshold (tests)
+ *   OTBX_Q9_BITMAP_BITS=N  — cap the Q9 part-bitmap slice width (bits);
+ *                            affects otbx_q9_workspace_bytes AND
+ *                            otbx_q9_partial identically (default 2^30 =
+ *                            single-pass for any realistic part table)
+ *   OTBX_DIRECT_CAP / OTBX_Q3_FORCE_HASH / OTBX_Q3_HASH_BUDGET
+ *                          — Q3 dense-direct vs hash+bloom path selection
+ *                            overrides (tests force the fallback)
  */
 #ifndef OTBX_H
 #define OTBX_H
