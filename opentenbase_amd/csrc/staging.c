@@ -89,6 +89,13 @@ otbx_status_i otbx_stage_pages(const void *pages, int64_t npages,
             uint8_t t_hoff = tup[22];
             uint32_t tup_natts = infomask2 & HEAP_NATTS_MASK;
             if (t_hoff > lp_len) return ST_ERR_INVALID;
+            /* the NULL bitmap (when present) lives at [23, t_hoff): a
+             * malformed t_hoff that doesn't cover (tup_natts+7)/8 bitmap
+             * bytes would send the bits[] reads past the tuple — and past
+             * the caller's buffer on the last page */
+            if ((infomask & HEAP_HASNULL) &&
+                (uint32_t)t_hoff < 23u + (tup_natts + 7u) / 8u)
+                return ST_ERR_INVALID;
             const uint8_t *bits =
                 (infomask & HEAP_HASNULL) ? tup + 23 : (const uint8_t *)0;
             if (row >= cap_rows) return ST_ERR_INVALID;

@@ -282,3 +282,24 @@ def test_staged_pages_q1_parity():
         assert r["count_order"] == e.count_order
         assert abs(r["sum_qty"] - e.sum_qty) <= 1e-9 * abs(e.sum_qty)
         assert abs(r["sum_charge"] - e.sum_charge) <= 1e-9 * abs(e.sum_charge)
+
+
+def test_null_bitmap_must_fit_under_hoff():
+    """Malformed tuple: HEAP_HASNULL set but t_hoff leaves no room for the
+    bitmap — the walker must reject it (ST_ERR_INVALID), not read past the
+    tuple (a 1..natts/8-byte OOB read before the fix, past the caller's
+    buffer when the tuple ends the last page)."""
+    atts = [(8, 8), (4, 4)]
+    # build a VALID 2-col tuple with a null, then corrupt t_hoff down to 23
+    tup = bytearray(pack_tuple([np.int64(7).tobytes(), None], atts,
+                               null_mask=[False, True]))
+    assert tup[22] == 24  # 23 + 1 bitmap byte, MAXALIGNed
+    tup[22] = 23          # t_hoff no longer covers the bitmap
+    page = pack_page([bytes(tup)])
+    st, nrows, _, _ = stage(page, 1, atts, 2, 8, with_nulls=True)
+    assert st == 3  # OTBX_ERR_INVALID
+    # the uncorrupted page stages fine
+    page_ok = pack_page([pack_tuple([np.int64(7).tobytes(), None], atts,
+                                    null_mask=[False, True])])
+    st, nrows, cols, nulls = stage(page_ok, 1, atts, 2, 8, with_nulls=True)
+    assert st == 0 and nrows == 1 and nulls[1][0] == 1
