@@ -81,8 +81,11 @@ otbx_status_i otbx_stage_pages(const void *pages, int64_t npages,
             uint32_t lp_len = (lp >> 17) & 0x7FFFu;
             if (lp_flags != LP_NORMAL)
                 continue; /* unused/dead/redirect: invisible post-heapgetpage */
-            if (lp_off + lp_len > page_size || lp_len < 23)
-                return ST_ERR_INVALID;
+            if (lp_off + lp_len > page_size || lp_len < 23 ||
+                lp_off < pd_upper || lp_off < PAGE_HEADER_BYTES)
+                return ST_ERR_INVALID; /* item must sit in the tuple area
+                                        * [pd_upper, page_size) —
+                                        * PageGetItemIdCareful analog */
             const uint8_t *tup = page + lp_off;
             uint16_t infomask2 = rd16(tup + 18);
             uint16_t infomask = rd16(tup + 20);

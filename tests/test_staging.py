@@ -303,3 +303,16 @@ def test_null_bitmap_must_fit_under_hoff():
                                     null_mask=[False, True])])
     st, nrows, cols, nulls = stage(page_ok, 1, atts, 2, 8, with_nulls=True)
     assert st == 0 and nrows == 1 and nulls[1][0] == 1
+
+
+def test_item_pointer_into_header_rejected():
+    """Malformed line pointer aiming into the page header / ItemId array
+    (lp_off < pd_upper) must be rejected — PageGetItemIdCareful analog."""
+    atts = [(8, 8)]
+    page = bytearray(pack_page([pack_tuple([np.int64(1).tobytes()], atts)]))
+    # corrupt item 0's lp_off to 24 (inside the ItemId area)
+    lp = struct.unpack_from("<I", page, 24)[0]
+    lp = (lp & ~0x7FFF) | 24
+    struct.pack_into("<I", page, 24, lp)
+    st, _, _, _ = stage(bytes(page), 1, atts, 1, 8)
+    assert st == 3
