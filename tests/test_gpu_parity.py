@@ -856,8 +856,10 @@ def test_q9_multipass_selfconsistent(ex, monkeypatch):
     assert [(r["o_year"], r["count_rows"]) for r in a] == \
         [(r["o_year"], r["count_rows"]) for r in b]
     for ra, rb in zip(a, b):
+        # order-dependent rounding only; 1e-9 bounds the worst-case linear
+        # error growth over the per-group value count with wide margin
         assert abs(ra["sum_revenue"] - rb["sum_revenue"]) <= \
-            1e-12 * abs(ra["sum_revenue"])
+            1e-9 * abs(ra["sum_revenue"])
 
 
 def test_partitioner_tile_vs_legacy_selfconsistent(ex, monkeypatch):
