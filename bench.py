@@ -71,9 +71,10 @@ def cpu_baseline(query, sample_rows, shards=1):
     reports whole-job rows/s = global rows / max-over-shards seconds,
     with cores = shards."""
     cli = os.path.join(REPO, "oracle", "oracle_cli")
-    if not os.path.exists(cli):
-        subprocess.run(["make", "-C", os.path.join(REPO, "oracle")],
-                       check=True, capture_output=True)
+    # always (re)build: an oracle.c edit with a stale committed binary would
+    # silently time old code (ADVICE r1); make is incremental and cheap
+    subprocess.run(["make", "-C", os.path.join(REPO, "oracle")],
+                   check=True, capture_output=True)
     # generators require rows % nranks == 0 for EVERY sharded table:
     # lineitem = rows, orders = rows/4, customer = rows/40
     sample_rows -= sample_rows % (40 * shards)
@@ -313,7 +314,10 @@ def main():
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": None,
+            # BASELINE.md publishes no reference numbers; its baseline IS the
+            # measured CPU-executor restatement (BASELINE.md "CPU-baseline
+            # plan"), so vs_baseline = GPU value / measured cpu_baseline
+            "vs_baseline": (value / cpu["value"]) if cpu else None,
             "dtype": "f64",
             "data": "synthetic (dbgen-shaped, seed 42, generated on-device)",
             "gb_per_s_scanned": gbps,

@@ -37,6 +37,15 @@
  *                            layer with RCCL (torch.distributed) on the
  *                            partial-state buffers these calls return.
  *
+ * THREADING / STREAM CONTRACT: the library is single-threaded per process,
+ * matching the reference's execution model (each DataNode backend is a
+ * single-threaded process). Entry points may be called from one
+ * thread at a time with one stream in flight; concurrent calls from
+ * multiple threads or interleaved streams are NOT supported (several entry
+ * points keep per-process cached scratch). Multiple backends on one host
+ * each load their own copy (process isolation). otbx_finish releases all
+ * cached scratch, so init → work → finish → init(other_device) is clean.
+ *
  * Tuning/test environment variables (read at call time; all optional — the
  * GUC analog of the provider shim, guc.c):
  *   OTBX_PART_TILE=0       — select the legacy cursor-scatter partitioner
@@ -282,7 +291,11 @@ otbx_status otbx_order_groups(const otbx_q3_group *groups_dev, int64_t n,
  * The GPU half of the reference's "Distribute results by H: col" exchange
  * (make_remotesubplan, optimizer/plan/createplan.c:8671; locator semantics
  * shardid → node, pgxc/shard/shardmap.c:2231 restated as key % nranks for
- * the dense-key locator of DESIGN.md §2): groups rows by owning rank into a
+ * the dense-key locator of DESIGN.md §2 — owner = (uint64_t)key % nranks,
+ * i.e. the UNSIGNED-cast modulo: negative keys map to
+ * (2^64 + key) % nranks, NOT the C signed remainder and NOT Python's
+ * floored modulo; server-side locator code must use the same cast): groups
+ * rows by owning rank into a
  * permutation with contiguous per-rank segments; the host layer then
  * all-to-alls the gathered segments over RCCL (fragment.py). nranks ≤ 64.
  * counts_host: int64[nranks], written synchronously (the call syncs). */
@@ -302,8 +315,15 @@ otbx_status otbx_gather_u8(const uint8_t *src, const int64_t *perm, int64_t n,
                            uint8_t *dst, void *stream);
 
 /* Inner hash join on i64 keys: emits (build_idx, probe_idx) pairs in
- * arbitrary order (result-set parity; SQL imposes no order). pairs capacity
- * cap_pairs; overflow → OTBX_ERR_INVALID reported via npairs_dev = -1. */
+ * arbitrary order (result-set parity; SQL imposes no order).
+ * OVERFLOW CONTRACT: *npairs_dev always receives the TRUE match count; if
+ * it exceeds cap_pairs the output arrays hold only a cap_pairs-bounded
+ * subset and the caller MUST treat the result as overflowed (re-run with
+ * cap_pairs >= *npairs_dev). The call itself still returns OTBX_OK — the
+ * count lives on the device and is not visible to the host entry point;
+ * this mirrors how the reference sizes hash tables from observed counts
+ * (ExecHashTableInsert growth, nodeHash.c:1876) rather than failing
+ * mid-scan. Pinned by tests/test_gpu_parity.py::test_join_overflow_contract. */
 otbx_status otbx_join_i64_workspace_bytes(int64_t nb, int64_t np,
                                            size_t *bytes);
 otbx_status otbx_join_i64(const int64_t *bkeys_dev, const uint8_t *bnull_dev,

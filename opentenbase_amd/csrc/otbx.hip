@@ -47,6 +47,33 @@
         }                                                                \
     } while (0)
 
+/* Lazily-cached per-process scratch (the function-local statics below):
+ * every allocation registers its slot here so otbx_finish can release it
+ * and null the static — re-initializing on a different device must not see
+ * stale allocations. The C-ABI is single-threaded per process (each
+ * reference backend is a single-threaded process; contract in otbx.h), so
+ * no locking. */
+#define OTBX_SCRATCH_MAX 64
+static void **g_scr_slot[OTBX_SCRATCH_MAX];
+static unsigned char g_scr_host[OTBX_SCRATCH_MAX];
+static int g_nscr = 0;
+
+static hipError_t otbx_scr_alloc(void **slot, size_t bytes, int host)
+{
+    hipError_t e = host ? hipHostMalloc(slot, bytes) : hipMalloc(slot, bytes);
+    if (e == hipSuccess && g_nscr < OTBX_SCRATCH_MAX) {
+        g_scr_slot[g_nscr] = slot;
+        g_scr_host[g_nscr] = (unsigned char)host;
+        g_nscr++;
+    }
+    return e;
+}
+
+#define SCR_ALLOC_DEV(p, bytes) \
+    HIP_CHECK(otbx_scr_alloc((void **)&(p), (bytes), 0))
+#define SCR_ALLOC_HOST(p, bytes) \
+    HIP_CHECK(otbx_scr_alloc((void **)&(p), (bytes), 1))
+
 extern "C" {
 
 const char *otbx_version(void) { return OTBX_VERSION_STR; }
@@ -76,6 +103,18 @@ otbx_status otbx_init(int device)
 otbx_status otbx_finish(void)
 {
     HIP_CHECK(hipDeviceSynchronize());
+    /* release the registered lazy scratch and null the owning statics so a
+     * later otbx_init on another device starts clean */
+    for (int i = 0; i < g_nscr; i++) {
+        if (*g_scr_slot[i]) {
+            if (g_scr_host[i])
+                HIP_CHECK(hipHostFree(*g_scr_slot[i]));
+            else
+                HIP_CHECK(hipFree(*g_scr_slot[i]));
+            *g_scr_slot[i] = nullptr;
+        }
+    }
+    g_nscr = 0;
     return OTBX_OK;
 }
 
@@ -1703,8 +1742,8 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
         static unsigned long long *d_ds = nullptr;
         static unsigned long long *h_ds = nullptr;
         if (!d_ds) {
-            HIP_CHECK(hipMalloc(&d_ds, 16));
-            HIP_CHECK(hipHostMalloc(&h_ds, 16));
+            SCR_ALLOC_DEV(d_ds, 16);
+            SCR_ALLOC_HOST(h_ds, 16);
         }
         /* sample set lives in the (still unwritten) perm region */
         int64_t *stab = (int64_t *)((char *)(nullgrp + 2));
@@ -1779,7 +1818,7 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
                            s, keys, knull, n, nb, cnts);
         static unsigned long long *h_cnts = nullptr;
         if (!h_cnts)
-            HIP_CHECK(hipHostMalloc(&h_cnts, AGGP_MAX_BUCKETS * 8 * 2));
+            SCR_ALLOC_HOST(h_cnts, AGGP_MAX_BUCKETS * 8 * 2);
         HIP_CHECK(hipMemcpyAsync(h_cnts, cnts, (size_t)nb * 8,
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipStreamSynchronize(s));
@@ -1810,8 +1849,8 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
                 static unsigned int *d_nflag = nullptr;
                 static unsigned int *h_nflag = nullptr;
                 if (!d_nflag) {
-                    HIP_CHECK(hipMalloc(&d_nflag, 4));
-                    HIP_CHECK(hipHostMalloc(&h_nflag, 4));
+                    SCR_ALLOC_DEV(d_nflag, 4);
+                    SCR_ALLOC_HOST(h_nflag, 4);
                 }
                 const ulonglong2 *frecs = recs;
                 const unsigned long long *foffs = offs, *fcnts = cnts;
@@ -1890,8 +1929,8 @@ otbx_status otbx_agg_i64(const int64_t *keys, const uint8_t *knull,
         static unsigned int *d_ab = nullptr;
         static unsigned int *h_ab = nullptr;
         if (!d_ab) {
-            HIP_CHECK(hipMalloc(&d_ab, 4));
-            HIP_CHECK(hipHostMalloc(&h_ab, 4));
+            SCR_ALLOC_DEV(d_ab, 4);
+            SCR_ALLOC_HOST(h_ab, 4);
         }
         int64_t cap_s = next_pow2_host(est * 64 < 4096 ? 4096 : est * 64);
         if (cap_s > cap) cap_s = cap;
@@ -2190,8 +2229,8 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
     static int64_t *h_mm = nullptr;
     static unsigned long long *d_mm = nullptr;
     if (!h_mm) {
-        HIP_CHECK(hipHostMalloc(&h_mm, 16));
-        HIP_CHECK(hipMalloc(&d_mm, 16));
+        SCR_ALLOC_HOST(h_mm, 16);
+        SCR_ALLOC_DEV(d_mm, 16);
     }
     int64_t mino, orange;
     if (o->n == 0) {
@@ -2617,7 +2656,7 @@ otbx_status otbx_partition_by_key(const int64_t *keys, int64_t n,
     hipStream_t s = (hipStream_t)stream;
     static unsigned long long *scratch = nullptr; /* counts + cursors */
     if (!scratch)
-        HIP_CHECK(hipMalloc(&scratch, PART_MAX_RANKS * 2 * 8));
+        SCR_ALLOC_DEV(scratch, PART_MAX_RANKS * 2 * 8);
     unsigned long long *counts = scratch, *cursor = scratch + PART_MAX_RANKS;
     HIP_CHECK(hipMemsetAsync(counts, 0, PART_MAX_RANKS * 8, s));
     if (n > 0)
@@ -2625,14 +2664,14 @@ otbx_status otbx_partition_by_key(const int64_t *keys, int64_t n,
                            s, keys, n, nranks, counts);
     static int64_t *h_counts = nullptr;
     if (!h_counts)
-        HIP_CHECK(hipHostMalloc(&h_counts, PART_MAX_RANKS * 8));
+        SCR_ALLOC_HOST(h_counts, PART_MAX_RANKS * 8);
     HIP_CHECK(hipMemcpyAsync(h_counts, counts, nranks * 8,
                              hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     int64_t off = 0;
     static int64_t *h_cursor = nullptr;
     if (!h_cursor)
-        HIP_CHECK(hipHostMalloc(&h_cursor, PART_MAX_RANKS * 8));
+        SCR_ALLOC_HOST(h_cursor, PART_MAX_RANKS * 8);
     for (uint32_t r = 0; r < nranks; r++) {
         counts_host[r] = h_counts[r];
         h_cursor[r] = off;
@@ -3131,7 +3170,7 @@ otbx_status otbx_join_i64(const int64_t *bkeys, const uint8_t *bnull, int64_t nb
         hipLaunchKernelGGL(k_aggp_count, dim3(grid_for(np, 256)), dim3(256),
                            0, s, pkeys, pnull, np, nbuk, cnts_p);
         static unsigned long long *h_j = nullptr;
-        if (!h_j) HIP_CHECK(hipHostMalloc(&h_j, AGGP_MAX_BUCKETS * 8 * 4));
+        if (!h_j) SCR_ALLOC_HOST(h_j, AGGP_MAX_BUCKETS * 8 * 4);
         HIP_CHECK(hipMemcpyAsync(h_j, cnts_b, (size_t)nbuk * 8,
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipMemcpyAsync(h_j + AGGP_MAX_BUCKETS, cnts_p,
@@ -3220,8 +3259,8 @@ otbx_status otbx_join_i64(const int64_t *bkeys, const uint8_t *bnull, int64_t nb
         static unsigned int *d_nflag = nullptr;
         static unsigned int *h_nflag = nullptr;
         if (!d_nflag) {
-            HIP_CHECK(hipMalloc(&d_nflag, 4));
-            HIP_CHECK(hipHostMalloc(&h_nflag, 4));
+            SCR_ALLOC_DEV(d_nflag, 4);
+            SCR_ALLOC_HOST(h_nflag, 4);
         }
         HIP_CHECK(hipMemsetAsync(d_nflag, 0, 4, s));
         HIP_CHECK(hipMemsetAsync(flags, 0, (size_t)fgrid, s));
@@ -4231,7 +4270,7 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
 
     static int64_t *h_cnt = nullptr;        /* pinned host readback */
     if (!h_cnt)
-        HIP_CHECK(hipHostMalloc(&h_cnt, 8 * sizeof(int64_t)));
+        SCR_ALLOC_HOST(h_cnt, 8 * sizeof(int64_t));
 
     HIP_CHECK(hipMemsetAsync(hdr, 0, 64, s));
     HIP_CHECK(hipMemsetAsync(&hdr[4], 0x7f, 8, s)); /* min orderkey */
@@ -4244,7 +4283,7 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     } else {
         static unsigned long long *scratch = nullptr;
         if (!scratch)
-            HIP_CHECK(hipMalloc(&scratch, sizeof(unsigned long long)));
+            SCR_ALLOC_DEV(scratch, sizeof(unsigned long long));
         nhits = scratch;
     }
 
@@ -4539,7 +4578,7 @@ extern "C" otbx_status otbx_topk_by_revenue(const otbx_q3_group *groups,
                        groups, n, hist);
     static uint32_t *h_hist = nullptr;
     if (!h_hist)
-        HIP_CHECK(hipHostMalloc(&h_hist, 16384 * 4));
+        SCR_ALLOC_HOST(h_hist, 16384 * 4);
     HIP_CHECK(hipMemcpyAsync(h_hist, hist, 16384 * 4, hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     int64_t cum = 0;

@@ -340,15 +340,22 @@ class GpuQ3Fragment(CustomScanState):
         if n == 0:
             self.groups = np.empty(0, dtype=np.dtype(self.NP_DTYPE))
             return []
+        # retry with a larger cap if the threshold histogram bin holds more
+        # candidates than fit (ncand receives the TRUE count; silently
+        # truncating could drop real top-k rows)
         cap_cand = 1 << 20
-        cand = torch.empty(cap_cand * 24, dtype=torch.uint8, device="cuda")
-        ncand = torch.zeros(1, dtype=torch.int64, device="cuda")
-        hist = torch.empty(16384, dtype=torch.int32, device="cuda")
-        call("otbx_topk_by_revenue", C.c_void_p(groups.data_ptr()),
-             C.c_int64(n), C.c_int64(self.k), C.c_void_p(cand.data_ptr()),
-             C.c_int64(cap_cand), C.c_void_p(ncand.data_ptr()),
-             C.c_void_p(hist.data_ptr()), _stream())
-        nc = min(int(ncand.cpu().item()), cap_cand)
+        while True:
+            cand = torch.empty(cap_cand * 24, dtype=torch.uint8, device="cuda")
+            ncand = torch.zeros(1, dtype=torch.int64, device="cuda")
+            hist = torch.empty(16384, dtype=torch.int32, device="cuda")
+            call("otbx_topk_by_revenue", C.c_void_p(groups.data_ptr()),
+                 C.c_int64(n), C.c_int64(self.k), C.c_void_p(cand.data_ptr()),
+                 C.c_int64(cap_cand), C.c_void_p(ncand.data_ptr()),
+                 C.c_void_p(hist.data_ptr()), _stream())
+            nc = int(ncand.cpu().item())
+            if nc <= cap_cand:
+                break
+            cap_cand = max(cap_cand * 2, nc)
         cands = cand[: nc * 24].cpu().numpy().view(np.dtype(self.NP_DTYPE))
         return [tuple(r) for r in q3_topk(cands, self.k)]
 

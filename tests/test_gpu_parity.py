@@ -884,3 +884,47 @@ def test_partitioner_tile_vs_legacy_selfconsistent(ex, monkeypatch):
         assert ra["count_star"] == rb["count_star"]
         assert abs(float(ra["sum_v"]) - float(rb["sum_v"])) <= \
             1e-12 * max(abs(float(rb["sum_v"])), 1e-12)
+
+
+def test_join_overflow_contract(ex):
+    """Pins the otbx.h overflow contract (ADVICE r1 / VERDICT r1 #6):
+    *npairs_dev receives the TRUE match count even when it exceeds
+    cap_pairs; the call returns OTBX_OK; the output arrays hold only a
+    bounded subset of VALID pairs; the host layer raises."""
+    import ctypes as C
+    from opentenbase_amd._lib import lib
+    from opentenbase_amd.executor import _stream
+
+    L = lib()
+    nb, npr = 1000, 50000
+    # every probe key matches exactly one build key -> 50000 true pairs
+    bk = torch.arange(nb, dtype=torch.int64, device="cuda")
+    pk = torch.arange(npr, dtype=torch.int64, device="cuda") % nb
+    ws_bytes = C.c_size_t(0)
+    assert L.otbx_join_i64_workspace_bytes(C.c_int64(nb), C.c_int64(npr),
+                                           C.byref(ws_bytes)) == 0
+    ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8, device="cuda")
+    cap = 128          # far below the 50000 true pairs
+    ob = torch.full((cap,), -1, dtype=torch.int64, device="cuda")
+    op = torch.full((cap,), -1, dtype=torch.int64, device="cuda")
+    npairs = torch.zeros(1, dtype=torch.int64, device="cuda")
+    st = L.otbx_join_i64(C.c_void_p(bk.data_ptr()), None, C.c_int64(nb),
+                         C.c_void_p(pk.data_ptr()), None, C.c_int64(npr),
+                         C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+                         C.c_void_p(ob.data_ptr()), C.c_void_p(op.data_ptr()),
+                         C.c_int64(cap), C.c_void_p(npairs.data_ptr()),
+                         _stream())
+    assert st == 0                      # OTBX_OK per the documented contract
+    torch.cuda.synchronize()
+    n_true = int(npairs.cpu().item())
+    assert n_true == npr                # TRUE count, not clamped to cap
+    # whatever subset was written must be valid join pairs
+    obh, oph = ob.cpu().numpy(), op.cpu().numpy()
+    written = obh >= 0
+    assert written.any()
+    assert np.array_equal(obh[written], oph[written] % nb)
+    # the host layer surfaces overflow as an error (never truncates silently)
+    node = ex.GpuHashJoin(bk, pk, cap_pairs=cap)
+    node.BeginCustomScan()
+    with pytest.raises(ex.OtbxError):
+        node._run()
