@@ -29,7 +29,7 @@ timeout 120 ./tools/microbench/append_ab | tee gpurun_out/r2_append_ab.txt
 timeout 180 python tools/generic_ops_bench.py 2>&1 | tee gpurun_out/r2_ops_baseline.txt
 
 # 4. Quick suite sanity on the fresh box.
-timeout 240 python -m pytest tests -q -m gpu -x 2>&1 | tail -3 | tee gpurun_out/r2_suite.txt
+timeout 700 python -m pytest tests -q -m gpu -x 2>&1 | tail -3 | tee gpurun_out/r2_suite.txt
 
 # 5. All-cores CPU baseline (report shard-per-process totals
 #    next to the 1-core scalar number; cores stated in the object).
