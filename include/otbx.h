@@ -60,6 +60,10 @@
  *   OTBX_DIRECT_CAP / OTBX_Q3_FORCE_HASH / OTBX_Q3_HASH_BUDGET
  *                          — Q3 dense-direct vs hash+bloom path selection
  *                            overrides (tests force the fallback)
+ *   OTBX_Q9_FILTER_WAVE=1  — legacy per-wave appender in the Q9 part filter
+ *                            (default is the tile-staged compaction; A/B)
+ *   OTBX_Q3_COMPACT_LEGACY=1 — legacy block-chunk Q3 group compaction
+ *                            (default is the tile-staged variant; A/B)
  */
 #ifndef OTBX_H
 #define OTBX_H

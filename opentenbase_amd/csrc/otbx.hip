@@ -1247,6 +1247,8 @@ __global__ void k_aggp_scatter_kv2(const ulonglong2 *__restrict__ recs,
 /* level 1: keys[] (+ optional knull skip list) -> (key, payload) records;
  * vals == NULL emits the row index as payload (the join shape), else the
  * double bits (the agg kv shape). nb ≤ 256. */
+typedef long long otbx_ll2 __attribute__((ext_vector_type(2)));
+
 __global__ __launch_bounds__(1024) void k_tile_scatter1(
     const int64_t *__restrict__ keys, const uint8_t *__restrict__ knull,
     const double *__restrict__ vals, int64_t n, uint32_t nb,
@@ -1255,15 +1257,35 @@ __global__ __launch_bounds__(1024) void k_tile_scatter1(
     __shared__ ulonglong2 stage[PT_TILE];
     __shared__ unsigned int hist[256], excl[256], delta[256];
     __shared__ unsigned int tot;
+    /* 16-B vector path needs 16-B-aligned bases (lo is even, so tile
+     * offsets preserve base alignment) and no null bitmap */
+    const bool vec = !knull &&
+                     (((uintptr_t)keys | (vals ? (uintptr_t)vals : 0)) & 15)
+                         == 0;
     int64_t ntiles = (n + PT_TILE - 1) / PT_TILE;
     for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
         int64_t lo = t * (int64_t)PT_TILE;
         int tn = (int)(n - lo < PT_TILE ? n - lo : PT_TILE);
+        int tq = tn / 2;
+        const otbx_ll2 *k2 = (const otbx_ll2 *)(keys + lo);
         for (int j = threadIdx.x; j < (int)nb; j += blockDim.x) hist[j] = 0;
         __syncthreads();
-        for (int i = threadIdx.x; i < tn; i += blockDim.x) {
-            if (knull && knull[lo + i]) continue;
-            atomicAdd(&hist[d_agg_bucket(keys[lo + i], nb)], 1u);
+        if (vec) {
+            /* microbenched fast path (scatter_ab v7, profiles/
+             * r2_scatter_ab.txt: 5.14 vs 7.34 ms at 600 M): 2 rows/thread,
+             * 16-B vector loads in the hist and stage passes */
+            for (int q = threadIdx.x; q < tq; q += blockDim.x) {
+                otbx_ll2 kk = k2[q];
+                atomicAdd(&hist[d_agg_bucket(kk.x, nb)], 1u);
+                atomicAdd(&hist[d_agg_bucket(kk.y, nb)], 1u);
+            }
+            if (threadIdx.x == 0 && (tn & 1))
+                atomicAdd(&hist[d_agg_bucket(keys[lo + tn - 1], nb)], 1u);
+        } else {
+            for (int i = threadIdx.x; i < tn; i += blockDim.x) {
+                if (knull && knull[lo + i]) continue;
+                atomicAdd(&hist[d_agg_bucket(keys[lo + i], nb)], 1u);
+            }
         }
         __syncthreads();
         if (threadIdx.x == 0) {
@@ -1284,17 +1306,51 @@ __global__ __launch_bounds__(1024) void k_tile_scatter1(
             hist[j] = 0;             /* reused as the stage cursor */
         }
         __syncthreads();
-        for (int i = threadIdx.x; i < tn; i += blockDim.x) {
-            if (knull && knull[lo + i]) continue;
-            int64_t k = keys[lo + i];
-            uint32_t b = d_agg_bucket(k, nb);
-            unsigned int r = excl[b] + atomicAdd(&hist[b], 1u);
-            ulonglong2 rec;
-            rec.x = (unsigned long long)k;
-            rec.y = vals ? (unsigned long long)__double_as_longlong(
-                               vals[lo + i])
-                         : (unsigned long long)(lo + i);
-            stage[r] = rec;
+        if (vec) {
+            for (int q = threadIdx.x; q < tq; q += blockDim.x) {
+                otbx_ll2 kk = k2[q];
+                uint32_t b0 = d_agg_bucket(kk.x, nb);
+                uint32_t b1 = d_agg_bucket(kk.y, nb);
+                unsigned int r0 = excl[b0] + atomicAdd(&hist[b0], 1u);
+                unsigned int r1 = excl[b1] + atomicAdd(&hist[b1], 1u);
+                ulonglong2 ra, rb;
+                ra.x = (unsigned long long)kk.x;
+                rb.x = (unsigned long long)kk.y;
+                if (vals) {
+                    double2 vv = ((const double2 *)(vals + lo))[q];
+                    ra.y = (unsigned long long)__double_as_longlong(vv.x);
+                    rb.y = (unsigned long long)__double_as_longlong(vv.y);
+                } else {
+                    ra.y = (unsigned long long)(lo + 2 * (int64_t)q);
+                    rb.y = (unsigned long long)(lo + 2 * (int64_t)q + 1);
+                }
+                stage[r0] = ra;
+                stage[r1] = rb;
+            }
+            if (threadIdx.x == 0 && (tn & 1)) {
+                int64_t k = keys[lo + tn - 1];
+                uint32_t b = d_agg_bucket(k, nb);
+                unsigned int r = excl[b] + atomicAdd(&hist[b], 1u);
+                ulonglong2 rr;
+                rr.x = (unsigned long long)k;
+                rr.y = vals ? (unsigned long long)__double_as_longlong(
+                                  vals[lo + tn - 1])
+                            : (unsigned long long)(lo + tn - 1);
+                stage[r] = rr;
+            }
+        } else {
+            for (int i = threadIdx.x; i < tn; i += blockDim.x) {
+                if (knull && knull[lo + i]) continue;
+                int64_t k = keys[lo + i];
+                uint32_t b = d_agg_bucket(k, nb);
+                unsigned int r = excl[b] + atomicAdd(&hist[b], 1u);
+                ulonglong2 rec;
+                rec.x = (unsigned long long)k;
+                rec.y = vals ? (unsigned long long)__double_as_longlong(
+                                   vals[lo + i])
+                             : (unsigned long long)(lo + i);
+                stage[r] = rec;
+            }
         }
         __syncthreads();
         int wtot = (int)tot;
@@ -2100,6 +2156,91 @@ __global__ void k_q9_filter(const int64_t *__restrict__ pk, int64_t n,
     }
 }
 
+/* phase 1 (tile variant, default): tile-staged compaction — the append_ab
+ * v6 pattern (profiles/r2_append_ab.txt: 0.22 vs 0.94 ms for the per-wave
+ * staged appender at 150 M rows). Per 8192-row tile: wave prefix sums →
+ * 16 wave totals scanned by thread 0 → stable ranks into an LDS stage →
+ * ONE cursor reservation per tile → linear write-out with consecutive
+ * lanes on consecutive addresses (the per-wave appender's run-strided
+ * writes, not its atomics, were the plateau). Survivor ids stay ascending
+ * within a tile; tiles land in arbitrary order (result-set parity only —
+ * the probe pass is order-free). */
+#define Q9T 8192
+__global__ __launch_bounds__(1024) void k_q9_filter_tile(
+    const int64_t *__restrict__ pk, int64_t n,
+    const unsigned long long *__restrict__ pbitmap, int64_t lo_k,
+    int64_t hi_k, uint32_t *__restrict__ hits, int64_t *nhits)
+{
+    __shared__ uint32_t stage[Q9T];
+    __shared__ int wtot[16];
+    __shared__ int woff[16];
+    __shared__ int sweepbase;
+    __shared__ long long gbase;
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    const v2l *pk2 = (const v2l *)pk;
+    int64_t ntiles = (n + Q9T - 1) / Q9T;
+    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+        int64_t tl = t * (int64_t)Q9T;
+        int64_t th = tl + Q9T < n ? tl + Q9T : n;
+        if (threadIdx.x == 0) sweepbase = 0;
+        __syncthreads();
+        for (int64_t s0 = tl; s0 < th; s0 += 4096) {
+            int64_t r0 = s0 + 4 * (int64_t)threadIdx.x;
+            bool m[4] = {false, false, false, false};
+            int mycnt = 0;
+            if (r0 + 3 < n) {
+                v2l pa = __builtin_nontemporal_load(&pk2[r0 / 2]);
+                v2l pb = __builtin_nontemporal_load(&pk2[r0 / 2 + 1]);
+                int64_t pks[4] = {pa.x, pa.y, pb.x, pb.y};
+#pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    int64_t pidx = pks[j] - 1 - lo_k;
+                    m[j] = pidx >= 0 && pidx < hi_k - lo_k &&
+                           ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
+                    mycnt += m[j];
+                }
+            } else {
+                for (int j = 0; j < 4 && r0 + j < n; j++) {
+                    int64_t pidx = pk[r0 + j] - 1 - lo_k;
+                    m[j] = pidx >= 0 && pidx < hi_k - lo_k &&
+                           ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
+                    mycnt += m[j];
+                }
+            }
+            int incl = mycnt;
+            for (int off = 1; off < WAVE; off <<= 1) {
+                int up = __shfl_up(incl, off, WAVE);
+                if (lane >= off) incl += up;
+            }
+            if (lane == WAVE - 1) wtot[wid] = incl;
+            __syncthreads();
+            if (threadIdx.x == 0) {
+                int acc = sweepbase;
+                for (int w = 0; w < 16; w++) {
+                    woff[w] = acc;
+                    acc += wtot[w];
+                }
+                sweepbase = acc;
+            }
+            __syncthreads();
+            int pos = woff[wid] + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j]) stage[pos++] = (uint32_t)(r0 + j);
+            __syncthreads();
+        }
+        int tot = sweepbase;
+        if (threadIdx.x == 0)
+            gbase = tot ? (long long)atomicAdd((unsigned long long *)nhits,
+                                               (unsigned long long)tot)
+                        : 0;
+        __syncthreads();
+        for (int p = threadIdx.x; p < tot; p += blockDim.x)
+            hits[gbase + p] = stage[p];
+        __syncthreads();
+    }
+}
+
 /* phase 2: dense pass over the survivors — orders date lookup + year
  * partial aggregate in per-lane registers (the Q1 pattern: 7-year domain,
  * compile-time indexed), wave+block reduce, one atomic per (year, block). */
@@ -2293,9 +2434,15 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
             HIP_CHECK(hipEventRecord(ev0, s));
             rec0 = true;
         }
-        hipLaunchKernelGGL(k_q9_filter, dim3(grid_for(l->n / 4, 256)),
-                           dim3(256), 0, s, l->l_partkey, l->n, pbitmap, lo,
-                           hi, hits, nhits);
+        const char *fw = getenv("OTBX_Q9_FILTER_WAVE");
+        if (fw && atoi(fw)) /* legacy per-wave appender (A/B) */
+            hipLaunchKernelGGL(k_q9_filter, dim3(grid_for(l->n / 4, 256)),
+                               dim3(256), 0, s, l->l_partkey, l->n, pbitmap,
+                               lo, hi, hits, nhits);
+        else
+            hipLaunchKernelGGL(k_q9_filter_tile, dim3(2048), dim3(1024), 0,
+                               s, l->l_partkey, l->n, pbitmap, lo, hi, hits,
+                               nhits);
         hipLaunchKernelGGL(k_q9_probe, dim3(grid_for(l->n / 8, 256)),
                            dim3(256), 0, s, *l, hits, nhits, dtab, mino,
                            orange, sums_dev, (unsigned long long *)counts_dev);
@@ -3745,6 +3892,96 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
         atomicAdd(nhits, myhits);
 }
 
+static inline bool q3_compact_legacy(void)
+{
+    const char *e = getenv("OTBX_Q3_COMPACT_LEGACY");
+    return e && atoi(e);
+}
+
+/* Tile-staged variant (default; append_ab v6 pattern, profiles/
+ * r2_append_ab.txt): per 8192-entry tile — count/rank via wave prefix sums,
+ * ONE group-counter reservation per tile, tile-relative survivor ids staged
+ * as u16 in LDS (16 KB, always fits), then a linear write-out with
+ * consecutive lanes on consecutive out positions; the rtab slice is L2-hot
+ * for the write-out's re-reads. The legacy kernel (below) re-reads its
+ * whole 73 k-row block chunk from HBM on the emit pass and writes
+ * per-thread runs. */
+#define Q3CT 8192
+__global__ __launch_bounds__(1024) void k_q3_compact_tile(
+    const double *__restrict__ rtab,
+    const unsigned long long *__restrict__ ptab, int64_t range, int64_t mino,
+    otbx_q3_group *out, int64_t cap_out, int64_t *ngroups)
+{
+    __shared__ uint16_t stage[Q3CT];
+    __shared__ int wtot[16];
+    __shared__ int woff[16];
+    __shared__ int sweepbase;
+    __shared__ long long gbase;
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    const v2d *rt2 = (const v2d *)rtab;
+    int64_t ntiles = (range + Q3CT - 1) / Q3CT;
+    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+        int64_t tl = t * (int64_t)Q3CT;
+        int64_t th = tl + Q3CT < range ? tl + Q3CT : range;
+        if (threadIdx.x == 0) sweepbase = 0;
+        __syncthreads();
+        for (int64_t s0 = tl; s0 < th; s0 += 4096) {
+            int64_t r0 = s0 + 4 * (int64_t)threadIdx.x;
+            bool m[4] = {false, false, false, false};
+            int mycnt = 0;
+            if (r0 + 3 < range) {
+                v2d ra = rt2[r0 / 2], rb = rt2[r0 / 2 + 1];
+                m[0] = ra.x != 0.0; m[1] = ra.y != 0.0;
+                m[2] = rb.x != 0.0; m[3] = rb.y != 0.0;
+                mycnt = m[0] + m[1] + m[2] + m[3];
+            } else {
+                for (int j = 0; j < 4 && r0 + j < range; j++) {
+                    m[j] = rtab[r0 + j] != 0.0;
+                    mycnt += m[j];
+                }
+            }
+            int incl = mycnt;
+            for (int off = 1; off < WAVE; off <<= 1) {
+                int up = __shfl_up(incl, off, WAVE);
+                if (lane >= off) incl += up;
+            }
+            if (lane == WAVE - 1) wtot[wid] = incl;
+            __syncthreads();
+            if (threadIdx.x == 0) {
+                int acc = sweepbase;
+                for (int w = 0; w < 16; w++) {
+                    woff[w] = acc;
+                    acc += wtot[w];
+                }
+                sweepbase = acc;
+            }
+            __syncthreads();
+            int pos = woff[wid] + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j]) stage[pos++] = (uint16_t)(r0 + j - tl);
+            __syncthreads();
+        }
+        int tot = sweepbase;
+        if (threadIdx.x == 0)
+            gbase = tot ? (long long)atomicAdd((unsigned long long *)ngroups,
+                                               (unsigned long long)tot)
+                        : 0;
+        __syncthreads();
+        for (int p = threadIdx.x; p < tot; p += blockDim.x) {
+            int64_t gp = gbase + p;
+            if (gp >= cap_out) continue;
+            int64_t i = tl + (int64_t)stage[p];
+            unsigned long long pl = ptab[i];
+            out[gp].l_orderkey = mino + i;
+            out[gp].revenue = rtab[i];
+            out[gp].o_orderdate = (int32_t)(pl & 0xffffffffull);
+            out[gp].o_shippriority = (int32_t)(pl >> 32);
+        }
+        __syncthreads();
+    }
+}
+
 __global__ void k_q3_compact_direct(const double *__restrict__ rtab,
                                     const unsigned long long *__restrict__ ptab,
                                     int64_t range, int64_t mino,
@@ -4374,10 +4611,15 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                            q3date, mino_all, range_all, dbitmap, dptab, drtab,
                            nhits);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
-        hipLaunchKernelGGL(k_q3_compact_direct,
-                           dim3(grid_for(range_all, 256)), dim3(256), 0, s,
-                           drtab, dptab, range_all, mino_all, groups_dev,
-                           cap_groups, ngroups_dev);
+        if (q3_compact_legacy())
+            hipLaunchKernelGGL(k_q3_compact_direct,
+                               dim3(grid_for(range_all, 256)), dim3(256), 0,
+                               s, drtab, dptab, range_all, mino_all,
+                               groups_dev, cap_groups, ngroups_dev);
+        else
+            hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024), 0,
+                               s, drtab, dptab, range_all, mino_all,
+                               groups_dev, cap_groups, ngroups_dev);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
         goto emit;
     }
@@ -4439,10 +4681,15 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                 HIP_CHECK(hipEventRecord(ev[3], s));
                 rec3 = true;
             }
-            hipLaunchKernelGGL(k_q3_compact_direct,
-                               dim3(grid_for(prange, 256)), dim3(256), 0, s,
-                               drtab, dptab, prange, pmin, groups_dev,
-                               cap_groups, ngroups_dev);
+            if (q3_compact_legacy())
+                hipLaunchKernelGGL(k_q3_compact_direct,
+                                   dim3(grid_for(prange, 256)), dim3(256), 0,
+                                   s, drtab, dptab, prange, pmin, groups_dev,
+                                   cap_groups, ngroups_dev);
+            else
+                hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024),
+                                   0, s, drtab, dptab, prange, pmin,
+                                   groups_dev, cap_groups, ngroups_dev);
         }
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
     } else {

@@ -928,3 +928,51 @@ def test_join_overflow_contract(ex):
     node.BeginCustomScan()
     with pytest.raises(ex.OtbxError):
         node._run()
+
+
+def test_q9_filter_tile_vs_wave(ex, monkeypatch):
+    """Tile-staged Q9 part filter (append_ab v6 pattern, default) vs the
+    legacy per-wave appender on the same tables: counts exact, sums within
+    tree-rounding tolerance (survivor order differs)."""
+    n = 2_000_000
+    li = ex.GpuLineitem.generate(n, with_partkey=True)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    pt = ex.GpuPart.generate(n // 30)
+
+    def run():
+        node = ex.GpuQ9Fragment(pt, od, li)
+        node.BeginCustomScan()
+        node._run()
+        s, c = node.partial_state_tensors()
+        return s.cpu().numpy().copy(), c.cpu().numpy().copy()
+
+    s_tile, c_tile = run()
+    monkeypatch.setenv("OTBX_Q9_FILTER_WAVE", "1")
+    s_wave, c_wave = run()
+    assert np.array_equal(c_tile, c_wave)
+    for a, b in zip(s_tile, s_wave):
+        assert abs(a - b) <= 1e-9 * max(abs(b), 1.0)
+
+
+def test_q3_compact_tile_vs_legacy(ex, monkeypatch):
+    """Tile-staged Q3 group compaction (default) vs the legacy block-chunk
+    kernel: identical group sets (revenues bit-exact — compaction reads the
+    same finished rtab either way)."""
+    n = 1_000_000
+    li = ex.GpuLineitem.generate(n)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    cu = ex.GpuCustomer.generate(n // 40)
+
+    def run():
+        node = ex.GpuQ3Fragment(cu, od, li)
+        node.BeginCustomScan()
+        node._run()
+        g = node.fetch_groups()
+        return sorted((int(r["l_orderkey"]), float(r["revenue"]),
+                       int(r["o_orderdate"]), int(r["o_shippriority"]))
+                      for r in g)
+
+    tile = run()
+    monkeypatch.setenv("OTBX_Q3_COMPACT_LEGACY", "1")
+    legacy = run()
+    assert tile == legacy
