@@ -3741,7 +3741,7 @@ __global__ void k_ord_filter_insert_fused(
     const unsigned long long *__restrict__ cbitmap, int64_t cmin,
     int64_t crange, int64_t mino, int64_t range,
     unsigned long long *__restrict__ bitmap,
-    unsigned long long *__restrict__ ptab)
+    unsigned long long *__restrict__ ptab, double *__restrict__ rtab)
 {
     int64_t nq = o.n / 4;
     const int4 *od4 = (const int4 *)o.o_orderdate;
@@ -3776,6 +3776,8 @@ __global__ void k_ord_filter_insert_fused(
             atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
             ptab[idx] = (unsigned long long)(uint32_t)ds[j] |
                         ((unsigned long long)(uint32_t)prio[j] << 32);
+            rtab[idx] = 0.0; /* survivor-slot zeroing (see
+                              * k_ord_insert_direct) */
         }
     }
     /* tail rows */
@@ -3807,7 +3809,7 @@ __global__ void k_ord_insert_direct(const otbx_orders_dev o,
                                     const int64_t *__restrict__ ncand_p,
                                     int64_t mino, int64_t range,
                                     unsigned long long *bitmap,
-                                    unsigned long long *ptab)
+                                    unsigned long long *ptab, double *rtab)
 {
     int64_t n = *ncand_p;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -3820,6 +3822,12 @@ __global__ void k_ord_insert_direct(const otbx_orders_dev o,
         atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
         ptab[idx] = (unsigned long long)(uint32_t)o.o_orderdate[i] |
                     ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
+        if (rtab)
+            rtab[idx] = 0.0; /* survivor-slot zeroing: replaces the whole-
+                              * range memset (orderkeys unique -> one writer
+                              * per idx; entries at unset-bit positions are
+                              * never read: compaction is
+                              * bitmap-authoritative) */
     }
 }
 
@@ -3909,8 +3917,9 @@ static inline bool q3_compact_legacy(void)
 #define Q3CT 8192
 __global__ __launch_bounds__(1024) void k_q3_compact_tile(
     const double *__restrict__ rtab,
-    const unsigned long long *__restrict__ ptab, int64_t range, int64_t mino,
-    otbx_q3_group *out, int64_t cap_out, int64_t *ngroups)
+    const unsigned long long *__restrict__ ptab,
+    const unsigned long long *__restrict__ bitmap, int64_t range,
+    int64_t mino, otbx_q3_group *out, int64_t cap_out, int64_t *ngroups)
 {
     __shared__ uint16_t stage[Q3CT];
     __shared__ int wtot[16];
@@ -3929,14 +3938,27 @@ __global__ __launch_bounds__(1024) void k_q3_compact_tile(
             int64_t r0 = s0 + 4 * (int64_t)threadIdx.x;
             bool m[4] = {false, false, false, false};
             int mycnt = 0;
+            /* bitmap is the presence authority (rtab/ptab are NOT
+             * memset per step; stale entries live at unset-bit
+             * positions); rtab != 0 then drops build rows with no
+             * probe match (revenue > 0 by domain: price > 0, disc < 1) */
             if (r0 + 3 < range) {
-                v2d ra = rt2[r0 / 2], rb = rt2[r0 / 2 + 1];
-                m[0] = ra.x != 0.0; m[1] = ra.y != 0.0;
-                m[2] = rb.x != 0.0; m[3] = rb.y != 0.0;
-                mycnt = m[0] + m[1] + m[2] + m[3];
+                unsigned long long w = bitmap[r0 >> 6] >> (r0 & 63);
+                if ((w & 0xfull) == 0) {
+                    /* fast skip: no bits set for these 4 entries */
+                } else {
+                    v2d ra = rt2[r0 / 2], rb = rt2[r0 / 2 + 1];
+                    m[0] = (w & 1ull) && ra.x != 0.0;
+                    m[1] = (w & 2ull) && ra.y != 0.0;
+                    m[2] = (w & 4ull) && rb.x != 0.0;
+                    m[3] = (w & 8ull) && rb.y != 0.0;
+                    mycnt = m[0] + m[1] + m[2] + m[3];
+                }
             } else {
                 for (int j = 0; j < 4 && r0 + j < range; j++) {
-                    m[j] = rtab[r0 + j] != 0.0;
+                    int64_t i = r0 + j;
+                    m[j] = ((bitmap[i >> 6] >> (i & 63)) & 1ull) &&
+                           rtab[i] != 0.0;
                     mycnt += m[j];
                 }
             }
@@ -4599,12 +4621,16 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     if (o->n > 0 && range_all > 0 && range_all <= dcap && !force_hash) {
         HIP_CHECK(hipMemsetAsync(dbitmap, 0,
                                  (size_t)(range_all / 64 + 8) * 8, s));
-        HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)range_all * 8, s));
-        HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)range_all * 8, s));
+        if (q3_compact_legacy()) {
+            /* legacy compaction scans rtab != 0 over the whole range, so
+             * it needs the full-range zeroing the default path skips */
+            HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)range_all * 8, s));
+            HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)range_all * 8, s));
+        }
         hipLaunchKernelGGL(k_ord_filter_insert_fused,
                            dim3(grid_for(o->n / 4, 256)), dim3(256), 0, s, *o,
                            q3date, ctab, ccap, cbitmap, cmin, crange,
-                           mino_all, range_all, dbitmap, dptab);
+                           mino_all, range_all, dbitmap, dptab, drtab);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
         hipLaunchKernelGGL(k_q3_scan_probe_agg_direct,
                            dim3(grid_for(l->n / 4, 256)), dim3(256), 0, s, *l,
@@ -4618,7 +4644,7 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                groups_dev, cap_groups, ngroups_dev);
         else
             hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024), 0,
-                               s, drtab, dptab, range_all, mino_all,
+                               s, drtab, dptab, dbitmap, range_all, mino_all,
                                groups_dev, cap_groups, ngroups_dev);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
         goto emit;
@@ -4662,13 +4688,16 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                                         : dcap;
             HIP_CHECK(hipMemsetAsync(dbitmap, 0,
                                      (size_t)(prange / 64 + 8) * 8, s));
-            HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)prange * 8, s));
-            HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)prange * 8, s));
+            if (q3_compact_legacy()) {
+                HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)prange * 8, s));
+                HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)prange * 8, s));
+            }
             if (pass > 0)
                 HIP_CHECK(hipMemsetAsync(&hdr[2], 0, 8, s)); /* reset cands */
             hipLaunchKernelGGL(k_ord_insert_direct, dim3(grid_for(o->n, 256)),
                                dim3(256), 0, s, *o, cand_o2, &hdr[1], pmin,
-                               prange, dbitmap, dptab);
+                               prange, dbitmap, dptab,
+                               q3_compact_legacy() ? NULL : drtab);
             if (kernel_ms && !rec2) {
                 HIP_CHECK(hipEventRecord(ev[2], s));
                 rec2 = true;
@@ -4688,7 +4717,7 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                    cap_groups, ngroups_dev);
             else
                 hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024),
-                                   0, s, drtab, dptab, prange, pmin,
+                                   0, s, drtab, dptab, dbitmap, prange, pmin,
                                    groups_dev, cap_groups, ngroups_dev);
         }
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));

@@ -721,3 +721,206 @@ int64_t ora_q3_topk(ora_q3_row *rows, int64_t n, int64_t k)
     qsort(rows, (size_t)n, sizeof(ora_q3_row), q3_cmp);
     return n < k ? n : k;
 }
+
+/* ================= extended joins (ora_join_ext) ================= */
+/* Restates the HJ_* FSM (nodeHashjoin.c:139-144) incl. the fill states —
+ * see oracle.h for the per-state citations. Chained-bucket table as
+ * ora_join_i64, generalized to an optional second key (multi-key combine
+ * semantics nodeHash.c:2059 restated at 64 bit). */
+
+static inline int ext_rownull(const uint8_t *n1, const uint8_t *n2,
+                              int has_k2, int64_t i)
+{
+    return (n1 && n1[i]) || (has_k2 && n2 && n2[i]);
+}
+
+static inline uint64_t ext_hash(int64_t k1, int64_t k2, int has_k2)
+{
+    uint64_t h = ora_hash_i64(k1);
+    if (has_k2) {
+        h = (h << 1) | (h >> 63);   /* rotate-left-1 (nodeHash.c:2059) */
+        h ^= ora_hash_i64(k2);
+    }
+    return h;
+}
+
+static ora_status ext_push(int64_t **bi, int64_t **pi, int64_t *n,
+                           int64_t *cap, int64_t b, int64_t p)
+{
+    if (*n == *cap) {
+        *cap *= 2;
+        int64_t *nb = realloc(*bi, (size_t)*cap * 8);
+        int64_t *np = realloc(*pi, (size_t)*cap * 8);
+        if (nb) *bi = nb;
+        if (np) *pi = np;
+        if (!nb || !np) return ORA_ERR_OOM;
+    }
+    (*bi)[*n] = b;
+    (*pi)[*n] = p;
+    (*n)++;
+    return ORA_OK;
+}
+
+ora_status ora_join_ext(const int64_t *bk1, const uint8_t *bn1,
+                        const int64_t *bk2, const uint8_t *bn2, int64_t nb,
+                        const int64_t *pk1, const uint8_t *pn1,
+                        const int64_t *pk2, const uint8_t *pn2, int64_t np,
+                        int join_type,
+                        int64_t **out_bidx, int64_t **out_pidx, int64_t *nout)
+{
+    const int has_k2 = bk2 != NULL;
+    const int jt = join_type;
+    if (jt < 0 || jt > 5 || (has_k2 != (pk2 != NULL)))
+        return ORA_ERR_INVALID;
+    const int fill_probe = (jt == 1 || jt == 5);      /* left, full  */
+    const int fill_build = (jt == 4 || jt == 5);      /* right, full */
+    const int emit_match = (jt == 0 || jt == 1 || jt == 4 || jt == 5);
+
+    /* chained-bucket build over non-null rows (ExecHashTableInsert
+     * nodeHash.c:1828; NULL keys dropped, nodeHash.c:2026) */
+    int64_t nbuckets = next_pow2(nb < 16 ? 16 : nb);
+    int64_t *head = malloc((size_t)nbuckets * 8);
+    int64_t *next = malloc((size_t)(nb > 0 ? nb : 1) * 8);
+    uint8_t *matched = NULL;
+    if (fill_build) matched = calloc((size_t)(nb > 0 ? nb : 1), 1);
+    if (!head || !next || (fill_build && !matched)) {
+        free(head); free(next); free(matched);
+        return ORA_ERR_OOM;
+    }
+    for (int64_t i = 0; i < nbuckets; i++) head[i] = -1;
+    for (int64_t i = 0; i < nb; i++) {
+        if (ext_rownull(bn1, bn2, has_k2, i)) continue;
+        int64_t b = (int64_t)(ext_hash(bk1[i], has_k2 ? bk2[i] : 0, has_k2) &
+                              (uint64_t)(nbuckets - 1));
+        next[i] = head[b];
+        head[b] = i;
+    }
+
+    int64_t cap = 64, n = 0;
+    int64_t *bi = malloc(cap * 8), *pi = malloc(cap * 8);
+    ora_status st = ORA_OK;
+    if (!bi || !pi) { st = ORA_ERR_OOM; goto done; }
+
+    for (int64_t p = 0; p < np; p++) {
+        if (ext_rownull(pn1, pn2, has_k2, p)) {
+            /* NULL-key outer tuple: cannot match; left/full fill
+             * (nodeHashjoin.c:543-552), anti emits (no match exists) */
+            if (fill_probe || jt == 3)
+                if ((st = ext_push(&bi, &pi, &n, &cap, -1, p))) goto done;
+            continue;
+        }
+        int64_t k1 = pk1[p], k2 = has_k2 ? pk2[p] : 0;
+        int64_t bkt = (int64_t)(ext_hash(k1, k2, has_k2) &
+                                (uint64_t)(nbuckets - 1));
+        int64_t nmatch = 0;
+        for (int64_t m = head[bkt]; m >= 0; m = next[m]) {
+            if (bk1[m] != k1 || (has_k2 && bk2[m] != k2)) continue;
+            nmatch++;
+            if (matched) matched[m] = 1;
+            if (emit_match) {
+                if ((st = ext_push(&bi, &pi, &n, &cap, m, p))) goto done;
+            }
+            if (jt == 2) break;     /* JOIN_SEMI: first match suffices
+                                     * (nodeHashjoin.c:572) */
+        }
+        if (jt == 2 && nmatch > 0) {
+            if ((st = ext_push(&bi, &pi, &n, &cap, -1, p))) goto done;
+        } else if ((jt == 3 || fill_probe) && nmatch == 0) {
+            if ((st = ext_push(&bi, &pi, &n, &cap, -1, p))) goto done;
+        }
+    }
+
+    /* HJ_FILL_INNER_TUPLES (nodeHashjoin.c:693) / unmatched-build scan
+     * (ExecScanHashTableForUnmatched nodeHash.c:2322): NULL-key build rows
+     * never entered the table, so they are unmatched by construction */
+    if (fill_build)
+        for (int64_t i = 0; i < nb; i++)
+            if (ext_rownull(bn1, bn2, has_k2, i) || !matched[i])
+                if ((st = ext_push(&bi, &pi, &n, &cap, i, -1))) goto done;
+
+done:
+    free(head); free(next); free(matched);
+    if (st) { free(bi); free(pi); return st; }
+    *out_bidx = bi; *out_pidx = pi; *nout = n;
+    return ORA_OK;
+}
+
+/* ================= two-key hash aggregate (ora_agg_i64x2) ================= */
+
+static int agg_group2_cmp(const void *pa, const void *pb)
+{
+    const ora_agg_group2 *a = pa, *b = pb;
+    if (a->key1_isnull != b->key1_isnull) return a->key1_isnull ? 1 : -1;
+    if (!a->key1_isnull) {
+        if (a->key1 != b->key1) return a->key1 < b->key1 ? -1 : 1;
+    }
+    if (a->key2_isnull != b->key2_isnull) return a->key2_isnull ? 1 : -1;
+    if (!a->key2_isnull) {
+        if (a->key2 != b->key2) return a->key2 < b->key2 ? -1 : 1;
+    }
+    return 0;
+}
+
+ora_status ora_agg_i64x2(const int64_t *k1, const uint8_t *k1null,
+                         const int64_t *k2, const uint8_t *k2null,
+                         const double *vals, const uint8_t *val_null,
+                         int64_t n, ora_agg_group2 **out, int64_t *ngroups)
+{
+    int64_t cap = next_pow2(n < 16 ? 16 : (int64_t)((double)n / 0.85) + 1);
+    ora_agg_group2 *slots = calloc(cap, sizeof(ora_agg_group2));
+    uint8_t *used = calloc(cap, 1);
+    if (!slots || !used) { free(slots); free(used); return ORA_ERR_OOM; }
+    int64_t ng = 0;
+    ora_status st;
+
+    for (int64_t i = 0; i < n; i++) {
+        int n1 = k1null && k1null[i], n2 = k2null && k2null[i];
+        int64_t a = n1 ? 0 : k1[i], b = n2 ? 0 : k2[i];
+        /* LookupTupleHashEntry (execGrouping.c:295): NULL==NULL for
+         * grouping (:525); per-key hash with NULL sentinel, combined by
+         * rotate-left-1 + xor (nodeHash.c:2059 restated) */
+        uint64_t h1 = n1 ? 0x9e3779b97f4a7c15ull : ora_hash_i64(a);
+        uint64_t h2 = n2 ? 0xc2b2ae3d27d4eb4full : ora_hash_i64(b);
+        uint64_t h = ((h1 << 1) | (h1 >> 63)) ^ h2;
+        int64_t s = (int64_t)(h & (uint64_t)(cap - 1));
+        for (;;) {
+            if (!used[s]) {
+                used[s] = 1;
+                slots[s].key1 = a; slots[s].key1_isnull = n1;
+                slots[s].key2 = b; slots[s].key2_isnull = n2;
+                slots[s].sum_isnull = 1;
+                ng++;
+                break;
+            }
+            if (slots[s].key1_isnull == n1 && slots[s].key2_isnull == n2 &&
+                (n1 || slots[s].key1 == a) && (n2 || slots[s].key2 == b))
+                break;
+            s = (s + 1) & (cap - 1);
+        }
+        ora_agg_group2 *e = &slots[s];
+        if ((st = int8inc(&e->count_star))) goto fail;
+        int vnull = val_null && val_null[i];
+        if (!vnull) {
+            if ((st = int8inc(&e->count_v))) goto fail;
+            if (e->sum_isnull) {
+                e->sum_v = vals[i];
+                e->sum_isnull = 0;
+            } else if ((st = float8pl(e->sum_v, vals[i], &e->sum_v))) goto fail;
+            if ((st = float8_accum(e->acc, vals[i]))) goto fail;
+        }
+    }
+
+    ora_agg_group2 *res = malloc((size_t)(ng > 0 ? ng : 1) *
+                                 sizeof(ora_agg_group2));
+    if (!res) { st = ORA_ERR_OOM; goto fail; }
+    int64_t j = 0;
+    for (int64_t s = 0; s < cap; s++)
+        if (used[s]) res[j++] = slots[s];
+    qsort(res, (size_t)ng, sizeof(ora_agg_group2), agg_group2_cmp);
+    free(slots); free(used);
+    *out = res; *ngroups = ng;
+    return ORA_OK;
+fail:
+    free(slots); free(used);
+    return st;
+}

@@ -158,6 +158,53 @@ ora_status ora_join_i64(const int64_t *bkeys, const uint8_t *bnull, int64_t nb,
                         const int64_t *pkeys, const uint8_t *pnull, int64_t np,
                         int64_t **out_bidx, int64_t **out_pidx, int64_t *nout);
 
+
+
+/* ---- extended join types (nodeHashjoin.c FSM fill states) ----
+ * join_type: 0 inner, 1 left, 2 semi, 3 anti, 4 right, 5 full — restating
+ * the HJ_* state machine (nodeHashjoin.c:139-144):
+ *   inner match                       -> (bidx, pidx)
+ *   left/full unmatched probe row     -> (-1, pidx)   HJ_FILL_OUTER_TUPLE
+ *     (incl. NULL-key probe rows)        (nodeHashjoin.c:142,668; NULL-key
+ *                                         outer path :543-552)
+ *   right/full unmatched build row    -> (bidx, -1)   HJ_FILL_INNER_TUPLES
+ *     (incl. NULL-key build rows)        (nodeHashjoin.c:143,693;
+ *                                         ExecScanHashTableForUnmatched
+ *                                         nodeHash.c:2322)
+ *   semi: (-1, pidx) once per probe row with >=1 match (JOIN_SEMI advances
+ *     to the next outer after the first match, nodeHashjoin.c:572)
+ *   anti: (-1, pidx) per probe row with no match (nodeHashjoin.c:631)
+ * bk2/pk2 NULL => single-key join. With two keys, the join key is the ROW
+ * (k1,k2): a row with EITHER key NULL never matches (strict equality,
+ * ExecHashGetHashValue nodeHash.c:2026 keep_nulls semantics; multi-key
+ * hash combine = rotate-left-1 then xor, nodeHash.c:2059 — restated at
+ * 64 bit; parity is on result sets, not hash values). Caller frees. */
+ora_status ora_join_ext(const int64_t *bk1, const uint8_t *bn1,
+                        const int64_t *bk2, const uint8_t *bn2, int64_t nb,
+                        const int64_t *pk1, const uint8_t *pn1,
+                        const int64_t *pk2, const uint8_t *pn2, int64_t np,
+                        int join_type,
+                        int64_t **out_bidx, int64_t **out_pidx, int64_t *nout);
+
+/* ---- two-key hash aggregate (multi-key GROUP BY) ----
+ * Group identity is (k1_isnull, k1, k2_isnull, k2) with NULL==NULL for
+ * grouping (execGrouping.c:295 + null-match semantics :525); aggregates as
+ * ora_agg_i64. Emitted sorted by (k1_isnull, k1, k2_isnull, k2). */
+typedef struct {
+    int64_t key1, key2;
+    int key1_isnull, key2_isnull;
+    int64_t count_star;
+    int64_t count_v;
+    double sum_v;
+    int sum_isnull;
+    double acc[3];
+} ora_agg_group2;
+
+ora_status ora_agg_i64x2(const int64_t *k1, const uint8_t *k1null,
+                         const int64_t *k2, const uint8_t *k2null,
+                         const double *vals, const uint8_t *val_null,
+                         int64_t n, ora_agg_group2 **out, int64_t *ngroups);
+
 #ifdef __cplusplus
 }
 #endif

@@ -107,6 +107,8 @@ def load():
     lib.ora_q3_topk.restype = C.c_int64
     lib.ora_agg_i64.restype = C.c_int
     lib.ora_join_i64.restype = C.c_int
+    lib.ora_join_ext.restype = C.c_int
+    lib.ora_agg_i64x2.restype = C.c_int
     lib.ora_q1_combine.restype = C.c_int
     return lib
 
@@ -295,3 +297,63 @@ def join_i64(bkeys, pkeys, bnull=None, pnull=None):
     bi = np.ctypeslib.as_array(ob, (nn,)).copy() if nn else np.empty(0, np.int64)
     pi = np.ctypeslib.as_array(op, (nn,)).copy() if nn else np.empty(0, np.int64)
     return bi, pi
+
+
+class OraAggGroup2(C.Structure):
+    _fields_ = [
+        ("key1", C.c_int64), ("key2", C.c_int64),
+        ("key1_isnull", C.c_int), ("key2_isnull", C.c_int),
+        ("count_star", C.c_int64), ("count_v", C.c_int64),
+        ("sum_v", C.c_double), ("sum_isnull", C.c_int),
+        ("acc", C.c_double * 3),
+    ]
+
+
+def join_ext(bkeys, pkeys, join_type, bnull=None, pnull=None,
+             bkeys2=None, pkeys2=None, bnull2=None, pnull2=None):
+    """Extended join (oracle.h ora_join_ext): join_type 0 inner / 1 left /
+    2 semi / 3 anti / 4 right / 5 full; optional second key column."""
+    L = lib()
+    bkeys = np.ascontiguousarray(bkeys, dtype=np.int64)
+    pkeys = np.ascontiguousarray(pkeys, dtype=np.int64)
+    arrs = {}
+    for nm, a, dt in [("bn", bnull, np.uint8), ("pn", pnull, np.uint8),
+                      ("bk2", bkeys2, np.int64), ("pk2", pkeys2, np.int64),
+                      ("bn2", bnull2, np.uint8), ("pn2", pnull2, np.uint8)]:
+        arrs[nm] = None if a is None else np.ascontiguousarray(a, dtype=dt)
+    ob = C.POINTER(C.c_int64)()
+    op = C.POINTER(C.c_int64)()
+    n = C.c_int64(0)
+    st = L.ora_join_ext(
+        _p(bkeys, C.c_int64), _p(arrs["bn"], C.c_uint8),
+        _p(arrs["bk2"], C.c_int64), _p(arrs["bn2"], C.c_uint8),
+        C.c_int64(len(bkeys)),
+        _p(pkeys, C.c_int64), _p(arrs["pn"], C.c_uint8),
+        _p(arrs["pk2"], C.c_int64), _p(arrs["pn2"], C.c_uint8),
+        C.c_int64(len(pkeys)), C.c_int(join_type),
+        C.byref(ob), C.byref(op), C.byref(n))
+    assert st == 0, st
+    nn = n.value
+    bi = np.ctypeslib.as_array(ob, (nn,)).copy() if nn else np.empty(0, np.int64)
+    pi = np.ctypeslib.as_array(op, (nn,)).copy() if nn else np.empty(0, np.int64)
+    return bi, pi
+
+
+def agg_i64x2(k1, k2, vals, k1null=None, k2null=None, val_null=None):
+    """Two-key hash aggregate (oracle.h ora_agg_i64x2); returns a list of
+    OraAggGroup2 sorted by (k1_isnull, k1, k2_isnull, k2)."""
+    L = lib()
+    k1 = np.ascontiguousarray(k1, dtype=np.int64)
+    k2 = np.ascontiguousarray(k2, dtype=np.int64)
+    vals = np.ascontiguousarray(vals, dtype=np.float64)
+    n1 = None if k1null is None else np.ascontiguousarray(k1null, np.uint8)
+    n2 = None if k2null is None else np.ascontiguousarray(k2null, np.uint8)
+    vn = None if val_null is None else np.ascontiguousarray(val_null, np.uint8)
+    out = C.POINTER(OraAggGroup2)()
+    ng = C.c_int64(0)
+    st = L.ora_agg_i64x2(_p(k1, C.c_int64), _p(n1, C.c_uint8),
+                         _p(k2, C.c_int64), _p(n2, C.c_uint8),
+                         _p(vals, C.c_double), _p(vn, C.c_uint8),
+                         C.c_int64(len(k1)), C.byref(out), C.byref(ng))
+    assert st == 0, st
+    return [out[i] for i in range(ng.value)]
