@@ -281,6 +281,71 @@ otbx_status otbx_agg_i64(const int64_t *keys_dev, const uint8_t *key_null_dev,
                          otbx_agg_group *groups_dev, int64_t *ngroups_dev,
                          void *stream);
 
+/* ---- extended join types + two-key variants ----
+ * The HJ_* fill-state FSM (executor/nodeHashjoin.c:139-144) on the generic
+ * join, and multi-key (2 x i64) variants of join and group-by.
+ * join_type: 0 inner, 1 left, 2 semi, 3 anti, 4 right, 5 full.
+ * Pair encoding in (out_bidx, out_pidx):
+ *   match                          -> (bidx, pidx)
+ *   left/full unmatched probe row  -> (-1, pidx)  [HJ_FILL_OUTER_TUPLE,
+ *     incl. NULL-key probe rows]      nodeHashjoin.c:142,668]
+ *   right/full unmatched build row -> (bidx, -1)  [HJ_FILL_INNER_TUPLES,
+ *     incl. NULL-key build rows]      nodeHashjoin.c:143,693;
+ *                                     ExecScanHashTableForUnmatched
+ *                                     nodeHash.c:2322]
+ *   semi: (-1, pidx) once per probe row with >= 1 match (JOIN_SEMI
+ *     advances after the first match, nodeHashjoin.c:572)
+ *   anti: (-1, pidx) per probe row with no match (nodeHashjoin.c:631)
+ * Two-key variants join/group on the ROW (k1,k2): a row with EITHER key
+ * NULL never matches (strict equality; multi-key hash combine =
+ * rotate-left-1 xor, nodeHash.c:2059, restated at 64 bit — parity is on
+ * result sets). Overflow contract identical to otbx_join_i64. */
+otbx_status otbx_join_ext_workspace_bytes(int64_t nb, int64_t np,
+                                          size_t *bytes);
+otbx_status otbx_join_i64_ext(const int64_t *bkeys_dev,
+                              const uint8_t *bnull_dev, int64_t nb,
+                              const int64_t *pkeys_dev,
+                              const uint8_t *pnull_dev, int64_t np,
+                              int32_t join_type, void *ws_dev,
+                              size_t ws_bytes, int64_t *out_bidx_dev,
+                              int64_t *out_pidx_dev, int64_t cap_pairs,
+                              int64_t *npairs_dev, void *stream);
+otbx_status otbx_join_i64x2(const int64_t *bk1_dev, const uint8_t *bn1_dev,
+                            const int64_t *bk2_dev, const uint8_t *bn2_dev,
+                            int64_t nb, const int64_t *pk1_dev,
+                            const uint8_t *pn1_dev, const int64_t *pk2_dev,
+                            const uint8_t *pn2_dev, int64_t np,
+                            int32_t join_type, void *ws_dev, size_t ws_bytes,
+                            int64_t *out_bidx_dev, int64_t *out_pidx_dev,
+                            int64_t cap_pairs, int64_t *npairs_dev,
+                            void *stream);
+
+/* two-key hash aggregate: group by (nullable k1, nullable k2) with
+ * NULL==NULL grouping (execGrouping.c:295,:525); aggregates as
+ * otbx_agg_i64. groups_dev capacity = n rows; *ngroups_dev receives the
+ * group count; emission order is arbitrary (hash-table order, as the
+ * reference's simplehash iteration). */
+typedef struct {
+    int64_t key1;
+    int64_t key2;
+    int64_t count_star;
+    int64_t count_v;
+    double sum_v;
+    int32_t key1_isnull;
+    int32_t key2_isnull;
+    int32_t sum_isnull;
+    int32_t _pad;
+} otbx_agg2_group; /* 56 B */
+
+otbx_status otbx_agg_i64x2_workspace_bytes(int64_t n, size_t *bytes);
+otbx_status otbx_agg_i64x2(const int64_t *k1_dev, const uint8_t *k1null_dev,
+                           const int64_t *k2_dev, const uint8_t *k2null_dev,
+                           const double *vals_dev,
+                           const uint8_t *val_null_dev, int64_t n,
+                           void *ws_dev, size_t ws_bytes,
+                           otbx_agg2_group *groups_dev, int64_t *ngroups_dev,
+                           void *stream);
+
 /* ---- GPU ORDER BY ----
  * Full sort of Q3 group rows by (revenue DESC, o_orderdate ASC) — the
  * tuplesort.c analog for the no-LIMIT ORDER BY case (LIMIT queries use

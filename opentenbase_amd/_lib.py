@@ -114,4 +114,6 @@ EXPORTED_SYMBOLS = [
     "otbx_partition_by_key", "otbx_gather_i64", "otbx_gather_f64",
     "otbx_gather_i32", "otbx_gather_u8",
     "otbx_join_i64_workspace_bytes", "otbx_join_i64",
+    "otbx_join_ext_workspace_bytes", "otbx_join_i64_ext", "otbx_join_i64x2",
+    "otbx_agg_i64x2_workspace_bytes", "otbx_agg_i64x2",
 ]

@@ -4870,3 +4870,451 @@ extern "C" otbx_status otbx_topk_by_revenue(const otbx_q3_group *groups,
 }
 
 } /* extern "C" */
+
+/* ================= extended joins (otbx_join_i64_ext / _i64x2) ===========
+ * The HJ_* fill-state FSM (nodeHashjoin.c:139-144) on the generic open-
+ * addressing join: inner/left/semi/anti/right/full, optionally on a
+ * two-column key (multi-key hash combine nodeHash.c:2059 restated at
+ * 64 bit — rotate-left-1 then xor; parity is on result sets).
+ * Pair encoding (matches oracle/oracle.h ora_join_ext):
+ *   match -> (bidx, pidx); left/full unmatched (or NULL-key) probe row and
+ *   every semi/anti emission -> (-1, pidx); right/full unmatched (or
+ *   NULL-key) build row -> (bidx, -1).
+ * Same overflow contract as otbx_join_i64 (npairs = TRUE count).
+ * Generality tier: correctness-first kernels, not the Q3 roofline path. */
+
+struct joinx_slot {
+    long long idx; /* claim word: -1 empty, else build row index */
+    long long k1;
+    long long k2;
+};
+
+__global__ void k_joinx_init(joinx_slot *tab, int64_t cap)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < cap;
+         i += stride)
+        tab[i].idx = -1;
+}
+
+__device__ __forceinline__ uint64_t d_jx_hash(int64_t k1, int64_t k2, int nk)
+{
+    uint64_t h = d_hash_i64(k1);
+    if (nk == 2) {
+        h = (h << 1) | (h >> 63); /* pg_rotate_left analog, nodeHash.c:2059 */
+        h ^= d_hash_i64(k2);
+    }
+    return h;
+}
+
+__global__ void k_joinx_build(const int64_t *__restrict__ k1,
+                              const uint8_t *__restrict__ n1,
+                              const int64_t *__restrict__ k2,
+                              const uint8_t *__restrict__ n2, int64_t nb,
+                              int nk, joinx_slot *tab, int64_t cap)
+{
+    int64_t mask = cap - 1;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nb;
+         i += stride) {
+        if ((n1 && n1[i]) || (nk == 2 && n2 && n2[i]))
+            continue; /* NULL join key never enters the table
+                       * (ExecHashGetHashValue nodeHash.c:2026) */
+        int64_t a = k1[i], b = nk == 2 ? k2[i] : 0;
+        int64_t s = (int64_t)(d_jx_hash(a, b, nk) & (uint64_t)mask);
+        while (atomicCAS((unsigned long long *)&tab[s].idx,
+                         (unsigned long long)(-1ll),
+                         (unsigned long long)i) != (unsigned long long)(-1ll))
+            s = (s + 1) & mask;
+        tab[s].k1 = a; /* plain stores: read only by the NEXT launch */
+        tab[s].k2 = b;
+    }
+}
+
+/* probe: wave-uniform slot walk (as k_join_probe) + per-row match count;
+ * the fill decisions are per-probe-row local except right/full, which mark
+ * a matched bitmap for the trailing unmatched-build sweep
+ * (ExecScanHashTableForUnmatched nodeHash.c:2322 analog). */
+__global__ void k_joinx_probe(const joinx_slot *__restrict__ tab, int64_t cap,
+                              const int64_t *__restrict__ k1,
+                              const uint8_t *__restrict__ n1,
+                              const int64_t *__restrict__ k2,
+                              const uint8_t *__restrict__ n2, int64_t np,
+                              int nk, int jt,
+                              unsigned long long *__restrict__ mbitmap,
+                              int64_t *__restrict__ out_b,
+                              int64_t *__restrict__ out_p, int64_t cap_pairs,
+                              int64_t *npairs)
+{
+    const bool emit_match = (jt == 0 || jt == 1 || jt == 4 || jt == 5);
+    const bool fill_probe = (jt == 1 || jt == 3 || jt == 5); /* left/anti/full */
+    const bool semi = jt == 2;
+    const int BUF = 512;
+    __shared__ int64_t bufb[256 / WAVE][BUF];
+    __shared__ int64_t bufp[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0; /* wave-uniform */
+    int64_t mask = cap - 1;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i0 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         i0 += stride) {
+        bool inrange = i0 < np;
+        bool rownull =
+            inrange && ((n1 && n1[i0]) || (nk == 2 && n2 && n2[i0]));
+        bool active = inrange && !rownull;
+        int64_t a = k1[active ? i0 : 0];
+        int64_t b = nk == 2 ? k2[active ? i0 : 0] : 0;
+        int64_t s = (int64_t)(d_jx_hash(a, b, nk) & (uint64_t)mask);
+        bool walking = active;
+        int nmatch = 0;
+        while (__any(walking)) {
+            long long bidx = walking ? tab[s].idx : -1;
+            bool have = walking && bidx >= 0;
+            bool match =
+                have && tab[s].k1 == a && (nk != 2 || tab[s].k2 == b);
+            if (match) {
+                nmatch++;
+                if (mbitmap)
+                    atomicOr(&mbitmap[bidx >> 6], 1ull << (bidx & 63));
+                if (semi && nmatch > 1)
+                    match = false; /* JOIN_SEMI: first match suffices
+                                    * (nodeHashjoin.c:572); keep walking the
+                                    * wave-converged loop but emit nothing */
+            }
+            bool emit = match && emit_match;
+            unsigned long long mmask = __ballot(emit);
+            int cnt = __popcll(mmask);
+            if (cnt) {
+                if (nbuf + cnt > BUF) {
+                    long long base = 0;
+                    if (lane == 0)
+                        base = (long long)atomicAdd(
+                            (unsigned long long *)npairs,
+                            (unsigned long long)nbuf);
+                    base = __shfl(base, 0, WAVE);
+                    for (int j = lane; j < nbuf; j += WAVE) {
+                        int64_t pos = base + j;
+                        if (pos < cap_pairs) {
+                            out_b[pos] = bufb[wid][j];
+                            out_p[pos] = bufp[wid][j];
+                        }
+                    }
+                    nbuf = 0;
+                }
+                if (emit) {
+                    int rank = __popcll(mmask & ((1ull << lane) - 1ull));
+                    bufb[wid][nbuf + rank] = bidx;
+                    bufp[wid][nbuf + rank] = i0;
+                }
+                nbuf += cnt;
+            }
+            s = (s + 1) & mask;
+            walking = have;
+        }
+        /* post-walk per-row fills (wave-converged point):
+         *   left/full: no match or NULL-key row -> (-1, i0)
+         *   anti:      same condition (NULL-key rows match nothing)
+         *   semi:      >=1 match -> (-1, i0) */
+        bool fill = false;
+        if (inrange) {
+            if (fill_probe)
+                fill = rownull || nmatch == 0;
+            else if (semi)
+                fill = nmatch > 0;
+        }
+        unsigned long long fmask = __ballot(fill);
+        int fcnt = __popcll(fmask);
+        if (fcnt) {
+            if (nbuf + fcnt > BUF) {
+                long long base = 0;
+                if (lane == 0)
+                    base = (long long)atomicAdd((unsigned long long *)npairs,
+                                                (unsigned long long)nbuf);
+                base = __shfl(base, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE) {
+                    int64_t pos = base + j;
+                    if (pos < cap_pairs) {
+                        out_b[pos] = bufb[wid][j];
+                        out_p[pos] = bufp[wid][j];
+                    }
+                }
+                nbuf = 0;
+            }
+            if (fill) {
+                int rank = __popcll(fmask & ((1ull << lane) - 1ull));
+                bufb[wid][nbuf + rank] = -1;
+                bufp[wid][nbuf + rank] = i0;
+            }
+            nbuf += fcnt;
+        }
+        if (__all(i0 >= np))
+            break;
+    }
+    if (nbuf) {
+        long long base = 0;
+        if (lane == 0)
+            base = (long long)atomicAdd((unsigned long long *)npairs,
+                                        (unsigned long long)nbuf);
+        base = __shfl(base, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE) {
+            int64_t pos = base + j;
+            if (pos < cap_pairs) {
+                out_b[pos] = bufb[wid][j];
+                out_p[pos] = bufp[wid][j];
+            }
+        }
+    }
+}
+
+/* trailing sweep for right/full: NULL-key build rows never entered the
+ * table, so they are unmatched by construction (HJ_FILL_INNER_TUPLES,
+ * nodeHashjoin.c:693). */
+__global__ void k_joinx_fill_build(const uint8_t *__restrict__ n1,
+                                   const uint8_t *__restrict__ n2, int64_t nb,
+                                   int nk,
+                                   const unsigned long long *__restrict__ mbitmap,
+                                   int64_t *__restrict__ out_b,
+                                   int64_t *__restrict__ out_p,
+                                   int64_t cap_pairs, int64_t *npairs)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i0 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         i0 += stride) {
+        bool fill = false;
+        if (i0 < nb) {
+            bool rownull = (n1 && n1[i0]) || (nk == 2 && n2 && n2[i0]);
+            fill = rownull || !((mbitmap[i0 >> 6] >> (i0 & 63)) & 1ull);
+        }
+        int64_t pos = wave_append(npairs, fill);
+        if (fill && pos < cap_pairs) {
+            out_b[pos] = i0;
+            out_p[pos] = -1;
+        }
+        if (__all(i0 >= nb))
+            break;
+    }
+}
+
+/* ================= two-key hash aggregate (otbx_agg_i64x2) ===============
+ * Group identity (k1_isnull, k1, k2_isnull, k2) with NULL==NULL for
+ * grouping (execGrouping.c:295,:525). Open-addressing slots claimed by row
+ * index (no reserved key value); the claiming row's key columns define the
+ * slot's group. Aggregates as otbx_agg_i64: count(*), count(v), sum(v)
+ * (avg = sum/count; the GPU keeps [N, Sx] — include/otbx.h note). */
+
+struct agg2_slot {
+    long long idx; /* claim word: -1 empty, else defining row index */
+    unsigned long long count_star;
+    unsigned long long count_v;
+    double sum;
+};
+
+__global__ void k_agg2_init(agg2_slot *tab, int64_t cap)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < cap;
+         i += stride) {
+        tab[i].idx = -1;
+        tab[i].count_star = 0;
+        tab[i].count_v = 0;
+        tab[i].sum = 0.0;
+    }
+}
+
+__global__ void k_agg2_build(const int64_t *__restrict__ k1,
+                             const uint8_t *__restrict__ n1,
+                             const int64_t *__restrict__ k2,
+                             const uint8_t *__restrict__ n2,
+                             const double *__restrict__ vals,
+                             const uint8_t *__restrict__ vnull, int64_t n,
+                             agg2_slot *tab, int64_t cap)
+{
+    int64_t mask = cap - 1;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        bool in1 = n1 && n1[i], in2 = n2 && n2[i];
+        int64_t a = in1 ? 0 : k1[i], b = in2 ? 0 : k2[i];
+        uint64_t h1 = in1 ? 0x9e3779b97f4a7c15ull : d_hash_i64(a);
+        uint64_t h2 = in2 ? 0xc2b2ae3d27d4eb4full : d_hash_i64(b);
+        uint64_t h = ((h1 << 1) | (h1 >> 63)) ^ h2;
+        int64_t s = (int64_t)(h & (uint64_t)mask);
+        for (;;) {
+            long long owner = tab[s].idx;
+            if (owner < 0) {
+                long long prev = (long long)atomicCAS(
+                    (unsigned long long *)&tab[s].idx,
+                    (unsigned long long)(-1ll), (unsigned long long)i);
+                owner = prev == -1ll ? i : prev;
+            }
+            bool on1 = n1 && n1[owner], on2 = n2 && n2[owner];
+            if (on1 == in1 && on2 == in2 && (in1 || k1[owner] == a) &&
+                (in2 || k2[owner] == b))
+                break;
+            s = (s + 1) & mask;
+        }
+        atomicAdd(&tab[s].count_star, 1ull);
+        if (!(vnull && vnull[i])) {
+            atomicAdd(&tab[s].count_v, 1ull);
+            atomicAdd(&tab[s].sum, vals[i]);
+        }
+    }
+}
+
+__global__ void k_agg2_compact(const agg2_slot *__restrict__ tab, int64_t cap,
+                               const int64_t *__restrict__ k1,
+                               const uint8_t *__restrict__ n1,
+                               const int64_t *__restrict__ k2,
+                               const uint8_t *__restrict__ n2,
+                               otbx_agg2_group *__restrict__ out,
+                               int64_t *ngroups)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         s += stride) {
+        bool used = s < cap && tab[s].idx >= 0;
+        int64_t pos = wave_append(ngroups, used);
+        if (used) {
+            long long i = tab[s].idx;
+            bool in1 = n1 && n1[i], in2 = n2 && n2[i];
+            out[pos].key1 = in1 ? 0 : k1[i];
+            out[pos].key2 = in2 ? 0 : k2[i];
+            out[pos].count_star = (int64_t)tab[s].count_star;
+            out[pos].count_v = (int64_t)tab[s].count_v;
+            out[pos].sum_v = tab[s].sum;
+            out[pos].key1_isnull = in1;
+            out[pos].key2_isnull = in2;
+            out[pos].sum_isnull = tab[s].count_v == 0;
+            out[pos]._pad = 0;
+        }
+        if (__all(s >= cap))
+            break;
+    }
+}
+
+extern "C" {
+
+static int64_t jx_cap_for(int64_t nb)
+{
+    return next_pow2_host(nb < 16 ? 16 : (int64_t)(nb / 0.7) + 1);
+}
+
+otbx_status otbx_join_ext_workspace_bytes(int64_t nb, int64_t np,
+                                          size_t *bytes)
+{
+    (void)np;
+    *bytes = (size_t)jx_cap_for(nb) * sizeof(joinx_slot) +
+             (size_t)((nb + 63) / 64 + 1) * 8;
+    return OTBX_OK;
+}
+
+static otbx_status joinx_run(const int64_t *bk1, const uint8_t *bn1,
+                             const int64_t *bk2, const uint8_t *bn2,
+                             int64_t nb, const int64_t *pk1,
+                             const uint8_t *pn1, const int64_t *pk2,
+                             const uint8_t *pn2, int64_t np, int nk,
+                             int32_t jt, void *ws, size_t ws_bytes,
+                             int64_t *out_b, int64_t *out_p,
+                             int64_t cap_pairs, int64_t *npairs_dev,
+                             void *stream)
+{
+    if (jt < 0 || jt > 5 || !ws || !npairs_dev)
+        return OTBX_ERR_INVALID;
+    hipStream_t s = (hipStream_t)stream;
+    int64_t cap = jx_cap_for(nb);
+    size_t need;
+    otbx_join_ext_workspace_bytes(nb, np, &need);
+    if (ws_bytes < need)
+        return OTBX_ERR_INVALID;
+    joinx_slot *tab = (joinx_slot *)ws;
+    unsigned long long *mbitmap =
+        (unsigned long long *)((char *)ws + (size_t)cap * sizeof(joinx_slot));
+    const bool fill_build = (jt == 4 || jt == 5);
+    HIP_CHECK(hipMemsetAsync(npairs_dev, 0, 8, s));
+    hipLaunchKernelGGL(k_joinx_init, dim3(grid_for(cap, 256)), dim3(256), 0,
+                       s, tab, cap);
+    if (fill_build)
+        HIP_CHECK(hipMemsetAsync(mbitmap, 0,
+                                 (size_t)((nb + 63) / 64 + 1) * 8, s));
+    if (nb > 0)
+        hipLaunchKernelGGL(k_joinx_build, dim3(grid_for(nb, 256)), dim3(256),
+                           0, s, bk1, bn1, bk2, bn2, nb, nk, tab, cap);
+    if (np > 0)
+        hipLaunchKernelGGL(k_joinx_probe, dim3(grid_for(np, 256)), dim3(256),
+                           0, s, tab, cap, pk1, pn1, pk2, pn2, np, nk, jt,
+                           fill_build ? mbitmap : NULL, out_b, out_p,
+                           cap_pairs, npairs_dev);
+    if (fill_build && nb > 0)
+        hipLaunchKernelGGL(k_joinx_fill_build, dim3(grid_for(nb, 256)),
+                           dim3(256), 0, s, bn1, bn2, nb, nk, mbitmap, out_b,
+                           out_p, cap_pairs, npairs_dev);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+otbx_status otbx_join_i64_ext(const int64_t *bkeys, const uint8_t *bnull,
+                              int64_t nb, const int64_t *pkeys,
+                              const uint8_t *pnull, int64_t np,
+                              int32_t join_type, void *ws, size_t ws_bytes,
+                              int64_t *out_b, int64_t *out_p,
+                              int64_t cap_pairs, int64_t *npairs_dev,
+                              void *stream)
+{
+    return joinx_run(bkeys, bnull, NULL, NULL, nb, pkeys, pnull, NULL, NULL,
+                     np, 1, join_type, ws, ws_bytes, out_b, out_p, cap_pairs,
+                     npairs_dev, stream);
+}
+
+otbx_status otbx_join_i64x2(const int64_t *bk1, const uint8_t *bn1,
+                            const int64_t *bk2, const uint8_t *bn2,
+                            int64_t nb, const int64_t *pk1,
+                            const uint8_t *pn1, const int64_t *pk2,
+                            const uint8_t *pn2, int64_t np,
+                            int32_t join_type, void *ws, size_t ws_bytes,
+                            int64_t *out_b, int64_t *out_p, int64_t cap_pairs,
+                            int64_t *npairs_dev, void *stream)
+{
+    if (!bk2 || !pk2)
+        return OTBX_ERR_INVALID;
+    return joinx_run(bk1, bn1, bk2, bn2, nb, pk1, pn1, pk2, pn2, np, 2,
+                     join_type, ws, ws_bytes, out_b, out_p, cap_pairs,
+                     npairs_dev, stream);
+}
+
+otbx_status otbx_agg_i64x2_workspace_bytes(int64_t n, size_t *bytes)
+{
+    *bytes = (size_t)jx_cap_for(n) * sizeof(agg2_slot);
+    return OTBX_OK;
+}
+
+otbx_status otbx_agg_i64x2(const int64_t *k1, const uint8_t *k1null,
+                           const int64_t *k2, const uint8_t *k2null,
+                           const double *vals, const uint8_t *val_null,
+                           int64_t n, void *ws, size_t ws_bytes,
+                           otbx_agg2_group *groups_dev, int64_t *ngroups_dev,
+                           void *stream)
+{
+    if (!ws || !ngroups_dev || !k1 || !k2)
+        return OTBX_ERR_INVALID;
+    hipStream_t s = (hipStream_t)stream;
+    int64_t cap = jx_cap_for(n);
+    size_t need;
+    otbx_agg_i64x2_workspace_bytes(n, &need);
+    if (ws_bytes < need)
+        return OTBX_ERR_INVALID;
+    agg2_slot *tab = (agg2_slot *)ws;
+    HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
+    hipLaunchKernelGGL(k_agg2_init, dim3(grid_for(cap, 256)), dim3(256), 0, s,
+                       tab, cap);
+    if (n > 0)
+        hipLaunchKernelGGL(k_agg2_build, dim3(grid_for(n, 256)), dim3(256), 0,
+                           s, k1, k1null, k2, k2null, vals, val_null, n, tab,
+                           cap);
+    hipLaunchKernelGGL(k_agg2_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
+                       s, tab, cap, k1, k1null, k2, k2null, groups_dev,
+                       ngroups_dev);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+} /* extern "C" */

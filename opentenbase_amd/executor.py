@@ -530,38 +530,111 @@ class GpuHashAgg(CustomScanState):
         return list(arr)
 
 
+JOIN_TYPES = {"inner": 0, "left": 1, "semi": 2, "anti": 3, "right": 4,
+              "full": 5}
+
+
 class GpuHashJoin(CustomScanState):
-    """Composable inner HashJoin on i64 keys (nodeHash.c/nodeHashjoin.c).
-    Emits (build_idx, probe_idx) pairs; result-set parity (order-free)."""
+    """Composable HashJoin on i64 keys (nodeHash.c/nodeHashjoin.c), all six
+    join types (HJ_* fill states, nodeHashjoin.c:139-144) and an optional
+    second key column (multi-key combine, nodeHash.c:2059). Emits
+    (build_idx, probe_idx) pairs, -1 = NULL-fill side (see include/otbx.h);
+    result-set parity (order-free). join_type 0/'inner' with a single key
+    routes through the optimized partitioned inner-join path."""
 
     def __init__(self, build_keys, probe_keys, build_null=None, probe_null=None,
-                 cap_pairs=None):
+                 cap_pairs=None, join_type=0, build_keys2=None,
+                 probe_keys2=None, build_null2=None, probe_null2=None):
         super().__init__()
         self.bk, self.pk = build_keys, probe_keys
         self.bn, self.pn = build_null, probe_null
+        self.bk2, self.pk2 = build_keys2, probe_keys2
+        self.bn2, self.pn2 = build_null2, probe_null2
         self.cap_pairs = cap_pairs
+        self.join_type = JOIN_TYPES.get(join_type, join_type) \
+            if isinstance(join_type, str) else join_type
 
     def _run(self):
-        import numpy as np
         L = lib()
         nb, npr = len(self.bk), len(self.pk)
+        jt = self.join_type
+        two_key = self.bk2 is not None
+        ext = jt != 0 or two_key
         ws_bytes = C.c_size_t(0)
-        check(L.otbx_join_i64_workspace_bytes(C.c_int64(nb), C.c_int64(npr),
-                                              C.byref(ws_bytes)))
+        if ext:
+            check(L.otbx_join_ext_workspace_bytes(
+                C.c_int64(nb), C.c_int64(npr), C.byref(ws_bytes)))
+        else:
+            check(L.otbx_join_i64_workspace_bytes(
+                C.c_int64(nb), C.c_int64(npr), C.byref(ws_bytes)))
         ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8, device="cuda")
-        cap = self.cap_pairs if self.cap_pairs else max(4 * max(nb, npr), 64)
+        # outer fills can exceed the match count: pairs <= matches + nb + np
+        cap = self.cap_pairs if self.cap_pairs \
+            else max(4 * max(nb, npr) + nb + npr, 64)
         ob = torch.empty(cap, dtype=torch.int64, device="cuda")
         op = torch.empty(cap, dtype=torch.int64, device="cuda")
         npairs = torch.zeros(1, dtype=torch.int64, device="cuda")
-        bn = C.c_void_p(self.bn.data_ptr()) if self.bn is not None else None
-        pn = C.c_void_p(self.pn.data_ptr()) if self.pn is not None else None
-        call("otbx_join_i64", C.c_void_p(self.bk.data_ptr()), bn, C.c_int64(nb),
-             C.c_void_p(self.pk.data_ptr()), pn, C.c_int64(npr),
-             C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
-             C.c_void_p(ob.data_ptr()), C.c_void_p(op.data_ptr()),
-             C.c_int64(cap), C.c_void_p(npairs.data_ptr()), _stream())
+
+        def vp(t):
+            return C.c_void_p(t.data_ptr()) if t is not None else None
+
+        if two_key:
+            call("otbx_join_i64x2", vp(self.bk), vp(self.bn), vp(self.bk2),
+                 vp(self.bn2), C.c_int64(nb), vp(self.pk), vp(self.pn),
+                 vp(self.pk2), vp(self.pn2), C.c_int64(npr), C.c_int32(jt),
+                 C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+                 vp(ob), vp(op), C.c_int64(cap), vp(npairs), _stream())
+        elif ext:
+            call("otbx_join_i64_ext", vp(self.bk), vp(self.bn), C.c_int64(nb),
+                 vp(self.pk), vp(self.pn), C.c_int64(npr), C.c_int32(jt),
+                 C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+                 vp(ob), vp(op), C.c_int64(cap), vp(npairs), _stream())
+        else:
+            call("otbx_join_i64", vp(self.bk), vp(self.bn), C.c_int64(nb),
+                 vp(self.pk), vp(self.pn), C.c_int64(npr),
+                 C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+                 vp(ob), vp(op), C.c_int64(cap), vp(npairs), _stream())
         n = int(npairs.cpu().item())
         if n > cap:
             raise OtbxError(3, f"join pair overflow: {n} > cap {cap}")
         return list(zip(ob[:n].cpu().numpy().tolist(),
                         op[:n].cpu().numpy().tolist()))
+
+
+class GpuHashAgg2(CustomScanState):
+    """Two-key composable HashAggregate (multi-key GROUP BY,
+    execGrouping.c:295 with NULL==NULL grouping :525). Returns rows sorted
+    by (k1_isnull, k1, k2_isnull, k2) for deterministic comparison."""
+
+    def __init__(self, keys1, keys2, vals, key1_null=None, key2_null=None,
+                 val_null=None):
+        super().__init__()
+        self.k1, self.k2, self.vals = keys1, keys2, vals
+        self.n1, self.n2, self.vn = key1_null, key2_null, val_null
+
+    def _run(self):
+        import numpy as np
+        L = lib()
+        n = len(self.k1)
+        ws_bytes = C.c_size_t(0)
+        check(L.otbx_agg_i64x2_workspace_bytes(C.c_int64(n),
+                                               C.byref(ws_bytes)))
+        ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8, device="cuda")
+        out = torch.empty(max(n, 1) * 56, dtype=torch.uint8, device="cuda")
+        ng = torch.zeros(1, dtype=torch.int64, device="cuda")
+
+        def vp(t):
+            return C.c_void_p(t.data_ptr()) if t is not None else None
+
+        call("otbx_agg_i64x2", vp(self.k1), vp(self.n1), vp(self.k2),
+             vp(self.n2), vp(self.vals), vp(self.vn), C.c_int64(n),
+             C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+             C.c_void_p(out.data_ptr()), C.c_void_p(ng.data_ptr()), _stream())
+        ngroups = int(ng.cpu().item())
+        dt = np.dtype([("key1", "i8"), ("key2", "i8"), ("count_star", "i8"),
+                       ("count_v", "i8"), ("sum_v", "f8"),
+                       ("key1_isnull", "i4"), ("key2_isnull", "i4"),
+                       ("sum_isnull", "i4"), ("_pad", "i4")])
+        arr = out[: ngroups * 56].cpu().numpy().view(dt).copy()
+        arr.sort(order=["key1_isnull", "key1", "key2_isnull", "key2"])
+        return list(arr)
