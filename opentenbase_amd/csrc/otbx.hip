@@ -3730,6 +3730,17 @@ __global__ void k_ord_count_pass(const otbx_orders_dev o,
  * date | prio<<32 with date ≥ 1, so 0 = empty. */
 #define OTBX_DIRECT_CAP_DEFAULT (1ll << 29)
 
+/* direct-table entry: revenue accumulator + packed (date|prio<<32) payload
+ * INTERLEAVED so the insert's zero+payload store, the probe's payload read
+ * + revenue atomicAdd, and the compaction's gather each touch ONE cache
+ * line per entry instead of two (the split ptab/rtab layout cost a line
+ * each). 0 payload/rev = empty under the legacy memset mode; the default
+ * mode is bitmap-authoritative and never reads unset entries. */
+struct q3_rec {
+    double rev;
+    unsigned long long pl;
+};
+
 /* single-pass orders side for the common dense case: when the UNFILTERED
  * orderkey range already fits the direct table (known from a cheap minmax
  * overlapped with the customer phase), the date qual + customer filter
@@ -3740,8 +3751,7 @@ __global__ void k_ord_filter_insert_fused(
     const unsigned long long *__restrict__ ckeys, int64_t ccap,
     const unsigned long long *__restrict__ cbitmap, int64_t cmin,
     int64_t crange, int64_t mino, int64_t range,
-    unsigned long long *__restrict__ bitmap,
-    unsigned long long *__restrict__ ptab, double *__restrict__ rtab)
+    unsigned long long *__restrict__ bitmap, q3_rec *__restrict__ grec)
 {
     int64_t nq = o.n / 4;
     const int4 *od4 = (const int4 *)o.o_orderdate;
@@ -3774,10 +3784,11 @@ __global__ void k_ord_filter_insert_fused(
             int64_t idx = ok[j] - mino;
             if (idx < 0 || idx >= range) continue;
             atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
-            ptab[idx] = (unsigned long long)(uint32_t)ds[j] |
-                        ((unsigned long long)(uint32_t)prio[j] << 32);
-            rtab[idx] = 0.0; /* survivor-slot zeroing (see
-                              * k_ord_insert_direct) */
+            q3_rec r;
+            r.rev = 0.0; /* survivor-slot zeroing (see k_ord_insert_direct) */
+            r.pl = (unsigned long long)(uint32_t)ds[j] |
+                   ((unsigned long long)(uint32_t)prio[j] << 32);
+            grec[idx] = r; /* one 16-B store on one line */
         }
     }
     /* tail rows */
@@ -3797,9 +3808,11 @@ __global__ void k_ord_filter_insert_fused(
             int64_t idx = o.o_orderkey[i] - mino;
             if (idx < 0 || idx >= range) continue;
             atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
-            ptab[idx] =
-                (unsigned long long)(uint32_t)o.o_orderdate[i] |
-                ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
+            q3_rec r;
+            r.rev = 0.0;
+            r.pl = (unsigned long long)(uint32_t)o.o_orderdate[i] |
+                   ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
+            grec[idx] = r;
         }
     }
 }
@@ -3809,7 +3822,7 @@ __global__ void k_ord_insert_direct(const otbx_orders_dev o,
                                     const int64_t *__restrict__ ncand_p,
                                     int64_t mino, int64_t range,
                                     unsigned long long *bitmap,
-                                    unsigned long long *ptab, double *rtab)
+                                    q3_rec *grec)
 {
     int64_t n = *ncand_p;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -3820,14 +3833,14 @@ __global__ void k_ord_insert_direct(const otbx_orders_dev o,
         if (idx < 0 || idx >= range)
             continue; /* outside this grace pass's key sub-range */
         atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
-        ptab[idx] = (unsigned long long)(uint32_t)o.o_orderdate[i] |
-                    ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
-        if (rtab)
-            rtab[idx] = 0.0; /* survivor-slot zeroing: replaces the whole-
-                              * range memset (orderkeys unique -> one writer
-                              * per idx; entries at unset-bit positions are
-                              * never read: compaction is
-                              * bitmap-authoritative) */
+        q3_rec r;
+        r.rev = 0.0; /* survivor-slot zeroing: replaces the whole-range
+                      * memset (orderkeys unique -> one writer per idx;
+                      * unset-bit entries are never read: compaction is
+                      * bitmap-authoritative) */
+        r.pl = (unsigned long long)(uint32_t)o.o_orderdate[i] |
+               ((unsigned long long)(uint32_t)o.o_shippriority[i] << 32);
+        grec[idx] = r;
     }
 }
 
@@ -3840,8 +3853,7 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
                                            int32_t q3date, int64_t mino,
                                            int64_t range,
                                            const unsigned long long *__restrict__ bitmap,
-                                           const unsigned long long *__restrict__ ptab,
-                                           double *__restrict__ rtab,
+                                           q3_rec *__restrict__ grec,
                                            unsigned long long *__restrict__ nhits)
 {
     int64_t nq = l.n / 4;
@@ -3866,7 +3878,7 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
             m[j] = pass && ((w >> (bidx & 63)) & 1ull);
             /* payload read only where the bitmap passed (exact filter):
              * clustered keys keep these on few lines */
-            pl[j] = m[j] ? ptab[idx] : 0ull;
+            pl[j] = m[j] ? grec[idx].pl : 0ull;
         }
 #pragma unroll
         for (int j = 0; j < 4; j++) {
@@ -3875,7 +3887,7 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
             myhits++;
             int64_t i = q * 4 + j;
             double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
-            atomicAdd(&rtab[ky[j] - mino], rev);
+            atomicAdd(&grec[ky[j] - mino].rev, rev); /* same line as pl */
         }
     }
     /* tail rows */
@@ -3886,12 +3898,12 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
                 continue;
             if (!((bitmap[idx >> 6] >> (idx & 63)) & 1ull))
                 continue;
-            unsigned long long pv = ptab[idx];
+            unsigned long long pv = grec[idx].pl;
             if (pv == 0ull)
                 continue;
             myhits++;
             double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
-            atomicAdd(&rtab[idx], rev);
+            atomicAdd(&grec[idx].rev, rev);
         }
     }
     for (int off = WAVE / 2; off > 0; off >>= 1)
@@ -3916,8 +3928,7 @@ static inline bool q3_compact_legacy(void)
  * per-thread runs. */
 #define Q3CT 8192
 __global__ __launch_bounds__(1024) void k_q3_compact_tile(
-    const double *__restrict__ rtab,
-    const unsigned long long *__restrict__ ptab,
+    const q3_rec *__restrict__ grec,
     const unsigned long long *__restrict__ bitmap, int64_t range,
     int64_t mino, otbx_q3_group *out, int64_t cap_out, int64_t *ngroups)
 {
@@ -3927,7 +3938,6 @@ __global__ __launch_bounds__(1024) void k_q3_compact_tile(
     __shared__ int sweepbase;
     __shared__ long long gbase;
     int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
-    const v2d *rt2 = (const v2d *)rtab;
     int64_t ntiles = (range + Q3CT - 1) / Q3CT;
     for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
         int64_t tl = t * (int64_t)Q3CT;
@@ -3938,27 +3948,25 @@ __global__ __launch_bounds__(1024) void k_q3_compact_tile(
             int64_t r0 = s0 + 4 * (int64_t)threadIdx.x;
             bool m[4] = {false, false, false, false};
             int mycnt = 0;
-            /* bitmap is the presence authority (rtab/ptab are NOT
-             * memset per step; stale entries live at unset-bit
-             * positions); rtab != 0 then drops build rows with no
-             * probe match (revenue > 0 by domain: price > 0, disc < 1) */
+            /* bitmap is the presence authority (records are NOT memset
+             * per step; stale entries live at unset-bit positions);
+             * rev != 0 then drops build rows with no probe match
+             * (revenue > 0 by domain: price > 0, disc < 1). The record
+             * loads only run near set bits — most quads fast-skip. */
             if (r0 + 3 < range) {
                 unsigned long long w = bitmap[r0 >> 6] >> (r0 & 63);
-                if ((w & 0xfull) == 0) {
-                    /* fast skip: no bits set for these 4 entries */
-                } else {
-                    v2d ra = rt2[r0 / 2], rb = rt2[r0 / 2 + 1];
-                    m[0] = (w & 1ull) && ra.x != 0.0;
-                    m[1] = (w & 2ull) && ra.y != 0.0;
-                    m[2] = (w & 4ull) && rb.x != 0.0;
-                    m[3] = (w & 8ull) && rb.y != 0.0;
-                    mycnt = m[0] + m[1] + m[2] + m[3];
+                if (w & 0xfull) {
+#pragma unroll
+                    for (int j = 0; j < 4; j++) {
+                        m[j] = ((w >> j) & 1ull) && grec[r0 + j].rev != 0.0;
+                        mycnt += m[j];
+                    }
                 }
             } else {
                 for (int j = 0; j < 4 && r0 + j < range; j++) {
                     int64_t i = r0 + j;
                     m[j] = ((bitmap[i >> 6] >> (i & 63)) & 1ull) &&
-                           rtab[i] != 0.0;
+                           grec[i].rev != 0.0;
                     mycnt += m[j];
                 }
             }
@@ -3994,39 +4002,31 @@ __global__ __launch_bounds__(1024) void k_q3_compact_tile(
             int64_t gp = gbase + p;
             if (gp >= cap_out) continue;
             int64_t i = tl + (int64_t)stage[p];
-            unsigned long long pl = ptab[i];
+            q3_rec r = grec[i];
             out[gp].l_orderkey = mino + i;
-            out[gp].revenue = rtab[i];
-            out[gp].o_orderdate = (int32_t)(pl & 0xffffffffull);
-            out[gp].o_shippriority = (int32_t)(pl >> 32);
+            out[gp].revenue = r.rev;
+            out[gp].o_orderdate = (int32_t)(r.pl & 0xffffffffull);
+            out[gp].o_shippriority = (int32_t)(r.pl >> 32);
         }
         __syncthreads();
     }
 }
 
-__global__ void k_q3_compact_direct(const double *__restrict__ rtab,
-                                    const unsigned long long *__restrict__ ptab,
+__global__ void k_q3_compact_direct(const q3_rec *__restrict__ grec,
                                     int64_t range, int64_t mino,
                                     otbx_q3_group *out, int64_t cap_out,
                                     int64_t *ngroups)
 {
-    /* 16-B vector loads on both passes: the scalar-8B version ran at 1.01
-     * ms for a 150 M-entry table — the guide's 0.54-0.7× scalar-load
-     * penalty — vs the ~0.45 ms two-pass stream floor */
-    int64_t nq = range / 2;
-    const v2d *rt2 = (const v2d *)rtab;
-    int64_t per_block = (nq + gridDim.x - 1) / gridDim.x;
+    /* legacy (memset-dependent) block-chunk compaction over the
+     * interleaved record table: one 16-B load per entry on each pass */
+    int64_t per_block = (range + gridDim.x - 1) / gridDim.x;
     int64_t lo = blockIdx.x * per_block;
-    int64_t hi = lo + per_block < nq ? lo + per_block : nq;
+    int64_t hi = lo + per_block < range ? lo + per_block : range;
     __shared__ int64_t tcnt[256];
     __shared__ int64_t tbase[257];
     int64_t my = 0;
-    for (int64_t q = lo + threadIdx.x; q < hi; q += blockDim.x) {
-        v2d r = rt2[q];
-        my += (r.x != 0.0) + (r.y != 0.0);
-    }
-    if ((range & 1) && blockIdx.x == 0 && threadIdx.x == 0)
-        my += rtab[range - 1] != 0.0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+        my += grec[i].rev != 0.0;
     tcnt[threadIdx.x] = my;
     __syncthreads();
     if (threadIdx.x == 0) {
@@ -4041,34 +4041,14 @@ __global__ void k_q3_compact_direct(const double *__restrict__ rtab,
     }
     __syncthreads();
     int64_t pos = tbase[256] + tbase[threadIdx.x];
-    for (int64_t q = lo + threadIdx.x; q < hi; q += blockDim.x) {
-        v2d r = rt2[q];
-        double revs[2] = {r.x, r.y};
-#pragma unroll
-        for (int j = 0; j < 2; j++) {
-            if (revs[j] != 0.0) {
-                int64_t i = 2 * q + j;
-                if (pos < cap_out) {
-                    unsigned long long pl = ptab[i];
-                    out[pos].l_orderkey = mino + i;
-                    out[pos].revenue = revs[j];
-                    out[pos].o_orderdate = (int32_t)(pl & 0xffffffffull);
-                    out[pos].o_shippriority = (int32_t)(pl >> 32);
-                }
-                pos++;
-            }
-        }
-    }
-    if ((range & 1) && blockIdx.x == 0 && threadIdx.x == 0) {
-        int64_t i = range - 1;
-        double rev = rtab[i];
-        if (rev != 0.0) {
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        q3_rec r = grec[i];
+        if (r.rev != 0.0) {
             if (pos < cap_out) {
-                unsigned long long pl = ptab[i];
                 out[pos].l_orderkey = mino + i;
-                out[pos].revenue = rev;
-                out[pos].o_orderdate = (int32_t)(pl & 0xffffffffull);
-                out[pos].o_shippriority = (int32_t)(pl >> 32);
+                out[pos].revenue = r.rev;
+                out[pos].o_orderdate = (int32_t)(r.pl & 0xffffffffull);
+                out[pos].o_shippriority = (int32_t)(r.pl >> 32);
             }
             pos++;
         }
@@ -4523,9 +4503,9 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
     int64_t dcap = direct_cap_for(o->n);
     int64_t dcap_c = direct_cap_for(ncust);
     unsigned long long *dbitmap = (unsigned long long *)(cand_o2 + o->n);
-    unsigned long long *dptab = dbitmap + dcap / 64 + 8;
-    double *drtab = (double *)(dptab + dcap);
-    unsigned long long *cbitmap_buf = (unsigned long long *)(drtab + dcap);
+    q3_rec *dgrec = (q3_rec *)(dbitmap + dcap / 64 + 8); /* 16 B/entry —
+        occupies exactly the former split ptab+rtab regions */
+    unsigned long long *cbitmap_buf = (unsigned long long *)(dgrec + dcap);
 
     static int64_t *h_cnt = nullptr;        /* pinned host readback */
     if (!h_cnt)
@@ -4622,29 +4602,28 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
         HIP_CHECK(hipMemsetAsync(dbitmap, 0,
                                  (size_t)(range_all / 64 + 8) * 8, s));
         if (q3_compact_legacy()) {
-            /* legacy compaction scans rtab != 0 over the whole range, so
+            /* legacy compaction scans rev != 0 over the whole range, so
              * it needs the full-range zeroing the default path skips */
-            HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)range_all * 8, s));
-            HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)range_all * 8, s));
+            HIP_CHECK(hipMemsetAsync(dgrec, 0, (size_t)range_all * 16, s));
         }
         hipLaunchKernelGGL(k_ord_filter_insert_fused,
                            dim3(grid_for(o->n / 4, 256)), dim3(256), 0, s, *o,
                            q3date, ctab, ccap, cbitmap, cmin, crange,
-                           mino_all, range_all, dbitmap, dptab, drtab);
+                           mino_all, range_all, dbitmap, dgrec);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
         hipLaunchKernelGGL(k_q3_scan_probe_agg_direct,
                            dim3(grid_for(l->n / 4, 256)), dim3(256), 0, s, *l,
-                           q3date, mino_all, range_all, dbitmap, dptab, drtab,
+                           q3date, mino_all, range_all, dbitmap, dgrec,
                            nhits);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
         if (q3_compact_legacy())
             hipLaunchKernelGGL(k_q3_compact_direct,
                                dim3(grid_for(range_all, 256)), dim3(256), 0,
-                               s, drtab, dptab, range_all, mino_all,
+                               s, dgrec, range_all, mino_all,
                                groups_dev, cap_groups, ngroups_dev);
         else
             hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024), 0,
-                               s, drtab, dptab, dbitmap, range_all, mino_all,
+                               s, dgrec, dbitmap, range_all, mino_all,
                                groups_dev, cap_groups, ngroups_dev);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
         goto emit;
@@ -4688,24 +4667,21 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                                         : dcap;
             HIP_CHECK(hipMemsetAsync(dbitmap, 0,
                                      (size_t)(prange / 64 + 8) * 8, s));
-            if (q3_compact_legacy()) {
-                HIP_CHECK(hipMemsetAsync(dptab, 0, (size_t)prange * 8, s));
-                HIP_CHECK(hipMemsetAsync(drtab, 0, (size_t)prange * 8, s));
-            }
+            if (q3_compact_legacy())
+                HIP_CHECK(hipMemsetAsync(dgrec, 0, (size_t)prange * 16, s));
             if (pass > 0)
                 HIP_CHECK(hipMemsetAsync(&hdr[2], 0, 8, s)); /* reset cands */
             hipLaunchKernelGGL(k_ord_insert_direct, dim3(grid_for(o->n, 256)),
                                dim3(256), 0, s, *o, cand_o2, &hdr[1], pmin,
-                               prange, dbitmap, dptab,
-                               q3_compact_legacy() ? NULL : drtab);
+                               prange, dbitmap, dgrec);
             if (kernel_ms && !rec2) {
                 HIP_CHECK(hipEventRecord(ev[2], s));
                 rec2 = true;
             }
             hipLaunchKernelGGL(k_q3_scan_probe_agg_direct,
                                dim3(grid_for(l->n / 4, 256)), dim3(256), 0, s,
-                               *l, q3date, pmin, prange, dbitmap, dptab,
-                               drtab, nhits);
+                               *l, q3date, pmin, prange, dbitmap, dgrec,
+                               nhits);
             if (kernel_ms && !rec3 && pass == npasses - 1) {
                 HIP_CHECK(hipEventRecord(ev[3], s));
                 rec3 = true;
@@ -4713,11 +4689,11 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
             if (q3_compact_legacy())
                 hipLaunchKernelGGL(k_q3_compact_direct,
                                    dim3(grid_for(prange, 256)), dim3(256), 0,
-                                   s, drtab, dptab, prange, pmin, groups_dev,
+                                   s, dgrec, prange, pmin, groups_dev,
                                    cap_groups, ngroups_dev);
             else
                 hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024),
-                                   0, s, drtab, dptab, dbitmap, prange, pmin,
+                                   0, s, dgrec, dbitmap, prange, pmin,
                                    groups_dev, cap_groups, ngroups_dev);
         }
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
