@@ -128,6 +128,12 @@ typedef struct {
     uint8_t *l_returnflag, *l_linestatus;
     int32_t *l_shipdate;
     int64_t *l_partkey;       /* appended; may be NULL (Q9-mix only) */
+    /* optional Q9 probe-side record cache {l_orderkey i64, l_extendedprice
+     * f64, l_discount f64, pad} (32 B/row), built ONCE at staging
+     * (otbx_build_q9recs — a derived layout like the zone-map metadata):
+     * the Q9 probe's per-survivor gather then touches one cache line
+     * instead of three column lines. NULL = probe gathers the columns. */
+    void *q9rec;
 } otbx_lineitem_dev;
 
 typedef struct {
@@ -157,6 +163,9 @@ typedef struct {
  * functions as the CPU oracle — oracle/otbx_gen.h — so tables are
  * bit-identical on both sides). Caller provides the device buffers
  * (column pointers in the struct, each sized for n_global/nranks rows). */
+/* build the q9rec cache from the staged columns (32 B/row into recs_dev) */
+otbx_status otbx_build_q9recs(const otbx_lineitem_dev *l, void *recs_dev,
+                              void *stream);
 otbx_status otbx_gen_lineitem_dev(const otbx_lineitem_dev *t, uint64_t seed,
                                   int64_t n_global, uint32_t rank,
                                   uint32_t nranks, void *stream);

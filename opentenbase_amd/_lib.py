@@ -34,6 +34,7 @@ class LineitemDev(C.Structure):
         ("l_returnflag", C.c_void_p), ("l_linestatus", C.c_void_p),
         ("l_shipdate", C.c_void_p),
         ("l_partkey", C.c_void_p),
+        ("q9rec", C.c_void_p),
     ]
 
 
@@ -105,7 +106,7 @@ EXPORTED_SYMBOLS = [
     "otbx_memcpy_d2h", "otbx_stream_sync",
     "otbx_gen_lineitem_dev", "otbx_gen_orders_dev", "otbx_gen_customer_dev",
     "otbx_gen_part_dev", "otbx_q9_workspace_bytes", "otbx_q9_partial",
-    "otbx_stage_pages",
+    "otbx_stage_pages", "otbx_build_q9recs",
     "otbx_scan_count", "otbx_q1_partial", "otbx_q1_partial_variant",
     "otbx_q3_workspace_bytes", "otbx_q3_partial", "otbx_filter_customer",
     "otbx_topk_by_revenue",

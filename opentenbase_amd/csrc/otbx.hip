@@ -2241,6 +2241,32 @@ __global__ __launch_bounds__(1024) void k_q9_filter_tile(
     }
 }
 
+/* Q9 probe-side AoS record (otbx.h q9rec): one 32-B record per lineitem
+ * row so a survivor's gather touches one cache line instead of three
+ * column lines (l_orderkey / l_extendedprice / l_discount). Built once at
+ * staging. */
+struct q9_rec {
+    long long okey;
+    double price;
+    double disc;
+    long long pad;
+};
+
+__global__ void k_q9_build_recs(const otbx_lineitem_dev l,
+                                q9_rec *__restrict__ r)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < l.n;
+         i += stride) {
+        q9_rec x;
+        x.okey = l.l_orderkey[i];
+        x.price = l.l_extendedprice[i];
+        x.disc = l.l_discount[i];
+        x.pad = 0;
+        r[i] = x;
+    }
+}
+
 /* phase 2: dense pass over the survivors — orders date lookup + year
  * partial aggregate in per-lane registers (the Q1 pattern: 7-year domain,
  * compile-time indexed), wave+block reduce, one atomic per (year, block). */
@@ -2263,14 +2289,27 @@ __global__ void k_q9_probe(const otbx_lineitem_dev l,
     for (int64_t h = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; h < nh;
          h += stride) {
         int64_t i = (int64_t)hits[h];
-        int64_t oidx = l.l_orderkey[i] - mino;
+        const q9_rec *recs = (const q9_rec *)l.q9rec;
+        int64_t okey;
+        double price, disc;
+        if (recs) { /* staged record cache: one line per survivor */
+            q9_rec rc = recs[i];
+            okey = rc.okey;
+            price = rc.price;
+            disc = rc.disc;
+        } else {
+            okey = l.l_orderkey[i];
+            price = l.l_extendedprice[i];
+            disc = l.l_discount[i];
+        }
+        int64_t oidx = okey - mino;
         if (oidx < 0 || oidx >= orange)
             continue;
         int32_t date = dtab[oidx];
         if (date == 0)
             continue;
         int32_t y = otbx_year_of_day(date);
-        double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
+        double rev = price * (1.0 - disc);
 #pragma unroll
         for (int yy = 0; yy < 7; yy++) {
             bool hit = yy == y;
@@ -5289,6 +5328,18 @@ otbx_status otbx_agg_i64x2(const int64_t *k1, const uint8_t *k1null,
     hipLaunchKernelGGL(k_agg2_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
                        s, tab, cap, k1, k1null, k2, k2null, groups_dev,
                        ngroups_dev);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+otbx_status otbx_build_q9recs(const otbx_lineitem_dev *l, void *recs_dev,
+                              void *stream)
+{
+    if (!l || !recs_dev || !l->l_orderkey || !l->l_extendedprice ||
+        !l->l_discount)
+        return OTBX_ERR_INVALID;
+    hipLaunchKernelGGL(k_q9_build_recs, dim3(grid_for(l->n, 256)), dim3(256),
+                       0, (hipStream_t)stream, *l, (q9_rec *)recs_dev);
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }

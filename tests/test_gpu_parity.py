@@ -1093,3 +1093,28 @@ def test_agg2_grid_golden_gpu(ex, ora):
     rows = drain(agg)
     assert len(rows) == 10
     assert all(int(r["count_star"]) == 1 for r in rows)
+
+
+def test_q9_rec_vs_columnar_selfconsistent(ex):
+    """The staged q9rec probe path (default) vs the columnar-gather
+    fallback (q9rec = NULL) on the same tables: counts exact, sums within
+    rounding tolerance."""
+    import ctypes as CT
+    n = 2_000_000
+    li = ex.GpuLineitem.generate(n, with_partkey=True)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    pt = ex.GpuPart.generate(n // 30)
+
+    def run():
+        node = ex.GpuQ9Fragment(pt, od, li)
+        node.BeginCustomScan()
+        node._run()
+        s, c = node.partial_state_tensors()
+        return s.cpu().numpy().copy(), c.cpu().numpy().copy()
+
+    s_rec, c_rec = run()
+    li.cstruct.q9rec = CT.c_void_p(0)      # force the columnar fallback
+    s_col, c_col = run()
+    assert np.array_equal(c_rec, c_col)
+    for a, b in zip(s_rec, s_col):
+        assert abs(a - b) <= 1e-9 * max(abs(b), 1.0)
