@@ -162,15 +162,16 @@ def main():
         bytes_per_row = SCAN_BYTES_PER_ROW
     elif args.workload == "tpch_q9mix":
         # BASELINE config 5's second shape: lineitem ⋈ part ⋈ orders,
-        # GROUP BY year. Streamed probe-side bytes: partkey 8 + orderkey 8 =
-        # 16 B/row (payload columns are read per part-filter hit only:
-        # dtab 4 + extendedprice 8 + discount 8 = 20 B/hit).
+        # GROUP BY year. Streamed probe-side bytes: l_partkey only =
+        # 8 B/row (the filter pass); per part-filter hit: survivor id
+        # written+read 8 + staged 32-B record (orderkey/price/discount,
+        # otbx.h q9rec) + dtab 4 = 44 B/hit.
         li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world,
                                      with_partkey=True)
         od = ex.GpuOrders.generate(n_global // 4, n_global // 40, rank=rank,
                                    nranks=world, skew=args.skew)
         pt = ex.GpuPart.generate(max(n_global // 30, 1))
-        bytes_per_row = 16
+        bytes_per_row = 8
     else:  # tpch_q3
         li = ex.GpuLineitem.generate(n_global, rank=rank, nranks=world)
         od = ex.GpuOrders.generate(n_global // 4, n_global // 40, rank=rank,
@@ -271,13 +272,14 @@ def main():
             hits = hits_acc[-1]
             algo_bytes = 28 * rows_per_gpu + 64 * hits
         elif args.workload == "tpch_q9mix":
-            # fused probe: 16 B/row streamed + 20 B per part-filter hit.
-            # NOTE: this kernel is bound by random-gather
+            # filter streams 8 B/row; per hit: id write+read 8 + 32-B
+            # staged record + 4-B dtab = 44 B.
+            # NOTE: the filter is bound by random-gather
             # REQUEST throughput (uniform partkeys), not HBM bytes — frac
             # vs the byte roofline understates it; the component microbench
             # (tools/microbench/q9_gather_ab) gives the real floor.
             hits = hits_acc[-1]
-            algo_bytes = 16 * rows_per_gpu + 20 * hits
+            algo_bytes = 8 * rows_per_gpu + 44 * hits
         else:
             algo_bytes = rows_per_gpu * bytes_per_row
         achieved = algo_bytes / (kmean_ms / 1e3)  # B/s, per GPU
