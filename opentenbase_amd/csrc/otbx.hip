@@ -2058,25 +2058,10 @@ __global__ void k_q9_part_bitmap(const otbx_part_dev p, uint8_t typemod,
 __global__ void k_q9_odate_build(const otbx_orders_dev o, int64_t mino,
                                  int32_t *dtab)
 {
-    /* 16-B vectorized + non-temporal: single-use streams, scattered but
-     * key-clustered dtab writes */
-    int64_t nq = o.n / 4;
-    const v2l *ok2 = (const v2l *)o.o_orderkey;
-    const v4i *od4 = (const v4i *)o.o_orderdate;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
-         q += stride) {
-        v2l ka = __builtin_nontemporal_load(&ok2[2 * q]);
-        v2l kb = __builtin_nontemporal_load(&ok2[2 * q + 1]);
-        v4i d = __builtin_nontemporal_load(&od4[q]);
-        dtab[ka.x - mino] = d.x;
-        dtab[ka.y - mino] = d.y;
-        dtab[kb.x - mino] = d.z;
-        dtab[kb.y - mino] = d.w;
-    }
-    if (blockIdx.x == 0 && threadIdx.x == 0)
-        for (int64_t i = nq * 4; i < o.n; i++)
-            dtab[o.o_orderkey[i] - mino] = o.o_orderdate[i];
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < o.n;
+         i += stride)
+        dtab[o.o_orderkey[i] - mino] = o.o_orderdate[i];
 }
 
 /* Q9 probe, split in two phases. The single fused kernel measured as the
@@ -3816,12 +3801,10 @@ __global__ void k_ord_filter_insert_fused(
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
          q += stride) {
-        v4i d = __builtin_nontemporal_load((const v4i *)&od4[q]);
-        v2l ca = __builtin_nontemporal_load((const v2l *)&ck2[2 * q]);
-        v2l cb = __builtin_nontemporal_load((const v2l *)&ck2[2 * q + 1]);
-        v2l ka = __builtin_nontemporal_load((const v2l *)&okk2[2 * q]);
-        v2l kb = __builtin_nontemporal_load((const v2l *)&okk2[2 * q + 1]);
-        v4i pr = __builtin_nontemporal_load((const v4i *)&op4[q]);
+        int4 d = od4[q];
+        longlong2 ca = ck2[2 * q], cb = ck2[2 * q + 1];
+        longlong2 ka = okk2[2 * q], kb = okk2[2 * q + 1];
+        int4 pr = op4[q];
         int32_t ds[4] = {d.x, d.y, d.z, d.w};
         int64_t ck[4] = {ca.x, ca.y, cb.x, cb.y};
         int64_t ok[4] = {ka.x, ka.y, kb.x, kb.y};
@@ -3920,12 +3903,11 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
     unsigned long long myhits = 0;
     for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
          q += stride) {
-        /* non-temporal: the single-use scan streams must not evict the
-         * hot bitmap/record lines they race against (the k_q9_filter
-         * finding; loads only — nt STORES regressed in r1) */
-        v4i d = __builtin_nontemporal_load((const v4i *)&sd4[q]);
-        v2l ka = __builtin_nontemporal_load((const v2l *)&ok2[2 * q]);
-        v2l kb = __builtin_nontemporal_load((const v2l *)&ok2[2 * q + 1]);
+        /* plain loads: nt loads here measured neutral-to-worse (r5 vs r4,
+         * profiles/r5_q3.json) — the streams already miss L2 and the nt
+         * hint bought nothing */
+        int4 d = sd4[q];
+        longlong2 ka = ok2[2 * q], kb = ok2[2 * q + 1];
         int32_t ds[4] = {d.x, d.y, d.z, d.w};
         int64_t ky[4] = {ka.x, ka.y, kb.x, kb.y};
         unsigned long long pl[4];
