@@ -3999,9 +3999,12 @@ __global__ __launch_bounds__(1024) void k_q3_compact_tile(
             if (r0 + 3 < range) {
                 unsigned long long w = bitmap[r0 >> 6] >> (r0 & 63);
                 if (w & 0xfull) {
+                    /* one 16-B load per candidate record (rev+pl share the
+                     * line anyway) instead of an 8-B stride-16 pick */
+                    const v2d *gr2 = (const v2d *)&grec[r0];
 #pragma unroll
                     for (int j = 0; j < 4; j++) {
-                        m[j] = ((w >> j) & 1ull) && grec[r0 + j].rev != 0.0;
+                        m[j] = ((w >> j) & 1ull) && gr2[j].x != 0.0;
                         mycnt += m[j];
                     }
                 }
@@ -4359,24 +4362,31 @@ __global__ void k_count_customer_seg(const otbx_customer_dev c, uint8_t want,
                                      unsigned long long *minkey,
                                      unsigned long long *maxkey)
 {
-    /* 4 rows/lane: uchar4 segment loads; custkey loaded only for matches */
+    /* 4 rows/lane; custkey loaded UNCONDITIONALLY as 16-B vectors — at the
+     * ~20 % segment selectivity the divergent conditional scalar gathers
+     * of the first version touched most custkey lines anyway but with poor
+     * MLP (measured 308 us for a 15 M-row table, ~8x its stream floor);
+     * select-based min/max keeps the lane code branchless */
     int64_t nq = c.n / 4;
     const uchar4 *seg4 = (const uchar4 *)c.c_mktsegment;
+    const v2l *ck2 = (const v2l *)c.c_custkey;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     unsigned long long my = 0, mymin = ~0ull >> 1, mymax = 0;
     for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
          q += stride) {
         uchar4 sv = seg4[q];
+        v2l ka = ck2[2 * q], kb = ck2[2 * q + 1];
         uint8_t ss[4] = {sv.x, sv.y, sv.z, sv.w};
+        unsigned long long ks[4] = {
+            (unsigned long long)ka.x, (unsigned long long)ka.y,
+            (unsigned long long)kb.x, (unsigned long long)kb.y};
 #pragma unroll
         for (int j = 0; j < 4; j++) {
-            if (ss[j] == want) {
-                my++;
-                unsigned long long k =
-                    (unsigned long long)c.c_custkey[q * 4 + j];
-                if (k < mymin) mymin = k;
-                if (k > mymax) mymax = k;
-            }
+            bool m = ss[j] == want;
+            my += m;
+            unsigned long long k = ks[j];
+            mymin = (m && k < mymin) ? k : mymin;
+            mymax = (m && k > mymax) ? k : mymax;
         }
     }
     /* tail */
@@ -4437,9 +4447,27 @@ __global__ void k_minmax_i64(const int64_t *__restrict__ keys, int64_t n,
 __global__ void k_cust_bitmap_filter(const otbx_customer_dev c, uint8_t want,
                                      int64_t minc, unsigned long long *bitmap)
 {
+    /* 4 rows/lane, unconditional vector loads (see k_count_customer_seg) */
+    int64_t nq = c.n / 4;
+    const uchar4 *seg4 = (const uchar4 *)c.c_mktsegment;
+    const v2l *ck2 = (const v2l *)c.c_custkey;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < c.n;
-         i += stride) {
+    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
+         q += stride) {
+        uchar4 sv = seg4[q];
+        v2l ka = ck2[2 * q], kb = ck2[2 * q + 1];
+        uint8_t ss[4] = {sv.x, sv.y, sv.z, sv.w};
+        long long ks[4] = {ka.x, ka.y, kb.x, kb.y};
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+            if (ss[j] == want) {
+                int64_t idx = ks[j] - minc;
+                atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
+            }
+        }
+    }
+    for (int64_t i = nq * 4 + blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         i < c.n; i += stride) {
         if (c.c_mktsegment[i] == want) {
             int64_t idx = c.c_custkey[i] - minc;
             atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
