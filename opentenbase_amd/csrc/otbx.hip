@@ -3923,6 +3923,13 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
              * clustered keys keep these on few lines */
             pl[j] = m[j] ? grec[idx].pl : 0ull;
         }
+        /* in-lane run combine: lineitem rows arrive clustered by orderkey
+         * (~4 rows/order), so a lane's quad usually hits one or two keys —
+         * summing runs locally cuts the f64 atomics (same-address atomics
+         * serialize) several-fold */
+        int64_t run_key = 0;
+        double run_sum = 0.0;
+        bool run_valid = false;
 #pragma unroll
         for (int j = 0; j < 4; j++) {
             if (!m[j] || pl[j] == 0ull)
@@ -3930,8 +3937,18 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
             myhits++;
             int64_t i = q * 4 + j;
             double rev = l.l_extendedprice[i] * (1.0 - l.l_discount[i]);
-            atomicAdd(&grec[ky[j] - mino].rev, rev); /* same line as pl */
+            if (run_valid && ky[j] == run_key) {
+                run_sum += rev;
+            } else {
+                if (run_valid)
+                    atomicAdd(&grec[run_key - mino].rev, run_sum);
+                run_key = ky[j];
+                run_sum = rev;
+                run_valid = true;
+            }
         }
+        if (run_valid)
+            atomicAdd(&grec[run_key - mino].rev, run_sum);
     }
     /* tail rows */
     if (blockIdx.x == 0 && threadIdx.x == 0) {
