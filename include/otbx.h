@@ -355,6 +355,33 @@ otbx_status otbx_agg_i64x2(const int64_t *k1_dev, const uint8_t *k1null_dev,
                            otbx_agg2_group *groups_dev, int64_t *ngroups_dev,
                            void *stream);
 
+/* ---- exact decimal (scaled-int64) aggregate with int128 sum ----
+ * The reference's HAVE_INT128 numeric aggregation: group state =
+ * Int128AggState {N, sumX} (utils/adt/numeric.c:5072; do_int128_accum
+ * :4998; transition int8_avg_accum :5365; sum(bigint) promotes to numeric
+ * and cannot overflow, int8_sum :6206). Values are scaled-decimal int64
+ * (e.g. NUMERIC(15,2) money in cents); the emitted two's-complement
+ * 128-bit sum is EXACT, so parity with the reference is bit-exact —
+ * no float tolerance. groups_dev capacity = n. */
+typedef struct {
+    int64_t key;
+    int64_t count_star;
+    int64_t count_v;
+    int64_t sum_hi;          /* int128 two's-complement high word */
+    uint64_t sum_lo;
+    int32_t key_isnull;
+    int32_t sum_isnull;
+} otbx_dec_group; /* 48 B */
+
+otbx_status otbx_agg_i64_dec_workspace_bytes(int64_t n, size_t *bytes);
+otbx_status otbx_agg_i64_dec(const int64_t *keys_dev,
+                             const uint8_t *key_null_dev,
+                             const int64_t *vals_dev,
+                             const uint8_t *val_null_dev, int64_t n,
+                             void *ws_dev, size_t ws_bytes,
+                             otbx_dec_group *groups_dev,
+                             int64_t *ngroups_dev, void *stream);
+
 /* ---- GPU ORDER BY ----
  * Full sort of Q3 group rows by (revenue DESC, o_orderdate ASC) — the
  * tuplesort.c analog for the no-LIMIT ORDER BY case (LIMIT queries use

@@ -117,4 +117,5 @@ EXPORTED_SYMBOLS = [
     "otbx_join_i64_workspace_bytes", "otbx_join_i64",
     "otbx_join_ext_workspace_bytes", "otbx_join_i64_ext", "otbx_join_i64x2",
     "otbx_agg_i64x2_workspace_bytes", "otbx_agg_i64x2",
+    "otbx_agg_i64_dec_workspace_bytes", "otbx_agg_i64_dec",
 ]

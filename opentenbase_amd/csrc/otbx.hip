@@ -5394,3 +5394,136 @@ otbx_status otbx_build_q9recs(const otbx_lineitem_dev *l, void *recs_dev,
 }
 
 } /* extern "C" */
+
+/* ============ exact decimal aggregate (otbx_agg_i64_dec) ============
+ * Int128AggState {N, sumX} semantics (numeric.c:5072, do_int128_accum
+ * :4998, int8_avg_accum :5365): 128-bit two's-complement sum accumulated
+ * with a carry-propagating pair of 64-bit atomics — lo += (u64)v returns
+ * the old word, carry-out = (old + v) wrapped; hi += sign_extend(v) +
+ * carry. Exact for any input (|sum| <= 2^126), bit-exact parity. */
+
+struct dec_slot {
+    long long idx; /* claim word: -1 empty, else defining row index */
+    unsigned long long count_star;
+    unsigned long long count_v;
+    unsigned long long sum_lo;
+    unsigned long long sum_hi; /* two's complement */
+};
+
+__global__ void k_dec_init(dec_slot *tab, int64_t cap)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < cap;
+         i += stride) {
+        tab[i].idx = -1;
+        tab[i].count_star = 0;
+        tab[i].count_v = 0;
+        tab[i].sum_lo = 0;
+        tab[i].sum_hi = 0;
+    }
+}
+
+__global__ void k_dec_build(const int64_t *__restrict__ keys,
+                            const uint8_t *__restrict__ knull,
+                            const int64_t *__restrict__ vals,
+                            const uint8_t *__restrict__ vnull, int64_t n,
+                            dec_slot *tab, int64_t cap)
+{
+    int64_t mask = cap - 1;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        bool kn = knull && knull[i];
+        int64_t k = kn ? 0 : keys[i];
+        uint64_t h = kn ? 0x9e3779b97f4a7c15ull : d_hash_i64(k);
+        int64_t s = (int64_t)(h & (uint64_t)mask);
+        for (;;) {
+            long long owner = tab[s].idx;
+            if (owner < 0) {
+                long long prev = (long long)atomicCAS(
+                    (unsigned long long *)&tab[s].idx,
+                    (unsigned long long)(-1ll), (unsigned long long)i);
+                owner = prev == -1ll ? i : prev;
+            }
+            bool on = knull && knull[owner];
+            if (on == kn && (kn || keys[owner] == k))
+                break;
+            s = (s + 1) & mask;
+        }
+        atomicAdd(&tab[s].count_star, 1ull);
+        if (!(vnull && vnull[i])) {
+            int64_t v = vals[i];
+            atomicAdd(&tab[s].count_v, 1ull);
+            unsigned long long uv = (unsigned long long)v;
+            unsigned long long old = atomicAdd(&tab[s].sum_lo, uv);
+            unsigned long long nw = old + uv;
+            long long hid = (v < 0 ? -1ll : 0ll) + (nw < old ? 1ll : 0ll);
+            if (hid)
+                atomicAdd(&tab[s].sum_hi, (unsigned long long)hid);
+        }
+    }
+}
+
+__global__ void k_dec_compact(const dec_slot *__restrict__ tab, int64_t cap,
+                              const int64_t *__restrict__ keys,
+                              const uint8_t *__restrict__ knull,
+                              otbx_dec_group *__restrict__ out,
+                              int64_t *ngroups)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         s += stride) {
+        bool used = s < cap && tab[s].idx >= 0;
+        int64_t pos = wave_append(ngroups, used);
+        if (used) {
+            long long i = tab[s].idx;
+            bool kn = knull && knull[i];
+            out[pos].key = kn ? 0 : keys[i];
+            out[pos].count_star = (int64_t)tab[s].count_star;
+            out[pos].count_v = (int64_t)tab[s].count_v;
+            out[pos].sum_hi = (int64_t)tab[s].sum_hi;
+            out[pos].sum_lo = tab[s].sum_lo;
+            out[pos].key_isnull = kn;
+            out[pos].sum_isnull = tab[s].count_v == 0;
+        }
+        if (__all(s >= cap))
+            break;
+    }
+}
+
+extern "C" {
+
+otbx_status otbx_agg_i64_dec_workspace_bytes(int64_t n, size_t *bytes)
+{
+    *bytes = (size_t)jx_cap_for(n) * sizeof(dec_slot);
+    return OTBX_OK;
+}
+
+otbx_status otbx_agg_i64_dec(const int64_t *keys, const uint8_t *knull,
+                             const int64_t *vals, const uint8_t *vnull,
+                             int64_t n, void *ws, size_t ws_bytes,
+                             otbx_dec_group *groups_dev, int64_t *ngroups_dev,
+                             void *stream)
+{
+    if (!ws || !ngroups_dev || !keys || !vals)
+        return OTBX_ERR_INVALID;
+    hipStream_t s = (hipStream_t)stream;
+    int64_t cap = jx_cap_for(n);
+    size_t need;
+    otbx_agg_i64_dec_workspace_bytes(n, &need);
+    if (ws_bytes < need)
+        return OTBX_ERR_INVALID;
+    dec_slot *tab = (dec_slot *)ws;
+    HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
+    hipLaunchKernelGGL(k_dec_init, dim3(grid_for(cap, 256)), dim3(256), 0, s,
+                       tab, cap);
+    if (n > 0)
+        hipLaunchKernelGGL(k_dec_build, dim3(grid_for(n, 256)), dim3(256), 0,
+                           s, keys, knull, vals, vnull, n, tab, cap);
+    hipLaunchKernelGGL(k_dec_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
+                       s, tab, cap, keys, knull, groups_dev, ngroups_dev);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+} /* extern "C" */

@@ -544,6 +544,49 @@ class GpuHashAgg(CustomScanState):
         return list(arr)
 
 
+class GpuHashAggDec(CustomScanState):
+    """Exact decimal aggregate: scaled-int64 values, int128 sum
+    (Int128AggState semantics, numeric.c:5072/:4998/:5365). Bit-exact —
+    rows carry .sum128 as a Python int."""
+
+    def __init__(self, keys, vals, key_null=None, val_null=None):
+        super().__init__()
+        self.keys, self.vals = keys, vals
+        self.key_null, self.val_null = key_null, val_null
+
+    def _run(self):
+        import numpy as np
+        L = lib()
+        n = len(self.keys)
+        ws_bytes = C.c_size_t(0)
+        check(L.otbx_agg_i64_dec_workspace_bytes(C.c_int64(n),
+                                                 C.byref(ws_bytes)))
+        ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8,
+                         device="cuda")
+        out = torch.empty(max(n, 1) * 48, dtype=torch.uint8, device="cuda")
+        ng = torch.zeros(1, dtype=torch.int64, device="cuda")
+
+        def vp(t):
+            return C.c_void_p(t.data_ptr()) if t is not None else None
+
+        call("otbx_agg_i64_dec", vp(self.keys), vp(self.key_null),
+             vp(self.vals), vp(self.val_null), C.c_int64(n),
+             C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+             C.c_void_p(out.data_ptr()), C.c_void_p(ng.data_ptr()), _stream())
+        ngroups = int(ng.cpu().item())
+        dt = np.dtype([("key", "i8"), ("count_star", "i8"), ("count_v", "i8"),
+                       ("sum_hi", "i8"), ("sum_lo", "u8"),
+                       ("key_isnull", "i4"), ("sum_isnull", "i4")])
+        arr = out[: ngroups * 48].cpu().numpy().view(dt).copy()
+        arr.sort(order=["key_isnull", "key"])
+        rows = []
+        for r in arr:
+            d = {k: int(r[k]) for k in dt.names}
+            d["sum128"] = (int(r["sum_hi"]) << 64) | int(r["sum_lo"])
+            rows.append(d)
+        return rows
+
+
 JOIN_TYPES = {"inner": 0, "left": 1, "semi": 2, "anti": 3, "right": 4,
               "full": 5}
 

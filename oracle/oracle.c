@@ -924,3 +924,86 @@ fail:
     free(slots); free(used);
     return st;
 }
+
+
+/* ============== exact decimal aggregate (ora_agg_i64_dec) ==============
+ * Int128AggState semantics (numeric.c:5072 / do_int128_accum :4998):
+ * strict transition skips NULL inputs; the int128 sum of int64 values
+ * cannot overflow (|sum| <= 2^63 * 2^63 = 2^126). */
+
+static int dec_group_cmp(const void *pa, const void *pb)
+{
+    const ora_dec_group *a = pa, *b = pb;
+    if (a->key_isnull != b->key_isnull) return a->key_isnull ? 1 : -1;
+    if (a->key_isnull) return 0;
+    return a->key < b->key ? -1 : a->key > b->key ? 1 : 0;
+}
+
+ora_status ora_agg_i64_dec(const int64_t *keys, const uint8_t *key_null,
+                           const int64_t *vals, const uint8_t *val_null,
+                           int64_t n, ora_dec_group **out, int64_t *ngroups)
+{
+    typedef struct {
+        int64_t key;
+        int key_isnull;
+        int64_t count_star, count_v;
+        __int128 sum;
+        int sum_isnull;
+    } slot_t;
+    int64_t cap = next_pow2(n < 16 ? 16 : (int64_t)((double)n / 0.85) + 1);
+    slot_t *slots = calloc(cap, sizeof(slot_t));
+    uint8_t *used = calloc(cap, 1);
+    if (!slots || !used) { free(slots); free(used); return ORA_ERR_OOM; }
+    int64_t ng = 0;
+    ora_status st;
+
+    for (int64_t i = 0; i < n; i++) {
+        int isnull = key_null && key_null[i];
+        int64_t k = isnull ? 0 : keys[i];
+        uint64_t h = isnull ? 0x9e3779b97f4a7c15ull : ora_hash_i64(k);
+        int64_t s = (int64_t)(h & (uint64_t)(cap - 1));
+        for (;;) {
+            if (!used[s]) {
+                used[s] = 1;
+                slots[s].key = k;
+                slots[s].key_isnull = isnull;
+                slots[s].sum_isnull = 1;
+                ng++;
+                break;
+            }
+            if (slots[s].key_isnull == isnull && (isnull || slots[s].key == k))
+                break;
+            s = (s + 1) & (cap - 1);
+        }
+        slot_t *e = &slots[s];
+        if ((st = int8inc(&e->count_star))) goto fail;
+        if (!(val_null && val_null[i])) {           /* strict transition */
+            if ((st = int8inc(&e->count_v))) goto fail;
+            e->sum += (__int128)vals[i];            /* do_int128_accum */
+            e->sum_isnull = 0;
+        }
+    }
+
+    ora_dec_group *res = malloc((size_t)(ng > 0 ? ng : 1) *
+                                sizeof(ora_dec_group));
+    if (!res) { st = ORA_ERR_OOM; goto fail; }
+    int64_t j = 0;
+    for (int64_t s = 0; s < cap; s++)
+        if (used[s]) {
+            res[j].key = slots[s].key;
+            res[j].key_isnull = slots[s].key_isnull;
+            res[j].count_star = slots[s].count_star;
+            res[j].count_v = slots[s].count_v;
+            res[j].sum_hi = (int64_t)(slots[s].sum >> 64);
+            res[j].sum_lo = (uint64_t)slots[s].sum;
+            res[j].sum_isnull = slots[s].sum_isnull;
+            j++;
+        }
+    qsort(res, (size_t)ng, sizeof(ora_dec_group), dec_group_cmp);
+    free(slots); free(used);
+    *out = res; *ngroups = ng;
+    return ORA_OK;
+fail:
+    free(slots); free(used);
+    return st;
+}

@@ -205,6 +205,30 @@ ora_status ora_agg_i64x2(const int64_t *k1, const uint8_t *k1null,
                          const double *vals, const uint8_t *val_null,
                          int64_t n, ora_agg_group2 **out, int64_t *ngroups);
 
+
+/* ---- exact decimal (scaled-int64) aggregate with int128 sum ----
+ * Restates the reference's HAVE_INT128 numeric aggregation: group state =
+ * Int128AggState {N, sumX} (numeric.c:5072, do_int128_accum :4998 region),
+ * transition int8_avg_accum (numeric.c:5365); sum(bigint) promotes to
+ * numeric and cannot overflow (int8_sum numeric.c:6206). Values are
+ * scaled-decimal int64 (e.g. NUMERIC(15,2) money in cents); the emitted
+ * 128-bit sum (two's-complement hi/lo) is exact, so parity is bit-exact.
+ * AVG finalization (numeric division, sum/N) is the caller's / combine
+ * phase's job — the partial state is what a DataNode ships. */
+typedef struct {
+    int64_t key;
+    int key_isnull;
+    int64_t count_star;
+    int64_t count_v;
+    int64_t sum_hi;          /* int128 two's-complement high word */
+    uint64_t sum_lo;
+    int sum_isnull;
+} ora_dec_group;
+
+ora_status ora_agg_i64_dec(const int64_t *keys, const uint8_t *key_null,
+                           const int64_t *vals, const uint8_t *val_null,
+                           int64_t n, ora_dec_group **out, int64_t *ngroups);
+
 #ifdef __cplusplus
 }
 #endif

@@ -109,6 +109,7 @@ def load():
     lib.ora_join_i64.restype = C.c_int
     lib.ora_join_ext.restype = C.c_int
     lib.ora_agg_i64x2.restype = C.c_int
+    lib.ora_agg_i64_dec.restype = C.c_int
     lib.ora_q1_combine.restype = C.c_int
     return lib
 
@@ -355,5 +356,35 @@ def agg_i64x2(k1, k2, vals, k1null=None, k2null=None, val_null=None):
                          _p(k2, C.c_int64), _p(n2, C.c_uint8),
                          _p(vals, C.c_double), _p(vn, C.c_uint8),
                          C.c_int64(len(k1)), C.byref(out), C.byref(ng))
+    assert st == 0, st
+    return [out[i] for i in range(ng.value)]
+
+
+class OraDecGroup(C.Structure):
+    _fields_ = [
+        ("key", C.c_int64), ("key_isnull", C.c_int),
+        ("count_star", C.c_int64), ("count_v", C.c_int64),
+        ("sum_hi", C.c_int64), ("sum_lo", C.c_uint64),
+        ("sum_isnull", C.c_int),
+    ]
+
+    @property
+    def sum128(self):
+        return (self.sum_hi << 64) | self.sum_lo
+
+
+def agg_i64_dec(keys, vals, key_null=None, val_null=None):
+    """Exact int128 decimal aggregate (oracle.h ora_agg_i64_dec); returns
+    OraDecGroup list sorted by (key_isnull, key); .sum128 is the exact sum."""
+    L = lib()
+    keys = np.ascontiguousarray(keys, dtype=np.int64)
+    vals = np.ascontiguousarray(vals, dtype=np.int64)
+    kn = None if key_null is None else np.ascontiguousarray(key_null, np.uint8)
+    vn = None if val_null is None else np.ascontiguousarray(val_null, np.uint8)
+    out = C.POINTER(OraDecGroup)()
+    ng = C.c_int64(0)
+    st = L.ora_agg_i64_dec(_p(keys, C.c_int64), _p(kn, C.c_uint8),
+                           _p(vals, C.c_int64), _p(vn, C.c_uint8),
+                           C.c_int64(len(keys)), C.byref(out), C.byref(ng))
     assert st == 0, st
     return [out[i] for i in range(ng.value)]

@@ -187,3 +187,64 @@ def test_agg2_grid_golden():
     g = ora.agg_i64x2(a, b, np.ones(10))
     assert len(g) == 10
     assert all(x.count_star == 1 for x in g)
+
+
+# ---------------- exact decimal (int128) aggregate ----------------
+
+def brute_dec(keys, vals, kn=None, vn=None):
+    groups = {}
+    for i in range(len(keys)):
+        a = (bool(kn[i]) if kn is not None else False,
+             0 if (kn is not None and kn[i]) else int(keys[i]))
+        g = groups.setdefault(a, {"cs": 0, "cv": 0, "sum": None})
+        g["cs"] += 1
+        if vn is None or not vn[i]:
+            g["cv"] += 1
+            g["sum"] = (g["sum"] or 0) + int(vals[i])   # exact Python int
+    return groups
+
+
+def test_agg_dec_exact_random():
+    rng = np.random.default_rng(31)
+    for trial in range(6):
+        n = int(rng.integers(1, 500))
+        k = rng.integers(-3, 4, n)
+        v = rng.integers(np.iinfo(np.int64).min // 2,
+                         np.iinfo(np.int64).max // 2, n)
+        kn = (rng.random(n) < 0.2).astype(np.uint8)
+        vn = (rng.random(n) < 0.2).astype(np.uint8)
+        got = ora.agg_i64_dec(k, v, key_null=kn, val_null=vn)
+        exp = brute_dec(k, v, kn, vn)
+        assert len(got) == len(exp)
+        for g in got:
+            key = (bool(g.key_isnull), 0 if g.key_isnull else int(g.key))
+            e = exp[key]
+            assert g.count_star == e["cs"] and g.count_v == e["cv"]
+            if e["sum"] is None:
+                assert g.sum_isnull
+            else:
+                assert g.sum128 == e["sum"]            # EXACT int128
+
+
+def test_agg_dec_beyond_int64():
+    """Sums that overflow int64 stay exact in the int128 state — the
+    numeric-promotion property of int8_sum (numeric.c:6206)."""
+    n = 10000
+    k = np.zeros(n, dtype=np.int64)
+    v = np.full(n, np.iinfo(np.int64).max, dtype=np.int64)
+    g = ora.agg_i64_dec(k, v)
+    assert len(g) == 1
+    assert g[0].sum128 == n * (2**63 - 1)              # > int64 range
+    vneg = np.full(n, np.iinfo(np.int64).min, dtype=np.int64)
+    g = ora.agg_i64_dec(k, vneg)
+    assert g[0].sum128 == n * (-2**63)
+
+
+def test_agg_dec_empty_allnull():
+    e = np.empty(0, dtype=np.int64)
+    assert ora.agg_i64_dec(e, e) == []
+    k = np.zeros(4, dtype=np.int64)
+    v = np.arange(4, dtype=np.int64)
+    g = ora.agg_i64_dec(k, v, val_null=np.ones(4, dtype=np.uint8))
+    assert len(g) == 1 and g[0].sum_isnull and g[0].count_star == 4 \
+        and g[0].count_v == 0
