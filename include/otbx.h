@@ -62,7 +62,6 @@
  *                            overrides (tests force the fallback)
  *   OTBX_Q9_FILTER_WAVE=1  — legacy per-wave appender in the Q9 part filter
  *                            (default is the tile-staged compaction; A/B)
- *   OTBX_Q9_FILTER8=1      — 8-rows/thread single-sweep tile filter (A/B)
  *   OTBX_Q3_COMPACT_LEGACY=1 — legacy block-chunk Q3 group compaction
  *                            (default is the tile-staged variant; A/B)
  */

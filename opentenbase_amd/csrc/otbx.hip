@@ -2268,83 +2268,6 @@ __global__ void k_q9_build_recs(const otbx_lineitem_dev l,
     }
 }
 
-/* phase 1 (8-rows/thread variant, OTBX_Q9_FILTER8=1): one 8192-row sweep
- * per tile — 8 concurrent bitmap gathers in flight per thread (vs 4) and
- * a third of the barriers. A/B against k_q9_filter_tile. */
-__global__ __launch_bounds__(1024) void k_q9_filter_tile8(
-    const int64_t *__restrict__ pk, int64_t n,
-    const unsigned long long *__restrict__ pbitmap, int64_t lo_k,
-    int64_t hi_k, uint32_t *__restrict__ hits, int64_t *nhits)
-{
-    __shared__ uint32_t stage[Q9T];
-    __shared__ int wtot[16];
-    __shared__ int woff[16];
-    __shared__ int tot_s;
-    __shared__ long long gbase;
-    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
-    const v2l *pk2 = (const v2l *)pk;
-    int64_t ntiles = (n + Q9T - 1) / Q9T;
-    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
-        int64_t tl = t * (int64_t)Q9T;
-        int64_t r0 = tl + 8 * (int64_t)threadIdx.x;
-        bool m[8];
-        int mycnt = 0;
-        if (r0 + 7 < n) {
-            v2l pa = __builtin_nontemporal_load(&pk2[r0 / 2]);
-            v2l pb = __builtin_nontemporal_load(&pk2[r0 / 2 + 1]);
-            v2l pc = __builtin_nontemporal_load(&pk2[r0 / 2 + 2]);
-            v2l pd = __builtin_nontemporal_load(&pk2[r0 / 2 + 3]);
-            int64_t pks[8] = {pa.x, pa.y, pb.x, pb.y, pc.x, pc.y, pd.x, pd.y};
-#pragma unroll
-            for (int j = 0; j < 8; j++) {
-                int64_t pidx = pks[j] - 1 - lo_k;
-                m[j] = pidx >= 0 && pidx < hi_k - lo_k &&
-                       ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
-                mycnt += m[j];
-            }
-        } else {
-#pragma unroll
-            for (int j = 0; j < 8; j++) {
-                m[j] = false;
-                if (r0 + j < n) {
-                    int64_t pidx = pk[r0 + j] - 1 - lo_k;
-                    m[j] = pidx >= 0 && pidx < hi_k - lo_k &&
-                           ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
-                    mycnt += m[j];
-                }
-            }
-        }
-        int incl = mycnt;
-        for (int off = 1; off < WAVE; off <<= 1) {
-            int up = __shfl_up(incl, off, WAVE);
-            if (lane >= off) incl += up;
-        }
-        if (lane == WAVE - 1) wtot[wid] = incl;
-        __syncthreads();
-        if (threadIdx.x == 0) {
-            int acc = 0;
-            for (int w = 0; w < 16; w++) {
-                woff[w] = acc;
-                acc += wtot[w];
-            }
-            tot_s = acc;
-            gbase = acc ? (long long)atomicAdd((unsigned long long *)nhits,
-                                               (unsigned long long)acc)
-                        : 0;
-        }
-        __syncthreads();
-        int pos = woff[wid] + incl - mycnt;
-#pragma unroll
-        for (int j = 0; j < 8; j++)
-            if (m[j]) stage[pos++] = (uint32_t)(r0 + j - tl);
-        __syncthreads();
-        int tot = tot_s;
-        for (int p = threadIdx.x; p < tot; p += blockDim.x)
-            hits[gbase + p] = stage[p];
-        __syncthreads();
-    }
-}
-
 /* phase 2: dense pass over the survivors — orders date lookup + year
  * partial aggregate in per-lane registers (the Q1 pattern: 7-year domain,
  * compile-time indexed), wave+block reduce, one atomic per (year, block). */
@@ -2552,15 +2475,10 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
             rec0 = true;
         }
         const char *fw = getenv("OTBX_Q9_FILTER_WAVE");
-        const char *f8 = getenv("OTBX_Q9_FILTER8");
         if (fw && atoi(fw)) /* legacy per-wave appender (A/B) */
             hipLaunchKernelGGL(k_q9_filter, dim3(grid_for(l->n / 4, 256)),
                                dim3(256), 0, s, l->l_partkey, l->n, pbitmap,
                                lo, hi, hits, nhits);
-        else if (f8 && atoi(f8)) /* 8-rows/thread single-sweep tile (A/B) */
-            hipLaunchKernelGGL(k_q9_filter_tile8, dim3(2048), dim3(1024), 0,
-                               s, l->l_partkey, l->n, pbitmap, lo, hi, hits,
-                               nhits);
         else
             hipLaunchKernelGGL(k_q9_filter_tile, dim3(2048), dim3(1024), 0,
                                s, l->l_partkey, l->n, pbitmap, lo, hi, hits,
