@@ -134,6 +134,11 @@ typedef struct {
      * the Q9 probe's per-survivor gather then touches one cache line
      * instead of three column lines. NULL = probe gathers the columns. */
     void *q9rec;
+    /* optional compact-key cache (otbx_build_key32, built once at staging):
+     * int32 copy of l_orderkey, valid iff every key fits [0, 2^31) — at
+     * TPC-H scales orderkeys do (SF100: max 6e8). Halves the probe's key
+     * stream. NULL = kernels read the i64 column. */
+    int32_t *l_orderkey32;
 } otbx_lineitem_dev;
 
 typedef struct {
@@ -145,6 +150,8 @@ typedef struct {
      * pages). has_minmax = 0 → executors run their own minmax kernel. */
     int64_t okey_min, okey_max;
     int32_t has_minmax;
+    /* optional compact-key caches (otbx_build_key32; see lineitem note) */
+    int32_t *o_orderkey32, *o_custkey32;
 } otbx_orders_dev;
 
 typedef struct {
@@ -163,6 +170,13 @@ typedef struct {
  * functions as the CPU oracle — oracle/otbx_gen.h — so tables are
  * bit-identical on both sides). Caller provides the device buffers
  * (column pointers in the struct, each sized for n_global/nranks rows). */
+/* build an int32 compact-key cache from an i64 key column: writes
+ * saturated casts into dst32_dev and *ok_host = 1 if every value fit
+ * [0, 2^31), else 0 (caller must then leave the cache pointer NULL).
+ * Synchronous (staging-time only). */
+otbx_status otbx_build_key32(const int64_t *src_dev, int64_t n,
+                             int32_t *dst32_dev, int32_t *ok_host,
+                             void *stream);
 /* build the q9rec cache from the staged columns (32 B/row into recs_dev) */
 otbx_status otbx_build_q9recs(const otbx_lineitem_dev *l, void *recs_dev,
                               void *stream);

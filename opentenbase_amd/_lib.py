@@ -35,6 +35,7 @@ class LineitemDev(C.Structure):
         ("l_shipdate", C.c_void_p),
         ("l_partkey", C.c_void_p),
         ("q9rec", C.c_void_p),
+        ("l_orderkey32", C.c_void_p),
     ]
 
 
@@ -45,6 +46,7 @@ class OrdersDev(C.Structure):
         ("o_orderdate", C.c_void_p), ("o_shippriority", C.c_void_p),
         ("okey_min", C.c_int64), ("okey_max", C.c_int64),
         ("has_minmax", C.c_int32),
+        ("o_orderkey32", C.c_void_p), ("o_custkey32", C.c_void_p),
     ]
 
 
@@ -106,7 +108,7 @@ EXPORTED_SYMBOLS = [
     "otbx_memcpy_d2h", "otbx_stream_sync",
     "otbx_gen_lineitem_dev", "otbx_gen_orders_dev", "otbx_gen_customer_dev",
     "otbx_gen_part_dev", "otbx_q9_workspace_bytes", "otbx_q9_partial",
-    "otbx_stage_pages", "otbx_build_q9recs",
+    "otbx_stage_pages", "otbx_build_q9recs", "otbx_build_key32",
     "otbx_scan_count", "otbx_q1_partial", "otbx_q1_partial_variant",
     "otbx_q3_workspace_bytes", "otbx_q3_partial", "otbx_filter_customer",
     "otbx_topk_by_revenue",

@@ -3786,6 +3786,7 @@ struct q3_rec {
  * overlapped with the customer phase), the date qual + customer filter
  * writes the bitmap/payload table directly — no candidate list and no
  * separate insert pass (measured 0.92 + 0.91 ms as two kernels). */
+template <bool K32>
 __global__ void k_ord_filter_insert_fused(
     const otbx_orders_dev o, int32_t q3date,
     const unsigned long long *__restrict__ ckeys, int64_t ccap,
@@ -3798,16 +3799,27 @@ __global__ void k_ord_filter_insert_fused(
     const int4 *op4 = (const int4 *)o.o_shippriority;
     const longlong2 *ck2 = (const longlong2 *)o.o_custkey;
     const longlong2 *okk2 = (const longlong2 *)o.o_orderkey;
+    /* K32 (otbx.h o_orderkey32/o_custkey32): 24 -> 16 B/row stream */
+    const int4 *ck4 = (const int4 *)o.o_custkey32;
+    const int4 *ok4 = (const int4 *)o.o_orderkey32;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
          q += stride) {
         int4 d = od4[q];
-        longlong2 ca = ck2[2 * q], cb = ck2[2 * q + 1];
-        longlong2 ka = okk2[2 * q], kb = okk2[2 * q + 1];
+        int64_t ck[4];
+        int64_t ok[4];
+        if (K32) {
+            int4 cc = ck4[q], kk = ok4[q];
+            ck[0] = cc.x; ck[1] = cc.y; ck[2] = cc.z; ck[3] = cc.w;
+            ok[0] = kk.x; ok[1] = kk.y; ok[2] = kk.z; ok[3] = kk.w;
+        } else {
+            longlong2 ca = ck2[2 * q], cb = ck2[2 * q + 1];
+            longlong2 ka = okk2[2 * q], kb = okk2[2 * q + 1];
+            ck[0] = ca.x; ck[1] = ca.y; ck[2] = cb.x; ck[3] = cb.y;
+            ok[0] = ka.x; ok[1] = ka.y; ok[2] = kb.x; ok[3] = kb.y;
+        }
         int4 pr = op4[q];
         int32_t ds[4] = {d.x, d.y, d.z, d.w};
-        int64_t ck[4] = {ca.x, ca.y, cb.x, cb.y};
-        int64_t ok[4] = {ka.x, ka.y, kb.x, kb.y};
         int32_t prio[4] = {pr.x, pr.y, pr.z, pr.w};
 #pragma unroll
         for (int j = 0; j < 4; j++) {
@@ -3889,6 +3901,7 @@ __global__ void k_ord_insert_direct(const otbx_orders_dev o,
  * HASH path's divergent chain walk; on the direct path the "probe" is a
  * single clustered table read + one atomic, so materializing candidates
  * (write + re-read + key re-gather) only costs bandwidth. */
+template <bool K32>
 __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
                                            int32_t q3date, int64_t mino,
                                            int64_t range,
@@ -3899,6 +3912,9 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
     int64_t nq = l.n / 4;
     const int4 *sd4 = (const int4 *)l.l_shipdate;
     const longlong2 *ok2 = (const longlong2 *)l.l_orderkey;
+    /* K32: the staged compact-key cache (otbx.h l_orderkey32) halves the
+     * orderkey stream — 12 → 8 B/row on this kernel's dominant traffic */
+    const int4 *ok4 = (const int4 *)l.l_orderkey32;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     unsigned long long myhits = 0;
     for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
@@ -3907,9 +3923,15 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
          * profiles/r5_q3.json) — the streams already miss L2 and the nt
          * hint bought nothing */
         int4 d = sd4[q];
-        longlong2 ka = ok2[2 * q], kb = ok2[2 * q + 1];
+        int64_t ky[4];
+        if (K32) {
+            int4 kk = ok4[q];
+            ky[0] = kk.x; ky[1] = kk.y; ky[2] = kk.z; ky[3] = kk.w;
+        } else {
+            longlong2 ka = ok2[2 * q], kb = ok2[2 * q + 1];
+            ky[0] = ka.x; ky[1] = ka.y; ky[2] = kb.x; ky[3] = kb.y;
+        }
         int32_t ds[4] = {d.x, d.y, d.z, d.w};
-        int64_t ky[4] = {ka.x, ka.y, kb.x, kb.y};
         unsigned long long pl[4];
         bool m[4];
 #pragma unroll
@@ -4694,15 +4716,27 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
              * it needs the full-range zeroing the default path skips */
             HIP_CHECK(hipMemsetAsync(dgrec, 0, (size_t)range_all * 16, s));
         }
-        hipLaunchKernelGGL(k_ord_filter_insert_fused,
-                           dim3(grid_for(o->n / 4, 256)), dim3(256), 0, s, *o,
-                           q3date, ctab, ccap, cbitmap, cmin, crange,
-                           mino_all, range_all, dbitmap, dgrec);
+        if (o->o_orderkey32 && o->o_custkey32)
+            hipLaunchKernelGGL((k_ord_filter_insert_fused<true>),
+                               dim3(grid_for(o->n / 4, 256)), dim3(256), 0, s,
+                               *o, q3date, ctab, ccap, cbitmap, cmin, crange,
+                               mino_all, range_all, dbitmap, dgrec);
+        else
+            hipLaunchKernelGGL((k_ord_filter_insert_fused<false>),
+                               dim3(grid_for(o->n / 4, 256)), dim3(256), 0, s,
+                               *o, q3date, ctab, ccap, cbitmap, cmin, crange,
+                               mino_all, range_all, dbitmap, dgrec);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[2], s));
-        hipLaunchKernelGGL(k_q3_scan_probe_agg_direct,
-                           dim3(grid_for(l->n / 4, 256)), dim3(256), 0, s, *l,
-                           q3date, mino_all, range_all, dbitmap, dgrec,
-                           nhits);
+        if (l->l_orderkey32)
+            hipLaunchKernelGGL((k_q3_scan_probe_agg_direct<true>),
+                               dim3(grid_for(l->n / 4, 256)), dim3(256), 0, s,
+                               *l, q3date, mino_all, range_all, dbitmap,
+                               dgrec, nhits);
+        else
+            hipLaunchKernelGGL((k_q3_scan_probe_agg_direct<false>),
+                               dim3(grid_for(l->n / 4, 256)), dim3(256), 0, s,
+                               *l, q3date, mino_all, range_all, dbitmap,
+                               dgrec, nhits);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[3], s));
         if (q3_compact_legacy())
             hipLaunchKernelGGL(k_q3_compact_direct,
@@ -4766,10 +4800,16 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                 HIP_CHECK(hipEventRecord(ev[2], s));
                 rec2 = true;
             }
-            hipLaunchKernelGGL(k_q3_scan_probe_agg_direct,
-                               dim3(grid_for(l->n / 4, 256)), dim3(256), 0, s,
-                               *l, q3date, pmin, prange, dbitmap, dgrec,
-                               nhits);
+            if (l->l_orderkey32)
+                hipLaunchKernelGGL((k_q3_scan_probe_agg_direct<true>),
+                                   dim3(grid_for(l->n / 4, 256)), dim3(256),
+                                   0, s, *l, q3date, pmin, prange, dbitmap,
+                                   dgrec, nhits);
+            else
+                hipLaunchKernelGGL((k_q3_scan_probe_agg_direct<false>),
+                                   dim3(grid_for(l->n / 4, 256)), dim3(256),
+                                   0, s, *l, q3date, pmin, prange, dbitmap,
+                                   dgrec, nhits);
             if (kernel_ms && !rec3 && pass == npasses - 1) {
                 HIP_CHECK(hipEventRecord(ev[3], s));
                 rec3 = true;
@@ -5522,6 +5562,45 @@ otbx_status otbx_agg_i64_dec(const int64_t *keys, const uint8_t *knull,
                            s, keys, knull, vals, vnull, n, tab, cap);
     hipLaunchKernelGGL(k_dec_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
                        s, tab, cap, keys, knull, groups_dev, ngroups_dev);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+__global__ void k_build_key32(const int64_t *__restrict__ src, int64_t n,
+                              int32_t *__restrict__ dst,
+                              unsigned int *bad)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    unsigned int mybad = 0;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        int64_t v = src[i];
+        mybad |= (v < 0 || v > 0x7fffffffll);
+        dst[i] = (int32_t)v;
+    }
+    if (__ballot(mybad) && (threadIdx.x % WAVE) == 0)
+        atomicOr(bad, 1u);
+}
+
+otbx_status otbx_build_key32(const int64_t *src, int64_t n, int32_t *dst,
+                             int32_t *ok_host, void *stream)
+{
+    if (!src || !dst || !ok_host)
+        return OTBX_ERR_INVALID;
+    hipStream_t s = (hipStream_t)stream;
+    static unsigned int *d_bad = nullptr;
+    static unsigned int *h_bad = nullptr;
+    if (!d_bad) {
+        SCR_ALLOC_DEV(d_bad, 4);
+        SCR_ALLOC_HOST(h_bad, 4);
+    }
+    HIP_CHECK(hipMemsetAsync(d_bad, 0, 4, s));
+    if (n > 0)
+        hipLaunchKernelGGL(k_build_key32, dim3(grid_for(n, 256)), dim3(256),
+                           0, s, src, n, dst, d_bad);
+    HIP_CHECK(hipMemcpyAsync(h_bad, d_bad, 4, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    *ok_host = *h_bad ? 0 : 1;
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }

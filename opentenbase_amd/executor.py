@@ -40,6 +40,17 @@ def device_synchronize():
 
 # ---------------- staged tables (device-resident column cache) -------------
 
+def _build_key32(src_tensor):
+    """Staging-time compact-key cache (otbx.h otbx_build_key32): int32 copy
+    of an i64 key column, or None if any key falls outside [0, 2^31)."""
+    n = src_tensor.numel()
+    dst = torch.empty(n, dtype=torch.int32, device="cuda")
+    ok = C.c_int32(0)
+    call("otbx_build_key32", C.c_void_p(src_tensor.data_ptr()), C.c_int64(n),
+         C.c_void_p(dst.data_ptr()), C.byref(ok), _stream())
+    return dst if ok.value else None
+
+
 class GpuLineitem:
     COLS = [("l_orderkey", torch.int64), ("l_quantity", torch.float64),
             ("l_extendedprice", torch.float64), ("l_discount", torch.float64),
@@ -95,6 +106,11 @@ class GpuLineitem:
     def _stage_q9recs(self):
         """Q9 probe-side record cache (otbx.h q9rec): built once at
         staging, outside any timed region, like the zone-map metadata."""
+        if self.t.get("l_orderkey") is not None:
+            self._okey32 = _build_key32(self.t["l_orderkey"])
+            if self._okey32 is not None:
+                self.cstruct.l_orderkey32 = C.c_void_p(
+                    self._okey32.data_ptr())
         if self.t.get("l_partkey") is None or self.t.get("l_orderkey") is None:
             return
         self._q9rec = torch.empty(self.n * 32, dtype=torch.uint8,
@@ -128,7 +144,16 @@ class GpuOrders:
         call("otbx_gen_orders_dev", C.byref(t.cstruct), C.c_uint64(seed),
              C.c_int64(n_global), C.c_int64(ncust_global), C.c_uint32(rank),
              C.c_uint32(nranks), C.c_int(1 if skew else 0), _stream())
+        t._stage_key32()
         return t
+
+    def _stage_key32(self):
+        self._okey32 = _build_key32(self.t["o_orderkey"])
+        self._ckey32 = _build_key32(self.t["o_custkey"])
+        if self._okey32 is not None:
+            self.cstruct.o_orderkey32 = C.c_void_p(self._okey32.data_ptr())
+        if self._ckey32 is not None:
+            self.cstruct.o_custkey32 = C.c_void_p(self._ckey32.data_ptr())
 
 
 class GpuCustomer:

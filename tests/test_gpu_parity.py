@@ -1153,3 +1153,42 @@ def test_agg_dec_parity_gpu(ex, ora):
     v1 = torch.full((m,), 2**63 - 1, dtype=torch.int64, device="cuda")
     rows = drain(ex.GpuHashAggDec(k1, v1))
     assert len(rows) == 1 and rows[0]["sum128"] == m * (2**63 - 1)
+
+
+def test_q3_key32_vs_i64_selfconsistent(ex):
+    """The staged compact-key (int32) path vs the i64 fallback (cache
+    pointers nulled) on the same tables: identical group sets and top-k."""
+    import ctypes as CT
+    n = 1_000_000
+    li = ex.GpuLineitem.generate(n)
+    od = ex.GpuOrders.generate(n // 4, n // 40)
+    cu = ex.GpuCustomer.generate(n // 40)
+    assert li.cstruct.l_orderkey32      # caches built at this scale
+    assert od.cstruct.o_orderkey32 and od.cstruct.o_custkey32
+
+    def run():
+        node = ex.GpuQ3Fragment(cu, od, li)
+        node.BeginCustomScan()
+        node._run()
+        g = node.fetch_groups()
+        return sorted((int(r["l_orderkey"]), float(r["revenue"]),
+                       int(r["o_orderdate"]), int(r["o_shippriority"]))
+                      for r in g)
+
+    k32 = run()
+    li.cstruct.l_orderkey32 = CT.c_void_p(0)
+    od.cstruct.o_orderkey32 = CT.c_void_p(0)
+    od.cstruct.o_custkey32 = CT.c_void_p(0)
+    i64 = run()
+    assert k32 == i64
+
+
+def test_build_key32_overflow_detect(ex):
+    """Keys outside [0, 2^31) must void the cache (staging-time check)."""
+    from opentenbase_amd.executor import _build_key32
+    ok = torch.tensor([1, 2, 3], dtype=torch.int64, device="cuda")
+    assert _build_key32(ok) is not None
+    bad = torch.tensor([1, 2**31, 3], dtype=torch.int64, device="cuda")
+    assert _build_key32(bad) is None
+    neg = torch.tensor([1, -2, 3], dtype=torch.int64, device="cuda")
+    assert _build_key32(neg) is None
