@@ -64,6 +64,7 @@
  *                            (default is the tile-staged compaction; A/B)
  *   OTBX_Q3_COMPACT_LEGACY=1 — legacy block-chunk Q3 group compaction
  *                            (default is the tile-staged variant; A/B)
+ *   OTBX_Q3_COMPACT_WORD=1 — word-granular bitmap-walk compaction (A/B)
  */
 #ifndef OTBX_H
 #define OTBX_H

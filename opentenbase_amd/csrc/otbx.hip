@@ -3994,9 +3994,118 @@ __global__ void k_q3_scan_probe_agg_direct(const otbx_lineitem_dev l,
         atomicAdd(nhits, myhits);
 }
 
+/* Word-granular tile compaction (A/B vs k_q3_compact_tile): each thread
+ * owns one 64-entry bitmap word per sweep — the bitmap is read ONCE (the
+ * quad-granular tile reads each word 16x) and the per-entry sweep
+ * machinery collapses to a popcount + a short set-bit loop. Tile = 1024
+ * words = 65536 entries; stage cap 32768 survivor ids (64 KB LDS) with a
+ * per-thread direct-write fallback for denser-than-50% tiles. */
+#define Q3CW (1024 * 64)
+#define Q3CW_STAGE 32768
+__global__ __launch_bounds__(1024) void k_q3_compact_word(
+    const q3_rec *__restrict__ grec,
+    const unsigned long long *__restrict__ bitmap, int64_t range,
+    int64_t mino, otbx_q3_group *out, int64_t cap_out, int64_t *ngroups)
+{
+    __shared__ uint16_t stage[Q3CW_STAGE];
+    __shared__ int wtot[16];
+    __shared__ int woff[16];
+    __shared__ int tot_s;
+    __shared__ long long gbase;
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int64_t nwords = (range + 63) / 64;
+    int64_t ntiles = (nwords + 1023) / 1024;
+    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+        int64_t w0 = t * 1024 + threadIdx.x;     /* my bitmap word */
+        int64_t tl = t * (int64_t)Q3CW;          /* tile base entry */
+        unsigned long long w = w0 < nwords ? bitmap[w0] : 0ull;
+        /* drop bits past range in the last word */
+        if (w0 == nwords - 1 && (range & 63))
+            w &= (1ull << (range & 63)) - 1ull;
+        /* count survivors in my word (rev != 0 filters no-match builds) */
+        int64_t base_e = w0 * 64;
+        int mycnt = 0;
+        unsigned long long wm = w;
+        unsigned long long keep = 0;
+        while (wm) {
+            int b = __builtin_ctzll(wm);
+            wm &= wm - 1;
+            if (grec[base_e + b].rev != 0.0) {
+                keep |= 1ull << b;
+                mycnt++;
+            }
+        }
+        int incl = mycnt;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            int up = __shfl_up(incl, off, WAVE);
+            if (lane >= off) incl += up;
+        }
+        if (lane == WAVE - 1) wtot[wid] = incl;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int acc = 0;
+            for (int ww = 0; ww < 16; ww++) {
+                woff[ww] = acc;
+                acc += wtot[ww];
+            }
+            tot_s = acc;
+            gbase = acc ? (long long)atomicAdd((unsigned long long *)ngroups,
+                                               (unsigned long long)acc)
+                        : 0;
+        }
+        __syncthreads();
+        int pos = woff[wid] + incl - mycnt;
+        int tot = tot_s;
+        if (tot <= Q3CW_STAGE) {
+            unsigned long long km = keep;
+            while (km) {
+                int b = __builtin_ctzll(km);
+                km &= km - 1;
+                stage[pos++] = (uint16_t)(base_e + b - tl);
+            }
+            __syncthreads();
+            for (int p = threadIdx.x; p < tot; p += blockDim.x) {
+                int64_t gp = gbase + p;
+                if (gp >= cap_out) continue;
+                int64_t i = tl + (int64_t)stage[p];
+                q3_rec r = grec[i];
+                out[gp].l_orderkey = mino + i;
+                out[gp].revenue = r.rev;
+                out[gp].o_orderdate = (int32_t)(r.pl & 0xffffffffull);
+                out[gp].o_shippriority = (int32_t)(r.pl >> 32);
+            }
+        } else {
+            /* dense tile (>50% survivors): direct per-thread writes */
+            unsigned long long km = keep;
+            int64_t gp = gbase + pos;
+            while (km) {
+                int b = __builtin_ctzll(km);
+                km &= km - 1;
+                if (gp < cap_out) {
+                    int64_t i = base_e + b;
+                    q3_rec r = grec[i];
+                    out[gp].l_orderkey = mino + i;
+                    out[gp].revenue = r.rev;
+                    out[gp].o_orderdate = (int32_t)(r.pl & 0xffffffffull);
+                    out[gp].o_shippriority = (int32_t)(r.pl >> 32);
+                }
+                gp++;
+            }
+            __syncthreads();
+        }
+        __syncthreads();
+    }
+}
+
 static inline bool q3_compact_legacy(void)
 {
     const char *e = getenv("OTBX_Q3_COMPACT_LEGACY");
+    return e && atoi(e);
+}
+
+static inline bool q3_compact_word(void)
+{
+    const char *e = getenv("OTBX_Q3_COMPACT_WORD");
     return e && atoi(e);
 }
 
@@ -4743,6 +4852,10 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                dim3(grid_for(range_all, 256)), dim3(256), 0,
                                s, dgrec, range_all, mino_all,
                                groups_dev, cap_groups, ngroups_dev);
+        else if (q3_compact_word())
+            hipLaunchKernelGGL(k_q3_compact_word, dim3(2048), dim3(1024), 0,
+                               s, dgrec, dbitmap, range_all, mino_all,
+                               groups_dev, cap_groups, ngroups_dev);
         else
             hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024), 0,
                                s, dgrec, dbitmap, range_all, mino_all,
@@ -4819,6 +4932,10 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                    dim3(grid_for(prange, 256)), dim3(256), 0,
                                    s, dgrec, prange, pmin, groups_dev,
                                    cap_groups, ngroups_dev);
+            else if (q3_compact_word())
+                hipLaunchKernelGGL(k_q3_compact_word, dim3(2048), dim3(1024),
+                                   0, s, dgrec, dbitmap, prange, pmin,
+                                   groups_dev, cap_groups, ngroups_dev);
             else
                 hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024),
                                    0, s, dgrec, dbitmap, prange, pmin,
