@@ -63,8 +63,8 @@
  *   OTBX_Q9_FILTER_WAVE=1  — legacy per-wave appender in the Q9 part filter
  *                            (default is the tile-staged compaction; A/B)
  *   OTBX_Q3_COMPACT_LEGACY=1 — legacy block-chunk Q3 group compaction
- *                            (default is the tile-staged variant; A/B)
- *   OTBX_Q3_COMPACT_WORD=1 — word-granular bitmap-walk compaction (A/B)
+ *                            (default is the word-granular bitmap walk)
+ *   OTBX_Q3_COMPACT_TILE=1 — quad-granular tile-staged compaction (A/B)
  */
 #ifndef OTBX_H
 #define OTBX_H

@@ -4103,9 +4103,9 @@ static inline bool q3_compact_legacy(void)
     return e && atoi(e);
 }
 
-static inline bool q3_compact_word(void)
+static inline bool q3_compact_tile_sel(void)
 {
-    const char *e = getenv("OTBX_Q3_COMPACT_WORD");
+    const char *e = getenv("OTBX_Q3_COMPACT_TILE");
     return e && atoi(e);
 }
 
@@ -4852,12 +4852,12 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                dim3(grid_for(range_all, 256)), dim3(256), 0,
                                s, dgrec, range_all, mino_all,
                                groups_dev, cap_groups, ngroups_dev);
-        else if (q3_compact_word())
-            hipLaunchKernelGGL(k_q3_compact_word, dim3(2048), dim3(1024), 0,
+        else if (q3_compact_tile_sel())
+            hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024), 0,
                                s, dgrec, dbitmap, range_all, mino_all,
                                groups_dev, cap_groups, ngroups_dev);
-        else
-            hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024), 0,
+        else /* default: word-granular (A/B winner, profiles/r12_*) */
+            hipLaunchKernelGGL(k_q3_compact_word, dim3(2048), dim3(1024), 0,
                                s, dgrec, dbitmap, range_all, mino_all,
                                groups_dev, cap_groups, ngroups_dev);
         if (kernel_ms) HIP_CHECK(hipEventRecord(ev[4], s));
@@ -4932,12 +4932,12 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                    dim3(grid_for(prange, 256)), dim3(256), 0,
                                    s, dgrec, prange, pmin, groups_dev,
                                    cap_groups, ngroups_dev);
-            else if (q3_compact_word())
-                hipLaunchKernelGGL(k_q3_compact_word, dim3(2048), dim3(1024),
+            else if (q3_compact_tile_sel())
+                hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024),
                                    0, s, dgrec, dbitmap, prange, pmin,
                                    groups_dev, cap_groups, ngroups_dev);
             else
-                hipLaunchKernelGGL(k_q3_compact_tile, dim3(2048), dim3(1024),
+                hipLaunchKernelGGL(k_q3_compact_word, dim3(2048), dim3(1024),
                                    0, s, dgrec, dbitmap, prange, pmin,
                                    groups_dev, cap_groups, ngroups_dev);
         }

@@ -972,10 +972,13 @@ def test_q3_compact_tile_vs_legacy(ex, monkeypatch):
                        int(r["o_orderdate"]), int(r["o_shippriority"]))
                       for r in g)
 
+    word = run()                      # default: word-granular
+    monkeypatch.setenv("OTBX_Q3_COMPACT_TILE", "1")
     tile = run()
+    monkeypatch.delenv("OTBX_Q3_COMPACT_TILE")
     monkeypatch.setenv("OTBX_Q3_COMPACT_LEGACY", "1")
     legacy = run()
-    assert tile == legacy
+    assert word == tile == legacy
 
 
 # ---------------- extended joins + two-key operators (r2 widening) --------
