@@ -295,3 +295,61 @@ def test_merge_q3_topk_empty_shard():
     top = _run(_worker_topk_empty)
     assert top["revenue"].tolist() == [9.0, 7.0]
     assert top["l_orderkey"].tolist() == [2, 3]
+
+
+def _worker_dist_leftjoin(rank, world, port, q):
+    """Repartitioned LEFT JOIN across ranks: both sides exchanged by
+    owner = key % world (NULL keys to rank 0 — the reference routes NULL
+    distribution keys to a designated node), rank-local ora_join_ext(left),
+    union of pair sets == the full-table oracle left join. Models the
+    'Distribute results by H' + outer-join plan shape (xl_join.out:13-20,
+    nodeHashjoin.c HJ_FILL_OUTER_TUPLE)."""
+    from opentenbase_amd import fragment
+    from oracle import oracle_py as ora
+    _init(rank, world, port)
+    rng = np.random.default_rng(5)
+    nb, npr = 300, 900
+    bk = rng.integers(0, 40, nb)
+    pk = rng.integers(0, 40, npr)
+    bn = (rng.random(nb) < 0.1).astype(np.uint8)
+    pn = (rng.random(npr) < 0.1).astype(np.uint8)
+
+    def shard_and_exchange(keys, nulls):
+        # this rank starts with the round-robin shard, then exchanges by
+        # owner = key % world (NULL rows -> rank 0)
+        mine = np.arange(len(keys)) % world == rank
+        ids = np.nonzero(mine)[0]
+        owner = np.where(nulls[ids] == 1, 0, keys[ids] % world)
+        order = np.argsort(owner, kind="stable")
+        ids = ids[order]
+        counts = [int((owner == r).sum()) for r in range(world)]
+        out_k = fragment.all_to_all_variable(
+            torch.as_tensor(keys[ids]), counts)
+        out_i = fragment.all_to_all_variable(
+            torch.as_tensor(ids), counts)
+        out_n = fragment.all_to_all_variable(
+            torch.as_tensor(nulls[ids]), counts)
+        return (out_k.numpy(), out_i.numpy(),
+                out_n.numpy().astype(np.uint8))
+
+    lbk, lbi, lbn = shard_and_exchange(bk, bn)
+    lpk, lpi, lpn = shard_and_exchange(pk, pn)
+    bi, pi = ora.join_ext(lbk, lpk, 1, bnull=lbn, pnull=lpn)  # left
+    # map local indices back to GLOBAL row ids
+    pairs = [(int(lbi[b]) if b >= 0 else -1, int(lpi[p])) for b, p in
+             zip(bi.tolist(), pi.tolist())]
+    gathered = fragment.allgather_variable(
+        torch.as_tensor(np.array(pairs, dtype=np.int64).reshape(-1)))
+    if rank == 0:
+        allp = gathered.numpy().reshape(-1, 2)
+        got = sorted(map(tuple, allp.tolist()))
+        obi, opi = ora.join_ext(bk, pk, 1, bnull=bn, pnull=pn)
+        exp = sorted(zip(obi.tolist(), opi.tolist()))
+        q.put((got, exp))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_left_join_two_ranks():
+    got, exp = _run(_worker_dist_leftjoin)
+    assert got == exp
