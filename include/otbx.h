@@ -370,6 +370,47 @@ otbx_status otbx_agg_i64x2(const int64_t *k1_dev, const uint8_t *k1null_dev,
                            otbx_agg2_group *groups_dev, int64_t *ngroups_dev,
                            void *stream);
 
+/* ---- N-key (1..8 columns) group-by and join ----
+ * Group/join identity is the ROW of key columns: NULL==NULL for grouping
+ * (execGrouping.c:295,:525), any-NULL-never-matches for joins
+ * (nodeHash.c:2026); hash = iterated rotate-left-1 xor over per-column
+ * hashes (nodeHash.c:2059). Groups carry the DEFINING ROW INDEX instead of
+ * N key values — the reference's hash table stores the representative
+ * tuple the same way (execGrouping.c firstTuple); the caller reads the key
+ * values back through the index. Joins emit (bidx, pidx) pairs exactly as
+ * otbx_join_i64_ext (same join_type codes, fills and overflow contract). */
+#define OTBX_MAX_KEYS 8
+typedef struct {
+    int32_t nkeys;                           /* 1..OTBX_MAX_KEYS */
+    const int64_t *keys[OTBX_MAX_KEYS];      /* device pointers */
+    const uint8_t *nulls[OTBX_MAX_KEYS];     /* per column; may be NULL */
+} otbx_keyset;
+
+typedef struct {
+    int64_t row_idx;          /* defining row (representative tuple) */
+    int64_t count_star;
+    int64_t count_v;
+    double sum_v;
+    int32_t sum_isnull;
+    int32_t _pad;
+} otbx_aggn_group; /* 40 B */
+
+otbx_status otbx_agg_i64n_workspace_bytes(int64_t n, size_t *bytes);
+otbx_status otbx_agg_i64n(const otbx_keyset *ks, const double *vals_dev,
+                          const uint8_t *val_null_dev, int64_t n,
+                          void *ws_dev, size_t ws_bytes,
+                          otbx_aggn_group *groups_dev, int64_t *ngroups_dev,
+                          void *stream);
+
+otbx_status otbx_join_i64n_workspace_bytes(int64_t nb, int64_t np,
+                                           size_t *bytes);
+otbx_status otbx_join_i64n(const otbx_keyset *bks, int64_t nb,
+                           const otbx_keyset *pks, int64_t np,
+                           int32_t join_type, void *ws_dev, size_t ws_bytes,
+                           int64_t *out_bidx_dev, int64_t *out_pidx_dev,
+                           int64_t cap_pairs, int64_t *npairs_dev,
+                           void *stream);
+
 /* ---- exact decimal (scaled-int64) aggregate with int128 sum ----
  * The reference's HAVE_INT128 numeric aggregation: group state =
  * Int128AggState {N, sumX} (utils/adt/numeric.c:5072; do_int128_accum

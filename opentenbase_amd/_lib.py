@@ -57,6 +57,13 @@ class CustomerDev(C.Structure):
     ]
 
 
+class KeysetDev(C.Structure):
+    """otbx_keyset (include/otbx.h): up to 8 key columns (device ptrs)."""
+    _fields_ = [("nkeys", C.c_int32),
+                ("keys", C.c_void_p * 8),
+                ("nulls", C.c_void_p * 8)]
+
+
 class PartDev(C.Structure):
     _fields_ = [
         ("n", C.c_int64),
@@ -120,4 +127,6 @@ EXPORTED_SYMBOLS = [
     "otbx_join_ext_workspace_bytes", "otbx_join_i64_ext", "otbx_join_i64x2",
     "otbx_agg_i64x2_workspace_bytes", "otbx_agg_i64x2",
     "otbx_agg_i64_dec_workspace_bytes", "otbx_agg_i64_dec",
+    "otbx_agg_i64n_workspace_bytes", "otbx_agg_i64n",
+    "otbx_join_i64n_workspace_bytes", "otbx_join_i64n",
 ]

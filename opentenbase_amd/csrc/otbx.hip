@@ -5723,3 +5723,319 @@ otbx_status otbx_build_key32(const int64_t *src, int64_t n, int32_t *dst,
 }
 
 } /* extern "C" */
+
+/* ============ N-key group-by / join (otbx_agg_i64n, otbx_join_i64n) ======
+ * See include/otbx.h for semantics + citations. The otbx_keyset is passed
+ * to kernels BY VALUE (132 B of kernarg — no device-side pointer array).
+ * Slots claim by defining row index (execGrouping.c firstTuple pattern);
+ * identity compares run over the keyset columns. Generality tier. */
+
+__device__ __forceinline__ bool d_nk_rownull(const otbx_keyset &ks, int64_t i)
+{
+    for (int c = 0; c < ks.nkeys; c++)
+        if (ks.nulls[c] && ks.nulls[c][i])
+            return true;
+    return false;
+}
+
+__device__ __forceinline__ uint64_t d_nk_hash(const otbx_keyset &ks,
+                                              int64_t i)
+{
+    uint64_t h = 0;
+    for (int c = 0; c < ks.nkeys; c++) {
+        h = (h << 1) | (h >> 63);
+        bool isnull = ks.nulls[c] && ks.nulls[c][i];
+        h ^= isnull ? (0x9e3779b97f4a7c15ull + (uint64_t)c)
+                    : d_hash_i64(ks.keys[c][i]);
+    }
+    return h;
+}
+
+__device__ __forceinline__ bool d_nk_row_eq(const otbx_keyset &ks, int64_t a,
+                                            int64_t b)
+{
+    for (int c = 0; c < ks.nkeys; c++) {
+        bool na = ks.nulls[c] && ks.nulls[c][a];
+        bool nb = ks.nulls[c] && ks.nulls[c][b];
+        if (na != nb) return false;
+        if (!na && ks.keys[c][a] != ks.keys[c][b]) return false;
+    }
+    return true;
+}
+
+__device__ __forceinline__ bool d_nk_match(const otbx_keyset &bks, int64_t b,
+                                           const otbx_keyset &pks, int64_t p)
+{
+    for (int c = 0; c < bks.nkeys; c++)
+        if (bks.keys[c][b] != pks.keys[c][p])
+            return false;
+    return true;
+}
+
+struct aggn_slot {
+    long long idx; /* claim word: -1 empty, else defining row index */
+    unsigned long long count_star;
+    unsigned long long count_v;
+    double sum;
+};
+
+__global__ void k_aggn_init(aggn_slot *tab, int64_t cap)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < cap;
+         i += stride) {
+        tab[i].idx = -1;
+        tab[i].count_star = 0;
+        tab[i].count_v = 0;
+        tab[i].sum = 0.0;
+    }
+}
+
+__global__ void k_aggn_build(const otbx_keyset ks,
+                             const double *__restrict__ vals,
+                             const uint8_t *__restrict__ vnull, int64_t n,
+                             aggn_slot *tab, int64_t cap)
+{
+    int64_t mask = cap - 1;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        int64_t s = (int64_t)(d_nk_hash(ks, i) & (uint64_t)mask);
+        for (;;) {
+            long long owner = tab[s].idx;
+            if (owner < 0) {
+                long long prev = (long long)atomicCAS(
+                    (unsigned long long *)&tab[s].idx,
+                    (unsigned long long)(-1ll), (unsigned long long)i);
+                owner = prev == -1ll ? i : prev;
+            }
+            if (d_nk_row_eq(ks, owner, i))
+                break;
+            s = (s + 1) & mask;
+        }
+        atomicAdd(&tab[s].count_star, 1ull);
+        if (!(vnull && vnull[i])) {
+            atomicAdd(&tab[s].count_v, 1ull);
+            atomicAdd(&tab[s].sum, vals[i]);
+        }
+    }
+}
+
+__global__ void k_aggn_compact(const aggn_slot *__restrict__ tab, int64_t cap,
+                               otbx_aggn_group *__restrict__ out,
+                               int64_t *ngroups)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         s += stride) {
+        bool used = s < cap && tab[s].idx >= 0;
+        int64_t pos = wave_append(ngroups, used);
+        if (used) {
+            out[pos].row_idx = tab[s].idx;
+            out[pos].count_star = (int64_t)tab[s].count_star;
+            out[pos].count_v = (int64_t)tab[s].count_v;
+            out[pos].sum_v = tab[s].sum;
+            out[pos].sum_isnull = tab[s].count_v == 0;
+            out[pos]._pad = 0;
+        }
+        if (__all(s >= cap))
+            break;
+    }
+}
+
+/* N-key join: same structure as k_joinx_* but identity through keysets */
+__global__ void k_joinn_build(const otbx_keyset ks, int64_t nb,
+                              long long *slot_idx, int64_t cap)
+{
+    int64_t mask = cap - 1;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nb;
+         i += stride) {
+        if (d_nk_rownull(ks, i))
+            continue;
+        int64_t s = (int64_t)(d_nk_hash(ks, i) & (uint64_t)mask);
+        while (atomicCAS((unsigned long long *)&slot_idx[s],
+                         (unsigned long long)(-1ll),
+                         (unsigned long long)i) != (unsigned long long)(-1ll))
+            s = (s + 1) & mask;
+    }
+}
+
+__global__ void k_joinn_probe(const otbx_keyset bks,
+                              const long long *__restrict__ slot_idx,
+                              int64_t cap, const otbx_keyset pks, int64_t np,
+                              int jt,
+                              unsigned long long *__restrict__ mbitmap,
+                              int64_t *__restrict__ out_b,
+                              int64_t *__restrict__ out_p, int64_t cap_pairs,
+                              int64_t *npairs)
+{
+    const bool emit_match = (jt == 0 || jt == 1 || jt == 4 || jt == 5);
+    const bool fill_probe = (jt == 1 || jt == 3 || jt == 5);
+    const bool semi = jt == 2;
+    int64_t mask = cap - 1;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i0 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         i0 += stride) {
+        bool inrange = i0 < np;
+        bool rownull = inrange && d_nk_rownull(pks, i0);
+        bool active = inrange && !rownull;
+        int64_t s = active
+                        ? (int64_t)(d_nk_hash(pks, i0) & (uint64_t)mask)
+                        : 0;
+        bool walking = active;
+        int nmatch = 0;
+        while (__any(walking)) {
+            long long bidx = walking ? slot_idx[s] : -1;
+            bool have = walking && bidx >= 0;
+            bool match = have && d_nk_match(bks, bidx, pks, i0);
+            if (match) {
+                nmatch++;
+                if (mbitmap)
+                    atomicOr(&mbitmap[bidx >> 6], 1ull << (bidx & 63));
+                if (semi && nmatch > 1)
+                    match = false;
+            }
+            bool emit = match && emit_match;
+            int64_t pos = wave_append(npairs, emit);
+            if (emit && pos < cap_pairs) {
+                out_b[pos] = bidx;
+                out_p[pos] = i0;
+            }
+            s = (s + 1) & mask;
+            walking = have;
+        }
+        bool fill = false;
+        if (inrange) {
+            if (fill_probe)
+                fill = rownull || nmatch == 0;
+            else if (semi)
+                fill = nmatch > 0;
+        }
+        int64_t pos = wave_append(npairs, fill);
+        if (fill && pos < cap_pairs) {
+            out_b[pos] = -1;
+            out_p[pos] = i0;
+        }
+        if (__all(i0 >= np))
+            break;
+    }
+}
+
+__global__ void k_joinn_fill_build(const otbx_keyset ks, int64_t nb,
+                                   const unsigned long long *__restrict__ mbitmap,
+                                   int64_t *__restrict__ out_b,
+                                   int64_t *__restrict__ out_p,
+                                   int64_t cap_pairs, int64_t *npairs)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i0 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         i0 += stride) {
+        bool fill = false;
+        if (i0 < nb)
+            fill = d_nk_rownull(ks, i0) ||
+                   !((mbitmap[i0 >> 6] >> (i0 & 63)) & 1ull);
+        int64_t pos = wave_append(npairs, fill);
+        if (fill && pos < cap_pairs) {
+            out_b[pos] = i0;
+            out_p[pos] = -1;
+        }
+        if (__all(i0 >= nb))
+            break;
+    }
+}
+
+extern "C" {
+
+static bool keyset_ok(const otbx_keyset *ks)
+{
+    if (!ks || ks->nkeys < 1 || ks->nkeys > OTBX_MAX_KEYS)
+        return false;
+    for (int c = 0; c < ks->nkeys; c++)
+        if (!ks->keys[c])
+            return false;
+    return true;
+}
+
+otbx_status otbx_agg_i64n_workspace_bytes(int64_t n, size_t *bytes)
+{
+    *bytes = (size_t)jx_cap_for(n) * sizeof(aggn_slot);
+    return OTBX_OK;
+}
+
+otbx_status otbx_agg_i64n(const otbx_keyset *ks, const double *vals,
+                          const uint8_t *vnull, int64_t n, void *ws,
+                          size_t ws_bytes, otbx_aggn_group *groups_dev,
+                          int64_t *ngroups_dev, void *stream)
+{
+    if (!keyset_ok(ks) || !ws || !ngroups_dev || !vals)
+        return OTBX_ERR_INVALID;
+    hipStream_t s = (hipStream_t)stream;
+    int64_t cap = jx_cap_for(n);
+    size_t need;
+    otbx_agg_i64n_workspace_bytes(n, &need);
+    if (ws_bytes < need)
+        return OTBX_ERR_INVALID;
+    aggn_slot *tab = (aggn_slot *)ws;
+    HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
+    hipLaunchKernelGGL(k_aggn_init, dim3(grid_for(cap, 256)), dim3(256), 0, s,
+                       tab, cap);
+    if (n > 0)
+        hipLaunchKernelGGL(k_aggn_build, dim3(grid_for(n, 256)), dim3(256), 0,
+                           s, *ks, vals, vnull, n, tab, cap);
+    hipLaunchKernelGGL(k_aggn_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
+                       s, tab, cap, groups_dev, ngroups_dev);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+otbx_status otbx_join_i64n_workspace_bytes(int64_t nb, int64_t np,
+                                           size_t *bytes)
+{
+    (void)np;
+    *bytes = (size_t)jx_cap_for(nb) * 8 + (size_t)((nb + 63) / 64 + 1) * 8;
+    return OTBX_OK;
+}
+
+otbx_status otbx_join_i64n(const otbx_keyset *bks, int64_t nb,
+                           const otbx_keyset *pks, int64_t np,
+                           int32_t join_type, void *ws, size_t ws_bytes,
+                           int64_t *out_b, int64_t *out_p, int64_t cap_pairs,
+                           int64_t *npairs_dev, void *stream)
+{
+    if (!keyset_ok(bks) || !keyset_ok(pks) || bks->nkeys != pks->nkeys ||
+        join_type < 0 || join_type > 5 || !ws || !npairs_dev)
+        return OTBX_ERR_INVALID;
+    hipStream_t s = (hipStream_t)stream;
+    int64_t cap = jx_cap_for(nb);
+    size_t need;
+    otbx_join_i64n_workspace_bytes(nb, np, &need);
+    if (ws_bytes < need)
+        return OTBX_ERR_INVALID;
+    long long *slot_idx = (long long *)ws;
+    unsigned long long *mbitmap =
+        (unsigned long long *)((char *)ws + (size_t)cap * 8);
+    const bool fill_build = (join_type == 4 || join_type == 5);
+    HIP_CHECK(hipMemsetAsync(npairs_dev, 0, 8, s));
+    hipLaunchKernelGGL(k_fill_i64, dim3(grid_for(cap, 256)), dim3(256), 0, s,
+                       (int64_t *)slot_idx, cap, -1ll);
+    if (fill_build)
+        HIP_CHECK(hipMemsetAsync(mbitmap, 0,
+                                 (size_t)((nb + 63) / 64 + 1) * 8, s));
+    if (nb > 0)
+        hipLaunchKernelGGL(k_joinn_build, dim3(grid_for(nb, 256)), dim3(256),
+                           0, s, *bks, nb, slot_idx, cap);
+    if (np > 0)
+        hipLaunchKernelGGL(k_joinn_probe, dim3(grid_for(np, 256)), dim3(256),
+                           0, s, *bks, slot_idx, cap, *pks, np, join_type,
+                           fill_build ? mbitmap : NULL, out_b, out_p,
+                           cap_pairs, npairs_dev);
+    if (fill_build && nb > 0)
+        hipLaunchKernelGGL(k_joinn_fill_build, dim3(grid_for(nb, 256)),
+                           dim3(256), 0, s, *bks, nb, mbitmap, out_b, out_p,
+                           cap_pairs, npairs_dev);
+    HIP_CHECK(hipGetLastError());
+    return OTBX_OK;
+}
+
+} /* extern "C" */

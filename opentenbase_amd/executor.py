@@ -15,8 +15,8 @@ import ctypes as C
 
 import torch
 
-from ._lib import (CustomerDev, LineitemDev, OrdersDev, OtbxError, PartDev,
-                   call, check, lib)
+from ._lib import (CustomerDev, KeysetDev, LineitemDev, OrdersDev,
+                   OtbxError, PartDev, call, check, lib)
 
 Q1_SLOT_ORDER = [(b"A", b"F"), (b"A", b"O"), (b"N", b"F"),
                  (b"N", b"O"), (b"R", b"F"), (b"R", b"O")]
@@ -612,6 +612,94 @@ class GpuHashAggDec(CustomScanState):
             d["sum128"] = (int(r["sum_hi"]) << 64) | int(r["sum_lo"])
             rows.append(d)
         return rows
+
+
+def _keyset_dev(key_tensors, null_tensors=None):
+    """Build an otbx_keyset (device pointers) from torch tensors."""
+    ks = KeysetDev()
+    ks.nkeys = len(key_tensors)
+    for c, k in enumerate(key_tensors):
+        ks.keys[c] = k.data_ptr()
+        nt = None if null_tensors is None else null_tensors[c]
+        ks.nulls[c] = nt.data_ptr() if nt is not None else None
+    return ks
+
+
+class GpuHashAggN(CustomScanState):
+    """N-key (1..8) composable HashAggregate: groups carry the defining
+    ROW INDEX (the representative-tuple pattern, execGrouping.c
+    firstTuple); the caller reads key values back through the index."""
+
+    def __init__(self, key_tensors, vals, null_tensors=None, val_null=None):
+        super().__init__()
+        self.keys, self.vals = key_tensors, vals
+        self.nulls, self.vn = null_tensors, val_null
+
+    def _run(self):
+        import numpy as np
+        L = lib()
+        n = len(self.vals)
+        ks = _keyset_dev(self.keys, self.nulls)
+        ws_bytes = C.c_size_t(0)
+        check(L.otbx_agg_i64n_workspace_bytes(C.c_int64(n),
+                                              C.byref(ws_bytes)))
+        ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8,
+                         device="cuda")
+        out = torch.empty(max(n, 1) * 40, dtype=torch.uint8, device="cuda")
+        ng = torch.zeros(1, dtype=torch.int64, device="cuda")
+        vn = C.c_void_p(self.vn.data_ptr()) if self.vn is not None else None
+        call("otbx_agg_i64n", C.byref(ks),
+             C.c_void_p(self.vals.data_ptr()), vn, C.c_int64(n),
+             C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+             C.c_void_p(out.data_ptr()), C.c_void_p(ng.data_ptr()), _stream())
+        ngroups = int(ng.cpu().item())
+        dt = np.dtype([("row_idx", "i8"), ("count_star", "i8"),
+                       ("count_v", "i8"), ("sum_v", "f8"),
+                       ("sum_isnull", "i4"), ("_pad", "i4")])
+        arr = out[: ngroups * 40].cpu().numpy().view(dt).copy()
+        arr.sort(order="row_idx")
+        return list(arr)
+
+
+class GpuHashJoinN(CustomScanState):
+    """N-key (1..8) HashJoin, all six join types; pair encoding and
+    overflow contract as GpuHashJoin."""
+
+    def __init__(self, bkeys, pkeys, join_type=0, bnulls=None, pnulls=None,
+                 cap_pairs=None):
+        super().__init__()
+        self.bkeys, self.pkeys = bkeys, pkeys
+        self.bnulls, self.pnulls = bnulls, pnulls
+        self.join_type = JOIN_TYPES.get(join_type, join_type) \
+            if isinstance(join_type, str) else join_type
+        self.cap_pairs = cap_pairs
+
+    def _run(self):
+        L = lib()
+        nb = len(self.bkeys[0]) if self.bkeys else 0
+        npr = len(self.pkeys[0]) if self.pkeys else 0
+        bks = _keyset_dev(self.bkeys, self.bnulls)
+        pks = _keyset_dev(self.pkeys, self.pnulls)
+        ws_bytes = C.c_size_t(0)
+        check(L.otbx_join_i64n_workspace_bytes(C.c_int64(nb), C.c_int64(npr),
+                                               C.byref(ws_bytes)))
+        ws = torch.empty(max(ws_bytes.value, 1), dtype=torch.uint8,
+                         device="cuda")
+        cap = self.cap_pairs if self.cap_pairs \
+            else max(4 * max(nb, npr) + nb + npr, 64)
+        ob = torch.empty(cap, dtype=torch.int64, device="cuda")
+        op = torch.empty(cap, dtype=torch.int64, device="cuda")
+        npairs = torch.zeros(1, dtype=torch.int64, device="cuda")
+        call("otbx_join_i64n", C.byref(bks), C.c_int64(nb), C.byref(pks),
+             C.c_int64(npr), C.c_int32(self.join_type),
+             C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+             C.c_void_p(ob.data_ptr()), C.c_void_p(op.data_ptr()),
+             C.c_int64(cap), C.c_void_p(npairs.data_ptr()), _stream())
+        n = int(npairs.cpu().item())
+        if n > cap:
+            raise OtbxError(3, f"join pair overflow: {n} > cap {cap}")
+        return list(zip(ob[:n].cpu().numpy().tolist(),
+                        op[:n].cpu().numpy().tolist()))
 
 
 JOIN_TYPES = {"inner": 0, "left": 1, "semi": 2, "anti": 3, "right": 4,

@@ -1007,3 +1007,179 @@ fail:
     free(slots); free(used);
     return st;
 }
+
+/* ============== N-key group-by / join (ora_agg_i64n, ora_join_i64n) ======
+ * See oracle.h for semantics + citations. The defining-row-index output is
+ * the representative-tuple pattern (execGrouping.c firstTuple). */
+
+static inline int nk_rownull(const ora_keyset *ks, int64_t i)
+{
+    for (int c = 0; c < ks->nkeys; c++)
+        if (ks->nulls[c] && ks->nulls[c][i])
+            return 1;
+    return 0;
+}
+
+static inline uint64_t nk_hash(const ora_keyset *ks, int64_t i)
+{
+    uint64_t h = 0;
+    for (int c = 0; c < ks->nkeys; c++) {
+        h = (h << 1) | (h >> 63);       /* rotate-left-1, nodeHash.c:2059 */
+        int isnull = ks->nulls[c] && ks->nulls[c][i];
+        h ^= isnull ? (0x9e3779b97f4a7c15ull + (uint64_t)c)
+                    : ora_hash_i64(ks->keys[c][i]);
+    }
+    return h;
+}
+
+/* group identity: rows a and b agree on every column's (isnull, value) */
+static inline int nk_row_eq(const ora_keyset *ks, int64_t a, int64_t b)
+{
+    for (int c = 0; c < ks->nkeys; c++) {
+        int na = ks->nulls[c] && ks->nulls[c][a];
+        int nb = ks->nulls[c] && ks->nulls[c][b];
+        if (na != nb) return 0;
+        if (!na && ks->keys[c][a] != ks->keys[c][b]) return 0;
+    }
+    return 1;
+}
+
+/* join match: no column NULL on either side, values equal */
+static inline int nk_match(const ora_keyset *bks, int64_t b,
+                           const ora_keyset *pks, int64_t p)
+{
+    for (int c = 0; c < bks->nkeys; c++)
+        if (bks->keys[c][b] != pks->keys[c][p])
+            return 0;
+    return 1;
+}
+
+static int aggn_cmp(const void *pa, const void *pb)
+{
+    const ora_aggn_group *a = pa, *b = pb;
+    return a->row_idx < b->row_idx ? -1 : a->row_idx > b->row_idx ? 1 : 0;
+}
+
+ora_status ora_agg_i64n(const ora_keyset *ks, const double *vals,
+                        const uint8_t *val_null, int64_t n,
+                        ora_aggn_group **out, int64_t *ngroups)
+{
+    if (!ks || ks->nkeys < 1 || ks->nkeys > ORA_MAX_KEYS)
+        return ORA_ERR_INVALID;
+    int64_t cap = next_pow2(n < 16 ? 16 : (int64_t)((double)n / 0.85) + 1);
+    int64_t *slot_row = malloc((size_t)cap * 8);       /* -1 empty */
+    ora_aggn_group *slots = calloc(cap, sizeof(ora_aggn_group));
+    if (!slot_row || !slots) { free(slot_row); free(slots); return ORA_ERR_OOM; }
+    for (int64_t s = 0; s < cap; s++) slot_row[s] = -1;
+    int64_t ng = 0;
+    ora_status st;
+
+    for (int64_t i = 0; i < n; i++) {
+        int64_t s = (int64_t)(nk_hash(ks, i) & (uint64_t)(cap - 1));
+        for (;;) {
+            if (slot_row[s] < 0) {
+                slot_row[s] = i;
+                slots[s].row_idx = i;
+                slots[s].sum_isnull = 1;
+                ng++;
+                break;
+            }
+            if (nk_row_eq(ks, slot_row[s], i))
+                break;
+            s = (s + 1) & (cap - 1);
+        }
+        ora_aggn_group *e = &slots[s];
+        if ((st = int8inc(&e->count_star))) goto fail;
+        if (!(val_null && val_null[i])) {
+            if ((st = int8inc(&e->count_v))) goto fail;
+            if (e->sum_isnull) {
+                e->sum_v = vals[i];
+                e->sum_isnull = 0;
+            } else if ((st = float8pl(e->sum_v, vals[i], &e->sum_v)))
+                goto fail;
+        }
+    }
+
+    ora_aggn_group *res = malloc((size_t)(ng > 0 ? ng : 1) *
+                                 sizeof(ora_aggn_group));
+    if (!res) { st = ORA_ERR_OOM; goto fail; }
+    int64_t j = 0;
+    for (int64_t s = 0; s < cap; s++)
+        if (slot_row[s] >= 0) res[j++] = slots[s];
+    qsort(res, (size_t)ng, sizeof(ora_aggn_group), aggn_cmp);
+    free(slot_row); free(slots);
+    *out = res; *ngroups = ng;
+    return ORA_OK;
+fail:
+    free(slot_row); free(slots);
+    return st;
+}
+
+ora_status ora_join_i64n(const ora_keyset *bks, int64_t nb,
+                         const ora_keyset *pks, int64_t np, int join_type,
+                         int64_t **out_bidx, int64_t **out_pidx,
+                         int64_t *nout)
+{
+    if (!bks || !pks || bks->nkeys != pks->nkeys || bks->nkeys < 1 ||
+        bks->nkeys > ORA_MAX_KEYS || join_type < 0 || join_type > 5)
+        return ORA_ERR_INVALID;
+    const int jt = join_type;
+    const int fill_probe = (jt == 1 || jt == 5);
+    const int fill_build = (jt == 4 || jt == 5);
+    const int emit_match = (jt == 0 || jt == 1 || jt == 4 || jt == 5);
+
+    int64_t nbuckets = next_pow2(nb < 16 ? 16 : nb);
+    int64_t *head = malloc((size_t)nbuckets * 8);
+    int64_t *next = malloc((size_t)(nb > 0 ? nb : 1) * 8);
+    uint8_t *matched = fill_build ? calloc((size_t)(nb > 0 ? nb : 1), 1)
+                                  : NULL;
+    if (!head || !next || (fill_build && !matched)) {
+        free(head); free(next); free(matched);
+        return ORA_ERR_OOM;
+    }
+    for (int64_t i = 0; i < nbuckets; i++) head[i] = -1;
+    for (int64_t i = 0; i < nb; i++) {
+        if (nk_rownull(bks, i)) continue;
+        int64_t b = (int64_t)(nk_hash(bks, i) & (uint64_t)(nbuckets - 1));
+        next[i] = head[b];
+        head[b] = i;
+    }
+
+    int64_t cap = 64, n = 0;
+    int64_t *bi = malloc(cap * 8), *pi = malloc(cap * 8);
+    ora_status st = ORA_OK;
+    if (!bi || !pi) { st = ORA_ERR_OOM; goto done; }
+
+    for (int64_t p = 0; p < np; p++) {
+        if (nk_rownull(pks, p)) {
+            if (fill_probe || jt == 3)
+                if ((st = ext_push(&bi, &pi, &n, &cap, -1, p))) goto done;
+            continue;
+        }
+        int64_t bkt = (int64_t)(nk_hash(pks, p) & (uint64_t)(nbuckets - 1));
+        int64_t nmatch = 0;
+        for (int64_t m = head[bkt]; m >= 0; m = next[m]) {
+            if (nk_rownull(bks, m) || !nk_match(bks, m, pks, p)) continue;
+            nmatch++;
+            if (matched) matched[m] = 1;
+            if (emit_match)
+                if ((st = ext_push(&bi, &pi, &n, &cap, m, p))) goto done;
+            if (jt == 2) break;
+        }
+        if (jt == 2 && nmatch > 0) {
+            if ((st = ext_push(&bi, &pi, &n, &cap, -1, p))) goto done;
+        } else if ((jt == 3 || fill_probe) && nmatch == 0) {
+            if ((st = ext_push(&bi, &pi, &n, &cap, -1, p))) goto done;
+        }
+    }
+    if (fill_build)
+        for (int64_t i = 0; i < nb; i++)
+            if (nk_rownull(bks, i) || !matched[i])
+                if ((st = ext_push(&bi, &pi, &n, &cap, i, -1))) goto done;
+
+done:
+    free(head); free(next); free(matched);
+    if (st) { free(bi); free(pi); return st; }
+    *out_bidx = bi; *out_pidx = pi; *nout = n;
+    return ORA_OK;
+}

@@ -229,6 +229,41 @@ ora_status ora_agg_i64_dec(const int64_t *keys, const uint8_t *key_null,
                            const int64_t *vals, const uint8_t *val_null,
                            int64_t n, ora_dec_group **out, int64_t *ngroups);
 
+
+/* ---- N-key (1..8 columns) group-by and join ----
+ * Group/join identity is the ROW of key columns with NULL==NULL for
+ * grouping (execGrouping.c:295,:525) and any-NULL-never-matches for joins
+ * (nodeHash.c:2026); hash = iterated rotate-left-1 xor over per-column
+ * hashes (ExecHashGetHashValue multi-key combine, nodeHash.c:2059).
+ * Groups carry the DEFINING ROW INDEX instead of N key values — the
+ * reference's hash table likewise stores the representative tuple
+ * (execGrouping.c firstTuple); the caller reads the keys back by index. */
+#define ORA_MAX_KEYS 8
+typedef struct {
+    int nkeys;
+    const int64_t *keys[ORA_MAX_KEYS];
+    const uint8_t *nulls[ORA_MAX_KEYS];   /* per column; entries may be NULL */
+} ora_keyset;
+
+typedef struct {
+    int64_t row_idx;          /* defining row (representative tuple) */
+    int64_t count_star;
+    int64_t count_v;
+    double sum_v;
+    int sum_isnull;
+} ora_aggn_group;
+
+/* emitted sorted by row_idx (deterministic; group identity is the row) */
+ora_status ora_agg_i64n(const ora_keyset *ks, const double *vals,
+                        const uint8_t *val_null, int64_t n,
+                        ora_aggn_group **out, int64_t *ngroups);
+
+/* join_type as ora_join_ext (0..5) */
+ora_status ora_join_i64n(const ora_keyset *bks, int64_t nb,
+                         const ora_keyset *pks, int64_t np, int join_type,
+                         int64_t **out_bidx, int64_t **out_pidx,
+                         int64_t *nout);
+
 #ifdef __cplusplus
 }
 #endif

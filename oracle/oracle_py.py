@@ -110,6 +110,8 @@ def load():
     lib.ora_join_ext.restype = C.c_int
     lib.ora_agg_i64x2.restype = C.c_int
     lib.ora_agg_i64_dec.restype = C.c_int
+    lib.ora_agg_i64n.restype = C.c_int
+    lib.ora_join_i64n.restype = C.c_int
     lib.ora_q1_combine.restype = C.c_int
     return lib
 
@@ -388,3 +390,68 @@ def agg_i64_dec(keys, vals, key_null=None, val_null=None):
                            C.c_int64(len(keys)), C.byref(out), C.byref(ng))
     assert st == 0, st
     return [out[i] for i in range(ng.value)]
+
+
+class OraKeyset(C.Structure):
+    _fields_ = [("nkeys", C.c_int),
+                ("keys", C.POINTER(C.c_int64) * 8),
+                ("nulls", C.POINTER(C.c_uint8) * 8)]
+
+
+class OraAggNGroup(C.Structure):
+    _fields_ = [("row_idx", C.c_int64), ("count_star", C.c_int64),
+                ("count_v", C.c_int64), ("sum_v", C.c_double),
+                ("sum_isnull", C.c_int)]
+
+
+def _keyset(key_cols, null_cols):
+    """key_cols: list of int64 arrays; null_cols: list (entries may be
+    None). Returns (OraKeyset, keepalive-list)."""
+    ks = OraKeyset()
+    ks.nkeys = len(key_cols)
+    keep = []
+    for c, k in enumerate(key_cols):
+        a = np.ascontiguousarray(k, dtype=np.int64)
+        keep.append(a)
+        ks.keys[c] = a.ctypes.data_as(C.POINTER(C.c_int64))
+        nc = None if null_cols is None else null_cols[c]
+        if nc is not None:
+            na = np.ascontiguousarray(nc, dtype=np.uint8)
+            keep.append(na)
+            ks.nulls[c] = na.ctypes.data_as(C.POINTER(C.c_uint8))
+    return ks, keep
+
+
+def agg_i64n(key_cols, vals, null_cols=None, val_null=None):
+    """N-key group-by (oracle.h ora_agg_i64n): groups carry the defining
+    ROW INDEX (representative tuple); sorted by row_idx."""
+    L = lib()
+    ks, keep = _keyset(key_cols, null_cols)
+    vals = np.ascontiguousarray(vals, dtype=np.float64)
+    vn = None if val_null is None else np.ascontiguousarray(val_null, np.uint8)
+    out = C.POINTER(OraAggNGroup)()
+    ng = C.c_int64(0)
+    st = L.ora_agg_i64n(C.byref(ks), _p(vals, C.c_double), _p(vn, C.c_uint8),
+                        C.c_int64(len(vals)), C.byref(out), C.byref(ng))
+    assert st == 0, st
+    return [out[i] for i in range(ng.value)]
+
+
+def join_i64n(bkey_cols, pkey_cols, join_type, bnull_cols=None,
+              pnull_cols=None):
+    L = lib()
+    bks, keepb = _keyset(bkey_cols, bnull_cols)
+    pks, keepp = _keyset(pkey_cols, pnull_cols)
+    nb = len(bkey_cols[0]) if len(bkey_cols) else 0
+    npr = len(pkey_cols[0]) if len(pkey_cols) else 0
+    ob = C.POINTER(C.c_int64)()
+    op = C.POINTER(C.c_int64)()
+    n = C.c_int64(0)
+    st = L.ora_join_i64n(C.byref(bks), C.c_int64(nb), C.byref(pks),
+                         C.c_int64(npr), C.c_int(join_type),
+                         C.byref(ob), C.byref(op), C.byref(n))
+    assert st == 0, st
+    nn = n.value
+    bi = np.ctypeslib.as_array(ob, (nn,)).copy() if nn else np.empty(0, np.int64)
+    pi = np.ctypeslib.as_array(op, (nn,)).copy() if nn else np.empty(0, np.int64)
+    return bi, pi

@@ -1195,3 +1195,64 @@ def test_build_key32_overflow_detect(ex):
     assert _build_key32(bad) is None
     neg = torch.tensor([1, -2, 3], dtype=torch.int64, device="cuda")
     assert _build_key32(neg) is None
+
+
+# ---------------- N-key operators ----------------
+
+@pytest.mark.parametrize("nk", [1, 3, 8])
+def test_aggn_parity_gpu(ex, ora, nk):
+    """N-key group-by GPU vs oracle: same group count and, keyed by each
+    group's identity tuple, identical counts and 1e-9-close sums; the
+    defining row must belong to its own group."""
+    rng = np.random.default_rng(60 + nk)
+    n = 200000
+    keys = [rng.integers(-3, 6, n) for _ in range(nk)]
+    nulls = [(rng.random(n) < 0.1).astype(np.uint8) if c == 0 else None
+             for c in range(nk)]
+    v = rng.standard_normal(n)
+    dev = lambda a, dt: torch.as_tensor(a, dtype=dt, device="cuda")  # noqa: E731
+    node = ex.GpuHashAggN([dev(k, torch.int64) for k in keys],
+                          dev(v, torch.float64),
+                          null_tensors=[dev(nc, torch.uint8)
+                                        if nc is not None else None
+                                        for nc in nulls])
+    got = drain(node)
+    exp = ora.agg_i64n(keys, v, null_cols=nulls)
+
+    def ident(i):
+        out = []
+        for c in range(nk):
+            isnull = nulls[c] is not None and nulls[c][i]
+            out.append((bool(isnull), 0 if isnull else int(keys[c][i])))
+        return tuple(out)
+
+    assert len(got) == len(exp)
+    em = {ident(o.row_idx): o for o in exp}
+    for g in got:
+        o = em[ident(int(g["row_idx"]))]
+        assert int(g["count_star"]) == o.count_star
+        assert int(g["count_v"]) == o.count_v
+        if o.sum_isnull:
+            assert g["sum_isnull"]
+        else:
+            assert abs(float(g["sum_v"]) - o.sum_v) <= \
+                1e-9 * max(abs(o.sum_v), 1.0)
+
+
+@pytest.mark.parametrize("jt", ["inner", "left", "semi", "anti", "full"])
+def test_joinn_parity_gpu(ex, ora, jt):
+    rng = np.random.default_rng(80)
+    nb, npr = 20000, 50000
+    bkeys = [rng.integers(0, 40, nb) for _ in range(3)]
+    pkeys = [rng.integers(0, 40, npr) for _ in range(3)]
+    bnulls = [(rng.random(nb) < 0.05).astype(np.uint8), None, None]
+    dev = lambda a: torch.as_tensor(a, dtype=torch.int64, device="cuda")  # noqa: E731
+    devn = lambda a: (torch.as_tensor(a, device="cuda")  # noqa: E731
+                      if a is not None else None)
+    node = ex.GpuHashJoinN([dev(k) for k in bkeys], [dev(k) for k in pkeys],
+                           join_type=jt,
+                           bnulls=[devn(x) for x in bnulls])
+    pairs = drain(node)
+    obi, opi = ora.join_i64n(bkeys, pkeys, ex.JOIN_TYPES[jt],
+                             bnull_cols=bnulls)
+    assert sorted(pairs) == sorted(zip(obi.tolist(), opi.tolist()))

@@ -248,3 +248,115 @@ def test_agg_dec_empty_allnull():
     g = ora.agg_i64_dec(k, v, val_null=np.ones(4, dtype=np.uint8))
     assert len(g) == 1 and g[0].sum_isnull and g[0].count_star == 4 \
         and g[0].count_v == 0
+
+
+# ---------------- N-key (1..8) group-by / join ----------------
+
+def brute_aggn(key_cols, vals, null_cols=None, vn=None):
+    groups = {}
+    nk = len(key_cols)
+    for i in range(len(vals)):
+        ident = []
+        for c in range(nk):
+            isnull = null_cols is not None and null_cols[c] is not None \
+                and null_cols[c][i]
+            ident.append((bool(isnull), 0 if isnull else int(key_cols[c][i])))
+        a = tuple(ident)
+        g = groups.setdefault(a, {"cs": 0, "cv": 0, "sum": None, "first": i})
+        g["cs"] += 1
+        if vn is None or not vn[i]:
+            g["cv"] += 1
+            g["sum"] = (g["sum"] or 0.0) + float(vals[i])
+    return groups
+
+
+def _ident_of(key_cols, null_cols, i):
+    out = []
+    for c in range(len(key_cols)):
+        isnull = null_cols is not None and null_cols[c] is not None \
+            and null_cols[c][i]
+        out.append((bool(isnull), 0 if isnull else int(key_cols[c][i])))
+    return tuple(out)
+
+
+@pytest.mark.parametrize("nk", [1, 3, 4, 8])
+def test_aggn_random(nk):
+    rng = np.random.default_rng(50 + nk)
+    n = 300
+    keys = [rng.integers(-2, 3, n) for _ in range(nk)]
+    nulls = [(rng.random(n) < 0.15).astype(np.uint8) if c % 2 == 0 else None
+             for c in range(nk)]
+    v = rng.standard_normal(n)
+    vn = (rng.random(n) < 0.2).astype(np.uint8)
+    got = ora.agg_i64n(keys, v, null_cols=nulls, val_null=vn)
+    exp = brute_aggn(keys, v, nulls, vn)
+    assert len(got) == len(exp)
+    for g in got:
+        ident = _ident_of(keys, nulls, g.row_idx)
+        e = exp[ident]
+        assert g.count_star == e["cs"] and g.count_v == e["cv"]
+        # the defining row must be a member of its own group
+        assert _ident_of(keys, nulls, e["first"]) == ident
+        if e["sum"] is None:
+            assert g.sum_isnull
+        else:
+            assert abs(g.sum_v - e["sum"]) <= 1e-9 * max(abs(e["sum"]), 1)
+
+
+@pytest.mark.parametrize("jt", list(JT.values()))
+def test_joinn_random_3key(jt):
+    rng = np.random.default_rng(70 + jt)
+    nb, npr = 120, 260
+    bkeys = [rng.integers(0, 4, nb) for _ in range(3)]
+    pkeys = [rng.integers(0, 4, npr) for _ in range(3)]
+    bnulls = [None, (rng.random(nb) < 0.1).astype(np.uint8), None]
+    pnulls = [(rng.random(npr) < 0.1).astype(np.uint8), None, None]
+    bi, pi = ora.join_i64n(bkeys, pkeys, jt, bnull_cols=bnulls,
+                           pnull_cols=pnulls)
+    got = sorted(zip(bi.tolist(), pi.tolist()))
+
+    def bnull(i):
+        return any(nc is not None and nc[i] for nc in bnulls)
+
+    def pnull(i):
+        return any(nc is not None and nc[i] for nc in pnulls)
+
+    pairs, matched = [], set()
+    for p in range(npr):
+        if pnull(p):
+            if jt in (1, 3, 5):
+                pairs.append((-1, p))
+            continue
+        ms = [b for b in range(nb) if not bnull(b) and
+              all(bkeys[c][b] == pkeys[c][p] for c in range(3))]
+        matched.update(ms)
+        if jt in (0, 1, 4, 5):
+            pairs.extend((b, p) for b in ms)
+        if jt == 2 and ms:
+            pairs.append((-1, p))
+        if jt in (1, 3, 5) and not ms:
+            pairs.append((-1, p))
+    if jt in (4, 5):
+        for b in range(nb):
+            if bnull(b) or b not in matched:
+                pairs.append((b, -1))
+    assert got == sorted(pairs)
+
+
+def test_aggn_matches_single_key_path():
+    """nkeys=1 N-key agg ≡ ora_agg_i64 on the same data."""
+    rng = np.random.default_rng(90)
+    n = 500
+    k = rng.integers(-5, 5, n)
+    v = rng.standard_normal(n)
+    kn = (rng.random(n) < 0.1).astype(np.uint8)
+    gn = ora.agg_i64n([k], v, null_cols=[kn])
+    g1 = ora.agg_i64(k, v, key_null=kn)
+    assert len(gn) == len(g1)
+    m = {}
+    for g in gn:
+        isnull = bool(kn[g.row_idx])
+        m[(isnull, 0 if isnull else int(k[g.row_idx]))] = g
+    for o in g1:
+        g = m[(bool(o.key_isnull), 0 if o.key_isnull else int(o.key))]
+        assert g.count_star == o.count_star and g.count_v == o.count_v
