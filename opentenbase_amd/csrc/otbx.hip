@@ -3821,18 +3821,28 @@ __global__ void k_ord_filter_insert_fused(
         int4 pr = op4[q];
         int32_t ds[4] = {d.x, d.y, d.z, d.w};
         int32_t prio[4] = {pr.x, pr.y, pr.z, pr.w};
+        /* branchless filter phase first: the 4 customer-bitmap gathers
+         * issue independently (the early-continue form serialized them
+         * behind each j's control flow) */
+        bool m[4];
+        if (cbitmap) {
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int64_t cidx = ck[j] - cmin;
+                bool in = (ds[j] < q3date) /* ExecQual on orders */ &&
+                          cidx >= 0 && cidx < crange;
+                unsigned long long w = cbitmap[in ? cidx >> 6 : 0];
+                m[j] = in && ((w >> (cidx & 63)) & 1ull);
+            }
+        } else {
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                m[j] = (ds[j] < q3date) &&
+                       d_keyset_probe(ckeys, ccap, ck[j]);
+        }
 #pragma unroll
         for (int j = 0; j < 4; j++) {
-            bool pass = ds[j] < q3date; /* ExecQual on orders */
-            if (!pass) continue;
-            if (cbitmap) {
-                int64_t cidx = ck[j] - cmin;
-                pass = cidx >= 0 && cidx < crange &&
-                       ((cbitmap[cidx >> 6] >> (cidx & 63)) & 1ull);
-            } else {
-                pass = d_keyset_probe(ckeys, ccap, ck[j]);
-            }
-            if (!pass) continue;
+            if (!m[j]) continue;
             int64_t idx = ok[j] - mino;
             if (idx < 0 || idx >= range) continue;
             atomicOr(&bitmap[idx >> 6], 1ull << (idx & 63));
