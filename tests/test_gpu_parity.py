@@ -1256,3 +1256,30 @@ def test_joinn_parity_gpu(ex, ora, jt):
     obi, opi = ora.join_i64n(bkeys, pkeys, ex.JOIN_TYPES[jt],
                              bnull_cols=bnulls)
     assert sorted(pairs) == sorted(zip(obi.tolist(), opi.tolist()))
+
+
+def test_finish_releases_scratch_and_reinit(ex):
+    """otbx_finish releases the per-process cached scratch and nulls the
+    owning statics: init -> ops that allocate lazy scratch
+    (partition, Q3 with its pinned readbacks) -> finish -> re-init ->
+    the same ops run correctly again on fresh scratch."""
+    from opentenbase_amd._lib import call
+
+    def work():
+        keys = torch.arange(100000, dtype=torch.int64, device="cuda") % 7
+        perm, counts = ex.partition_by_key(keys)
+        assert sum(counts) == 100000
+        n = 200000
+        li = ex.GpuLineitem.generate(n)
+        od = ex.GpuOrders.generate(n // 4, n // 40)
+        cu = ex.GpuCustomer.generate(n // 40)
+        node = ex.GpuQ3Fragment(cu, od, li)
+        node.BeginCustomScan()
+        node._run()
+        return node.ngroups, counts
+
+    a = work()
+    call("otbx_finish")
+    ex.init_device(0)
+    b = work()
+    assert a == b
