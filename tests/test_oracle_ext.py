@@ -360,3 +360,46 @@ def test_aggn_matches_single_key_path():
     for o in g1:
         g = m[(bool(o.key_isnull), 0 if o.key_isnull else int(o.key))]
         assert g.count_star == o.count_star and g.count_v == o.count_v
+
+
+# ---------------- cross-type join invariants ----------------
+
+def _pairs(bk, pk, jt, **kw):
+    bi, pi = ora.join_ext(bk, pk, jt, **kw)
+    return sorted(zip(bi.tolist(), pi.tolist()))
+
+
+def test_join_type_algebra():
+    """Relational identities across the six types (random inputs):
+    left = inner + probe-fills; right = inner + build-fills;
+    full = inner + probe-fills + build-fills;
+    semi and anti partition the non-NULL probe rows;
+    anti fills + semi fills = left fills' NULL-extended probe set."""
+    rng = np.random.default_rng(99)
+    for trial in range(6):
+        nb, npr = rng.integers(0, 120, 2)
+        bk = rng.integers(-4, 10, nb)
+        pk = rng.integers(-4, 10, npr)
+        bn = (rng.random(nb) < 0.15).astype(np.uint8)
+        pn = (rng.random(npr) < 0.15).astype(np.uint8)
+        kw = dict(bnull=bn, pnull=pn)
+        inner = _pairs(bk, pk, JT["inner"], **kw)
+        left = _pairs(bk, pk, JT["left"], **kw)
+        right = _pairs(bk, pk, JT["right"], **kw)
+        full = _pairs(bk, pk, JT["full"], **kw)
+        semi = _pairs(bk, pk, JT["semi"], **kw)
+        anti = _pairs(bk, pk, JT["anti"], **kw)
+
+        probe_fills = [p for p in left if p[0] == -1]
+        build_fills = [p for p in right if p[1] == -1]
+        assert sorted(inner + probe_fills) == left
+        assert sorted(inner + build_fills) == right
+        assert sorted(inner + probe_fills + build_fills) == full
+        # semi ∪ anti = every probe row exactly once... except NULL-key
+        # probe rows, which appear only on the anti side
+        semi_rows = {p for _, p in semi}
+        anti_rows = {p for _, p in anti}
+        assert not (semi_rows & anti_rows)
+        assert semi_rows | anti_rows == set(range(npr))
+        # matched probe rows in semi == distinct probe ids in inner
+        assert semi_rows == {p for _, p in inner}
