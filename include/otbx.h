@@ -60,11 +60,8 @@
  *   OTBX_DIRECT_CAP / OTBX_Q3_FORCE_HASH / OTBX_Q3_HASH_BUDGET
  *                          — Q3 dense-direct vs hash+bloom path selection
  *                            overrides (tests force the fallback)
- *   OTBX_Q9_SPLIT=1        — split filter+probe Q9 pipeline (default with a
- *                            staged q9rec cache is ONE fused pass; the
- *                            split path also serves q9rec == NULL)
- *   OTBX_Q9_FILTER_WAVE=1  — legacy per-wave appender in the split Q9 part
- *                            filter (default split filter is tile-staged)
+ *   OTBX_Q9_FILTER_WAVE=1  — legacy per-wave appender in the Q9 part filter
+ *                            (default is the tile-staged compaction; A/B)
  *   OTBX_Q3_COMPACT_LEGACY=1 — legacy block-chunk Q3 group compaction
  *                            (default is the word-granular bitmap walk)
  *   OTBX_Q3_COMPACT_TILE=1 — quad-granular tile-staged compaction (A/B)

@@ -2268,121 +2268,6 @@ __global__ void k_q9_build_recs(const otbx_lineitem_dev l,
     }
 }
 
-/* FUSED single-pass variant (default when the q9rec cache is staged):
- * streams ONLY l_partkey, tests the part bitmap, and for the ~6%
- * survivors gathers the 32-B record + dtab inline and accumulates the
- * 7-year register partials — no hit-list write+read and one kernel
- * launch. Viable only since the record cache collapsed the survivor
- * payload to one line (the pre-record fused kernel lost because it
- * streamed every payload column; the request budget is additive —
- * DESIGN §7 — and the hit-list round trip is pure overhead on top).
- * Split path kept for q9rec == NULL and OTBX_Q9_SPLIT=1. */
-__global__ void k_q9_fused(const otbx_lineitem_dev l,
-                           const unsigned long long *__restrict__ pbitmap,
-                           int64_t lo_k, int64_t hi_k,
-                           const int32_t *__restrict__ dtab, int64_t mino,
-                           int64_t orange, double *__restrict__ out_sums,
-                           unsigned long long *__restrict__ out_counts)
-{
-    const q9_rec *__restrict__ recs = (const q9_rec *)l.q9rec;
-    double acc[7];
-    uint32_t cnt[7];
-#pragma unroll
-    for (int y = 0; y < 7; y++) {
-        acc[y] = 0.0;
-        cnt[y] = 0;
-    }
-    int64_t nq = l.n / 4;
-    const v2l *pk2 = (const v2l *)l.l_partkey;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t q = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; q < nq;
-         q += stride) {
-        v2l pa = __builtin_nontemporal_load(&pk2[2 * q]);
-        v2l pb = __builtin_nontemporal_load(&pk2[2 * q + 1]);
-        int64_t pks[4] = {pa.x, pa.y, pb.x, pb.y};
-        bool m[4];
-#pragma unroll
-        for (int j = 0; j < 4; j++) {
-            int64_t pidx = pks[j] - 1 - lo_k;
-            m[j] = pidx >= 0 && pidx < hi_k - lo_k &&
-                   ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull);
-        }
-#pragma unroll
-        for (int j = 0; j < 4; j++) {
-            if (!m[j])
-                continue;
-            q9_rec rc = recs[q * 4 + j];
-            int64_t oidx = rc.okey - mino;
-            if (oidx < 0 || oidx >= orange)
-                continue;
-            int32_t date = dtab[oidx];
-            if (date == 0)
-                continue;
-            int32_t y = otbx_year_of_day(date);
-            double rev = rc.price * (1.0 - rc.disc);
-#pragma unroll
-            for (int yy = 0; yy < 7; yy++) {
-                bool hit = yy == y;
-                acc[yy] += hit ? rev : 0.0;
-                cnt[yy] += hit;
-            }
-        }
-    }
-    /* tail rows */
-    if (blockIdx.x == 0 && threadIdx.x == 0) {
-        for (int64_t i = nq * 4; i < l.n; i++) {
-            int64_t pidx = l.l_partkey[i] - 1 - lo_k;
-            if (!(pidx >= 0 && pidx < hi_k - lo_k &&
-                  ((pbitmap[pidx >> 6] >> (pidx & 63)) & 1ull)))
-                continue;
-            q9_rec rc = recs[i];
-            int64_t oidx = rc.okey - mino;
-            if (oidx < 0 || oidx >= orange)
-                continue;
-            int32_t date = dtab[oidx];
-            if (date == 0)
-                continue;
-            int32_t y = otbx_year_of_day(date);
-            double rev = rc.price * (1.0 - rc.disc);
-            acc[y] += rev;
-            cnt[y]++;
-        }
-    }
-    /* wave + block reduction, one atomic per (year) per block */
-#pragma unroll
-    for (int y = 0; y < 7; y++) {
-        for (int off = WAVE / 2; off > 0; off >>= 1) {
-            acc[y] += __shfl_down(acc[y], off, WAVE);
-            cnt[y] += __shfl_down(cnt[y], off, WAVE);
-        }
-    }
-    __shared__ double lacc[256 / WAVE][7];
-    __shared__ uint32_t lcnt[256 / WAVE][7];
-    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
-    if (lane == 0) {
-#pragma unroll
-        for (int y = 0; y < 7; y++) {
-            lacc[wid][y] = acc[y];
-            lcnt[wid][y] = cnt[y];
-        }
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-        int nw = (int)(blockDim.x / WAVE);
-#pragma unroll
-        for (int y = 0; y < 7; y++) {
-            double v = 0;
-            unsigned long long c = 0;
-            for (int w = 0; w < nw; w++) {
-                v += lacc[w][y];
-                c += lcnt[w][y];
-            }
-            if (v != 0.0) atomicAdd(&out_sums[y], v);
-            if (c) atomicAdd(&out_counts[y], c);
-        }
-    }
-}
-
 /* phase 2: dense pass over the survivors — orders date lookup + year
  * partial aggregate in per-lane registers (the Q1 pattern: 7-year domain,
  * compile-time indexed), wave+block reduce, one atomic per (year, block). */
@@ -2590,14 +2475,6 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
             rec0 = true;
         }
         const char *fw = getenv("OTBX_Q9_FILTER_WAVE");
-        const char *fs = getenv("OTBX_Q9_SPLIT");
-        if (l->q9rec && !(fs && atoi(fs)) && !(fw && atoi(fw))) {
-            /* fused single pass (default with the staged record cache) */
-            hipLaunchKernelGGL(k_q9_fused, dim3(grid_for(l->n / 4, 256)),
-                               dim3(256), 0, s, *l, pbitmap, lo, hi, dtab,
-                               mino, orange, sums_dev,
-                               (unsigned long long *)counts_dev);
-        } else {
         if (fw && atoi(fw)) /* legacy per-wave appender (A/B) */
             hipLaunchKernelGGL(k_q9_filter, dim3(grid_for(l->n / 4, 256)),
                                dim3(256), 0, s, l->l_partkey, l->n, pbitmap,
@@ -2609,7 +2486,6 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
         hipLaunchKernelGGL(k_q9_probe, dim3(grid_for(l->n / 8, 256)),
                            dim3(256), 0, s, *l, hits, nhits, dtab, mino,
                            orange, sums_dev, (unsigned long long *)counts_dev);
-        }
     }
     HIP_CHECK(hipGetLastError());
     if (kernel_ms) {
