@@ -183,3 +183,92 @@ def test_q3_partial_matches_numpy_any_size(n, segment, date):
                        rows["revenue"].tolist()))
         for k, v in byk.items():
             assert abs(got[k] - v) <= 1e-9 * max(abs(v), 1e-9)
+
+
+# ---------------- round-2 operators: hypothesis pinning ----------------
+
+@st.composite
+def dec_input(draw):
+    n = draw(st.integers(min_value=0, max_value=60))
+    wild_val = st.one_of(st.integers(min_value=-100, max_value=100),
+                         st.sampled_from([I64_MIN, I64_MIN + 1, 2**63 - 1]))
+    keys = np.array(draw(st.lists(small_key, min_size=n, max_size=n)),
+                    dtype=np.int64)
+    vals = np.array(draw(st.lists(wild_val, min_size=n, max_size=n)),
+                    dtype=np.int64)
+    kn = np.array(draw(st.lists(st.booleans(), min_size=n, max_size=n)),
+                  dtype=np.uint8)
+    vn = np.array(draw(st.lists(st.booleans(), min_size=n, max_size=n)),
+                  dtype=np.uint8)
+    return keys, vals, kn, vn
+
+
+@settings(max_examples=80, deadline=None, derandomize=True)
+@given(dec_input())
+def test_dec_agg_matches_bigint_bruteforce(inp):
+    """Exact int128 decimal agg vs Python's unbounded ints — including
+    INT64_MIN/MAX values whose sums leave the int64 range."""
+    keys, vals, kn, vn = inp
+    got = ora.agg_i64_dec(keys, vals, key_null=kn, val_null=vn)
+    groups = {}
+    for i in range(len(keys)):
+        a = (bool(kn[i]), 0 if kn[i] else int(keys[i]))
+        g = groups.setdefault(a, [0, 0, None])
+        g[0] += 1
+        if not vn[i]:
+            g[1] += 1
+            g[2] = (g[2] or 0) + int(vals[i])
+    assert len(got) == len(groups)
+    for o in got:
+        key = (bool(o.key_isnull), 0 if o.key_isnull else int(o.key))
+        cs, cv, sm = groups[key]
+        assert o.count_star == cs and o.count_v == cv
+        if sm is None:
+            assert o.sum_isnull
+        else:
+            assert o.sum128 == sm
+
+
+@st.composite
+def joinext_input(draw):
+    nb = draw(st.integers(min_value=0, max_value=40))
+    npr = draw(st.integers(min_value=0, max_value=60))
+    bk = np.array(draw(st.lists(wild_key, min_size=nb, max_size=nb)),
+                  dtype=np.int64)
+    pk = np.array(draw(st.lists(wild_key, min_size=npr, max_size=npr)),
+                  dtype=np.int64)
+    bn = np.array(draw(st.lists(st.booleans(), min_size=nb, max_size=nb)),
+                  dtype=np.uint8)
+    pn = np.array(draw(st.lists(st.booleans(), min_size=npr, max_size=npr)),
+                  dtype=np.uint8)
+    jt = draw(st.integers(min_value=0, max_value=5))
+    return bk, pk, bn, pn, jt
+
+
+@settings(max_examples=80, deadline=None, derandomize=True)
+@given(joinext_input())
+def test_join_ext_matches_bruteforce(inp):
+    """All six join types vs brute force under hypothesis — INT64_MIN
+    keys, NULL masks, empty sides, duplicates."""
+    bk, pk, bn, pn, jt = inp
+    bi, pi = ora.join_ext(bk, pk, jt, bnull=bn, pnull=pn)
+    got = sorted(zip(bi.tolist(), pi.tolist()))
+    pairs, matched = [], set()
+    for p in range(len(pk)):
+        if pn[p]:
+            if jt in (1, 3, 5):
+                pairs.append((-1, p))
+            continue
+        ms = [b for b in range(len(bk)) if not bn[b] and bk[b] == pk[p]]
+        matched.update(ms)
+        if jt in (0, 1, 4, 5):
+            pairs.extend((b, p) for b in ms)
+        if jt == 2 and ms:
+            pairs.append((-1, p))
+        if jt in (1, 3, 5) and not ms:
+            pairs.append((-1, p))
+    if jt in (4, 5):
+        for b in range(len(bk)):
+            if bn[b] or b not in matched:
+                pairs.append((b, -1))
+    assert got == sorted(pairs)
