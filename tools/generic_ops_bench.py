@@ -70,6 +70,49 @@ def bench_join(nb, np_):
           f"({np_/best/1e9:6.1f} Gprobes/s), pairs={int(npairs.cpu().item())}")
 
 
+def bench_node(label, make_node, n_probe):
+    """Scale timing through an executor wrapper (generality-tier ops)."""
+    torch.cuda.synchronize()
+    best = 1e9
+    for _ in range(3):
+        t0 = time.time()
+        node = make_node()
+        node.BeginCustomScan()
+        node._run()
+        torch.cuda.synchronize()
+        best = min(best, time.time() - t0)
+    print(f"{label}: {best*1e3:8.2f} ms ({n_probe/best/1e9:6.1f} Grows/s)")
+
+
+def bench_round2_ops():
+    """Generality-tier round-2 operators at scale (correctness-first
+    designs; numbers are evidence-of-function, not roofline targets)."""
+    g = torch.Generator(device="cuda").manual_seed(3)
+    n = 100_000_000
+    k1 = torch.randint(0, 100_000, (n,), dtype=torch.int64, device="cuda",
+                       generator=g)
+    k2 = torch.randint(0, 50, (n,), dtype=torch.int64, device="cuda",
+                       generator=g)
+    v = torch.rand(n, dtype=torch.float64, device="cuda", generator=g)
+    vi = torch.randint(-10**12, 10**12, (n,), dtype=torch.int64,
+                       device="cuda", generator=g)
+    bench_node("agg2  n=100M groups=5M       ",
+               lambda: ex.GpuHashAgg2(k1, k2, v), n)
+    bench_node("dec   n=100M groups=100k     ",
+               lambda: ex.GpuHashAggDec(k1, vi), n)
+    bench_node("aggn4 n=100M groups=5M       ",
+               lambda: ex.GpuHashAggN([k1, k2, k2, k2], v), n)
+    nb = 10_000_000
+    bk = torch.randint(0, nb, (nb,), dtype=torch.int64, device="cuda",
+                       generator=g)
+    pk = torch.randint(0, 2 * nb, (n,), dtype=torch.int64, device="cuda",
+                       generator=g)
+    for jt in ("left", "semi", "anti"):
+        bench_node(f"joinx {jt:<5} nb=10M np=100M  ",
+                   lambda jt=jt: ex.GpuHashJoin(bk, pk, join_type=jt,
+                                                cap_pairs=2 * n), n)
+
+
 if __name__ == "__main__":
     ex.init_device(0)
     bench_agg(600_000_000, 4)
@@ -77,3 +120,5 @@ if __name__ == "__main__":
     bench_agg(600_000_000, 100_000_000)
     bench_join(15_000_000, 600_000_000)
     bench_join(150_000_000, 600_000_000)
+    if "--round2-ops" in sys.argv:
+        bench_round2_ops()
