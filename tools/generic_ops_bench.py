@@ -70,23 +70,14 @@ def bench_join(nb, np_):
           f"({np_/best/1e9:6.1f} Gprobes/s), pairs={int(npairs.cpu().item())}")
 
 
-def bench_node(label, make_node, n_probe):
-    """Scale timing through an executor wrapper (generality-tier ops)."""
-    torch.cuda.synchronize()
-    best = 1e9
-    for _ in range(3):
-        t0 = time.time()
-        node = make_node()
-        node.BeginCustomScan()
-        node._run()
-        torch.cuda.synchronize()
-        best = min(best, time.time() - t0)
-    print(f"{label}: {best*1e3:8.2f} ms ({n_probe/best/1e9:6.1f} Grows/s)")
-
-
 def bench_round2_ops():
-    """Generality-tier round-2 operators at scale (correctness-first
-    designs; numbers are evidence-of-function, not roofline targets)."""
+    """Generality-tier round-2 operators at scale, timed at the C-ABI
+    (device outputs; no host materialization — the executor wrappers
+    additionally convert results to Python objects, which dominates at
+    100 M-pair scale and is not kernel time). Correctness-first designs;
+    numbers are evidence-of-function, not roofline targets."""
+    L = lib()
+    stream = C.c_void_p(torch.cuda.current_stream().cuda_stream)
     g = torch.Generator(device="cuda").manual_seed(3)
     n = 100_000_000
     k1 = torch.randint(0, 100_000, (n,), dtype=torch.int64, device="cuda",
@@ -96,21 +87,68 @@ def bench_round2_ops():
     v = torch.rand(n, dtype=torch.float64, device="cuda", generator=g)
     vi = torch.randint(-10**12, 10**12, (n,), dtype=torch.int64,
                        device="cuda", generator=g)
-    bench_node("agg2  n=100M groups=5M       ",
-               lambda: ex.GpuHashAgg2(k1, k2, v), n)
-    bench_node("dec   n=100M groups=100k     ",
-               lambda: ex.GpuHashAggDec(k1, vi), n)
-    bench_node("aggn4 n=100M groups=5M       ",
-               lambda: ex.GpuHashAggN([k1, k2, k2, k2], v), n)
+
+    def timed(label, fn, units):
+        torch.cuda.synchronize()
+        best = 1e9
+        for _ in range(3):
+            t0 = time.time()
+            fn()
+            torch.cuda.synchronize()
+            best = min(best, time.time() - t0)
+        print(f"{label}: {best*1e3:8.2f} ms ({units/best/1e9:6.1f} Grows/s)")
+
+    ws_bytes = C.c_size_t(0)
+    L.otbx_agg_i64x2_workspace_bytes(C.c_int64(n), C.byref(ws_bytes))
+    ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
+    out = torch.empty(n * 56, dtype=torch.uint8, device="cuda")
+    ng = torch.zeros(1, dtype=torch.int64, device="cuda")
+    timed("agg2  n=100M groups=5M       ", lambda: call(
+        "otbx_agg_i64x2", C.c_void_p(k1.data_ptr()), None,
+        C.c_void_p(k2.data_ptr()), None, C.c_void_p(v.data_ptr()), None,
+        C.c_int64(n), C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+        C.c_void_p(out.data_ptr()), C.c_void_p(ng.data_ptr()), stream), n)
+
+    L.otbx_agg_i64_dec_workspace_bytes(C.c_int64(n), C.byref(ws_bytes))
+    ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
+    timed("dec   n=100M groups=100k     ", lambda: call(
+        "otbx_agg_i64_dec", C.c_void_p(k1.data_ptr()), None,
+        C.c_void_p(vi.data_ptr()), None, C.c_int64(n),
+        C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+        C.c_void_p(out.data_ptr()), C.c_void_p(ng.data_ptr()), stream), n)
+
+    from opentenbase_amd._lib import KeysetDev
+    ks = KeysetDev()
+    ks.nkeys = 4
+    for c, t in enumerate([k1, k2, k2, k2]):
+        ks.keys[c] = t.data_ptr()
+    L.otbx_agg_i64n_workspace_bytes(C.c_int64(n), C.byref(ws_bytes))
+    ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
+    timed("aggn4 n=100M groups=5M       ", lambda: call(
+        "otbx_agg_i64n", C.byref(ks), C.c_void_p(v.data_ptr()), None,
+        C.c_int64(n), C.c_void_p(ws.data_ptr()), C.c_size_t(ws_bytes.value),
+        C.c_void_p(out.data_ptr()), C.c_void_p(ng.data_ptr()), stream), n)
+
     nb = 10_000_000
     bk = torch.randint(0, nb, (nb,), dtype=torch.int64, device="cuda",
                        generator=g)
     pk = torch.randint(0, 2 * nb, (n,), dtype=torch.int64, device="cuda",
                        generator=g)
-    for jt in ("left", "semi", "anti"):
-        bench_node(f"joinx {jt:<5} nb=10M np=100M  ",
-                   lambda jt=jt: ex.GpuHashJoin(bk, pk, join_type=jt,
-                                                cap_pairs=2 * n), n)
+    L.otbx_join_ext_workspace_bytes(C.c_int64(nb), C.c_int64(n),
+                                    C.byref(ws_bytes))
+    ws = torch.empty(ws_bytes.value, dtype=torch.uint8, device="cuda")
+    cap = 2 * n
+    ob = torch.empty(cap, dtype=torch.int64, device="cuda")
+    op = torch.empty(cap, dtype=torch.int64, device="cuda")
+    npairs = torch.zeros(1, dtype=torch.int64, device="cuda")
+    for jt, name in [(1, "left"), (2, "semi"), (3, "anti")]:
+        timed(f"joinx {name:<5} nb=10M np=100M  ", lambda jt=jt: call(
+            "otbx_join_i64_ext", C.c_void_p(bk.data_ptr()), None,
+            C.c_int64(nb), C.c_void_p(pk.data_ptr()), None, C.c_int64(n),
+            C.c_int32(jt), C.c_void_p(ws.data_ptr()),
+            C.c_size_t(ws_bytes.value), C.c_void_p(ob.data_ptr()),
+            C.c_void_p(op.data_ptr()), C.c_int64(cap),
+            C.c_void_p(npairs.data_ptr()), stream), n)
 
 
 if __name__ == "__main__":
