@@ -65,6 +65,9 @@
  *   OTBX_Q3_COMPACT_LEGACY=1 — legacy block-chunk Q3 group compaction
  *                            (default is the word-granular bitmap walk)
  *   OTBX_Q3_COMPACT_TILE=1 — quad-granular tile-staged compaction (A/B)
+ *   OTBX_NK_FORCE_CAP=N    — force a tiny first-attempt table in the
+ *                            generality-tier aggregates (tests the
+ *                            estimator-overflow abort + full-cap rerun)
  */
 #ifndef OTBX_H
 #define OTBX_H

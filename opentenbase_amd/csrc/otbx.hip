@@ -5102,6 +5102,141 @@ extern "C" otbx_status otbx_topk_by_revenue(const otbx_q3_group *groups,
 
 } /* extern "C" */
 
+__device__ __forceinline__ bool d_nk_rownull(const otbx_keyset &ks, int64_t i)
+{
+    for (int c = 0; c < ks.nkeys; c++)
+        if (ks.nulls[c] && ks.nulls[c][i])
+            return true;
+    return false;
+}
+
+__device__ __forceinline__ uint64_t d_nk_hash(const otbx_keyset &ks,
+                                              int64_t i)
+{
+    uint64_t h = 0;
+    for (int c = 0; c < ks.nkeys; c++) {
+        h = (h << 1) | (h >> 63);
+        bool isnull = ks.nulls[c] && ks.nulls[c][i];
+        h ^= isnull ? (0x9e3779b97f4a7c15ull + (uint64_t)c)
+                    : d_hash_i64(ks.keys[c][i]);
+    }
+    return h;
+}
+
+__device__ __forceinline__ bool d_nk_row_eq(const otbx_keyset &ks, int64_t a,
+                                            int64_t b)
+{
+    for (int c = 0; c < ks.nkeys; c++) {
+        bool na = ks.nulls[c] && ks.nulls[c][a];
+        bool nb = ks.nulls[c] && ks.nulls[c][b];
+        if (na != nb) return false;
+        if (!na && ks.keys[c][a] != ks.keys[c][b]) return false;
+    }
+    return true;
+}
+
+__device__ __forceinline__ bool d_nk_match(const otbx_keyset &bks, int64_t b,
+                                           const otbx_keyset &pks, int64_t p)
+{
+    for (int c = 0; c < bks.nkeys; c++)
+        if (bks.keys[c][b] != pks.keys[c][p])
+            return false;
+    return true;
+}
+
+
+/* Distinct-group estimator for the generality-tier aggregates: samples
+ * AGG_SAMPLE rows' COMBINED group hashes (d_nk_hash over the keyset view
+ * of the op's columns — any group-consistent hash estimates group
+ * cardinality) into a small exact table, then the 1-key path's birthday
+ * formula (see otbx_agg_i64). Right-sizing the open-addressing tables is
+ * the same lever that took the 1-key aggregate from 1.9 to 20+ Grows/s:
+ * a full-n table is both probe-collision-heavy and init/compact-heavy. */
+__global__ void k_nk_sample_hash(const otbx_keyset ks, int64_t n,
+                                 int64_t *__restrict__ stab,
+                                 unsigned long long *out2)
+{
+    int64_t stride_n = n > AGG_SAMPLE ? n / AGG_SAMPLE : 1;
+    unsigned long long mynew = 0, myproc = 0;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         j < AGG_SAMPLE; j += stride) {
+        int64_t i = j * stride_n;
+        if (i >= n) break;
+        myproc++;
+        int64_t k = (int64_t)d_nk_hash(ks, i);
+        if (k == AGG_EMPTY) k ^= 1; /* sentinel remap (1-in-2^64 case) */
+        int64_t pos = (int64_t)((uint64_t)k & (uint64_t)(AGG_SCAP - 1));
+        for (;;) {
+            long long cur = __hip_atomic_load(
+                (long long *)&stab[pos], __ATOMIC_RELAXED,
+                __HIP_MEMORY_SCOPE_AGENT);
+            if (cur == k) break;
+            if (cur == AGG_EMPTY) {
+                long long old = atomicCAS((unsigned long long *)&stab[pos],
+                                          (unsigned long long)AGG_EMPTY,
+                                          (unsigned long long)k);
+                if (old == AGG_EMPTY) {
+                    mynew++;
+                    break;
+                }
+                if (old == k) break;
+                continue;
+            }
+            pos = (pos + 1) & (AGG_SCAP - 1);
+        }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        mynew += __shfl_down(mynew, off, WAVE);
+        myproc += __shfl_down(myproc, off, WAVE);
+    }
+    if ((threadIdx.x % WAVE) == 0 && (mynew | myproc)) {
+        atomicAdd(&out2[0], mynew);
+        atomicAdd(&out2[1], myproc);
+    }
+}
+
+/* returns the estimated distinct-group count (<= n); stab is an
+ * AGG_SCAP-entry device scratch region; synchronous (reads the counters
+ * back), matching the 1-key aggregate's estimator step */
+static int64_t nk_estimate_groups(const otbx_keyset &ks, int64_t n,
+                                  int64_t *stab, hipStream_t s)
+{
+    static unsigned long long *d_ds = nullptr;
+    static unsigned long long *h_ds = nullptr;
+    if (!d_ds) {
+        if (otbx_scr_alloc((void **)&d_ds, 16, 0) != hipSuccess)
+            return n;
+        if (otbx_scr_alloc((void **)&h_ds, 16, 1) != hipSuccess)
+            return n;
+    }
+    if (hipMemsetAsync(d_ds, 0, 16, s) != hipSuccess)
+        return n;
+    hipLaunchKernelGGL(k_fill_i64, dim3(grid_for(AGG_SCAP, 256)), dim3(256),
+                       0, s, stab, AGG_SCAP, AGG_EMPTY);
+    hipLaunchKernelGGL(k_nk_sample_hash, dim3(grid_for(AGG_SAMPLE, 256)),
+                       dim3(256), 0, s, ks, n, stab, d_ds);
+    if (hipMemcpyAsync(h_ds, d_ds, 16, hipMemcpyDeviceToHost, s) !=
+            hipSuccess ||
+        hipStreamSynchronize(s) != hipSuccess)
+        return n;
+    double d = (double)h_ds[0], se = (double)h_ds[1];
+    if (se < 1.0) se = 1.0;
+    int64_t est;
+    if (d < se / 2.0) {
+        est = (int64_t)d;
+    } else {
+        double c = se - d;
+        if (c < 1.0) c = 1.0;
+        est = (int64_t)(se * se / (2.0 * c)); /* birthday */
+    }
+    if (est > n) est = n;
+    if (est < 1) est = 1;
+    return est;
+}
+
+#define NK_PROBE_BOUND 128
+
 /* ================= extended joins (otbx_join_i64_ext / _i64x2) ===========
  * The HJ_* fill-state FSM (nodeHashjoin.c:139-144) on the generic open-
  * addressing join: inner/left/semi/anti/right/full, optionally on a
@@ -5358,19 +5493,28 @@ __global__ void k_agg2_build(const int64_t *__restrict__ k1,
                              const uint8_t *__restrict__ n2,
                              const double *__restrict__ vals,
                              const uint8_t *__restrict__ vnull, int64_t n,
-                             agg2_slot *tab, int64_t cap)
+                             agg2_slot *tab, int64_t cap,
+                             unsigned int *abortf)
 {
     int64_t mask = cap - 1;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride) {
+        if (abortf && __hip_atomic_load(abortf, __ATOMIC_RELAXED,
+                                        __HIP_MEMORY_SCOPE_AGENT))
+            return; /* estimator-sized table overflowed: full-cap rerun */
         bool in1 = n1 && n1[i], in2 = n2 && n2[i];
         int64_t a = in1 ? 0 : k1[i], b = in2 ? 0 : k2[i];
         uint64_t h1 = in1 ? 0x9e3779b97f4a7c15ull : d_hash_i64(a);
         uint64_t h2 = in2 ? 0xc2b2ae3d27d4eb4full : d_hash_i64(b);
         uint64_t h = ((h1 << 1) | (h1 >> 63)) ^ h2;
         int64_t s = (int64_t)(h & (uint64_t)mask);
+        int steps = 0;
         for (;;) {
+            if (abortf && ++steps > NK_PROBE_BOUND) {
+                atomicOr(abortf, 1u);
+                break;
+            }
             long long owner = tab[s].idx;
             if (owner < 0) {
                 long long prev = (long long)atomicCAS(
@@ -5384,6 +5528,8 @@ __global__ void k_agg2_build(const int64_t *__restrict__ k1,
                 break;
             s = (s + 1) & mask;
         }
+        if (abortf && steps > NK_PROBE_BOUND)
+            continue; /* row unplaced; results discarded by the rerun */
         atomicAdd(&tab[s].count_star, 1ull);
         if (!(vnull && vnull[i])) {
             atomicAdd(&tab[s].count_v, 1ull);
@@ -5514,7 +5660,10 @@ otbx_status otbx_join_i64x2(const int64_t *bk1, const uint8_t *bn1,
 
 otbx_status otbx_agg_i64x2_workspace_bytes(int64_t n, size_t *bytes)
 {
-    *bytes = (size_t)jx_cap_for(n) * sizeof(agg2_slot);
+    /* full-cap table (the abort-fallback path) + the estimator's sample
+     * table (AGG_SCAP i64 hashes) */
+    *bytes = (size_t)jx_cap_for(n) * sizeof(agg2_slot) +
+             (size_t)AGG_SCAP * 8;
     return OTBX_OK;
 }
 
@@ -5534,16 +5683,54 @@ otbx_status otbx_agg_i64x2(const int64_t *k1, const uint8_t *k1null,
     if (ws_bytes < need)
         return OTBX_ERR_INVALID;
     agg2_slot *tab = (agg2_slot *)ws;
-    HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
-    hipLaunchKernelGGL(k_agg2_init, dim3(grid_for(cap, 256)), dim3(256), 0, s,
-                       tab, cap);
-    if (n > 0)
-        hipLaunchKernelGGL(k_agg2_build, dim3(grid_for(n, 256)), dim3(256), 0,
-                           s, k1, k1null, k2, k2null, vals, val_null, n, tab,
-                           cap);
-    hipLaunchKernelGGL(k_agg2_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
-                       s, tab, cap, k1, k1null, k2, k2null, groups_dev,
-                       ngroups_dev);
+    int64_t *stab = (int64_t *)((char *)ws + (size_t)cap * sizeof(agg2_slot));
+    /* estimator-sized attempt first (1-key recipe): the full-n table is
+     * probe- and init/compact-heavy when groups << rows */
+    int64_t cap_use = cap;
+    if (n >= AGGP_THRESHOLD) {
+        otbx_keyset ks;
+        ks.nkeys = 2;
+        ks.keys[0] = k1; ks.keys[1] = k2;
+        ks.nulls[0] = k1null; ks.nulls[1] = k2null;
+        int64_t est = nk_estimate_groups(ks, n, stab, s);
+        int64_t cs = next_pow2_host(est * 8 < 4096 ? 4096 : est * 8);
+        if (cs < cap) cap_use = cs;
+    }
+    const char *fcap = getenv("OTBX_NK_FORCE_CAP"); /* test hook: force a
+        tiny first-attempt table so the abort+full-rerun path executes */
+    if (fcap) {
+        int64_t v = atoll(fcap);
+        if (v >= 16 && v < cap)
+            cap_use = next_pow2_host(v);
+    }
+    static unsigned int *d_ab2 = nullptr;
+    static unsigned int *h_ab2 = nullptr;
+    if (!d_ab2) {
+        SCR_ALLOC_DEV(d_ab2, 4);
+        SCR_ALLOC_HOST(h_ab2, 4);
+    }
+    for (int attempt = 0; attempt < 2; attempt++) {
+        bool bounded = cap_use < cap;
+        HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
+        HIP_CHECK(hipMemsetAsync(d_ab2, 0, 4, s));
+        hipLaunchKernelGGL(k_agg2_init, dim3(grid_for(cap_use, 256)),
+                           dim3(256), 0, s, tab, cap_use);
+        if (n > 0)
+            hipLaunchKernelGGL(k_agg2_build, dim3(grid_for(n, 256)),
+                               dim3(256), 0, s, k1, k1null, k2, k2null, vals,
+                               val_null, n, tab, cap_use,
+                               bounded ? d_ab2 : NULL);
+        if (!bounded)
+            break;
+        HIP_CHECK(hipMemcpyAsync(h_ab2, d_ab2, 4, hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        if (!*h_ab2)
+            break;
+        cap_use = cap; /* rare-tail overflow: redo against the full table */
+    }
+    hipLaunchKernelGGL(k_agg2_compact, dim3(grid_for(cap_use, 256)),
+                       dim3(256), 0, s, tab, cap_use, k1, k1null, k2, k2null,
+                       groups_dev, ngroups_dev);
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }
@@ -5594,17 +5781,26 @@ __global__ void k_dec_build(const int64_t *__restrict__ keys,
                             const uint8_t *__restrict__ knull,
                             const int64_t *__restrict__ vals,
                             const uint8_t *__restrict__ vnull, int64_t n,
-                            dec_slot *tab, int64_t cap)
+                            dec_slot *tab, int64_t cap,
+                            unsigned int *abortf)
 {
     int64_t mask = cap - 1;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride) {
+        if (abortf && __hip_atomic_load(abortf, __ATOMIC_RELAXED,
+                                        __HIP_MEMORY_SCOPE_AGENT))
+            return; /* estimator-sized table overflowed: full-cap rerun */
         bool kn = knull && knull[i];
         int64_t k = kn ? 0 : keys[i];
         uint64_t h = kn ? 0x9e3779b97f4a7c15ull : d_hash_i64(k);
         int64_t s = (int64_t)(h & (uint64_t)mask);
+        int steps = 0;
         for (;;) {
+            if (abortf && ++steps > NK_PROBE_BOUND) {
+                atomicOr(abortf, 1u);
+                break;
+            }
             long long owner = tab[s].idx;
             if (owner < 0) {
                 long long prev = (long long)atomicCAS(
@@ -5617,6 +5813,8 @@ __global__ void k_dec_build(const int64_t *__restrict__ keys,
                 break;
             s = (s + 1) & mask;
         }
+        if (abortf && steps > NK_PROBE_BOUND)
+            continue; /* row unplaced; results discarded by the rerun */
         atomicAdd(&tab[s].count_star, 1ull);
         if (!(vnull && vnull[i])) {
             int64_t v = vals[i];
@@ -5662,7 +5860,8 @@ extern "C" {
 
 otbx_status otbx_agg_i64_dec_workspace_bytes(int64_t n, size_t *bytes)
 {
-    *bytes = (size_t)jx_cap_for(n) * sizeof(dec_slot);
+    *bytes = (size_t)jx_cap_for(n) * sizeof(dec_slot) +
+             (size_t)AGG_SCAP * 8; /* + estimator sample table */
     return OTBX_OK;
 }
 
@@ -5681,14 +5880,51 @@ otbx_status otbx_agg_i64_dec(const int64_t *keys, const uint8_t *knull,
     if (ws_bytes < need)
         return OTBX_ERR_INVALID;
     dec_slot *tab = (dec_slot *)ws;
-    HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
-    hipLaunchKernelGGL(k_dec_init, dim3(grid_for(cap, 256)), dim3(256), 0, s,
-                       tab, cap);
-    if (n > 0)
-        hipLaunchKernelGGL(k_dec_build, dim3(grid_for(n, 256)), dim3(256), 0,
-                           s, keys, knull, vals, vnull, n, tab, cap);
-    hipLaunchKernelGGL(k_dec_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
-                       s, tab, cap, keys, knull, groups_dev, ngroups_dev);
+    int64_t *stab = (int64_t *)((char *)ws + (size_t)cap * sizeof(dec_slot));
+    int64_t cap_use = cap;
+    if (n >= AGGP_THRESHOLD) {
+        otbx_keyset ks;
+        ks.nkeys = 1;
+        ks.keys[0] = keys;
+        ks.nulls[0] = knull;
+        int64_t est = nk_estimate_groups(ks, n, stab, s);
+        int64_t cs = next_pow2_host(est * 8 < 4096 ? 4096 : est * 8);
+        if (cs < cap) cap_use = cs;
+    }
+    const char *fcap = getenv("OTBX_NK_FORCE_CAP"); /* test hook: force a
+        tiny first-attempt table so the abort+full-rerun path executes */
+    if (fcap) {
+        int64_t v = atoll(fcap);
+        if (v >= 16 && v < cap)
+            cap_use = next_pow2_host(v);
+    }
+    static unsigned int *d_abd = nullptr;
+    static unsigned int *h_abd = nullptr;
+    if (!d_abd) {
+        SCR_ALLOC_DEV(d_abd, 4);
+        SCR_ALLOC_HOST(h_abd, 4);
+    }
+    for (int attempt = 0; attempt < 2; attempt++) {
+        bool bounded = cap_use < cap;
+        HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
+        HIP_CHECK(hipMemsetAsync(d_abd, 0, 4, s));
+        hipLaunchKernelGGL(k_dec_init, dim3(grid_for(cap_use, 256)),
+                           dim3(256), 0, s, tab, cap_use);
+        if (n > 0)
+            hipLaunchKernelGGL(k_dec_build, dim3(grid_for(n, 256)),
+                               dim3(256), 0, s, keys, knull, vals, vnull, n,
+                               tab, cap_use, bounded ? d_abd : NULL);
+        if (!bounded)
+            break;
+        HIP_CHECK(hipMemcpyAsync(h_abd, d_abd, 4, hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        if (!*h_abd)
+            break;
+        cap_use = cap;
+    }
+    hipLaunchKernelGGL(k_dec_compact, dim3(grid_for(cap_use, 256)),
+                       dim3(256), 0, s, tab, cap_use, keys, knull,
+                       groups_dev, ngroups_dev);
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }
@@ -5740,48 +5976,6 @@ otbx_status otbx_build_key32(const int64_t *src, int64_t n, int32_t *dst,
  * Slots claim by defining row index (execGrouping.c firstTuple pattern);
  * identity compares run over the keyset columns. Generality tier. */
 
-__device__ __forceinline__ bool d_nk_rownull(const otbx_keyset &ks, int64_t i)
-{
-    for (int c = 0; c < ks.nkeys; c++)
-        if (ks.nulls[c] && ks.nulls[c][i])
-            return true;
-    return false;
-}
-
-__device__ __forceinline__ uint64_t d_nk_hash(const otbx_keyset &ks,
-                                              int64_t i)
-{
-    uint64_t h = 0;
-    for (int c = 0; c < ks.nkeys; c++) {
-        h = (h << 1) | (h >> 63);
-        bool isnull = ks.nulls[c] && ks.nulls[c][i];
-        h ^= isnull ? (0x9e3779b97f4a7c15ull + (uint64_t)c)
-                    : d_hash_i64(ks.keys[c][i]);
-    }
-    return h;
-}
-
-__device__ __forceinline__ bool d_nk_row_eq(const otbx_keyset &ks, int64_t a,
-                                            int64_t b)
-{
-    for (int c = 0; c < ks.nkeys; c++) {
-        bool na = ks.nulls[c] && ks.nulls[c][a];
-        bool nb = ks.nulls[c] && ks.nulls[c][b];
-        if (na != nb) return false;
-        if (!na && ks.keys[c][a] != ks.keys[c][b]) return false;
-    }
-    return true;
-}
-
-__device__ __forceinline__ bool d_nk_match(const otbx_keyset &bks, int64_t b,
-                                           const otbx_keyset &pks, int64_t p)
-{
-    for (int c = 0; c < bks.nkeys; c++)
-        if (bks.keys[c][b] != pks.keys[c][p])
-            return false;
-    return true;
-}
-
 struct aggn_slot {
     long long idx; /* claim word: -1 empty, else defining row index */
     unsigned long long count_star;
@@ -5804,14 +5998,23 @@ __global__ void k_aggn_init(aggn_slot *tab, int64_t cap)
 __global__ void k_aggn_build(const otbx_keyset ks,
                              const double *__restrict__ vals,
                              const uint8_t *__restrict__ vnull, int64_t n,
-                             aggn_slot *tab, int64_t cap)
+                             aggn_slot *tab, int64_t cap,
+                             unsigned int *abortf)
 {
     int64_t mask = cap - 1;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride) {
+        if (abortf && __hip_atomic_load(abortf, __ATOMIC_RELAXED,
+                                        __HIP_MEMORY_SCOPE_AGENT))
+            return;
         int64_t s = (int64_t)(d_nk_hash(ks, i) & (uint64_t)mask);
+        int steps = 0;
         for (;;) {
+            if (abortf && ++steps > NK_PROBE_BOUND) {
+                atomicOr(abortf, 1u);
+                break;
+            }
             long long owner = tab[s].idx;
             if (owner < 0) {
                 long long prev = (long long)atomicCAS(
@@ -5823,6 +6026,8 @@ __global__ void k_aggn_build(const otbx_keyset ks,
                 break;
             s = (s + 1) & mask;
         }
+        if (abortf && steps > NK_PROBE_BOUND)
+            continue;
         atomicAdd(&tab[s].count_star, 1ull);
         if (!(vnull && vnull[i])) {
             atomicAdd(&tab[s].count_v, 1ull);
@@ -5969,7 +6174,8 @@ static bool keyset_ok(const otbx_keyset *ks)
 
 otbx_status otbx_agg_i64n_workspace_bytes(int64_t n, size_t *bytes)
 {
-    *bytes = (size_t)jx_cap_for(n) * sizeof(aggn_slot);
+    *bytes = (size_t)jx_cap_for(n) * sizeof(aggn_slot) +
+             (size_t)AGG_SCAP * 8; /* + estimator sample table */
     return OTBX_OK;
 }
 
@@ -5987,14 +6193,47 @@ otbx_status otbx_agg_i64n(const otbx_keyset *ks, const double *vals,
     if (ws_bytes < need)
         return OTBX_ERR_INVALID;
     aggn_slot *tab = (aggn_slot *)ws;
-    HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
-    hipLaunchKernelGGL(k_aggn_init, dim3(grid_for(cap, 256)), dim3(256), 0, s,
-                       tab, cap);
-    if (n > 0)
-        hipLaunchKernelGGL(k_aggn_build, dim3(grid_for(n, 256)), dim3(256), 0,
-                           s, *ks, vals, vnull, n, tab, cap);
-    hipLaunchKernelGGL(k_aggn_compact, dim3(grid_for(cap, 256)), dim3(256), 0,
-                       s, tab, cap, groups_dev, ngroups_dev);
+    int64_t *stab = (int64_t *)((char *)ws + (size_t)cap * sizeof(aggn_slot));
+    int64_t cap_use = cap;
+    if (n >= AGGP_THRESHOLD) {
+        int64_t est = nk_estimate_groups(*ks, n, stab, s);
+        int64_t cs = next_pow2_host(est * 8 < 4096 ? 4096 : est * 8);
+        if (cs < cap) cap_use = cs;
+    }
+    const char *fcap = getenv("OTBX_NK_FORCE_CAP"); /* test hook: force a
+        tiny first-attempt table so the abort+full-rerun path executes */
+    if (fcap) {
+        int64_t v = atoll(fcap);
+        if (v >= 16 && v < cap)
+            cap_use = next_pow2_host(v);
+    }
+    static unsigned int *d_abn = nullptr;
+    static unsigned int *h_abn = nullptr;
+    if (!d_abn) {
+        SCR_ALLOC_DEV(d_abn, 4);
+        SCR_ALLOC_HOST(h_abn, 4);
+    }
+    for (int attempt = 0; attempt < 2; attempt++) {
+        bool bounded = cap_use < cap;
+        HIP_CHECK(hipMemsetAsync(ngroups_dev, 0, 8, s));
+        HIP_CHECK(hipMemsetAsync(d_abn, 0, 4, s));
+        hipLaunchKernelGGL(k_aggn_init, dim3(grid_for(cap_use, 256)),
+                           dim3(256), 0, s, tab, cap_use);
+        if (n > 0)
+            hipLaunchKernelGGL(k_aggn_build, dim3(grid_for(n, 256)),
+                               dim3(256), 0, s, *ks, vals, vnull, n, tab,
+                               cap_use, bounded ? d_abn : NULL);
+        if (!bounded)
+            break;
+        HIP_CHECK(hipMemcpyAsync(h_abn, d_abn, 4, hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        if (!*h_abn)
+            break;
+        cap_use = cap;
+    }
+    hipLaunchKernelGGL(k_aggn_compact, dim3(grid_for(cap_use, 256)),
+                       dim3(256), 0, s, tab, cap_use, groups_dev,
+                       ngroups_dev);
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
 }

@@ -1283,3 +1283,40 @@ def test_finish_releases_scratch_and_reinit(ex):
     ex.init_device(0)
     b = work()
     assert a == b
+
+
+def test_nk_agg_estimator_fallback(ex, monkeypatch):
+    """The estimator-sized first attempt vs a forced-overflow run (tiny
+    first table -> abort -> full-cap rerun): identical results. Covers the
+    bounded-probe abort path of agg2/dec/aggn."""
+    g = torch.Generator(device="cuda").manual_seed(13)
+    n = 9_000_000                       # above AGGP_THRESHOLD
+    k1 = torch.randint(0, 100_000, (n,), dtype=torch.int64, device="cuda",
+                       generator=g)
+    k2 = torch.randint(0, 10, (n,), dtype=torch.int64, device="cuda",
+                       generator=g)
+    v = torch.rand(n, dtype=torch.float64, device="cuda", generator=g)
+    vi = torch.randint(-10**9, 10**9, (n,), dtype=torch.int64,
+                       device="cuda", generator=g)
+
+    def run_all():
+        a2 = ex.GpuHashAgg2(k1, k2, v)
+        a2.BeginCustomScan()
+        r2 = [(int(r["key1"]), int(r["key2"]), int(r["count_star"]))
+              for r in a2._run()]
+        dc = ex.GpuHashAggDec(k1, vi)
+        dc.BeginCustomScan()
+        rd = [(int(r["key"]), int(r["count_star"]), r["sum128"])
+              for r in dc._run()]
+        an = ex.GpuHashAggN([k1, k2], v)
+        an.BeginCustomScan()
+        rn = sorted((int(r["count_star"]), int(r["count_v"]))
+                    for r in an._run())
+        return r2, rd, rn
+
+    normal = run_all()
+    monkeypatch.setenv("OTBX_NK_FORCE_CAP", "4096")
+    forced = run_all()
+    assert normal[0] == forced[0]
+    assert normal[1] == forced[1]
+    assert normal[2] == forced[2]
