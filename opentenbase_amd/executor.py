@@ -702,6 +702,26 @@ class GpuHashJoinN(CustomScanState):
                         op[:n].cpu().numpy().tolist()))
 
 
+def dec_avg(sum128, count, extra_scale=0):
+    """Exact AVG finalizer for the decimal aggregate's int128 state
+    (Int128AggState {N, sumX}, numeric.c:5072): returns the scaled-int64
+    average round-half-up away from zero — the reference's numeric
+    rounding (round_var HALF_ADJUST_ROUND, numeric.c:724,1743,7145 via
+    div_var). extra_scale shifts the result scale by 10^extra_scale
+    relative to the input scale (e.g. cents in -> tenth-cents out with
+    extra_scale=1). Python ints are unbounded, so this is exact for any
+    int128 state; the combine phase sums states exactly first. Raises on
+    count == 0 (AVG over no rows is NULL at the SQL level — the caller
+    checks sum_isnull/count first, nodeAgg.c finalize semantics)."""
+    if count <= 0:
+        raise OtbxError(3, "dec_avg: count must be positive (NULL AVG is "
+                           "decided by the caller)")
+    num = sum128 * (10 ** extra_scale)
+    if num >= 0:
+        return (2 * num + count) // (2 * count)
+    return -((2 * (-num) + count) // (2 * count))
+
+
 JOIN_TYPES = {"inner": 0, "left": 1, "semi": 2, "anti": 3, "right": 4,
               "full": 5}
 

@@ -403,3 +403,52 @@ def test_join_type_algebra():
         assert semi_rows | anti_rows == set(range(npr))
         # matched probe rows in semi == distinct probe ids in inner
         assert semi_rows == {p for _, p in inner}
+
+
+# ---------------- decimal AVG finalizer ----------------
+
+def test_dec_avg_half_up():
+    """Exact round-half-up-away-from-zero division (numeric.c round_var
+    HALF_ADJUST_ROUND) vs Fraction, incl. negatives and exact halves."""
+    from fractions import Fraction
+    from opentenbase_amd.executor import dec_avg
+    import pytest as _pt
+
+    cases = [(7, 2), (-7, 2), (5, 2), (-5, 2), (1, 3), (-1, 3),
+             (10**30 + 5, 10), (-(10**30 + 5), 10), (0, 5),
+             (2**100 + 1, 7), (-(2**100) - 1, 7)]
+    for s, c in cases:
+        got = dec_avg(s, c)
+        f = Fraction(s, c)
+        # round half away from zero
+        import math
+        exp = int(math.floor(f + Fraction(1, 2))) if f >= 0 \
+            else -int(math.floor(-f + Fraction(1, 2)))
+        assert got == exp, (s, c, got, exp)
+    # extra scale: cents -> hundredth-cents
+    assert dec_avg(1, 3, extra_scale=2) == 33
+    assert dec_avg(2, 3, extra_scale=2) == 67      # .666.. -> 67
+    assert dec_avg(1, 2, extra_scale=2) == 50
+    assert dec_avg(-1, 2, extra_scale=2) == -50
+    with _pt.raises(Exception):
+        dec_avg(1, 0)
+
+
+def test_dec_avg_roundtrip_with_agg():
+    """AVG through the oracle dec state: exact vs Fraction on the real
+    aggregate output."""
+    from fractions import Fraction
+    from opentenbase_amd.executor import dec_avg
+    rng = np.random.default_rng(123)
+    k = rng.integers(0, 5, 300)
+    v = rng.integers(-10**15, 10**15, 300)
+    for g in ora.agg_i64_dec(k, v):
+        if g.sum_isnull:
+            continue
+        got = dec_avg(g.sum128, g.count_v)
+        f = Fraction(g.sum128, g.count_v)
+        half = Fraction(1, 2)
+        import math
+        exp = int(math.floor(f + half)) if f >= 0 \
+            else -int(math.floor(-f + half))
+        assert got == exp
