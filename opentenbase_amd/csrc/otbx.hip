@@ -5461,6 +5461,65 @@ __global__ void k_joinx_fill_build(const uint8_t *__restrict__ n1,
     }
 }
 
+/* ---- outer joins via the FAST inner join + mark-and-fill ----
+ * For 1-key left/right/full with a large build side, the FSM table above
+ * is dominated by random probes; instead: run otbx_join_i64 (dense-direct
+ * or partitioned — 18-28 Gprobes/s), then mark matched probe/build rows
+ * from the emitted pairs and sweep the unmatched (+ NULL-key) rows as
+ * fills. Exactly the reference's split between the probe loop and
+ * HJ_FILL_* states, just batched. Overflow caveat: if the TRUE pair count
+ * exceeds cap_pairs the marks are incomplete and the fills with them —
+ * the documented contract already requires the caller to re-run on
+ * overflow (include/otbx.h). */
+__global__ void k_joinx_mark_pairs(const int64_t *__restrict__ out_b,
+                                   const int64_t *__restrict__ out_p,
+                                   const int64_t *__restrict__ npairs_p,
+                                   int64_t cap_pairs,
+                                   unsigned long long *__restrict__ bbm,
+                                   unsigned long long *__restrict__ pbm)
+{
+    int64_t n = *npairs_p;
+    if (n > cap_pairs) n = cap_pairs;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        if (bbm) {
+            int64_t b = out_b[i];
+            atomicOr(&bbm[b >> 6], 1ull << (b & 63));
+        }
+        if (pbm) {
+            int64_t p = out_p[i];
+            atomicOr(&pbm[p >> 6], 1ull << (p & 63));
+        }
+    }
+}
+
+__global__ void k_joinx_fill_probe(const uint8_t *__restrict__ n1,
+                                   const uint8_t *__restrict__ n2, int64_t np,
+                                   int nk,
+                                   const unsigned long long *__restrict__ pbm,
+                                   int64_t *__restrict__ out_b,
+                                   int64_t *__restrict__ out_p,
+                                   int64_t cap_pairs, int64_t *npairs)
+{
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i0 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
+         i0 += stride) {
+        bool fill = false;
+        if (i0 < np) {
+            bool rownull = (n1 && n1[i0]) || (nk == 2 && n2 && n2[i0]);
+            fill = rownull || !((pbm[i0 >> 6] >> (i0 & 63)) & 1ull);
+        }
+        int64_t pos = wave_append(npairs, fill);
+        if (fill && pos < cap_pairs) {
+            out_b[pos] = -1;
+            out_p[pos] = i0;
+        }
+        if (__all(i0 >= np))
+            break;
+    }
+}
+
 /* ================= two-key hash aggregate (otbx_agg_i64x2) ===============
  * Group identity (k1_isnull, k1, k2_isnull, k2) with NULL==NULL for
  * grouping (execGrouping.c:295,:525). Open-addressing slots claimed by row
@@ -5579,9 +5638,15 @@ static int64_t jx_cap_for(int64_t nb)
 otbx_status otbx_join_ext_workspace_bytes(int64_t nb, int64_t np,
                                           size_t *bytes)
 {
-    (void)np;
-    *bytes = (size_t)jx_cap_for(nb) * sizeof(joinx_slot) +
-             (size_t)((nb + 63) / 64 + 1) * 8;
+    size_t fsm = (size_t)jx_cap_for(nb) * sizeof(joinx_slot) +
+                 (size_t)((nb + 63) / 64 + 1) * 8;
+    /* the via-inner outer-join route additionally needs the inner join's
+     * workspace plus a matched-probe bitmap */
+    size_t inner;
+    otbx_join_i64_workspace_bytes(nb, np, &inner);
+    size_t via = inner + (size_t)((nb + 63) / 64 + 1) * 8 +
+                 (size_t)((np + 63) / 64 + 1) * 8;
+    *bytes = fsm > via ? fsm : via;
     return OTBX_OK;
 }
 
@@ -5607,6 +5672,58 @@ static otbx_status joinx_run(const int64_t *bk1, const uint8_t *bn1,
     unsigned long long *mbitmap =
         (unsigned long long *)((char *)ws + (size_t)cap * sizeof(joinx_slot));
     const bool fill_build = (jt == 4 || jt == 5);
+
+    /* 1-key left/right/full with a non-trivial build side: fast inner
+     * join + mark-and-fill (see k_joinx_mark_pairs). Threshold overridable
+     * for tests (OTBX_JOINX_VIA_INNER: 1 = always, 0 = never). */
+    if (nk == 1 && (jt == 1 || jt == 4 || jt == 5)) {
+        bool via = nb >= 65536;
+        const char *ov = getenv("OTBX_JOINX_VIA_INNER");
+        if (ov) via = atoi(ov) != 0;
+        if (via) {
+            size_t inner_ws;
+            otbx_join_i64_workspace_bytes(nb, np, &inner_ws);
+            unsigned long long *bbm =
+                (unsigned long long *)((char *)ws + inner_ws);
+            unsigned long long *pbm = bbm + ((nb + 63) / 64 + 1);
+            size_t need_v = inner_ws + (size_t)((nb + 63) / 64 + 1) * 8 +
+                            (size_t)((np + 63) / 64 + 1) * 8;
+            if (ws_bytes < need_v)
+                return OTBX_ERR_INVALID;
+            otbx_status st = otbx_join_i64(bk1, bn1, nb, pk1, pn1, np, ws,
+                                           inner_ws, out_b, out_p, cap_pairs,
+                                           npairs_dev, stream);
+            if (st != OTBX_OK)
+                return st;
+            const bool mark_b = (jt == 4 || jt == 5);
+            const bool mark_p = (jt == 1 || jt == 5);
+            if (mark_b)
+                HIP_CHECK(hipMemsetAsync(bbm, 0,
+                                         (size_t)((nb + 63) / 64 + 1) * 8,
+                                         s));
+            if (mark_p)
+                HIP_CHECK(hipMemsetAsync(pbm, 0,
+                                         (size_t)((np + 63) / 64 + 1) * 8,
+                                         s));
+            hipLaunchKernelGGL(k_joinx_mark_pairs,
+                               dim3(grid_for(np, 256)), dim3(256), 0, s,
+                               out_b, out_p, npairs_dev, cap_pairs,
+                               mark_b ? bbm : NULL, mark_p ? pbm : NULL);
+            if (mark_p && np > 0)
+                hipLaunchKernelGGL(k_joinx_fill_probe,
+                                   dim3(grid_for(np, 256)), dim3(256), 0, s,
+                                   pn1, NULL, np, 1, pbm, out_b, out_p,
+                                   cap_pairs, npairs_dev);
+            if (mark_b && nb > 0)
+                hipLaunchKernelGGL(k_joinx_fill_build,
+                                   dim3(grid_for(nb, 256)), dim3(256), 0, s,
+                                   bn1, NULL, nb, 1, bbm, out_b, out_p,
+                                   cap_pairs, npairs_dev);
+            HIP_CHECK(hipGetLastError());
+            return OTBX_OK;
+        }
+    }
+
     HIP_CHECK(hipMemsetAsync(npairs_dev, 0, 8, s));
     hipLaunchKernelGGL(k_joinx_init, dim3(grid_for(cap, 256)), dim3(256), 0,
                        s, tab, cap);

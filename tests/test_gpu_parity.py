@@ -1320,3 +1320,31 @@ def test_nk_agg_estimator_fallback(ex, monkeypatch):
     assert normal[0] == forced[0]
     assert normal[1] == forced[1]
     assert normal[2] == forced[2]
+
+
+@pytest.mark.parametrize("jt", ["left", "right", "full"])
+def test_join_ext_via_inner_parity(ex, ora, jt, monkeypatch):
+    """The via-inner outer-join route (fast inner join + mark-and-fill)
+    vs the oracle AND vs the FSM-table route on the same inputs."""
+    rng = np.random.default_rng(55)
+    nb, npr = 40000, 120000
+    bk = rng.integers(-10, 25000, nb)
+    pk = rng.integers(-10, 25000, npr)
+    bn = (rng.random(nb) < 0.08).astype(np.uint8)
+    pn = (rng.random(npr) < 0.08).astype(np.uint8)
+    dev = lambda a, dt: torch.as_tensor(a, dtype=dt, device="cuda")  # noqa: E731
+
+    def run():
+        node = ex.GpuHashJoin(dev(bk, torch.int64), dev(pk, torch.int64),
+                              dev(bn, torch.uint8), dev(pn, torch.uint8),
+                              join_type=jt)
+        return sorted(drain(node))
+
+    monkeypatch.setenv("OTBX_JOINX_VIA_INNER", "1")
+    via = run()
+    monkeypatch.setenv("OTBX_JOINX_VIA_INNER", "0")
+    fsm = run()
+    obi, opi = ora.join_ext(bk, pk, ex.JOIN_TYPES[jt], bnull=bn, pnull=pn)
+    exp = sorted(zip(obi.tolist(), opi.tolist()))
+    assert via == exp
+    assert fsm == exp
