@@ -68,10 +68,10 @@
  *   OTBX_NK_FORCE_CAP=N    — force a tiny first-attempt table in the
  *                            generality-tier aggregates (tests the
  *                            estimator-overflow abort + full-cap rerun)
- *   OTBX_JOINX_VIA_INNER=0|1 — force the outer-join route: 1 = always via
- *                            the fast inner join + mark-and-fill, 0 =
- *                            always the FSM table (default: via-inner for
- *                            1-key left/right/full with build >= 64 k)
+ *   OTBX_JOINX_VIA_INNER=1 — route 1-key left/right/full via the inner
+ *                            join + mark-and-fill (measured slower than
+ *                            the FSM table at all tested shapes; kept
+ *                            parity-tested for future shapes)
  */
 #ifndef OTBX_H
 #define OTBX_H

@@ -5677,7 +5677,13 @@ static otbx_status joinx_run(const int64_t *bk1, const uint8_t *bn1,
      * join + mark-and-fill (see k_joinx_mark_pairs). Threshold overridable
      * for tests (OTBX_JOINX_VIA_INNER: 1 = always, 0 = never). */
     if (nk == 1 && (jt == 1 || jt == 4 || jt == 5)) {
-        bool via = nb >= 65536;
+        /* measured OFF by default: at both tested shapes the FSM table
+         * beat this route (nb=15M/np=600M: 55.7 vs 143.9 ms; nb=10M/
+         * np=100M: 16.8 vs 24.3 — profiles/r23 notes). The pair
+         * materialization + mark-pass traffic outweighs the faster inner
+         * probe. Kept behind OTBX_JOINX_VIA_INNER=1 (parity-tested) as
+         * the starting point if a future shape favors it. */
+        bool via = false;
         const char *ov = getenv("OTBX_JOINX_VIA_INNER");
         if (ov) via = atoi(ov) != 0;
         if (via) {
