@@ -5640,13 +5640,20 @@ otbx_status otbx_join_ext_workspace_bytes(int64_t nb, int64_t np,
 {
     size_t fsm = (size_t)jx_cap_for(nb) * sizeof(joinx_slot) +
                  (size_t)((nb + 63) / 64 + 1) * 8;
-    /* the via-inner outer-join route additionally needs the inner join's
-     * workspace plus a matched-probe bitmap */
-    size_t inner;
-    otbx_join_i64_workspace_bytes(nb, np, &inner);
-    size_t via = inner + (size_t)((nb + 63) / 64 + 1) * 8 +
-                 (size_t)((np + 63) / 64 + 1) * 8;
-    *bytes = fsm > via ? fsm : via;
+    *bytes = fsm;
+    /* OTBX_JOINX_VIA_INNER=1 (env-gated, default off) additionally needs
+     * the inner join's workspace plus matched bitmaps; the env affects
+     * this sizing AND the route identically (the OTBX_Q9_BITMAP_BITS
+     * convention) so default callers never over-allocate the ~30x larger
+     * via-route scratch */
+    const char *ov = getenv("OTBX_JOINX_VIA_INNER");
+    if (ov && atoi(ov)) {
+        size_t inner;
+        otbx_join_i64_workspace_bytes(nb, np, &inner);
+        size_t via = inner + (size_t)((nb + 63) / 64 + 1) * 8 +
+                     (size_t)((np + 63) / 64 + 1) * 8;
+        if (via > *bytes) *bytes = via;
+    }
     return OTBX_OK;
 }
 
