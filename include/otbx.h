@@ -147,6 +147,8 @@ typedef struct {
      * TPC-H scales orderkeys do (SF100: max 6e8). Halves the probe's key
      * stream. NULL = kernels read the i64 column. */
     int32_t *l_orderkey32;
+    int32_t *l_partkey32;     /* same compact-key cache for l_partkey
+                               * (halves the Q9 filter's key stream) */
 } otbx_lineitem_dev;
 
 typedef struct {

@@ -36,6 +36,7 @@ class LineitemDev(C.Structure):
         ("l_partkey", C.c_void_p),
         ("q9rec", C.c_void_p),
         ("l_orderkey32", C.c_void_p),
+        ("l_partkey32", C.c_void_p),
     ]
 
 

@@ -2059,9 +2059,15 @@ __global__ void k_q9_odate_build(const otbx_orders_dev o, int64_t mino,
                                  int32_t *dtab)
 {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < o.n;
-         i += stride)
-        dtab[o.o_orderkey[i] - mino] = o.o_orderdate[i];
+    if (o.o_orderkey32) { /* compact-key cache: 12 -> 8 B/row stream */
+        for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+             i < o.n; i += stride)
+            dtab[(int64_t)o.o_orderkey32[i] - mino] = o.o_orderdate[i];
+    } else {
+        for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+             i < o.n; i += stride)
+            dtab[o.o_orderkey[i] - mino] = o.o_orderdate[i];
+    }
 }
 
 /* Q9 probe, split in two phases. The single fused kernel measured as the
@@ -2167,9 +2173,10 @@ __global__ void k_q9_filter(const int64_t *__restrict__ pk, int64_t n,
  * within a tile; tiles land in arbitrary order (result-set parity only —
  * the probe pass is order-free). */
 #define Q9T 8192
+template <bool K32>
 __global__ __launch_bounds__(1024) void k_q9_filter_tile(
-    const int64_t *__restrict__ pk, int64_t n,
-    const unsigned long long *__restrict__ pbitmap, int64_t lo_k,
+    const int64_t *__restrict__ pk, const int32_t *__restrict__ pk32,
+    int64_t n, const unsigned long long *__restrict__ pbitmap, int64_t lo_k,
     int64_t hi_k, uint32_t *__restrict__ hits, int64_t *nhits)
 {
     __shared__ uint32_t stage[Q9T];
@@ -2190,9 +2197,18 @@ __global__ __launch_bounds__(1024) void k_q9_filter_tile(
             bool m[4] = {false, false, false, false};
             int mycnt = 0;
             if (r0 + 3 < n) {
-                v2l pa = __builtin_nontemporal_load(&pk2[r0 / 2]);
-                v2l pb = __builtin_nontemporal_load(&pk2[r0 / 2 + 1]);
-                int64_t pks[4] = {pa.x, pa.y, pb.x, pb.y};
+                int64_t pks[4];
+                if (K32) { /* staged compact-key cache: 4 B/row stream */
+                    v4i pp = __builtin_nontemporal_load(
+                        (const v4i *)&pk32[r0]);
+                    pks[0] = pp.x; pks[1] = pp.y;
+                    pks[2] = pp.z; pks[3] = pp.w;
+                } else {
+                    v2l pa = __builtin_nontemporal_load(&pk2[r0 / 2]);
+                    v2l pb = __builtin_nontemporal_load(&pk2[r0 / 2 + 1]);
+                    pks[0] = pa.x; pks[1] = pa.y;
+                    pks[2] = pb.x; pks[3] = pb.y;
+                }
 #pragma unroll
                 for (int j = 0; j < 4; j++) {
                     int64_t pidx = pks[j] - 1 - lo_k;
@@ -2479,10 +2495,15 @@ otbx_status otbx_q9_partial(const otbx_part_dev *p, const otbx_orders_dev *o,
             hipLaunchKernelGGL(k_q9_filter, dim3(grid_for(l->n / 4, 256)),
                                dim3(256), 0, s, l->l_partkey, l->n, pbitmap,
                                lo, hi, hits, nhits);
-        else
-            hipLaunchKernelGGL(k_q9_filter_tile, dim3(2048), dim3(1024), 0,
-                               s, l->l_partkey, l->n, pbitmap, lo, hi, hits,
+        else if (l->l_partkey32)
+            hipLaunchKernelGGL((k_q9_filter_tile<true>), dim3(2048),
+                               dim3(1024), 0, s, l->l_partkey,
+                               l->l_partkey32, l->n, pbitmap, lo, hi, hits,
                                nhits);
+        else
+            hipLaunchKernelGGL((k_q9_filter_tile<false>), dim3(2048),
+                               dim3(1024), 0, s, l->l_partkey, NULL, l->n,
+                               pbitmap, lo, hi, hits, nhits);
         hipLaunchKernelGGL(k_q9_probe, dim3(grid_for(l->n / 8, 256)),
                            dim3(256), 0, s, *l, hits, nhits, dtab, mino,
                            orange, sums_dev, (unsigned long long *)counts_dev);

@@ -113,6 +113,11 @@ class GpuLineitem:
             if self._okey32 is not None:
                 self.cstruct.l_orderkey32 = C.c_void_p(
                     self._okey32.data_ptr())
+        if self.t.get("l_partkey") is not None:
+            self._pkey32 = _build_key32(self.t["l_partkey"])
+            if self._pkey32 is not None:
+                self.cstruct.l_partkey32 = C.c_void_p(
+                    self._pkey32.data_ptr())
         if self.t.get("l_partkey") is None or self.t.get("l_orderkey") is None:
             return
         self._q9rec = torch.empty(self.n * 32, dtype=torch.uint8,

@@ -1117,6 +1117,7 @@ def test_q9_rec_vs_columnar_selfconsistent(ex):
 
     s_rec, c_rec = run()
     li.cstruct.q9rec = CT.c_void_p(0)      # force the columnar fallback
+    li.cstruct.l_partkey32 = CT.c_void_p(0)  # ... and the i64 key stream
     s_col, c_col = run()
     assert np.array_equal(c_rec, c_col)
     for a, b in zip(s_rec, s_col):
