@@ -71,7 +71,9 @@
  *   OTBX_JOINX_VIA_INNER=1 — route 1-key left/right/full via the inner
  *                            join + mark-and-fill (measured slower than
  *                            the FSM table at all tested shapes; kept
- *                            parity-tested for future shapes)
+ *                            parity-tested for future shapes). Affects
+ *                            otbx_join_ext_workspace_bytes AND the route
+ *                            identically (query sizing under the same env)
  */
 #ifndef OTBX_H
 #define OTBX_H
