@@ -129,6 +129,10 @@ def main():
 
     import torch
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.gpus != world:
+        log(f"note: --gpus {args.gpus} but WORLD_SIZE={world} — rank count "
+            "comes from the torchrun environment (contract); running "
+            f"{world} rank(s)")
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     if world > 1:
