@@ -4521,16 +4521,83 @@ __global__ void k_q3_compact(const q3g_slot *gtab, int64_t gcap,
 }
 
 /* compact filtered customer keys for the broadcast build side */
-__global__ void k_filter_customer(const otbx_customer_dev c, uint8_t want,
-                                  int64_t *out_keys, int64_t *nkeys)
+/* tile-staged compaction (the append_ab v6 pattern, as k_q9_filter_tile):
+ * the original bare wave_append ran at 48 GB/s — LINEAR in n, i.e. the
+ * measured ~88-reservations/us single-counter cap, not a fixed cost
+ * (profiles/r29 diagnosis) — which would put ~2.6 ms/step on every rank
+ * of the multi-GPU Q3 broadcast path at SF100. */
+#define FCT 8192
+__global__ __launch_bounds__(1024) void k_filter_customer(
+    const otbx_customer_dev c, uint8_t want, int64_t *__restrict__ out_keys,
+    int64_t *nkeys)
 {
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < c.n;
-         i += stride) {
-        bool m = c.c_mktsegment[i] == want;
-        int64_t pos = wave_append(nkeys, m);
-        if (m)
-            out_keys[pos] = c.c_custkey[i];
+    __shared__ int64_t stage[FCT];
+    __shared__ int wtot[16];
+    __shared__ int woff[16];
+    __shared__ int sweepbase;
+    __shared__ long long gbase;
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int64_t n = c.n;
+    int64_t ntiles = (n + FCT - 1) / FCT;
+    for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+        int64_t tl = t * (int64_t)FCT;
+        int64_t th = tl + FCT < n ? tl + FCT : n;
+        if (threadIdx.x == 0) sweepbase = 0;
+        __syncthreads();
+        for (int64_t s0 = tl; s0 < th; s0 += 4096) {
+            int64_t r0 = s0 + 4 * (int64_t)threadIdx.x;
+            bool m[4] = {false, false, false, false};
+            int64_t ks[4];
+            int mycnt = 0;
+            if (r0 + 3 < n) {
+                uchar4 sv = *(const uchar4 *)&c.c_mktsegment[r0];
+                v2l ka = *(const v2l *)&c.c_custkey[r0];
+                v2l kb = *(const v2l *)&c.c_custkey[r0 + 2];
+                uint8_t ss[4] = {sv.x, sv.y, sv.z, sv.w};
+                ks[0] = ka.x; ks[1] = ka.y; ks[2] = kb.x; ks[3] = kb.y;
+#pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    m[j] = ss[j] == want;
+                    mycnt += m[j];
+                }
+            } else {
+                for (int j = 0; j < 4 && r0 + j < n; j++) {
+                    m[j] = c.c_mktsegment[r0 + j] == want;
+                    ks[j] = c.c_custkey[r0 + j];
+                    mycnt += m[j];
+                }
+            }
+            int incl = mycnt;
+            for (int off = 1; off < WAVE; off <<= 1) {
+                int up = __shfl_up(incl, off, WAVE);
+                if (lane >= off) incl += up;
+            }
+            if (lane == WAVE - 1) wtot[wid] = incl;
+            __syncthreads();
+            if (threadIdx.x == 0) {
+                int acc = sweepbase;
+                for (int w = 0; w < 16; w++) {
+                    woff[w] = acc;
+                    acc += wtot[w];
+                }
+                sweepbase = acc;
+            }
+            __syncthreads();
+            int pos = woff[wid] + incl - mycnt;
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                if (m[j]) stage[pos++] = ks[j];
+            __syncthreads();
+        }
+        int tot = sweepbase;
+        if (threadIdx.x == 0)
+            gbase = tot ? (long long)atomicAdd((unsigned long long *)nkeys,
+                                               (unsigned long long)tot)
+                        : 0;
+        __syncthreads();
+        for (int p = threadIdx.x; p < tot; p += blockDim.x)
+            out_keys[gbase + p] = stage[p];
+        __syncthreads();
     }
 }
 
@@ -4673,7 +4740,7 @@ otbx_status otbx_filter_customer(const otbx_customer_dev *c, uint8_t segment,
 {
     hipStream_t s = (hipStream_t)stream;
     HIP_CHECK(hipMemsetAsync(nkeys_dev, 0, sizeof(int64_t), s));
-    hipLaunchKernelGGL(k_filter_customer, dim3(grid_for(c->n, 256)), dim3(256),
+    hipLaunchKernelGGL(k_filter_customer, dim3(2048), dim3(1024),
                        0, s, *c, segment, keys_out_dev, nkeys_dev);
     HIP_CHECK(hipGetLastError());
     return OTBX_OK;
