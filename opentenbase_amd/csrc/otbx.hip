@@ -5693,26 +5693,56 @@ __global__ void k_agg2_compact(const agg2_slot *__restrict__ tab, int64_t cap,
                                otbx_agg2_group *__restrict__ out,
                                int64_t *ngroups)
 {
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
-         s += stride) {
-        bool used = s < cap && tab[s].idx >= 0;
-        int64_t pos = wave_append(ngroups, used);
+    /* block-aggregated two-phase emission — ONE global reservation per
+     * block (the k_agg_compact recipe; a per-wave wave_append over a
+     * large slot scan serializes on the single group counter). */
+    __shared__ unsigned long long lbase;
+    __shared__ unsigned int lcnt, ltot;
+    if (threadIdx.x == 0) lcnt = 0;
+    __syncthreads();
+    int64_t per_block = (cap + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < cap ? lo + per_block : cap;
+    int lane = (int)(threadIdx.x % WAVE);
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        unsigned long long mask = __ballot(tab[i].idx >= 0);
+        if (lane == 0 && mask)
+            atomicAdd(&lcnt, (unsigned int)__popcll(mask));
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        ltot = lcnt;
+        lbase = lcnt ? (unsigned long long)atomicAdd(
+                           (unsigned long long *)ngroups,
+                           (unsigned long long)lcnt)
+                     : 0;
+        lcnt = 0;
+    }
+    __syncthreads();
+    if (ltot == 0) return;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        bool used = tab[i].idx >= 0;
+        unsigned long long mask = __ballot(used);
+        if (!mask) continue;
+        unsigned int wbase = 0;
+        if (lane == 0)
+            wbase = atomicAdd(&lcnt, (unsigned int)__popcll(mask));
+        wbase = (unsigned int)__shfl((int)wbase, 0, WAVE);
         if (used) {
-            long long i = tab[s].idx;
-            bool in1 = n1 && n1[i], in2 = n2 && n2[i];
-            out[pos].key1 = in1 ? 0 : k1[i];
-            out[pos].key2 = in2 ? 0 : k2[i];
-            out[pos].count_star = (int64_t)tab[s].count_star;
-            out[pos].count_v = (int64_t)tab[s].count_v;
-            out[pos].sum_v = tab[s].sum;
+            int64_t pos = (int64_t)lbase + wbase +
+                          __popcll(mask & ((1ull << lane) - 1ull));
+            long long r = tab[i].idx;
+            bool in1 = n1 && n1[r], in2 = n2 && n2[r];
+            out[pos].key1 = in1 ? 0 : k1[r];
+            out[pos].key2 = in2 ? 0 : k2[r];
+            out[pos].count_star = (int64_t)tab[i].count_star;
+            out[pos].count_v = (int64_t)tab[i].count_v;
+            out[pos].sum_v = tab[i].sum;
             out[pos].key1_isnull = in1;
             out[pos].key2_isnull = in2;
-            out[pos].sum_isnull = tab[s].count_v == 0;
+            out[pos].sum_isnull = tab[i].count_v == 0;
             out[pos]._pad = 0;
         }
-        if (__all(s >= cap))
-            break;
     }
 }
 
@@ -6053,24 +6083,54 @@ __global__ void k_dec_compact(const dec_slot *__restrict__ tab, int64_t cap,
                               otbx_dec_group *__restrict__ out,
                               int64_t *ngroups)
 {
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
-         s += stride) {
-        bool used = s < cap && tab[s].idx >= 0;
-        int64_t pos = wave_append(ngroups, used);
+    /* block-aggregated two-phase emission — ONE global reservation per
+     * block (the k_agg_compact recipe; a per-wave wave_append over a
+     * large slot scan serializes on the single group counter). */
+    __shared__ unsigned long long lbase;
+    __shared__ unsigned int lcnt, ltot;
+    if (threadIdx.x == 0) lcnt = 0;
+    __syncthreads();
+    int64_t per_block = (cap + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < cap ? lo + per_block : cap;
+    int lane = (int)(threadIdx.x % WAVE);
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        unsigned long long mask = __ballot(tab[i].idx >= 0);
+        if (lane == 0 && mask)
+            atomicAdd(&lcnt, (unsigned int)__popcll(mask));
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        ltot = lcnt;
+        lbase = lcnt ? (unsigned long long)atomicAdd(
+                           (unsigned long long *)ngroups,
+                           (unsigned long long)lcnt)
+                     : 0;
+        lcnt = 0;
+    }
+    __syncthreads();
+    if (ltot == 0) return;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        bool used = tab[i].idx >= 0;
+        unsigned long long mask = __ballot(used);
+        if (!mask) continue;
+        unsigned int wbase = 0;
+        if (lane == 0)
+            wbase = atomicAdd(&lcnt, (unsigned int)__popcll(mask));
+        wbase = (unsigned int)__shfl((int)wbase, 0, WAVE);
         if (used) {
-            long long i = tab[s].idx;
-            bool kn = knull && knull[i];
-            out[pos].key = kn ? 0 : keys[i];
-            out[pos].count_star = (int64_t)tab[s].count_star;
-            out[pos].count_v = (int64_t)tab[s].count_v;
-            out[pos].sum_hi = (int64_t)tab[s].sum_hi;
-            out[pos].sum_lo = tab[s].sum_lo;
+            int64_t pos = (int64_t)lbase + wbase +
+                          __popcll(mask & ((1ull << lane) - 1ull));
+            long long r = tab[i].idx;
+            bool kn = knull && knull[r];
+            out[pos].key = kn ? 0 : keys[r];
+            out[pos].count_star = (int64_t)tab[i].count_star;
+            out[pos].count_v = (int64_t)tab[i].count_v;
+            out[pos].sum_hi = (int64_t)tab[i].sum_hi;
+            out[pos].sum_lo = tab[i].sum_lo;
             out[pos].key_isnull = kn;
-            out[pos].sum_isnull = tab[s].count_v == 0;
+            out[pos].sum_isnull = tab[i].count_v == 0;
         }
-        if (__all(s >= cap))
-            break;
     }
 }
 
@@ -6258,21 +6318,51 @@ __global__ void k_aggn_compact(const aggn_slot *__restrict__ tab, int64_t cap,
                                otbx_aggn_group *__restrict__ out,
                                int64_t *ngroups)
 {
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t s = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
-         s += stride) {
-        bool used = s < cap && tab[s].idx >= 0;
-        int64_t pos = wave_append(ngroups, used);
+    /* block-aggregated two-phase emission — ONE global reservation per
+     * block (the k_agg_compact recipe; a per-wave wave_append over a
+     * large slot scan serializes on the single group counter). */
+    __shared__ unsigned long long lbase;
+    __shared__ unsigned int lcnt, ltot;
+    if (threadIdx.x == 0) lcnt = 0;
+    __syncthreads();
+    int64_t per_block = (cap + gridDim.x - 1) / gridDim.x;
+    int64_t lo = blockIdx.x * per_block;
+    int64_t hi = lo + per_block < cap ? lo + per_block : cap;
+    int lane = (int)(threadIdx.x % WAVE);
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        unsigned long long mask = __ballot(tab[i].idx >= 0);
+        if (lane == 0 && mask)
+            atomicAdd(&lcnt, (unsigned int)__popcll(mask));
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        ltot = lcnt;
+        lbase = lcnt ? (unsigned long long)atomicAdd(
+                           (unsigned long long *)ngroups,
+                           (unsigned long long)lcnt)
+                     : 0;
+        lcnt = 0;
+    }
+    __syncthreads();
+    if (ltot == 0) return;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        bool used = tab[i].idx >= 0;
+        unsigned long long mask = __ballot(used);
+        if (!mask) continue;
+        unsigned int wbase = 0;
+        if (lane == 0)
+            wbase = atomicAdd(&lcnt, (unsigned int)__popcll(mask));
+        wbase = (unsigned int)__shfl((int)wbase, 0, WAVE);
         if (used) {
-            out[pos].row_idx = tab[s].idx;
-            out[pos].count_star = (int64_t)tab[s].count_star;
-            out[pos].count_v = (int64_t)tab[s].count_v;
-            out[pos].sum_v = tab[s].sum;
-            out[pos].sum_isnull = tab[s].count_v == 0;
+            int64_t pos = (int64_t)lbase + wbase +
+                          __popcll(mask & ((1ull << lane) - 1ull));
+            out[pos].row_idx = tab[i].idx;
+            out[pos].count_star = (int64_t)tab[i].count_star;
+            out[pos].count_v = (int64_t)tab[i].count_v;
+            out[pos].sum_v = tab[i].sum;
+            out[pos].sum_isnull = tab[i].count_v == 0;
             out[pos]._pad = 0;
         }
-        if (__all(s >= cap))
-            break;
     }
 }
 
