@@ -6393,9 +6393,17 @@ __global__ void k_joinn_probe(const otbx_keyset bks,
                               int64_t *__restrict__ out_p, int64_t cap_pairs,
                               int64_t *npairs)
 {
+    /* per-wave LDS pair buffer (the k_joinx_probe form): one global
+     * reservation per ~BUF pairs instead of one per walk step — the
+     * per-step wave_append serialized on the single pair counter. */
     const bool emit_match = (jt == 0 || jt == 1 || jt == 4 || jt == 5);
     const bool fill_probe = (jt == 1 || jt == 3 || jt == 5);
     const bool semi = jt == 2;
+    const int BUF = 512;
+    __shared__ int64_t bufb[256 / WAVE][BUF];
+    __shared__ int64_t bufp[256 / WAVE][BUF];
+    int wid = (int)(threadIdx.x / WAVE), lane = (int)(threadIdx.x % WAVE);
+    int nbuf = 0; /* wave-uniform */
     int64_t mask = cap - 1;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i0 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;;
@@ -6420,10 +6428,31 @@ __global__ void k_joinn_probe(const otbx_keyset bks,
                     match = false;
             }
             bool emit = match && emit_match;
-            int64_t pos = wave_append(npairs, emit);
-            if (emit && pos < cap_pairs) {
-                out_b[pos] = bidx;
-                out_p[pos] = i0;
+            unsigned long long mmask = __ballot(emit);
+            int cnt = __popcll(mmask);
+            if (cnt) {
+                if (nbuf + cnt > BUF) {
+                    long long base = 0;
+                    if (lane == 0)
+                        base = (long long)atomicAdd(
+                            (unsigned long long *)npairs,
+                            (unsigned long long)nbuf);
+                    base = __shfl(base, 0, WAVE);
+                    for (int j = lane; j < nbuf; j += WAVE) {
+                        int64_t pos = base + j;
+                        if (pos < cap_pairs) {
+                            out_b[pos] = bufb[wid][j];
+                            out_p[pos] = bufp[wid][j];
+                        }
+                    }
+                    nbuf = 0;
+                }
+                if (emit) {
+                    int rank = __popcll(mmask & ((1ull << lane) - 1ull));
+                    bufb[wid][nbuf + rank] = bidx;
+                    bufp[wid][nbuf + rank] = i0;
+                }
+                nbuf += cnt;
             }
             s = (s + 1) & mask;
             walking = have;
@@ -6435,13 +6464,47 @@ __global__ void k_joinn_probe(const otbx_keyset bks,
             else if (semi)
                 fill = nmatch > 0;
         }
-        int64_t pos = wave_append(npairs, fill);
-        if (fill && pos < cap_pairs) {
-            out_b[pos] = -1;
-            out_p[pos] = i0;
+        unsigned long long fmask = __ballot(fill);
+        int fcnt = __popcll(fmask);
+        if (fcnt) {
+            if (nbuf + fcnt > BUF) {
+                long long base = 0;
+                if (lane == 0)
+                    base = (long long)atomicAdd((unsigned long long *)npairs,
+                                                (unsigned long long)nbuf);
+                base = __shfl(base, 0, WAVE);
+                for (int j = lane; j < nbuf; j += WAVE) {
+                    int64_t pos = base + j;
+                    if (pos < cap_pairs) {
+                        out_b[pos] = bufb[wid][j];
+                        out_p[pos] = bufp[wid][j];
+                    }
+                }
+                nbuf = 0;
+            }
+            if (fill) {
+                int rank = __popcll(fmask & ((1ull << lane) - 1ull));
+                bufb[wid][nbuf + rank] = -1;
+                bufp[wid][nbuf + rank] = i0;
+            }
+            nbuf += fcnt;
         }
         if (__all(i0 >= np))
             break;
+    }
+    if (nbuf) {
+        long long base = 0;
+        if (lane == 0)
+            base = (long long)atomicAdd((unsigned long long *)npairs,
+                                        (unsigned long long)nbuf);
+        base = __shfl(base, 0, WAVE);
+        for (int j = lane; j < nbuf; j += WAVE) {
+            int64_t pos = base + j;
+            if (pos < cap_pairs) {
+                out_b[pos] = bufb[wid][j];
+                out_p[pos] = bufp[wid][j];
+            }
+        }
     }
 }
 
