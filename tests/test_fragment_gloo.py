@@ -353,3 +353,47 @@ def _worker_dist_leftjoin(rank, world, port, q):
 def test_distributed_left_join_two_ranks():
     got, exp = _run(_worker_dist_leftjoin)
     assert got == exp
+
+
+def _worker_dec_combine(rank, world, port, q):
+    """Decimal partial states combine EXACTLY across ranks: each rank
+    aggregates its shard with the int128 oracle, states are all-gathered
+    (as the CN merge ships them) and summed with unbounded ints; result
+    must equal the whole-table aggregate bit-exactly (the numeric
+    combine-phase analog: int128 state addition cannot lose precision)."""
+    from opentenbase_amd import fragment
+    from oracle import oracle_py as ora
+    _init(rank, world, port)
+    rng = np.random.default_rng(7)
+    n = 4000
+    keys = rng.integers(0, 6, n)
+    vals = rng.integers(-2**62, 2**62, n)     # sums leave int64 quickly
+    mine = np.arange(n) % world == rank
+    local = ora.agg_i64_dec(keys[mine], vals[mine])
+    # ship (key, count_star, count_v, sum_hi, sum_lo) rows
+    rows = np.array([(g.key, g.count_star, g.count_v, g.sum_hi,
+                      np.int64(g.sum_lo - 2**63)) for g in local],
+                    dtype=np.int64).reshape(-1)
+    gathered = fragment.allgather_variable(torch.as_tensor(rows))
+    if rank == 0:
+        allrows = gathered.numpy().reshape(-1, 5)
+        comb = {}
+        for k, cs, cv, hi, lo_biased in allrows.tolist():
+            s128 = (int(hi) << 64) | (int(lo_biased) + 2**63)
+            c = comb.setdefault(int(k), [0, 0, 0])
+            c[0] += int(cs)
+            c[1] += int(cv)
+            c[2] += s128
+        full = ora.agg_i64_dec(keys, vals)
+        ok = len(full) == len(comb)
+        for g in full:
+            cs, cv, s = comb[int(g.key)]
+            ok = ok and cs == g.count_star and cv == g.count_v \
+                and s == g.sum128
+        q.put(ok)
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_distributed_dec_combine_exact():
+    assert _run(_worker_dec_combine)
