@@ -55,18 +55,34 @@ def _worker_topk(rank, world, port, q):
     torch.distributed.destroy_process_group()
 
 
-def _run(worker):
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    port = 29511 + np.random.randint(0, 400)
-    procs = [ctx.Process(target=worker, args=(r, 2, port, q)) for r in range(2)]
-    for p in procs:
-        p.start()
-    res = q.get(timeout=120)
-    for p in procs:
-        p.join(timeout=60)
-        assert p.exitcode == 0
-    return res
+def _run(worker, world=2):
+    # retry with fresh ports: a lingering TIME_WAIT on a reused MASTER_PORT
+    # can fail the rendezvous (rare flake under suite load)
+    last = None
+    for attempt in range(3):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        port = 29511 + np.random.randint(0, 2000)
+        procs = [ctx.Process(target=worker, args=(r, world, port, q))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        try:
+            res = q.get(timeout=120)
+        except Exception as e:
+            last = e
+            for p in procs:
+                p.terminate()
+                p.join(timeout=30)
+            continue
+        ok = True
+        for p in procs:
+            p.join(timeout=60)
+            ok = ok and p.exitcode == 0
+        if ok:
+            return res
+        last = AssertionError([p.exitcode for p in procs])
+    raise last
 
 
 @pytest.mark.timeout(180)
@@ -133,18 +149,7 @@ def _worker_a2a(rank, world, port, q):
 
 
 def _run_n(worker, world):
-    ctx = mp.get_context("spawn")
-    q = ctx.Queue()
-    port = 29911 + np.random.randint(0, 400)
-    procs = [ctx.Process(target=worker, args=(r, world, port, q))
-             for r in range(world)]
-    for p in procs:
-        p.start()
-    res = q.get(timeout=180)
-    for p in procs:
-        p.join(timeout=60)
-        assert p.exitcode == 0
-    return res
+    return _run(worker, world=world)
 
 
 def _worker_q1_w4(rank, world, port, q):
