@@ -4861,23 +4861,12 @@ otbx_status otbx_q3_partial(const otbx_customer_dev *c, const otbx_orders_dev *o
                                (unsigned long long *)&hdr[6],
                                (unsigned long long *)&hdr[7]);
     } else {
-        /* 1024-thread blocks: the sibling tile kernel moves the same two
-         * columns 6x faster at this shape; A/B via OTBX_CUSTSEG_256=1 */
-        {
-            const char *c256 = getenv("OTBX_CUSTSEG_256");
-            if (c256 && atoi(c256))
-                hipLaunchKernelGGL(k_count_customer_seg,
-                                   dim3(grid_for(c->n, 256)), dim3(256), 0, s,
-                                   *c, segment, &hdr[0],
-                                   (unsigned long long *)&hdr[6],
-                                   (unsigned long long *)&hdr[7]);
-            else
-                hipLaunchKernelGGL(k_count_customer_seg,
-                                   dim3(grid_for(c->n, 4096)), dim3(1024), 0,
-                                   s, *c, segment, &hdr[0],
-                                   (unsigned long long *)&hdr[6],
-                                   (unsigned long long *)&hdr[7]);
-        }
+        /* 256-thread launch (measured: a 1024-thread variant tripled the
+         * phase — r32 probe A/B; the ~0.3 ms residual stays a round-3 item) */
+        hipLaunchKernelGGL(k_count_customer_seg, dim3(grid_for(c->n, 256)),
+                           dim3(256), 0, s, *c, segment, &hdr[0],
+                           (unsigned long long *)&hdr[6],
+                           (unsigned long long *)&hdr[7]);
     }
     /* UNFILTERED orderkey range (staged zone-map metadata when available,
      * else a minmax kernel read back with the same sync): decides the
